@@ -1,0 +1,96 @@
+"""Execution engine (parity with src/engine/ — ThreadedEnginePerDevice /
+NaiveEngine and include/mxnet/engine.h).
+
+MI355X-native design: the reference's engine exists to order CUDA work from
+many frontend threads by variable dependencies.  On this stack, ordering is
+the HIP stream model itself — ops are stream-ordered by the HIP runtime, and
+the engine's job reduces to (a) stream management per device (compute /
+H2D / D2H copy / comm streams, so copies and RCCL collectives overlap
+compute — reference: MXNET_GPU_COPY_NTHREADS workers + FnProperty lanes),
+(b) sync points (WaitForVar == tensor-ready event, WaitForAll == device
+sync), and (c) a serial debug oracle (NaiveEngine semantics:
+MXA_ENGINE_TYPE=NaiveEngine syncs after every op for deterministic
+debugging, reference env MXNET_ENGINE_TYPE).
+
+The native C++ side (src/engine.cpp) owns the raw hipStream_t pool; Python
+wraps them as torch ExternalStreams so torch ops run on engine streams.
+"""
+from __future__ import annotations
+
+import os
+import torch
+
+__all__ = ["Engine", "get", "naive_mode", "wait_all", "copy_stream",
+           "comm_stream", "compute_stream"]
+
+
+class Engine:
+    _inst = None
+
+    def __init__(self):
+        self.engine_type = os.environ.get("MXA_ENGINE_TYPE",
+                                          os.environ.get("MXNET_ENGINE_TYPE",
+                                                         "StreamEngine"))
+        self._copy_streams = {}
+        self._comm_streams = {}
+        self._native = None
+        if torch.cuda.is_available():
+            from .base import lib
+            self._native = lib()
+
+    @property
+    def is_naive(self) -> bool:
+        return self.engine_type == "NaiveEngine"
+
+    def _get_stream(self, cache, device):
+        if device not in cache:
+            if self._native is not None:
+                handle = self._native.engine_new_stream(device)
+                cache[device] = torch.cuda.ExternalStream(handle, device=device)
+            else:
+                cache[device] = torch.cuda.Stream(device=device)
+        return cache[device]
+
+    def copy_stream(self, device: int):
+        """Dedicated H2D/D2H stream (reference: kCopyFromGPU/kCopyToGPU lanes)."""
+        return self._get_stream(self._copy_streams, device)
+
+    def comm_stream(self, device: int):
+        """Dedicated RCCL stream so collectives overlap backward compute."""
+        return self._get_stream(self._comm_streams, device)
+
+    def wait_all(self):
+        if torch.cuda.is_available():
+            for i in range(torch.cuda.device_count()):
+                torch.cuda.synchronize(i)
+
+    def after_op(self):
+        """NaiveEngine oracle: serialize after every op."""
+        if self.is_naive and torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+
+def get() -> Engine:
+    if Engine._inst is None:
+        Engine._inst = Engine()
+    return Engine._inst
+
+
+def naive_mode() -> bool:
+    return get().is_naive
+
+
+def wait_all():
+    get().wait_all()
+
+
+def copy_stream(device=0):
+    return get().copy_stream(device)
+
+
+def comm_stream(device=0):
+    return get().comm_stream(device)
+
+
+def compute_stream(device=0):
+    return torch.cuda.current_stream(device)

@@ -1,0 +1,83 @@
+"""VGG (parity with gluon/model_zoo/vision/vgg.py)."""
+from ... import nn
+from ...block import HybridBlock
+
+__all__ = ["VGG", "vgg11", "vgg13", "vgg16", "vgg19", "vgg11_bn", "vgg13_bn",
+           "vgg16_bn", "vgg19_bn"]
+
+
+class VGG(HybridBlock):
+    def __init__(self, layers, filters, classes=1000, batch_norm=False,
+                 **kwargs):
+        super().__init__(**kwargs)
+        assert len(layers) == len(filters)
+        with self.name_scope():
+            self.features = self._make_features(layers, filters, batch_norm)
+            self.features.add(nn.Dense(4096, activation="relu",
+                                       weight_initializer="normal"))
+            self.features.add(nn.Dropout(rate=0.5))
+            self.features.add(nn.Dense(4096, activation="relu",
+                                       weight_initializer="normal"))
+            self.features.add(nn.Dropout(rate=0.5))
+            self.output = nn.Dense(classes, weight_initializer="normal")
+
+    def _make_features(self, layers, filters, batch_norm):
+        featurizer = nn.HybridSequential(prefix="")
+        for i, num in enumerate(layers):
+            for _ in range(num):
+                featurizer.add(nn.Conv2D(filters[i], kernel_size=3,
+                                         padding=1,
+                                         weight_initializer="xavier"))
+                if batch_norm:
+                    featurizer.add(nn.BatchNorm())
+                featurizer.add(nn.Activation("relu"))
+            featurizer.add(nn.MaxPool2D(strides=2))
+        return featurizer
+
+    def forward(self, x):
+        return self.output(self.features(x))
+
+    hybrid_forward = None
+
+
+vgg_spec = {11: ([1, 1, 2, 2, 2], [64, 128, 256, 512, 512]),
+            13: ([2, 2, 2, 2, 2], [64, 128, 256, 512, 512]),
+            16: ([2, 2, 3, 3, 3], [64, 128, 256, 512, 512]),
+            19: ([2, 2, 4, 4, 4], [64, 128, 256, 512, 512])}
+
+
+def get_vgg(num_layers, **kwargs):
+    layers, filters = vgg_spec[num_layers]
+    return VGG(layers, filters, **kwargs)
+
+
+def vgg11(**kw):
+    return get_vgg(11, **kw)
+
+
+def vgg13(**kw):
+    return get_vgg(13, **kw)
+
+
+def vgg16(**kw):
+    return get_vgg(16, **kw)
+
+
+def vgg19(**kw):
+    return get_vgg(19, **kw)
+
+
+def vgg11_bn(**kw):
+    return get_vgg(11, batch_norm=True, **kw)
+
+
+def vgg13_bn(**kw):
+    return get_vgg(13, batch_norm=True, **kw)
+
+
+def vgg16_bn(**kw):
+    return get_vgg(16, batch_norm=True, **kw)
+
+
+def vgg19_bn(**kw):
+    return get_vgg(19, batch_norm=True, **kw)

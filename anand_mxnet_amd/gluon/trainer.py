@@ -1,0 +1,159 @@
+"""Gluon Trainer (parity with python/mxnet/gluon/trainer.py).
+
+MI355X-native data-parallel path: in the one-process-per-GPU regime
+(torch.distributed initialized, RCCL backend) gradients are averaged by
+``parallel.bucketed.BucketedAllReducer`` — persistent flat buckets,
+all-reduce launched as backward fills each bucket (overlap), no per-key
+python loop.  Single-process multi-ctx values go through the KVStore
+('local'/'device' CommCPU/CommDevice semantics).  Optimizer updates are
+always worker-side (the reference's update_on_kvstore PS mode is replaced
+by all-reduce)."""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+from ..base import MXNetError
+from .. import optimizer as opt
+from .. import kvstore as kvs
+from .parameter import ParameterDict, Parameter
+
+__all__ = ["Trainer"]
+
+
+class Trainer:
+    def __init__(self, params, optimizer, optimizer_params=None, kvstore="device",
+                 compression_params=None, update_on_kvstore=None):
+        if isinstance(params, (dict, ParameterDict)):
+            params = list(params.values())
+        if not isinstance(params, (list, tuple)):
+            raise ValueError("params must be a ParameterDict / list of Parameter")
+        self._params = []
+        self._param2idx = {}
+        for i, param in enumerate(params):
+            if not isinstance(param, Parameter):
+                raise ValueError(f"invalid param {param}")
+            self._param2idx[param.name] = i
+            self._params.append(param)
+        self._compression_params = compression_params
+        optimizer_params = optimizer_params or {}
+        self._scale = float(optimizer_params.get("rescale_grad", 1.0))
+        self._init_optimizer(optimizer, optimizer_params)
+        self._kvstore_type = kvstore
+        self._kvstore = None
+        self._kv_initialized = False
+        self._reducer = None
+        self._distributed = dist.is_initialized() and dist.get_world_size() > 1
+        self._params_to_init = []
+        self._contains_sparse_grad = False
+        self._update_on_kvstore = False if update_on_kvstore is None \
+            else update_on_kvstore
+
+    def _init_optimizer(self, optimizer, optimizer_params):
+        param_dict = {i: param for i, param in enumerate(self._params)}
+        if isinstance(optimizer, opt.Optimizer):
+            assert not optimizer_params or \
+                list(optimizer_params.keys()) == ["rescale_grad"], \
+                "optimizer_params must be None if optimizer is an Optimizer " \
+                "instance"
+            self._optimizer = optimizer
+            self._optimizer.param_dict = param_dict
+        else:
+            self._optimizer = opt.create(optimizer, param_dict=param_dict,
+                                         **optimizer_params)
+        self._updaters = [opt.get_updater(self._optimizer)]
+
+    # ---------------- kvstore / reducer init ----------------
+    def _init_kvstore(self):
+        if self._distributed:
+            # bucketed RCCL all-reduce path; grads must exist on one device
+            self._init_params_if_needed()
+            tensors = []
+            for p in self._params:
+                if p.grad_req != "null":
+                    for arr in p.list_data():
+                        tensors.append(arr._t)
+                        arr._zero_in_place = True
+            from ..parallel.bucketed import BucketedAllReducer
+            self._reducer = BucketedAllReducer(tensors, average=False)
+            # broadcast initial params from rank 0 so replicas agree
+            with torch.no_grad():
+                for p in self._params:
+                    for arr in p.list_data():
+                        dist.broadcast(arr._t.detach(), src=0)
+        else:
+            self._kvstore = kvs.create(
+                self._kvstore_type if self._kvstore_type else "local")
+        self._kv_initialized = True
+
+    def _init_params_if_needed(self):
+        for p in self._params:
+            p._check_initialized()
+
+    # ---------------- public API ----------------
+    @property
+    def learning_rate(self):
+        return self._optimizer.learning_rate
+
+    @property
+    def optimizer(self):
+        return self._optimizer
+
+    def set_learning_rate(self, lr):
+        self._optimizer.set_learning_rate(lr)
+
+    def step(self, batch_size, ignore_stale_grad=False):
+        """rescale grads by 1/batch_size, all-reduce, apply optimizer."""
+        if not self._kv_initialized:
+            self._init_kvstore()
+        self._optimizer.rescale_grad = self._scale / batch_size
+        self._allreduce_grads()
+        self._update(ignore_stale_grad)
+        if self._reducer is not None:
+            self._reducer.zero_grads()
+
+    def allreduce_grads(self):
+        if not self._kv_initialized:
+            self._init_kvstore()
+        self._allreduce_grads()
+
+    def _allreduce_grads(self):
+        if self._reducer is not None:
+            self._reducer.finish()
+            # sum across ranks; rescale_grad handles 1/batch (global batch)
+            return
+        if self._kvstore is not None:
+            for i, param in enumerate(self._params):
+                if param.grad_req != "null" and len(param.list_data()) > 1:
+                    self._kvstore.allreduce([param.list_grad()])
+
+    def update(self, batch_size, ignore_stale_grad=False):
+        if not self._kv_initialized:
+            self._init_kvstore()
+        self._optimizer.rescale_grad = self._scale / batch_size
+        self._update(ignore_stale_grad)
+
+    def _update(self, ignore_stale_grad=False):
+        updater = self._updaters[0]
+        for i, param in enumerate(self._params):
+            if param.grad_req == "null":
+                continue
+            for arr, grad in zip(param.list_data(), param.list_grad()):
+                updater(i, grad, arr)
+
+    # ---------------- states ----------------
+    def save_states(self, fname):
+        assert self._optimizer is not None
+        if not self._kv_initialized:
+            self._init_kvstore()
+        with open(fname, "wb") as f:
+            f.write(self._updaters[0].get_states(dump_optimizer=True))
+
+    def load_states(self, fname):
+        if not self._kv_initialized:
+            self._init_kvstore()
+        with open(fname, "rb") as f:
+            states = f.read()
+        for updater in self._updaters:
+            updater.set_states(states)
+            updater.optimizer = self._optimizer

@@ -1,0 +1,664 @@
+"""Core tensor ops: elementwise, broadcast, reductions, matrix manipulation,
+indexing, random sampling.
+
+Parity targets (reference layout): src/operator/tensor/elemwise_*_op*,
+broadcast_reduce_op*, matrix_op*, indexing_op.*, random/sample_op.* and the
+mshadow_op.h functor zoo.  On GPU the elementwise family executes through
+torch-ROCm's HIP kernels (the sanctioned substrate); the *hot* fused ops live
+in ops/nn.py + ops/hip with hand-written CDNA4 kernels.
+"""
+from __future__ import annotations
+
+import math
+import torch
+
+from .registry import register
+from ..base import dtype_from_any
+
+# ---------------- unary elementwise ----------------
+_UNARY = {
+    "relu": lambda x: torch.relu(x),
+    "sigmoid": torch.sigmoid,
+    "tanh": torch.tanh,
+    "exp": torch.exp,
+    "log": torch.log,
+    "log2": torch.log2,
+    "log10": torch.log10,
+    "sqrt": torch.sqrt,
+    "rsqrt": torch.rsqrt,
+    "square": torch.square,
+    "abs": torch.abs,
+    "sign": torch.sign,
+    "floor": torch.floor,
+    "ceil": torch.ceil,
+    "round": torch.round,
+    "trunc": torch.trunc,
+    "negative": torch.neg,
+    "reciprocal": torch.reciprocal,
+    "erf": torch.erf,
+    "erfinv": torch.erfinv,
+    "sin": torch.sin,
+    "cos": torch.cos,
+    "tan": torch.tan,
+    "arcsin": torch.asin,
+    "arccos": torch.acos,
+    "arctan": torch.atan,
+    "sinh": torch.sinh,
+    "cosh": torch.cosh,
+    "arcsinh": torch.asinh,
+    "arccosh": torch.acosh,
+    "arctanh": torch.atanh,
+    "expm1": torch.expm1,
+    "log1p": torch.log1p,
+    "gamma": lambda x: torch.exp(torch.lgamma(x)),
+    "gammaln": torch.lgamma,
+    "logical_not": lambda x: (x == 0).to(x.dtype),
+}
+for _name, _fn in _UNARY.items():
+    register(_name)(_fn)
+
+
+@register("softsign")
+def softsign(x):
+    return x / (1 + torch.abs(x))
+
+
+@register("softrelu")
+def softrelu(x):
+    return torch.nn.functional.softplus(x)
+
+
+@register("gelu")
+def gelu(x):
+    return torch.nn.functional.gelu(x)
+
+
+@register("gelu_tanh")
+def gelu_tanh(x):
+    return torch.nn.functional.gelu(x, approximate="tanh")
+
+
+@register("silu")
+def silu(x):
+    return torch.nn.functional.silu(x)
+
+
+@register("LeakyReLU")
+def leaky_relu(x, gamma=None, act_type="leaky", slope=0.25, lower_bound=None,
+               upper_bound=None):
+    if act_type == "leaky":
+        return torch.nn.functional.leaky_relu(x, slope)
+    if act_type == "prelu":
+        return torch.nn.functional.prelu(x, gamma)
+    if act_type == "elu":
+        return torch.nn.functional.elu(x, slope)
+    if act_type == "selu":
+        return torch.nn.functional.selu(x)
+    if act_type == "gelu":
+        return torch.nn.functional.gelu(x)
+    raise ValueError(f"unknown act_type {act_type}")
+
+
+# ---------------- binary / broadcast ----------------
+_BINARY = {
+    "elemwise_add": torch.add, "broadcast_add": torch.add,
+    "elemwise_sub": torch.sub, "broadcast_sub": torch.sub,
+    "elemwise_mul": torch.mul, "broadcast_mul": torch.mul,
+    "elemwise_div": torch.div, "broadcast_div": torch.div,
+    "broadcast_power": torch.pow, "power": torch.pow,
+    "broadcast_maximum": torch.maximum, "maximum": torch.maximum,
+    "broadcast_minimum": torch.minimum, "minimum": torch.minimum,
+    "broadcast_mod": torch.remainder, "mod": torch.remainder,
+    "broadcast_hypot": torch.hypot,
+    "arctan2": torch.atan2,
+}
+for _name, _fn in _BINARY.items():
+    register(_name)(_fn)
+
+for _name, _cmp in [("broadcast_equal", torch.eq), ("broadcast_not_equal", torch.ne),
+                    ("broadcast_greater", torch.gt), ("broadcast_greater_equal", torch.ge),
+                    ("broadcast_lesser", torch.lt), ("broadcast_lesser_equal", torch.le),
+                    ("broadcast_logical_and", torch.logical_and),
+                    ("broadcast_logical_or", torch.logical_or),
+                    ("broadcast_logical_xor", torch.logical_xor)]:
+    register(_name)(lambda a, b, _c=_cmp: _c(a, b).to(
+        a.dtype if isinstance(a, torch.Tensor) else b.dtype))
+
+
+@register("add_n")
+def add_n(*args):
+    """Multi-input sum (reference: ElementwiseSum, ndarray_function.cu)."""
+    if len(args) == 1 and isinstance(args[0], (list, tuple)):
+        args = args[0]
+    out = args[0].clone() if len(args) > 1 else args[0]
+    for a in args[1:]:
+        out = out + a
+    return out
+
+
+@register("clip")
+def clip(x, a_min=None, a_max=None):
+    return torch.clamp(x, a_min, a_max)
+
+
+@register("where")
+def where(cond, a, b):
+    return torch.where(cond.bool(), a, b)
+
+
+# ---------------- reductions ----------------
+def _axes(axis):
+    if axis is None:
+        return None
+    if isinstance(axis, int):
+        return (axis,)
+    return tuple(axis)
+
+
+@register("sum")
+def sum_(x, axis=None, keepdims=False, exclude=False):
+    ax = _axes(axis)
+    if exclude and ax is not None:
+        ax = tuple(i for i in range(x.dim()) if i not in
+                   tuple(a % x.dim() for a in ax))
+    if ax is None:
+        out = x.sum()
+        return out.reshape((1,) * x.dim()) if keepdims else out
+    return x.sum(dim=ax, keepdim=keepdims)
+
+
+@register("mean")
+def mean(x, axis=None, keepdims=False, exclude=False):
+    ax = _axes(axis)
+    if exclude and ax is not None:
+        ax = tuple(i for i in range(x.dim()) if i not in
+                   tuple(a % x.dim() for a in ax))
+    if ax is None:
+        out = x.float().mean().to(x.dtype)
+        return out.reshape((1,) * x.dim()) if keepdims else out
+    return x.float().mean(dim=ax, keepdim=keepdims).to(x.dtype)
+
+
+@register("max")
+def max_(x, axis=None, keepdims=False):
+    ax = _axes(axis)
+    if ax is None:
+        out = x.max()
+        return out.reshape((1,) * x.dim()) if keepdims else out
+    out = x
+    for a in sorted((a % x.dim() for a in ax), reverse=True):
+        out = out.max(dim=a, keepdim=keepdims).values
+    return out
+
+
+@register("min")
+def min_(x, axis=None, keepdims=False):
+    ax = _axes(axis)
+    if ax is None:
+        out = x.min()
+        return out.reshape((1,) * x.dim()) if keepdims else out
+    out = x
+    for a in sorted((a % x.dim() for a in ax), reverse=True):
+        out = out.min(dim=a, keepdim=keepdims).values
+    return out
+
+
+@register("prod")
+def prod(x, axis=None, keepdims=False):
+    ax = _axes(axis)
+    if ax is None:
+        out = x.prod()
+        return out.reshape((1,) * x.dim()) if keepdims else out
+    out = x
+    for a in sorted((a % x.dim() for a in ax), reverse=True):
+        out = out.prod(dim=a, keepdim=keepdims)
+    return out
+
+
+@register("norm")
+def norm(x, ord=2, axis=None, keepdims=False):
+    """fp32-accumulated norm (reference MXNET_SAFE_ACCUMULATION semantics)."""
+    ax = _axes(axis)
+    xf = x.float()
+    if ax is None:
+        out = torch.linalg.vector_norm(xf, ord=ord)
+        out = out.reshape((1,)) if not keepdims else out.reshape((1,) * x.dim())
+        return out.to(x.dtype)
+    return torch.linalg.vector_norm(xf, ord=ord, dim=ax, keepdim=keepdims).to(x.dtype)
+
+
+@register("argmax")
+def argmax(x, axis=None, keepdims=False):
+    if axis is None:
+        return x.flatten().argmax().to(torch.float32)
+    return x.argmax(dim=axis, keepdim=keepdims).to(torch.float32)
+
+
+@register("argmin")
+def argmin(x, axis=None, keepdims=False):
+    if axis is None:
+        return x.flatten().argmin().to(torch.float32)
+    return x.argmin(dim=axis, keepdim=keepdims).to(torch.float32)
+
+
+@register("topk")
+def topk(x, axis=-1, k=1, ret_typ="indices", is_ascend=False):
+    vals, idx = torch.topk(x, k, dim=axis, largest=not is_ascend)
+    if ret_typ == "indices":
+        return idx.to(torch.float32)
+    if ret_typ == "value":
+        return vals
+    return [vals, idx.to(torch.float32)]
+
+
+@register("sort")
+def sort(x, axis=-1, is_ascend=True):
+    return torch.sort(x, dim=axis, descending=not is_ascend).values
+
+
+@register("argsort")
+def argsort(x, axis=-1, is_ascend=True):
+    return torch.argsort(x, dim=axis, descending=not is_ascend).to(torch.float32)
+
+
+# ---------------- matrix / shape manipulation ----------------
+@register("dot")
+def dot(a, b, transpose_a=False, transpose_b=False):
+    """GEMM. GPU path: hand-written MFMA kernel via ops.hip (see nn.py dot)."""
+    from . import nn as _nn
+    return _nn.dot_impl(a, b, transpose_a, transpose_b)
+
+
+@register("batch_dot")
+def batch_dot(a, b, transpose_a=False, transpose_b=False):
+    from . import nn as _nn
+    return _nn.batch_dot_impl(a, b, transpose_a, transpose_b)
+
+
+@register("transpose")
+def transpose(x, axes=None):
+    if axes is None:
+        axes = tuple(reversed(range(x.dim())))
+    return x.permute(tuple(axes))
+
+
+@register("reshape")
+def reshape(x, shape):
+    out = []
+    for i, s in enumerate(shape):
+        if s == 0:
+            out.append(x.shape[i])
+        else:
+            out.append(int(s))
+    return x.reshape(out)
+
+
+@register("expand_dims")
+def expand_dims(x, axis):
+    return x.unsqueeze(axis)
+
+
+@register("squeeze")
+def squeeze(x, axis=None):
+    return x.squeeze() if axis is None else x.squeeze(axis)
+
+
+@register("Flatten")
+def flatten(x):
+    return x.reshape(x.shape[0], -1)
+
+
+@register("concat")
+def concat(*args, dim=1):
+    if len(args) == 1 and isinstance(args[0], (list, tuple)):
+        args = args[0]
+    return torch.cat(list(args), dim=dim)
+
+
+@register("stack")
+def stack(*args, axis=0):
+    if len(args) == 1 and isinstance(args[0], (list, tuple)):
+        args = args[0]
+    return torch.stack(list(args), dim=axis)
+
+
+@register("split")
+def split(x, num_outputs, axis=1, squeeze_axis=False):
+    outs = torch.chunk(x, num_outputs, dim=axis)
+    if squeeze_axis:
+        outs = [o.squeeze(axis) for o in outs]
+    return list(outs)
+
+
+@register("slice")
+def slice_(x, begin, end, step=None):
+    idx = []
+    for i in range(len(begin)):
+        b = begin[i]
+        e = end[i]
+        s = step[i] if step is not None else None
+        idx.append(slice(b, e, s))
+    return x[tuple(idx)]
+
+
+@register("slice_axis")
+def slice_axis(x, axis, begin, end):
+    if end is None:
+        end = x.shape[axis]
+    return x.narrow(axis, begin, end - begin)
+
+
+@register("slice_like")
+def slice_like(x, like, axes=None):
+    if axes is None:
+        axes = range(min(x.dim(), like.dim()))
+    out = x
+    for a in axes:
+        out = out.narrow(a, 0, like.shape[a])
+    return out
+
+
+@register("tile")
+def tile(x, reps):
+    return x.repeat(tuple(reps))
+
+
+@register("repeat")
+def repeat(x, repeats, axis=None):
+    if axis is None:
+        return x.flatten().repeat_interleave(repeats)
+    return x.repeat_interleave(repeats, dim=axis)
+
+
+@register("flip")
+def flip(x, axis):
+    if isinstance(axis, int):
+        axis = (axis,)
+    return torch.flip(x, tuple(axis))
+
+
+@register("pad")
+def pad(x, mode="constant", pad_width=None, constant_value=0):
+    # MXNet pad_width: (before, after) per axis, first dims first
+    pw = []
+    for i in range(len(pad_width) // 2 - 1, -1, -1):
+        pw += [pad_width[2 * i], pad_width[2 * i + 1]]
+    return torch.nn.functional.pad(x, pw, mode=mode if mode != "edge" else "replicate",
+                                   value=constant_value)
+
+
+@register("Cast")
+def cast(x, dtype):
+    return x.to(dtype_from_any(dtype))
+
+
+cast_op = cast
+register("cast")(cast)
+
+
+@register("amp_cast")
+def amp_cast(x, dtype):
+    return x.to(dtype_from_any(dtype))
+
+
+@register("amp_multicast")
+def amp_multicast(*args, num_outputs=None):
+    xs = list(args)
+    widest = xs[0].dtype
+    for x in xs[1:]:
+        widest = torch.promote_types(widest, x.dtype)
+    return [x.to(widest) for x in xs]
+
+
+@register("broadcast_to")
+def broadcast_to(x, shape):
+    shp = [x.shape[i] if s == 0 else s for i, s in enumerate(shape)]
+    return x.expand(shp)
+
+
+@register("broadcast_axis")
+def broadcast_axis(x, axis, size):
+    if isinstance(axis, int):
+        axis, size = (axis,), (size,)
+    shape = list(x.shape)
+    for a, s in zip(axis, size):
+        shape[a] = s
+    return x.expand(shape)
+
+
+@register("zeros_like")
+def zeros_like(x):
+    return torch.zeros_like(x)
+
+
+@register("ones_like")
+def ones_like(x):
+    return torch.ones_like(x)
+
+
+@register("full_like")
+def full_like(x, fill_value):
+    return torch.full_like(x, fill_value)
+
+
+@register("identity")
+def identity(x):
+    return x.clone()
+
+
+@register("BlockGrad")
+def block_grad(x):
+    return x.detach()
+
+
+@register("stop_gradient")
+def stop_gradient(x):
+    return x.detach()
+
+
+@register("make_loss")
+def make_loss(x):
+    return x
+
+
+@register("L2Normalization")
+def l2_normalization(x, eps=1e-10, mode="instance"):
+    if mode == "instance":
+        n = torch.sqrt((x.float() ** 2).sum(dim=tuple(range(1, x.dim())),
+                                            keepdim=True) + eps)
+        return (x.float() / n).to(x.dtype)
+    if mode == "channel":
+        n = torch.sqrt((x.float() ** 2).sum(dim=1, keepdim=True) + eps)
+        return (x.float() / n).to(x.dtype)
+    raise ValueError(mode)
+
+
+@register("smooth_l1")
+def smooth_l1(x, scalar=1.0):
+    s2 = scalar * scalar
+    ax = torch.abs(x)
+    return torch.where(ax < 1.0 / s2, 0.5 * s2 * x * x, ax - 0.5 / s2)
+
+
+# ---------------- indexing ----------------
+@register("Embedding")
+def embedding(data, weight, input_dim=None, output_dim=None, dtype=None,
+              sparse_grad=False):
+    from . import nn as _nn
+    return _nn.embedding_impl(data, weight)
+
+
+@register("take")
+def take(a, indices, axis=0, mode="clip"):
+    idx = indices.long()
+    if mode == "clip":
+        idx = idx.clamp(0, a.shape[axis] - 1)
+    elif mode == "wrap":
+        idx = idx % a.shape[axis]
+    return torch.index_select(a, axis, idx.flatten()).reshape(
+        a.shape[:axis] + tuple(indices.shape) + a.shape[axis + 1:])
+
+
+@register("batch_take")
+def batch_take(a, indices):
+    return a.gather(1, indices.long().unsqueeze(1)).squeeze(1)
+
+
+@register("pick")
+def pick(data, index, axis=-1, keepdims=False, mode="clip"):
+    idx = index.long()
+    if mode == "clip":
+        idx = idx.clamp(0, data.shape[axis] - 1)
+    out = data.gather(axis, idx.unsqueeze(axis))
+    return out if keepdims else out.squeeze(axis)
+
+
+@register("one_hot")
+def one_hot(indices, depth, on_value=1.0, off_value=0.0, dtype="float32"):
+    oh = torch.nn.functional.one_hot(indices.long(), depth).to(dtype_from_any(dtype))
+    if on_value != 1.0 or off_value != 0.0:
+        oh = oh * (on_value - off_value) + off_value
+    return oh
+
+
+@register("gather_nd")
+def gather_nd(data, indices):
+    idx = indices.long()
+    m = idx.shape[0]
+    out_shape = tuple(idx.shape[1:]) + tuple(data.shape[m:])
+    flat_idx = [idx[i].flatten() for i in range(m)]
+    return data[tuple(flat_idx)].reshape(out_shape)
+
+
+@register("scatter_nd")
+def scatter_nd(data, indices, shape):
+    idx = indices.long()
+    out = torch.zeros(tuple(shape), dtype=data.dtype, device=data.device)
+    m = idx.shape[0]
+    flat_idx = tuple(idx[i].flatten() for i in range(m))
+    out.index_put_(flat_idx, data.reshape((-1,) + tuple(shape[m:])),
+                   accumulate=True)
+    return out
+
+
+@register("SequenceMask")
+def sequence_mask(data, sequence_length=None, use_sequence_length=False,
+                  value=0.0, axis=0):
+    if not use_sequence_length or sequence_length is None:
+        return data
+    maxlen = data.shape[axis]
+    pos = torch.arange(maxlen, device=data.device)
+    if axis == 0:
+        mask = pos.unsqueeze(1) < sequence_length.unsqueeze(0)  # (T, B)
+        for _ in range(data.dim() - 2):
+            mask = mask.unsqueeze(-1)
+    else:  # axis == 1: (B, T, ...)
+        mask = pos.unsqueeze(0) < sequence_length.unsqueeze(1)
+        for _ in range(data.dim() - 2):
+            mask = mask.unsqueeze(-1)
+    return torch.where(mask, data, torch.as_tensor(value, dtype=data.dtype,
+                                                   device=data.device))
+
+
+@register("SequenceLast")
+def sequence_last(data, sequence_length=None, use_sequence_length=False, axis=0):
+    if not use_sequence_length:
+        return data.select(axis, data.shape[axis] - 1)
+    idx = (sequence_length.long() - 1)
+    if axis == 0:
+        return data[idx, torch.arange(data.shape[1], device=data.device)]
+    return data[torch.arange(data.shape[0], device=data.device), idx]
+
+
+@register("SequenceReverse")
+def sequence_reverse(data, sequence_length=None, use_sequence_length=False, axis=0):
+    if not use_sequence_length:
+        return torch.flip(data, (axis,))
+    out = data.clone()
+    for b in range(data.shape[1]):
+        n = int(sequence_length[b])
+        out[:n, b] = torch.flip(data[:n, b], (0,))
+    return out
+
+
+@register("boolean_mask")
+def boolean_mask(data, index, axis=0):
+    return data[index.bool()] if axis == 0 else data.transpose(0, axis)[index.bool()].transpose(0, axis)
+
+
+# ---------------- random ----------------
+@register("random_uniform")
+def random_uniform(low=0.0, high=1.0, shape=(1,), dtype="float32", device=None):
+    td = dtype_from_any(dtype)
+    return torch.empty(tuple(shape), dtype=td, device=device).uniform_(low, high)
+
+
+@register("random_normal")
+def random_normal(loc=0.0, scale=1.0, shape=(1,), dtype="float32", device=None):
+    td = dtype_from_any(dtype)
+    return torch.empty(tuple(shape), dtype=td, device=device).normal_(loc, scale)
+
+
+@register("random_randint")
+def random_randint(low, high, shape=(1,), dtype="int32", device=None):
+    return torch.randint(low, high, tuple(shape), dtype=dtype_from_any(dtype),
+                         device=device)
+
+
+@register("random_gamma")
+def random_gamma(alpha=1.0, beta=1.0, shape=(1,), dtype="float32", device=None):
+    d = torch.distributions.Gamma(alpha, 1.0 / beta)
+    return d.sample(tuple(shape)).to(dtype_from_any(dtype))
+
+
+@register("random_exponential")
+def random_exponential(lam=1.0, shape=(1,), dtype="float32", device=None):
+    return torch.empty(tuple(shape), dtype=dtype_from_any(dtype),
+                       device=device).exponential_(lam)
+
+
+@register("random_poisson")
+def random_poisson(lam=1.0, shape=(1,), dtype="float32", device=None):
+    return torch.poisson(torch.full(tuple(shape), float(lam),
+                                    device=device)).to(dtype_from_any(dtype))
+
+
+@register("sample_multinomial")
+def sample_multinomial(data, shape=1, get_prob=False, dtype="int32"):
+    n = shape if isinstance(shape, int) else shape[0]
+    idx = torch.multinomial(data, n, replacement=True)
+    if get_prob:
+        lp = torch.log(data.gather(-1, idx))
+        return [idx.to(dtype_from_any(dtype)), lp]
+    return idx.to(dtype_from_any(dtype))
+
+
+@register("shuffle")
+def shuffle(x):
+    return x[torch.randperm(x.shape[0], device=x.device)]
+
+
+# ---------------- linalg (subset) ----------------
+@register("linalg_gemm2")
+def linalg_gemm2(a, b, transpose_a=False, transpose_b=False, alpha=1.0):
+    from . import nn as _nn
+    return alpha * _nn.batch_dot_impl(a, b, transpose_a, transpose_b) \
+        if a.dim() > 2 else alpha * _nn.dot_impl(a, b, transpose_a, transpose_b)
+
+
+@register("linalg_potrf")
+def linalg_potrf(a):
+    return torch.linalg.cholesky(a)
+
+
+@register("linalg_trsm")
+def linalg_trsm(a, b, transpose=False, rightside=False, lower=True, alpha=1.0):
+    return alpha * torch.linalg.solve_triangular(
+        a.transpose(-1, -2) if transpose else a, b,
+        upper=(not lower) ^ transpose, left=not rightside)
+
+
+@register("diag")
+def diag(x, k=0):
+    return torch.diagonal(x, offset=k, dim1=-2, dim2=-1) if x.dim() >= 2 \
+        else torch.diag(x, k)

@@ -1,0 +1,411 @@
+"""Neural-net hot ops: GEMM/FC, Convolution, BatchNorm, LayerNorm, softmax,
+Dropout, Pooling, Embedding, attention.
+
+Parity targets: src/operator/nn/* and src/operator/contrib/transformer.* in
+the reference.  Dispatch contract (MI355X-first):
+  * tensors on ``gpu(i)``  -> hand-written HIP/CDNA4 kernels registered in
+    ``ops.hip`` (MFMA/LDS-tiled, NHWC-native).  If the native extension is
+    missing on a GPU box this raises — no silent eager fallback
+    (set MXA_ALLOW_TORCH_FALLBACK=1 explicitly for bring-up only).
+  * tensors on cpu -> torch compositions; these are the numerics oracle the
+    GPU kernels are tested against (reference pattern: check_consistency in
+    tests/python/gpu/test_operator_gpu.py).
+"""
+from __future__ import annotations
+
+import os
+import torch
+import torch.nn.functional as F
+
+from .registry import register, gpu_impl, has_gpu_impl
+from .. import autograd as _ag
+
+
+def _allow_fallback() -> bool:
+    return os.environ.get("MXA_ALLOW_TORCH_FALLBACK", "0") not in ("0", "")
+
+
+def _use_hip(x: torch.Tensor, name: str) -> bool:
+    if not x.is_cuda:
+        return False
+    if has_gpu_impl(name):
+        return True
+    if _allow_fallback():
+        return False
+    gpu_impl(name)  # raises loudly
+    return False
+
+
+# ---------------- GEMM ----------------
+
+def dot_impl(a, b, transpose_a=False, transpose_b=False):
+    if a.dim() == 1 and b.dim() == 1:
+        return (a * b).sum()
+    if _use_hip(a, "gemm") and a.dtype == torch.bfloat16:
+        return gpu_impl("gemm")(a, b, transpose_a, transpose_b)
+    x = a.transpose(-1, -2) if transpose_a else a
+    y = b.transpose(-1, -2) if transpose_b else b
+    return torch.matmul(x, y)
+
+
+def batch_dot_impl(a, b, transpose_a=False, transpose_b=False):
+    if _use_hip(a, "batch_gemm") and a.dtype == torch.bfloat16:
+        return gpu_impl("batch_gemm")(a, b, transpose_a, transpose_b)
+    x = a.transpose(-1, -2) if transpose_a else a
+    y = b.transpose(-1, -2) if transpose_b else b
+    return torch.matmul(x, y)
+
+
+@register("FullyConnected")
+def fully_connected(data, weight, bias=None, num_hidden=None, no_bias=False,
+                    flatten=True):
+    """y = x W^T + b ; weight is (num_hidden, in) (reference convention)."""
+    x = data.reshape(data.shape[0], -1) if flatten and data.dim() > 2 else data
+    if _use_hip(x, "linear") and x.dtype == torch.bfloat16:
+        return gpu_impl("linear")(x, weight, None if no_bias else bias)
+    out = F.linear(x, weight, None if no_bias else bias)
+    return out
+
+
+# ---------------- Convolution ----------------
+
+@register("Convolution")
+def convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
+                dilate=(1, 1), pad=(0, 0), num_filter=None, num_group=1,
+                no_bias=False, layout="NCHW", cudnn_tune=None, cudnn_off=False,
+                workspace=None):
+    """2-D convolution.  GPU path: implicit-GEMM MFMA kernel, NHWC-native.
+
+    Weight conventions (reference parity): layout NCHW -> (K, C, R, S);
+    layout NHWC -> (K, R, S, C).
+    """
+    b = None if no_bias else bias
+    if layout == "NHWC":
+        if _use_hip(data, "conv2d_nhwc") and data.dtype == torch.bfloat16:
+            return gpu_impl("conv2d_nhwc")(data, weight, b, stride, pad,
+                                           dilate, num_group)
+        # CPU oracle: permute to NCHW for torch
+        x = data.permute(0, 3, 1, 2)
+        w = weight.permute(0, 3, 1, 2)  # (K,R,S,C) -> (K,C,R,S)
+        out = F.conv2d(x, w, b, stride=tuple(stride), padding=tuple(pad),
+                       dilation=tuple(dilate), groups=num_group)
+        return out.permute(0, 2, 3, 1).contiguous()
+    if data.is_cuda and data.dtype == torch.bfloat16 and has_gpu_impl("conv2d_nhwc"):
+        # NCHW request on GPU: run the NHWC kernel with layout conversion
+        x = data.permute(0, 2, 3, 1).contiguous()
+        w = weight.permute(0, 2, 3, 1).contiguous()
+        out = gpu_impl("conv2d_nhwc")(x, w, b, stride, pad, dilate, num_group)
+        return out.permute(0, 3, 1, 2)
+    _use_hip(data, "conv2d_nhwc")  # raises on GPU without native ext
+    return F.conv2d(data, weight, b, stride=tuple(stride), padding=tuple(pad),
+                    dilation=tuple(dilate), groups=num_group)
+
+
+@register("Deconvolution")
+def deconvolution(data, weight, bias=None, kernel=None, stride=(1, 1),
+                  dilate=(1, 1), pad=(0, 0), adj=(0, 0), num_filter=None,
+                  num_group=1, no_bias=True, layout="NCHW"):
+    b = None if no_bias else bias
+    return F.conv_transpose2d(data, weight, b, stride=tuple(stride),
+                              padding=tuple(pad), output_padding=tuple(adj),
+                              dilation=tuple(dilate), groups=num_group)
+
+
+# ---------------- Pooling ----------------
+
+@register("Pooling")
+def pooling(data, kernel=(2, 2), pool_type="max", global_pool=False,
+            stride=None, pad=(0, 0), layout="NCHW", count_include_pad=True,
+            pooling_convention="valid"):
+    nhwc = layout == "NHWC"
+    if nhwc and _use_hip(data, "pool_nhwc"):
+        return gpu_impl("pool_nhwc")(data, kernel, pool_type, global_pool,
+                                     stride, pad, count_include_pad)
+    x = data.permute(0, 3, 1, 2) if nhwc else data
+    if not nhwc and x.is_cuda:
+        _use_hip(x.permute(0, 2, 3, 1), "pool_nhwc")
+    if global_pool:
+        if pool_type == "max":
+            out = F.adaptive_max_pool2d(x, 1)
+        else:
+            out = F.adaptive_avg_pool2d(x, 1)
+    else:
+        stride = stride or kernel
+        ceil = pooling_convention == "full"
+        if pool_type == "max":
+            out = F.max_pool2d(x, tuple(kernel), tuple(stride), tuple(pad),
+                               ceil_mode=ceil)
+        elif pool_type == "avg":
+            out = F.avg_pool2d(x, tuple(kernel), tuple(stride), tuple(pad),
+                               ceil_mode=ceil, count_include_pad=count_include_pad)
+        elif pool_type == "sum":
+            out = F.avg_pool2d(x, tuple(kernel), tuple(stride), tuple(pad),
+                               ceil_mode=ceil, divisor_override=1)
+        else:
+            raise ValueError(pool_type)
+    return out.permute(0, 2, 3, 1).contiguous() if nhwc else out
+
+
+# ---------------- Normalization ----------------
+
+@register("BatchNorm")
+def batch_norm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
+               momentum=0.9, fix_gamma=False, use_global_stats=False,
+               output_mean_var=False, axis=1, cudnn_off=False,
+               fuse_relu=False):
+    """BatchNorm; training mode from autograd.is_training() (reference
+    semantics).  MXNet momentum: moving = m*moving + (1-m)*batch.
+    GPU path: fused single-pass Welford NHWC kernel (axis=-1/3)."""
+    training = _ag.is_training() and not use_global_stats
+    g = torch.ones_like(gamma) if fix_gamma else gamma
+    ch_axis = axis % data.dim()
+    if ch_axis == data.dim() - 1 and data.dim() == 4 and \
+            _use_hip(data, "batchnorm_nhwc"):
+        return gpu_impl("batchnorm_nhwc")(data, g, beta, moving_mean,
+                                          moving_var, eps, momentum, training,
+                                          fuse_relu)
+    if data.is_cuda:
+        _use_hip(data, "batchnorm_nhwc")
+    # torch F.batch_norm expects channel at dim1
+    perm = None
+    x = data
+    if ch_axis != 1:
+        perm = list(range(data.dim()))
+        perm.insert(1, perm.pop(ch_axis))
+        x = data.permute(perm).contiguous()
+    out = F.batch_norm(x, moving_mean, moving_var, g, beta, training,
+                       1.0 - momentum, eps)
+    if fuse_relu:
+        out = torch.relu(out)
+    if perm is not None:
+        inv = [0] * len(perm)
+        for i, p in enumerate(perm):
+            inv[p] = i
+        out = out.permute(inv).contiguous()
+    return out
+
+
+@register("LayerNorm")
+def layer_norm(data, gamma, beta, axis=-1, eps=1e-5):
+    ax = axis % data.dim()
+    if ax == data.dim() - 1:
+        if _use_hip(data, "layernorm"):
+            return gpu_impl("layernorm")(data, gamma, beta, eps)
+        return F.layer_norm(data, (data.shape[-1],), gamma, beta, eps)
+    # general-axis fallback: move axis last
+    x = data.transpose(ax, -1)
+    out = F.layer_norm(x, (x.shape[-1],), gamma, beta, eps)
+    return out.transpose(ax, -1)
+
+
+@register("InstanceNorm")
+def instance_norm(data, gamma, beta, eps=1e-5):
+    return F.instance_norm(data, weight=gamma, bias=beta, eps=eps)
+
+
+@register("GroupNorm")
+def group_norm(data, gamma, beta, num_groups=1, eps=1e-5):
+    return F.group_norm(data, num_groups, gamma, beta, eps)
+
+
+@register("RMSNorm")
+def rms_norm(data, gamma, eps=1e-6):
+    if _use_hip(data, "rmsnorm"):
+        return gpu_impl("rmsnorm")(data, gamma, eps)
+    xf = data.float()
+    r = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * r).to(data.dtype) * gamma
+
+
+# ---------------- softmax family ----------------
+
+@register("softmax")
+def softmax(data, axis=-1, temperature=None, length=None):
+    x = data if temperature in (None, 1.0) else data / temperature
+    if length is not None:
+        # masked softmax over `axis` with valid lengths
+        maxlen = x.shape[axis]
+        pos = torch.arange(maxlen, device=x.device)
+        shape = [1] * x.dim()
+        shape[axis] = maxlen
+        mask = pos.reshape(shape) < length.reshape(
+            list(length.shape) + [1] * (x.dim() - length.dim()))
+        x = x.masked_fill(~mask, float("-inf"))
+    if _use_hip(x, "softmax") and axis in (-1, x.dim() - 1):
+        return gpu_impl("softmax")(x)
+    return F.softmax(x.float(), dim=axis).to(data.dtype)
+
+
+@register("log_softmax")
+def log_softmax(data, axis=-1, temperature=None):
+    x = data if temperature in (None, 1.0) else data / temperature
+    if _use_hip(x, "log_softmax") and axis in (-1, x.dim() - 1):
+        return gpu_impl("log_softmax")(x)
+    return F.log_softmax(x.float(), dim=axis).to(data.dtype)
+
+
+@register("softmax_cross_entropy")
+def softmax_cross_entropy(data, label):
+    ls = F.log_softmax(data.float(), dim=-1)
+    return -(ls.gather(1, label.long().unsqueeze(1))).sum()
+
+
+class _SoftmaxOutputFn(torch.autograd.Function):
+    """Legacy fused softmax + CE gradient (reference softmax_output.cu)."""
+
+    @staticmethod
+    def forward(ctx, data, label, grad_scale, normalization, ignore_label,
+                use_ignore):
+        prob = F.softmax(data.float(), dim=-1)
+        ctx.save_for_backward(prob, label)
+        ctx.grad_scale = grad_scale
+        ctx.normalization = normalization
+        ctx.ignore_label = ignore_label
+        ctx.use_ignore = use_ignore
+        return prob.to(data.dtype)
+
+    @staticmethod
+    def backward(ctx, gout):
+        prob, label = ctx.saved_tensors
+        lab = label.long()
+        grad = prob.clone()
+        oh = F.one_hot(lab.clamp(min=0), prob.shape[-1]).to(prob.dtype)
+        grad -= oh
+        if ctx.use_ignore:
+            keep = (lab != ctx.ignore_label).to(prob.dtype).unsqueeze(-1)
+            grad *= keep
+        scale = ctx.grad_scale
+        if ctx.normalization == "batch":
+            scale /= prob.shape[0]
+        elif ctx.normalization == "valid" and ctx.use_ignore:
+            valid = (lab != ctx.ignore_label).sum().clamp(min=1)
+            scale = scale / valid
+        grad = grad * scale
+        return grad.to(gout.dtype), None, None, None, None, None
+
+
+@register("SoftmaxOutput")
+def softmax_output(data, label, grad_scale=1.0, ignore_label=-1,
+                   use_ignore=False, normalization="null",
+                   multi_output=False, preserve_shape=False,
+                   out_grad=False, smooth_alpha=0.0):
+    return _SoftmaxOutputFn.apply(data, label, grad_scale, normalization,
+                                  ignore_label, use_ignore)
+
+
+# ---------------- Dropout ----------------
+
+@register("Dropout")
+def dropout(data, p=0.5, mode="training", axes=None, cudnn_off=False):
+    training = _ag.is_training() or mode == "always"
+    if not training or p == 0:
+        return data
+    if _use_hip(data, "dropout"):
+        return gpu_impl("dropout")(data, p)
+    if axes:
+        shape = list(data.shape)
+        for a in axes:
+            shape[a] = 1
+        mask = torch.bernoulli(
+            torch.full(shape, 1 - p, device=data.device)).to(data.dtype) / (1 - p)
+        return data * mask
+    return F.dropout(data, p, training=True)
+
+
+# ---------------- Activation (legacy op) ----------------
+
+@register("Activation")
+def activation(data, act_type="relu"):
+    if _use_hip(data, "act_" + act_type):
+        return gpu_impl("act_" + act_type)(data)
+    if act_type == "relu":
+        return torch.relu(data)
+    if act_type == "sigmoid":
+        return torch.sigmoid(data)
+    if act_type == "tanh":
+        return torch.tanh(data)
+    if act_type == "softrelu":
+        return F.softplus(data)
+    if act_type == "softsign":
+        return data / (1 + torch.abs(data))
+    if act_type == "gelu":
+        return F.gelu(data)
+    raise ValueError(f"unknown act_type {act_type}")
+
+
+# ---------------- Embedding ----------------
+
+def embedding_impl(data, weight):
+    if _use_hip(weight, "embedding"):
+        return gpu_impl("embedding")(data, weight)
+    return F.embedding(data.long(), weight)
+
+
+# ---------------- attention (contrib/transformer parity) ----------------
+
+@register("multi_head_attention")
+def multi_head_attention(q, k, v, num_heads, mask=None, dropout_p=0.0,
+                         causal=False, softmax_scale=None):
+    """Fused MHA.  q/k/v: (B, T, H*D).  GPU: flash-style CDNA4 kernel.
+
+    Reference parity: the _contrib_interleaved_matmul_selfatt_* pair
+    (src/operator/contrib/transformer.cu) composed into one op.
+    """
+    B, Tq, HD = q.shape
+    D = HD // num_heads
+    scale = softmax_scale if softmax_scale is not None else D ** -0.5
+    if _use_hip(q, "mha") and q.dtype == torch.bfloat16 and mask is None \
+            and dropout_p == 0.0:
+        return gpu_impl("mha")(q, k, v, num_heads, causal, scale)
+    qh = q.reshape(B, Tq, num_heads, D).transpose(1, 2)
+    kh = k.reshape(B, -1, num_heads, D).transpose(1, 2)
+    vh = v.reshape(B, -1, num_heads, D).transpose(1, 2)
+    s = torch.matmul(qh.float(), kh.float().transpose(-1, -2)) * scale
+    if causal:
+        Tk = kh.shape[2]
+        cm = torch.ones(Tq, Tk, dtype=torch.bool, device=q.device).tril()
+        s = s.masked_fill(~cm, float("-inf"))
+    if mask is not None:
+        s = s + mask.float()
+    p = F.softmax(s, dim=-1).to(q.dtype)
+    if dropout_p > 0 and _ag.is_training():
+        p = F.dropout(p, dropout_p, training=True)
+    out = torch.matmul(p, vh)
+    return out.transpose(1, 2).reshape(B, Tq, HD)
+
+
+@register("_contrib_interleaved_matmul_selfatt_qk")
+def interleaved_matmul_selfatt_qk(queries_keys_values, heads):
+    """(T, B, 3*H*D) interleaved QKV -> (B*H, T, T) scores (scaled)."""
+    T, B, P = queries_keys_values.shape
+    D = P // (3 * heads)
+    x = queries_keys_values.reshape(T, B, heads, 3, D)
+    q = x[:, :, :, 0].permute(1, 2, 0, 3).reshape(B * heads, T, D)
+    k = x[:, :, :, 1].permute(1, 2, 0, 3).reshape(B * heads, T, D)
+    return torch.matmul(q, k.transpose(-1, -2)) * (D ** -0.5)
+
+
+@register("_contrib_interleaved_matmul_selfatt_valatt")
+def interleaved_matmul_selfatt_valatt(queries_keys_values, attention, heads):
+    T, B, P = queries_keys_values.shape
+    D = P // (3 * heads)
+    x = queries_keys_values.reshape(T, B, heads, 3, D)
+    v = x[:, :, :, 2].permute(1, 2, 0, 3).reshape(B * heads, T, D)
+    out = torch.matmul(attention, v)  # (B*H, T, D)
+    return out.reshape(B, heads, T, D).permute(2, 0, 1, 3).reshape(T, B, heads * D)
+
+
+@register("_contrib_div_sqrt_dim")
+def div_sqrt_dim(x):
+    return x / (x.shape[-1] ** 0.5)
+
+
+# ---------------- RNN (unfused cells live in gluon.rnn) ----------------
+
+@register("RNN")
+def rnn(data, parameters, state, state_cell=None, state_size=None,
+        num_layers=1, mode="lstm", bidirectional=False, p=0.0,
+        state_outputs=False, projection_size=None):
+    raise NotImplementedError(
+        "fused RNN op not implemented; use gluon.rnn cells "
+        "(reference rnn.cu is cuDNN-bound; see docs/PARITY.md)")

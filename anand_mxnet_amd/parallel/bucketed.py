@@ -1,0 +1,119 @@
+"""Bucketed gradient all-reduce overlapped with backward.
+
+MI355X-native replacement for the reference's per-key KVStore push/pull loop
+(gluon/trainer.py -> kvstore_nccl.h grouped ncclAllReduce): gradients live in
+persistent flat buckets (param.grad tensors are VIEWS into the bucket, so
+backward accumulates in place with zero copies); when the last grad of a
+bucket lands (post-accumulate-grad hook), the bucket is all-reduced
+asynchronously on RCCL.  Bucket size is tuned for the xGMI fabric: each of
+the 7 links carries ≈153 GB/s point-to-point, and a ring all-reduce is
+per-link bound, so buckets are kept large (default 50 MiB) to amortize
+per-collective latency while still overlapping with backward.
+
+Buckets are filled in REVERSE registration order (backward produces last
+layers first), matching the reference's sorted-by-size lazy init intent and
+DDP practice.
+"""
+from __future__ import annotations
+
+import os
+from typing import List
+
+import torch
+import torch.distributed as dist
+
+DEFAULT_BUCKET_BYTES = int(os.environ.get("MXA_BUCKET_KB", str(50 * 1024))) * 1024
+
+
+class GradBucket:
+    def __init__(self, tensors: List[torch.Tensor], dtype, device):
+        numel = sum(t.numel() for t in tensors)
+        self.flat = torch.zeros(numel, dtype=dtype, device=device)
+        self.views = []
+        offset = 0
+        for t in tensors:
+            v = self.flat.narrow(0, offset, t.numel()).view(t.shape)
+            self.views.append(v)
+            offset += t.numel()
+        self.tensors = tensors
+        self.pending = 0
+        self.expected = len(tensors)
+        self.work = None
+
+    def attach(self):
+        for t, v in zip(self.tensors, self.views):
+            t.grad = v
+
+    def ready(self):
+        self.pending += 1
+        return self.pending >= self.expected
+
+    def reset(self):
+        self.pending = 0
+        self.work = None
+
+
+class BucketedAllReducer:
+    """Overlap-capable gradient averaging over torch.distributed (RCCL)."""
+
+    def __init__(self, param_tensors: List[torch.Tensor],
+                 bucket_bytes: int = DEFAULT_BUCKET_BYTES,
+                 average: bool = True, overlap: bool = True):
+        assert dist.is_initialized()
+        self.world_size = dist.get_world_size()
+        self.average = average
+        self.overlap = overlap and all(t.is_cuda for t in param_tensors)
+        self.buckets: List[GradBucket] = []
+        self._t2bucket = {}
+
+        # reverse order: backward computes last-registered params first
+        rev = list(reversed(param_tensors))
+        cur, cur_bytes = [], 0
+        for t in rev:
+            nbytes = t.numel() * t.element_size()
+            if cur and cur_bytes + nbytes > bucket_bytes:
+                self._make_bucket(cur)
+                cur, cur_bytes = [], 0
+            cur.append(t)
+            cur_bytes += nbytes
+        if cur:
+            self._make_bucket(cur)
+
+        if self.overlap:
+            for t in param_tensors:
+                t.register_post_accumulate_grad_hook(self._hook)
+
+    def _make_bucket(self, tensors):
+        b = GradBucket(tensors, tensors[0].dtype, tensors[0].device)
+        b.attach()
+        self.buckets.append(b)
+        for t in tensors:
+            self._t2bucket[t] = b
+
+    def _launch(self, bucket: GradBucket):
+        if self.average:
+            bucket.flat.div_(self.world_size)
+        bucket.work = dist.all_reduce(bucket.flat, op=dist.ReduceOp.SUM,
+                                      async_op=True)
+
+    def _hook(self, param: torch.Tensor):
+        b = self._t2bucket[param]
+        if b.ready():
+            self._launch(b)
+
+    def zero_grads(self):
+        for b in self.buckets:
+            b.flat.zero_()
+            b.reset()
+            b.attach()  # re-attach in case autograd replaced .grad
+
+    def finish(self):
+        """Complete all reductions (called from Trainer before update)."""
+        for b in self.buckets:
+            if b.work is None:  # non-overlap mode or hook missed (grad_req null)
+                if b.pending > 0 or not self.overlap:
+                    self._launch(b)
+            if b.work is not None:
+                b.work.wait()
+        for b in self.buckets:
+            b.reset()

@@ -1,0 +1,128 @@
+import numpy as np
+import pytest
+
+import anand_mxnet_amd as mx
+from anand_mxnet_amd import nd, autograd
+from anand_mxnet_amd.test_utils import assert_almost_equal, \
+    check_numeric_gradient
+
+
+def test_basic_grad():
+    x = nd.array([1.0, 2.0, 3.0])
+    x.attach_grad()
+    with autograd.record():
+        y = (x * x).sum()
+    y.backward()
+    assert_almost_equal(x.grad.asnumpy(), [2, 4, 6])
+
+
+def test_write_semantics():
+    """grad_req='write': second backward overwrites, not accumulates."""
+    x = nd.array([1.0, 2.0])
+    x.attach_grad()
+    for _ in range(2):
+        with autograd.record():
+            y = (3 * x).sum()
+        y.backward()
+    assert_almost_equal(x.grad.asnumpy(), [3, 3])
+
+
+def test_no_record_no_graph():
+    x = nd.array([1.0])
+    x.attach_grad()
+    y = x * 2  # outside record
+    with pytest.raises(Exception):
+        y.backward()
+
+
+def test_pause():
+    x = nd.array([2.0])
+    x.attach_grad()
+    with autograd.record():
+        y = x * x
+        with autograd.pause():
+            z = x * 10  # not recorded
+        w = y + z.detach()
+    w.backward()
+    assert_almost_equal(x.grad.asnumpy(), [4.0])
+
+
+def test_is_training():
+    assert not autograd.is_training()
+    with autograd.record():
+        assert autograd.is_training()
+    with autograd.record(train_mode=False):
+        assert not autograd.is_training()
+    with autograd.train_mode():
+        assert autograd.is_training()
+
+
+def test_head_grad():
+    x = nd.array([1.0, 2.0])
+    x.attach_grad()
+    with autograd.record():
+        y = x * 3
+    y.backward(nd.array([10.0, 100.0]))
+    assert_almost_equal(x.grad.asnumpy(), [30, 300])
+
+
+def test_chain_rule_ops():
+    x = nd.array(np.random.uniform(0.5, 2, (3, 4)).astype(np.float32))
+    x.attach_grad()
+    with autograd.record():
+        y = nd.exp(nd.log(x) * 2)  # = x^2
+        loss = y.sum()
+    loss.backward()
+    assert_almost_equal(x.grad.asnumpy(), 2 * x.asnumpy(), rtol=1e-4,
+                        atol=1e-5)
+
+
+def test_numeric_gradient_dense():
+    x = mx.test_utils.rand_ndarray((3, 4))
+    w = mx.test_utils.rand_ndarray((5, 4))
+    check_numeric_gradient(lambda a, b: nd.FullyConnected(
+        a, b, None, num_hidden=5, no_bias=True), [x, w])
+
+
+def test_numeric_gradient_softmax():
+    x = mx.test_utils.rand_ndarray((4, 6))
+    check_numeric_gradient(lambda a: (nd.softmax(a) ** 2), [x])
+
+
+def test_mark_variables():
+    x = nd.array([1.0, 2.0])
+    g = nd.zeros((2,))
+    autograd.mark_variables([x], [g])
+    with autograd.record():
+        y = (x * 5).sum()
+    y.backward()
+    assert_almost_equal(x.grad.asnumpy(), [5, 5])
+
+
+def test_custom_function():
+    class Square(autograd.Function):
+        def forward(self, x):
+            self.save_for_backward(x)
+            return x * x
+
+        def backward(self, dy):
+            (x,) = self.saved_tensors
+            return 2 * x * dy
+
+    x = nd.array([1.0, 3.0])
+    x.attach_grad()
+    f = Square()
+    with autograd.record():
+        y = f(x).sum()
+    y.backward()
+    assert_almost_equal(x.grad.asnumpy(), [2, 6])
+
+
+def test_grad_add_req():
+    x = nd.array([1.0])
+    x.attach_grad(grad_req="add")
+    for _ in range(3):
+        with autograd.record():
+            y = (x * 2).sum()
+        y.backward()
+    assert_almost_equal(x.grad.asnumpy(), [6.0])

@@ -1,0 +1,214 @@
+import numpy as np
+import pytest
+
+import anand_mxnet_amd as mx
+from anand_mxnet_amd import nd, gluon, autograd
+from anand_mxnet_amd.gluon import nn
+from anand_mxnet_amd.test_utils import assert_almost_equal
+
+
+def test_dense_shapes():
+    net = nn.Dense(10, in_units=4)
+    net.initialize()
+    out = net(nd.ones((2, 4)))
+    assert out.shape == (2, 10)
+    assert net.weight.shape == (10, 4)
+    assert net.bias.shape == (10,)
+
+
+def test_deferred_init():
+    net = nn.Dense(7)
+    net.initialize()
+    with pytest.raises(gluon.DeferredInitializationError):
+        net.weight.data()
+    out = net(nd.ones((3, 5)))
+    assert out.shape == (3, 7)
+    assert net.weight.shape == (7, 5)
+
+
+def test_sequential():
+    net = nn.HybridSequential()
+    net.add(nn.Dense(16, activation="relu"),
+            nn.Dropout(0.5),
+            nn.Dense(4))
+    net.initialize()
+    out = net(nd.ones((2, 8)))
+    assert out.shape == (2, 4)
+    assert len(net) == 3
+    assert isinstance(net[0], nn.Dense)
+
+
+def test_conv_block_nchw():
+    net = nn.Conv2D(8, kernel_size=3, padding=1, in_channels=3)
+    net.initialize()
+    out = net(nd.ones((2, 3, 8, 8)))
+    assert out.shape == (2, 8, 8, 8)
+    assert net.weight.shape == (8, 3, 3, 3)
+
+
+def test_conv_block_nhwc():
+    net = nn.Conv2D(8, kernel_size=3, padding=1, strides=2, layout="NHWC")
+    net.initialize()
+    out = net(nd.ones((2, 9, 9, 4)))
+    assert out.shape == (2, 5, 5, 8)
+    assert net.weight.shape == (8, 3, 3, 4)
+
+
+def test_conv_nhwc_matches_nchw():
+    np.random.seed(0)
+    x = np.random.uniform(-1, 1, (2, 5, 7, 3)).astype(np.float32)  # NHWC
+    w = np.random.uniform(-1, 1, (6, 3, 3, 3)).astype(np.float32)  # KCRS
+    out_nchw = nd.Convolution(nd.array(x.transpose(0, 3, 1, 2)),
+                              nd.array(w), None, kernel=(3, 3), pad=(1, 1),
+                              num_filter=6, no_bias=True, layout="NCHW")
+    out_nhwc = nd.Convolution(nd.array(x),
+                              nd.array(w.transpose(0, 2, 3, 1)), None,
+                              kernel=(3, 3), pad=(1, 1), num_filter=6,
+                              no_bias=True, layout="NHWC")
+    assert_almost_equal(out_nhwc.asnumpy(),
+                        out_nchw.asnumpy().transpose(0, 2, 3, 1),
+                        rtol=1e-4, atol=1e-5)
+
+
+def test_batchnorm_train_vs_eval():
+    bn = nn.BatchNorm(in_channels=4)
+    bn.initialize()
+    x = nd.array(np.random.uniform(-1, 1, (8, 4, 3, 3)).astype(np.float32))
+    with autograd.record():
+        y_train = bn(x)
+    m = bn.running_mean.data().asnumpy()
+    assert not np.allclose(m, 0)  # running stats updated
+    y_eval = bn(x)
+    assert y_train.shape == y_eval.shape
+
+
+def test_pooling_blocks():
+    x = nd.ones((2, 3, 8, 8))
+    assert nn.MaxPool2D(2, 2)(x).shape == (2, 3, 4, 4)
+    assert nn.AvgPool2D(2, 2)(x).shape == (2, 3, 4, 4)
+    assert nn.GlobalAvgPool2D()(x).shape == (2, 3, 1, 1)
+
+
+def test_embedding_block():
+    emb = nn.Embedding(20, 8)
+    emb.initialize()
+    out = emb(nd.array([[1, 2], [3, 4]]))
+    assert out.shape == (2, 2, 8)
+
+
+def test_layernorm_block():
+    ln = nn.LayerNorm(in_channels=6)
+    ln.initialize()
+    x = nd.array(np.random.uniform(-1, 1, (4, 6)).astype(np.float32))
+    y = ln(x).asnumpy()
+    assert np.allclose(y.mean(axis=-1), 0, atol=1e-5)
+    assert np.allclose(y.std(axis=-1), 1, atol=1e-2)
+
+
+def test_collect_params_select():
+    net = nn.HybridSequential(prefix="net_")
+    with net.name_scope():
+        net.add(nn.Dense(4, prefix="fc1_"), nn.Dense(2, prefix="fc2_"))
+    params = net.collect_params()
+    assert len(params) == 4
+    weights = net.collect_params(".*weight")
+    assert len(weights) == 2
+
+
+def test_trainer_updates_params():
+    net = nn.Dense(1, in_units=1, use_bias=False)
+    net.initialize(mx.init.Constant(2.0))
+    trainer = gluon.Trainer(net.collect_params(), "sgd",
+                            {"learning_rate": 0.1})
+    x = nd.array([[1.0]])
+    with autograd.record():
+        loss = (net(x) ** 2).sum()
+    loss.backward()
+    trainer.step(1)
+    w = net.weight.data().asscalar()
+    # dL/dw = 2*w*x^2 = 4; w_new = 2 - 0.1*4 = 1.6
+    assert abs(w - 1.6) < 1e-5
+
+
+def test_cast_bf16():
+    net = nn.Dense(4, in_units=3)
+    net.initialize()
+    net.cast("bfloat16")
+    out = net(nd.ones((2, 3), dtype="bfloat16"))
+    assert out.dtype == "bfloat16"
+
+
+def test_save_load_parameters(tmp_path):
+    net = nn.HybridSequential()
+    net.add(nn.Dense(5, activation="relu"), nn.Dense(3))
+    net.initialize()
+    x = nd.random.uniform(shape=(2, 4))
+    y1 = net(x)
+    f = str(tmp_path / "net.params")
+    net.save_parameters(f)
+    net2 = nn.HybridSequential()
+    net2.add(nn.Dense(5, activation="relu"), nn.Dense(3))
+    net2.load_parameters(f)
+    assert_almost_equal(net2(x).asnumpy(), y1.asnumpy())
+
+
+def test_constant_param():
+    class Net(gluon.HybridBlock):
+        def __init__(self):
+            super().__init__()
+            self.c = self.params.get_constant("c", nd.array([1.0, 2.0]))
+
+        def hybrid_forward(self, F, x, c):
+            return x * c
+
+    net = Net()
+    net.initialize()
+    out = net(nd.ones((2,)))
+    assert_almost_equal(out.asnumpy(), [1, 2])
+
+
+def test_grad_req_null():
+    net = nn.Dense(2, in_units=2)
+    net.initialize()
+    net.weight.grad_req = "null"
+    x = nd.ones((1, 2))
+    with autograd.record():
+        y = net(x).sum()
+    y.backward()
+    assert net.bias.grad(mx.cpu()) is not None
+
+
+def test_lstm_cell():
+    from anand_mxnet_amd.gluon import rnn
+    cell = rnn.LSTMCell(10)
+    cell.initialize()
+    x = nd.random.uniform(shape=(4, 6))
+    states = cell.begin_state(4)
+    out, new_states = cell(x, states)
+    assert out.shape == (4, 10)
+    assert len(new_states) == 2
+
+
+def test_rnn_layer_unroll():
+    from anand_mxnet_amd.gluon import rnn
+    lstm = rnn.LSTM(8, num_layers=2, layout="NTC")
+    lstm.initialize()
+    x = nd.random.uniform(shape=(3, 5, 4))  # N,T,C
+    out = lstm(x)
+    assert out.shape == (3, 5, 8)
+
+
+def test_split_and_load():
+    from anand_mxnet_amd.gluon.utils import split_and_load
+    data = nd.ones((8, 2))
+    parts = split_and_load(data, [mx.cpu(), mx.cpu()])
+    assert len(parts) == 2 and parts[0].shape == (4, 2)
+
+
+def test_clip_global_norm():
+    from anand_mxnet_amd.gluon.utils import clip_global_norm
+    arrays = [nd.full((2,), 3.0), nd.full((2,), 4.0)]
+    norm = clip_global_norm(arrays, 1.0)
+    assert abs(norm - np.sqrt(9 * 2 + 16 * 2)) < 1e-4
+    total = sum((a.asnumpy() ** 2).sum() for a in arrays)
+    assert abs(np.sqrt(total) - 1.0) < 1e-3

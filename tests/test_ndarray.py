@@ -1,0 +1,161 @@
+import numpy as np
+import pytest
+
+import anand_mxnet_amd as mx
+from anand_mxnet_amd import nd
+from anand_mxnet_amd.test_utils import assert_almost_equal
+
+
+def test_creation():
+    a = nd.zeros((2, 3))
+    assert a.shape == (2, 3)
+    assert a.dtype == np.float32
+    b = nd.ones((4,), dtype="float64")
+    assert b.asnumpy().sum() == 4.0
+    c = nd.full((2, 2), 7)
+    assert c.asnumpy().tolist() == [[7, 7], [7, 7]]
+    d = nd.arange(0, 10, 2)
+    assert d.asnumpy().tolist() == [0, 2, 4, 6, 8]
+    e = nd.array(np.eye(3))
+    assert e.dtype == np.float32
+
+
+def test_arithmetic():
+    a = nd.array([[1.0, 2.0], [3.0, 4.0]])
+    b = nd.array([[10.0, 20.0], [30.0, 40.0]])
+    assert_almost_equal((a + b).asnumpy(), [[11, 22], [33, 44]])
+    assert_almost_equal((b - a).asnumpy(), [[9, 18], [27, 36]])
+    assert_almost_equal((a * 2).asnumpy(), [[2, 4], [6, 8]])
+    assert_almost_equal((2 * a).asnumpy(), [[2, 4], [6, 8]])
+    assert_almost_equal((b / a).asnumpy(), [[10, 10], [10, 10]])
+    assert_almost_equal((1.0 / a).asnumpy(), 1.0 / a.asnumpy())
+    assert_almost_equal((-a).asnumpy(), -a.asnumpy())
+    assert_almost_equal((a ** 2).asnumpy(), a.asnumpy() ** 2)
+
+
+def test_inplace():
+    a = nd.ones((2, 2))
+    a += 1
+    assert a.asnumpy().tolist() == [[2, 2], [2, 2]]
+    a *= 3
+    assert a.asnumpy().tolist() == [[6, 6], [6, 6]]
+
+
+def test_indexing():
+    a = nd.array(np.arange(12).reshape(3, 4))
+    assert a[1].asnumpy().tolist() == [4, 5, 6, 7]
+    assert a[1, 2].asscalar() == 6
+    assert a[0:2, 1:3].shape == (2, 2)
+    a[0] = 100.0
+    assert a[0].asnumpy().tolist() == [100] * 4
+    a[1, 1] = -1
+    assert a[1, 1].asscalar() == -1
+
+
+def test_reshape_transpose():
+    a = nd.array(np.arange(6).reshape(2, 3))
+    assert a.reshape(3, 2).shape == (3, 2)
+    assert a.reshape((-1,)).shape == (6,)
+    assert a.reshape(0, 3).shape == (2, 3)
+    assert a.T.shape == (3, 2)
+    assert a.transpose((1, 0)).shape == (3, 2)
+    assert a.expand_dims(0).shape == (1, 2, 3)
+    assert a.expand_dims(0).squeeze(0).shape == (2, 3)
+    assert a.flatten().shape == (2, 3)
+
+
+def test_reductions():
+    a = nd.array(np.arange(6, dtype=np.float32).reshape(2, 3))
+    assert a.sum().asscalar() == 15
+    assert_almost_equal(a.sum(axis=0).asnumpy(), [3, 5, 7])
+    assert_almost_equal(a.mean(axis=1).asnumpy(), [1, 4])
+    assert a.max().asscalar() == 5
+    assert a.min().asscalar() == 0
+    assert nd.sum(a, axis=1, keepdims=True).shape == (2, 1)
+    assert nd.sum(a, axis=0, exclude=True).shape == (2,)
+    assert abs(a.norm().asscalar() - np.sqrt((np.arange(6) ** 2).sum())) < 1e-5
+
+
+def test_broadcast_ops():
+    a = nd.ones((2, 1, 3))
+    b = nd.ones((1, 4, 3))
+    assert nd.broadcast_add(a, b).shape == (2, 4, 3)
+    assert nd.broadcast_maximum(a, b * 2).asnumpy().max() == 2
+    c = nd.broadcast_to(nd.ones((1, 3)), (5, 3))
+    assert c.shape == (5, 3)
+
+
+def test_concat_split_stack():
+    a = nd.ones((2, 3))
+    b = nd.zeros((2, 3))
+    c = nd.concat(a, b, dim=0)
+    assert c.shape == (4, 3)
+    parts = nd.split(c, num_outputs=2, axis=0)
+    assert parts[0].shape == (2, 3)
+    s = nd.stack(a, b, axis=0)
+    assert s.shape == (2, 2, 3)
+
+
+def test_dot():
+    a = np.random.uniform(-1, 1, (4, 5)).astype(np.float32)
+    b = np.random.uniform(-1, 1, (5, 6)).astype(np.float32)
+    assert_almost_equal(nd.dot(nd.array(a), nd.array(b)).asnumpy(),
+                        a @ b, rtol=1e-4, atol=1e-5)
+    assert_almost_equal(
+        nd.dot(nd.array(a.T), nd.array(b), transpose_a=True).asnumpy(),
+        a @ b, rtol=1e-4, atol=1e-5)
+    assert_almost_equal(
+        nd.dot(nd.array(a), nd.array(b.T), transpose_b=True).asnumpy(),
+        a @ b, rtol=1e-4, atol=1e-5)
+
+
+def test_batch_dot():
+    a = np.random.uniform(-1, 1, (3, 4, 5)).astype(np.float32)
+    b = np.random.uniform(-1, 1, (3, 5, 6)).astype(np.float32)
+    assert_almost_equal(nd.batch_dot(nd.array(a), nd.array(b)).asnumpy(),
+                        a @ b, rtol=1e-4, atol=1e-5)
+
+
+def test_indexing_ops():
+    w = nd.array(np.arange(12).reshape(4, 3))
+    idx = nd.array([0, 2])
+    assert nd.take(w, idx).asnumpy().tolist() == [[0, 1, 2], [6, 7, 8]]
+    assert nd.Embedding(idx, w).shape == (2, 3)
+    data = nd.array([[0.1, 0.9], [0.8, 0.2]])
+    picked = nd.pick(data, nd.array([1, 0]))
+    assert_almost_equal(picked.asnumpy(), [0.9, 0.8])
+    oh = nd.one_hot(nd.array([0, 2]), 3)
+    assert oh.asnumpy().tolist() == [[1, 0, 0], [0, 0, 1]]
+
+
+def test_astype_copy_context():
+    a = nd.ones((2, 2))
+    b = a.astype("float64")
+    assert b.dtype == np.float64
+    c = a.copy()
+    c[0, 0] = 5
+    assert a[0, 0].asscalar() == 1
+    assert a.context == mx.cpu()
+    d = a.as_in_context(mx.cpu())
+    assert d is a
+
+
+def test_bf16_roundtrip():
+    a = nd.ones((2, 2), dtype="bfloat16")
+    assert a.dtype == "bfloat16"
+    assert a.asnumpy().dtype == np.float32
+    b = a.astype("float32") * 3
+    assert b.asnumpy().tolist() == [[3, 3], [3, 3]]
+
+
+def test_clip_where():
+    a = nd.array([-2.0, 0.5, 3.0])
+    assert nd.clip(a, -1, 1).asnumpy().tolist() == [-1, 0.5, 1]
+    w = nd.where(a > 0, nd.ones_like(a), nd.zeros_like(a))
+    assert w.asnumpy().tolist() == [0, 1, 1]
+
+
+def test_waitall():
+    nd.waitall()
+    a = nd.ones((4,))
+    a.wait_to_read()

@@ -1,0 +1,93 @@
+"""Golden-byte tests for the .params wire format (reference
+NDArray::Save/Load — survey hard requirement: byte-exact layout)."""
+import struct
+
+import numpy as np
+import pytest
+
+import anand_mxnet_amd as mx
+from anand_mxnet_amd import nd
+from anand_mxnet_amd.serializer import (MAGIC_LIST, NDARRAY_V2_MAGIC,
+                                        save_ndarrays, load_ndarrays)
+
+
+def test_file_header_bytes(tmp_path):
+    f = str(tmp_path / "x.params")
+    nd.save(f, {"w": nd.zeros((2, 3))})
+    raw = open(f, "rb").read()
+    # uint64 list magic 0x112, uint64 reserved 0
+    assert struct.unpack("<Q", raw[0:8])[0] == 0x112
+    assert struct.unpack("<Q", raw[8:16])[0] == 0
+    # uint64 count
+    assert struct.unpack("<Q", raw[16:24])[0] == 1
+    # ndarray v2 magic
+    assert struct.unpack("<I", raw[24:28])[0] == 0xF993FAC9
+    # stype dense = 0
+    assert struct.unpack("<i", raw[28:32])[0] == 0
+    # ndim=2, dims 2,3
+    assert struct.unpack("<III", raw[32:44]) == (2, 2, 3)
+    # ctx cpu(1,0), dtype f32 (0)
+    assert struct.unpack("<iii", raw[44:56]) == (1, 0, 0)
+    # 24 bytes of zeros data
+    assert raw[56:80] == b"\x00" * 24
+
+
+def test_roundtrip_dict(tmp_path):
+    f = str(tmp_path / "d.params")
+    arrays = {
+        "a": nd.array(np.random.uniform(-1, 1, (3, 4)).astype(np.float32)),
+        "b": nd.array(np.arange(5).astype(np.int64), dtype="int64"),
+        "c": nd.ones((2,), dtype="float16"),
+        "arg:fc_weight": nd.array(np.random.randn(4, 4).astype(np.float64),
+                                  dtype="float64"),
+    }
+    save_ndarrays(f, arrays)
+    loaded = load_ndarrays(f)
+    assert set(loaded.keys()) == set(arrays.keys())
+    for k in arrays:
+        np.testing.assert_allclose(loaded[k].asnumpy(), arrays[k].asnumpy())
+        assert loaded[k].dtype == arrays[k].dtype
+
+
+def test_roundtrip_list(tmp_path):
+    f = str(tmp_path / "l.params")
+    arrays = [nd.ones((2, 2)), nd.zeros((3,))]
+    save_ndarrays(f, arrays)
+    loaded = load_ndarrays(f)
+    assert isinstance(loaded, list) and len(loaded) == 2
+    np.testing.assert_allclose(loaded[0].asnumpy(), np.ones((2, 2)))
+
+
+def test_roundtrip_bf16(tmp_path):
+    f = str(tmp_path / "bf.params")
+    a = nd.array(np.random.uniform(-2, 2, (4, 4)).astype(np.float32))
+    a16 = a.astype("bfloat16")
+    save_ndarrays(f, {"w": a16})
+    loaded = load_ndarrays(f)
+    assert loaded["w"].dtype == "bfloat16"
+    np.testing.assert_allclose(loaded["w"].asnumpy(), a16.asnumpy())
+
+
+def test_scalar_and_empty(tmp_path):
+    f = str(tmp_path / "s.params")
+    save_ndarrays(f, {"s": nd.array(np.float32(3.5).reshape(1))})
+    loaded = load_ndarrays(f)
+    assert loaded["s"].asscalar() == 3.5
+
+
+def test_checkpoint_prefixes(tmp_path):
+    from anand_mxnet_amd.model import save_checkpoint, load_checkpoint
+    prefix = str(tmp_path / "model")
+    arg = {"fc1_weight": nd.ones((2, 2))}
+    aux = {"bn_moving_mean": nd.zeros((2,))}
+    save_checkpoint(prefix, 3, None, arg, aux)
+    sym, a, x = load_checkpoint(prefix, 3)
+    assert "fc1_weight" in a and "bn_moving_mean" in x
+    np.testing.assert_allclose(a["fc1_weight"].asnumpy(), np.ones((2, 2)))
+
+
+def test_bad_magic(tmp_path):
+    f = str(tmp_path / "bad.params")
+    open(f, "wb").write(b"\x00" * 32)
+    with pytest.raises(mx.MXNetError):
+        load_ndarrays(f)
