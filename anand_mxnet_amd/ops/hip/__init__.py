@@ -1,0 +1,503 @@
+"""GPU dispatch: hand-written CDNA4 kernels wrapped as torch.autograd
+Functions and registered into the op registry (ops.registry.GPU_IMPL).
+
+Contract: these run for bf16 tensors on gpu(i).  fp32 GPU tensors use the
+torch-ROCm substrate (documented in docs/PARITY.md); bf16 is the
+performance dtype of every target config.
+"""
+from __future__ import annotations
+
+import torch
+
+from ..registry import register_gpu
+from ...base import lib
+
+_C = lib()
+if _C is None:  # pragma: no cover
+    raise ImportError("native extension not built")
+
+_seed_counter = [12345]
+
+
+def _next_seed():
+    _seed_counter[0] = (_seed_counter[0] * 6364136223846793005 + 1442695040888963407) % (1 << 63)
+    return _seed_counter[0]
+
+
+# ---------------------------------------------------------------------------
+# GEMM / linear
+# ---------------------------------------------------------------------------
+
+class _GemmTN(torch.autograd.Function):
+    """C = A @ B^T over the MFMA TN kernel; A [.., M, K], B [.., N, K]."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        ctx.save_for_backward(a, b)
+        return _C.gemm_tn(a.contiguous(), b.contiguous(), None, False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, b = ctx.saved_tensors
+        dy = dy.contiguous()
+        da = db = None
+        if ctx.needs_input_grad[0]:
+            # dA[M,K] = dY[M,N] @ B[N,K] -> TN with B^T[K,N]
+            da = _C.gemm_tn(dy, b.transpose(-1, -2).contiguous(), None,
+                            False)
+        if ctx.needs_input_grad[1]:
+            if a.dim() == 2:
+                db = _C.gemm_wg(dy, a).to(a.dtype)  # [N,K]
+            else:
+                db = _C.gemm_tn(dy.transpose(-1, -2).contiguous(),
+                                a.transpose(-1, -2).contiguous(), None,
+                                False)
+        return da, db
+
+
+@register_gpu("gemm")
+def gemm(a, b, transpose_a, transpose_b):
+    A = a.transpose(-1, -2) if transpose_a else a
+    Bt = b if transpose_b else b.transpose(-1, -2)
+    if A.size(-1) % 64 != 0:
+        # reduce-dim not tileable: fall back to library GEMM (plain GEMMs
+        # may use rocBLAS; the tiled hot paths are all %64)
+        return torch.matmul(A, Bt.transpose(-1, -2))
+    return _GemmTN.apply(A.contiguous(), Bt.contiguous())
+
+
+@register_gpu("batch_gemm")
+def batch_gemm(a, b, transpose_a, transpose_b):
+    A = a.transpose(-1, -2) if transpose_a else a
+    Bt = b if transpose_b else b.transpose(-1, -2)
+    if A.size(-1) % 64 != 0:
+        return torch.matmul(A, Bt.transpose(-1, -2))
+    return _GemmTN.apply(A.contiguous(), Bt.contiguous())
+
+
+class _Linear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        return _C.gemm_tn(x.contiguous(), w.contiguous(),
+                          bias.contiguous() if bias is not None else None,
+                          False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = _C.gemm_tn(dy, w.t().contiguous(), None, False)
+        if ctx.needs_input_grad[1]:
+            dw = _C.gemm_wg(dy, x).to(w.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy.float().sum(0)
+        return dx, dw, db
+
+
+@register_gpu("linear")
+def linear(x, w, bias):
+    if x.size(-1) % 64 != 0 or (x.dim() == 2 and x.size(0) % 8 != 0):
+        return torch.nn.functional.linear(x, w, bias)
+    x2 = x.reshape(-1, x.shape[-1])
+    out = _Linear.apply(x2.contiguous(), w,
+                        bias if bias is None else bias.to(x.dtype))
+    return out.reshape(x.shape[:-1] + (w.shape[0],))
+
+
+# ---------------------------------------------------------------------------
+# Convolution (NHWC bf16)
+# ---------------------------------------------------------------------------
+
+class _Conv2dNHWC(torch.autograd.Function):
+    """Implicit-GEMM conv; weight logical layout (K, R, S, C)."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pad, dil):
+        x = x.contiguous()
+        K, R, S, C = w.shape
+        w_rskc = w.permute(1, 2, 0, 3).contiguous()
+        ctx.save_for_backward(x, w)
+        ctx.conf = (stride, pad, dil)
+        y = _C.conv2d_fwd(x, w_rskc,
+                          bias.float() if bias is not None else None,
+                          list(stride), list(pad), list(dil), False)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        stride, pad, dil = ctx.conf
+        dy = dy.contiguous()
+        K, R, S, C = w.shape
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            w_rsck = w.permute(1, 2, 3, 0).contiguous()
+            dx = _C.conv2d_dgrad(dy, w_rsck, list(x.shape), list(stride),
+                                 list(pad), list(dil))
+        if ctx.needs_input_grad[1]:
+            dwf = _C.conv2d_wgrad(x, dy, [R, S], list(stride), list(pad),
+                                  list(dil))  # [R,S,C,K] f32
+            dw = dwf.permute(3, 0, 1, 2).to(w.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy.float().sum(dim=(0, 1, 2))
+        return dx, dw, db, None, None, None
+
+
+class _Conv2dIm2col(torch.autograd.Function):
+    """Small-C path (the 7x7 C=3 stem): materialized im2col + TN/WG GEMM."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pad, dil):
+        K, R, S, C = w.shape
+        N, H, W_, _ = x.shape
+        sh, sw = stride
+        ph, pw = pad
+        P = (H + 2 * ph - (R - 1) * dil[0] - 1) // sh + 1
+        Q = (W_ + 2 * pw - (S - 1) * dil[1] - 1) // sw + 1
+        xp = torch.nn.functional.pad(x, (0, 0, pw, pw, ph, ph))
+        dev = x.device
+        pi = torch.arange(P, device=dev) * sh
+        qi = torch.arange(Q, device=dev) * sw
+        ri = torch.arange(R, device=dev) * dil[0]
+        si = torch.arange(S, device=dev) * dil[1]
+        IH = (pi[:, None, None, None] + ri[None, None, :, None])
+        IW = (qi[None, :, None, None] + si[None, None, None, :])
+        IH = IH.expand(P, Q, R, S)
+        IW = IW.expand(P, Q, R, S)
+        cols = xp[:, IH, IW, :]                       # [N,P,Q,R,S,C]
+        A = cols.reshape(N * P * Q, R * S * C)
+        kdim = R * S * C
+        kpad = (64 - kdim % 64) % 64
+        if kpad:
+            A = torch.nn.functional.pad(A, (0, kpad))
+        B = w.reshape(K, kdim)
+        if kpad:
+            B = torch.nn.functional.pad(B, (0, kpad))
+        A = A.contiguous()
+        ctx.save_for_backward(A, w)
+        ctx.dims = (N, P, Q, K, kdim)
+        ctx.has_bias = bias is not None
+        y = _C.gemm_tn(A, B.contiguous(),
+                       bias.float() if bias is not None else None, False)
+        return y.reshape(N, P, Q, K)
+
+    @staticmethod
+    def backward(ctx, dy):
+        A, w = ctx.saved_tensors
+        N, P, Q, K, kdim = ctx.dims
+        dy2 = dy.reshape(N * P * Q, K).contiguous()
+        dw = db = None
+        if ctx.needs_input_grad[1]:
+            dwf = _C.gemm_wg(dy2, A)       # [K, kdim_padded] f32
+            dw = dwf[:, :kdim].reshape(w.shape).to(w.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy2.float().sum(0)
+        if ctx.needs_input_grad[0]:
+            raise RuntimeError(
+                "im2col conv path has no dgrad (stem layer input needs no "
+                "gradient); use C%64 shapes for inner convs")
+        return None, dw, db, None, None, None
+
+
+@register_gpu("conv2d_nhwc")
+def conv2d_nhwc(x, w, bias, stride, pad, dil, groups):
+    if groups != 1:
+        raise NotImplementedError(
+            "grouped/depthwise conv HIP kernel not implemented yet "
+            "(docs/PARITY.md)")
+    stride, pad, dil = tuple(stride), tuple(pad), tuple(dil)
+    if w.shape[3] % 64 == 0:
+        return _Conv2dNHWC.apply(x, w, bias, stride, pad, dil)
+    return _Conv2dIm2col.apply(x, w, bias, stride, pad, dil)
+
+
+# ---------------------------------------------------------------------------
+# BatchNorm (NHWC)
+# ---------------------------------------------------------------------------
+
+class _BatchNormNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, rmean, rvar, eps, momentum, training,
+                fuse_relu):
+        x = x.contiguous()
+        g32 = gamma.float().contiguous()
+        b32 = beta.float().contiguous()
+        if training:
+            y, save_mean, save_rstd = _C.bn_fwd_train(
+                x, g32, b32, rmean, rvar, eps, momentum, fuse_relu)
+            ctx.save_for_backward(x, g32, b32, save_mean, save_rstd)
+            ctx.relu = fuse_relu
+            ctx.gdtype = gamma.dtype
+            return y
+        ctx.inference = True
+        return _C.bn_fwd_infer(x, g32, b32, rmean.contiguous(),
+                               rvar.contiguous(), eps, fuse_relu)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if getattr(ctx, "inference", False):
+            raise RuntimeError("BatchNorm inference backward not supported")
+        x, g32, b32, save_mean, save_rstd = ctx.saved_tensors
+        dx, dgamma, dbeta = _C.bn_bwd(x, dy.contiguous(), g32, b32,
+                                      save_mean, save_rstd, ctx.relu)
+        return (dx, dgamma.to(ctx.gdtype), dbeta.to(ctx.gdtype), None, None,
+                None, None, None, None)
+
+
+@register_gpu("batchnorm_nhwc")
+def batchnorm_nhwc(x, gamma, beta, rmean, rvar, eps, momentum, training,
+                   fuse_relu):
+    if x.dtype != torch.bfloat16 or x.shape[-1] % 8 != 0:
+        # torch path for non-native dtype/shape
+        out = torch.nn.functional.batch_norm(
+            x.permute(0, 3, 1, 2), rmean, rvar, gamma, beta, training,
+            1.0 - momentum if training else 0.0, eps)
+        if fuse_relu:
+            out = torch.relu(out)
+        return out.permute(0, 2, 3, 1).contiguous()
+    return _BatchNormNHWC.apply(x, gamma, beta, rmean, rvar, eps, momentum,
+                                training, fuse_relu)
+
+
+# ---------------------------------------------------------------------------
+# Pooling (NHWC)
+# ---------------------------------------------------------------------------
+
+class _MaxPoolNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, pad):
+        x = x.contiguous()
+        y, arg = _C.maxpool(x, list(kernel), list(stride), list(pad))
+        ctx.save_for_backward(arg)
+        ctx.conf = (tuple(x.shape), kernel, stride, pad)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (arg,) = ctx.saved_tensors
+        xshape, kernel, stride, pad = ctx.conf
+        dx = _C.maxpool_bwd(dy.contiguous(), arg, list(xshape),
+                            list(kernel), list(stride), list(pad))
+        return dx, None, None, None
+
+
+class _GlobalAvgPoolNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.xshape = tuple(x.shape)
+        return _C.global_avgpool(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return _C.global_avgpool_bwd(dy.contiguous(), list(ctx.xshape))
+
+
+@register_gpu("pool_nhwc")
+def pool_nhwc(x, kernel, pool_type, global_pool, stride, pad,
+              count_include_pad):
+    if x.dtype != torch.bfloat16 or x.shape[-1] % 8 != 0:
+        import torch.nn.functional as F
+        xc = x.permute(0, 3, 1, 2)
+        if global_pool:
+            out = (F.adaptive_max_pool2d(xc, 1) if pool_type == "max"
+                   else F.adaptive_avg_pool2d(xc, 1))
+        elif pool_type == "max":
+            out = F.max_pool2d(xc, tuple(kernel), tuple(stride or kernel),
+                               tuple(pad))
+        else:
+            out = F.avg_pool2d(xc, tuple(kernel), tuple(stride or kernel),
+                               tuple(pad),
+                               count_include_pad=count_include_pad)
+        return out.permute(0, 2, 3, 1).contiguous()
+    if global_pool:
+        if pool_type == "avg":
+            return _GlobalAvgPoolNHWC.apply(x)
+        kernel = (x.shape[1], x.shape[2])
+        stride, pad = (1, 1), (0, 0)
+    stride = tuple(stride) if stride is not None else tuple(kernel)
+    if pool_type == "max":
+        return _MaxPoolNHWC.apply(x, tuple(kernel), stride, tuple(pad))
+    # average pooling: no stored state needed; use autograd-free fwd +
+    # custom bwd via the gap/avg kernels
+    class _Avg(torch.autograd.Function):
+        @staticmethod
+        def forward(ctx2, xx):
+            ctx2.conf = (tuple(xx.shape), tuple(kernel), stride, tuple(pad),
+                         count_include_pad)
+            return _C.avgpool(xx.contiguous(), list(kernel), list(stride),
+                              list(pad), count_include_pad)
+
+        @staticmethod
+        def backward(ctx2, dy):
+            import torch.nn.functional as F
+            xshape, kern, strd, pd, cip = ctx2.conf
+            # exact bwd via torch avg_pool grad on permuted layout
+            dxc = torch.ops.aten.avg_pool2d_backward(
+                dy.permute(0, 3, 1, 2), torch.empty(
+                    (xshape[0], xshape[3], xshape[1], xshape[2]),
+                    device=dy.device, dtype=dy.dtype), list(kern),
+                list(strd), list(pd), False, cip, None)
+            return dxc.permute(0, 2, 3, 1).contiguous()
+    return _Avg.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# softmax / layernorm / dropout / embedding / activations
+# ---------------------------------------------------------------------------
+
+class _Softmax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, log_):
+        y = _C.softmax_fwd(x.contiguous(), log_)
+        ctx.save_for_backward(y)
+        ctx.log = log_
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        return _C.softmax_bwd(y, dy.contiguous(), ctx.log), None
+
+
+@register_gpu("softmax")
+def softmax(x):
+    if x.dtype != torch.bfloat16 or x.shape[-1] % 8 != 0:
+        return torch.nn.functional.softmax(x.float(), dim=-1).to(x.dtype)
+    return _Softmax.apply(x, False)
+
+
+@register_gpu("log_softmax")
+def log_softmax(x):
+    if x.dtype != torch.bfloat16 or x.shape[-1] % 8 != 0:
+        return torch.nn.functional.log_softmax(x.float(), dim=-1).to(x.dtype)
+    return _Softmax.apply(x, True)
+
+
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        x = x.contiguous()
+        g32 = gamma.float().contiguous()
+        y, mean, rstd = _C.layernorm_fwd(x, g32, beta.float().contiguous(),
+                                         eps)
+        ctx.save_for_backward(x, g32, mean, rstd)
+        ctx.gdtype = gamma.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, g32, mean, rstd = ctx.saved_tensors
+        dx, dgamma, dbeta = _C.layernorm_bwd(x, dy.contiguous(), g32, mean,
+                                             rstd)
+        return dx, dgamma.to(ctx.gdtype), dbeta.to(ctx.gdtype), None
+
+
+@register_gpu("layernorm")
+def layernorm(x, gamma, beta, eps):
+    if x.dtype != torch.bfloat16 or x.shape[-1] % 8 != 0:
+        return torch.nn.functional.layer_norm(x, (x.shape[-1],), gamma,
+                                              beta, eps)
+    return _LayerNorm.apply(x, gamma, beta, eps)
+
+
+class _Dropout(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p):
+        y, mask = _C.dropout_fwd(x.contiguous(), p, _next_seed())
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        return _C.dropout_bwd(dy.contiguous(), mask, ctx.p), None
+
+
+@register_gpu("dropout")
+def dropout(x, p):
+    if x.dtype != torch.bfloat16 or x.numel() % 8 != 0:
+        return torch.nn.functional.dropout(x, p, training=True)
+    return _Dropout.apply(x, p)
+
+
+class _Embedding(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, idx, W):
+        ctx.save_for_backward(idx)
+        ctx.vocab = W.shape[0]
+        ctx.wdtype = W.dtype
+        return _C.embedding_fwd(idx, W.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        dW = _C.embedding_bwd(idx, dy.contiguous(), ctx.vocab)
+        return None, dW.to(ctx.wdtype)
+
+
+@register_gpu("embedding")
+def embedding(idx, W):
+    if W.dtype != torch.bfloat16 or W.shape[1] % 8 != 0:
+        return torch.nn.functional.embedding(idx.long(), W)
+    return _Embedding.apply(idx, W)
+
+
+_ACT_IDS = {"relu": 0, "sigmoid": 1, "tanh": 2, "gelu": 3, "silu": 4,
+            "gelu_tanh": 5}
+
+
+class _Act(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, op):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        ctx.op = op
+        return _C.act_fwd(op, x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        return _C.act_bwd(ctx.op, x, dy.contiguous()), None
+
+
+def _make_act(name, opid):
+    def fn(x):
+        if x.dtype != torch.bfloat16 or x.numel() % 8 != 0:
+            return {"relu": torch.relu, "sigmoid": torch.sigmoid,
+                    "tanh": torch.tanh,
+                    "gelu": torch.nn.functional.gelu,
+                    "silu": torch.nn.functional.silu,
+                    "gelu_tanh": lambda t: torch.nn.functional.gelu(
+                        t, approximate="tanh")}[name](x)
+        return _Act.apply(x, opid)
+    register_gpu("act_" + name)(fn)
+    return fn
+
+
+for _n, _i in _ACT_IDS.items():
+    _make_act(_n, _i)
+
+
+# ---------------------------------------------------------------------------
+# fused multi-tensor optimizer hooks (used by gluon.Trainer fused path)
+# ---------------------------------------------------------------------------
+
+def build_opt_descs(ws, gs, ms, vs, w32s):
+    return _C.build_opt_descs(ws, gs, ms, vs, w32s)
+
+
+def multi_opt_run(blob, ntens, nchunks, mode, wbf16, lr, momentum, beta1,
+                  beta2, eps, wd, rescale, clip):
+    _C.multi_opt_run(blob, ntens, nchunks, mode, wbf16, lr, momentum, beta1,
+                     beta2, eps, wd, rescale, clip)
+
+
+def multi_sumsq(blob, ntens, nchunks, bf16in):
+    return _C.multi_sumsq_run(blob, ntens, nchunks, bf16in)

@@ -1,0 +1,555 @@
+// Python bindings for the MI355X-native kernel library (_C extension).
+// Plays the role of the reference's flat C ABI (src/c_api/) — here the
+// native boundary is a torch extension: tensors in, stream-ordered HIP
+// kernel launches out, zero Python in the hot loop.
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <unordered_map>
+#include <vector>
+
+typedef __bf16 bf16;
+
+namespace mxa {
+// gemm_tn.hip
+void gemm_tn_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
+                  int, long, long, long, long, long, long, int, bool,
+                  const bf16*, hipStream_t);
+void conv2d_fwd_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
+                     int, int, int, int, int, int, int, int, int, int, int,
+                     int, int, bool, const bf16*, hipStream_t);
+void conv2d_dgrad_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
+                       int, int, int, int, int, int, int, int, int, int,
+                       int, const bf16*, hipStream_t);
+// gemm_wg.hip
+void gemm_wg_bf16(const bf16*, const bf16*, float*, long, int, int, long,
+                  long, long, const bf16*, hipStream_t);
+void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, int, int, int, int,
+                       int, int, int, int, int, int, int, int, int, int,
+                       int, const bf16*, hipStream_t);
+// bn_pool.hip
+void bn_fwd_train(const bf16*, bf16*, float*, float*, float*, float*, float*,
+                  float*, const float*, const float*, long, int, float,
+                  float, bool, hipStream_t);
+void bn_fwd_infer(const bf16*, bf16*, const float*, const float*,
+                  const float*, const float*, long, int, float, bool,
+                  hipStream_t);
+void bn_bwd(const bf16*, const bf16*, bf16*, const float*, const float*,
+            const float*, const float*, float*, float*, long, int, bool,
+            hipStream_t);
+void maxpool_nhwc(const bf16*, bf16*, unsigned char*, int, int, int, int,
+                  int, int, int, int, int, int, int, int, hipStream_t);
+void maxpool_nhwc_bwd(const bf16*, const unsigned char*, float*, int, int,
+                      int, int, int, int, int, int, int, int, int, int,
+                      hipStream_t);
+void avgpool_nhwc(const bf16*, bf16*, int, int, int, int, int, int, int,
+                  int, int, int, int, int, bool, hipStream_t);
+void gap_nhwc(const bf16*, bf16*, int, long, int, hipStream_t);
+void gap_nhwc_bwd(const bf16*, bf16*, int, long, int, hipStream_t);
+// softmax_norm.hip
+void softmax_fwd_bf16(const bf16*, bf16*, long, int, bool, hipStream_t);
+void softmax_bwd_bf16(const bf16*, const bf16*, bf16*, long, int, bool,
+                      hipStream_t);
+void layernorm_fwd_bf16(const bf16*, bf16*, const float*, const float*,
+                        float*, float*, long, int, float, hipStream_t);
+void layernorm_bwd_bf16(const bf16*, const bf16*, bf16*, const float*,
+                        const float*, const float*, float*, float*, long,
+                        int, hipStream_t);
+// misc.hip
+void dropout_fwd_bf16(const bf16*, bf16*, unsigned char*, long,
+                      unsigned long long, float, hipStream_t);
+void dropout_bwd_bf16(const bf16*, const unsigned char*, bf16*, long, float,
+                      hipStream_t);
+void embedding_fwd_bf16(const long*, const bf16*, bf16*, long, int, long,
+                        hipStream_t);
+void embedding_bwd_bf16(const long*, const bf16*, float*, long, int, long,
+                        hipStream_t);
+void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
+void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
+// optim.hip
+struct TensorDesc {
+  void* w;
+  void* g;
+  float* m;
+  float* v;
+  float* w32;
+  long n;
+  long chunk0;
+};
+void multi_opt(const TensorDesc*, int, long, int, bool, float, float, float,
+               float, float, float, float, float, hipStream_t);
+void multi_sumsq(const TensorDesc*, int, long, bool, float*, hipStream_t);
+}  // namespace mxa
+
+namespace {
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+const bf16* bfp(const at::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, "expected bf16 tensor");
+  return reinterpret_cast<const bf16*>(t.data_ptr());
+}
+bf16* bfp_mut(at::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, "expected bf16 tensor");
+  return reinterpret_cast<bf16*>(t.data_ptr());
+}
+
+// 128-byte zero page per device (OOB rows of tile staging read this)
+at::Tensor& zpage_for(const at::Tensor& like) {
+  static std::unordered_map<int, at::Tensor> cache;
+  int dev = like.get_device();
+  auto it = cache.find(dev);
+  if (it == cache.end()) {
+    auto z = at::zeros({128}, like.options().dtype(at::kBFloat16));
+    it = cache.emplace(dev, z).first;
+  }
+  return it->second;
+}
+
+const bf16* zp(const at::Tensor& like) { return bfp(zpage_for(like)); }
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// GEMM / linear
+// ---------------------------------------------------------------------------
+
+// C[M,N] = A[M,K] @ B[N,K]^T (+bias) (+relu); 3-d inputs batch over dim 0.
+at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
+                   c10::optional<at::Tensor> bias, bool relu) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous(), "gemm_tn: contiguous");
+  bool batched = A.dim() == 3;
+  long M = A.size(-2), K = A.size(-1), N = B.size(-2);
+  TORCH_CHECK(B.size(-1) == K, "gemm_tn: K mismatch");
+  TORCH_CHECK(K % 64 == 0, "gemm_tn: K must be a multiple of 64, got ", K);
+  int batches = batched ? A.size(0) : 1;
+  auto out = batched ? at::empty({batches, M, N}, A.options())
+                     : at::empty({M, N}, A.options());
+  const float* bptr = nullptr;
+  at::Tensor bias_f;
+  if (bias.has_value() && bias->defined()) {
+    bias_f = bias->to(at::kFloat).contiguous();
+    bptr = bias_f.data_ptr<float>();
+  }
+  long sA = batched ? M * K : 0;
+  long sB = (batched && B.dim() == 3) ? N * K : 0;
+  long sC = batched ? M * N : 0;
+  mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
+                    (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
+                    cur_stream());
+  return out;
+}
+
+// C[Mo,No] (f32) = sum_m A[m,Mo] * B[m,No]
+at::Tensor gemm_wg(const at::Tensor& A, const at::Tensor& B) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous(), "gemm_wg: contiguous");
+  long m = A.size(0), Mo = A.size(1), No = B.size(1);
+  TORCH_CHECK(B.size(0) == m, "gemm_wg: reduce dim mismatch");
+  TORCH_CHECK(Mo % 8 == 0 && No % 8 == 0, "gemm_wg: dims must be %8");
+  auto out = at::empty({Mo, No}, A.options().dtype(at::kFloat));
+  mxa::gemm_wg_bf16(bfp(A), bfp(B), out.data_ptr<float>(), m, (int)Mo,
+                    (int)No, Mo, No, No, zp(A), cur_stream());
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// Convolution (NHWC)
+// ---------------------------------------------------------------------------
+
+at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w_rskc,
+                      c10::optional<at::Tensor> bias,
+                      std::vector<long> stride, std::vector<long> pad,
+                      std::vector<long> dil, bool relu) {
+  TORCH_CHECK(x.is_contiguous() && w_rskc.is_contiguous());
+  int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int R = w_rskc.size(0), S = w_rskc.size(1), Kout = w_rskc.size(2);
+  TORCH_CHECK(w_rskc.size(3) == C, "conv2d_fwd: channel mismatch");
+  TORCH_CHECK(C % 64 == 0, "conv2d_fwd: C must be %64 (stem uses im2col)");
+  int sh = stride[0], sw = stride[1], ph = pad[0], pw = pad[1],
+      dh = dil[0], dw = dil[1];
+  int P = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
+  int Q = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
+  auto y = at::empty({Nb, P, Q, Kout}, x.options());
+  const float* bptr = nullptr;
+  at::Tensor bias_f;
+  if (bias.has_value() && bias->defined()) {
+    bias_f = bias->to(at::kFloat).contiguous();
+    bptr = bias_f.data_ptr<float>();
+  }
+  mxa::conv2d_fwd_bf16(bfp(x), bfp(w_rskc), bfp_mut(y), bptr, Nb, H, W, C,
+                       Kout, R, S, P, Q, sh, sw, ph, pw, dh, dw, relu,
+                       zp(x), cur_stream());
+  return y;
+}
+
+at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w_rsck,
+                        std::vector<long> xshape, std::vector<long> stride,
+                        std::vector<long> pad, std::vector<long> dil) {
+  TORCH_CHECK(dy.is_contiguous() && w_rsck.is_contiguous());
+  int Nb = xshape[0], H = xshape[1], W = xshape[2], C = xshape[3];
+  int R = w_rsck.size(0), S = w_rsck.size(1);
+  TORCH_CHECK(w_rsck.size(2) == C, "conv2d_dgrad: channel mismatch");
+  int Kout = w_rsck.size(3);
+  TORCH_CHECK(Kout % 64 == 0, "conv2d_dgrad: K must be %64");
+  int P = dy.size(1), Q = dy.size(2);
+  auto dx = at::empty({Nb, H, W, C}, dy.options());
+  mxa::conv2d_dgrad_bf16(bfp(dy), bfp(w_rsck), bfp_mut(dx), Nb, H, W, C,
+                         Kout, R, S, P, Q, (int)stride[0], (int)stride[1],
+                         (int)pad[0], (int)pad[1], (int)dil[0], (int)dil[1],
+                         zp(dy), cur_stream());
+  return dx;
+}
+
+at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy,
+                        std::vector<long> kernel, std::vector<long> stride,
+                        std::vector<long> pad, std::vector<long> dil) {
+  TORCH_CHECK(x.is_contiguous() && dy.is_contiguous());
+  int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
+  int R = kernel[0], S = kernel[1];
+  auto dw = at::empty({R, S, C, Kout}, x.options().dtype(at::kFloat));
+  mxa::conv2d_wgrad_bf16(bfp(x), bfp(dy), dw.data_ptr<float>(), Nb, H, W, C,
+                         Kout, R, S, P, Q, (int)stride[0], (int)stride[1],
+                         (int)pad[0], (int)pad[1], (int)dil[0], (int)dil[1],
+                         zp(x), cur_stream());
+  return dw;
+}
+
+// ---------------------------------------------------------------------------
+// BatchNorm (NHWC) / pooling
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
+                                     const at::Tensor& gamma,
+                                     const at::Tensor& beta,
+                                     at::Tensor running_mean,
+                                     at::Tensor running_var, double eps,
+                                     double momentum, bool relu) {
+  int C = x.size(-1);
+  long M = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto psum = at::empty({C}, opts), psumsq = at::empty({C}, opts);
+  auto save_mean = at::empty({C}, opts), save_rstd = at::empty({C}, opts);
+  mxa::bn_fwd_train(bfp(x), bfp_mut(y), psum.data_ptr<float>(),
+                    psumsq.data_ptr<float>(), save_mean.data_ptr<float>(),
+                    save_rstd.data_ptr<float>(),
+                    running_mean.data_ptr<float>(),
+                    running_var.data_ptr<float>(), gamma.data_ptr<float>(),
+                    beta.data_ptr<float>(), M, C, (float)eps,
+                    (float)momentum, relu, cur_stream());
+  return {y, save_mean, save_rstd};
+}
+
+at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
+                        const at::Tensor& beta, const at::Tensor& rmean,
+                        const at::Tensor& rvar, double eps, bool relu) {
+  int C = x.size(-1);
+  long M = x.numel() / C;
+  auto y = at::empty_like(x);
+  mxa::bn_fwd_infer(bfp(x), bfp_mut(y), rmean.data_ptr<float>(),
+                    rvar.data_ptr<float>(), gamma.data_ptr<float>(),
+                    beta.data_ptr<float>(), M, C, (float)eps, relu,
+                    cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
+                               const at::Tensor& gamma,
+                               const at::Tensor& beta,
+                               const at::Tensor& save_mean,
+                               const at::Tensor& save_rstd, bool relu) {
+  int C = x.size(-1);
+  long M = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto dgamma = at::empty({C}, opts), dbeta = at::empty({C}, opts);
+  mxa::bn_bwd(bfp(x), bfp(dy), bfp_mut(dx), save_mean.data_ptr<float>(),
+              save_rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+              beta.data_ptr<float>(), dgamma.data_ptr<float>(),
+              dbeta.data_ptr<float>(), M, C, relu, cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+std::vector<at::Tensor> maxpool(const at::Tensor& x,
+                                std::vector<long> kernel,
+                                std::vector<long> stride,
+                                std::vector<long> pad) {
+  int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int R = kernel[0], S = kernel[1], sh = stride[0], sw = stride[1],
+      ph = pad[0], pw = pad[1];
+  int P = (H + 2 * ph - R) / sh + 1;
+  int Q = (W + 2 * pw - S) / sw + 1;
+  auto y = at::empty({Nb, P, Q, C}, x.options());
+  auto arg = at::empty({Nb, P, Q, C}, x.options().dtype(at::kByte));
+  mxa::maxpool_nhwc(bfp(x), bfp_mut(y), arg.data_ptr<unsigned char>(), Nb,
+                    H, W, C, P, Q, R, S, sh, sw, ph, pw, cur_stream());
+  return {y, arg};
+}
+
+at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& arg,
+                       std::vector<long> xshape, std::vector<long> kernel,
+                       std::vector<long> stride, std::vector<long> pad) {
+  int Nb = xshape[0], H = xshape[1], W = xshape[2], C = xshape[3];
+  int P = dy.size(1), Q = dy.size(2);
+  auto dxf = at::empty({Nb, H, W, C}, dy.options().dtype(at::kFloat));
+  mxa::maxpool_nhwc_bwd(bfp(dy), arg.data_ptr<unsigned char>(),
+                        dxf.data_ptr<float>(), Nb, H, W, C, P, Q,
+                        (int)kernel[0], (int)kernel[1], (int)stride[0],
+                        (int)stride[1], (int)pad[0], (int)pad[1],
+                        cur_stream());
+  return dxf.to(at::kBFloat16);
+}
+
+at::Tensor avgpool(const at::Tensor& x, std::vector<long> kernel,
+                   std::vector<long> stride, std::vector<long> pad,
+                   bool count_pad) {
+  int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int R = kernel[0], S = kernel[1], sh = stride[0], sw = stride[1],
+      ph = pad[0], pw = pad[1];
+  int P = (H + 2 * ph - R) / sh + 1;
+  int Q = (W + 2 * pw - S) / sw + 1;
+  auto y = at::empty({Nb, P, Q, C}, x.options());
+  mxa::avgpool_nhwc(bfp(x), bfp_mut(y), Nb, H, W, C, P, Q, R, S, sh, sw,
+                    ph, pw, count_pad, cur_stream());
+  return y;
+}
+
+at::Tensor global_avgpool(const at::Tensor& x) {
+  int Nb = x.size(0), C = x.size(3);
+  long HW = (long)x.size(1) * x.size(2);
+  auto y = at::empty({Nb, 1, 1, C}, x.options());
+  mxa::gap_nhwc(bfp(x), bfp_mut(y), Nb, HW, C, cur_stream());
+  return y;
+}
+
+at::Tensor global_avgpool_bwd(const at::Tensor& dy,
+                              std::vector<long> xshape) {
+  int Nb = xshape[0], C = xshape[3];
+  long HW = (long)xshape[1] * xshape[2];
+  auto dx = at::empty({Nb, xshape[1], xshape[2], C}, dy.options());
+  mxa::gap_nhwc_bwd(bfp(dy), bfp_mut(dx), Nb, HW, C, cur_stream());
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+// softmax / layernorm / dropout / embedding / activations
+// ---------------------------------------------------------------------------
+
+at::Tensor softmax_fwd(const at::Tensor& x, bool log_) {
+  int D = x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "softmax: last dim must be %8");
+  long rows = x.numel() / D;
+  auto y = at::empty_like(x);
+  mxa::softmax_fwd_bf16(bfp(x), bfp_mut(y), rows, D, log_, cur_stream());
+  return y;
+}
+
+at::Tensor softmax_bwd(const at::Tensor& y, const at::Tensor& dy,
+                       bool log_) {
+  int D = y.size(-1);
+  long rows = y.numel() / D;
+  auto dx = at::empty_like(y);
+  mxa::softmax_bwd_bf16(bfp(y), bfp(dy), bfp_mut(dx), rows, D, log_,
+                        cur_stream());
+  return dx;
+}
+
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& beta, double eps) {
+  int D = x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "layernorm: last dim must be %8");
+  long rows = x.numel() / D;
+  auto y = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto mean = at::empty({rows}, opts), rstd = at::empty({rows}, opts);
+  mxa::layernorm_fwd_bf16(bfp(x), bfp_mut(y), gamma.data_ptr<float>(),
+                          beta.data_ptr<float>(), mean.data_ptr<float>(),
+                          rstd.data_ptr<float>(), rows, D, (float)eps,
+                          cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x,
+                                      const at::Tensor& dy,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& mean,
+                                      const at::Tensor& rstd) {
+  int D = x.size(-1);
+  long rows = x.numel() / D;
+  auto dx = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto dgamma = at::empty({D}, opts), dbeta = at::empty({D}, opts);
+  mxa::layernorm_bwd_bf16(bfp(x), bfp(dy), bfp_mut(dx),
+                          gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                          rstd.data_ptr<float>(), dgamma.data_ptr<float>(),
+                          dbeta.data_ptr<float>(), rows, D, cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+std::vector<at::Tensor> dropout_fwd(const at::Tensor& x, double p,
+                                    long seed) {
+  long n = x.numel();
+  TORCH_CHECK(n % 8 == 0, "dropout: numel must be %8");
+  auto y = at::empty_like(x);
+  auto mask = at::empty({n}, x.options().dtype(at::kByte));
+  mxa::dropout_fwd_bf16(bfp(x), bfp_mut(y), mask.data_ptr<unsigned char>(),
+                        n, (unsigned long long)seed, (float)p,
+                        cur_stream());
+  return {y, mask};
+}
+
+at::Tensor dropout_bwd(const at::Tensor& dy, const at::Tensor& mask,
+                       double p) {
+  long n = dy.numel();
+  auto dx = at::empty_like(dy);
+  mxa::dropout_bwd_bf16(bfp(dy), mask.data_ptr<unsigned char>(),
+                        bfp_mut(dx), n, (float)p, cur_stream());
+  return dx;
+}
+
+at::Tensor embedding_fwd(const at::Tensor& idx, const at::Tensor& W) {
+  auto idx_l = idx.to(at::kLong).contiguous();
+  long ntok = idx_l.numel();
+  int D = W.size(1);
+  TORCH_CHECK(D % 8 == 0, "embedding: dim must be %8");
+  auto sizes = idx.sizes().vec();
+  sizes.push_back(D);
+  auto out = at::empty(sizes, W.options());
+  mxa::embedding_fwd_bf16(idx_l.data_ptr<long>(), bfp(W), bfp_mut(out),
+                          ntok, D, W.size(0), cur_stream());
+  return out;
+}
+
+at::Tensor embedding_bwd(const at::Tensor& idx, const at::Tensor& dy,
+                         long vocab) {
+  auto idx_l = idx.to(at::kLong).contiguous();
+  long ntok = idx_l.numel();
+  int D = dy.size(-1);
+  auto dW = at::empty({vocab, D}, dy.options().dtype(at::kFloat));
+  mxa::embedding_bwd_bf16(idx_l.data_ptr<long>(), bfp(dy),
+                          dW.data_ptr<float>(), ntok, D, vocab,
+                          cur_stream());
+  return dW;
+}
+
+at::Tensor act_fwd(long op, const at::Tensor& x) {
+  long n = x.numel();
+  TORCH_CHECK(n % 8 == 0, "act: numel must be %8");
+  auto y = at::empty_like(x);
+  mxa::act_fwd_bf16((int)op, bfp(x), bfp_mut(y), n, cur_stream());
+  return y;
+}
+
+at::Tensor act_bwd(long op, const at::Tensor& x, const at::Tensor& dy) {
+  long n = x.numel();
+  auto dx = at::empty_like(x);
+  mxa::act_bwd_bf16((int)op, bfp(x), bfp(dy), bfp_mut(dx), n, cur_stream());
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+// multi-tensor optimizer
+// ---------------------------------------------------------------------------
+
+// Packs descriptors; returns (blob_tensor_on_device, nchunks).
+std::vector<at::Tensor> build_opt_descs(std::vector<at::Tensor> ws,
+                                        std::vector<at::Tensor> gs,
+                                        std::vector<at::Tensor> ms,
+                                        std::vector<at::Tensor> vs,
+                                        std::vector<at::Tensor> w32s) {
+  int n = ws.size();
+  std::vector<mxa::TensorDesc> descs(n);
+  long chunks = 0;
+  for (int i = 0; i < n; ++i) {
+    descs[i].w = ws[i].data_ptr();
+    descs[i].g = gs[i].data_ptr();
+    descs[i].m = ms.empty() || !ms[i].defined() ? nullptr
+                                                : ms[i].data_ptr<float>();
+    descs[i].v = vs.empty() || !vs[i].defined() ? nullptr
+                                                : vs[i].data_ptr<float>();
+    descs[i].w32 = w32s.empty() || !w32s[i].defined()
+                       ? nullptr
+                       : w32s[i].data_ptr<float>();
+    descs[i].n = ws[i].numel();
+    descs[i].chunk0 = chunks;
+    chunks += (descs[i].n + 65535) / 65536;
+  }
+  auto blob = at::from_blob(descs.data(),
+                            {(long)(n * sizeof(mxa::TensorDesc))},
+                            at::TensorOptions().dtype(at::kByte))
+                  .clone()
+                  .to(ws[0].device());
+  auto meta = at::tensor({(long)n, chunks}, at::kLong);
+  return {blob, meta};
+}
+
+void multi_opt_run(const at::Tensor& blob, long ntens, long nchunks,
+                   long mode, bool wbf16, double lr, double momentum,
+                   double beta1, double beta2, double eps, double wd,
+                   double rescale, double clip) {
+  mxa::multi_opt(
+      reinterpret_cast<const mxa::TensorDesc*>(blob.data_ptr()), (int)ntens,
+      nchunks, (int)mode, wbf16, (float)lr, (float)momentum, (float)beta1,
+      (float)beta2, (float)eps, (float)wd, (float)rescale, (float)clip,
+      cur_stream());
+}
+
+at::Tensor multi_sumsq_run(const at::Tensor& blob, long ntens, long nchunks,
+                           bool bf16in) {
+  auto out = at::zeros({ntens}, blob.options().dtype(at::kFloat));
+  mxa::multi_sumsq(
+      reinterpret_cast<const mxa::TensorDesc*>(blob.data_ptr()), (int)ntens,
+      nchunks, bf16in, out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// engine: raw HIP stream management (wrapped by torch ExternalStream)
+// ---------------------------------------------------------------------------
+
+long engine_new_stream(long device) {
+  hipStream_t s;
+  int prev;
+  (void)hipGetDevice(&prev);
+  (void)hipSetDevice((int)device);
+  auto err = hipStreamCreateWithFlags(&s, hipStreamNonBlocking);
+  (void)hipSetDevice(prev);
+  TORCH_CHECK(err == hipSuccess, "hipStreamCreate failed");
+  return (long)(uintptr_t)s;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm_tn", &gemm_tn, "C = A @ B^T (bf16, MFMA)");
+  m.def("gemm_wg", &gemm_wg, "C(f32) = sum_m A[m,:] outer B[m,:]");
+  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_wgrad", &conv2d_wgrad);
+  m.def("bn_fwd_train", &bn_fwd_train);
+  m.def("bn_fwd_infer", &bn_fwd_infer);
+  m.def("bn_bwd", &bn_bwd);
+  m.def("maxpool", &maxpool);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("avgpool", &avgpool);
+  m.def("global_avgpool", &global_avgpool);
+  m.def("global_avgpool_bwd", &global_avgpool_bwd);
+  m.def("softmax_fwd", &softmax_fwd);
+  m.def("softmax_bwd", &softmax_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_bwd", &dropout_bwd);
+  m.def("embedding_fwd", &embedding_fwd);
+  m.def("embedding_bwd", &embedding_bwd);
+  m.def("act_fwd", &act_fwd);
+  m.def("act_bwd", &act_bwd);
+  m.def("build_opt_descs", &build_opt_descs);
+  m.def("multi_opt_run", &multi_opt_run);
+  m.def("multi_sumsq_run", &multi_sumsq_run);
+  m.def("engine_new_stream", &engine_new_stream);
+}

@@ -1,0 +1,461 @@
+#include "hip/hip_runtime.h"
+// Fused BatchNorm (NHWC, bf16 data / fp32 stats) and pooling kernels for
+// gfx950.  Reference parity: src/operator/nn/batch_norm.cu (+cudnn_batch_norm)
+// and nn/pooling.cu — re-designed CDNA4-native: NHWC channel-contiguous
+// bf16x8 vector loads (guide G13), fp32 accumulation, fused ReLU epilogue,
+// per-channel partial sums reduced with fp32 atomics (memory-bound ops; the
+// 8 TB/s HBM3E is the roofline, so one pass per tensor read is the goal).
+#include "common.h"
+
+namespace mxa {
+
+// ---------------------------------------------------------------------------
+// BatchNorm forward: stats (partial) -> finalize -> apply(+relu)
+// x is [M][C] with M = N*H*W.  Blocks: 256 thr = 32 rows x 8 col-chunks of
+// 8 channels; grid.x covers channel blocks of 64, grid.y strides rows.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void bn_stats_partial(
+    const bf16* __restrict__ x, float* __restrict__ psum,
+    float* __restrict__ psumsq, long M, int C) {
+  const int cblk = blockIdx.x * 64;
+  const int t = threadIdx.x;
+  const int chunk = t & 7;           // 8 chunks x 8 channels = 64 channels
+  const int rlane = t >> 3;          // 32 row lanes
+  const int c0 = cblk + chunk * 8;
+  const bool active = c0 + 8 <= C;
+  float s[8] = {0}, ss[8] = {0};
+  for (long m = blockIdx.y * 32 + rlane; active && m < M;
+       m += 32L * gridDim.y) {
+    bf16x8 v = *(const bf16x8*)&x[m * C + c0];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      float f = bf2f(v[i]);
+      s[i] += f;
+      ss[i] += f * f;
+    }
+  }
+  // reduce across the 32 row-lanes holding the same channels via LDS
+  __shared__ float red[2][32][64];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    red[0][rlane][chunk * 8 + i] = s[i];
+    red[1][rlane][chunk * 8 + i] = ss[i];
+  }
+  __syncthreads();
+  // 64 threads finalize 64 channels
+  if (t < 64) {
+    float fs = 0, fss = 0;
+#pragma unroll
+    for (int r = 0; r < 32; ++r) {
+      fs += red[0][r][t];
+      fss += red[1][r][t];
+    }
+    int c = cblk + t;
+    if (c < C) {
+      atomicAdd(&psum[c], fs);
+      atomicAdd(&psumsq[c], fss);
+    }
+  }
+}
+
+__global__ void bn_finalize(const float* __restrict__ psum,
+                            const float* __restrict__ psumsq,
+                            float* __restrict__ save_mean,
+                            float* __restrict__ save_rstd,
+                            float* __restrict__ running_mean,
+                            float* __restrict__ running_var, long M, int C,
+                            float eps, float momentum, int update_running) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mean = psum[c] / M;
+  float var = psumsq[c] / M - mean * mean;
+  if (var < 0) var = 0;
+  save_mean[c] = mean;
+  save_rstd[c] = rsqrtf(var + eps);
+  if (update_running) {
+    running_mean[c] = momentum * running_mean[c] + (1.f - momentum) * mean;
+    float ub = M > 1 ? var * M / (M - 1) : var;  // unbiased (torch parity)
+    running_var[c] = momentum * running_var[c] + (1.f - momentum) * ub;
+  }
+}
+
+template <bool RELU>
+__global__ __launch_bounds__(256) void bn_apply(
+    const bf16* __restrict__ x, bf16* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    long M, int C) {
+  long nvec = M * (C / 8);
+  int cv = C / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    bf16x8 v = *(const bf16x8*)&x[i * 8];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float f = (bf2f(v[j]) - mean[c]) * rstd[c] * gamma[c] + beta[c];
+      if (RELU) f = f > 0 ? f : 0;
+      o[j] = f2bf(f);
+    }
+    *(bf16x8*)&y[i * 8] = o;
+  }
+}
+
+// inference: normalize with running stats directly
+template <bool RELU>
+__global__ __launch_bounds__(256) void bn_infer(
+    const bf16* __restrict__ x, bf16* __restrict__ y,
+    const float* __restrict__ rmean, const float* __restrict__ rvar,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    long M, int C, float eps) {
+  long nvec = M * (C / 8);
+  int cv = C / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    bf16x8 v = *(const bf16x8*)&x[i * 8];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float f = (bf2f(v[j]) - rmean[c]) * rsqrtf(rvar[c] + eps) * gamma[c] +
+                beta[c];
+      if (RELU) f = f > 0 ? f : 0;
+      o[j] = f2bf(f);
+    }
+    *(bf16x8*)&y[i * 8] = o;
+  }
+}
+
+// backward reduction: dbeta = sum dy_eff, dgamma = sum dy_eff * xhat
+// (RELU fused: dy_eff = dy * (y > 0), y reconstructed from xhat)
+template <bool RELU>
+__global__ __launch_bounds__(256) void bn_bwd_reduce(
+    const bf16* __restrict__ x, const bf16* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ dbeta, float* __restrict__ dgamma, long M, int C) {
+  const int cblk = blockIdx.x * 64;
+  const int t = threadIdx.x;
+  const int chunk = t & 7, rlane = t >> 3;
+  const int c0 = cblk + chunk * 8;
+  const bool active = c0 + 8 <= C;
+  float sb[8] = {0}, sg[8] = {0};
+  for (long m = blockIdx.y * 32 + rlane; active && m < M;
+       m += 32L * gridDim.y) {
+    bf16x8 vx = *(const bf16x8*)&x[m * C + c0];
+    bf16x8 vdy = *(const bf16x8*)&dy[m * C + c0];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int c = c0 + i;
+      float xh = (bf2f(vx[i]) - mean[c]) * rstd[c];
+      float g = bf2f(vdy[i]);
+      if (RELU && (xh * gamma[c] + beta[c]) <= 0) g = 0;
+      sb[i] += g;
+      sg[i] += g * xh;
+    }
+  }
+  __shared__ float red[2][32][64];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    red[0][rlane][chunk * 8 + i] = sb[i];
+    red[1][rlane][chunk * 8 + i] = sg[i];
+  }
+  __syncthreads();
+  if (t < 64) {
+    float fb = 0, fg = 0;
+#pragma unroll
+    for (int r = 0; r < 32; ++r) {
+      fb += red[0][r][t];
+      fg += red[1][r][t];
+    }
+    int c = cblk + t;
+    if (c < C) {
+      atomicAdd(&dbeta[c], fb);
+      atomicAdd(&dgamma[c], fg);
+    }
+  }
+}
+
+template <bool RELU>
+__global__ __launch_bounds__(256) void bn_bwd_apply(
+    const bf16* __restrict__ x, const bf16* __restrict__ dy,
+    bf16* __restrict__ dx, const float* __restrict__ mean,
+    const float* __restrict__ rstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, const float* __restrict__ dbeta,
+    const float* __restrict__ dgamma, long M, int C) {
+  long nvec = M * (C / 8);
+  int cv = C / 8;
+  float invM = 1.f / (float)M;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    bf16x8 vx = *(const bf16x8*)&x[i * 8];
+    bf16x8 vdy = *(const bf16x8*)&dy[i * 8];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float xh = (bf2f(vx[j]) - mean[c]) * rstd[c];
+      float g = bf2f(vdy[j]);
+      if (RELU && (xh * gamma[c] + beta[c]) <= 0) g = 0;
+      float v = (g - dbeta[c] * invM - xh * dgamma[c] * invM) * gamma[c] *
+                rstd[c];
+      o[j] = f2bf(v);
+    }
+    *(bf16x8*)&dx[i * 8] = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Pooling (NHWC): max with argmax (u8), avg, global-avg; backwards.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void maxpool_fwd(
+    const bf16* __restrict__ x, bf16* __restrict__ y,
+    unsigned char* __restrict__ arg, int Nb, int H, int W, int C, int P,
+    int Q, int R, int S, int sh, int sw, int ph, int pw) {
+  long total = (long)Nb * P * Q * (C / 8);
+  int cv = C / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    long npq = i / cv;
+    int q = (int)(npq % Q);
+    int p = (int)((npq / Q) % P);
+    int n = (int)(npq / ((long)P * Q));
+    float best[8];
+    int bidx[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      best[j] = -3.0e38f;
+      bidx[j] = 0;
+    }
+    for (int r = 0; r < R; ++r) {
+      int ih = p * sh - ph + r;
+      if (ih < 0 || ih >= H) continue;
+      for (int s = 0; s < S; ++s) {
+        int iw = q * sw - pw + s;
+        if (iw < 0 || iw >= W) continue;
+        bf16x8 v =
+            *(const bf16x8*)&x[(((long)n * H + ih) * W + iw) * C + c0];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf2f(v[j]);
+          if (f > best[j]) {
+            best[j] = f;
+            bidx[j] = r * S + s;
+          }
+        }
+      }
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      o[j] = f2bf(best[j]);
+      arg[(npq * C + c0) + j] = (unsigned char)bidx[j];
+    }
+    *(bf16x8*)&y[npq * C + c0] = o;
+  }
+}
+
+__global__ __launch_bounds__(256) void maxpool_bwd(
+    const bf16* __restrict__ dy, const unsigned char* __restrict__ arg,
+    float* __restrict__ dx, int Nb, int H, int W, int C, int P, int Q,
+    int R, int S, int sh, int sw, int ph, int pw) {
+  long total = (long)Nb * P * Q * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long npq = i / C;
+    int q = (int)(npq % Q);
+    int p = (int)((npq / Q) % P);
+    int n = (int)(npq / ((long)P * Q));
+    int a = arg[i];
+    int r = a / S, s = a % S;
+    int ih = p * sh - ph + r, iw = q * sw - pw + s;
+    if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
+    atomicAdd(&dx[(((long)n * H + ih) * W + iw) * C + c], bf2f(dy[i]));
+  }
+}
+
+template <bool COUNT_PAD>
+__global__ __launch_bounds__(256) void avgpool_fwd(
+    const bf16* __restrict__ x, bf16* __restrict__ y, int Nb, int H, int W,
+    int C, int P, int Q, int R, int S, int sh, int sw, int ph, int pw) {
+  long total = (long)Nb * P * Q * (C / 8);
+  int cv = C / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    long npq = i / cv;
+    int q = (int)(npq % Q);
+    int p = (int)((npq / Q) % P);
+    int n = (int)(npq / ((long)P * Q));
+    float acc[8] = {0};
+    int cnt = 0;
+    for (int r = 0; r < R; ++r) {
+      int ih = p * sh - ph + r;
+      if (ih < 0 || ih >= H) continue;
+      for (int s = 0; s < S; ++s) {
+        int iw = q * sw - pw + s;
+        if (iw < 0 || iw >= W) continue;
+        bf16x8 v =
+            *(const bf16x8*)&x[(((long)n * H + ih) * W + iw) * C + c0];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += bf2f(v[j]);
+        ++cnt;
+      }
+    }
+    float div = COUNT_PAD ? (float)(R * S) : (float)max(cnt, 1);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[j] / div);
+    *(bf16x8*)&y[npq * C + c0] = o;
+  }
+}
+
+// global average pool over H*W: y[n, c] (+ backward broadcast)
+__global__ __launch_bounds__(256) void gap_fwd(const bf16* __restrict__ x,
+                                               bf16* __restrict__ y, int Nb,
+                                               long HW, int C) {
+  // one block per n, threads cover channels in chunks of 8
+  int n = blockIdx.x;
+  for (int c0 = threadIdx.x * 8; c0 < C; c0 += blockDim.x * 8) {
+    float acc[8] = {0};
+    const bf16* base = x + (long)n * HW * C + c0;
+    for (long m = 0; m < HW; ++m) {
+      bf16x8 v = *(const bf16x8*)&base[m * C];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += bf2f(v[j]);
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[j] / (float)HW);
+    *(bf16x8*)&y[(long)n * C + c0] = o;
+  }
+}
+
+__global__ __launch_bounds__(256) void gap_bwd(const bf16* __restrict__ dy,
+                                               bf16* __restrict__ dx, int Nb,
+                                               long HW, int C) {
+  long total = (long)Nb * HW * (C / 8);
+  int cv = C / 8;
+  float inv = 1.f / (float)HW;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    long nm = i / cv;
+    int n = (int)(nm / HW);
+    bf16x8 g = *(const bf16x8*)&dy[(long)n * C + c0];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(g[j]) * inv);
+    *(bf16x8*)&dx[i * 8] = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host launchers
+// ---------------------------------------------------------------------------
+
+void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
+                  float* save_mean, float* save_rstd, float* running_mean,
+                  float* running_var, const float* gamma, const float* beta,
+                  long M, int C, float eps, float momentum, bool relu,
+                  hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(psum, 0, C * sizeof(float), s));
+  HIP_CHECK(hipMemsetAsync(psumsq, 0, C * sizeof(float), s));
+  int gy = (int)min((M + 8191) / 8192, (long)64);
+  dim3 g1(cdiv_h(C, 64), gy < 1 ? 1 : gy);
+ hipLaunchKernelGGL(( bn_stats_partial), dim3(g1), dim3(256), 0, s, x, psum, psumsq, M, C);
+ hipLaunchKernelGGL(( bn_finalize), dim3(cdiv_h(C, 256)), dim3(256), 0, s, 
+      psum, psumsq, save_mean, save_rstd, running_mean, running_var, M, C,
+      eps, momentum, 1);
+  dim3 g2 = ew_grid(M * (C / 8));
+  if (relu)
+   hipLaunchKernelGGL(( bn_apply<true>), dim3(g2), dim3(256), 0, s, x, y, save_mean, save_rstd, gamma,
+                                      beta, M, C);
+  else
+   hipLaunchKernelGGL(( bn_apply<false>), dim3(g2), dim3(256), 0, s, x, y, save_mean, save_rstd, gamma,
+                                       beta, M, C);
+}
+
+void bn_fwd_infer(const bf16* x, bf16* y, const float* rmean,
+                  const float* rvar, const float* gamma, const float* beta,
+                  long M, int C, float eps, bool relu, hipStream_t s) {
+  dim3 g = ew_grid(M * (C / 8));
+  if (relu)
+   hipLaunchKernelGGL(( bn_infer<true>), dim3(g), dim3(256), 0, s, x, y, rmean, rvar, gamma, beta, M, C,
+                                     eps);
+  else
+   hipLaunchKernelGGL(( bn_infer<false>), dim3(g), dim3(256), 0, s, x, y, rmean, rvar, gamma, beta, M, C,
+                                      eps);
+}
+
+void bn_bwd(const bf16* x, const bf16* dy, bf16* dx, const float* save_mean,
+            const float* save_rstd, const float* gamma, const float* beta,
+            float* dgamma, float* dbeta, long M, int C, bool relu,
+            hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(dgamma, 0, C * sizeof(float), s));
+  HIP_CHECK(hipMemsetAsync(dbeta, 0, C * sizeof(float), s));
+  int gy = (int)min((M + 8191) / 8192, (long)64);
+  dim3 g1(cdiv_h(C, 64), gy < 1 ? 1 : gy);
+  dim3 g2 = ew_grid(M * (C / 8));
+  if (relu) {
+   hipLaunchKernelGGL(( bn_bwd_reduce<true>), dim3(g1), dim3(256), 0, s, x, dy, save_mean, save_rstd,
+                                           gamma, beta, dbeta, dgamma, M, C);
+   hipLaunchKernelGGL(( bn_bwd_apply<true>), dim3(g2), dim3(256), 0, s, x, dy, dx, save_mean, save_rstd,
+                                          gamma, beta, dbeta, dgamma, M, C);
+  } else {
+   hipLaunchKernelGGL(( bn_bwd_reduce<false>), dim3(g1), dim3(256), 0, s, x, dy, save_mean, save_rstd,
+                                            gamma, beta, dbeta, dgamma, M, C);
+   hipLaunchKernelGGL(( bn_bwd_apply<false>), dim3(g2), dim3(256), 0, s, x, dy, dx, save_mean, save_rstd,
+                                           gamma, beta, dbeta, dgamma, M, C);
+  }
+}
+
+void maxpool_nhwc(const bf16* x, bf16* y, unsigned char* arg, int Nb, int H,
+                  int W, int C, int P, int Q, int R, int S, int sh, int sw,
+                  int ph, int pw, hipStream_t s) {
+  dim3 g = ew_grid((long)Nb * P * Q * (C / 8));
+ hipLaunchKernelGGL(( maxpool_fwd), dim3(g), dim3(256), 0, s, x, y, arg, Nb, H, W, C, P, Q, R, S, sh, sw,
+                                ph, pw);
+}
+
+void maxpool_nhwc_bwd(const bf16* dy, const unsigned char* arg, float* dx,
+                      int Nb, int H, int W, int C, int P, int Q, int R,
+                      int S, int sh, int sw, int ph, int pw, hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(dx, 0, (long)Nb * H * W * C * sizeof(float), s));
+  dim3 g = ew_grid((long)Nb * P * Q * C);
+ hipLaunchKernelGGL(( maxpool_bwd), dim3(g), dim3(256), 0, s, dy, arg, dx, Nb, H, W, C, P, Q, R, S, sh,
+                                sw, ph, pw);
+}
+
+void avgpool_nhwc(const bf16* x, bf16* y, int Nb, int H, int W, int C,
+                  int P, int Q, int R, int S, int sh, int sw, int ph,
+                  int pw, bool count_pad, hipStream_t s) {
+  dim3 g = ew_grid((long)Nb * P * Q * (C / 8));
+  if (count_pad)
+   hipLaunchKernelGGL(( avgpool_fwd<true>), dim3(g), dim3(256), 0, s, x, y, Nb, H, W, C, P, Q, R, S, sh,
+                                        sw, ph, pw);
+  else
+   hipLaunchKernelGGL(( avgpool_fwd<false>), dim3(g), dim3(256), 0, s, x, y, Nb, H, W, C, P, Q, R, S, sh,
+                                         sw, ph, pw);
+}
+
+void gap_nhwc(const bf16* x, bf16* y, int Nb, long HW, int C,
+              hipStream_t s) {
+ hipLaunchKernelGGL(( gap_fwd), dim3(Nb), dim3(256), 0, s, x, y, Nb, HW, C);
+}
+
+void gap_nhwc_bwd(const bf16* dy, bf16* dx, int Nb, long HW, int C,
+                  hipStream_t s) {
+  dim3 g = ew_grid((long)Nb * HW * (C / 8));
+ hipLaunchKernelGGL(( gap_bwd), dim3(g), dim3(256), 0, s, dy, dx, Nb, HW, C);
+}
+
+}  // namespace mxa

@@ -1,0 +1,383 @@
+// TN-form MFMA GEMM + implicit-GEMM convolution for gfx950 (MI355X).
+//
+//   C[M][N] = sum_k A[M][k] * B[N][k]   (both operands [out][reduce-k],
+//                                        k contiguous in memory)
+//
+// Every dense hot op maps onto this form (reference parity:
+// src/operator/linalg.h linalg_gemm, nn/fully_connected.cu,
+// nn/convolution.cu — here one CDNA4-native kernel family instead of
+// cuBLAS/cuDNN):
+//   * FullyConnected fwd : A=x[m,in],  B=W[out,in]
+//   * FullyConnected dgrad: A=dy[m,out], B=W^T[in,out]
+//   * Conv2D fwd  (NHWC) : A=im2col gather of x, B=W[R,S,K,C] (per-(r,s)
+//                          [K][C] slices), K-dim = (r,s,c)
+//   * Conv2D dgrad (NHWC): A=scatter-inverse gather of dy (divisibility
+//                          predicates), B=W[R,S,C,K], K-dim = (r,s,k)
+//
+// Structure (guide §5 ladder): 128x128 tile, BK=64, 4 waves (2x2 of 64x64),
+// mfma_f32_16x16x32_bf16, double-buffered LDS, register staging with the
+// T14 write-after-barrier schedule (one barrier per K-step), T2 XOR LDS
+// swizzle (byte ^= (row&7)<<4) on store+read so ds_read_b128 fragment
+// reads are conflict-free, T1 bijective XCD-aware block swizzle.
+// OOB rows (tile tails, conv padding) read a 128-byte zero page instead of
+// branching in the inner loop.
+#include "common.h"
+
+namespace mxa {
+
+constexpr int BM = 128, BN = 128, BK = 64, THREADS = 256;
+
+DEVINL int lds_off(int row, int chunk16) {
+  // bf16 element offset of 16-byte chunk `chunk16` of row `row` in a
+  // [128][64] bf16 tile with the T2 XOR swizzle (8 chunks of 8 bf16 / row).
+  return row * BK + ((chunk16 ^ (row & 7)) << 3);
+}
+
+// ---------------------------------------------------------------------------
+// Loaders: produce the per-lane 16-byte source pointer for tile staging.
+// Each thread stages 4 rows (passes); row = pass*32 + t/8, chunk16 = t%8.
+// ---------------------------------------------------------------------------
+
+struct PlainLoader {
+  const bf16* base;
+  const bf16* zpage;
+  int rows;
+  long ld;        // elements between rows
+  long stride_z;  // batch stride (elements)
+  int ksteps;     // K / 64
+  // device state
+  const bf16* rp[4];
+  long koff;
+
+  DEVINL void init(int row0, int t, int z) {
+    const bf16* b = base + (long)z * stride_z;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int r = row0 + p * 32 + (t >> 3);
+      rp[p] = (r < rows) ? b + (long)r * ld : nullptr;
+    }
+  }
+  DEVINL void set_kstep(int ks) { koff = (long)ks * BK; }
+  DEVINL const bf16* src(int p, int chunk16) const {
+    return rp[p] ? rp[p] + koff + chunk16 * 8 : zpage + chunk16 * 8;
+  }
+};
+
+// im2col gather over x[N,H,W,C] for conv forward; K-dim order (r, s, c).
+struct ConvALoader {
+  const bf16* x;
+  const bf16* zpage;
+  int Nb, H, W, C, P, Q, S;
+  int sh, sw, ph, pw, dh, dw;
+  int csteps;  // C/64
+  // device state
+  long nb[4];
+  int ih0[4], iw0[4];
+  int rowok;  // bitmask
+  const bf16* rp[4];
+
+  DEVINL void init(int row0, int t, int) {
+    rowok = 0;
+    int PQ = P * Q;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int m = row0 + p * 32 + (t >> 3);
+      int ok = m < Nb * PQ;
+      int mm = ok ? m : 0;
+      int n = mm / PQ, pq = mm % PQ;
+      int pi = pq / Q, qi = pq % Q;
+      nb[p] = (long)n * H * W * C;
+      ih0[p] = pi * sh - ph;
+      iw0[p] = qi * sw - pw;
+      rowok |= ok << p;
+    }
+  }
+  DEVINL void set_kstep(int ks) {
+    int seg = ks / csteps, cb = ks % csteps;
+    int r = seg / S, s = seg % S;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int ih = ih0[p] + r * dh, iw = iw0[p] + s * dw;
+      bool ok = ((rowok >> p) & 1) && ih >= 0 && ih < H && iw >= 0 && iw < W;
+      rp[p] = ok ? x + nb[p] + ((long)ih * W + iw) * C + cb * BK : nullptr;
+    }
+  }
+  DEVINL const bf16* src(int p, int chunk16) const {
+    return rp[p] ? rp[p] + chunk16 * 8 : zpage + chunk16 * 8;
+  }
+};
+
+// forward weights stored [R,S,K,C]: per-(r,s) a [K][C] slice, C contiguous.
+struct ConvWLoader {
+  const bf16* w;
+  const bf16* zpage;
+  int Kout, C, csteps;
+  // device state
+  int nrow[4];
+  const bf16* rp[4];
+
+  DEVINL void init(int row0, int t, int) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int n = row0 + p * 32 + (t >> 3);
+      nrow[p] = (n < Kout) ? n : -1;
+    }
+  }
+  DEVINL void set_kstep(int ks) {
+    int seg = ks / csteps, cb = ks % csteps;
+    const bf16* base = w + (long)seg * Kout * C + cb * BK;
+#pragma unroll
+    for (int p = 0; p < 4; ++p)
+      rp[p] = (nrow[p] >= 0) ? base + (long)nrow[p] * C : nullptr;
+  }
+  DEVINL const bf16* src(int p, int chunk16) const {
+    return rp[p] ? rp[p] + chunk16 * 8 : zpage + chunk16 * 8;
+  }
+};
+
+// dgrad A: gather dy[N,P,Q,K] rows for dx[m=(n,h,w)]; K-dim order (r, s, k).
+struct DgradALoader {
+  const bf16* dy;
+  const bf16* zpage;
+  int Nb, H, W, P, Q, Kout, S;
+  int sh, sw, ph, pw, dh, dw;
+  int ksteps;  // Kout/64
+  // device state
+  long nb[4];
+  int h0[4], w0[4];
+  int rowok;
+  const bf16* rp[4];
+
+  DEVINL void init(int row0, int t, int) {
+    rowok = 0;
+    int HW = H * W;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int m = row0 + p * 32 + (t >> 3);
+      int ok = m < Nb * HW;
+      int mm = ok ? m : 0;
+      int n = mm / HW, hw = mm % HW;
+      nb[p] = (long)n * P * Q * Kout;
+      h0[p] = hw / W + ph;
+      w0[p] = hw % W + pw;
+      rowok |= ok << p;
+    }
+  }
+  DEVINL void set_kstep(int ks) {
+    int seg = ks / ksteps, kb = ks % ksteps;
+    int r = seg / S, s = seg % S;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int pn = h0[p] - r * dh, qn = w0[p] - s * dw;
+      int pi = pn / sh, qi = qn / sw;
+      bool ok = ((rowok >> p) & 1) && pn >= 0 && qn >= 0 &&
+                pn % sh == 0 && qn % sw == 0 && pi < P && qi < Q;
+      rp[p] = ok ? dy + nb[p] + ((long)pi * Q + qi) * Kout + kb * BK
+                 : nullptr;
+    }
+  }
+  DEVINL const bf16* src(int p, int chunk16) const {
+    return rp[p] ? rp[p] + chunk16 * 8 : zpage + chunk16 * 8;
+  }
+};
+
+// dgrad weights stored [R,S,C,K]: per-(r,s) a [C][K] slice, K contiguous.
+struct DgradWLoader {
+  const bf16* w;
+  const bf16* zpage;
+  int C, Kout, ksteps;
+  // device state
+  int nrow[4];
+  const bf16* rp[4];
+
+  DEVINL void init(int row0, int t, int) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int n = row0 + p * 32 + (t >> 3);
+      nrow[p] = (n < C) ? n : -1;
+    }
+  }
+  DEVINL void set_kstep(int ks) {
+    int seg = ks / ksteps, kb = ks % ksteps;
+    const bf16* base = w + (long)seg * C * Kout + kb * BK;
+#pragma unroll
+    for (int p = 0; p < 4; ++p)
+      rp[p] = (nrow[p] >= 0) ? base + (long)nrow[p] * Kout : nullptr;
+  }
+  DEVINL const bf16* src(int p, int chunk16) const {
+    return rp[p] ? rp[p] + chunk16 * 8 : zpage + chunk16 * 8;
+  }
+};
+
+// ---------------------------------------------------------------------------
+// Core kernel
+// ---------------------------------------------------------------------------
+
+template <typename AL, typename BL, bool RELU, bool BIAS>
+__global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
+    AL aload, BL bload, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, long strideC,
+    int ksteps) {
+  __shared__ __align__(16) bf16 smem[2][2][BM * BK];
+
+  const int ntn = (N + BN - 1) / BN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * BM, bn = (bid % ntn) * BN;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+
+  aload.init(bm, t, z);
+  bload.init(bn, t, z);
+
+  i32x4 areg[4], breg[4];
+  const int srow = t >> 3, schunk = t & 7;  // staging row-slot base / chunk
+
+  auto stage_load = [&](int ks) {
+    aload.set_kstep(ks);
+    bload.set_kstep(ks);
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      areg[p] = *(const i32x4*)aload.src(p, schunk);
+      breg[p] = *(const i32x4*)bload.src(p, schunk);
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int row = p * 32 + srow;
+      *(i32x4*)&smem[buf][0][lds_off(row, schunk)] = areg[p];
+      *(i32x4*)&smem[buf][1][lds_off(row, schunk)] = breg[p];
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int w = t / WAVE, l = t % WAVE;
+  const int wm = (w >> 1) * 64, wn = (w & 1) * 64;
+  const int lrow = l & 15, lk = l >> 4;  // fragment row / k-group
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bfr[4];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        int row = wm + mf * 16 + lrow;
+        af[mf] = *(const bf16x8*)&smem[buf][0][lds_off(row, chunk)];
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int row = wn + nf * 16 + lrow;
+        bfr[nf] = *(const bf16x8*)&smem[buf][1][lds_off(row, chunk)];
+      }
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+    }
+  };
+
+  // T14 schedule: one barrier per K-step, loads for step t+2 issued right
+  // after the step-(t+1) tile is written to LDS.
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  if (ksteps > 1) stage_load(1);
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    compute(buf);
+    if (ks + 1 < ksteps) {
+      stage_write(buf ^ 1);
+      if (ks + 2 < ksteps) stage_load(ks + 2);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C/D fragment layout (guide §3): col = l&15, row = (l>>4)*4+i
+  bf16* cz = Cout + (long)z * strideC;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wn + nf * 16 + (l & 15);
+      if (gcol >= N) continue;
+      float bv = BIAS ? bias[gcol] : 0.f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (l >> 4) * 4 + i;
+        if (grow >= M) continue;
+        float v = acc[mf][nf][i] + bv;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        cz[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host launchers
+// ---------------------------------------------------------------------------
+
+template <typename AL, typename BL>
+static void launch(const AL& a, const BL& b, bf16* c, const float* bias,
+                   int M, int N, long ldc, long strideC, int ksteps,
+                   bool relu, int batches, hipStream_t stream) {
+  dim3 grid(cdiv_h(M, BM) * cdiv_h(N, BN), 1, batches);
+  if (bias) {
+    if (relu)
+      gemm_tn_core<AL, BL, true, true><<<grid, THREADS, 0, stream>>>(
+          a, b, c, bias, M, N, ldc, strideC, ksteps);
+    else
+      gemm_tn_core<AL, BL, false, true><<<grid, THREADS, 0, stream>>>(
+          a, b, c, bias, M, N, ldc, strideC, ksteps);
+  } else {
+    if (relu)
+      gemm_tn_core<AL, BL, true, false><<<grid, THREADS, 0, stream>>>(
+          a, b, c, bias, M, N, ldc, strideC, ksteps);
+    else
+      gemm_tn_core<AL, BL, false, false><<<grid, THREADS, 0, stream>>>(
+          a, b, c, bias, M, N, ldc, strideC, ksteps);
+  }
+}
+
+void gemm_tn_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
+                  int M, int N, int K, long lda, long ldb, long ldc,
+                  long strideA, long strideB, long strideC, int batches,
+                  bool relu, const bf16* zpage, hipStream_t stream) {
+  PlainLoader a{A, zpage, M, lda, strideA, K / BK};
+  PlainLoader b{B, zpage, N, ldb, strideB, K / BK};
+  launch(a, b, C, bias, M, N, ldc, strideC, K / BK, relu, batches, stream);
+}
+
+void conv2d_fwd_bf16(const bf16* x, const bf16* w_rskc, bf16* y,
+                     const float* bias, int Nb, int H, int W, int C, int Kout,
+                     int R, int S, int P, int Q, int sh, int sw, int ph,
+                     int pw, int dh, int dw, bool relu, const bf16* zpage,
+                     hipStream_t stream) {
+  ConvALoader a{x, zpage, Nb, H, W, C, P, Q, S, sh, sw, ph, pw, dh, dw,
+                C / BK};
+  ConvWLoader b{w_rskc, zpage, Kout, C, C / BK};
+  int M = Nb * P * Q;
+  launch(a, b, y, bias, M, Kout, Kout, 0, R * S * (C / BK), relu, 1, stream);
+}
+
+void conv2d_dgrad_bf16(const bf16* dy, const bf16* w_rsck, bf16* dx, int Nb,
+                       int H, int W, int C, int Kout, int R, int S, int P,
+                       int Q, int sh, int sw, int ph, int pw, int dh, int dw,
+                       const bf16* zpage, hipStream_t stream) {
+  DgradALoader a{dy, zpage, Nb, H, W, P, Q, Kout, S, sh, sw, ph, pw, dh, dw,
+                 Kout / BK};
+  DgradWLoader b{w_rsck, zpage, C, Kout, Kout / BK};
+  int M = Nb * H * W;
+  launch(a, b, dx, (const float*)nullptr, M, C, C, 0,
+         R * S * (Kout / BK), false, 1, stream);
+}
+
+}  // namespace mxa

@@ -1,0 +1,309 @@
+// WG-form MFMA GEMM for gfx950: weight gradients (reduce over the MAJOR
+// dimension of both operands):
+//
+//   C[Mo][No] += sum_m A[m][Mo] * B[m][No]
+//
+// Used for (reference parity: FullyConnectedGrad / conv wgrad in
+// nn/fully_connected.cu, nn/convolution.cu):
+//   * FullyConnected wgrad: dW[out][in] = sum_m dy[m][out] * x[m][in]
+//   * Conv2D wgrad (NHWC) : dW[r,s,c,k] = sum_m xg[m][c] * dy[m][k]
+//     where xg is the im2col gather; the (r,s) slice and the reduce-split
+//     are encoded in blockIdx.y.
+//
+// Both operands are staged [m][cols] (coalesced 16-B row chunks) and
+// written to LDS TRANSPOSED via in-register bf16 pair interleave (two
+// m-rows zipped into b32 writes), producing a [cols][m] swizzled image —
+// fragment reads are then plain ds_read_b128 rows exactly like the TN
+// kernel (no transpose-read instruction).  The reduce dimension is split
+// across blocks; f32 output accumulates with atomicAdd.
+#include "common.h"
+
+namespace mxa {
+
+constexpr int WBM = 128, WBN = 128, WBK = 64, WTHREADS = 256;
+
+DEVINL int colkey(int col) { return (col ^ (col >> 3)) & 7; }
+
+DEVINL int wg_byte(int col, int mbyte) {
+  // byte offset of m-byte `mbyte` (0..127) in image column `col` of a
+  // [128 cols][64 m] bf16 image; XOR swizzle on the 16-B chunk index.
+  return col * 128 + ((((mbyte >> 4) ^ colkey(col)) << 4) | (mbyte & 15));
+}
+
+struct WgPlainRows {
+  const bf16* base;
+  const bf16* zpage;
+  long mtotal;
+  long ld;
+  int rowlen;  // valid elements per row (col tail guard; must be %8)
+  // device state
+  long cur;
+  int col0;
+
+  DEVINL void init(int c0, long m0) {
+    col0 = c0;
+    cur = m0;
+  }
+  DEVINL void step() { cur += WBK; }
+  DEVINL const bf16* src(int mrow, int colchunk) const {
+    long m = cur + mrow;
+    int c = col0 + colchunk * 8;
+    return (m < mtotal && c + 8 <= rowlen)
+               ? base + m * ld + c
+               : zpage + colchunk * 8;
+  }
+};
+
+struct WgIm2colRows {
+  const bf16* x;
+  const bf16* zpage;
+  int Nb, H, W, C, P, Q;
+  int sh, sw, ph, pw;
+  // device state
+  int roff, soff;
+  int n_[8], p_[8], q_[8];
+  int col0;
+
+  DEVINL void init_rows(int c0, long m0, const int* rows) {
+    col0 = c0;
+    int PQ = P * Q;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      long m = m0 + rows[i];
+      n_[i] = (int)(m / PQ);
+      int pq = (int)(m % PQ);
+      p_[i] = pq / Q;
+      q_[i] = pq % Q;
+    }
+  }
+  DEVINL void step() {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      q_[i] += WBK;
+      while (q_[i] >= Q) {
+        q_[i] -= Q;
+        p_[i] += 1;
+      }
+      while (p_[i] >= P) {
+        p_[i] -= P;
+        n_[i] += 1;
+      }
+    }
+  }
+  DEVINL const bf16* src(int slot, int colchunk) const {
+    int n = n_[slot];
+    if (n >= Nb) return zpage + colchunk * 8;
+    int ih = p_[slot] * sh - ph + roff;
+    int iw = q_[slot] * sw - pw + soff;
+    if (ih < 0 || ih >= H || iw < 0 || iw >= W) return zpage + colchunk * 8;
+    return x + (((long)n * H + ih) * W + iw) * C + col0 + colchunk * 8;
+  }
+};
+
+template <bool A_IM2COL>
+__global__ __launch_bounds__(WTHREADS, 1) void gemm_wg_core(
+    WgPlainRows aplain, WgIm2colRows aim, WgPlainRows bload,
+    float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
+    int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic) {
+  __shared__ __align__(16) bf16 smem[2][2][WBM * WBK];
+
+  const int seg = blockIdx.y / nsplit, split = blockIdx.y % nsplit;
+  const int per = (ksteps_total + nsplit - 1) / nsplit;
+  const int kstep0 = split * per;
+  const int ksteps = min(per, ksteps_total - kstep0);
+  if (ksteps <= 0) return;
+
+  const int ntn = (No + WBN - 1) / WBN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * WBM, bn = (bid % ntn) * WBN;
+  const int t = threadIdx.x;
+  const bool isB = t >= 128;
+  const int tt = t & 127;
+  const int pair = tt & 7, colchunk = tt >> 3;  // 0..15
+
+  int rows_local[8];
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    rows_local[2 * p] = p * 16 + 2 * pair;
+    rows_local[2 * p + 1] = p * 16 + 2 * pair + 1;
+  }
+
+  long m0 = (long)kstep0 * WBK;
+  if (!isB) {
+    if (A_IM2COL) {
+      aim.roff = (seg / S) * dh;
+      aim.soff = (seg % S) * dw;
+      aim.init_rows(bm, m0, rows_local);
+    } else {
+      aplain.init(bm, m0);
+    }
+  } else {
+    bload.init(bn, m0);
+  }
+
+  i32x4 sreg[8];
+
+  auto stage_load = [&]() {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      if (!isB) {
+        if (A_IM2COL)
+          sreg[i] = *(const i32x4*)aim.src(i, colchunk);
+        else
+          sreg[i] = *(const i32x4*)aplain.src(rows_local[i], colchunk);
+      } else {
+        sreg[i] = *(const i32x4*)bload.src(rows_local[i], colchunk);
+      }
+    }
+    if (!isB) {
+      if (A_IM2COL)
+        aim.step();
+      else
+        aplain.step();
+    } else {
+      bload.step();
+    }
+  };
+
+  auto stage_write = [&](int buf) {
+    bf16* img = smem[buf][isB ? 1 : 0];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const u32* lo = (const u32*)&sreg[2 * p];      // row r0, 8 bf16
+      const u32* hi = (const u32*)&sreg[2 * p + 1];  // row r0+1
+      int r0 = rows_local[2 * p];
+#pragma unroll
+      for (int jw = 0; jw < 4; ++jw) {
+        u32 a01 = lo[jw];  // cols (2jw, 2jw+1) of row r0
+        u32 b01 = hi[jw];
+        u32 pair0 = (a01 & 0xffffu) | (b01 << 16);
+        u32 pair1 = (a01 >> 16) | (b01 & 0xffff0000u);
+        int c0 = colchunk * 8 + 2 * jw;
+        *(u32*)((char*)img + wg_byte(c0, r0 * 2)) = pair0;
+        *(u32*)((char*)img + wg_byte(c0 + 1, r0 * 2)) = pair1;
+      }
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int w = t / WAVE, l = t % WAVE;
+  const int wm = (w >> 1) * 64, wn = (w & 1) * 64;
+  const int lcol = l & 15, lk = l >> 4;
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bfr[4];
+      int mbyte = (kk * 4 + lk) << 4;
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        int col = wm + mf * 16 + lcol;
+        af[mf] =
+            *(const bf16x8*)((const char*)smem[buf][0] + wg_byte(col, mbyte));
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int col = wn + nf * 16 + lcol;
+        bfr[nf] =
+            *(const bf16x8*)((const char*)smem[buf][1] + wg_byte(col, mbyte));
+      }
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+    }
+  };
+
+  stage_load();
+  stage_write(0);
+  __syncthreads();
+  if (ksteps > 1) stage_load();
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    compute(buf);
+    if (ks + 1 < ksteps) {
+      stage_write(buf ^ 1);
+      if (ks + 2 < ksteps) stage_load();
+    }
+    __syncthreads();
+  }
+
+  float* cseg = Cout + (long)seg * strideSeg;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wn + nf * 16 + (l & 15);
+      if (gcol >= No) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (l >> 4) * 4 + i;
+        if (grow >= Mo) continue;
+        float v = acc[mf][nf][i];
+        if (use_atomic)
+          atomicAdd(&cseg[(long)grow * ldc + gcol], v);
+        else
+          cseg[(long)grow * ldc + gcol] = v;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host launchers
+// ---------------------------------------------------------------------------
+
+static int pick_split(long blocks_xy, long ksteps) {
+  // enough blocks to fill 256 CUs a few times over, bounded by the reduce
+  int target = 1;
+  while (blocks_xy * target < 640 && target < ksteps && target < 64)
+    target *= 2;
+  return target;
+}
+
+void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, long mtotal,
+                  int Mo, int No, long lda, long ldb, long ldc,
+                  const bf16* zpage, hipStream_t stream) {
+  int ksteps = (int)((mtotal + WBK - 1) / WBK);
+  long bxy = (long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN);
+  int nsplit = pick_split(bxy, ksteps);
+  WgPlainRows a{A, zpage, mtotal, lda, Mo};
+  WgPlainRows b{B, zpage, mtotal, ldb, No};
+  WgIm2colRows dummy{};
+  dim3 grid((unsigned)bxy, nsplit, 1);
+  int use_atomic = nsplit > 1;
+  if (use_atomic) HIP_CHECK(hipMemsetAsync(C, 0, sizeof(float) * Mo * ldc,
+                                           stream));
+  gemm_wg_core<false><<<grid, WTHREADS, 0, stream>>>(
+      a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic);
+}
+
+void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
+                       int Nb, int H, int W, int C, int Kout, int R, int S,
+                       int P, int Q, int sh, int sw, int ph, int pw, int dh,
+                       int dw_, const bf16* zpage, hipStream_t stream) {
+  long mtotal = (long)Nb * P * Q;
+  int ksteps = (int)((mtotal + WBK - 1) / WBK);
+  long bxy = (long)cdiv_h(C, WBM) * cdiv_h(Kout, WBN);
+  int nsplit = pick_split(bxy * R * S, ksteps);
+  WgIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
+  WgPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
+  WgPlainRows dummy{};
+  dim3 grid((unsigned)bxy, R * S * nsplit, 1);
+  int use_atomic = nsplit > 1;
+  if (use_atomic)
+    HIP_CHECK(hipMemsetAsync(dw_rsck, 0,
+                             sizeof(float) * (long)R * S * C * Kout, stream));
+  gemm_wg_core<true><<<grid, WTHREADS, 0, stream>>>(
+      dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
+      ksteps, nsplit, use_atomic);
+}
+
+}  // namespace mxa

@@ -1,0 +1,295 @@
+// Row softmax / log-softmax and LayerNorm for gfx950 (bf16 data, fp32
+// accumulation).  Reference parity: src/operator/nn/softmax.cu (warp-level
+// row softmax) and nn/layer_norm.cu (LayerNormFusedForwardKernel) —
+// CDNA4-native: one wave64 (or one 256-thr block for wide rows) per row,
+// online max+sum single pass where the row fits in registers, bf16x8
+// vector loads, __shfl_xor wave reductions.
+#include "common.h"
+
+namespace mxa {
+
+DEVINL float wave_red_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+DEVINL float wave_red_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+// block-level reductions via LDS (256 threads = 4 waves)
+DEVINL float block_red_sum(float v, float* scratch) {
+  v = wave_red_sum(v);
+  int w = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) scratch[w] = v;
+  __syncthreads();
+  float r = scratch[0] + scratch[1] + scratch[2] + scratch[3];
+  __syncthreads();
+  return r;
+}
+DEVINL float block_red_max(float v, float* scratch) {
+  v = wave_red_max(v);
+  int w = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) scratch[w] = v;
+  __syncthreads();
+  float r = fmaxf(fmaxf(scratch[0], scratch[1]),
+                  fmaxf(scratch[2], scratch[3]));
+  __syncthreads();
+  return r;
+}
+
+// ---------------------------------------------------------------------------
+// softmax / log_softmax forward: one 256-thr block per row (D % 8 == 0)
+// ---------------------------------------------------------------------------
+
+template <bool LOG>
+__global__ __launch_bounds__(256) void softmax_fwd(
+    const bf16* __restrict__ x, bf16* __restrict__ y, long rows, int D) {
+  __shared__ float scratch[4];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const bf16* xr = x + row * D;
+    bf16* yr = y + row * D;
+    float mx = -3.0e38f;
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) mx = fmaxf(mx, bf2f(v[j]));
+    }
+    mx = block_red_max(mx, scratch);
+    float sum = 0;
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) sum += __expf(bf2f(v[j]) - mx);
+    }
+    sum = block_red_sum(sum, scratch);
+    float inv = 1.f / sum, lse = __logf(sum);
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float e = bf2f(v[j]) - mx;
+        o[j] = f2bf(LOG ? e - lse : __expf(e) * inv);
+      }
+      *(bf16x8*)&yr[i] = o;
+    }
+  }
+}
+
+// softmax bwd: dx = (dy - sum(dy*y)) * y ; log_softmax bwd: dx = dy - exp(y)*sum(dy)
+template <bool LOG>
+__global__ __launch_bounds__(256) void softmax_bwd(
+    const bf16* __restrict__ y, const bf16* __restrict__ dy,
+    bf16* __restrict__ dx, long rows, int D) {
+  __shared__ float scratch[4];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const bf16* yr = y + row * D;
+    const bf16* gr = dy + row * D;
+    bf16* or_ = dx + row * D;
+    float dot = 0;
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 vy = *(const bf16x8*)&yr[i];
+      bf16x8 vg = *(const bf16x8*)&gr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += LOG ? bf2f(vg[j]) : bf2f(vg[j]) * bf2f(vy[j]);
+    }
+    dot = block_red_sum(dot, scratch);
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 vy = *(const bf16x8*)&yr[i];
+      bf16x8 vg = *(const bf16x8*)&gr[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float yv = bf2f(vy[j]), gv = bf2f(vg[j]);
+        o[j] = f2bf(LOG ? gv - __expf(yv) * dot : (gv - dot) * yv);
+      }
+      *(bf16x8*)&or_[i] = o;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm forward/backward: block per row, saves mean/rstd for bwd
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void layernorm_fwd(
+    const bf16* __restrict__ x, bf16* __restrict__ y,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ save_mean, float* __restrict__ save_rstd, long rows,
+    int D, float eps) {
+  __shared__ float scratch[4];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const bf16* xr = x + row * D;
+    bf16* yr = y + row * D;
+    float s = 0, ss = 0;
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(v[j]);
+        s += f;
+        ss += f * f;
+      }
+    }
+    s = block_red_sum(s, scratch);
+    ss = block_red_sum(ss, scratch);
+    float mean = s / D;
+    float var = ss / D - mean * mean;
+    float rstd = rsqrtf(var > 0 ? var + eps : eps);
+    if (threadIdx.x == 0) {
+      save_mean[row] = mean;
+      save_rstd[row] = rstd;
+    }
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = (bf2f(v[j]) - mean) * rstd;
+        o[j] = f2bf(xh * gamma[i + j] + beta[i + j]);
+      }
+      *(bf16x8*)&yr[i] = o;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void layernorm_bwd_dx(
+    const bf16* __restrict__ x, const bf16* __restrict__ dy,
+    bf16* __restrict__ dx, const float* __restrict__ gamma,
+    const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
+    long rows, int D) {
+  __shared__ float scratch[4];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const bf16* xr = x + row * D;
+    const bf16* gr = dy + row * D;
+    bf16* or_ = dx + row * D;
+    float mean = save_mean[row], rstd = save_rstd[row];
+    float c1 = 0, c2 = 0;
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 vx = *(const bf16x8*)&xr[i];
+      bf16x8 vg = *(const bf16x8*)&gr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float gg = bf2f(vg[j]) * gamma[i + j];
+        float xh = (bf2f(vx[j]) - mean) * rstd;
+        c1 += gg;
+        c2 += gg * xh;
+      }
+    }
+    c1 = block_red_sum(c1, scratch) / D;
+    c2 = block_red_sum(c2, scratch) / D;
+    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+      bf16x8 vx = *(const bf16x8*)&xr[i];
+      bf16x8 vg = *(const bf16x8*)&gr[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float gg = bf2f(vg[j]) * gamma[i + j];
+        float xh = (bf2f(vx[j]) - mean) * rstd;
+        o[j] = f2bf((gg - c1 - xh * c2) * rstd);
+      }
+      *(bf16x8*)&or_[i] = o;
+    }
+  }
+}
+
+// column reduce for dgamma/dbeta: dgamma = sum dy*xhat, dbeta = sum dy
+__global__ __launch_bounds__(256) void layernorm_bwd_dparams(
+    const bf16* __restrict__ x, const bf16* __restrict__ dy,
+    const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
+    float* __restrict__ dgamma, float* __restrict__ dbeta, long rows,
+    int D) {
+  // blocks over column chunks of 64; 256 thr = 32 rows x 8 chunks of 8
+  const int cblk = blockIdx.x * 64;
+  const int t = threadIdx.x;
+  const int chunk = t & 7, rlane = t >> 3;
+  const int c0 = cblk + chunk * 8;
+  const bool active = c0 + 8 <= D;
+  float sg[8] = {0}, sb[8] = {0};
+  for (long m = blockIdx.y * 32 + rlane; active && m < rows;
+       m += 32L * gridDim.y) {
+    float mean = save_mean[m], rstd = save_rstd[m];
+    bf16x8 vx = *(const bf16x8*)&x[m * D + c0];
+    bf16x8 vg = *(const bf16x8*)&dy[m * D + c0];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      float g = bf2f(vg[i]);
+      float xh = (bf2f(vx[i]) - mean) * rstd;
+      sg[i] += g * xh;
+      sb[i] += g;
+    }
+  }
+  __shared__ float red[2][32][64];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    red[0][rlane][chunk * 8 + i] = sg[i];
+    red[1][rlane][chunk * 8 + i] = sb[i];
+  }
+  __syncthreads();
+  if (t < 64) {
+    float fg = 0, fb = 0;
+#pragma unroll
+    for (int r = 0; r < 32; ++r) {
+      fg += red[0][r][t];
+      fb += red[1][r][t];
+    }
+    int c = cblk + t;
+    if (c < D) {
+      atomicAdd(&dgamma[c], fg);
+      atomicAdd(&dbeta[c], fb);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host launchers
+// ---------------------------------------------------------------------------
+
+static int rows_grid(long rows) {
+  return (int)(rows < 2048 ? (rows < 1 ? 1 : rows) : 2048);
+}
+
+void softmax_fwd_bf16(const bf16* x, bf16* y, long rows, int D, bool log_,
+                      hipStream_t s) {
+  if (log_)
+    softmax_fwd<true><<<rows_grid(rows), 256, 0, s>>>(x, y, rows, D);
+  else
+    softmax_fwd<false><<<rows_grid(rows), 256, 0, s>>>(x, y, rows, D);
+}
+
+void softmax_bwd_bf16(const bf16* y, const bf16* dy, bf16* dx, long rows,
+                      int D, bool log_, hipStream_t s) {
+  if (log_)
+    softmax_bwd<true><<<rows_grid(rows), 256, 0, s>>>(y, dy, dx, rows, D);
+  else
+    softmax_bwd<false><<<rows_grid(rows), 256, 0, s>>>(y, dy, dx, rows, D);
+}
+
+void layernorm_fwd_bf16(const bf16* x, bf16* y, const float* gamma,
+                        const float* beta, float* save_mean,
+                        float* save_rstd, long rows, int D, float eps,
+                        hipStream_t s) {
+  layernorm_fwd<<<rows_grid(rows), 256, 0, s>>>(x, y, gamma, beta,
+                                                save_mean, save_rstd, rows,
+                                                D, eps);
+}
+
+void layernorm_bwd_bf16(const bf16* x, const bf16* dy, bf16* dx,
+                        const float* gamma, const float* save_mean,
+                        const float* save_rstd, float* dgamma, float* dbeta,
+                        long rows, int D, hipStream_t s) {
+  layernorm_bwd_dx<<<rows_grid(rows), 256, 0, s>>>(
+      x, dy, dx, gamma, save_mean, save_rstd, rows, D);
+  HIP_CHECK(hipMemsetAsync(dgamma, 0, D * sizeof(float), s));
+  HIP_CHECK(hipMemsetAsync(dbeta, 0, D * sizeof(float), s));
+  int gy = (int)min((rows + 8191) / 8192, (long)64);
+  dim3 g(cdiv_h(D, 64), gy < 1 ? 1 : gy);
+  layernorm_bwd_dparams<<<g, 256, 0, s>>>(x, dy, save_mean, save_rstd,
+                                          dgamma, dbeta, rows, D);
+}
+
+}  // namespace mxa

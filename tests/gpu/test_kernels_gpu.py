@@ -1,0 +1,301 @@
+"""GPU numerics tests: every HIP kernel vs a plain torch fp32 reference
+(reference pattern: tests/python/gpu/test_operator_gpu.py check_consistency
+— the CPU/fp32 implementation is the oracle)."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from anand_mxnet_amd import _C
+else:  # collection on CPU boxes; tests are skipped by conftest
+    _C = None
+
+BF = torch.bfloat16
+
+
+def _dev():
+    return torch.device("cuda", 0)
+
+
+def _rt(shape, scale=1.0):
+    return (torch.rand(shape, device=_dev(), dtype=torch.float32) * 2 - 1) \
+        * scale
+
+
+def assert_close(a, b, rtol=3e-2, atol=3e-2, msg=""):
+    a = a.float().cpu()
+    b = b.float().cpu()
+    err = (a - b).abs().max().item()
+    denom = b.abs().max().item() + 1e-6
+    assert torch.allclose(a, b, rtol=rtol, atol=atol), \
+        f"{msg}: max abs err {err:.4e} (ref max {denom:.3e})"
+
+
+def test_gemm_tn_vs_fp32():
+    for (M, N, K) in [(128, 128, 64), (256, 512, 192), (100, 130, 64),
+                      (1024, 1000, 512)]:
+        a = _rt((M, K))
+        b = _rt((N, K))
+        ref = a @ b.t()
+        out = _C.gemm_tn(a.to(BF).contiguous(), b.to(BF).contiguous(),
+                         None, False)
+        assert_close(out, ref, msg=f"gemm {M}x{N}x{K}")
+
+
+def test_gemm_tn_bias_relu():
+    a, b = _rt((256, 128)), _rt((64, 128))
+    bias = _rt((64,))
+    ref = torch.relu(a @ b.t() + bias)
+    out = _C.gemm_tn(a.to(BF), b.to(BF), bias, True)
+    assert_close(out, ref, msg="gemm bias+relu")
+
+
+def test_gemm_tn_batched():
+    a, b = _rt((4, 128, 64)), _rt((4, 96, 64))
+    ref = a @ b.transpose(-1, -2)
+    out = _C.gemm_tn(a.to(BF).contiguous(), b.to(BF).contiguous(), None,
+                     False)
+    assert_close(out, ref, msg="batched gemm")
+
+
+def test_gemm_wg_vs_fp32():
+    for (m, Mo, No) in [(512, 128, 128), (1000, 64, 256), (4096, 256, 104)]:
+        a, b = _rt((m, Mo)), _rt((m, No))
+        ref = a.t() @ b
+        out = _C.gemm_wg(a.to(BF).contiguous(), b.to(BF).contiguous())
+        assert_close(out, ref, rtol=3e-2, atol=5e-2, msg=f"wg {m}x{Mo}x{No}")
+
+
+def _conv_ref(x, w, stride, pad):
+    # x NHWC f32, w (K,R,S,C): torch conv in NCHW
+    xc = x.permute(0, 3, 1, 2)
+    wc = w.permute(0, 3, 1, 2)
+    y = torch.nn.functional.conv2d(xc, wc, None, stride, pad)
+    return y.permute(0, 2, 3, 1)
+
+
+@pytest.mark.parametrize("cfg", [
+    ((2, 16, 16, 64), 64, 3, 1, 1),     # 3x3 s1
+    ((2, 16, 16, 64), 128, 3, 2, 1),    # 3x3 s2
+    ((2, 14, 14, 128), 256, 1, 1, 0),   # 1x1
+    ((2, 14, 14, 128), 256, 1, 2, 0),   # 1x1 s2
+])
+def test_conv2d_fwd(cfg):
+    xs, K, R, s, p = cfg
+    x = _rt(xs)
+    w = _rt((K, R, R, xs[3]), 0.2)
+    ref = _conv_ref(x, w, (s, s), (p, p))
+    w_rskc = w.to(BF).permute(1, 2, 0, 3).contiguous()
+    y = _C.conv2d_fwd(x.to(BF).contiguous(), w_rskc, None, [s, s], [p, p],
+                      [1, 1], False)
+    assert_close(y, ref, msg=f"conv fwd {cfg}")
+
+
+@pytest.mark.parametrize("cfg", [
+    ((2, 16, 16, 64), 64, 3, 1, 1),
+    ((2, 16, 16, 64), 128, 3, 2, 1),
+    ((2, 14, 14, 128), 256, 1, 2, 0),
+])
+def test_conv2d_dgrad_wgrad(cfg):
+    xs, K, R, s, p = cfg
+    x = _rt(xs).requires_grad_(True)
+    w = _rt((K, R, R, xs[3]), 0.2).requires_grad_(True)
+    y = _conv_ref(x, w, (s, s), (p, p))
+    dy = _rt(tuple(y.shape))
+    y.backward(dy)
+
+    w_rsck = w.detach().to(BF).permute(1, 2, 3, 0).contiguous()
+    dx = _C.conv2d_dgrad(dy.to(BF).contiguous(), w_rsck, list(xs), [s, s],
+                         [p, p], [1, 1])
+    assert_close(dx, x.grad, msg=f"conv dgrad {cfg}")
+
+    dwf = _C.conv2d_wgrad(x.detach().to(BF).contiguous(),
+                          dy.to(BF).contiguous(), [R, R], [s, s], [p, p],
+                          [1, 1])
+    dw_ref = w.grad.permute(1, 2, 3, 0)  # (R,S,C,K)
+    assert_close(dwf, dw_ref, rtol=3e-2, atol=5e-2, msg=f"conv wgrad {cfg}")
+
+
+def test_bn_train_fwd_bwd():
+    N, H, W, C = 4, 8, 8, 64
+    x = _rt((N, H, W, C))
+    gamma = torch.rand(C, device=_dev()) + 0.5
+    beta = _rt((C,))
+    rmean = torch.zeros(C, device=_dev())
+    rvar = torch.ones(C, device=_dev())
+    xt = x.clone().requires_grad_(True)
+    ref = torch.nn.functional.batch_norm(
+        xt.permute(0, 3, 1, 2), None, None, gamma, beta, True, 0.1,
+        1e-5).permute(0, 2, 3, 1)
+    y, sm, sr = _C.bn_fwd_train(x.to(BF).contiguous(), gamma, beta, rmean,
+                                rvar, 1e-5, 0.9, False)
+    assert_close(y, ref, msg="bn fwd")
+    # bwd
+    dy = _rt((N, H, W, C))
+    ref.backward(dy)
+    dx, dg, db = _C.bn_bwd(x.to(BF).contiguous(), dy.to(BF).contiguous(),
+                           gamma, beta, sm, sr, False)
+    assert_close(dx, xt.grad, rtol=5e-2, atol=5e-2, msg="bn dx")
+    # running stats: mxnet momentum 0.9 == torch momentum 0.1
+    ref_mean = 0.1 * x.reshape(-1, C).mean(0)
+    assert_close(rmean, ref_mean, rtol=2e-2, atol=2e-3, msg="bn rmean")
+
+
+def test_bn_relu_fused():
+    N, H, W, C = 2, 4, 4, 64
+    x = _rt((N, H, W, C))
+    gamma = torch.ones(C, device=_dev())
+    beta = torch.zeros(C, device=_dev())
+    rmean = torch.zeros(C, device=_dev())
+    rvar = torch.ones(C, device=_dev())
+    y, sm, sr = _C.bn_fwd_train(x.to(BF).contiguous(), gamma, beta, rmean,
+                                rvar, 1e-5, 0.9, True)
+    assert (y.float() >= 0).all()
+
+
+def test_maxpool():
+    x = _rt((2, 16, 16, 64))
+    xt = x.clone().requires_grad_(True)
+    ref = torch.nn.functional.max_pool2d(xt.permute(0, 3, 1, 2), 3, 2,
+                                         1).permute(0, 2, 3, 1)
+    y, arg = _C.maxpool(x.to(BF).contiguous(), [3, 3], [2, 2], [1, 1])
+    assert_close(y, ref, msg="maxpool fwd")
+    dy = _rt(tuple(ref.shape))
+    ref.backward(dy)
+    dx = _C.maxpool_bwd(dy.to(BF).contiguous(), arg, list(x.shape), [3, 3],
+                        [2, 2], [1, 1])
+    assert_close(dx, xt.grad, rtol=5e-2, atol=5e-2, msg="maxpool bwd")
+
+
+def test_global_avgpool():
+    x = _rt((3, 7, 7, 128))
+    ref = x.mean(dim=(1, 2), keepdim=True)
+    y = _C.global_avgpool(x.to(BF).contiguous())
+    assert_close(y, ref, msg="gap fwd")
+    dy = _rt((3, 1, 1, 128))
+    dx = _C.global_avgpool_bwd(dy.to(BF).contiguous(), list(x.shape))
+    ref_dx = (dy / 49.0).expand(3, 7, 7, 128)
+    assert_close(dx, ref_dx, msg="gap bwd")
+
+
+def test_softmax_fwd_bwd():
+    x = _rt((64, 128), 4.0)
+    xt = x.clone().requires_grad_(True)
+    ref = torch.softmax(xt, dim=-1)
+    y = _C.softmax_fwd(x.to(BF).contiguous(), False)
+    assert_close(y, ref, rtol=2e-2, atol=2e-3, msg="softmax fwd")
+    dy = _rt((64, 128))
+    ref.backward(dy)
+    dx = _C.softmax_bwd(y, dy.to(BF).contiguous(), False)
+    assert_close(dx, xt.grad, rtol=5e-2, atol=1e-2, msg="softmax bwd")
+
+
+def test_log_softmax():
+    x = _rt((32, 1000), 3.0)
+    # pad to %8 is satisfied (1000%8==0)
+    ref = torch.log_softmax(x, dim=-1)
+    y = _C.softmax_fwd(x.to(BF).contiguous(), True)
+    assert_close(y, ref, rtol=2e-2, atol=3e-2, msg="log_softmax")
+
+
+def test_layernorm_fwd_bwd():
+    x = _rt((128, 768), 2.0)
+    gamma = torch.rand(768, device=_dev()) + 0.5
+    beta = _rt((768,))
+    xt = x.clone().requires_grad_(True)
+    gt = gamma.clone().requires_grad_(True)
+    bt = beta.clone().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(xt, (768,), gt, bt, 1e-5)
+    y, mean, rstd = _C.layernorm_fwd(x.to(BF).contiguous(), gamma, beta,
+                                     1e-5)
+    assert_close(y, ref, msg="ln fwd")
+    dy = _rt((128, 768))
+    ref.backward(dy)
+    dx, dg, db = _C.layernorm_bwd(x.to(BF).contiguous(),
+                                  dy.to(BF).contiguous(), gamma, mean, rstd)
+    assert_close(dx, xt.grad, rtol=5e-2, atol=5e-2, msg="ln dx")
+    assert_close(dg, gt.grad, rtol=5e-2, atol=5e-1, msg="ln dgamma")
+    assert_close(db, bt.grad, rtol=5e-2, atol=5e-1, msg="ln dbeta")
+
+
+def test_dropout():
+    x = torch.ones((1024, 64), device=_dev(), dtype=BF)
+    y, mask = _C.dropout_fwd(x, 0.5, 1234)
+    keep = mask.float().mean().item()
+    assert 0.4 < keep < 0.6
+    yv = y.float()
+    assert ((yv == 0) | ((yv - 2.0).abs() < 1e-2)).all()
+    dy = torch.ones_like(x)
+    dx = _C.dropout_bwd(dy, mask, 0.5)
+    assert_close(dx, y, msg="dropout bwd == fwd pattern for ones")
+
+
+def test_embedding():
+    W = _rt((1000, 64))
+    idx = torch.randint(0, 1000, (4, 37), device=_dev())
+    ref = torch.nn.functional.embedding(idx, W)
+    out = _C.embedding_fwd(idx, W.to(BF).contiguous())
+    assert_close(out, ref, msg="embedding fwd")
+    dy = _rt((4, 37, 64))
+    dW = _C.embedding_bwd(idx, dy.to(BF).contiguous(), 1000)
+    ref_dW = torch.zeros_like(W)
+    ref_dW.index_add_(0, idx.flatten(), dy.reshape(-1, 64))
+    assert_close(dW, ref_dW, rtol=3e-2, atol=3e-2, msg="embedding bwd")
+
+
+def test_activations():
+    x = _rt((512, 64), 2.0)
+    for name, opid, fn in [
+        ("relu", 0, torch.relu), ("sigmoid", 1, torch.sigmoid),
+        ("tanh", 2, torch.tanh),
+        ("gelu", 3, torch.nn.functional.gelu),
+        ("silu", 4, torch.nn.functional.silu),
+    ]:
+        xt = x.clone().requires_grad_(True)
+        ref = fn(xt)
+        y = _C.act_fwd(opid, x.to(BF).contiguous())
+        assert_close(y, ref, msg=f"act {name} fwd")
+        dy = _rt((512, 64))
+        ref.backward(dy)
+        dx = _C.act_bwd(opid, x.to(BF).contiguous(), dy.to(BF).contiguous())
+        assert_close(dx, xt.grad, rtol=5e-2, atol=2e-2,
+                     msg=f"act {name} bwd")
+
+
+def test_multi_opt_sgd_mom():
+    torch.manual_seed(0)
+    shapes = [(64, 64), (1000,), (3, 3, 64, 64)]
+    ws32 = [torch.randn(s, device=_dev()) for s in shapes]
+    gs32 = [torch.randn(s, device=_dev()) for s in shapes]
+    ws = [w.to(BF) for w in ws32]
+    gs = [g.to(BF) for g in gs32]
+    moms = [torch.zeros(s, device=_dev()) for s in shapes]
+    w32s = [w.float().clone() for w in ws]
+    blob, meta = _C.build_opt_descs(ws, gs, moms, [], w32s)
+    ntens, nchunks = int(meta[0]), int(meta[1])
+    lr, mom, wd = 0.1, 0.9, 1e-4
+    _C.multi_opt_run(blob, ntens, nchunks, 1, True, lr, mom, 0.9, 0.999,
+                     1e-8, wd, 1.0, -1.0)
+    for w, g, w32 in zip(ws, gs, w32s):
+        gref = g.float() + wd * (w32 + 0)  # note: w32 already updated
+    # reference recompute from scratch
+    for i, s in enumerate(shapes):
+        w0 = ws32[i].to(BF).float()
+        g = gs[i].float()
+        m = -lr * (g + wd * w0)
+        ref = w0 + m
+        assert_close(w32s[i], ref, rtol=1e-4, atol=1e-4,
+                     msg=f"mp_sgd tensor {i}")
+        assert_close(ws[i].float(), ref, rtol=1e-2, atol=1e-2)
+
+
+def test_multi_sumsq():
+    ws = [torch.randn((100,), device=_dev()).to(BF),
+          torch.randn((70000,), device=_dev()).to(BF)]
+    blob, meta = _C.build_opt_descs(ws, ws, [], [], [])
+    out = _C.multi_sumsq_run(blob, int(meta[0]), int(meta[1]), True)
+    for i, w in enumerate(ws):
+        ref = w.float().pow(2).sum()
+        assert_close(out[i], ref, rtol=1e-2, atol=1e-2, msg=f"sumsq {i}")
