@@ -84,6 +84,23 @@ class Trainer:
         else:
             self._kvstore = kvs.create(
                 self._kvstore_type if self._kvstore_type else "local")
+            # single-process GPU: persistent flat grads so the fused
+            # multi-tensor optimizer kernel can cache its descriptors
+            try:
+                self._init_params_if_needed()
+                tensors = [arr._t for p in self._params
+                           if p.grad_req != "null"
+                           for arr in p.list_data()]
+                if tensors and all(t.is_cuda for t in tensors) and \
+                        len({t.device for t in tensors}) == 1:
+                    from ..parallel.bucketed import GradFlattener
+                    self._reducer = GradFlattener(tensors)
+                    for p in self._params:
+                        if p.grad_req != "null":
+                            for arr in p.list_data():
+                                arr._zero_in_place = True
+            except Exception:
+                pass
         self._kv_initialized = True
 
     def _init_params_if_needed(self):
@@ -134,12 +151,127 @@ class Trainer:
         self._update(ignore_stale_grad)
 
     def _update(self, ignore_stale_grad=False):
+        if self._try_fused_update():
+            return
         updater = self._updaters[0]
         for i, param in enumerate(self._params):
             if param.grad_req == "null":
                 continue
             for arr, grad in zip(param.list_data(), param.list_grad()):
                 updater(i, grad, arr)
+
+    # ---------------- fused multi-tensor optimizer path ----------------
+    # One HIP kernel per parameter group per step (reference parity:
+    # multi_mp_sgd_mom_update / MXNET_OPTIMIZER_AGGREGATION_SIZE — here the
+    # aggregation covers ALL params, chunked inside the kernel).
+    _fused = None
+
+    def _try_fused_update(self):
+        if self._fused is False:
+            return False
+        if self._fused is None:
+            self._fused = self._build_fused()
+            if self._fused is False:
+                return False
+        fused = self._fused
+        import torch as _t
+        # grads must still be the cached buffers (bucket views are stable)
+        for g, ptr in zip(fused["grads"], fused["gptrs"]):
+            if g.data_ptr() != ptr:
+                self._fused = None  # rebuild next time
+                return self._try_fused_update()
+        opt = self._optimizer
+        opt._update_count(fused["indices"])
+        tcount = opt._index_update_count[fused["indices"][0]]
+        from ..ops import hip as hipops
+        for grp in fused["groups"]:
+            lr = opt.learning_rate
+            if grp["mode"] >= 2:  # adam: fold bias correction into lr
+                import math
+                lr = lr * math.sqrt(1 - opt.beta2 ** tcount) \
+                    / (1 - opt.beta1 ** tcount)
+            hipops.multi_opt_run(
+                grp["blob"], grp["ntens"], grp["nchunks"], grp["mode"],
+                grp["wbf16"], lr, getattr(opt, "momentum", 0.0),
+                getattr(opt, "beta1", 0.9), getattr(opt, "beta2", 0.999),
+                getattr(opt, "epsilon", 1e-8), opt.wd, opt.rescale_grad,
+                opt.clip_gradient if opt.clip_gradient else -1.0)
+        return True
+
+    def _build_fused(self):
+        import torch as _t
+        try:
+            from ..ops import hip as hipops
+        except Exception:
+            return False
+        opt = self._optimizer
+        if type(opt).__name__ not in ("SGD", "Adam"):
+            return False
+        if opt.lr_mult or opt.wd_mult or opt.lr_scheduler is not None:
+            return False  # per-param lr/wd or schedule: per-param path
+        is_adam = type(opt).__name__ == "Adam"
+        updater = self._updaters[0]
+        groups = {}
+        grads, gptrs, indices = [], [], []
+        for i, p in enumerate(self._params):
+            if p.grad_req == "null":
+                continue
+            data = p.list_data()
+            if len(data) != 1 or not data[0]._t.is_cuda:
+                return False
+            arr = data[0]
+            w = arr._t.detach()
+            g = arr._t.grad
+            if g is None:
+                return False  # no backward yet; retry later
+            if i not in updater.states:
+                updater.states[i] = opt.create_state_multi_precision(i, arr)
+            st = updater.states[i]
+            mp = opt._needs_master(arr)
+            if is_adam:
+                if mp:
+                    w32, (m, v) = st
+                    mode = 3
+                else:
+                    m, v = st
+                    w32 = None
+                    mode = 2
+            else:
+                if mp:
+                    w32, m = st
+                    if m is None:
+                        m = _t.zeros_like(w32)
+                        updater.states[i] = (w32, m)
+                    mode = 1
+                else:
+                    m = st
+                    if m is None:
+                        m = _t.zeros_like(w, dtype=_t.float32)
+                        updater.states[i] = m
+                    w32 = None
+                    mode = 0
+                v = None
+            wbf16 = w.dtype == _t.bfloat16
+            key = (mode, wbf16)
+            grp = groups.setdefault(key, {"ws": [], "gs": [], "ms": [],
+                                          "vs": [], "w32s": []})
+            grp["ws"].append(w)
+            grp["gs"].append(g)
+            grp["ms"].append(m)
+            grp["vs"].append(v if v is not None else m)
+            grp["w32s"].append(w32 if w32 is not None else m)
+            grads.append(g)
+            gptrs.append(g.data_ptr())
+            indices.append(i)
+        built = []
+        for (mode, wbf16), grp in groups.items():
+            blob, meta = hipops.build_opt_descs(grp["ws"], grp["gs"],
+                                                grp["ms"], grp["vs"],
+                                                grp["w32s"])
+            built.append({"mode": mode, "wbf16": wbf16, "blob": blob,
+                          "ntens": int(meta[0]), "nchunks": int(meta[1])})
+        return {"groups": built, "grads": grads, "gptrs": gptrs,
+                "indices": indices}
 
     # ---------------- states ----------------
     def save_states(self, fname):

@@ -100,7 +100,7 @@ class _Linear(torch.autograd.Function):
 
 @register_gpu("linear")
 def linear(x, w, bias):
-    if x.size(-1) % 64 != 0 or (x.dim() == 2 and x.size(0) % 8 != 0):
+    if x.size(-1) % 64 != 0:
         return torch.nn.functional.linear(x, w, bias)
     x2 = x.reshape(-1, x.shape[-1])
     out = _Linear.apply(x2.contiguous(), w,

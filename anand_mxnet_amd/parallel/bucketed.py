@@ -53,19 +53,17 @@ class GradBucket:
         self.work = None
 
 
-class BucketedAllReducer:
-    """Overlap-capable gradient averaging over torch.distributed (RCCL)."""
+class GradFlattener:
+    """Persistent flat gradient storage (param.grad = bucket views).
+
+    Used on its own for single-process training (stable grad pointers ->
+    the fused multi-tensor optimizer kernel caches its descriptor blob),
+    and as the base of the distributed all-reducer."""
 
     def __init__(self, param_tensors: List[torch.Tensor],
-                 bucket_bytes: int = DEFAULT_BUCKET_BYTES,
-                 average: bool = True, overlap: bool = True):
-        assert dist.is_initialized()
-        self.world_size = dist.get_world_size()
-        self.average = average
-        self.overlap = overlap and all(t.is_cuda for t in param_tensors)
+                 bucket_bytes: int = DEFAULT_BUCKET_BYTES):
         self.buckets: List[GradBucket] = []
         self._t2bucket = {}
-
         # reverse order: backward computes last-registered params first
         rev = list(reversed(param_tensors))
         cur, cur_bytes = [], 0
@@ -79,16 +77,41 @@ class BucketedAllReducer:
         if cur:
             self._make_bucket(cur)
 
-        if self.overlap:
-            for t in param_tensors:
-                t.register_post_accumulate_grad_hook(self._hook)
-
     def _make_bucket(self, tensors):
         b = GradBucket(tensors, tensors[0].dtype, tensors[0].device)
         b.attach()
         self.buckets.append(b)
         for t in tensors:
             self._t2bucket[t] = b
+
+    def grad_view(self, t: torch.Tensor) -> torch.Tensor:
+        b = self._t2bucket[t]
+        return b.views[b.tensors.index(t)]
+
+    def zero_grads(self):
+        for b in self.buckets:
+            b.flat.zero_()
+            b.reset()
+            b.attach()
+
+    def finish(self):
+        pass
+
+
+class BucketedAllReducer(GradFlattener):
+    """Overlap-capable gradient averaging over torch.distributed (RCCL)."""
+
+    def __init__(self, param_tensors: List[torch.Tensor],
+                 bucket_bytes: int = DEFAULT_BUCKET_BYTES,
+                 average: bool = True, overlap: bool = True):
+        assert dist.is_initialized()
+        self.world_size = dist.get_world_size()
+        self.average = average
+        self.overlap = overlap and all(t.is_cuda for t in param_tensors)
+        super().__init__(param_tensors, bucket_bytes)
+        if self.overlap:
+            for t in param_tensors:
+                t.register_post_accumulate_grad_hook(self._hook)
 
     def _launch(self, bucket: GradBucket):
         if self.average:
@@ -100,12 +123,6 @@ class BucketedAllReducer:
         b = self._t2bucket[param]
         if b.ready():
             self._launch(b)
-
-    def zero_grads(self):
-        for b in self.buckets:
-            b.flat.zero_()
-            b.reset()
-            b.attach()  # re-attach in case autograd replaced .grad
 
     def finish(self):
         """Complete all reductions (called from Trainer before update)."""

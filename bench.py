@@ -1,0 +1,126 @@
+#!/usr/bin/env python3
+"""Flagship benchmark (driver contract): ResNet-50 v1.5 bf16 NHWC training
+step on synthetic ImageNet-shape data, N GPUs of one node via RCCL
+(one process per GPU; launched by torch.distributed.run for N>1).
+
+Prints ONE JSON line from rank 0:
+  {"metric": "samples/sec", "value": <whole-job>, ...}
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch-size", type=int, default=256,
+                    help="per-GPU batch size")
+    ap.add_argument("--model", default="resnet50_v1_5")
+    ap.add_argument("--image-size", type=int, default=224)
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world_size > 1
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl")
+
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import gluon, autograd, nd
+    from anand_mxnet_amd.gluon.model_zoo import vision as zoo
+
+    ctx = mx.gpu(local_rank)
+    torch.cuda.set_device(local_rank)
+
+    net = zoo.get_model(args.model, layout="NHWC")
+    net.cast("bfloat16")
+    net.initialize(mx.init.Xavier(), ctx=ctx)
+    loss_fn = gluon.loss.SoftmaxCrossEntropyLoss()
+    trainer = gluon.Trainer(
+        net.collect_params(), "sgd",
+        {"learning_rate": 0.1, "momentum": 0.9, "wd": 1e-4,
+         "multi_precision": True},
+        kvstore="device")
+
+    B, S = args.batch_size, args.image_size
+    global_batch = B * world_size
+    # synthetic ImageNet-shape data, random values (bench on random data —
+    # zero-filled inputs inflate clocks), resident on device
+    rng = np.random.RandomState(1234 + rank)
+    x = nd.array(rng.uniform(-1, 1, (B, S, S, 3)).astype(np.float32),
+                 ctx=ctx, dtype="bfloat16")
+    y = nd.array(rng.randint(0, 1000, (B,)).astype(np.int32), ctx=ctx)
+
+    def step():
+        with autograd.record():
+            out = net(x)
+            loss = loss_fn(out, y)
+        loss.backward()
+        trainer.step(global_batch)
+        return loss
+
+    # warmup
+    for _ in range(args.warmup):
+        loss = step()
+    torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines the job)
+    if distributed:
+        t = torch.tensor([elapsed], device=f"cuda:{local_rank}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    lval = float(loss.mean().asscalar())
+    samples_per_sec = global_batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "samples/sec (whole node) ResNet-50 ImageNet-shape "
+                      "bf16 at 1/2/4/8 MI355X",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/sec",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {"model": "resnet50_v1.5", "global_batch": global_batch,
+                       "image": S, "layout": "NHWC",
+                       "parallelism": f"dp{world_size}",
+                       "loss": round(lval, 4)},
+        }))
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

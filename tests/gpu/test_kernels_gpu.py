@@ -36,8 +36,8 @@ def assert_close(a, b, rtol=3e-2, atol=3e-2, msg=""):
 def test_gemm_tn_vs_fp32():
     for (M, N, K) in [(128, 128, 64), (256, 512, 192), (100, 130, 64),
                       (1024, 1000, 512)]:
-        a = _rt((M, K))
-        b = _rt((N, K))
+        a = _rt((M, K)).to(BF).float()   # round inputs: isolates accum error
+        b = _rt((N, K)).to(BF).float()
         ref = a @ b.t()
         out = _C.gemm_tn(a.to(BF).contiguous(), b.to(BF).contiguous(),
                          None, False)
@@ -62,7 +62,8 @@ def test_gemm_tn_batched():
 
 def test_gemm_wg_vs_fp32():
     for (m, Mo, No) in [(512, 128, 128), (1000, 64, 256), (4096, 256, 104)]:
-        a, b = _rt((m, Mo)), _rt((m, No))
+        a = _rt((m, Mo)).to(BF).float()
+        b = _rt((m, No)).to(BF).float()
         ref = a.t() @ b
         out = _C.gemm_wg(a.to(BF).contiguous(), b.to(BF).contiguous())
         assert_close(out, ref, rtol=3e-2, atol=5e-2, msg=f"wg {m}x{Mo}x{No}")
@@ -156,7 +157,7 @@ def test_bn_relu_fused():
 
 
 def test_maxpool():
-    x = _rt((2, 16, 16, 64))
+    x = _rt((2, 16, 16, 64)).to(BF).float()  # same quantization as kernel
     xt = x.clone().requires_grad_(True)
     ref = torch.nn.functional.max_pool2d(xt.permute(0, 3, 1, 2), 3, 2,
                                          1).permute(0, 2, 3, 1)
