@@ -166,10 +166,10 @@ class Block:
         for _, param in self.params.items():
             param.cast(dtype)
 
-    def __call__(self, *args):
+    def __call__(self, *args, **kwargs):
         for hook in self._forward_pre_hooks.values():
             hook(self, args)
-        out = self.forward(*args)
+        out = self.forward(*args, **kwargs)
         for hook in self._forward_hooks.values():
             hook(self, args, out)
         return out

@@ -150,6 +150,8 @@ class Parameter:
                 ctx = self._deferred_init[1]
             elif ctx is None:
                 ctx = [cpu()]
+            if isinstance(ctx, Context):
+                ctx = [ctx]
             self._init_impl(None, ctx, None, data.astype(self.dtype)
                             if not cast_dtype else data)
         else:

@@ -40,18 +40,25 @@ class _GemmTN(torch.autograd.Function):
     def backward(ctx, dy):
         a, b = ctx.saved_tensors
         dy = dy.contiguous()
+        N = b.shape[-2]
+        M = a.shape[-2]
         da = db = None
         if ctx.needs_input_grad[0]:
             # dA[M,K] = dY[M,N] @ B[N,K] -> TN with B^T[K,N]
-            da = _C.gemm_tn(dy, b.transpose(-1, -2).contiguous(), None,
-                            False)
+            if N % 64 == 0:
+                da = _C.gemm_tn(dy, b.transpose(-1, -2).contiguous(), None,
+                                False)
+            else:  # non-tileable reduce dim: library GEMM
+                da = torch.matmul(dy, b)
         if ctx.needs_input_grad[1]:
             if a.dim() == 2:
                 db = _C.gemm_wg(dy, a).to(a.dtype)  # [N,K]
-            else:
+            elif M % 64 == 0:
                 db = _C.gemm_tn(dy.transpose(-1, -2).contiguous(),
                                 a.transpose(-1, -2).contiguous(), None,
                                 False)
+            else:
+                db = torch.matmul(dy.transpose(-1, -2), a)
         return da, db
 
 
@@ -90,7 +97,10 @@ class _Linear(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = _C.gemm_tn(dy, w.t().contiguous(), None, False)
+            if w.shape[0] % 64 == 0:
+                dx = _C.gemm_tn(dy, w.t().contiguous(), None, False)
+            else:  # e.g. the 1000-class head: library GEMM for dgrad
+                dx = torch.matmul(dy, w)
         if ctx.needs_input_grad[1]:
             dw = _C.gemm_wg(dy, x).to(w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
@@ -501,3 +511,59 @@ def multi_opt_run(blob, ntens, nchunks, mode, wbf16, lr, momentum, beta1,
 
 def multi_sumsq(blob, ntens, nchunks, bf16in):
     return _C.multi_sumsq_run(blob, ntens, nchunks, bf16in)
+
+
+# ---------------------------------------------------------------------------
+# Multi-head attention (composed from the MFMA batched GEMM + masked-softmax
+# kernels; reference parity: contrib/transformer.cu interleaved matmuls.
+# A fully-fused flash-style kernel is the planned fast path.)
+# ---------------------------------------------------------------------------
+
+class _MaskedSoftmax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, vlen):
+        y = _C.softmax_masked_fwd(x.contiguous(), vlen)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        # masked cols have y == 0 -> zero grad automatically
+        return _C.softmax_bwd(y, dy.contiguous(), False), None
+
+
+_causal_vlen_cache = {}
+
+
+def _causal_vlen(BH, T, device):
+    key = (BH, T, device)
+    v = _causal_vlen_cache.get(key)
+    if v is None:
+        v = torch.arange(1, T + 1, dtype=torch.int32,
+                         device=device).repeat(BH)
+        _causal_vlen_cache[key] = v
+    return v
+
+
+@register_gpu("mha")
+def mha(q, k, v, num_heads, causal, scale):
+    B, Tq, HD = q.shape
+    Tk = k.shape[1]
+    D = HD // num_heads
+    qh = q.reshape(B, Tq, num_heads, D).transpose(1, 2) \
+        .reshape(B * num_heads, Tq, D).contiguous()
+    kh = k.reshape(B, Tk, num_heads, D).transpose(1, 2) \
+        .reshape(B * num_heads, Tk, D).contiguous()
+    vh = v.reshape(B, Tk, num_heads, D).transpose(1, 2) \
+        .reshape(B * num_heads, Tk, D).contiguous()
+    s = _GemmTN.apply(qh, kh) * scale            # [BH, Tq, Tk]
+    if causal:
+        p = _MaskedSoftmax.apply(s, _causal_vlen(B * num_heads, Tq,
+                                                 s.device))
+    else:
+        p = _Softmax.apply(s, False)
+    vt = vh.transpose(-1, -2).contiguous()       # [BH, D, Tk]
+    out = _GemmTN.apply(p.contiguous(), vt)      # [BH, Tq, D]
+    return out.reshape(B, num_heads, Tq, D).transpose(1, 2) \
+        .reshape(B, Tq, HD)

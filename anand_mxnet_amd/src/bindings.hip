@@ -50,6 +50,8 @@ void gap_nhwc(const bf16*, bf16*, int, long, int, hipStream_t);
 void gap_nhwc_bwd(const bf16*, bf16*, int, long, int, hipStream_t);
 // softmax_norm.hip
 void softmax_fwd_bf16(const bf16*, bf16*, long, int, bool, hipStream_t);
+void softmax_masked_fwd_bf16(const bf16*, bf16*, long, int, const int*,
+                             hipStream_t);
 void softmax_bwd_bf16(const bf16*, const bf16*, bf16*, long, int, bool,
                       hipStream_t);
 void layernorm_fwd_bf16(const bf16*, bf16*, const float*, const float*,
@@ -349,6 +351,17 @@ at::Tensor softmax_fwd(const at::Tensor& x, bool log_) {
   return y;
 }
 
+at::Tensor softmax_masked_fwd(const at::Tensor& x, const at::Tensor& vlen) {
+  int D = x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "softmax: last dim must be %8");
+  long rows = x.numel() / D;
+  TORCH_CHECK(vlen.numel() == rows && vlen.scalar_type() == at::kInt);
+  auto y = at::empty_like(x);
+  mxa::softmax_masked_fwd_bf16(bfp(x), bfp_mut(y), rows, D,
+                               vlen.data_ptr<int>(), cur_stream());
+  return y;
+}
+
 at::Tensor softmax_bwd(const at::Tensor& y, const at::Tensor& dy,
                        bool log_) {
   int D = y.size(-1);
@@ -539,6 +552,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("global_avgpool", &global_avgpool);
   m.def("global_avgpool_bwd", &global_avgpool_bwd);
   m.def("softmax_fwd", &softmax_fwd);
+  m.def("softmax_masked_fwd", &softmax_masked_fwd);
   m.def("softmax_bwd", &softmax_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);

@@ -44,25 +44,31 @@ DEVINL float block_red_max(float v, float* scratch) {
 // softmax / log_softmax forward: one 256-thr block per row (D % 8 == 0)
 // ---------------------------------------------------------------------------
 
+// vlen: optional per-row valid length (masked/causal attention); columns
+// >= vlen[row] get probability 0 (log: -inf -> stored as large negative).
 template <bool LOG>
 __global__ __launch_bounds__(256) void softmax_fwd(
-    const bf16* __restrict__ x, bf16* __restrict__ y, long rows, int D) {
+    const bf16* __restrict__ x, bf16* __restrict__ y, long rows, int D,
+    const int* __restrict__ vlen) {
   __shared__ float scratch[4];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const bf16* xr = x + row * D;
     bf16* yr = y + row * D;
+    const int V = vlen ? vlen[row] : D;
     float mx = -3.0e38f;
     for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
       bf16x8 v = *(const bf16x8*)&xr[i];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) mx = fmaxf(mx, bf2f(v[j]));
+      for (int j = 0; j < 8; ++j)
+        if (i + j < V) mx = fmaxf(mx, bf2f(v[j]));
     }
     mx = block_red_max(mx, scratch);
     float sum = 0;
     for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
       bf16x8 v = *(const bf16x8*)&xr[i];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) sum += __expf(bf2f(v[j]) - mx);
+      for (int j = 0; j < 8; ++j)
+        if (i + j < V) sum += __expf(bf2f(v[j]) - mx);
     }
     sum = block_red_sum(sum, scratch);
     float inv = 1.f / sum, lse = __logf(sum);
@@ -71,8 +77,12 @@ __global__ __launch_bounds__(256) void softmax_fwd(
       bf16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float e = bf2f(v[j]) - mx;
-        o[j] = f2bf(LOG ? e - lse : __expf(e) * inv);
+        if (i + j >= V) {
+          o[j] = LOG ? f2bf(-3.0e38f) : (bf16)0.f;
+        } else {
+          float e = bf2f(v[j]) - mx;
+          o[j] = f2bf(LOG ? e - lse : __expf(e) * inv);
+        }
       }
       *(bf16x8*)&yr[i] = o;
     }
@@ -256,9 +266,16 @@ static int rows_grid(long rows) {
 void softmax_fwd_bf16(const bf16* x, bf16* y, long rows, int D, bool log_,
                       hipStream_t s) {
   if (log_)
-    softmax_fwd<true><<<rows_grid(rows), 256, 0, s>>>(x, y, rows, D);
+    softmax_fwd<true><<<rows_grid(rows), 256, 0, s>>>(x, y, rows, D,
+                                                      nullptr);
   else
-    softmax_fwd<false><<<rows_grid(rows), 256, 0, s>>>(x, y, rows, D);
+    softmax_fwd<false><<<rows_grid(rows), 256, 0, s>>>(x, y, rows, D,
+                                                       nullptr);
+}
+
+void softmax_masked_fwd_bf16(const bf16* x, bf16* y, long rows, int D,
+                             const int* vlen, hipStream_t s) {
+  softmax_fwd<false><<<rows_grid(rows), 256, 0, s>>>(x, y, rows, D, vlen);
 }
 
 void softmax_bwd_bf16(const bf16* y, const bf16* dy, bf16* dx, long rows,

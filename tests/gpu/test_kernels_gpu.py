@@ -85,8 +85,8 @@ def _conv_ref(x, w, stride, pad):
 ])
 def test_conv2d_fwd(cfg):
     xs, K, R, s, p = cfg
-    x = _rt(xs)
-    w = _rt((K, R, R, xs[3]), 0.2)
+    x = _rt(xs).to(BF).float()
+    w = _rt((K, R, R, xs[3]), 0.2).to(BF).float()
     ref = _conv_ref(x, w, (s, s), (p, p))
     w_rskc = w.to(BF).permute(1, 2, 0, 3).contiguous()
     y = _C.conv2d_fwd(x.to(BF).contiguous(), w_rskc, None, [s, s], [p, p],
@@ -101,10 +101,10 @@ def test_conv2d_fwd(cfg):
 ])
 def test_conv2d_dgrad_wgrad(cfg):
     xs, K, R, s, p = cfg
-    x = _rt(xs).requires_grad_(True)
-    w = _rt((K, R, R, xs[3]), 0.2).requires_grad_(True)
+    x = _rt(xs).to(BF).float().requires_grad_(True)
+    w = _rt((K, R, R, xs[3]), 0.2).to(BF).float().requires_grad_(True)
     y = _conv_ref(x, w, (s, s), (p, p))
-    dy = _rt(tuple(y.shape))
+    dy = _rt(tuple(y.shape)).to(BF).float()
     y.backward(dy)
 
     w_rsck = w.detach().to(BF).permute(1, 2, 3, 0).contiguous()

@@ -6,8 +6,12 @@ import argparse
 import json
 import time
 
+import os
+import sys
+
 import torch
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from anand_mxnet_amd import _C
 
 BF = torch.bfloat16
