@@ -42,6 +42,10 @@ class GradBucket:
 
     def attach(self):
         for t, v in zip(self.tensors, self.views):
+            if t.grad is not None and t.grad.data_ptr() != v.data_ptr():
+                # lazy init after a backward already produced grads:
+                # preserve them in the persistent view
+                v.copy_(t.grad)
             t.grad = v
 
     def ready(self):

@@ -1,0 +1,145 @@
+"""Metric + loss tests (reference: test_metric.py / test_loss.py)."""
+import numpy as np
+import pytest
+
+import anand_mxnet_amd as mx
+from anand_mxnet_amd import nd, metric, gluon, autograd
+from anand_mxnet_amd.test_utils import assert_almost_equal
+
+
+def test_accuracy():
+    m = metric.Accuracy()
+    pred = nd.array([[0.1, 0.9], [0.8, 0.2], [0.3, 0.7]])
+    label = nd.array([1, 0, 0])
+    m.update([label], [pred])
+    name, acc = m.get()
+    assert name == "accuracy"
+    assert abs(acc - 2.0 / 3) < 1e-6
+
+
+def test_topk():
+    m = metric.TopKAccuracy(top_k=2)
+    pred = nd.array([[0.1, 0.5, 0.4], [0.6, 0.3, 0.1]])
+    label = nd.array([2, 1])
+    m.update([label], [pred])
+    assert abs(m.get()[1] - 1.0) < 1e-6
+
+
+def test_f1_mse_rmse_mae():
+    f1 = metric.F1()
+    f1.update([nd.array([1, 0, 1])], [nd.array([[0.2, 0.8], [0.9, 0.1],
+                                                [0.4, 0.6]])])
+    assert f1.get()[1] == 1.0
+    for cls, val in [(metric.MSE, 0.01), (metric.RMSE, 0.1),
+                     (metric.MAE, 0.1)]:
+        m = cls()
+        m.update([nd.zeros((4,))], [nd.full((4,), 0.1)])
+        assert abs(m.get()[1] - val) < 1e-6
+
+
+def test_perplexity_ce():
+    pred = nd.array([[0.5, 0.5], [0.9, 0.1]])
+    label = nd.array([0, 0])
+    ce = metric.CrossEntropy()
+    ce.update([label], [pred])
+    expected = -(np.log(0.5) + np.log(0.9)) / 2
+    assert abs(ce.get()[1] - expected) < 1e-5
+    p = metric.Perplexity(ignore_label=None)
+    p.update([label], [pred])
+    assert abs(p.get()[1] - np.exp(expected)) < 1e-4
+
+
+def test_composite_and_custom():
+    comp = metric.CompositeEvalMetric()
+    comp.add(metric.Accuracy())
+    comp.add(metric.CrossEntropy())
+    pred = nd.array([[0.0, 1.0]])
+    comp.update([nd.array([1])], [pred])
+    names, vals = comp.get()
+    assert len(names) == 2
+    m = metric.np_fn(lambda l, p: float(np.sum(l == p.argmax(-1))),
+                     name="mymetric")
+    m.update([nd.array([1])], [pred])
+    assert m.get()[1] == 1.0
+
+
+def test_metric_create_by_name():
+    assert isinstance(metric.create("acc"), metric.Accuracy)
+    assert isinstance(metric.create(["acc", "mse"]),
+                      metric.CompositeEvalMetric)
+
+
+# ---------------- losses ----------------
+
+def _check_loss_grad(loss_fn, pred_shape, label, **kw):
+    pred = mx.test_utils.rand_ndarray(pred_shape)
+    pred.attach_grad()
+    with autograd.record():
+        loss = loss_fn(pred, label)
+    loss.backward()
+    assert loss.shape[0] == pred_shape[0]
+    assert np.all(np.isfinite(pred.grad.asnumpy()))
+    return loss
+
+
+def test_softmax_ce_loss():
+    loss_fn = gluon.loss.SoftmaxCrossEntropyLoss()
+    label = nd.array([1, 0, 2, 1])
+    loss = _check_loss_grad(loss_fn, (4, 5), label)
+    # cross-check vs numpy
+    pred = nd.array(np.random.randn(4, 5).astype(np.float32))
+    l = loss_fn(pred, label).asnumpy()
+    p = pred.asnumpy()
+    e = np.exp(p - p.max(-1, keepdims=True))
+    sm = e / e.sum(-1, keepdims=True)
+    ref = -np.log(sm[np.arange(4), label.asnumpy().astype(int)])
+    assert_almost_equal(l, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_l1_l2_huber():
+    label = nd.zeros((4, 3))
+    for fn in [gluon.loss.L2Loss(), gluon.loss.L1Loss(),
+               gluon.loss.HuberLoss()]:
+        _check_loss_grad(fn, (4, 3), label)
+
+
+def test_sigmoid_bce():
+    label = nd.array(np.random.randint(0, 2, (4, 3)).astype(np.float32))
+    loss_fn = gluon.loss.SigmoidBCELoss()
+    l = _check_loss_grad(loss_fn, (4, 3), label)
+    pred = nd.zeros((2, 2))
+    l0 = loss_fn(pred, nd.zeros((2, 2))).asnumpy()
+    assert_almost_equal(l0, np.full((2,), np.log(2)), rtol=1e-5, atol=1e-6)
+
+
+def test_kl_hinge_logistic_triplet():
+    _check_loss_grad(gluon.loss.KLDivLoss(),
+                     (4, 5), nd.softmax(mx.test_utils.rand_ndarray((4, 5))))
+    label = nd.array(np.random.choice([-1, 1], (4, 3)).astype(np.float32))
+    _check_loss_grad(gluon.loss.HingeLoss(), (4, 3), label)
+    _check_loss_grad(gluon.loss.SquaredHingeLoss(), (4, 3), label)
+    _check_loss_grad(gluon.loss.LogisticLoss(), (4, 3), label)
+    t = gluon.loss.TripletLoss()
+    p = mx.test_utils.rand_ndarray((4, 8))
+    pos = mx.test_utils.rand_ndarray((4, 8))
+    neg = mx.test_utils.rand_ndarray((4, 8))
+    p.attach_grad()
+    with autograd.record():
+        loss = t(p, pos, neg)
+    loss.backward()
+    assert np.all(np.isfinite(p.grad.asnumpy()))
+
+
+def test_softmax_output_op():
+    """Legacy fused softmax+CE gradient op (Module-era classifier)."""
+    data = mx.test_utils.rand_ndarray((4, 10))
+    label = nd.array([0, 3, 2, 9])
+    data.attach_grad()
+    with autograd.record():
+        out = nd.SoftmaxOutput(data, label, normalization="batch")
+        s = out.sum()
+    s.backward()
+    prob = out.asnumpy()
+    oh = np.eye(10)[label.asnumpy().astype(int)]
+    assert_almost_equal(data.grad.asnumpy(), (prob - oh) / 4,
+                        rtol=1e-4, atol=1e-5)
