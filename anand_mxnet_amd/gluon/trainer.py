@@ -86,21 +86,18 @@ class Trainer:
                 self._kvstore_type if self._kvstore_type else "local")
             # single-process GPU: persistent flat grads so the fused
             # multi-tensor optimizer kernel can cache its descriptors
-            try:
-                self._init_params_if_needed()
-                tensors = [arr._t for p in self._params
-                           if p.grad_req != "null"
-                           for arr in p.list_data()]
-                if tensors and all(t.is_cuda for t in tensors) and \
-                        len({t.device for t in tensors}) == 1:
-                    from ..parallel.bucketed import GradFlattener
-                    self._reducer = GradFlattener(tensors)
-                    for p in self._params:
-                        if p.grad_req != "null":
-                            for arr in p.list_data():
-                                arr._zero_in_place = True
-            except Exception:
-                pass
+            self._init_params_if_needed()
+            tensors = [arr._t for p in self._params
+                       if p.grad_req != "null"
+                       for arr in p.list_data()]
+            if tensors and all(t.is_cuda for t in tensors) and \
+                    len({t.device for t in tensors}) == 1:
+                from ..parallel.bucketed import GradFlattener
+                self._reducer = GradFlattener(tensors)
+                for p in self._params:
+                    if p.grad_req != "null":
+                        for arr in p.list_data():
+                            arr._zero_in_place = True
         self._kv_initialized = True
 
     def _init_params_if_needed(self):

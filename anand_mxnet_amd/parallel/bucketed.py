@@ -68,12 +68,14 @@ class GradFlattener:
                  bucket_bytes: int = DEFAULT_BUCKET_BYTES):
         self.buckets: List[GradBucket] = []
         self._t2bucket = {}
-        # reverse order: backward computes last-registered params first
+        # reverse order: backward computes last-registered params first.
+        # Buckets are homogeneous in dtype (bf16 weights vs fp32 BN params).
         rev = list(reversed(param_tensors))
         cur, cur_bytes = [], 0
         for t in rev:
             nbytes = t.numel() * t.element_size()
-            if cur and cur_bytes + nbytes > bucket_bytes:
+            if cur and (cur_bytes + nbytes > bucket_bytes
+                        or t.dtype != cur[0].dtype):
                 self._make_bucket(cur)
                 cur, cur_bytes = [], 0
             cur.append(t)
