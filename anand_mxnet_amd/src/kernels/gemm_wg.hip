@@ -91,12 +91,15 @@ struct WgIm2colRows {
     }
   }
   DEVINL const bf16* src(int slot, int colchunk) const {
+    int c = col0 + colchunk * 8;
     int n = n_[slot];
-    if (n >= Nb) return zpage + colchunk * 8;
+    // column tail (tile wider than C, e.g. C=64 under a 128-wide tile)
+    // must read the zero page — past-row reads can cross the allocation.
+    if (n >= Nb || c + 8 > C) return zpage + colchunk * 8;
     int ih = p_[slot] * sh - ph + roff;
     int iw = q_[slot] * sw - pw + soff;
     if (ih < 0 || ih >= H || iw < 0 || iw >= W) return zpage + colchunk * 8;
-    return x + (((long)n * H + ih) * W + iw) * C + col0 + colchunk * 8;
+    return x + (((long)n * H + ih) * W + iw) * C + c;
   }
 };
 
