@@ -105,7 +105,7 @@ struct WgIm2colRows {
 };
 
 template <bool A_IM2COL>
-__global__ __launch_bounds__(WTHREADS, 1) void gemm_wg_core(
+__global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_core(
     WgPlainRows aplain, WgIm2colRows aim, WgPlainRows bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
     int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic) {

@@ -50,7 +50,7 @@ def main():
     loss_fn = gluon.loss.SoftmaxCrossEntropyLoss()
     trainer = gluon.Trainer(
         net.collect_params(), "sgd",
-        {"learning_rate": 0.1, "momentum": 0.9, "wd": 1e-4,
+        {"learning_rate": 0.02, "momentum": 0.9, "wd": 1e-4,
          "multi_precision": True},
         kvstore="device")
 

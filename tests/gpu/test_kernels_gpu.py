@@ -279,8 +279,6 @@ def test_multi_opt_sgd_mom():
     lr, mom, wd = 0.1, 0.9, 1e-4
     _C.multi_opt_run(blob, ntens, nchunks, 1, True, lr, mom, 0.9, 0.999,
                      1e-8, wd, 1.0, -1.0)
-    for w, g, w32 in zip(ws, gs, w32s):
-        gref = g.float() + wd * (w32 + 0)  # note: w32 already updated
     # reference recompute from scratch
     for i, s in enumerate(shapes):
         w0 = ws32[i].to(BF).float()

@@ -89,3 +89,47 @@ def test_mlp_train_gpu():
         trainer.step(64)
         losses.append(float(loss.mean().asscalar()))
     assert losses[-1] < losses[0] * 0.9, losses
+
+
+def test_resnet18_grad_matches_cpu():
+    """Backward cross-check: GPU bf16 HIP grads vs CPU fp32 torch grads
+    (catches systematic dgrad/wgrad errors unit tolerances can hide)."""
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import gluon, nd, autograd
+    from anand_mxnet_amd.gluon.model_zoo.vision import resnet18_v1
+    import os, tempfile
+
+    np.random.seed(3)
+    net = resnet18_v1(layout="NHWC")
+    net.initialize(mx.init.Xavier(), ctx=mx.cpu())
+    f = os.path.join(tempfile.mkdtemp(), "w.params")
+    net.save_parameters(f)
+    x_np = np.random.uniform(-1, 1, (4, 64, 64, 3)).astype(np.float32)
+    y_np = np.random.randint(0, 1000, (4,))
+    loss_fn = gluon.loss.SoftmaxCrossEntropyLoss()
+
+    def run(ctx, dtype):
+        net2 = resnet18_v1(layout="NHWC")
+        net2.load_parameters(f, ctx=ctx)
+        if dtype != "float32":
+            net2.cast(dtype)
+        x = nd.array(x_np, ctx=ctx, dtype=dtype)
+        y = nd.array(y_np, ctx=ctx)
+        with autograd.record():
+            loss = loss_fn(net2(x), y).mean()
+        loss.backward()
+        return {name: p.grad(ctx).asnumpy().astype(np.float64)
+                for name, p in net2.collect_params().items()
+                if p.grad_req != "null"}
+
+    g_cpu = run(mx.cpu(), "float32")
+    g_gpu = run(mx.gpu(0), "bfloat16")
+    bad = []
+    for name in g_cpu:
+        a, b = g_cpu[name].ravel(), g_gpu[name].ravel()
+        denom = np.abs(a).max() + 1e-8
+        corr = np.corrcoef(a, b)[0, 1] if a.std() > 0 else 1.0
+        relerr = np.abs(a - b).max() / denom
+        if corr < 0.98 or relerr > 0.35:
+            bad.append((name, corr, relerr))
+    assert not bad, f"grad mismatches: {bad[:8]}"

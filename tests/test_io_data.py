@@ -1,0 +1,155 @@
+"""IO tests: NDArrayIter, RecordIO byte format, DataLoader (incl. worker
+processes), datasets/transforms (reference: test_io.py, test_recordio.py,
+test_gluon_data.py)."""
+import os
+import struct
+
+import numpy as np
+import pytest
+
+import anand_mxnet_amd as mx
+from anand_mxnet_amd import nd, recordio
+from anand_mxnet_amd.io import NDArrayIter, ResizeIter, PrefetchingIter, \
+    DataDesc
+from anand_mxnet_amd.gluon.data import (ArrayDataset, DataLoader,
+                                        SimpleDataset, SyntheticDataset)
+
+
+def test_ndarray_iter():
+    x = np.arange(40).reshape(10, 4).astype(np.float32)
+    y = np.arange(10).astype(np.float32)
+    it = NDArrayIter(x, y, batch_size=3, last_batch_handle="pad")
+    batches = list(it)
+    assert len(batches) == 4
+    assert batches[0].data[0].shape == (3, 4)
+    assert batches[-1].pad == 2
+    it.reset()
+    assert len(list(it)) == 4
+    descs = it.provide_data
+    assert isinstance(descs[0], DataDesc)
+    assert descs[0].shape == (3, 4)
+
+
+def test_ndarray_iter_discard_shuffle():
+    x = np.arange(10).astype(np.float32)
+    it = NDArrayIter(x, None, batch_size=3, shuffle=True,
+                     last_batch_handle="discard")
+    batches = list(it)
+    assert len(batches) == 3
+
+
+def test_resize_prefetch():
+    x = np.arange(8).astype(np.float32)
+    base = NDArrayIter(x, None, batch_size=4)
+    r = ResizeIter(base, 5)
+    assert len(list(r)) == 5
+    base.reset()
+    p = PrefetchingIter(NDArrayIter(x, None, batch_size=4))
+    assert len(list(p)) == 2
+    p.reset()
+    assert len(list(p)) == 2
+
+
+def test_recordio_roundtrip(tmp_path):
+    f = str(tmp_path / "t.rec")
+    w = recordio.MXRecordIO(f, "w")
+    for i in range(5):
+        w.write(f"record{i}".encode() * (i + 1))
+    w.close()
+    r = recordio.MXRecordIO(f, "r")
+    for i in range(5):
+        assert r.read() == f"record{i}".encode() * (i + 1)
+    assert r.read() is None
+    # byte-format check: magic
+    raw = open(f, "rb").read()
+    assert struct.unpack("<I", raw[:4])[0] == 0xCED7230A
+
+
+def test_indexed_recordio(tmp_path):
+    f = str(tmp_path / "t.rec")
+    idx = str(tmp_path / "t.idx")
+    w = recordio.MXIndexedRecordIO(idx, f, "w")
+    for i in range(5):
+        w.write_idx(i, f"rec{i}".encode())
+    w.close()
+    r = recordio.MXIndexedRecordIO(idx, f, "r")
+    assert r.read_idx(3) == b"rec3"
+    assert r.read_idx(0) == b"rec0"
+    assert r.keys == [0, 1, 2, 3, 4]
+
+
+def test_recordio_pack_unpack():
+    hdr = recordio.IRHeader(0, 3.0, 7, 0)
+    s = recordio.pack(hdr, b"payload")
+    h2, data = recordio.unpack(s)
+    assert h2.label == 3.0 and h2.id == 7 and data == b"payload"
+    # float-array label
+    hdr = recordio.IRHeader(0, np.array([1.0, 2.0], np.float32), 1, 0)
+    s = recordio.pack(hdr, b"x")
+    h2, data = recordio.unpack(s)
+    assert h2.flag == 2 and list(h2.label) == [1.0, 2.0]
+
+
+def test_pack_img_raw(tmp_path):
+    img = np.random.randint(0, 255, (4, 6, 3)).astype(np.uint8)
+    s = recordio.pack_img(recordio.IRHeader(0, 1.0, 0, 0), img)
+    h, img2 = recordio.unpack_img(s)
+    np.testing.assert_array_equal(img, img2)
+
+
+def test_dataloader_basic():
+    ds = ArrayDataset(np.arange(20).reshape(10, 2).astype(np.float32),
+                      np.arange(10).astype(np.int32))
+    dl = DataLoader(ds, batch_size=4, shuffle=False, last_batch="keep")
+    batches = list(dl)
+    assert len(batches) == 3
+    data, label = batches[0]
+    assert data.shape == (4, 2) and label.shape == (4,)
+    assert label.asnumpy().tolist() == [0, 1, 2, 3]
+
+
+def test_dataloader_workers():
+    ds = ArrayDataset(np.arange(64).reshape(32, 2).astype(np.float32))
+    dl = DataLoader(ds, batch_size=8, num_workers=2)
+    batches = list(dl)
+    assert len(batches) == 4
+    total = np.concatenate([b.asnumpy() for b in batches])
+    np.testing.assert_array_equal(total, np.arange(64).reshape(32, 2))
+
+
+def test_dataset_transform():
+    ds = SimpleDataset(list(range(8))).transform(lambda x: x * 2)
+    assert ds[3] == 6
+    ds2 = ArrayDataset(np.ones((4, 2), np.float32),
+                       np.zeros(4, np.int32)).transform_first(
+        lambda x: x + 1)
+    d, l = ds2[0]
+    assert float(d[0]) == 2.0
+
+
+def test_synthetic_dataset():
+    ds = SyntheticDataset([(3, 8, 8), ()], [np.float32, np.int64],
+                          length=16, classes=10)
+    img, lbl = ds[0]
+    assert img.shape == (3, 8, 8)
+    assert 0 <= int(lbl) < 10
+    assert len(ds) == 16
+
+
+def test_mnist_iter_from_raw(tmp_path):
+    # synthesize tiny idx files (reference MNISTIter reads raw idx)
+    imgs = np.random.randint(0, 255, (6, 28, 28)).astype(np.uint8)
+    labels = np.arange(6).astype(np.uint8)
+    ipath = str(tmp_path / "img-idx3-ubyte")
+    lpath = str(tmp_path / "lab-idx1-ubyte")
+    with open(ipath, "wb") as f:
+        f.write(struct.pack(">IIII", 2051, 6, 28, 28))
+        f.write(imgs.tobytes())
+    with open(lpath, "wb") as f:
+        f.write(struct.pack(">II", 2049, 6))
+        f.write(labels.tobytes())
+    from anand_mxnet_amd.io import MNISTIter
+    it = MNISTIter(ipath, lpath, batch_size=2, shuffle=False)
+    b = next(it)
+    assert b.data[0].shape == (2, 1, 28, 28)
+    assert b.label[0].asnumpy().tolist() == [0, 1]
