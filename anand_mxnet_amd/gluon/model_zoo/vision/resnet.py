@@ -34,8 +34,7 @@ class BasicBlockV1(HybridBlock):
         ax = _bn_axis(layout)
         self.body = nn.HybridSequential(prefix="")
         self.body.add(_conv3x3(channels, stride, in_channels, layout))
-        self.body.add(nn.BatchNorm(axis=ax))
-        self.body.add(nn.Activation("relu"))
+        self.body.add(nn.BatchNorm(axis=ax, fuse_relu=True))
         self.body.add(_conv3x3(channels, 1, channels, layout))
         self.body.add(nn.BatchNorm(axis=ax))
         if downsample:
@@ -60,7 +59,10 @@ class BasicBlockV1(HybridBlock):
 
 
 class BottleneckV1(HybridBlock):
-    """v1: stride in first 1x1; v1.5 (stride_in_3x3): stride in the 3x3."""
+    """v1: stride in first 1x1; v1.5 (stride_in_3x3): stride in the 3x3.
+
+    BN+ReLU pairs use the fused-ReLU BatchNorm (CDNA4 kernel epilogue —
+    one less full-tensor round trip per pair)."""
 
     def __init__(self, channels, stride, downsample=False, in_channels=0,
                  layout="NCHW", stride_in_3x3=False, **kwargs):
@@ -70,11 +72,9 @@ class BottleneckV1(HybridBlock):
         self.body = nn.HybridSequential(prefix="")
         self.body.add(nn.Conv2D(channels // 4, kernel_size=1, strides=s1,
                                 use_bias=False, layout=layout))
-        self.body.add(nn.BatchNorm(axis=ax))
-        self.body.add(nn.Activation("relu"))
+        self.body.add(nn.BatchNorm(axis=ax, fuse_relu=True))
         self.body.add(_conv3x3(channels // 4, s3, channels // 4, layout))
-        self.body.add(nn.BatchNorm(axis=ax))
-        self.body.add(nn.Activation("relu"))
+        self.body.add(nn.BatchNorm(axis=ax, fuse_relu=True))
         self.body.add(nn.Conv2D(channels, kernel_size=1, strides=1,
                                 use_bias=False, layout=layout))
         self.body.add(nn.BatchNorm(axis=ax))
@@ -183,8 +183,7 @@ class ResNetV1(HybridBlock):
             else:
                 self.features.add(nn.Conv2D(channels[0], 7, 2, 3,
                                             use_bias=False, layout=layout))
-                self.features.add(nn.BatchNorm(axis=ax))
-                self.features.add(nn.Activation("relu"))
+                self.features.add(nn.BatchNorm(axis=ax, fuse_relu=True))
                 self.features.add(nn.MaxPool2D(3, 2, 1, layout=layout))
             for i, num_layer in enumerate(layers):
                 stride = 1 if i == 0 else 2

@@ -70,6 +70,11 @@ void embedding_bwd_bf16(const long*, const bf16*, float*, long, int, long,
                         hipStream_t);
 void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
 void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
+// gemm_fp8.hip
+void gemm_tn_fp8(const unsigned char*, const unsigned char*, bf16*,
+                 const float*, float, int, int, int, long, long, long, long,
+                 long, long, int, bool, const unsigned char*, hipStream_t);
+void cast_to_fp8(const bf16*, unsigned char*, long, float, hipStream_t);
 // optim.hip
 struct TensorDesc {
   void* w;
@@ -156,6 +161,45 @@ at::Tensor gemm_wg(const at::Tensor& A, const at::Tensor& B) {
   mxa::gemm_wg_bf16(bfp(A), bfp(B), out.data_ptr<float>(), m, (int)Mo,
                     (int)No, Mo, No, No, zp(A), cur_stream());
   return out;
+}
+
+// fp8 (e4m3) GEMM: A8 [.., M, K] u8, B8 [.., N, K] u8 -> bf16 C
+at::Tensor gemm_tn_fp8(const at::Tensor& A8, const at::Tensor& B8,
+                       c10::optional<at::Tensor> bias, double inv_scale,
+                       bool relu) {
+  TORCH_CHECK(A8.scalar_type() == at::kByte && B8.scalar_type() == at::kByte);
+  TORCH_CHECK(A8.is_contiguous() && B8.is_contiguous());
+  bool batched = A8.dim() == 3;
+  long M = A8.size(-2), K = A8.size(-1), N = B8.size(-2);
+  TORCH_CHECK(B8.size(-1) == K && K % 128 == 0,
+              "gemm_tn_fp8: K must be %128");
+  int batches = batched ? A8.size(0) : 1;
+  auto out = batched
+                 ? at::empty({batches, M, N}, A8.options().dtype(at::kBFloat16))
+                 : at::empty({M, N}, A8.options().dtype(at::kBFloat16));
+  const float* bptr = nullptr;
+  at::Tensor bias_f;
+  if (bias.has_value() && bias->defined()) {
+    bias_f = bias->to(at::kFloat).contiguous();
+    bptr = bias_f.data_ptr<float>();
+  }
+  mxa::gemm_tn_fp8(A8.data_ptr<unsigned char>(),
+                   B8.data_ptr<unsigned char>(), bfp_mut(out), bptr,
+                   (float)inv_scale, (int)M, (int)N, (int)K, K, K, N,
+                   batched ? M * K : 0,
+                   (batched && B8.dim() == 3) ? N * K : 0,
+                   batched ? M * N : 0, batches, relu,
+                   (const unsigned char*)zp(zpage_for(out)), cur_stream());
+  return out;
+}
+
+at::Tensor cast_to_fp8(const at::Tensor& x, double scale) {
+  long n = x.numel();
+  TORCH_CHECK(n % 8 == 0);
+  auto y = at::empty_like(x, x.options().dtype(at::kByte));
+  mxa::cast_to_fp8(bfp(x), y.data_ptr<unsigned char>(), n, (float)scale,
+                   cur_stream());
+  return y;
 }
 
 // ---------------------------------------------------------------------------
@@ -540,6 +584,8 @@ long engine_new_stream(long device) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn", &gemm_tn, "C = A @ B^T (bf16, MFMA)");
   m.def("gemm_wg", &gemm_wg, "C(f32) = sum_m A[m,:] outer B[m,:]");
+  m.def("gemm_tn_fp8", &gemm_tn_fp8);
+  m.def("cast_to_fp8", &cast_to_fp8);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);

@@ -369,8 +369,11 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
                   hipStream_t s) {
   HIP_CHECK(hipMemsetAsync(psum, 0, C * sizeof(float), s));
   HIP_CHECK(hipMemsetAsync(psumsq, 0, C * sizeof(float), s));
-  int gy = (int)min((M + 8191) / 8192, (long)64);
-  dim3 g1(cdiv_h(C, 64), gy < 1 ? 1 : gy);
+  // fill the chip: ~768 blocks total regardless of C (narrow-C layers
+  // otherwise leave most CUs idle in the reduce)
+  int cb = cdiv_h(C, 64);
+  long gy = min((M + 255) / 256, (long)(768 / cb + 1));
+  dim3 g1(cb, (unsigned)(gy < 1 ? 1 : gy));
  hipLaunchKernelGGL(( bn_stats_partial), dim3(g1), dim3(256), 0, s, x, psum, psumsq, M, C);
  hipLaunchKernelGGL(( bn_finalize), dim3(cdiv_h(C, 256)), dim3(256), 0, s, 
       psum, psumsq, save_mean, save_rstd, running_mean, running_var, M, C,
@@ -402,8 +405,9 @@ void bn_bwd(const bf16* x, const bf16* dy, bf16* dx, const float* save_mean,
             hipStream_t s) {
   HIP_CHECK(hipMemsetAsync(dgamma, 0, C * sizeof(float), s));
   HIP_CHECK(hipMemsetAsync(dbeta, 0, C * sizeof(float), s));
-  int gy = (int)min((M + 8191) / 8192, (long)64);
-  dim3 g1(cdiv_h(C, 64), gy < 1 ? 1 : gy);
+  int cb = cdiv_h(C, 64);
+  long gy = min((M + 255) / 256, (long)(768 / cb + 1));
+  dim3 g1(cb, (unsigned)(gy < 1 ? 1 : gy));
   dim3 g2 = ew_grid(M * (C / 8));
   if (relu) {
    hipLaunchKernelGGL(( bn_bwd_reduce<true>), dim3(g1), dim3(256), 0, s, x, dy, save_mean, save_rstd,

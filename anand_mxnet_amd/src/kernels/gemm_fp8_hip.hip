@@ -1,0 +1,194 @@
+#include "hip/hip_runtime.h"
+// FP8 (OCP e4m3) TN-form MFMA GEMM for gfx950 + scaled cast kernels.
+// Config #5 (BASELINE.json): Transformer-big fp8 mixed precision.
+//
+//   C[M][N] = (1/(sa*sb)) * sum_k A8[M][k] * B8[N][k]
+//
+// A8/B8 are e4m3 with per-tensor scales sa/sb (amax-based, applied at cast
+// time); accumulation fp32; output bf16.  Same 128x128 tile / double-
+// buffered LDS / T14 register-staging structure as gemm_tn.hip, but
+// BK=128 fp8 per 128-byte row (half the staged bytes per MAC of bf16) and
+// 4 MFMA K-substeps per tile via v_mfma_f32_16x16x32_fp8_fp8.
+// gfx950 fp8 is OCP e4m3fn (NOT MI300X fnuz).
+#include "common.h"
+#include <hip/hip_fp8.h>
+
+namespace mxa {
+
+constexpr int F8_BM = 128, F8_BN = 128, F8_BK = 128, F8_THREADS = 256;
+
+typedef unsigned char u8;
+
+DEVINL int f8_lds_off(int row, int chunk16) {
+  // byte offset in a [128][128] u8 tile (128-B rows), T2 XOR swizzle
+  return row * F8_BK + ((chunk16 ^ (row & 7)) << 4);
+}
+
+struct F8PlainLoader {
+  const u8* base;
+  const u8* zpage;
+  int rows;
+  long ld;
+  long stride_z;
+  int ksteps;
+  // device state
+  const u8* rp[4];
+  long koff;
+
+  DEVINL void init(int row0, int t, int z) {
+    const u8* b = base + (long)z * stride_z;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int r = row0 + p * 32 + (t >> 3);
+      rp[p] = (r < rows) ? b + (long)r * ld : nullptr;
+    }
+  }
+  DEVINL void set_kstep(int ks) { koff = (long)ks * F8_BK; }
+  DEVINL const u8* src(int p, int chunk16) const {
+    return rp[p] ? rp[p] + koff + chunk16 * 16 : zpage + chunk16 * 16;
+  }
+};
+
+__global__ __launch_bounds__(F8_THREADS, 2) void gemm_tn_fp8_core(
+    F8PlainLoader aload, F8PlainLoader bload, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, float inv_scale, int M, int N, long ldc,
+    long strideC, int ksteps, int relu) {
+  __shared__ __align__(16) u8 smem[2][2][F8_BM * F8_BK];
+
+  const int ntn = (N + F8_BN - 1) / F8_BN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * F8_BM, bn = (bid % ntn) * F8_BN;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+
+  aload.init(bm, t, z);
+  bload.init(bn, t, z);
+
+  i32x4 areg[4], breg[4];
+  const int srow = t >> 3, schunk = t & 7;
+
+  auto stage_load = [&](int ks) {
+    aload.set_kstep(ks);
+    bload.set_kstep(ks);
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      areg[p] = *(const i32x4*)aload.src(p, schunk);
+      breg[p] = *(const i32x4*)bload.src(p, schunk);
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int row = p * 32 + srow;
+      *(i32x4*)&smem[buf][0][f8_lds_off(row, schunk)] = areg[p];
+      *(i32x4*)&smem[buf][1][f8_lds_off(row, schunk)] = breg[p];
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int w = t / WAVE, l = t % WAVE;
+  const int wm = (w >> 1) * 64, wn = (w & 1) * 64;
+  const int lrow = l & 15, lk = l >> 4;
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {  // 4 x K=32 fp8 substeps
+      long af[4], bfr[4];
+      int kbyte = kk * 32 + lk * 8;
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        int row = wm + mf * 16 + lrow;
+        int chunk = kbyte >> 4, within = kbyte & 15;
+        af[mf] = *(const long*)&smem[buf][0][row * F8_BK +
+                                             (((chunk ^ (row & 7)) << 4) |
+                                              within)];
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int row = wn + nf * 16 + lrow;
+        int chunk = kbyte >> 4, within = kbyte & 15;
+        bfr[nf] = *(const long*)&smem[buf][1][row * F8_BK +
+                                              (((chunk ^ (row & 7)) << 4) |
+                                               within)];
+      }
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+    }
+  };
+
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  if (ksteps > 1) stage_load(1);
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    compute(buf);
+    if (ks + 1 < ksteps) {
+      stage_write(buf ^ 1);
+      if (ks + 2 < ksteps) stage_load(ks + 2);
+    }
+    __syncthreads();
+  }
+
+  bf16* cz = Cout + (long)z * strideC;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wn + nf * 16 + (l & 15);
+      if (gcol >= N) continue;
+      float bv = bias ? bias[gcol] : 0.f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (l >> 4) * 4 + i;
+        if (grow >= M) continue;
+        float v = acc[mf][nf][i] * inv_scale + bv;
+        if (relu) v = v > 0.f ? v : 0.f;
+        cz[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
+// scaled cast bf16 -> e4m3 (vectorized 8-wide)
+__global__ __launch_bounds__(256) void cast_bf16_to_fp8(
+    const bf16* __restrict__ x, u8* __restrict__ y, long nvec, float scale) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 v = *(const bf16x8*)&x[i * 8];
+    u8 o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = __hip_fp8_e4m3(bf2f(v[j]) * scale).__x;
+    *(u64*)&y[i * 8] = *(const u64*)o;
+  }
+}
+
+void gemm_tn_fp8(const u8* A, const u8* B, bf16* C, const float* bias,
+                 float inv_scale, int M, int N, int K, long lda, long ldb,
+                 long ldc, long strideA, long strideB, long strideC,
+                 int batches, bool relu, const u8* zpage,
+                 hipStream_t stream) {
+  F8PlainLoader a{A, zpage, M, lda, strideA, K / F8_BK};
+  F8PlainLoader b{B, zpage, N, ldb, strideB, K / F8_BK};
+  dim3 grid(cdiv_h(M, F8_BM) * cdiv_h(N, F8_BN), 1, batches);
+ hipLaunchKernelGGL(( gemm_tn_fp8_core), dim3(grid), dim3(F8_THREADS), 0, stream, 
+      a, b, C, bias, inv_scale, M, N, ldc, strideC, K / F8_BK,
+      relu ? 1 : 0);
+}
+
+void cast_to_fp8(const bf16* x, u8* y, long n, float scale,
+                 hipStream_t stream) {
+ hipLaunchKernelGGL(( cast_bf16_to_fp8), dim3(ew_grid(n / 8)), dim3(256), 0, stream, x, y, n / 8, scale);
+}
+
+}  // namespace mxa

@@ -266,7 +266,7 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_core(
 static int pick_split(long blocks_xy, long ksteps) {
   // enough blocks to fill 256 CUs a few times over, bounded by the reduce
   int target = 1;
-  while (blocks_xy * target < 640 && target < ksteps && target < 64)
+  while (blocks_xy * target < 1024 && target < ksteps && target < 128)
     target *= 2;
   return target;
 }

@@ -303,8 +303,9 @@ void layernorm_bwd_bf16(const bf16* x, const bf16* dy, bf16* dx,
       x, dy, dx, gamma, save_mean, save_rstd, rows, D);
   HIP_CHECK(hipMemsetAsync(dgamma, 0, D * sizeof(float), s));
   HIP_CHECK(hipMemsetAsync(dbeta, 0, D * sizeof(float), s));
-  int gy = (int)min((rows + 8191) / 8192, (long)64);
-  dim3 g(cdiv_h(D, 64), gy < 1 ? 1 : gy);
+  int cb = cdiv_h(D, 64);
+  long gy = min((rows + 255) / 256, (long)(768 / cb + 1));
+  dim3 g(cb, (unsigned)(gy < 1 ? 1 : gy));
   layernorm_bwd_dparams<<<g, 256, 0, s>>>(x, dy, save_mean, save_rstd,
                                           dgamma, dbeta, rows, D);
 }
