@@ -11,6 +11,24 @@ from .. import nn
 from ..block import HybridBlock
 from .bert import MultiHeadAttentionCell, PositionwiseFFN
 from ... import ndarray as nd
+from .. import nn as _gnn
+
+
+def _set_fp8(use_fp8):
+    """Route the Dense layers built inside bert/transformer cells through
+    Fp8Dense while constructing an fp8 model."""
+    import anand_mxnet_amd.gluon.model_zoo.bert as _bert
+    _bert.nn = _FP8NS if use_fp8 else _gnn
+
+
+class _Fp8Namespace:
+    def __getattr__(self, name):
+        if name == "Dense":
+            return _gnn.Fp8Dense
+        return getattr(_gnn, name)
+
+
+_FP8NS = _Fp8Namespace()
 
 __all__ = ["TransformerEncoder", "TransformerDecoder", "TransformerModel",
            "transformer_big", "transformer_base"]
@@ -121,11 +139,18 @@ class TransformerModel(HybridBlock):
     hybrid_forward = None
 
 
-def transformer_big(src_vocab=32768, tgt_vocab=32768, **kwargs):
-    """WMT transformer-big: 6+6 layers, d=1024, 16 heads, ffn 4096."""
-    return TransformerModel(src_vocab, tgt_vocab, units=1024, num_layers=6,
-                            hidden_size=4096, num_heads=16, dropout=0.1,
-                            **kwargs)
+def transformer_big(src_vocab=32768, tgt_vocab=32768, use_fp8=False,
+                    **kwargs):
+    """WMT transformer-big: 6+6 layers, d=1024, 16 heads, ffn 4096.
+    use_fp8=True builds the attention/FFN projections as Fp8Dense
+    (CDNA4 fp8 e4m3 MFMA forward — config #5)."""
+    _set_fp8(use_fp8)
+    try:
+        return TransformerModel(src_vocab, tgt_vocab, units=1024,
+                                num_layers=6, hidden_size=4096,
+                                num_heads=16, dropout=0.1, **kwargs)
+    finally:
+        _set_fp8(False)
 
 
 def transformer_base(src_vocab=32768, tgt_vocab=32768, **kwargs):

@@ -6,7 +6,7 @@ import numpy as np
 from ..block import Block, HybridBlock
 from ...base import MXNetError
 
-__all__ = ["Sequential", "HybridSequential", "Dense", "Dropout", "BatchNorm",
+__all__ = ["Sequential", "HybridSequential", "Dense", "Fp8Dense", "Dropout", "BatchNorm",
            "LayerNorm", "GroupNorm", "InstanceNorm", "Embedding", "Flatten",
            "Lambda", "HybridLambda", "Activation", "LeakyReLU", "PReLU",
            "ELU", "SELU", "Swish", "GELU"]
@@ -104,6 +104,18 @@ class Dense(HybridBlock):
     def hybrid_forward(self, F, x, weight, bias=None):
         out = F.FullyConnected(x, weight, bias, num_hidden=self._units,
                                no_bias=bias is None, flatten=self._flatten)
+        if self.act is not None:
+            out = self.act(out)
+        return out
+
+
+class Fp8Dense(Dense):
+    """Dense whose forward GEMM runs in fp8 e4m3 on the CDNA4 fp8 MFMA
+    path (per-tensor amax scaling; grads bf16).  Config #5 building block."""
+
+    def hybrid_forward(self, F, x, weight, bias=None):
+        out = F.fp8_linear(x, weight, bias, num_hidden=self._units,
+                           no_bias=bias is None, flatten=self._flatten)
         if self.act is not None:
             out = self.act(out)
         return out

@@ -567,3 +567,58 @@ def mha(q, k, v, num_heads, causal, scale):
     out = _GemmTN.apply(p.contiguous(), vt)      # [BH, Tq, D]
     return out.reshape(B, num_heads, Tq, D).transpose(1, 2) \
         .reshape(B, Tq, HD)
+
+
+# ---------------------------------------------------------------------------
+# fp8 (OCP e4m3) linear path — config #5 (Transformer-big fp8).
+# Forward runs the fp8 MFMA GEMM with per-tensor amax scaling; backward
+# runs bf16 (standard fp8 training recipe: e4m3 fwd, higher-precision grads).
+# ---------------------------------------------------------------------------
+
+E4M3_MAX = 448.0
+
+
+def _amax_scale(t):
+    amax = t.detach().abs().amax().clamp(min=1e-12)
+    return (E4M3_MAX / amax).item()
+
+
+class _Fp8Linear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        sx = _amax_scale(x)
+        sw = _amax_scale(w)
+        x8 = _C.cast_to_fp8(x.contiguous(), sx)
+        w8 = _C.cast_to_fp8(w.contiguous(), sw)
+        return _C.gemm_tn_fp8(x8, w8,
+                              bias.float() if bias is not None else None,
+                              1.0 / (sx * sw), False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            if w.shape[0] % 64 == 0:
+                dx = _C.gemm_tn(dy, w.t().contiguous(), None, False)
+            else:
+                dx = torch.matmul(dy, w)
+        if ctx.needs_input_grad[1]:
+            dw = _C.gemm_wg(dy, x).to(w.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy.float().sum(0)
+        return dx, dw, db
+
+
+@register_gpu("fp8_linear")
+def fp8_linear(x, w, bias):
+    if x.size(-1) % 128 != 0:
+        # K not tileable for the fp8 kernel: bf16 MFMA path
+        return linear(x, w, bias)
+    x2 = x.reshape(-1, x.shape[-1])
+    out = _Fp8Linear.apply(x2.contiguous(), w,
+                           bias if bias is None else bias.to(x.dtype))
+    return out.reshape(x.shape[:-1] + (w.shape[0],))

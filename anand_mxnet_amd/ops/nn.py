@@ -67,6 +67,23 @@ def fully_connected(data, weight, bias=None, num_hidden=None, no_bias=False,
     return out
 
 
+@register("fp8_linear")
+def fp8_linear_op(data, weight, bias=None, num_hidden=None, no_bias=False,
+                  flatten=True):
+    """FullyConnected with fp8 (e4m3) forward (config #5).  CPU oracle:
+    torch float8_e4m3fn round-trip emulation of the same quantization."""
+    x = data.reshape(data.shape[0], -1) if flatten and data.dim() > 2 else data
+    b = None if no_bias else bias
+    if _use_hip(x, "fp8_linear"):
+        return gpu_impl("fp8_linear")(x, weight, b)
+    E4M3_MAX = 448.0
+    sx = (E4M3_MAX / x.detach().abs().amax().clamp(min=1e-12))
+    sw = (E4M3_MAX / weight.detach().abs().amax().clamp(min=1e-12))
+    xq = ((x * sx).to(torch.float8_e4m3fn).to(torch.float32)) / sx
+    wq = ((weight * sw).to(torch.float8_e4m3fn).to(torch.float32)) / sw
+    return F.linear(xq.to(x.dtype), wq.to(weight.dtype), b)
+
+
 # ---------------- Convolution ----------------
 
 @register("Convolution")

@@ -298,3 +298,30 @@ def test_multi_sumsq():
     for i, w in enumerate(ws):
         ref = w.float().pow(2).sum()
         assert_close(out[i], ref, rtol=1e-2, atol=1e-2, msg=f"sumsq {i}")
+
+
+def test_gemm_fp8():
+    M, N, K = 256, 192, 256
+    a = _rt((M, K))
+    b = _rt((N, K), 0.3)
+    E4 = 448.0
+    sa = (E4 / a.abs().max()).item()
+    sb = (E4 / b.abs().max()).item()
+    a8 = _C.cast_to_fp8(a.to(BF).contiguous(), sa)
+    b8 = _C.cast_to_fp8(b.to(BF).contiguous(), sb)
+    out = _C.gemm_tn_fp8(a8, b8, None, 1.0 / (sa * sb), False)
+    ref = a @ b.t()
+    # fp8 e4m3 has ~2 mantissa bits: loose elementwise, tight correlation
+    import numpy as np
+    o, r = out.float().cpu().numpy(), ref.cpu().numpy()
+    corr = np.corrcoef(o.ravel(), r.ravel())[0, 1]
+    assert corr > 0.995, corr
+    rel = np.abs(o - r).max() / (np.abs(r).max() + 1e-6)
+    assert rel < 0.15, rel
+
+
+def test_cast_fp8_roundtrip():
+    x = torch.linspace(-400, 400, 1024, device=_dev()).to(BF)
+    y8 = _C.cast_to_fp8(x.contiguous(), 1.0)
+    ref = x.float().to(torch.float8_e4m3fn).view(torch.uint8)
+    assert (y8 == ref).float().mean().item() > 0.99

@@ -27,6 +27,8 @@ def main():
                     help="per-GPU sequences")
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--dtype", default="bfloat16")
+    ap.add_argument("--fp8", action="store_true",
+                    help="transformer_big only: fp8 e4m3 MFMA projections")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -82,7 +84,7 @@ def main():
             transformer_big
         vocab = 32768
         net = transformer_big(src_vocab=vocab, tgt_vocab=vocab,
-                              max_length=max(T, 1024))
+                              max_length=max(T, 1024), use_fp8=args.fp8)
         net.cast(args.dtype)
         net.initialize(mx.init.Normal(0.02), ctx=ctx)
         trainer = gluon.Trainer(net.collect_params(), "adam",
@@ -102,7 +104,7 @@ def main():
             trainer.step(B * world_size)
             return l
         tokens_per_step = B * T * world_size
-        cfg_model = "transformer_big"
+        cfg_model = "transformer_big_fp8" if args.fp8 else "transformer_big"
 
     for _ in range(args.warmup):
         loss = step()
