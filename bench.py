@@ -26,6 +26,8 @@ def main():
                     help="per-GPU batch size")
     ap.add_argument("--model", default="resnet50_v1_5")
     ap.add_argument("--image-size", type=int, default=224)
+    ap.add_argument("--no-capture", action="store_true",
+                    help="disable hipGraph step capture (eager launches)")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -75,13 +77,35 @@ def main():
     for _ in range(args.warmup):
         loss = step()
     torch.cuda.synchronize()
+
+    # hipGraph step capture (the CachedOp/static_alloc fast path: one graph
+    # replay per step instead of per-op launches).  Graph capture composes
+    # with RCCL collectives; fall back to eager if capture fails.
+    graph = None
+    if not args.no_capture:
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                loss = step()
+            g.replay()
+            torch.cuda.synchronize()
+            graph = g
+        except Exception as e:
+            if rank == 0:
+                print(f"# graph capture unavailable ({e}); eager",
+                      file=sys.stderr)
+            graph = None
+    run_step = graph.replay if graph is not None else step
+    for _ in range(2):
+        run_step()
+    torch.cuda.synchronize()
     if distributed:
         dist.barrier()
     torch.cuda.synchronize()
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        loss = step()
+        run_step()
     torch.cuda.synchronize()
     if distributed:
         dist.barrier()
