@@ -237,7 +237,21 @@ class HybridBlock(Block):
         super().__init__(prefix=prefix, params=params)
         self._active = False
         self._cached_graph = None
+        self._cached_graph_obj = None
         self._graph_kwargs = {}
+
+    def _eager_forward(self, *args):
+        return super(HybridBlock, type(self)).forward(self, *args) \
+            if type(self).forward is HybridBlock.forward \
+            else type(self).forward(self, *args)
+
+    def __call__(self, *args, **kwargs):
+        if self._active and not kwargs:
+            from .cached_op import maybe_cached_forward
+            out = maybe_cached_forward(self, args)
+            if out is not None:
+                return out
+        return super().__call__(*args, **kwargs)
 
     def hybridize(self, active=True, static_alloc=False, static_shape=False,
                   **kwargs):
@@ -245,6 +259,7 @@ class HybridBlock(Block):
         self._graph_kwargs = dict(static_alloc=static_alloc,
                                   static_shape=static_shape, **kwargs)
         self._cached_graph = None
+        self._cached_graph_obj = None
         super().hybridize(active, static_alloc=static_alloc,
                           static_shape=static_shape, **kwargs)
 

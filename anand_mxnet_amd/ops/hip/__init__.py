@@ -52,7 +52,7 @@ class _GemmTN(torch.autograd.Function):
                 da = torch.matmul(dy, b)
         if ctx.needs_input_grad[1]:
             if a.dim() == 2:
-                db = _C.gemm_wg(dy, a).to(a.dtype)  # [N,K]
+                db = _wg_or_matmul(dy, a, a.dtype)  # [N,K]
             elif M % 64 == 0:
                 db = _C.gemm_tn(dy.transpose(-1, -2).contiguous(),
                                 a.transpose(-1, -2).contiguous(), None,
@@ -60,6 +60,21 @@ class _GemmTN(torch.autograd.Function):
             else:
                 db = torch.matmul(dy.transpose(-1, -2), a)
         return da, db
+
+
+
+def _tn_or_matmul(a, b):
+    """TN MFMA kernel when the reduce dim tiles, else library GEMM."""
+    if a.size(-1) % 64 == 0:
+        return _GemmTN.apply(a.contiguous(), b.contiguous())
+    return torch.matmul(a, b.transpose(-1, -2))
+
+
+def _wg_or_matmul(dy, x, out_dtype):
+    """WG MFMA kernel when output dims are %8, else library GEMM."""
+    if dy.size(-1) % 8 == 0 and x.size(-1) % 8 == 0:
+        return _C.gemm_wg(dy, x).to(out_dtype)
+    return torch.matmul(dy.t(), x).to(out_dtype)
 
 
 @register_gpu("gemm")
@@ -102,7 +117,7 @@ class _Linear(torch.autograd.Function):
             else:  # e.g. the 1000-class head: library GEMM for dgrad
                 dx = torch.matmul(dy, w)
         if ctx.needs_input_grad[1]:
-            dw = _C.gemm_wg(dy, x).to(w.dtype)
+            dw = _wg_or_matmul(dy, x, w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(0)
         return dx, dw, db
@@ -557,14 +572,14 @@ def mha(q, k, v, num_heads, causal, scale):
         .reshape(B * num_heads, Tk, D).contiguous()
     vh = v.reshape(B, Tk, num_heads, D).transpose(1, 2) \
         .reshape(B * num_heads, Tk, D).contiguous()
-    s = _GemmTN.apply(qh, kh) * scale            # [BH, Tq, Tk]
+    s = _tn_or_matmul(qh, kh) * scale            # [BH, Tq, Tk]
     if causal:
         p = _MaskedSoftmax.apply(s, _causal_vlen(B * num_heads, Tq,
                                                  s.device))
     else:
         p = _Softmax.apply(s, False)
     vt = vh.transpose(-1, -2).contiguous()       # [BH, D, Tk]
-    out = _GemmTN.apply(p.contiguous(), vt)      # [BH, Tq, D]
+    out = _tn_or_matmul(p.contiguous(), vt)      # [BH, Tq, D]
     return out.reshape(B, num_heads, Tq, D).transpose(1, 2) \
         .reshape(B, Tq, HD)
 
@@ -607,7 +622,7 @@ class _Fp8Linear(torch.autograd.Function):
             else:
                 dx = torch.matmul(dy, w)
         if ctx.needs_input_grad[1]:
-            dw = _C.gemm_wg(dy, x).to(w.dtype)
+            dw = _wg_or_matmul(dy, x, w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(0)
         return dx, dw, db

@@ -121,7 +121,7 @@ def test_resnet18_grad_matches_cpu():
             loss = loss_fn(net2(x), y).mean()
         loss.backward()
         return {name: p.grad(ctx).asnumpy().astype(np.float64)
-                for name, p in net2.collect_params().items()
+                for name, p in net2._collect_params_with_prefix().items()
                 if p.grad_req != "null"}
 
     g_cpu = run(mx.cpu(), "float32")
