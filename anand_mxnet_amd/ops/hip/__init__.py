@@ -594,8 +594,9 @@ E4M3_MAX = 448.0
 
 
 def _amax_scale(t):
-    amax = t.detach().abs().amax().clamp(min=1e-12)
-    return (E4M3_MAX / amax).item()
+    """Device-resident scale (no host sync — the kernels read it)."""
+    amax = t.detach().abs().amax().clamp(min=1e-12).float()
+    return E4M3_MAX / amax
 
 
 class _Fp8Linear(torch.autograd.Function):
@@ -609,7 +610,7 @@ class _Fp8Linear(torch.autograd.Function):
         w8 = _C.cast_to_fp8(w.contiguous(), sw)
         return _C.gemm_tn_fp8(x8, w8,
                               bias.float() if bias is not None else None,
-                              1.0 / (sx * sw), False)
+                              sx, sw, False)
 
     @staticmethod
     def backward(ctx, dy):

@@ -72,9 +72,11 @@ void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
 void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
 // gemm_fp8.hip
 void gemm_tn_fp8(const unsigned char*, const unsigned char*, bf16*,
-                 const float*, float, int, int, int, long, long, long, long,
-                 long, long, int, bool, const unsigned char*, hipStream_t);
-void cast_to_fp8(const bf16*, unsigned char*, long, float, hipStream_t);
+                 const float*, const float*, const float*, int, int, int,
+                 long, long, long, long, long, long, int, bool,
+                 const unsigned char*, hipStream_t);
+void cast_to_fp8(const bf16*, unsigned char*, long, const float*,
+                 hipStream_t);
 // optim.hip
 struct TensorDesc {
   void* w;
@@ -165,8 +167,9 @@ at::Tensor gemm_wg(const at::Tensor& A, const at::Tensor& B) {
 
 // fp8 (e4m3) GEMM: A8 [.., M, K] u8, B8 [.., N, K] u8 -> bf16 C
 at::Tensor gemm_tn_fp8(const at::Tensor& A8, const at::Tensor& B8,
-                       c10::optional<at::Tensor> bias, double inv_scale,
-                       bool relu) {
+                       c10::optional<at::Tensor> bias, const at::Tensor& sa,
+                       const at::Tensor& sb, bool relu) {
+  TORCH_CHECK(sa.scalar_type() == at::kFloat && sb.scalar_type() == at::kFloat);
   TORCH_CHECK(A8.scalar_type() == at::kByte && B8.scalar_type() == at::kByte);
   TORCH_CHECK(A8.is_contiguous() && B8.is_contiguous());
   bool batched = A8.dim() == 3;
@@ -185,7 +188,8 @@ at::Tensor gemm_tn_fp8(const at::Tensor& A8, const at::Tensor& B8,
   }
   mxa::gemm_tn_fp8(A8.data_ptr<unsigned char>(),
                    B8.data_ptr<unsigned char>(), bfp_mut(out), bptr,
-                   (float)inv_scale, (int)M, (int)N, (int)K, K, K, N,
+                   sa.data_ptr<float>(), sb.data_ptr<float>(),
+                   (int)M, (int)N, (int)K, K, K, N,
                    batched ? M * K : 0,
                    (batched && B8.dim() == 3) ? N * K : 0,
                    batched ? M * N : 0, batches, relu,
@@ -193,12 +197,13 @@ at::Tensor gemm_tn_fp8(const at::Tensor& A8, const at::Tensor& B8,
   return out;
 }
 
-at::Tensor cast_to_fp8(const at::Tensor& x, double scale) {
+at::Tensor cast_to_fp8(const at::Tensor& x, const at::Tensor& scale) {
   long n = x.numel();
   TORCH_CHECK(n % 8 == 0);
+  TORCH_CHECK(scale.scalar_type() == at::kFloat);
   auto y = at::empty_like(x, x.options().dtype(at::kByte));
-  mxa::cast_to_fp8(bfp(x), y.data_ptr<unsigned char>(), n, (float)scale,
-                   cur_stream());
+  mxa::cast_to_fp8(bfp(x), y.data_ptr<unsigned char>(), n,
+                   scale.data_ptr<float>(), cur_stream());
   return y;
 }
 
