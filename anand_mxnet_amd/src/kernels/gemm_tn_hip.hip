@@ -34,6 +34,12 @@ DEVINL int lds_off(int row, int chunk16) {
   return row * BK + ((chunk16 ^ (row & 7)) << 3);
 }
 
+DEVINL int stage_row(int p, int t) {
+  // row a thread's lane-linear glds slot covers: wave w, pass p -> LDS
+  // bytes (w*4+p)*1024 + lane*16, i.e. row = (w*4+p)*8 + lane/8.
+  return (((t >> 6) * 4 + p) << 3) + ((t & 63) >> 3);
+}
+
 // ---------------------------------------------------------------------------
 // Loaders: produce the per-lane 16-byte source pointer for tile staging.
 // Each thread stages 4 rows (passes); row = pass*32 + t/8, chunk16 = t%8.
@@ -54,7 +60,7 @@ struct PlainLoader {
     const bf16* b = base + (long)z * stride_z;
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      int r = row0 + p * 32 + (t >> 3);
+      int r = row0 + stage_row(p, t);
       rp[p] = (r < rows) ? b + (long)r * ld : nullptr;
     }
   }
@@ -82,7 +88,7 @@ struct ConvALoader {
     int PQ = P * Q;
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      int m = row0 + p * 32 + (t >> 3);
+      int m = row0 + stage_row(p, t);
       int ok = m < Nb * PQ;
       int mm = ok ? m : 0;
       int n = mm / PQ, pq = mm % PQ;
@@ -120,7 +126,7 @@ struct ConvWLoader {
   DEVINL void init(int row0, int t, int) {
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      int n = row0 + p * 32 + (t >> 3);
+      int n = row0 + stage_row(p, t);
       nrow[p] = (n < Kout) ? n : -1;
     }
   }
@@ -154,7 +160,7 @@ struct DgradALoader {
     int HW = H * W;
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      int m = row0 + p * 32 + (t >> 3);
+      int m = row0 + stage_row(p, t);
       int ok = m < Nb * HW;
       int mm = ok ? m : 0;
       int n = mm / HW, hw = mm % HW;
@@ -194,7 +200,7 @@ struct DgradWLoader {
   DEVINL void init(int row0, int t, int) {
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      int n = row0 + p * 32 + (t >> 3);
+      int n = row0 + stage_row(p, t);
       nrow[p] = (n < C) ? n : -1;
     }
   }
@@ -230,24 +236,29 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
   aload.init(bm, t, z);
   bload.init(bn, t, z);
 
-  i32x4 areg[4], breg[4];
-  const int srow = t >> 3, schunk = t & 7;  // staging row-slot base / chunk
-
-  auto stage_load = [&](int ks) {
+  const int lw = t / WAVE, ll = t % WAVE;
+  // glds staging (guide §5 step 3): LDS dest is lane-linear (wave base +
+  // lane*16); the T2 swizzle moves to the per-lane SOURCE chunk (rule 21),
+  // so the LDS image matches lds_off()'s read XOR.
+  auto stage_glds = [&](int buf, int ks) {
     aload.set_kstep(ks);
     bload.set_kstep(ks);
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      areg[p] = *(const i32x4*)aload.src(p, schunk);
-      breg[p] = *(const i32x4*)bload.src(p, schunk);
-    }
-  };
-  auto stage_write = [&](int buf) {
-#pragma unroll
-    for (int p = 0; p < 4; ++p) {
-      int row = p * 32 + srow;
-      *(i32x4*)&smem[buf][0][lds_off(row, schunk)] = areg[p];
-      *(i32x4*)&smem[buf][1][lds_off(row, schunk)] = breg[p];
+      int row = stage_row(p, t);
+      int cg = (ll & 7) ^ (row & 7);  // source chunk for this lane's slot
+      unsigned lbase = __builtin_amdgcn_readfirstlane(
+          (unsigned)((lw * 4 + p) * 1024));
+      auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[buf][0][0] + lbase);
+      auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[buf][1][0] + lbase);
+      auto gA = (const __attribute__((address_space(1))) unsigned int*)
+          aload.src(p, cg);
+      auto gB = (const __attribute__((address_space(1))) unsigned int*)
+          bload.src(p, cg);
+      __builtin_amdgcn_global_load_lds(gA, ldsA, 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(gB, ldsB, 16, 0, 0);
     }
   };
 
@@ -257,7 +268,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int w = t / WAVE, l = t % WAVE;
+  const int w = lw, l = ll;
   const int wm = (w >> 1) * 64, wn = (w & 1) * 64;
   const int lrow = l & 15, lk = l >> 4;  // fragment row / k-group
 
@@ -285,19 +296,15 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
     }
   };
 
-  // T14 schedule: one barrier per K-step, loads for step t+2 issued right
-  // after the step-(t+1) tile is written to LDS.
-  stage_load(0);
-  stage_write(0);
+  // 2-phase glds schedule (guide T3 minimum form): issue next-tile DMA
+  // before computing the current tile; __syncthreads() drains the DMA
+  // (vmcnt(0) is folded into the barrier while a glds is in flight).
+  stage_glds(0, 0);
   __syncthreads();
-  if (ksteps > 1) stage_load(1);
   for (int ks = 0; ks < ksteps; ++ks) {
     int buf = ks & 1;
+    if (ks + 1 < ksteps) stage_glds(buf ^ 1, ks + 1);
     compute(buf);
-    if (ks + 1 < ksteps) {
-      stage_write(buf ^ 1);
-      if (ks + 2 < ksteps) stage_load(ks + 2);
-    }
     __syncthreads();
   }
 
