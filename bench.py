@@ -26,8 +26,9 @@ def main():
                     help="per-GPU batch size")
     ap.add_argument("--model", default="resnet50_v1_5")
     ap.add_argument("--image-size", type=int, default=224)
-    ap.add_argument("--no-capture", action="store_true",
-                    help="disable hipGraph step capture (eager launches)")
+    ap.add_argument("--capture", action="store_true",
+                    help="hipGraph whole-step capture (experimental: "
+                         "crashes under rocprofv3 on this stack)")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -82,7 +83,7 @@ def main():
     # replay per step instead of per-op launches).  Graph capture composes
     # with RCCL collectives; fall back to eager if capture fails.
     graph = None
-    if not args.no_capture:
+    if args.capture:
         try:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
