@@ -36,6 +36,10 @@ from . import callback
 from . import profiler
 from . import symbol
 from . import symbol as sym
+from . import rtc
+from . import contrib
+from . import monitor
+from . import visualization
 from . import test_utils
 from . import util
 
