@@ -1,0 +1,33 @@
+#!/usr/bin/env python3
+"""Single-kernel probe for PMC counter runs (keeps dispatch count small)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from anand_mxnet_amd import _C
+
+BF = torch.bfloat16
+DEV = torch.device("cuda", 0)
+
+which = sys.argv[1] if len(sys.argv) > 1 else "wg"
+iters = int(sys.argv[2]) if len(sys.argv) > 2 else 3
+
+if which == "wg":
+    m, Mo, No = 100352, 128, 128
+    a = (torch.rand((m, Mo), device=DEV) * 2 - 1).to(BF)
+    b = (torch.rand((m, No), device=DEV) * 2 - 1).to(BF)
+    for _ in range(iters):
+        _C.gemm_wg(a, b)
+elif which == "wg_conv":
+    x = (torch.rand((64, 28, 28, 128), device=DEV) * 2 - 1).to(BF)
+    dy = (torch.rand((64, 28, 28, 128), device=DEV) * 2 - 1).to(BF)
+    for _ in range(iters):
+        _C.conv2d_wgrad(x, dy, [3, 3], [1, 1], [1, 1], [1, 1])
+elif which == "tn":
+    a = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
+    b = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
+    for _ in range(iters):
+        _C.gemm_tn(a, b, None, False)
+torch.cuda.synchronize()
+print("done", which)
