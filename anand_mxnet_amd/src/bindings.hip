@@ -37,11 +37,12 @@ void bn_fwd_infer(const bf16*, bf16*, const float*, const float*,
                   const float*, const float*, long, int, float, bool,
                   hipStream_t);
 void bn_bwd(const bf16*, const bf16*, bf16*, const float*, const float*,
-            const float*, const float*, float*, float*, long, int, bool,
-            hipStream_t);
+            const float*, const float*, float*, float*, float*, long, int,
+            bool, hipStream_t);
+int bn_gy(long, int);
 void maxpool_nhwc(const bf16*, bf16*, unsigned char*, int, int, int, int,
                   int, int, int, int, int, int, int, int, hipStream_t);
-void maxpool_nhwc_bwd(const bf16*, const unsigned char*, float*, int, int,
+void maxpool_nhwc_bwd(const bf16*, const unsigned char*, bf16*, int, int,
                       int, int, int, int, int, int, int, int, int, int,
                       hipStream_t);
 void avgpool_nhwc(const bf16*, bf16*, int, int, int, int, int, int, int,
@@ -284,7 +285,8 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
   long M = x.numel() / C;
   auto y = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
-  auto psum = at::empty({C}, opts), psumsq = at::empty({C}, opts);
+  int gy = mxa::bn_gy(M, C);
+  auto psum = at::empty({gy, C}, opts), psumsq = at::empty({gy, C}, opts);
   auto save_mean = at::empty({C}, opts), save_rstd = at::empty({C}, opts);
   mxa::bn_fwd_train(bfp(x), bfp_mut(y), psum.data_ptr<float>(),
                     psumsq.data_ptr<float>(), save_mean.data_ptr<float>(),
@@ -319,10 +321,13 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   auto dx = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
   auto dgamma = at::empty({C}, opts), dbeta = at::empty({C}, opts);
+  int gy = mxa::bn_gy(M, C);
+  auto slab = at::empty({2L * gy, C}, opts);
   mxa::bn_bwd(bfp(x), bfp(dy), bfp_mut(dx), save_mean.data_ptr<float>(),
               save_rstd.data_ptr<float>(), gamma.data_ptr<float>(),
               beta.data_ptr<float>(), dgamma.data_ptr<float>(),
-              dbeta.data_ptr<float>(), M, C, relu, cur_stream());
+              dbeta.data_ptr<float>(), slab.data_ptr<float>(), M, C, relu,
+              cur_stream());
   return {dx, dgamma, dbeta};
 }
 
@@ -347,13 +352,13 @@ at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& arg,
                        std::vector<long> stride, std::vector<long> pad) {
   int Nb = xshape[0], H = xshape[1], W = xshape[2], C = xshape[3];
   int P = dy.size(1), Q = dy.size(2);
-  auto dxf = at::empty({Nb, H, W, C}, dy.options().dtype(at::kFloat));
+  auto dx = at::empty({Nb, H, W, C}, dy.options());
   mxa::maxpool_nhwc_bwd(bfp(dy), arg.data_ptr<unsigned char>(),
-                        dxf.data_ptr<float>(), Nb, H, W, C, P, Q,
+                        bfp_mut(dx), Nb, H, W, C, P, Q,
                         (int)kernel[0], (int)kernel[1], (int)stride[0],
                         (int)stride[1], (int)pad[0], (int)pad[1],
                         cur_stream());
-  return dxf.to(at::kBFloat16);
+  return dx;
 }
 
 at::Tensor avgpool(const at::Tensor& x, std::vector<long> kernel,

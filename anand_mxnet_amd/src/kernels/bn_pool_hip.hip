@@ -43,7 +43,7 @@ __global__ __launch_bounds__(256) void bn_stats_partial(
     red[1][rlane][chunk * 8 + i] = ss[i];
   }
   __syncthreads();
-  // 64 threads finalize 64 channels
+  // 64 threads write this block's partials to its slab row (no atomics)
   if (t < 64) {
     float fs = 0, fss = 0;
 #pragma unroll
@@ -53,8 +53,8 @@ __global__ __launch_bounds__(256) void bn_stats_partial(
     }
     int c = cblk + t;
     if (c < C) {
-      atomicAdd(&psum[c], fs);
-      atomicAdd(&psumsq[c], fss);
+      psum[(long)blockIdx.y * C + c] = fs;
+      psumsq[(long)blockIdx.y * C + c] = fss;
     }
   }
 }
@@ -65,11 +65,17 @@ __global__ void bn_finalize(const float* __restrict__ psum,
                             float* __restrict__ save_rstd,
                             float* __restrict__ running_mean,
                             float* __restrict__ running_var, long M, int C,
-                            float eps, float momentum, int update_running) {
+                            int gy, float eps, float momentum,
+                            int update_running) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = psum[c] / M;
-  float var = psumsq[c] / M - mean * mean;
+  float s = 0, ss = 0;
+  for (int g = 0; g < gy; ++g) {
+    s += psum[(long)g * C + c];
+    ss += psumsq[(long)g * C + c];
+  }
+  float mean = s / M;
+  float var = ss / M - mean * mean;
   if (var < 0) var = 0;
   save_mean[c] = mean;
   save_rstd[c] = rsqrtf(var + eps);
@@ -174,10 +180,26 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce(
     }
     int c = cblk + t;
     if (c < C) {
-      atomicAdd(&dbeta[c], fb);
-      atomicAdd(&dgamma[c], fg);
+      dbeta[(long)blockIdx.y * C + c] = fb;
+      dgamma[(long)blockIdx.y * C + c] = fg;
     }
   }
+}
+
+// reduce the per-block slabs into the final dgamma/dbeta vectors
+__global__ void bn_bwd_finalize(const float* __restrict__ dbeta_slab,
+                                const float* __restrict__ dgamma_slab,
+                                float* __restrict__ dbeta,
+                                float* __restrict__ dgamma, int C, int gy) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float fb = 0, fg = 0;
+  for (int g = 0; g < gy; ++g) {
+    fb += dbeta_slab[(long)g * C + c];
+    fg += dgamma_slab[(long)g * C + c];
+  }
+  dbeta[c] = fb;
+  dgamma[c] = fg;
 }
 
 template <bool RELU>
@@ -262,23 +284,37 @@ __global__ __launch_bounds__(256) void maxpool_fwd(
   }
 }
 
+// gather form: each INPUT position sums the windows whose argmax points
+// at it (no atomics, no zero-fill, bf16 out directly)
 __global__ __launch_bounds__(256) void maxpool_bwd(
     const bf16* __restrict__ dy, const unsigned char* __restrict__ arg,
-    float* __restrict__ dx, int Nb, int H, int W, int C, int P, int Q,
+    bf16* __restrict__ dx, int Nb, int H, int W, int C, int P, int Q,
     int R, int S, int sh, int sw, int ph, int pw) {
-  long total = (long)Nb * P * Q * C;
+  long total = (long)Nb * H * W * C;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     int c = (int)(i % C);
-    long npq = i / C;
-    int q = (int)(npq % Q);
-    int p = (int)((npq / Q) % P);
-    int n = (int)(npq / ((long)P * Q));
-    int a = arg[i];
-    int r = a / S, s = a % S;
-    int ih = p * sh - ph + r, iw = q * sw - pw + s;
-    if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
-    atomicAdd(&dx[(((long)n * H + ih) * W + iw) * C + c], bf2f(dy[i]));
+    long nhw = i / C;
+    int w_ = (int)(nhw % W);
+    int h = (int)((nhw / W) % H);
+    int n = (int)(nhw / ((long)H * W));
+    float acc = 0;
+    // output windows covering (h, w): p in [ceil((h+ph-R+1)/sh), (h+ph)/sh]
+    int p_lo = (h + ph - R + sh) >= 0 ? (h + ph - R + sh) / sh : 0;
+    int p_hi = min((h + ph) / sh, P - 1);
+    int q_lo = (w_ + pw - S + sw) >= 0 ? (w_ + pw - S + sw) / sw : 0;
+    int q_hi = min((w_ + pw) / sw, Q - 1);
+    for (int p = p_lo; p <= p_hi; ++p) {
+      int r = h + ph - p * sh;
+      if (r < 0 || r >= R) continue;
+      for (int q = q_lo; q <= q_hi; ++q) {
+        int sx = w_ + pw - q * sw;
+        if (sx < 0 || sx >= S) continue;
+        long oi = (((long)n * P + p) * Q + q) * C + c;
+        if (arg[oi] == r * S + sx) acc += bf2f(dy[oi]);
+      }
+    }
+    dx[i] = f2bf(acc);
   }
 }
 
@@ -362,22 +398,26 @@ __global__ __launch_bounds__(256) void gap_bwd(const bf16* __restrict__ dy,
 // host launchers
 // ---------------------------------------------------------------------------
 
+int bn_gy(long M, int C) {
+  // fill the chip: ~768 blocks total regardless of C (narrow-C layers
+  // otherwise leave most CUs idle in the reduce)
+  int cb = cdiv_h(C, 64);
+  long gy = min((M + 255) / 256, (long)(768 / cb + 1));
+  return (int)(gy < 1 ? 1 : gy);
+}
+
 void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
                   float* save_mean, float* save_rstd, float* running_mean,
                   float* running_var, const float* gamma, const float* beta,
                   long M, int C, float eps, float momentum, bool relu,
                   hipStream_t s) {
-  HIP_CHECK(hipMemsetAsync(psum, 0, C * sizeof(float), s));
-  HIP_CHECK(hipMemsetAsync(psumsq, 0, C * sizeof(float), s));
-  // fill the chip: ~768 blocks total regardless of C (narrow-C layers
-  // otherwise leave most CUs idle in the reduce)
   int cb = cdiv_h(C, 64);
-  long gy = min((M + 255) / 256, (long)(768 / cb + 1));
-  dim3 g1(cb, (unsigned)(gy < 1 ? 1 : gy));
+  int gy = bn_gy(M, C);
+  dim3 g1(cb, gy);
  hipLaunchKernelGGL(( bn_stats_partial), dim3(g1), dim3(256), 0, s, x, psum, psumsq, M, C);
  hipLaunchKernelGGL(( bn_finalize), dim3(cdiv_h(C, 256)), dim3(256), 0, s, 
       psum, psumsq, save_mean, save_rstd, running_mean, running_var, M, C,
-      eps, momentum, 1);
+      gy, eps, momentum, 1);
   dim3 g2 = ew_grid(M * (C / 8));
   if (relu)
    hipLaunchKernelGGL(( bn_apply<true>), dim3(g2), dim3(256), 0, s, x, y, save_mean, save_rstd, gamma,
@@ -401,25 +441,30 @@ void bn_fwd_infer(const bf16* x, bf16* y, const float* rmean,
 
 void bn_bwd(const bf16* x, const bf16* dy, bf16* dx, const float* save_mean,
             const float* save_rstd, const float* gamma, const float* beta,
-            float* dgamma, float* dbeta, long M, int C, bool relu,
-            hipStream_t s) {
-  HIP_CHECK(hipMemsetAsync(dgamma, 0, C * sizeof(float), s));
-  HIP_CHECK(hipMemsetAsync(dbeta, 0, C * sizeof(float), s));
+            float* dgamma, float* dbeta, float* slab, long M, int C,
+            bool relu, hipStream_t s) {
   int cb = cdiv_h(C, 64);
-  long gy = min((M + 255) / 256, (long)(768 / cb + 1));
-  dim3 g1(cb, (unsigned)(gy < 1 ? 1 : gy));
+  int gy = bn_gy(M, C);
+  dim3 g1(cb, gy);
   dim3 g2 = ew_grid(M * (C / 8));
-  if (relu) {
+  float* dbeta_slab = slab;
+  float* dgamma_slab = slab + (long)gy * C;
+  if (relu)
    hipLaunchKernelGGL(( bn_bwd_reduce<true>), dim3(g1), dim3(256), 0, s, x, dy, save_mean, save_rstd,
-                                           gamma, beta, dbeta, dgamma, M, C);
+                                           gamma, beta, dbeta_slab,
+                                           dgamma_slab, M, C);
+  else
+   hipLaunchKernelGGL(( bn_bwd_reduce<false>), dim3(g1), dim3(256), 0, s, x, dy, save_mean, save_rstd,
+                                            gamma, beta, dbeta_slab,
+                                            dgamma_slab, M, C);
+ hipLaunchKernelGGL(( bn_bwd_finalize), dim3(cdiv_h(C, 256)), dim3(256), 0, s, dbeta_slab, dgamma_slab,
+                                                 dbeta, dgamma, C, gy);
+  if (relu)
    hipLaunchKernelGGL(( bn_bwd_apply<true>), dim3(g2), dim3(256), 0, s, x, dy, dx, save_mean, save_rstd,
                                           gamma, beta, dbeta, dgamma, M, C);
-  } else {
-   hipLaunchKernelGGL(( bn_bwd_reduce<false>), dim3(g1), dim3(256), 0, s, x, dy, save_mean, save_rstd,
-                                            gamma, beta, dbeta, dgamma, M, C);
+  else
    hipLaunchKernelGGL(( bn_bwd_apply<false>), dim3(g2), dim3(256), 0, s, x, dy, dx, save_mean, save_rstd,
                                            gamma, beta, dbeta, dgamma, M, C);
-  }
 }
 
 void maxpool_nhwc(const bf16* x, bf16* y, unsigned char* arg, int Nb, int H,
@@ -430,11 +475,10 @@ void maxpool_nhwc(const bf16* x, bf16* y, unsigned char* arg, int Nb, int H,
                                 ph, pw);
 }
 
-void maxpool_nhwc_bwd(const bf16* dy, const unsigned char* arg, float* dx,
+void maxpool_nhwc_bwd(const bf16* dy, const unsigned char* arg, bf16* dx,
                       int Nb, int H, int W, int C, int P, int Q, int R,
                       int S, int sh, int sw, int ph, int pw, hipStream_t s) {
-  HIP_CHECK(hipMemsetAsync(dx, 0, (long)Nb * H * W * C * sizeof(float), s));
-  dim3 g = ew_grid((long)Nb * P * Q * C);
+  dim3 g = ew_grid((long)Nb * H * W * C);
  hipLaunchKernelGGL(( maxpool_bwd), dim3(g), dim3(256), 0, s, dy, arg, dx, Nb, H, W, C, P, Q, R, S, sh,
                                 sw, ph, pw);
 }
