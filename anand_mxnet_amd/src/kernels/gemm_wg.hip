@@ -61,14 +61,14 @@ struct WgIm2colRows {
   int sh, sw, ph, pw;
   // device state
   int roff, soff;
-  int n_[8], p_[8], q_[8];
+  int n_[4], p_[4], q_[4];
   int col0;
 
   DEVINL void init_rows(int c0, long m0, const int* rows) {
     col0 = c0;
     int PQ = P * Q;
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < 4; ++i) {
       long m = m0 + rows[i];
       n_[i] = (int)(m / PQ);
       int pq = (int)(m % PQ);
@@ -78,7 +78,7 @@ struct WgIm2colRows {
   }
   DEVINL void step() {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < 4; ++i) {
       q_[i] += WBK;
       while (q_[i] >= Q) {
         q_[i] -= Q;
@@ -120,70 +120,67 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_core(
   int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int bm = (bid / ntn) * WBM, bn = (bid % ntn) * WBN;
   const int t = threadIdx.x;
-  const bool isB = t >= 128;
-  const int tt = t & 127;
-  const int pair = tt & 7, colchunk = tt >> 3;  // 0..15
+  // Balanced staging: EVERY thread stages half of A and half of B (a
+  // wave that staged only the contiguous operand would park at the
+  // barrier while the gather waves straggle — measured 41% SQ_WAIT_ANY).
+  // Per thread: 2 row-pairs per operand; pair = t&15, colchunk = t>>4.
+  const int pair = t & 15, colchunk = t >> 4;  // colchunk 0..15
 
-  int rows_local[8];
+  int rows_local[4];
 #pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    rows_local[2 * p] = p * 16 + 2 * pair;
-    rows_local[2 * p + 1] = p * 16 + 2 * pair + 1;
+  for (int p = 0; p < 2; ++p) {
+    rows_local[2 * p] = (p * 16 + pair) * 2;
+    rows_local[2 * p + 1] = (p * 16 + pair) * 2 + 1;
   }
 
   long m0 = (long)kstep0 * WBK;
-  if (!isB) {
-    if (A_IM2COL) {
-      aim.roff = (seg / S) * dh;
-      aim.soff = (seg % S) * dw;
-      aim.init_rows(bm, m0, rows_local);
-    } else {
-      aplain.init(bm, m0);
-    }
+  if (A_IM2COL) {
+    aim.roff = (seg / S) * dh;
+    aim.soff = (seg % S) * dw;
+    aim.init_rows(bm, m0, rows_local);
   } else {
-    bload.init(bn, m0);
+    aplain.init(bm, m0);
   }
+  bload.init(bn, m0);
 
-  i32x4 sreg[8];
+  i32x4 sregA[4], sregB[4];
 
   auto stage_load = [&]() {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      if (!isB) {
-        if (A_IM2COL)
-          sreg[i] = *(const i32x4*)aim.src(i, colchunk);
-        else
-          sreg[i] = *(const i32x4*)aplain.src(rows_local[i], colchunk);
-      } else {
-        sreg[i] = *(const i32x4*)bload.src(rows_local[i], colchunk);
-      }
-    }
-    if (!isB) {
+    for (int i = 0; i < 4; ++i) {
       if (A_IM2COL)
-        aim.step();
+        sregA[i] = *(const i32x4*)aim.src(i, colchunk);
       else
-        aplain.step();
-    } else {
-      bload.step();
+        sregA[i] = *(const i32x4*)aplain.src(rows_local[i], colchunk);
+      sregB[i] = *(const i32x4*)bload.src(rows_local[i], colchunk);
     }
+    if (A_IM2COL)
+      aim.step();
+    else
+      aplain.step();
+    bload.step();
   };
 
   auto stage_write = [&](int buf) {
-    bf16* img = smem[buf][isB ? 1 : 0];
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
-      const u32* lo = (const u32*)&sreg[2 * p];      // row r0, 8 bf16
-      const u32* hi = (const u32*)&sreg[2 * p + 1];  // row r0+1
-      int r0 = rows_local[2 * p];
+    for (int op = 0; op < 2; ++op) {
+      bf16* img = smem[buf][op];
+      const i32x4* sreg = op ? sregB : sregA;
 #pragma unroll
-      for (int jw = 0; jw < 4; ++jw) {
-        u32 a01 = lo[jw];  // cols (2jw, 2jw+1) of row r0
-        u32 b01 = hi[jw];
-        u32 pair0 = (a01 & 0xffffu) | (b01 << 16);
-        u32 pair1 = (a01 >> 16) | (b01 & 0xffff0000u);
-        int c0 = colchunk * 8 + 2 * jw;
-        *(u32*)((char*)img + wg_byte(c0, r0 * 2)) = pair0;
-        *(u32*)((char*)img + wg_byte(c0 + 1, r0 * 2)) = pair1;
+      for (int p = 0; p < 2; ++p) {
+        const u32* lo = (const u32*)&sreg[2 * p];      // row r0, 8 bf16
+        const u32* hi = (const u32*)&sreg[2 * p + 1];  // row r0+1
+        int r0 = rows_local[2 * p];
+#pragma unroll
+        for (int jw = 0; jw < 4; ++jw) {
+          u32 a01 = lo[jw];  // cols (2jw, 2jw+1) of row r0
+          u32 b01 = hi[jw];
+          u32 pair0 = (a01 & 0xffffu) | (b01 << 16);
+          u32 pair1 = (a01 >> 16) | (b01 & 0xffff0000u);
+          int c0 = colchunk * 8 + 2 * jw;
+          *(u32*)((char*)img + wg_byte(c0, r0 * 2)) = pair0;
+          *(u32*)((char*)img + wg_byte(c0 + 1, r0 * 2)) = pair1;
+        }
       }
     }
   };
