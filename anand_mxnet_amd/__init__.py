@@ -39,6 +39,7 @@ from . import symbol as sym
 from . import rtc
 from . import contrib
 from . import monitor
+from . import image
 from . import visualization
 from . import test_utils
 from . import util
