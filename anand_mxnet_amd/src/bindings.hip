@@ -17,12 +17,12 @@ namespace mxa {
 void gemm_tn_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
                   int, long, long, long, long, long, long, int, bool,
                   const bf16*, hipStream_t);
-void conv2d_fwd_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
+void conv2d_fwd_bf16(const bf16*, const bf16*, bf16*, float*, const float*,
                      int, int, int, int, int, int, int, int, int, int, int,
-                     int, int, bool, const bf16*, hipStream_t);
-void conv2d_dgrad_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
+                     int, int, int, int, bool, const bf16*, hipStream_t);
+void conv2d_dgrad_bf16(const bf16*, const bf16*, bf16*, float*, int, int,
                        int, int, int, int, int, int, int, int, int, int,
-                       int, const bf16*, hipStream_t);
+                       int, int, int, const bf16*, hipStream_t);
 // gemm_wg.hip
 void gemm_wg_bf16(const bf16*, const bf16*, float*, long, int, int, long,
                   long, long, const bf16*, hipStream_t);
@@ -37,8 +37,8 @@ void bn_fwd_infer(const bf16*, bf16*, const float*, const float*,
                   const float*, const float*, long, int, float, bool,
                   hipStream_t);
 void bn_bwd(const bf16*, const bf16*, bf16*, const float*, const float*,
-            const float*, const float*, float*, float*, float*, long, int,
-            bool, hipStream_t);
+            const float*, const float*, float*, float*, long, int, bool,
+            hipStream_t);
 int bn_gy(long, int);
 void maxpool_nhwc(const bf16*, bf16*, unsigned char*, int, int, int, int,
                   int, int, int, int, int, int, int, int, hipStream_t);
@@ -232,8 +232,16 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w_rskc,
     bias_f = bias->to(at::kFloat).contiguous();
     bptr = bias_f.data_ptr<float>();
   }
-  mxa::conv2d_fwd_bf16(bfp(x), bfp(w_rskc), bfp_mut(y), bptr, Nb, H, W, C,
-                       Kout, R, S, P, Q, sh, sw, ph, pw, dh, dw, relu,
+  long M = (long)Nb * P * Q;
+  long blocks = ((M + 127) / 128) * ((Kout + 127) / 128);
+  at::Tensor accbuf;
+  float* accptr = nullptr;
+  if (blocks < 384 && R * S * (C / 64) > 1) {
+    accbuf = at::empty({M, Kout}, x.options().dtype(at::kFloat));
+    accptr = accbuf.data_ptr<float>();
+  }
+  mxa::conv2d_fwd_bf16(bfp(x), bfp(w_rskc), bfp_mut(y), accptr, bptr, Nb, H,
+                       W, C, Kout, R, S, P, Q, sh, sw, ph, pw, dh, dw, relu,
                        zp(x), cur_stream());
   return y;
 }
@@ -249,10 +257,18 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w_rsck,
   TORCH_CHECK(Kout % 64 == 0, "conv2d_dgrad: K must be %64");
   int P = dy.size(1), Q = dy.size(2);
   auto dx = at::empty({Nb, H, W, C}, dy.options());
-  mxa::conv2d_dgrad_bf16(bfp(dy), bfp(w_rsck), bfp_mut(dx), Nb, H, W, C,
-                         Kout, R, S, P, Q, (int)stride[0], (int)stride[1],
-                         (int)pad[0], (int)pad[1], (int)dil[0], (int)dil[1],
-                         zp(dy), cur_stream());
+  long M = (long)Nb * H * W;
+  long blocks = ((M + 127) / 128) * ((C + 127) / 128);
+  at::Tensor accbuf;
+  float* accptr = nullptr;
+  if (blocks < 384 && R * S * (Kout / 64) > 1) {
+    accbuf = at::empty({M, C}, dy.options().dtype(at::kFloat));
+    accptr = accbuf.data_ptr<float>();
+  }
+  mxa::conv2d_dgrad_bf16(bfp(dy), bfp(w_rsck), bfp_mut(dx), accptr, Nb, H,
+                         W, C, Kout, R, S, P, Q, (int)stride[0],
+                         (int)stride[1], (int)pad[0], (int)pad[1],
+                         (int)dil[0], (int)dil[1], zp(dy), cur_stream());
   return dx;
 }
 
@@ -285,8 +301,8 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
   long M = x.numel() / C;
   auto y = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
-  int gy = mxa::bn_gy(M, C);
-  auto psum = at::empty({gy, C}, opts), psumsq = at::empty({gy, C}, opts);
+  auto acc = at::empty({2, C}, opts);   // [psum; psumsq], one memset
+  auto psum = acc[0], psumsq = acc[1];
   auto save_mean = at::empty({C}, opts), save_rstd = at::empty({C}, opts);
   mxa::bn_fwd_train(bfp(x), bfp_mut(y), psum.data_ptr<float>(),
                     psumsq.data_ptr<float>(), save_mean.data_ptr<float>(),
@@ -320,14 +336,12 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   long M = x.numel() / C;
   auto dx = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
-  auto dgamma = at::empty({C}, opts), dbeta = at::empty({C}, opts);
-  int gy = mxa::bn_gy(M, C);
-  auto slab = at::empty({2L * gy, C}, opts);
+  auto dacc = at::empty({2, C}, opts);  // [dbeta; dgamma], one memset
+  auto dbeta = dacc[0], dgamma = dacc[1];
   mxa::bn_bwd(bfp(x), bfp(dy), bfp_mut(dx), save_mean.data_ptr<float>(),
               save_rstd.data_ptr<float>(), gamma.data_ptr<float>(),
               beta.data_ptr<float>(), dgamma.data_ptr<float>(),
-              dbeta.data_ptr<float>(), slab.data_ptr<float>(), M, C, relu,
-              cur_stream());
+              dbeta.data_ptr<float>(), M, C, relu, cur_stream());
   return {dx, dgamma, dbeta};
 }
 

@@ -42,7 +42,7 @@ __global__ __launch_bounds__(256) void bn_stats_partial(
     red[1][rlane][chunk * 8 + i] = ss[i];
   }
   __syncthreads();
-  // 64 threads write this block's partials to its slab row (no atomics)
+  // 64 threads accumulate 64 channels (one atomic per block per channel)
   if (t < 64) {
     float fs = 0, fss = 0;
 #pragma unroll
@@ -52,8 +52,8 @@ __global__ __launch_bounds__(256) void bn_stats_partial(
     }
     int c = cblk + t;
     if (c < C) {
-      psum[(long)blockIdx.y * C + c] = fs;
-      psumsq[(long)blockIdx.y * C + c] = fss;
+      atomicAdd(&psum[c], fs);
+      atomicAdd(&psumsq[c], fss);
     }
   }
 }
@@ -68,13 +68,8 @@ __global__ void bn_finalize(const float* __restrict__ psum,
                             int update_running) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float s = 0, ss = 0;
-  for (int g = 0; g < gy; ++g) {
-    s += psum[(long)g * C + c];
-    ss += psumsq[(long)g * C + c];
-  }
-  float mean = s / M;
-  float var = ss / M - mean * mean;
+  float mean = psum[c] / M;
+  float var = psumsq[c] / M - mean * mean;
   if (var < 0) var = 0;
   save_mean[c] = mean;
   save_rstd[c] = rsqrtf(var + eps);
@@ -179,26 +174,10 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce(
     }
     int c = cblk + t;
     if (c < C) {
-      dbeta[(long)blockIdx.y * C + c] = fb;
-      dgamma[(long)blockIdx.y * C + c] = fg;
+      atomicAdd(&dbeta[c], fb);
+      atomicAdd(&dgamma[c], fg);
     }
   }
-}
-
-// reduce the per-block slabs into the final dgamma/dbeta vectors
-__global__ void bn_bwd_finalize(const float* __restrict__ dbeta_slab,
-                                const float* __restrict__ dgamma_slab,
-                                float* __restrict__ dbeta,
-                                float* __restrict__ dgamma, int C, int gy) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float fb = 0, fg = 0;
-  for (int g = 0; g < gy; ++g) {
-    fb += dbeta_slab[(long)g * C + c];
-    fg += dgamma_slab[(long)g * C + c];
-  }
-  dbeta[c] = fb;
-  dgamma[c] = fg;
 }
 
 template <bool RELU>
@@ -410,6 +389,8 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
                   float* running_var, const float* gamma, const float* beta,
                   long M, int C, float eps, float momentum, bool relu,
                   hipStream_t s) {
+  // psum/psumsq are one contiguous [2,C] buffer: single fill
+  HIP_CHECK(hipMemsetAsync(psum, 0, 2L * C * sizeof(float), s));
   int cb = cdiv_h(C, 64);
   int gy = bn_gy(M, C);
   dim3 g1(cb, gy);
@@ -440,24 +421,22 @@ void bn_fwd_infer(const bf16* x, bf16* y, const float* rmean,
 
 void bn_bwd(const bf16* x, const bf16* dy, bf16* dx, const float* save_mean,
             const float* save_rstd, const float* gamma, const float* beta,
-            float* dgamma, float* dbeta, float* slab, long M, int C,
-            bool relu, hipStream_t s) {
+            float* dgamma, float* dbeta, long M, int C, bool relu,
+            hipStream_t s) {
+  // dgamma/dbeta are one contiguous [2,C] buffer: single fill
+  HIP_CHECK(hipMemsetAsync(dbeta < dgamma ? dbeta : dgamma, 0,
+                           2L * C * sizeof(float), s));
   int cb = cdiv_h(C, 64);
   int gy = bn_gy(M, C);
   dim3 g1(cb, gy);
   dim3 g2 = ew_grid(M * (C / 8));
-  float* dbeta_slab = slab;
-  float* dgamma_slab = slab + (long)gy * C;
   if (relu)
     bn_bwd_reduce<true><<<g1, 256, 0, s>>>(x, dy, save_mean, save_rstd,
-                                           gamma, beta, dbeta_slab,
-                                           dgamma_slab, M, C);
+                                           gamma, beta, dbeta, dgamma, M, C);
   else
     bn_bwd_reduce<false><<<g1, 256, 0, s>>>(x, dy, save_mean, save_rstd,
-                                            gamma, beta, dbeta_slab,
-                                            dgamma_slab, M, C);
-  bn_bwd_finalize<<<cdiv_h(C, 256), 256, 0, s>>>(dbeta_slab, dgamma_slab,
-                                                 dbeta, dgamma, C, gy);
+                                            gamma, beta, dbeta, dgamma, M,
+                                            C);
   if (relu)
     bn_bwd_apply<true><<<g2, 256, 0, s>>>(x, dy, dx, save_mean, save_rstd,
                                           gamma, beta, dbeta, dgamma, M, C);

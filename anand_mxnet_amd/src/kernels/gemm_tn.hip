@@ -219,11 +219,21 @@ struct DgradWLoader {
 // Core kernel
 // ---------------------------------------------------------------------------
 
-template <typename AL, typename BL, bool RELU, bool BIAS>
+template <typename AL, typename BL, bool RELU, bool BIAS,
+          bool ACCF32 = false>
 __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
     AL aload, BL bload, bf16* __restrict__ Cout,
     const float* __restrict__ bias, int M, int N, long ldc, long strideC,
     int ksteps) {
+  // split-K over blockIdx.y (small-M convs underfill the chip otherwise):
+  // each y-slice covers a contiguous k-step range, accumulating fp32.
+  int kstep0 = 0;
+  if (ACCF32) {
+    int per = (ksteps + gridDim.y - 1) / gridDim.y;
+    kstep0 = blockIdx.y * per;
+    ksteps = min(per, ksteps - kstep0);
+    if (ksteps <= 0) return;
+  }
   __shared__ __align__(16) bf16 smem[2][2][BM * BK];
 
   const int ntn = (N + BN - 1) / BN;
@@ -298,11 +308,11 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
   // 2-phase glds schedule (guide T3 minimum form): issue next-tile DMA
   // before computing the current tile; __syncthreads() drains the DMA
   // (vmcnt(0) is folded into the barrier while a glds is in flight).
-  stage_glds(0, 0);
+  stage_glds(0, kstep0);
   __syncthreads();
   for (int ks = 0; ks < ksteps; ++ks) {
     int buf = ks & 1;
-    if (ks + 1 < ksteps) stage_glds(buf ^ 1, ks + 1);
+    if (ks + 1 < ksteps) stage_glds(buf ^ 1, kstep0 + ks + 1);
     compute(buf);
     __syncthreads();
   }
@@ -320,11 +330,28 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
       for (int i = 0; i < 4; ++i) {
         int grow = bm + wm + mf * 16 + (l >> 4) * 4 + i;
         if (grow >= M) continue;
-        float v = acc[mf][nf][i] + bv;
-        if (RELU) v = v > 0.f ? v : 0.f;
-        cz[(long)grow * ldc + gcol] = f2bf(v);
+        if (ACCF32) {
+          atomicAdd((float*)Cout + (long)grow * ldc + gcol,
+                    acc[mf][nf][i]);
+        } else {
+          float v = acc[mf][nf][i] + bv;
+          if (RELU) v = v > 0.f ? v : 0.f;
+          cz[(long)grow * ldc + gcol] = f2bf(v);
+        }
       }
     }
+  }
+}
+
+// fp32 accum -> bf16 (+bias/relu) finish for the split-K path
+__global__ __launch_bounds__(256) void splitk_finish(
+    const float* __restrict__ accbuf, bf16* __restrict__ out,
+    const float* __restrict__ bias, long total, int N, int relu) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    float v = accbuf[i] + (bias ? bias[i % N] : 0.f);
+    if (relu) v = v > 0.f ? v : 0.f;
+    out[i] = f2bf(v);
   }
 }
 
@@ -364,27 +391,60 @@ void gemm_tn_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
 }
 
 void conv2d_fwd_bf16(const bf16* x, const bf16* w_rskc, bf16* y,
-                     const float* bias, int Nb, int H, int W, int C, int Kout,
-                     int R, int S, int P, int Q, int sh, int sw, int ph,
-                     int pw, int dh, int dw, bool relu, const bf16* zpage,
-                     hipStream_t stream) {
+                     float* accbuf, const float* bias, int Nb, int H, int W,
+                     int C, int Kout, int R, int S, int P, int Q, int sh,
+                     int sw, int ph, int pw, int dh, int dw, bool relu,
+                     const bf16* zpage, hipStream_t stream) {
   ConvALoader a{x, zpage, Nb, H, W, C, P, Q, S, sh, sw, ph, pw, dh, dw,
                 C / BK};
   ConvWLoader b{w_rskc, zpage, Kout, C, C / BK};
   int M = Nb * P * Q;
-  launch(a, b, y, bias, M, Kout, Kout, 0, R * S * (C / BK), relu, 1, stream);
+  int ksteps = R * S * (C / BK);
+  long blocks = (long)cdiv_h(M, BM) * cdiv_h(Kout, BN);
+  int split = accbuf ? (int)min((long)ksteps,
+                                max(1L, 512 / (blocks < 1 ? 1 : blocks)))
+                     : 1;
+  if (split > 1) {
+    long total = (long)M * Kout;
+    HIP_CHECK(hipMemsetAsync(accbuf, 0, total * sizeof(float), stream));
+    dim3 grid((unsigned)blocks, split, 1);
+    gemm_tn_core<ConvALoader, ConvWLoader, false, false, true>
+        <<<grid, THREADS, 0, stream>>>(a, b, (bf16*)accbuf, nullptr, M,
+                                       Kout, Kout, 0, ksteps);
+    splitk_finish<<<ew_grid(total / 1), 256, 0, stream>>>(
+        accbuf, y, bias, total, Kout, relu ? 1 : 0);
+    return;
+  }
+  launch(a, b, y, bias, M, Kout, Kout, 0, ksteps, relu, 1, stream);
 }
 
-void conv2d_dgrad_bf16(const bf16* dy, const bf16* w_rsck, bf16* dx, int Nb,
-                       int H, int W, int C, int Kout, int R, int S, int P,
-                       int Q, int sh, int sw, int ph, int pw, int dh, int dw,
-                       const bf16* zpage, hipStream_t stream) {
+void conv2d_dgrad_bf16(const bf16* dy, const bf16* w_rsck, bf16* dx,
+                       float* accbuf, int Nb, int H, int W, int C, int Kout,
+                       int R, int S, int P, int Q, int sh, int sw, int ph,
+                       int pw, int dh, int dw, const bf16* zpage,
+                       hipStream_t stream) {
   DgradALoader a{dy, zpage, Nb, H, W, P, Q, Kout, S, sh, sw, ph, pw, dh, dw,
                  Kout / BK};
   DgradWLoader b{w_rsck, zpage, C, Kout, Kout / BK};
   int M = Nb * H * W;
-  launch(a, b, dx, (const float*)nullptr, M, C, C, 0,
-         R * S * (Kout / BK), false, 1, stream);
+  int ksteps = R * S * (Kout / BK);
+  long blocks = (long)cdiv_h(M, BM) * cdiv_h(C, BN);
+  int split = accbuf ? (int)min((long)ksteps,
+                                max(1L, 512 / (blocks < 1 ? 1 : blocks)))
+                     : 1;
+  if (split > 1) {
+    long total = (long)M * C;
+    HIP_CHECK(hipMemsetAsync(accbuf, 0, total * sizeof(float), stream));
+    dim3 grid((unsigned)blocks, split, 1);
+    gemm_tn_core<DgradALoader, DgradWLoader, false, false, true>
+        <<<grid, THREADS, 0, stream>>>(a, b, (bf16*)accbuf, nullptr, M, C,
+                                       C, 0, ksteps);
+    splitk_finish<<<ew_grid(total), 256, 0, stream>>>(
+        accbuf, dx, nullptr, total, C, 0);
+    return;
+  }
+  launch(a, b, dx, (const float*)nullptr, M, C, C, 0, ksteps, false, 1,
+         stream);
 }
 
 }  // namespace mxa
