@@ -17,7 +17,8 @@ from . import engine
 from . import autograd
 from . import ndarray
 from . import ndarray as nd
-from .ndarray import random
+from . import random
+from . import random as random_mod
 from . import initializer
 from . import initializer as init
 from . import optimizer

@@ -1,2 +1,3 @@
 from .base_module import BaseModule
 from .module import Module
+from .bucketing_module import BucketingModule
