@@ -236,7 +236,7 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w_rskc,
   long blocks = ((M + 127) / 128) * ((Kout + 127) / 128);
   at::Tensor accbuf;
   float* accptr = nullptr;
-  if (blocks < 384 && R * S * (C / 64) > 1) {
+  if (blocks < 150 && R * S * (C / 64) > 1) {
     accbuf = at::empty({M, Kout}, x.options().dtype(at::kFloat));
     accptr = accbuf.data_ptr<float>();
   }
@@ -261,7 +261,7 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w_rsck,
   long blocks = ((M + 127) / 128) * ((C + 127) / 128);
   at::Tensor accbuf;
   float* accptr = nullptr;
-  if (blocks < 384 && R * S * (Kout / 64) > 1) {
+  if (blocks < 150 && R * S * (Kout / 64) > 1) {
     accbuf = at::empty({M, C}, dy.options().dtype(at::kFloat));
     accptr = accbuf.data_ptr<float>();
   }
