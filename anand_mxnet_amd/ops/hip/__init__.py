@@ -562,7 +562,7 @@ def _causal_vlen(BH, T, device):
 
 
 @register_gpu("mha")
-def mha(q, k, v, num_heads, causal, scale):
+def mha(q, k, v, num_heads, causal, scale, dropout_p=0.0, training=False):
     B, Tq, HD = q.shape
     Tk = k.shape[1]
     D = HD // num_heads
@@ -578,6 +578,8 @@ def mha(q, k, v, num_heads, causal, scale):
                                                  s.device))
     else:
         p = _Softmax.apply(s, False)
+    if dropout_p > 0.0 and training:
+        p = _Dropout.apply(p, dropout_p)
     vt = vh.transpose(-1, -2).contiguous()       # [BH, D, Tk]
     out = _tn_or_matmul(p.contiguous(), vt)      # [BH, Tq, D]
     return out.reshape(B, num_heads, Tq, D).transpose(1, 2) \

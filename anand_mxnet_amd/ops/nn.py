@@ -371,9 +371,9 @@ def multi_head_attention(q, k, v, num_heads, mask=None, dropout_p=0.0,
     B, Tq, HD = q.shape
     D = HD // num_heads
     scale = softmax_scale if softmax_scale is not None else D ** -0.5
-    if _use_hip(q, "mha") and q.dtype == torch.bfloat16 and mask is None \
-            and dropout_p == 0.0:
-        return gpu_impl("mha")(q, k, v, num_heads, causal, scale)
+    if _use_hip(q, "mha") and q.dtype == torch.bfloat16 and mask is None:
+        return gpu_impl("mha")(q, k, v, num_heads, causal, scale,
+                               dropout_p, _ag.is_training())
     qh = q.reshape(B, Tq, num_heads, D).transpose(1, 2)
     kh = k.reshape(B, -1, num_heads, D).transpose(1, 2)
     vh = v.reshape(B, -1, num_heads, D).transpose(1, 2)

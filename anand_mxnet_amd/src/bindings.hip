@@ -23,6 +23,10 @@ void conv2d_fwd_bf16(const bf16*, const bf16*, bf16*, float*, const float*,
 void conv2d_dgrad_bf16(const bf16*, const bf16*, bf16*, float*, int, int,
                        int, int, int, int, int, int, int, int, int, int,
                        int, int, int, const bf16*, hipStream_t);
+// gemm_tn8.hip
+void gemm_tn8_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
+                   int, long, long, long, long, long, long, int, bool,
+                   const bf16*, hipStream_t);
 // gemm_wg.hip
 void gemm_wg_bf16(const bf16*, const bf16*, float*, long, int, int, long,
                   long, long, const bf16*, hipStream_t);
@@ -148,9 +152,16 @@ at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
   long sA = batched ? M * K : 0;
   long sB = (batched && B.dim() == 3) ? N * K : 0;
   long sC = batched ? M * N : 0;
-  mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
-                    (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
-                    cur_stream());
+  if (M >= 256 && N >= 256) {
+    // deep-pipelined 256² 8-phase kernel (NLP GEMM shapes)
+    mxa::gemm_tn8_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
+                       (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
+                       cur_stream());
+  } else {
+    mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
+                      (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
+                      cur_stream());
+  }
   return out;
 }
 

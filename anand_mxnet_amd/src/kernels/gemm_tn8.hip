@@ -1,0 +1,215 @@
+// 256x256-tile 8-phase MFMA GEMM for gfx950 — the deep-pipelined dense
+// path (guide §5 "256² 8-phase template"): 512 threads (8 waves as 2Mx4N),
+// BK=64, half-tile glds ring (2 tile-buffers x 2 halves per operand,
+// 128 KiB LDS), COUNTED s_waitcnt vmcnt across raw barriers (never 0 in
+// the main loop), quadrant-phased compute with s_setprio around each MFMA
+// cluster, T2 source-side XOR swizzle, T1 bijective XCD remap.
+//
+// Phase schedule (tile T, quadrants q = (A-half, B-half)):
+//   p0 (A0,B0): issue T+1.A0 ; vmcnt(6) ; barrier ; 16 MFMA
+//   p1 (A1,B0): issue T+1.B0 ; vmcnt(6) ; barrier ; 16 MFMA
+//   p2 (A1,B1): issue T+1.A1 ; vmcnt(6) ; barrier ; 16 MFMA
+//   p3 (A0,B1): issue T+1.B1 ;  (no wait) ; barrier ; 16 MFMA
+// so three halves stay in flight at every wait (vmcnt counts 2 glds per
+// half).  Slot of tile T's half is reused by tile T+2, whose issue happens
+// after T's last-reader barrier.
+//
+// Used for the dense NLP GEMMs (Linear/FFN/attention, K % 128 == 0);
+// convolution keeps the 128² kernel in gemm_tn.hip.
+#include "common.h"
+
+namespace mxa {
+
+constexpr int T8_BM = 256, T8_BN = 256, T8_BK = 64, T8_THREADS = 512;
+
+DEVINL int t8_lds_byte(int row, int chunk) {
+  // byte offset within a [128][64] bf16 half-image, T2 row-XOR swizzle
+  return row * 128 + ((chunk ^ (row & 7)) << 4);
+}
+
+struct Plain8Loader {
+  const bf16* base;
+  const bf16* zpage;
+  int rows;
+  long ld;
+  long stride_z;
+  // device state: 4 row-slots (half a = p>>1, instr j = p&1)
+  const bf16* rp[4];
+  int cg[4];
+
+  DEVINL void init(int row0, int t, int z) {
+    const bf16* b = base + (long)z * stride_z;
+    int w = t / WAVE, l = t % WAVE;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int local = (p >> 1) * 128 + w * 16 + (p & 1) * 8 + (l >> 3);
+      int r = row0 + local;
+      rp[p] = (r < rows) ? b + (long)r * ld : nullptr;
+      cg[p] = (l & 7) ^ (local & 7);  // source chunk for lane's slot
+    }
+  }
+  DEVINL const bf16* src(int p, long koff) const {
+    return rp[p] ? rp[p] + koff + cg[p] * 8 : zpage + cg[p] * 8;
+  }
+};
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(T8_THREADS, 2) void gemm_tn8_core(
+    Plain8Loader aload, Plain8Loader bload, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, long strideC,
+    int ksteps, long kstride) {
+  // [tilebuf][op][half][128x64 bf16]
+  __shared__ __align__(16) bf16 smem[2][2][2][128 * 64];
+
+  const int ntn = (N + T8_BN - 1) / T8_BN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * T8_BM, bn = (bid % ntn) * T8_BN;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+  const int w = t / WAVE, l = t % WAVE;
+
+  aload.init(bm, t, z);
+  bload.init(bn, t, z);
+
+  // one glds pair: half `hf` (0=A0,1=B0,2=A1,3=B1 order by issue index) of
+  // tile `tile` into its ring slot.
+  auto issue_half = [&](int issue_idx, int tile) {
+    int op = issue_idx & 1;          // A0,B0,A1,B1
+    int half = issue_idx >> 1;
+    long koff = (long)tile * kstride;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;  // loader slot
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[tile & 1][op][half]
+              [0] + seg);
+      const bf16* gp = op ? bload.src(p, koff) : aload.src(p, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+
+  f32x4 acc[4][8];  // [quadrant][mf*2+nf]
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[q][i] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int lrow = l & 15, lk = l >> 4;
+  const int wmq = (w >> 2) * 64;  // within-quadrant M offset of this wave
+  const int wnq = (w & 3) * 32;   // within-quadrant N offset
+
+  // one phase: quadrant (qa, qb) of tile `tile`
+  auto phase = [&](int tile, int qa, int qb, int q, int wait6) {
+    if (wait6 == 1)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if (wait6 == 0)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const bf16* As = smem[tile & 1][0][qa];
+    const bf16* Bs = smem[tile & 1][1][qb];
+    bf16x8 af[4][2], bfr[2][2];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+        af[mf][kk] = *(const bf16x8*)((const char*)As +
+                                      t8_lds_byte(wmq + mf * 16 + lrow,
+                                                  chunk));
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        bfr[nf][kk] = *(const bf16x8*)((const char*)Bs +
+                                       t8_lds_byte(wnq + nf * 16 + lrow,
+                                                   chunk));
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[q][mf * 2 + nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf][kk], bfr[nf][kk], acc[q][mf * 2 + nf], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  };
+
+  // prologue: tile 0's 4 halves
+#pragma unroll
+  for (int h = 0; h < 4; ++h) issue_half(h, 0);
+
+  // quadrant order: (A0,B0) (A1,B0) (A1,B1) (A0,B1); q index = phase index
+  for (int tile = 0; tile < ksteps; ++tile) {
+    int nxt = tile + 1;
+    bool pre = nxt < ksteps;
+    // p0
+    if (pre) issue_half(0, nxt);
+    phase(tile, 0, 0, 0, pre ? 1 : 0);
+    // p1
+    if (pre) issue_half(1, nxt);
+    phase(tile, 1, 0, 1, pre ? 1 : -1);
+    // p2
+    if (pre) issue_half(2, nxt);
+    phase(tile, 1, 1, 2, pre ? 1 : -1);
+    // p3
+    if (pre) issue_half(3, nxt);
+    phase(tile, 0, 1, 3, -1);
+  }
+
+  // epilogue
+  bf16* cz = Cout + (long)z * strideC;
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    int qa = (q == 1 || q == 2) ? 1 : 0;  // A-half of quadrant q
+    int qb = (q >= 2) ? 1 : 0;            // B-half
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int gcol = bn + qb * 128 + wnq + nf * 16 + (l & 15);
+        if (gcol >= N) continue;
+        float bv = BIAS ? bias[gcol] : 0.f;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int grow = bm + qa * 128 + wmq + mf * 16 + (l >> 4) * 4 + i;
+          if (grow >= M) continue;
+          float v = acc[q][mf * 2 + nf][i] + bv;
+          if (RELU) v = v > 0.f ? v : 0.f;
+          cz[(long)grow * ldc + gcol] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+void gemm_tn8_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
+                   int M, int N, int K, long lda, long ldb, long ldc,
+                   long strideA, long strideB, long strideC, int batches,
+                   bool relu, const bf16* zpage, hipStream_t stream) {
+  Plain8Loader a{A, zpage, M, lda, strideA};
+  Plain8Loader b{B, zpage, N, ldb, strideB};
+  dim3 grid(cdiv_h(M, T8_BM) * cdiv_h(N, T8_BN), 1, batches);
+  int ksteps = K / T8_BK;
+  if (bias) {
+    if (relu)
+      gemm_tn8_core<true, true><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    else
+      gemm_tn8_core<false, true><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+  } else {
+    if (relu)
+      gemm_tn8_core<true, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    else
+      gemm_tn8_core<false, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+  }
+}
+
+}  // namespace mxa
