@@ -40,6 +40,7 @@ from . import symbol as sym
 from . import rtc
 from . import contrib
 from . import monitor
+from . import operator
 from . import image
 from . import visualization
 from . import test_utils

@@ -145,3 +145,31 @@ def test_rtc_module_compiles_source():
         pytest.skip(f"hiprtc unavailable here: {e}")
     k = mod.get_kernel("add1", "float*, int")
     assert k is not None
+
+
+def test_custom_op():
+    from anand_mxnet_amd import operator as op_mod
+
+    class Sigmoid(op_mod.CustomOp):
+        def forward(self, is_train, req, in_data, out_data, aux):
+            x = in_data[0]
+            self.assign(out_data[0], req[0], nd.sigmoid(x))
+
+        def backward(self, req, out_grad, in_data, out_data, in_grad, aux):
+            y = out_data[0]
+            self.assign(in_grad[0], req[0], out_grad[0] * y * (1 - y))
+
+    @op_mod.register("test_sigmoid")
+    class SigmoidProp(op_mod.CustomOpProp):
+        def create_operator(self, ctx, shapes, dtypes):
+            return Sigmoid()
+
+    x = nd.array([[0.0, 1.0], [-1.0, 2.0]])
+    x.attach_grad()
+    with autograd.record():
+        y = nd.Custom(x, op_type="test_sigmoid")
+        loss = y.sum()
+    loss.backward()
+    ref = 1 / (1 + np.exp(-x.asnumpy()))
+    np.testing.assert_allclose(y.asnumpy(), ref, rtol=1e-5)
+    np.testing.assert_allclose(x.grad.asnumpy(), ref * (1 - ref), rtol=1e-5)
