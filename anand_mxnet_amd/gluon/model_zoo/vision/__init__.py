@@ -12,6 +12,7 @@ from .mobilenet import (mobilenet1_0, mobilenet0_75, mobilenet0_5,
 from .squeezenet import squeezenet1_0, squeezenet1_1, SqueezeNet  # noqa: F401
 from .densenet import (densenet121, densenet161, densenet169, densenet201,
                        DenseNet)  # noqa: F401
+from .inception import inception_v3, Inception3  # noqa: F401
 from . import resnet  # noqa: F401
 
 _models = {}
@@ -26,9 +27,12 @@ def _register_models():
                  "vgg11", "vgg13", "vgg16", "vgg19", "vgg11_bn", "vgg13_bn",
                  "vgg16_bn", "vgg19_bn", "mobilenet1.0", "mobilenet0.75",
                  "mobilenet0.5", "mobilenet0.25", "mobilenetv2_1.0",
-                 "squeezenet1.0", "squeezenet1.1", "densenet121",
+                 "squeezenet1.0", "squeezenet1.1", "inceptionv3",
+                 "densenet121",
                  "densenet161", "densenet169", "densenet201"]:
         attr = name.replace(".", "_").replace("mobilenetv2", "mobilenet_v2")
+        if attr == "inceptionv3":
+            attr = "inception_v3"
         if hasattr(mod, attr):
             _models[name] = getattr(mod, attr)
 

@@ -152,8 +152,10 @@ at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
   long sA = batched ? M * K : 0;
   long sB = (batched && B.dim() == 3) ? N * K : 0;
   long sC = batched ? M * N : 0;
-  if (M >= 256 && N >= 256) {
-    // deep-pipelined 256² 8-phase kernel (NLP GEMM shapes)
+  long blocks8 = ((M + 255) / 256) * ((N + 255) / 256) * (long)batches;
+  if (M >= 256 && N >= 256 && blocks8 >= 240) {
+    // deep-pipelined 256² 8-phase kernel (needs a full chip of tiles;
+    // smaller grids run the 128² kernel which fills better)
     mxa::gemm_tn8_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
                        (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
                        cur_stream());
