@@ -59,6 +59,27 @@ class Engine:
         """Dedicated RCCL stream so collectives overlap backward compute."""
         return self._get_stream(self._comm_streams, device)
 
+    def new_event(self):
+        """VarHandle-style sync primitive (reference Engine::NewVariable/
+        WaitForVar): record on a producing stream, wait from a consumer."""
+        if self._native is not None:
+            return self._native.engine_new_event()
+        return None
+
+    def record_event(self, ev, stream=None):
+        if self._native is not None and ev is not None:
+            s = stream.cuda_stream if stream is not None else                 torch.cuda.current_stream().cuda_stream
+            self._native.engine_record_event(ev, s)
+
+    def stream_wait_event(self, stream, ev):
+        if self._native is not None and ev is not None:
+            self._native.engine_stream_wait_event(stream.cuda_stream, ev)
+
+    def wait_for_var(self, ev):
+        """Block the host until the producer stream passed the event."""
+        if self._native is not None and ev is not None:
+            self._native.engine_sync_event(ev)
+
     def wait_all(self):
         if torch.cuda.is_available():
             for i in range(torch.cuda.device_count()):

@@ -607,6 +607,34 @@ at::Tensor multi_sumsq_run(const at::Tensor& blob, long ntens, long nchunks,
 // engine: raw HIP stream management (wrapped by torch ExternalStream)
 // ---------------------------------------------------------------------------
 
+// hipEvent pool: WaitForVar/cross-stream ordering primitives (reference
+// engine.h VarHandle semantics collapse to event waits under stream order)
+long engine_new_event() {
+  hipEvent_t e;
+  TORCH_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming) ==
+              hipSuccess);
+  return (long)(uintptr_t)e;
+}
+
+void engine_record_event(long ev, long stream) {
+  TORCH_CHECK(hipEventRecord((hipEvent_t)(uintptr_t)ev,
+                             (hipStream_t)(uintptr_t)stream) == hipSuccess);
+}
+
+void engine_stream_wait_event(long stream, long ev) {
+  TORCH_CHECK(hipStreamWaitEvent((hipStream_t)(uintptr_t)stream,
+                                 (hipEvent_t)(uintptr_t)ev, 0) ==
+              hipSuccess);
+}
+
+void engine_sync_event(long ev) {
+  TORCH_CHECK(hipEventSynchronize((hipEvent_t)(uintptr_t)ev) == hipSuccess);
+}
+
+void engine_destroy_event(long ev) {
+  (void)hipEventDestroy((hipEvent_t)(uintptr_t)ev);
+}
+
 long engine_new_stream(long device) {
   hipStream_t s;
   int prev;
@@ -649,4 +677,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_opt_run", &multi_opt_run);
   m.def("multi_sumsq_run", &multi_sumsq_run);
   m.def("engine_new_stream", &engine_new_stream);
+  m.def("engine_new_event", &engine_new_event);
+  m.def("engine_record_event", &engine_record_event);
+  m.def("engine_stream_wait_event", &engine_stream_wait_event);
+  m.def("engine_sync_event", &engine_sync_event);
+  m.def("engine_destroy_event", &engine_destroy_event);
 }
