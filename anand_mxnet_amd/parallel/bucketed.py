@@ -128,15 +128,25 @@ class BucketedAllReducer(GradFlattener):
     def _hook(self, param: torch.Tensor):
         b = self._t2bucket[param]
         if b.ready():
-            self._launch(b)
+            b.ready_flag = True
+            # launch in fixed bucket order: every rank must issue the same
+            # collective sequence on the communicator (DDP-style guard
+            # against autograd scheduling divergence -> NCCL hang)
+            while self._next_launch < len(self.buckets) and \
+                    getattr(self.buckets[self._next_launch], "ready_flag",
+                            False):
+                self._launch(self.buckets[self._next_launch])
+                self._next_launch += 1
+
+    _next_launch = 0
 
     def finish(self):
         """Complete all reductions (called from Trainer before update)."""
         for b in self.buckets:
-            if b.work is None:  # non-overlap mode or hook missed (grad_req null)
-                if b.pending > 0 or not self.overlap:
-                    self._launch(b)
-            if b.work is not None:
-                b.work.wait()
+            if b.work is None:  # non-overlap mode / hook missed / stragglers
+                self._launch(b)
+            b.work.wait()
         for b in self.buckets:
             b.reset()
+            b.ready_flag = False
+        self._next_launch = 0
