@@ -257,6 +257,257 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_core(
 }
 
 // ---------------------------------------------------------------------------
+// tr-variant: hardware-transpose fragment reads (ds_read_b64_tr_b16)
+// ---------------------------------------------------------------------------
+// Empirically probed semantics (tools/probe_tr.hip on gfx950): within each
+// 16-lane group, writing the in-group lane index as (r,c) = (l>>2, l&3),
+// output slot j of lane (r*4+c) is slot c of the 8-byte segment addressed
+// by lane (j*4+r) — a rotation of the (lane_hi2, lane_lo2, slot) cube.
+// Consequence: tiles can be staged ROW-major [m][cols] straight from
+// global memory with global_load_lds (no in-register zip, no LDS write
+// instructions at all), and MFMA k-major fragments come out of two tr
+// reads per lane.  The zip kernel above is kept as a fallback
+// (MXA_WG_ZIP=1) and an A/B baseline.
+//
+// LDS image per operand: [64 m-rows][128 cols] bf16, 256-B rows, chunk16
+// XOR swizzle  chunk' = chunk ^ ((row&3)<<1)  so the 4 rows of one tr
+// read land in disjoint 32-B bank windows (even bank coverage).
+
+DEVINL int tr_byte(int row, int cbyte) {
+  return row * 256 + ((((cbyte >> 4) ^ ((row & 3) << 1)) << 4) | (cbyte & 15));
+}
+
+struct TrPlainRows {
+  const bf16* base;
+  const bf16* zpage;
+  long mtotal;
+  long ld;
+  int rowlen;
+  // device state
+  const bf16* ptr[4];  // pass p -> this thread's source row pointer
+  long mrem[4];
+  bool colok;
+  int cgbyte;
+
+  DEVINL void init(int col0, long m0, const int* myrow, int cg) {
+    cgbyte = cg * 16;
+    int col = col0 + cg * 8;
+    colok = col + 8 <= rowlen;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      ptr[p] = base + (m0 + myrow[p]) * ld + col;
+      mrem[p] = mtotal - (m0 + myrow[p]);
+    }
+  }
+  DEVINL void step() {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      ptr[p] += (long)WBK * ld;
+      mrem[p] -= WBK;
+    }
+  }
+  DEVINL const bf16* src(int p) const {
+    return (colok && mrem[p] > 0) ? ptr[p]
+                                  : (const bf16*)((const char*)zpage + cgbyte);
+  }
+};
+
+struct TrIm2colRows {
+  const bf16* x;
+  const bf16* zpage;
+  int Nb, H, W, C, P, Q;
+  int sh, sw, ph, pw;
+  // device state
+  int roff, soff;
+  int n_[4], p_[4], q_[4];
+  bool colok;
+  int col, cgbyte;
+
+  DEVINL void init(int col0, long m0, const int* myrow, int cg) {
+    cgbyte = cg * 16;
+    col = col0 + cg * 8;
+    colok = col + 8 <= C;
+    int PQ = P * Q;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      long m = m0 + myrow[p];
+      n_[p] = (int)(m / PQ);
+      int pq = (int)(m % PQ);
+      p_[p] = pq / Q;
+      q_[p] = pq % Q;
+    }
+  }
+  DEVINL void step() {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      q_[i] += WBK;
+      while (q_[i] >= Q) {
+        q_[i] -= Q;
+        p_[i] += 1;
+      }
+      while (p_[i] >= P) {
+        p_[i] -= P;
+        n_[i] += 1;
+      }
+    }
+  }
+  DEVINL const bf16* src(int p) const {
+    const bf16* zp = (const bf16*)((const char*)zpage + cgbyte);
+    if (n_[p] >= Nb || !colok) return zp;
+    int ih = p_[p] * sh - ph + roff;
+    int iw = q_[p] * sw - pw + soff;
+    if (ih < 0 || ih >= H || iw < 0 || iw >= W) return zp;
+    return x + (((long)n_[p] * H + ih) * W + iw) * C + col;
+  }
+};
+
+template <bool A_IM2COL>
+__global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
+    TrPlainRows aplain, TrIm2colRows aim, TrPlainRows bload,
+    float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
+    int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic) {
+  __shared__ __align__(16) bf16 smem[2][2][WBK * WBM];  // [buf][op][64][128]
+
+  const int seg = blockIdx.y / nsplit, split = blockIdx.y % nsplit;
+  const int per = (ksteps_total + nsplit - 1) / nsplit;
+  const int kstep0 = split * per;
+  const int ksteps = min(per, ksteps_total - kstep0);
+  if (ksteps <= 0) return;
+
+  const int ntn = (No + WBN - 1) / WBN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * WBM, bn = (bid % ntn) * WBN;
+  const int t = threadIdx.x;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  // glds slot of pass p: LDS bytes (lw*4+p)*1024 + ll*16
+  //   -> row = lw*16 + p*4 + (ll>>4), chunk position ll&15.
+  // Swizzle moves to the SOURCE chunk: cg = (ll&15) ^ key(row),
+  // key(row) = (row&3)<<1 = (ll>>4)<<1 (uniform per lane).
+  const int cg = (ll & 15) ^ ((ll >> 4) << 1);
+  int myrow[4];
+#pragma unroll
+  for (int p = 0; p < 4; ++p) myrow[p] = lw * 16 + p * 4 + (ll >> 4);
+
+  long m0 = (long)kstep0 * WBK;
+  if (A_IM2COL) {
+    aim.roff = (seg / S) * dh;
+    aim.soff = (seg % S) * dw;
+    aim.init(bm, m0, myrow, cg);
+  } else {
+    aplain.init(bm, m0, myrow, cg);
+  }
+  bload.init(bn, m0, myrow, cg);
+
+  auto stage_glds = [&](int buf) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      unsigned lbase =
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 4 + p) * 1024));
+      auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[buf][0][0] + lbase);
+      auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[buf][1][0] + lbase);
+      auto gA = (const __attribute__((address_space(1))) unsigned int*)(
+          A_IM2COL ? aim.src(p) : aplain.src(p));
+      auto gB =
+          (const __attribute__((address_space(1))) unsigned int*)bload.src(p);
+      __builtin_amdgcn_global_load_lds(gA, ldsA, 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(gB, ldsB, 16, 0, 0);
+    }
+    if (A_IM2COL)
+      aim.step();
+    else
+      aplain.step();
+    bload.step();
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int wm = (lw >> 1) * 64, wn = (lw & 1) * 64;
+  const int ig = ll & 15, khi = ll >> 4;
+
+  auto trread = [&](const bf16* img, int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)img + tr_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      // fragment k-rows this lane addresses: kk*32 + khi*8 + h*4 + (ig>>2)
+      int rbase = kk * 32 + khi * 8 + (ig >> 2);
+      int cb = (ig & 3) * 8;
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        int cbyte = (wm + mf * 16) * 2 + cb;
+        bf16x4 lo = trread(smem[buf][0], rbase, cbyte);
+        bf16x4 hi = trread(smem[buf][0], rbase + 4, cbyte);
+        af[mf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int cbyte = (wn + nf * 16) * 2 + cb;
+        bf16x4 lo = trread(smem[buf][1], rbase, cbyte);
+        bf16x4 hi = trread(smem[buf][1], rbase + 4, cbyte);
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+    }
+  };
+
+  // 2-phase glds schedule (same as gemm_tn): issue next-tile DMA before
+  // computing the current tile; __syncthreads() drains the DMA.
+  stage_glds(0);
+  __syncthreads();
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    if (ks + 1 < ksteps) stage_glds(buf ^ 1);
+    compute(buf);
+    __syncthreads();
+  }
+
+  float* cseg = Cout + (long)seg * strideSeg;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wn + nf * 16 + (ll & 15);
+      if (gcol >= No) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (ll >> 4) * 4 + i;
+        if (grow >= Mo) continue;
+        float v = acc[mf][nf][i];
+        if (use_atomic)
+          atomicAdd(&cseg[(long)grow * ldc + gcol], v);
+        else
+          cseg[(long)grow * ldc + gcol] = v;
+      }
+    }
+  }
+}
+
+static bool wg_use_zip() {
+  static const bool z = [] {
+    const char* e = getenv("MXA_WG_ZIP");
+    return e && e[0] == '1';
+  }();
+  return z;
+}
+
+// ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
 
@@ -274,13 +525,21 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, long mtotal,
   int ksteps = (int)((mtotal + WBK - 1) / WBK);
   long bxy = (long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN);
   int nsplit = pick_split(bxy, ksteps);
-  WgPlainRows a{A, zpage, mtotal, lda, Mo};
-  WgPlainRows b{B, zpage, mtotal, ldb, No};
-  WgIm2colRows dummy{};
   dim3 grid((unsigned)bxy, nsplit, 1);
   int use_atomic = nsplit > 1;
   if (use_atomic) HIP_CHECK(hipMemsetAsync(C, 0, sizeof(float) * Mo * ldc,
                                            stream));
+  if (!wg_use_zip() && lda % 8 == 0 && ldb % 8 == 0) {
+    TrPlainRows a{A, zpage, mtotal, lda, Mo};
+    TrPlainRows b{B, zpage, mtotal, ldb, No};
+    TrIm2colRows dummy{};
+    gemm_wg_tr_core<false><<<grid, WTHREADS, 0, stream>>>(
+        a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic);
+    return;
+  }
+  WgPlainRows a{A, zpage, mtotal, lda, Mo};
+  WgPlainRows b{B, zpage, mtotal, ldb, No};
+  WgIm2colRows dummy{};
   gemm_wg_core<false><<<grid, WTHREADS, 0, stream>>>(
       a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic);
 }
@@ -293,14 +552,23 @@ void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
   int ksteps = (int)((mtotal + WBK - 1) / WBK);
   long bxy = (long)cdiv_h(C, WBM) * cdiv_h(Kout, WBN);
   int nsplit = pick_split(bxy * R * S, ksteps);
-  WgIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
-  WgPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
-  WgPlainRows dummy{};
   dim3 grid((unsigned)bxy, R * S * nsplit, 1);
   int use_atomic = nsplit > 1;
   if (use_atomic)
     HIP_CHECK(hipMemsetAsync(dw_rsck, 0,
                              sizeof(float) * (long)R * S * C * Kout, stream));
+  if (!wg_use_zip() && C % 8 == 0 && Kout % 8 == 0) {
+    TrIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
+    TrPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
+    TrPlainRows dummy{};
+    gemm_wg_tr_core<true><<<grid, WTHREADS, 0, stream>>>(
+        dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
+        ksteps, nsplit, use_atomic);
+    return;
+  }
+  WgIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
+  WgPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
+  WgPlainRows dummy{};
   gemm_wg_core<true><<<grid, WTHREADS, 0, stream>>>(
       dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
       ksteps, nsplit, use_atomic);

@@ -91,8 +91,31 @@ def bench_softmax(rows, D):
             "ms": round(t * 1e3, 3)}
 
 
+def bench_conv_wg(N, H, W, C, K, R, s, p, name=""):
+    x = (torch.rand((N, H, W, C), device=DEV) * 2 - 1).to(BF)
+    P = (H + 2 * p - R) // s + 1
+    flops = 2.0 * N * P * P * K * R * R * C
+    dy = (torch.rand((N, P, P, K), device=DEV) * 2 - 1).to(BF)
+    tw = timeit(lambda: _C.conv2d_wgrad(x, dy, [R, R], [s, s], [p, p],
+                                        [1, 1]))
+    return {"op": f"convwg {name} N{N} {H}x{W}x{C}->K{K} r{R}s{s}",
+            "wgrad_TF": round(flops / tw / 1e12, 1)}
+
+
 def main():
     results = []
+    if "--wg-only" in sys.argv:
+        results.append(bench_wg(100352, 128, 128))
+        results.append(bench_wg(25088, 512, 256))
+        results.append(bench_wg(8192, 4096, 4096))
+        results.append(bench_conv_wg(64, 56, 56, 64, 64, 3, 1, 1, "res2-3x3"))
+        results.append(bench_conv_wg(64, 56, 56, 64, 256, 1, 1, 0, "res2-1x1"))
+        results.append(bench_conv_wg(64, 28, 28, 128, 128, 3, 1, 1, "res3-3x3"))
+        results.append(bench_conv_wg(64, 14, 14, 256, 256, 3, 1, 1, "res4-3x3"))
+        results.append(bench_conv_wg(64, 7, 7, 512, 512, 3, 1, 1, "res5-3x3"))
+        for r in results:
+            print(json.dumps(r))
+        return
     for shape in [(4096, 4096, 4096), (8192, 8192, 8192),
                   (12544, 256, 1152), (3136, 512, 2304), (25088, 128, 576)]:
         results.append(bench_gemm(*shape))
