@@ -44,8 +44,11 @@ class _GemmTN(torch.autograd.Function):
         M = a.shape[-2]
         da = db = None
         if ctx.needs_input_grad[0]:
-            # dA[M,K] = dY[M,N] @ B[N,K] -> TN with B^T[K,N]
-            if N % 64 == 0:
+            # dA[M,K] = dY[M,N] @ B[N,K] — the NN-form kernel consumes
+            # dY (N-contiguous) and B (K-contiguous) in place, no B^T copy
+            if a.dim() == 2 and N % 64 == 0 and a.shape[-1] % 8 == 0:
+                da = _C.gemm_nn(dy, b)
+            elif N % 64 == 0:
                 da = _C.gemm_tn(dy, b.transpose(-1, -2).contiguous(), None,
                                 False)
             else:  # non-tileable reduce dim: library GEMM
@@ -112,14 +115,14 @@ class _Linear(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            if w.shape[0] % 64 == 0:
-                dx = _C.gemm_tn(dy, w.t().contiguous(), None, False)
+            if w.shape[0] % 64 == 0 and w.shape[1] % 8 == 0:
+                dx = _C.gemm_nn(dy, w)  # no W^T copy (NN-form kernel)
             else:  # e.g. the 1000-class head: library GEMM for dgrad
                 dx = torch.matmul(dy, w)
         if ctx.needs_input_grad[1]:
             dw = _wg_or_matmul(dy, x, w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = dy.float().sum(0)
+            db = _C.colsum(dy)
         return dx, dw, db
 
 
@@ -169,7 +172,7 @@ class _Conv2dNHWC(torch.autograd.Function):
                                   list(dil))  # [R,S,C,K] f32
             dw = dwf.permute(3, 0, 1, 2).to(w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = dy.float().sum(dim=(0, 1, 2))
+            db = _C.colsum(dy)
         return dx, dw, db, None, None, None
 
 
@@ -221,7 +224,7 @@ class _Conv2dIm2col(torch.autograd.Function):
             dwf = _C.gemm_wg(dy2, A)       # [K, kdim_padded] f32
             dw = dwf[:, :kdim].reshape(w.shape).to(w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = dy2.float().sum(0)
+            db = _C.colsum(dy2)
         if ctx.needs_input_grad[0]:
             raise RuntimeError(
                 "im2col conv path has no dgrad (stem layer input needs no "
@@ -379,7 +382,7 @@ def pool_nhwc(x, kernel, pool_type, global_pool, stride, pad,
 class _Softmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, log_):
-        y = _C.softmax_fwd(x.contiguous(), log_)
+        y = _C.softmax_fwd(x.contiguous(), log_, 1.0)
         ctx.save_for_backward(y)
         ctx.log = log_
         return y
@@ -387,7 +390,7 @@ class _Softmax(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (y,) = ctx.saved_tensors
-        return _C.softmax_bwd(y, dy.contiguous(), ctx.log), None
+        return _C.softmax_bwd(y, dy.contiguous(), ctx.log, 1.0), None
 
 
 @register_gpu("softmax")
@@ -537,7 +540,7 @@ def multi_sumsq(blob, ntens, nchunks, bf16in):
 class _MaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, vlen):
-        y = _C.softmax_masked_fwd(x.contiguous(), vlen)
+        y = _C.softmax_masked_fwd(x.contiguous(), vlen, 1.0)
         ctx.save_for_backward(y)
         return y
 
@@ -545,7 +548,7 @@ class _MaskedSoftmax(torch.autograd.Function):
     def backward(ctx, dy):
         (y,) = ctx.saved_tensors
         # masked cols have y == 0 -> zero grad automatically
-        return _C.softmax_bwd(y, dy.contiguous(), False), None
+        return _C.softmax_bwd(y, dy.contiguous(), False, 1.0), None
 
 
 _causal_vlen_cache = {}
@@ -561,11 +564,61 @@ def _causal_vlen(BH, T, device):
     return v
 
 
+class _MhaCore(torch.autograd.Function):
+    """Zero-copy multi-head attention: heads are read from / written to the
+    [B, S, H*D] activation IN PLACE by the three layout-form GEMM kernels
+    (attn_scores = TN, attn_ctx = NN with tr-reads, attn_wg = WG) — no
+    [B,H,S,D] permute copies anywhere in fwd or bwd, and the 1/sqrt(d)
+    scale is folded into the softmax kernel."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, H, causal, scale, dropout_p, training):
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        B, Tq, HD = q.shape
+        s = _C.attn_scores(q, k, H)                  # [BH, Tq, Tk] logits
+        if causal:
+            p = _C.softmax_masked_fwd(
+                s, _causal_vlen(B * H, Tq, s.device), scale)
+        else:
+            p = _C.softmax_fwd(s, False, scale)
+        if dropout_p > 0.0 and training:
+            pd, mask = _C.dropout_fwd(p, dropout_p, _next_seed())
+        else:
+            pd, mask = p, None
+        out = _C.attn_ctx(pd, v, H)                  # [B, Tq, HD]
+        ctx.save_for_backward(q, k, v, p, mask)
+        ctx.H, ctx.scale, ctx.dropout_p = H, scale, dropout_p
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, p, mask = ctx.saved_tensors
+        H, scale, dropout_p = ctx.H, ctx.scale, ctx.dropout_p
+        dout = dout.contiguous()
+        dpd = _C.attn_scores(dout, v, H)             # dP (post-dropout)
+        if mask is not None:
+            pd = _C.dropout_bwd(p, mask, dropout_p)  # recompute dropped P
+            dv = _C.attn_wg(pd, dout, H).to(v.dtype)
+            dp = _C.dropout_bwd(dpd, mask, dropout_p)
+        else:
+            dv = _C.attn_wg(p, dout, H).to(v.dtype)
+            dp = dpd
+        ds = _C.softmax_bwd(p, dp, False, scale)     # [BH, Tq, Tk]
+        dq = _C.attn_ctx(ds, k, H)
+        dk = _C.attn_wg(ds, q, H).to(k.dtype)
+        return dq, dk, dv, None, None, None, None, None
+
+
 @register_gpu("mha")
 def mha(q, k, v, num_heads, causal, scale, dropout_p=0.0, training=False):
     B, Tq, HD = q.shape
     Tk = k.shape[1]
     D = HD // num_heads
+    if (D % 64 == 0 and Tk % 64 == 0 and Tq % 8 == 0
+            and q.dtype == torch.bfloat16):
+        return _MhaCore.apply(q, k, v, num_heads, causal, scale,
+                              dropout_p, training)
+    # fallback (odd head dims / seq lens): materialized head layout
     qh = q.reshape(B, Tq, num_heads, D).transpose(1, 2) \
         .reshape(B * num_heads, Tq, D).contiguous()
     kh = k.reshape(B, Tk, num_heads, D).transpose(1, 2) \
@@ -620,14 +673,14 @@ class _Fp8Linear(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            if w.shape[0] % 64 == 0:
-                dx = _C.gemm_tn(dy, w.t().contiguous(), None, False)
+            if w.shape[0] % 64 == 0 and w.shape[1] % 8 == 0:
+                dx = _C.gemm_nn(dy, w)
             else:
                 dx = torch.matmul(dy, w)
         if ctx.needs_input_grad[1]:
             dw = _wg_or_matmul(dy, x, w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = dy.float().sum(0)
+            db = _C.colsum(dy)
         return dx, dw, db
 
 

@@ -10,7 +10,7 @@
 #include <unordered_map>
 #include <vector>
 
-typedef __bf16 bf16;
+#include "kernels/common.h"  // bf16 typedef, ZMap batch map
 
 namespace mxa {
 // gemm_tn.hip
@@ -54,11 +54,13 @@ void avgpool_nhwc(const bf16*, bf16*, int, int, int, int, int, int, int,
 void gap_nhwc(const bf16*, bf16*, int, long, int, hipStream_t);
 void gap_nhwc_bwd(const bf16*, bf16*, int, long, int, hipStream_t);
 // softmax_norm.hip
-void softmax_fwd_bf16(const bf16*, bf16*, long, int, bool, hipStream_t);
+void softmax_fwd_bf16(const bf16*, bf16*, long, int, bool, float,
+                      hipStream_t);
 void softmax_masked_fwd_bf16(const bf16*, bf16*, long, int, const int*,
+                             float,
                              hipStream_t);
 void softmax_bwd_bf16(const bf16*, const bf16*, bf16*, long, int, bool,
-                      hipStream_t);
+                      float, hipStream_t);
 void layernorm_fwd_bf16(const bf16*, bf16*, const float*, const float*,
                         float*, float*, long, int, float, hipStream_t);
 void layernorm_bwd_bf16(const bf16*, const bf16*, bf16*, const float*,
@@ -73,6 +75,15 @@ void embedding_fwd_bf16(const long*, const bf16*, bf16*, long, int, long,
                         hipStream_t);
 void embedding_bwd_bf16(const long*, const bf16*, float*, long, int, long,
                         hipStream_t);
+void colsum_bf16(const bf16*, float*, long, int, hipStream_t);
+// gemm_nn.hip / attention entry points (gemm_tn.hip, gemm_wg.hip)
+void gemm_nn_bf16(const bf16*, ZMap, long, int, const bf16*, ZMap, long,
+                  int, bf16*, ZMap, long, int, int, const bf16*,
+                  hipStream_t);
+void attn_scores_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
+                      int, const bf16*, hipStream_t);
+void attn_wg_bf16(const bf16*, const bf16*, float*, int, int, int, int,
+                  int, const bf16*, hipStream_t);
 void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
 void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
 // gemm_fp8.hip
@@ -423,33 +434,36 @@ at::Tensor global_avgpool_bwd(const at::Tensor& dy,
 // softmax / layernorm / dropout / embedding / activations
 // ---------------------------------------------------------------------------
 
-at::Tensor softmax_fwd(const at::Tensor& x, bool log_) {
+at::Tensor softmax_fwd(const at::Tensor& x, bool log_, double alpha) {
   int D = x.size(-1);
   TORCH_CHECK(D % 8 == 0, "softmax: last dim must be %8");
   long rows = x.numel() / D;
   auto y = at::empty_like(x);
-  mxa::softmax_fwd_bf16(bfp(x), bfp_mut(y), rows, D, log_, cur_stream());
+  mxa::softmax_fwd_bf16(bfp(x), bfp_mut(y), rows, D, log_, (float)alpha,
+                        cur_stream());
   return y;
 }
 
-at::Tensor softmax_masked_fwd(const at::Tensor& x, const at::Tensor& vlen) {
+at::Tensor softmax_masked_fwd(const at::Tensor& x, const at::Tensor& vlen,
+                              double alpha) {
   int D = x.size(-1);
   TORCH_CHECK(D % 8 == 0, "softmax: last dim must be %8");
   long rows = x.numel() / D;
   TORCH_CHECK(vlen.numel() == rows && vlen.scalar_type() == at::kInt);
   auto y = at::empty_like(x);
   mxa::softmax_masked_fwd_bf16(bfp(x), bfp_mut(y), rows, D,
-                               vlen.data_ptr<int>(), cur_stream());
+                               vlen.data_ptr<int>(), (float)alpha,
+                               cur_stream());
   return y;
 }
 
 at::Tensor softmax_bwd(const at::Tensor& y, const at::Tensor& dy,
-                       bool log_) {
+                       bool log_, double alpha) {
   int D = y.size(-1);
   long rows = y.numel() / D;
   auto dx = at::empty_like(y);
   mxa::softmax_bwd_bf16(bfp(y), bfp(dy), bfp_mut(dx), rows, D, log_,
-                        cur_stream());
+                        (float)alpha, cur_stream());
   return dx;
 }
 
@@ -484,6 +498,78 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x,
                           rstd.data_ptr<float>(), dgamma.data_ptr<float>(),
                           dbeta.data_ptr<float>(), rows, D, cur_stream());
   return {dx, dgamma, dbeta};
+}
+
+at::Tensor gemm_nn(const at::Tensor& a, const at::Tensor& b) {
+  // C[M][N] = A[M][k] @ B[k][N], A k-contiguous, B out-contiguous — the
+  // no-transpose dgrad form (dx = dy @ W without W^T copy).
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.is_contiguous() &&
+              b.is_contiguous());
+  long M = a.size(0), K = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == K && K % 64 == 0 && N % 8 == 0,
+              "gemm_nn: K%64, N%8 required");
+  auto out = at::empty({M, N}, a.options());
+  mxa::gemm_nn_bf16(bfp(a), ZMap{1, 0, 0}, K, (int)M, bfp(b),
+                    ZMap{1, 0, 0}, N, (int)N, bfp_mut(out),
+                    ZMap{1, 0, 0}, N, (int)K, 1, zp(a), cur_stream());
+  return out;
+}
+
+at::Tensor attn_scores(const at::Tensor& q, const at::Tensor& k, long H) {
+  // S[b*H+h][tq][tk] = sum_d q[b,tq,h*D+d] k[b,tk,h*D+d] — heads read in
+  // place from [B,S,H*D] (no permute copies).  Also dP with (dO, V).
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && q.is_contiguous() &&
+              k.is_contiguous());
+  long B = q.size(0), Tq = q.size(1), HD = q.size(2), Tk = k.size(1);
+  long D = HD / H;
+  TORCH_CHECK(D * H == HD && D % 64 == 0, "attn_scores: D%64 required");
+  auto out = at::empty({B * H, Tq, Tk}, q.options());
+  mxa::attn_scores_bf16(bfp(q), bfp(k), bfp_mut(out), (int)B, (int)H,
+                        (int)Tq, (int)Tk, (int)D, zp(q), cur_stream());
+  return out;
+}
+
+at::Tensor attn_ctx(const at::Tensor& p, const at::Tensor& v, long H) {
+  // O[b,tq,h*D+d] = sum_tk P[b*H+h][tq][tk] v[b,tk,h*D+d] (NN form,
+  // strided head output written in place).  Also dQ with (dS, K).
+  TORCH_CHECK(p.dim() == 3 && v.dim() == 3 && p.is_contiguous() &&
+              v.is_contiguous());
+  long BH = p.size(0), Tq = p.size(1), Tk = p.size(2);
+  long B = BH / H, HD = v.size(2), D = HD / H;
+  TORCH_CHECK(B * H == BH && D * H == HD && Tk % 64 == 0 && D % 8 == 0,
+              "attn_ctx: Tk%64, D%8 required");
+  auto out = at::empty({B, Tq, HD}, p.options());
+  mxa::gemm_nn_bf16(bfp(p), ZMap{1, Tq * Tk, 0}, Tk, (int)Tq, bfp(v),
+                    ZMap{(int)H, Tk * HD, D}, HD, (int)D,
+                    bfp_mut(out), ZMap{(int)H, Tq * HD, D}, HD,
+                    (int)Tk, (int)BH, zp(p), cur_stream());
+  return out;
+}
+
+at::Tensor attn_wg(const at::Tensor& p, const at::Tensor& q, long H) {
+  // dK[b,tk,h*D+d] = sum_tq dS[b*H+h][tq][tk] q[b,tq,h*D+d] (WG form,
+  // f32 out in [B,S,H*D] layout).  Also dV with (P, dO).
+  TORCH_CHECK(p.dim() == 3 && q.dim() == 3 && p.is_contiguous() &&
+              q.is_contiguous());
+  long BH = p.size(0), Tq = p.size(1), Tk = p.size(2);
+  long B = BH / H, HD = q.size(2), D = HD / H;
+  TORCH_CHECK(B * H == BH && D * H == HD && Tk % 8 == 0 && D % 8 == 0,
+              "attn_wg: Tk%8, D%8 required");
+  auto out = at::empty({B, Tk, HD}, q.options().dtype(at::kFloat));
+  mxa::attn_wg_bf16(bfp(p), bfp(q), out.data_ptr<float>(), (int)B, (int)H,
+                    (int)Tq, (int)Tk, (int)D, zp(p), cur_stream());
+  return out;
+}
+
+at::Tensor colsum(const at::Tensor& x) {
+  // bias gradient: sum over all leading dims of a [..., N] bf16 tensor,
+  // f32 result (replaces .float().sum(0) copy+reduce pair).
+  TORCH_CHECK(x.is_contiguous(), "colsum: contiguous input required");
+  int N = (int)x.size(-1);
+  long m = x.numel() / N;
+  auto out = at::empty({N}, x.options().dtype(at::kFloat));
+  mxa::colsum_bf16(bfp(x), out.data_ptr<float>(), m, N, cur_stream());
+  return out;
 }
 
 std::vector<at::Tensor> dropout_fwd(const at::Tensor& x, double p,
@@ -667,6 +753,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_bwd", &softmax_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("colsum", &colsum);
+  m.def("gemm_nn", &gemm_nn);
+  m.def("attn_scores", &attn_scores);
+  m.def("attn_ctx", &attn_ctx);
+  m.def("attn_wg", &attn_wg);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("embedding_fwd", &embedding_fwd);

@@ -25,6 +25,18 @@ typedef unsigned long long u64;
 DEVINL float bf2f(bf16 x) { return __bfloat162float(x); }
 DEVINL bf16 f2bf(float x) { return __float2bfloat16(x); }
 
+// Two-level batch offset map: z = zo * zdiv + zi -> zo*souter + zi*sinner.
+// Used for per-(batch, head) slices of a [B, S, H*D] activation so batched
+// attention GEMMs read heads IN PLACE (no [B,H,S,D] permute copies).
+// ZMap{1, stride, 0} is a plain contiguous batch; ZMap{1, 0, 0} unbatched.
+struct ZMap {
+  int zdiv;
+  long souter, sinner;
+  DEVINL long off(int z) const {
+    return (long)(z / zdiv) * souter + (long)(z % zdiv) * sinner;
+  }
+};
+
 // ceil-div
 DEVINL constexpr int cdiv(int a, int b) { return (a + b - 1) / b; }
 static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }

@@ -50,14 +50,14 @@ struct PlainLoader {
   const bf16* zpage;
   int rows;
   long ld;        // elements between rows
-  long stride_z;  // batch stride (elements)
+  ZMap zmap;      // batch offset map (ZMap{1, stride, 0} = plain batch)
   int ksteps;     // K / 64
   // device state
   const bf16* rp[4];
   long koff;
 
   DEVINL void init(int row0, int t, int z) {
-    const bf16* b = base + (long)z * stride_z;
+    const bf16* b = base + zmap.off(z);
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       int r = row0 + stage_row(p, t);
@@ -386,9 +386,23 @@ void gemm_tn_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
                   int M, int N, int K, long lda, long ldb, long ldc,
                   long strideA, long strideB, long strideC, int batches,
                   bool relu, const bf16* zpage, hipStream_t stream) {
-  PlainLoader a{A, zpage, M, lda, strideA, K / BK};
-  PlainLoader b{B, zpage, N, ldb, strideB, K / BK};
+  PlainLoader a{A, zpage, M, lda, ZMap{1, strideA, 0}, K / BK};
+  PlainLoader b{B, zpage, N, ldb, ZMap{1, strideB, 0}, K / BK};
   launch(a, b, C, bias, M, N, ldc, strideC, K / BK, relu, batches, stream);
+}
+
+// Batched attention scores/probs GEMM over per-(batch, head) slices of
+// [B, S, H*D] activations read IN PLACE (z = b*H + h):
+//   S[z][tq][tk] = sum_d q[b, tq, h*D + d] * k[b, tk, h*D + d]
+// Also computes dP (= dO @ V^T) with (q := dO, k := V).
+void attn_scores_bf16(const bf16* q, const bf16* k, bf16* s, int B, int H,
+                      int Tq, int Tk, int D, const bf16* zpage,
+                      hipStream_t stream) {
+  long hd = (long)H * D;
+  PlainLoader a{q, zpage, Tq, hd, ZMap{H, (long)Tq * hd, D}, D / BK};
+  PlainLoader b{k, zpage, Tk, hd, ZMap{H, (long)Tk * hd, D}, D / BK};
+  launch(a, b, s, nullptr, Tq, Tk, Tk, (long)Tq * Tk, D / BK, false, B * H,
+         stream);
 }
 
 void conv2d_fwd_bf16(const bf16* x, const bf16* w_rskc, bf16* y,

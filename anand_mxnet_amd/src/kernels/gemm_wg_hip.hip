@@ -366,8 +366,17 @@ template <bool A_IM2COL>
 __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
     TrPlainRows aplain, TrIm2colRows aim, TrPlainRows bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
-    int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic) {
+    int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic,
+    ZMap za, ZMap zb, ZMap zc) {
   __shared__ __align__(16) bf16 smem[2][2][WBK * WBM];  // [buf][op][64][128]
+
+  const int zb_ = blockIdx.z;
+  if (A_IM2COL)
+    aim.x += za.off(zb_);
+  else
+    aplain.base += za.off(zb_);
+  bload.base += zb.off(zb_);
+  Cout += zc.off(zb_);
 
   const int seg = blockIdx.y / nsplit, split = blockIdx.y % nsplit;
   const int per = (ksteps_total + nsplit - 1) / nsplit;
@@ -535,7 +544,8 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, long mtotal,
     TrPlainRows b{B, zpage, mtotal, ldb, No};
     TrIm2colRows dummy{};
    hipLaunchKernelGGL(( gemm_wg_tr_core<false>), dim3(grid), dim3(WTHREADS), 0, stream, 
-        a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic);
+        a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic,
+        ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0});
     return;
   }
   WgPlainRows a{A, zpage, mtotal, lda, Mo};
@@ -564,7 +574,8 @@ void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
     TrPlainRows dummy{};
    hipLaunchKernelGGL(( gemm_wg_tr_core<true>), dim3(grid), dim3(WTHREADS), 0, stream, 
         dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
-        ksteps, nsplit, use_atomic);
+        ksteps, nsplit, use_atomic,
+        ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0});
     return;
   }
   WgIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
@@ -573,6 +584,26 @@ void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
  hipLaunchKernelGGL(( gemm_wg_core<true>), dim3(grid), dim3(WTHREADS), 0, stream, 
       dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
       ksteps, nsplit, use_atomic);
+}
+
+// Batched attention dK/dV over per-(batch, head) slices (z = b*H + h):
+//   dK[z][tk][d] = sum_tq dS[z][tq][tk] * Q[b, tq, h*D + d]   (and
+//   dV with dS := P, Q := dO).  A is the contiguous [BH, Tq, X] probs /
+//   score-grads tensor; B and the f32 output live in [B, S, H*D] layout
+//   read/written in place.
+void attn_wg_bf16(const bf16* A, const bf16* Bq, float* C, int B, int H,
+                  int Tq, int Tk, int D, const bf16* zpage,
+                  hipStream_t stream) {
+  long hd = (long)H * D;
+  int ksteps = cdiv_h(Tq, WBK);
+  long bxy = (long)cdiv_h(Tk, WBM) * cdiv_h(D, WBN);
+  TrPlainRows a{A, zpage, Tq, (long)Tk, Tk};
+  TrPlainRows b{Bq, zpage, Tq, hd, (int)hd};
+  TrIm2colRows dummy{};
+  dim3 grid((unsigned)bxy, 1, (unsigned)(B * H));
+ hipLaunchKernelGGL(( gemm_wg_tr_core<false>), dim3(grid), dim3(WTHREADS), 0, stream, 
+      a, dummy, b, C, Tk, D, hd, 0, 1, 1, 1, ksteps, 1, 0,
+      ZMap{1, (long)Tq * Tk, 0}, ZMap{H, Tq * hd, D}, ZMap{H, Tk * hd, D});
 }
 
 }  // namespace mxa

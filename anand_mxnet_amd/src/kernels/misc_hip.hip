@@ -218,6 +218,33 @@ void embedding_bwd_bf16(const long* idx, const bf16* dy, float* dW,
                                                   vocab);
 }
 
+// Column sum (bias gradient): out[c] = sum_m x[m][c], bf16 in / f32 out.
+// One thread per column, rows split over blockIdx.y, f32 atomics to finish
+// (replaces torch's materialized .float() copy + reduce_kernel pair).
+__global__ void colsum_kernel(const bf16* __restrict__ x,
+                              float* __restrict__ out, long m, int N,
+                              long rows_per) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= N) return;
+  long r0 = blockIdx.y * rows_per;
+  long r1 = min(m, r0 + rows_per);
+  float acc = 0.f;
+  for (long r = r0; r < r1; ++r) acc += bf2f(x[r * N + col]);
+  if (gridDim.y > 1)
+    atomicAdd(&out[col], acc);
+  else
+    out[col] = acc;
+}
+
+void colsum_bf16(const bf16* x, float* out, long m, int N, hipStream_t s) {
+  int bx = cdiv_h(N, 256);
+  int nsplit = 1;
+  while ((long)bx * nsplit < 1024 && nsplit < m && nsplit < 512) nsplit *= 2;
+  long rows_per = (m + nsplit - 1) / nsplit;
+  if (nsplit > 1) HIP_CHECK(hipMemsetAsync(out, 0, N * sizeof(float), s));
+ hipLaunchKernelGGL(( colsum_kernel), dim3(dim3(bx, nsplit)), dim3(256), 0, s, x, out, m, N, rows_per);
+}
+
 void act_fwd_bf16(int op, const bf16* x, bf16* y, long n, hipStream_t s) {
   long nvec = n / 8;
   dim3 g = ew_grid(nvec);

@@ -50,7 +50,7 @@ DEVINL float block_red_max(float v, float* scratch) {
 template <bool LOG>
 __global__ __launch_bounds__(256) void softmax_fwd(
     const bf16* __restrict__ x, bf16* __restrict__ y, long rows, int D,
-    const int* __restrict__ vlen) {
+    const int* __restrict__ vlen, float alpha) {
   __shared__ float scratch[4];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const bf16* xr = x + row * D;
@@ -61,7 +61,7 @@ __global__ __launch_bounds__(256) void softmax_fwd(
       bf16x8 v = *(const bf16x8*)&xr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        if (i + j < V) mx = fmaxf(mx, bf2f(v[j]));
+        if (i + j < V) mx = fmaxf(mx, bf2f(v[j]) * alpha);
     }
     mx = block_red_max(mx, scratch);
     float sum = 0;
@@ -69,7 +69,7 @@ __global__ __launch_bounds__(256) void softmax_fwd(
       bf16x8 v = *(const bf16x8*)&xr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        if (i + j < V) sum += __expf(bf2f(v[j]) - mx);
+        if (i + j < V) sum += __expf(bf2f(v[j]) * alpha - mx);
     }
     sum = block_red_sum(sum, scratch);
     float inv = 1.f / sum, lse = __logf(sum);
@@ -81,7 +81,7 @@ __global__ __launch_bounds__(256) void softmax_fwd(
         if (i + j >= V) {
           o[j] = LOG ? f2bf(-3.0e38f) : (bf16)0.f;
         } else {
-          float e = bf2f(v[j]) - mx;
+          float e = bf2f(v[j]) * alpha - mx;
           o[j] = f2bf(LOG ? e - lse : __expf(e) * inv);
         }
       }
@@ -94,7 +94,7 @@ __global__ __launch_bounds__(256) void softmax_fwd(
 template <bool LOG>
 __global__ __launch_bounds__(256) void softmax_bwd(
     const bf16* __restrict__ y, const bf16* __restrict__ dy,
-    bf16* __restrict__ dx, long rows, int D) {
+    bf16* __restrict__ dx, long rows, int D, float alpha) {
   __shared__ float scratch[4];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const bf16* yr = y + row * D;
@@ -116,7 +116,7 @@ __global__ __launch_bounds__(256) void softmax_bwd(
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float yv = bf2f(vy[j]), gv = bf2f(vg[j]);
-        o[j] = f2bf(LOG ? gv - __expf(yv) * dot : (gv - dot) * yv);
+        o[j] = f2bf((LOG ? gv - __expf(yv) * dot : (gv - dot) * yv) * alpha);
       }
       *(bf16x8*)&or_[i] = o;
     }
@@ -265,26 +265,30 @@ static int rows_grid(long rows) {
 }
 
 void softmax_fwd_bf16(const bf16* x, bf16* y, long rows, int D, bool log_,
+                      float alpha,
                       hipStream_t s) {
   if (log_)
    hipLaunchKernelGGL(( softmax_fwd<true>), dim3(rows_grid(rows)), dim3(256), 0, s, x, y, rows, D,
-                                                      nullptr);
+                                                      nullptr, alpha);
   else
    hipLaunchKernelGGL(( softmax_fwd<false>), dim3(rows_grid(rows)), dim3(256), 0, s, x, y, rows, D,
-                                                       nullptr);
+                                                       nullptr, alpha);
 }
 
 void softmax_masked_fwd_bf16(const bf16* x, bf16* y, long rows, int D,
-                             const int* vlen, hipStream_t s) {
- hipLaunchKernelGGL(( softmax_fwd<false>), dim3(rows_grid(rows)), dim3(256), 0, s, x, y, rows, D, vlen);
+                             const int* vlen, float alpha, hipStream_t s) {
+ hipLaunchKernelGGL(( softmax_fwd<false>), dim3(rows_grid(rows)), dim3(256), 0, s, x, y, rows, D, vlen,
+                                                     alpha);
 }
 
 void softmax_bwd_bf16(const bf16* y, const bf16* dy, bf16* dx, long rows,
-                      int D, bool log_, hipStream_t s) {
+                      int D, bool log_, float alpha, hipStream_t s) {
   if (log_)
-   hipLaunchKernelGGL(( softmax_bwd<true>), dim3(rows_grid(rows)), dim3(256), 0, s, y, dy, dx, rows, D);
+   hipLaunchKernelGGL(( softmax_bwd<true>), dim3(rows_grid(rows)), dim3(256), 0, s, y, dy, dx, rows, D,
+                                                      alpha);
   else
-   hipLaunchKernelGGL(( softmax_bwd<false>), dim3(rows_grid(rows)), dim3(256), 0, s, y, dy, dx, rows, D);
+   hipLaunchKernelGGL(( softmax_bwd<false>), dim3(rows_grid(rows)), dim3(256), 0, s, y, dy, dx, rows, D,
+                                                       alpha);
 }
 
 void layernorm_fwd_bf16(const bf16* x, bf16* y, const float* gamma,

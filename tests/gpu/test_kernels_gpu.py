@@ -325,3 +325,87 @@ def test_cast_fp8_roundtrip():
     y8 = _C.cast_to_fp8(x.contiguous(), torch.ones((), device=_dev()))
     ref = x.float().to(torch.float8_e4m3fn).view(torch.uint8)
     assert (y8 == ref).float().mean().item() > 0.99
+
+
+def test_gemm_nn_vs_fp32():
+    for (M, N, K) in [(128, 128, 64), (256, 512, 192), (100, 136, 64),
+                      (512, 1000, 256)]:
+        if N % 8:
+            continue
+        a = _rt((M, K)).to(BF).float()
+        b = _rt((K, N)).to(BF).float()
+        ref = a @ b
+        out = _C.gemm_nn(a.to(BF).contiguous(), b.to(BF).contiguous())
+        assert_close(out, ref, msg=f"gemm_nn {M}x{N}x{K}")
+
+
+def test_colsum():
+    for (m, N) in [(1024, 128), (8192, 768), (777, 1000)]:
+        x = _rt((m, N)).to(BF)
+        ref = x.float().sum(0)
+        out = _C.colsum(x.contiguous())
+        assert_close(out, ref, rtol=1e-3, atol=1e-2, msg=f"colsum {m}x{N}")
+
+
+def _heads(x, H):
+    B, T, HD = x.shape
+    return x.reshape(B, T, H, HD // H).transpose(1, 2)  # [B,H,T,D]
+
+
+def test_attn_scores_ctx_wg():
+    B, H, Tq, Tk, D = 3, 4, 128, 192, 64
+    q = _rt((B, Tq, H * D)).to(BF)
+    k = _rt((B, Tk, H * D)).to(BF)
+    v = _rt((B, Tk, H * D)).to(BF)
+    s = _C.attn_scores(q.contiguous(), k.contiguous(), H)
+    ref_s = (_heads(q, H).float() @ _heads(k, H).float().transpose(-1, -2)) \
+        .reshape(B * H, Tq, Tk)
+    assert_close(s, ref_s, msg="attn_scores")
+
+    p = torch.nn.functional.softmax(ref_s.to(BF).float(), dim=-1).to(BF)
+    o = _C.attn_ctx(p.contiguous(), v.contiguous(), H)
+    ref_o = (p.float().reshape(B, H, Tq, Tk) @ _heads(v, H).float()) \
+        .transpose(1, 2).reshape(B, Tq, H * D)
+    assert_close(o, ref_o, msg="attn_ctx")
+
+    dk = _C.attn_wg(p.contiguous(), q.contiguous(), H)
+    ref_dk = (p.float().reshape(B, H, Tq, Tk).transpose(-1, -2)
+              @ _heads(q, H).float()).transpose(1, 2).reshape(B, Tk, H * D)
+    assert_close(dk, ref_dk, msg="attn_wg")
+
+
+def test_mha_fwd_bwd_vs_torch():
+    from anand_mxnet_amd.ops.hip import mha
+    B, H, T, D = 2, 4, 128, 64
+    scale = 1.0 / (D ** 0.5)
+    for causal in (False, True):
+        q0 = _rt((B, T, H * D)).to(BF)
+        k0 = _rt((B, T, H * D)).to(BF)
+        v0 = _rt((B, T, H * D)).to(BF)
+        q1 = q0.clone().requires_grad_(True)
+        k1 = k0.clone().requires_grad_(True)
+        v1 = v0.clone().requires_grad_(True)
+        out = mha(q1, k1, v1, H, causal, scale, 0.0, True)
+        g = _rt((B, T, H * D)).to(BF)
+        out.backward(g)
+
+        q2 = q0.clone().float().requires_grad_(True)
+        k2 = k0.clone().float().requires_grad_(True)
+        v2 = v0.clone().float().requires_grad_(True)
+        qh, kh, vh = _heads(q2, H), _heads(k2, H), _heads(v2, H)
+        sc = qh @ kh.transpose(-1, -2) * scale
+        if causal:
+            mask = torch.triu(torch.ones(T, T, device=_dev(), dtype=torch.bool),
+                              diagonal=1)
+            sc = sc.masked_fill(mask, float("-inf"))
+        pr = torch.nn.functional.softmax(sc, dim=-1)
+        ref = (pr @ vh).transpose(1, 2).reshape(B, T, H * D)
+        ref.backward(g.float())
+
+        assert_close(out, ref, rtol=5e-2, atol=5e-2, msg=f"mha out c={causal}")
+        assert_close(q1.grad, q2.grad, rtol=6e-2, atol=6e-2,
+                     msg=f"mha dq c={causal}")
+        assert_close(k1.grad, k2.grad, rtol=6e-2, atol=6e-2,
+                     msg=f"mha dk c={causal}")
+        assert_close(v1.grad, v2.grad, rtol=6e-2, atol=6e-2,
+                     msg=f"mha dv c={causal}")
