@@ -748,9 +748,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avgpool", &avgpool);
   m.def("global_avgpool", &global_avgpool);
   m.def("global_avgpool_bwd", &global_avgpool_bwd);
-  m.def("softmax_fwd", &softmax_fwd);
-  m.def("softmax_masked_fwd", &softmax_masked_fwd);
-  m.def("softmax_bwd", &softmax_bwd);
+  m.def("softmax_fwd", &softmax_fwd, py::arg("x"), py::arg("log_"),
+        py::arg("alpha") = 1.0);
+  m.def("softmax_masked_fwd", &softmax_masked_fwd, py::arg("x"),
+        py::arg("vlen"), py::arg("alpha") = 1.0);
+  m.def("softmax_bwd", &softmax_bwd, py::arg("y"), py::arg("dy"),
+        py::arg("log_"), py::arg("alpha") = 1.0);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("colsum", &colsum);

@@ -598,7 +598,10 @@ void attn_wg_bf16(const bf16* A, const bf16* Bq, float* C, int B, int H,
   int ksteps = cdiv_h(Tq, WBK);
   long bxy = (long)cdiv_h(Tk, WBM) * cdiv_h(D, WBN);
   TrPlainRows a{A, zpage, Tq, (long)Tk, Tk};
-  TrPlainRows b{Bq, zpage, Tq, hd, (int)hd};
+  // rowlen D (not H*D): the per-head slice is only D columns wide — a
+  // wider guard would let edge lanes read past the allocation on the
+  // final row of the final batch.
+  TrPlainRows b{Bq, zpage, Tq, hd, D};
   TrIm2colRows dummy{};
   dim3 grid((unsigned)bxy, 1, (unsigned)(B * H));
  hipLaunchKernelGGL(( gemm_wg_tr_core<false>), dim3(grid), dim3(WTHREADS), 0, stream, 
