@@ -36,7 +36,8 @@ class BasicBlockV1(HybridBlock):
         self.body.add(_conv3x3(channels, stride, in_channels, layout))
         self.body.add(nn.BatchNorm(axis=ax, fuse_relu=True))
         self.body.add(_conv3x3(channels, 1, channels, layout))
-        self.body.add(nn.BatchNorm(axis=ax))
+        # junction BN: relu(bn(conv_out) + residual) fused in one kernel
+        self.bn_join = nn.BatchNorm(axis=ax)
         if downsample:
             self.downsample = nn.HybridSequential(prefix="")
             self.downsample.add(nn.Conv2D(channels, kernel_size=1,
@@ -52,8 +53,7 @@ class BasicBlockV1(HybridBlock):
         x_out = self.body(x)
         if self.downsample is not None:
             residual = self.downsample(residual)
-        from .... import ndarray as F
-        return F.Activation(x_out + residual, act_type="relu")
+        return self.bn_join(x_out, residual)
 
     hybrid_forward = None
 
@@ -77,7 +77,8 @@ class BottleneckV1(HybridBlock):
         self.body.add(nn.BatchNorm(axis=ax, fuse_relu=True))
         self.body.add(nn.Conv2D(channels, kernel_size=1, strides=1,
                                 use_bias=False, layout=layout))
-        self.body.add(nn.BatchNorm(axis=ax))
+        # junction BN: relu(bn(conv_out) + residual) fused in one kernel
+        self.bn_join = nn.BatchNorm(axis=ax)
         if downsample:
             self.downsample = nn.HybridSequential(prefix="")
             self.downsample.add(nn.Conv2D(channels, kernel_size=1,
@@ -93,8 +94,7 @@ class BottleneckV1(HybridBlock):
         x_out = self.body(x)
         if self.downsample is not None:
             residual = self.downsample(residual)
-        from .... import ndarray as F
-        return F.Activation(x_out + residual, act_type="relu")
+        return self.bn_join(x_out, residual)
 
     hybrid_forward = None
 

@@ -173,9 +173,12 @@ class BatchNorm(HybridBlock):
             dtype = "float32"
         super().cast(dtype)
 
-    def hybrid_forward(self, F, x, gamma, beta, running_mean, running_var):
+    def hybrid_forward(self, F, x, addend=None, gamma=None, beta=None,
+                       running_mean=None, running_var=None):
+        # optional second data arg: residual-junction form
+        # relu(bn(x) + addend) fused in the kernel epilogue
         return F.BatchNorm(x, gamma, beta, running_mean, running_var,
-                           **self._kwargs)
+                           addend=addend, **self._kwargs)
 
 
 class LayerNorm(HybridBlock):

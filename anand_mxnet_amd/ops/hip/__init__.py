@@ -187,26 +187,16 @@ class _Conv2dIm2col(torch.autograd.Function):
         ph, pw = pad
         P = (H + 2 * ph - (R - 1) * dil[0] - 1) // sh + 1
         Q = (W_ + 2 * pw - (S - 1) * dil[1] - 1) // sw + 1
-        xp = torch.nn.functional.pad(x, (0, 0, pw, pw, ph, ph))
-        dev = x.device
-        pi = torch.arange(P, device=dev) * sh
-        qi = torch.arange(Q, device=dev) * sw
-        ri = torch.arange(R, device=dev) * dil[0]
-        si = torch.arange(S, device=dev) * dil[1]
-        IH = (pi[:, None, None, None] + ri[None, None, :, None])
-        IW = (qi[None, :, None, None] + si[None, None, None, :])
-        IH = IH.expand(P, Q, R, S)
-        IW = IW.expand(P, Q, R, S)
-        cols = xp[:, IH, IW, :]                       # [N,P,Q,R,S,C]
-        A = cols.reshape(N * P * Q, R * S * C)
         kdim = R * S * C
-        kpad = (64 - kdim % 64) % 64
-        if kpad:
-            A = torch.nn.functional.pad(A, (0, kpad))
+        kpadded = ((kdim + 63) // 64) * 64
+        # dedicated gather kernel: writes the padded GEMM operand directly
+        # (replaces torch pad + advanced-indexing + pad + contiguous chain)
+        A = _C.im2col_pad(x.contiguous(), [R, S], list(stride), list(pad),
+                          list(dil), kpadded)
+        kpad = kpadded - kdim
         B = w.reshape(K, kdim)
         if kpad:
             B = torch.nn.functional.pad(B, (0, kpad))
-        A = A.contiguous()
         ctx.save_for_backward(A, w)
         ctx.dims = (N, P, Q, K, kdim)
         ctx.has_bias = bias is not None
@@ -250,46 +240,64 @@ def conv2d_nhwc(x, w, bias, stride, pad, dil, groups):
 
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, rmean, rvar, eps, momentum, training,
-                fuse_relu):
+    def forward(ctx, x, resid, gamma, beta, rmean, rvar, eps, momentum,
+                training, fuse_relu):
         x = x.contiguous()
         g32 = gamma.float().contiguous()
         b32 = beta.float().contiguous()
+        fuse_add = resid is not None
         if training:
             y, save_mean, save_rstd = _C.bn_fwd_train(
-                x, g32, b32, rmean, rvar, eps, momentum, fuse_relu)
-            ctx.save_for_backward(x, g32, b32, save_mean, save_rstd)
+                x, g32, b32, rmean, rvar, eps, momentum, fuse_relu,
+                resid.contiguous() if fuse_add else None)
+            if fuse_add:
+                # junction form relu(bn(x)+r): mask comes from y > 0
+                ctx.save_for_backward(x, g32, b32, save_mean, save_rstd, y)
+            else:
+                ctx.save_for_backward(x, g32, b32, save_mean, save_rstd)
             ctx.relu = fuse_relu
+            ctx.fuse_add = fuse_add
             ctx.gdtype = gamma.dtype
             return y
         ctx.inference = True
         return _C.bn_fwd_infer(x, g32, b32, rmean.contiguous(),
-                               rvar.contiguous(), eps, fuse_relu)
+                               rvar.contiguous(), eps, fuse_relu,
+                               resid.contiguous() if fuse_add else None)
 
     @staticmethod
     def backward(ctx, dy):
         if getattr(ctx, "inference", False):
             raise RuntimeError("BatchNorm inference backward not supported")
+        if ctx.fuse_add:
+            x, g32, b32, save_mean, save_rstd, y = ctx.saved_tensors
+            dr = _C.act_bwd(0, y, dy.contiguous())  # relu mask from y
+            dx, dgamma, dbeta = _C.bn_bwd(x, dr, g32, b32,
+                                          save_mean, save_rstd, False)
+            return (dx, dr, dgamma.to(ctx.gdtype), dbeta.to(ctx.gdtype),
+                    None, None, None, None, None, None)
         x, g32, b32, save_mean, save_rstd = ctx.saved_tensors
         dx, dgamma, dbeta = _C.bn_bwd(x, dy.contiguous(), g32, b32,
                                       save_mean, save_rstd, ctx.relu)
-        return (dx, dgamma.to(ctx.gdtype), dbeta.to(ctx.gdtype), None, None,
-                None, None, None, None)
+        return (dx, None, dgamma.to(ctx.gdtype), dbeta.to(ctx.gdtype), None,
+                None, None, None, None, None)
 
 
 @register_gpu("batchnorm_nhwc")
 def batchnorm_nhwc(x, gamma, beta, rmean, rvar, eps, momentum, training,
-                   fuse_relu):
+                   fuse_relu, addend=None):
     if x.dtype != torch.bfloat16 or x.shape[-1] % 8 != 0:
         # torch path for non-native dtype/shape
         out = torch.nn.functional.batch_norm(
             x.permute(0, 3, 1, 2), rmean, rvar, gamma, beta, training,
             1.0 - momentum if training else 0.0, eps)
-        if fuse_relu:
+        if addend is not None:
+            out = out + addend.permute(0, 3, 1, 2)
+            out = torch.relu(out)
+        elif fuse_relu:
             out = torch.relu(out)
         return out.permute(0, 2, 3, 1).contiguous()
-    return _BatchNormNHWC.apply(x, gamma, beta, rmean, rvar, eps, momentum,
-                                training, fuse_relu)
+    return _BatchNormNHWC.apply(x, addend, gamma, beta, rmean, rvar, eps,
+                                momentum, training, fuse_relu)
 
 
 # ---------------------------------------------------------------------------

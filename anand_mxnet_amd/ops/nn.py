@@ -169,10 +169,12 @@ def pooling(data, kernel=(2, 2), pool_type="max", global_pool=False,
 def batch_norm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
                momentum=0.9, fix_gamma=False, use_global_stats=False,
                output_mean_var=False, axis=1, cudnn_off=False,
-               fuse_relu=False):
+               fuse_relu=False, addend=None):
     """BatchNorm; training mode from autograd.is_training() (reference
     semantics).  MXNet momentum: moving = m*moving + (1-m)*batch.
-    GPU path: fused single-pass Welford NHWC kernel (axis=-1/3)."""
+    GPU path: fused single-pass Welford NHWC kernel (axis=-1/3).
+    addend: residual-junction form relu(bn(data) + addend) fused into the
+    apply kernel (ResNet block outputs)."""
     training = _ag.is_training() and not use_global_stats
     g = torch.ones_like(gamma) if fix_gamma else gamma
     ch_axis = axis % data.dim()
@@ -180,7 +182,7 @@ def batch_norm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
             _use_hip(data, "batchnorm_nhwc"):
         return gpu_impl("batchnorm_nhwc")(data, g, beta, moving_mean,
                                           moving_var, eps, momentum, training,
-                                          fuse_relu)
+                                          fuse_relu, addend)
     if data.is_cuda:
         _use_hip(data, "batchnorm_nhwc")
     # torch F.batch_norm expects channel at dim1
@@ -192,7 +194,10 @@ def batch_norm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
         x = data.permute(perm).contiguous()
     out = F.batch_norm(x, moving_mean, moving_var, g, beta, training,
                        1.0 - momentum, eps)
-    if fuse_relu:
+    if addend is not None:
+        a = addend.permute(perm).contiguous() if perm is not None else addend
+        out = torch.relu(out + a)
+    elif fuse_relu:
         out = torch.relu(out)
     if perm is not None:
         inv = [0] * len(perm)

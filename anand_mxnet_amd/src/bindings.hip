@@ -36,10 +36,10 @@ void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, int, int, int, int,
 // bn_pool.hip
 void bn_fwd_train(const bf16*, bf16*, float*, float*, float*, float*, float*,
                   float*, const float*, const float*, long, int, float,
-                  float, bool, hipStream_t);
+                  float, bool, const bf16*, hipStream_t);
 void bn_fwd_infer(const bf16*, bf16*, const float*, const float*,
                   const float*, const float*, long, int, float, bool,
-                  hipStream_t);
+                  const bf16*, hipStream_t);
 void bn_bwd(const bf16*, const bf16*, bf16*, const float*, const float*,
             const float*, const float*, float*, float*, long, int, bool,
             hipStream_t);
@@ -76,6 +76,9 @@ void embedding_fwd_bf16(const long*, const bf16*, bf16*, long, int, long,
 void embedding_bwd_bf16(const long*, const bf16*, float*, long, int, long,
                         hipStream_t);
 void colsum_bf16(const bf16*, float*, long, int, hipStream_t);
+void im2col_stem_bf16(const bf16*, bf16*, int, int, int, int, int, int,
+                      int, int, int, int, int, int, int, int, int, int,
+                      hipStream_t);
 // gemm_nn.hip / attention entry points (gemm_tn.hip, gemm_wg.hip)
 void gemm_nn_bf16(const bf16*, ZMap, long, int, const bf16*, ZMap, long,
                   int, bf16*, ZMap, long, int, int, const bf16*,
@@ -320,7 +323,8 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                                      const at::Tensor& beta,
                                      at::Tensor running_mean,
                                      at::Tensor running_var, double eps,
-                                     double momentum, bool relu) {
+                                     double momentum, bool relu,
+                                     const c10::optional<at::Tensor>& resid) {
   int C = x.size(-1);
   long M = x.numel() / C;
   auto y = at::empty_like(x);
@@ -334,19 +338,23 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                     running_mean.data_ptr<float>(),
                     running_var.data_ptr<float>(), gamma.data_ptr<float>(),
                     beta.data_ptr<float>(), M, C, (float)eps,
-                    (float)momentum, relu, cur_stream());
+                    (float)momentum, relu,
+                    resid.has_value() ? bfp(*resid) : nullptr,
+                    cur_stream());
   return {y, save_mean, save_rstd};
 }
 
 at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
                         const at::Tensor& beta, const at::Tensor& rmean,
-                        const at::Tensor& rvar, double eps, bool relu) {
+                        const at::Tensor& rvar, double eps, bool relu,
+                        const c10::optional<at::Tensor>& resid) {
   int C = x.size(-1);
   long M = x.numel() / C;
   auto y = at::empty_like(x);
   mxa::bn_fwd_infer(bfp(x), bfp_mut(y), rmean.data_ptr<float>(),
                     rvar.data_ptr<float>(), gamma.data_ptr<float>(),
                     beta.data_ptr<float>(), M, C, (float)eps, relu,
+                    resid.has_value() ? bfp(*resid) : nullptr,
                     cur_stream());
   return y;
 }
@@ -561,6 +569,25 @@ at::Tensor attn_wg(const at::Tensor& p, const at::Tensor& q, long H) {
   return out;
 }
 
+at::Tensor im2col_pad(const at::Tensor& x, std::vector<long> rs,
+                      std::vector<long> stride, std::vector<long> pad,
+                      std::vector<long> dil, long kpadded) {
+  // stem conv operand: [N*P*Q, kpadded] padded im2col (k order (r,s,c))
+  TORCH_CHECK(x.dim() == 4 && x.is_contiguous());
+  int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int R = rs[0], S = rs[1], sh = stride[0], sw = stride[1];
+  int ph = pad[0], pw = pad[1], dh = dil[0], dw = dil[1];
+  int P = (H + 2 * ph - (R - 1) * dh - 1) / sh + 1;
+  int Q = (W + 2 * pw - (S - 1) * dw - 1) / sw + 1;
+  int kdim = R * S * C;
+  TORCH_CHECK(kpadded % 64 == 0 && kpadded >= kdim);
+  auto out = at::empty({(long)Nb * P * Q, kpadded}, x.options());
+  mxa::im2col_stem_bf16(bfp(x), bfp_mut(out), Nb, H, W, C, P, Q, R, S, sh,
+                        sw, ph, pw, dh, dw, kdim, (int)kpadded,
+                        cur_stream());
+  return out;
+}
+
 at::Tensor colsum(const at::Tensor& x) {
   // bias gradient: sum over all leading dims of a [..., N] bf16 tensor,
   // f32 result (replaces .float().sum(0) copy+reduce pair).
@@ -740,8 +767,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
-  m.def("bn_fwd_train", &bn_fwd_train);
-  m.def("bn_fwd_infer", &bn_fwd_infer);
+  m.def("bn_fwd_train", &bn_fwd_train, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("running_mean"), py::arg("running_var"),
+        py::arg("eps"), py::arg("momentum"), py::arg("relu"),
+        py::arg("resid") = py::none());
+  m.def("bn_fwd_infer", &bn_fwd_infer, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("rmean"), py::arg("rvar"),
+        py::arg("eps"), py::arg("relu"), py::arg("resid") = py::none());
   m.def("bn_bwd", &bn_bwd);
   m.def("maxpool", &maxpool);
   m.def("maxpool_bwd", &maxpool_bwd);
@@ -757,6 +789,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("colsum", &colsum);
+  m.def("im2col_pad", &im2col_pad);
   m.def("gemm_nn", &gemm_nn);
   m.def("attn_scores", &attn_scores);
   m.def("attn_ctx", &attn_ctx);

@@ -105,6 +105,57 @@ __global__ __launch_bounds__(256) void bn_apply(
   }
 }
 
+// junction form: y = relu(bn(x) + residual) — fuses the residual add
+// and the block-output ReLU into the BN epilogue (one HBM round trip
+// instead of three; backward reconstructs the mask from y > 0).
+__global__ __launch_bounds__(256) void bn_apply_add(
+    const bf16* __restrict__ x, const bf16* __restrict__ r,
+    bf16* __restrict__ y, const float* __restrict__ mean,
+    const float* __restrict__ rstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, long M, int C) {
+  long nvec = M * (C / 8);
+  int cv = C / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    bf16x8 v = *(const bf16x8*)&x[i * 8];
+    bf16x8 rv = *(const bf16x8*)&r[i * 8];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float f = (bf2f(v[j]) - mean[c]) * rstd[c] * gamma[c] + beta[c] +
+                bf2f(rv[j]);
+      o[j] = f2bf(f > 0 ? f : 0);
+    }
+    *(bf16x8*)&y[i * 8] = o;
+  }
+}
+
+__global__ __launch_bounds__(256) void bn_infer_add(
+    const bf16* __restrict__ x, const bf16* __restrict__ r,
+    bf16* __restrict__ y, const float* __restrict__ rmean,
+    const float* __restrict__ rvar, const float* __restrict__ gamma,
+    const float* __restrict__ beta, long M, int C, float eps) {
+  long nvec = M * (C / 8);
+  int cv = C / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    bf16x8 v = *(const bf16x8*)&x[i * 8];
+    bf16x8 rv = *(const bf16x8*)&r[i * 8];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float f = (bf2f(v[j]) - rmean[c]) * rsqrtf(rvar[c] + eps) * gamma[c] +
+                beta[c] + bf2f(rv[j]);
+      o[j] = f2bf(f > 0 ? f : 0);
+    }
+    *(bf16x8*)&y[i * 8] = o;
+  }
+}
+
 // inference: normalize with running stats directly
 template <bool RELU>
 __global__ __launch_bounds__(256) void bn_infer(
@@ -269,16 +320,21 @@ __global__ __launch_bounds__(256) void maxpool_bwd(
     const bf16* __restrict__ dy, const unsigned char* __restrict__ arg,
     bf16* __restrict__ dx, int Nb, int H, int W, int C, int P, int Q,
     int R, int S, int sh, int sw, int ph, int pw) {
-  long total = (long)Nb * H * W * C;
+  // gather form, 8 channels per thread (no atomics): for each input
+  // pixel, scan the few output windows covering it and claim the grads
+  // whose argmax points here.
+  long total = (long)Nb * H * W * (C / 8);
+  int cv = C / 8;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    long nhw = i / C;
+    int c0 = (int)(i % cv) * 8;
+    long nhw = i / cv;
     int w_ = (int)(nhw % W);
     int h = (int)((nhw / W) % H);
     int n = (int)(nhw / ((long)H * W));
-    float acc = 0;
-    // output windows covering (h, w): p in [ceil((h+ph-R+1)/sh), (h+ph)/sh]
+    float acc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = 0.f;
     int p_lo = (h + ph - R + sh) >= 0 ? (h + ph - R + sh) / sh : 0;
     int p_hi = min((h + ph) / sh, P - 1);
     int q_lo = (w_ + pw - S + sw) >= 0 ? (w_ + pw - S + sw) / sw : 0;
@@ -289,11 +345,58 @@ __global__ __launch_bounds__(256) void maxpool_bwd(
       for (int q = q_lo; q <= q_hi; ++q) {
         int sx = w_ + pw - q * sw;
         if (sx < 0 || sx >= S) continue;
-        long oi = (((long)n * P + p) * Q + q) * C + c;
-        if (arg[oi] == r * S + sx) acc += bf2f(dy[oi]);
+        long oi = (((long)n * P + p) * Q + q) * C + c0;
+        unsigned char want = (unsigned char)(r * S + sx);
+        u64 a8 = *(const u64*)&arg[oi];
+        bf16x8 g8 = *(const bf16x8*)&dy[oi];
+        const unsigned char* ab = (const unsigned char*)&a8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (ab[j] == want) acc[j] += bf2f(g8[j]);
       }
     }
-    dx[i] = f2bf(acc);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[j]);
+    *(bf16x8*)&dx[(((long)n * H + h) * W + w_) * C + c0] = o;
+  }
+}
+
+// Materialized im2col for the small-C stem conv (7x7 C=3): writes the
+// padded [N*P*Q, kdim_pad] GEMM operand directly (k order (r, s, c),
+// zeros in the %64 pad tail and outside the image) — replaces a chain of
+// torch pad + advanced-indexing + pad + contiguous full-tensor copies.
+__global__ __launch_bounds__(256) void im2col_stem(
+    const bf16* __restrict__ x, bf16* __restrict__ out, int Nb, int H,
+    int W, int C, int P, int Q, int R, int S, int sh, int sw, int ph,
+    int pw, int dh, int dw, int kdim, int kpadded) {
+  int kv = kpadded / 8;
+  long total = (long)Nb * P * Q * kv;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int j0 = (int)(i % kv) * 8;
+    long npq = i / kv;
+    int q = (int)(npq % Q);
+    int p = (int)((npq / Q) % P);
+    int n = (int)(npq / ((long)P * Q));
+    int ih0 = p * sh - ph, iw0 = q * sw - pw;
+    const bf16* xn = x + (long)n * H * W * C;
+    bf16x8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int k = j0 + e;
+      float v = 0.f;
+      if (k < kdim) {
+        int c = k % C;
+        int rs = k / C;
+        int sx = rs % S, r = rs / S;
+        int ih = ih0 + r * dh, iw = iw0 + sx * dw;
+        if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+          v = bf2f(xn[((long)ih * W + iw) * C + c]);
+      }
+      o[e] = f2bf(v);
+    }
+    *(bf16x8*)&out[npq * kpadded + j0] = o;
   }
 }
 
@@ -389,7 +492,7 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
                   float* save_mean, float* save_rstd, float* running_mean,
                   float* running_var, const float* gamma, const float* beta,
                   long M, int C, float eps, float momentum, bool relu,
-                  hipStream_t s) {
+                  const bf16* resid, hipStream_t s) {
   // psum/psumsq are one contiguous [2,C] buffer: single fill
   HIP_CHECK(hipMemsetAsync(psum, 0, 2L * C * sizeof(float), s));
   int cb = cdiv_h(C, 64);
@@ -400,7 +503,10 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
       psum, psumsq, save_mean, save_rstd, running_mean, running_var, M, C,
       gy, eps, momentum, 1);
   dim3 g2 = ew_grid(M * (C / 8));
-  if (relu)
+  if (resid)
+   hipLaunchKernelGGL(( bn_apply_add), dim3(g2), dim3(256), 0, s, x, resid, y, save_mean, save_rstd,
+                                    gamma, beta, M, C);
+  else if (relu)
    hipLaunchKernelGGL(( bn_apply<true>), dim3(g2), dim3(256), 0, s, x, y, save_mean, save_rstd, gamma,
                                       beta, M, C);
   else
@@ -410,9 +516,13 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
 
 void bn_fwd_infer(const bf16* x, bf16* y, const float* rmean,
                   const float* rvar, const float* gamma, const float* beta,
-                  long M, int C, float eps, bool relu, hipStream_t s) {
+                  long M, int C, float eps, bool relu, const bf16* resid,
+                  hipStream_t s) {
   dim3 g = ew_grid(M * (C / 8));
-  if (relu)
+  if (resid)
+   hipLaunchKernelGGL(( bn_infer_add), dim3(g), dim3(256), 0, s, x, resid, y, rmean, rvar, gamma, beta,
+                                   M, C, eps);
+  else if (relu)
    hipLaunchKernelGGL(( bn_infer<true>), dim3(g), dim3(256), 0, s, x, y, rmean, rvar, gamma, beta, M, C,
                                      eps);
   else
@@ -454,10 +564,19 @@ void maxpool_nhwc(const bf16* x, bf16* y, unsigned char* arg, int Nb, int H,
                                 ph, pw);
 }
 
+void im2col_stem_bf16(const bf16* x, bf16* out, int Nb, int H, int W,
+                      int C, int P, int Q, int R, int S, int sh, int sw,
+                      int ph, int pw, int dh, int dw, int kdim, int kpadded,
+                      hipStream_t s) {
+  dim3 g = ew_grid((long)Nb * P * Q * (kpadded / 8), 256, 8192);
+ hipLaunchKernelGGL(( im2col_stem), dim3(g), dim3(256), 0, s, x, out, Nb, H, W, C, P, Q, R, S, sh, sw,
+                                ph, pw, dh, dw, kdim, kpadded);
+}
+
 void maxpool_nhwc_bwd(const bf16* dy, const unsigned char* arg, bf16* dx,
                       int Nb, int H, int W, int C, int P, int Q, int R,
                       int S, int sh, int sw, int ph, int pw, hipStream_t s) {
-  dim3 g = ew_grid((long)Nb * H * W * C);
+  dim3 g = ew_grid((long)Nb * H * W * (C / 8), 256, 8192);
  hipLaunchKernelGGL(( maxpool_bwd), dim3(g), dim3(256), 0, s, dy, arg, dx, Nb, H, W, C, P, Q, R, S, sh,
                                 sw, ph, pw);
 }

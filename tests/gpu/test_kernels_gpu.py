@@ -409,3 +409,54 @@ def test_mha_fwd_bwd_vs_torch():
                      msg=f"mha dk c={causal}")
         assert_close(v1.grad, v2.grad, rtol=6e-2, atol=6e-2,
                      msg=f"mha dv c={causal}")
+
+
+def test_im2col_pad_stem():
+    N, H, W, C, R, S = 2, 20, 20, 3, 7, 7
+    sh = sw = 2
+    ph = pw = 3
+    x = _rt((N, H, W, C)).to(BF)
+    kdim = R * S * C
+    kpadded = ((kdim + 63) // 64) * 64
+    out = _C.im2col_pad(x.contiguous(), [R, S], [sh, sw], [ph, pw], [1, 1],
+                        kpadded)
+    # torch reference via unfold on NCHW
+    xn = x.float().permute(0, 3, 1, 2)
+    cols = torch.nn.functional.unfold(xn, (R, S), padding=(ph, pw),
+                                      stride=(sh, sw))  # [N, C*R*S, P*Q]
+    P = (H + 2 * ph - R) // sh + 1
+    # unfold k order is (c, r, s); ours is (r, s, c)
+    ref = cols.reshape(N, C, R * S, -1).permute(0, 3, 2, 1) \
+        .reshape(-1, R * S * C)
+    assert_close(out[:, :kdim], ref, msg="im2col_pad")
+    assert out[:, kdim:].abs().max().item() == 0.0
+
+
+def test_bn_add_relu_fused():
+    from anand_mxnet_amd.ops.hip import batchnorm_nhwc
+    N, H, W, C = 4, 8, 8, 64
+    x0 = _rt((N, H, W, C)).to(BF)
+    r0 = _rt((N, H, W, C)).to(BF)
+    gamma = torch.rand(C, device=_dev()) + 0.5
+    beta = torch.rand(C, device=_dev()) - 0.5
+    rm = torch.zeros(C, device=_dev())
+    rv = torch.ones(C, device=_dev())
+
+    x1 = x0.clone().requires_grad_(True)
+    r1 = r0.clone().requires_grad_(True)
+    y = batchnorm_nhwc(x1, gamma, beta, rm.clone(), rv.clone(), 1e-5, 0.9,
+                       True, False, r1)
+    g = _rt((N, H, W, C)).to(BF)
+    y.backward(g)
+
+    x2 = x0.clone().float().requires_grad_(True)
+    r2 = r0.clone().float().requires_grad_(True)
+    xn = x2.permute(0, 3, 1, 2)
+    yn = torch.nn.functional.batch_norm(xn, rm.clone(), rv.clone(), gamma,
+                                        beta, True, 0.1, 1e-5)
+    ref = torch.relu(yn.permute(0, 2, 3, 1) + r2)
+    ref.backward(g.float())
+
+    assert_close(y, ref, msg="bn_add_relu fwd")
+    assert_close(x1.grad, x2.grad, rtol=5e-2, atol=5e-2, msg="bn_add dx")
+    assert_close(r1.grad, r2.grad, rtol=5e-2, atol=5e-2, msg="bn_add dr")
