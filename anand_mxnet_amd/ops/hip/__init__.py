@@ -606,14 +606,14 @@ class _MhaCore(torch.autograd.Function):
         dpd = _C.attn_scores(dout, v, H)             # dP (post-dropout)
         if mask is not None:
             pd = _C.dropout_bwd(p, mask, dropout_p)  # recompute dropped P
-            dv = _C.attn_wg(pd, dout, H).to(v.dtype)
+            dv = _C.attn_wg(pd, dout, H)
             dp = _C.dropout_bwd(dpd, mask, dropout_p)
         else:
-            dv = _C.attn_wg(p, dout, H).to(v.dtype)
+            dv = _C.attn_wg(p, dout, H)
             dp = dpd
         ds = _C.softmax_bwd(p, dp, False, scale)     # [BH, Tq, Tk]
         dq = _C.attn_ctx(ds, k, H)
-        dk = _C.attn_wg(ds, q, H).to(k.dtype)
+        dk = _C.attn_wg(ds, q, H)
         return dq, dk, dv, None, None, None, None, None
 
 

@@ -85,7 +85,7 @@ void gemm_nn_bf16(const bf16*, ZMap, long, int, const bf16*, ZMap, long,
                   hipStream_t);
 void attn_scores_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
                       int, const bf16*, hipStream_t);
-void attn_wg_bf16(const bf16*, const bf16*, float*, int, int, int, int,
+void attn_wg_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
                   int, const bf16*, hipStream_t);
 void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
 void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
@@ -563,8 +563,8 @@ at::Tensor attn_wg(const at::Tensor& p, const at::Tensor& q, long H) {
   long B = BH / H, HD = q.size(2), D = HD / H;
   TORCH_CHECK(B * H == BH && D * H == HD && Tk % 8 == 0 && D % 8 == 0,
               "attn_wg: Tk%8, D%8 required");
-  auto out = at::empty({B, Tk, HD}, q.options().dtype(at::kFloat));
-  mxa::attn_wg_bf16(bfp(p), bfp(q), out.data_ptr<float>(), (int)B, (int)H,
+  auto out = at::empty({B, Tk, HD}, q.options());
+  mxa::attn_wg_bf16(bfp(p), bfp(q), bfp_mut(out), (int)B, (int)H,
                     (int)Tq, (int)Tk, (int)D, zp(p), cur_stream());
   return out;
 }

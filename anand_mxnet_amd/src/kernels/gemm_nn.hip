@@ -164,11 +164,155 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn_core(
   }
 }
 
+// 64-wide-N variant: attention ctx/dQ have N = D = 64 — the 128-wide
+// tile would run half-empty.  B image is [64 k][64 cols] (128-B rows),
+// each wave computes a 64x32 quadrant (4 m-frags x 2 n-frags).
+DEVINL int nn_b64_byte(int row, int cbyte) {
+  return row * 128 + ((((cbyte >> 4) ^ ((row & 3) << 1)) << 4) | (cbyte & 15));
+}
+
+__global__ __launch_bounds__(NTHREADS, 2) void gemm_nn64_core(
+    const bf16* __restrict__ A, ZMap za, long lda, int M,
+    const bf16* __restrict__ B, ZMap zb, long ldb, int N,
+    bf16* __restrict__ Cout, ZMap zc, long ldc,
+    int ksteps, const bf16* __restrict__ zpage) {
+  __shared__ __align__(16) bf16 smemA[2][NBM * NBK];
+  __shared__ __align__(16) bf16 smemB[2][NBK * 64];
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = bid * NBM;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  const bf16* abase = A + za.off(z);
+  const bf16* arp[4];
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int r = bm + nn_a_row(p, t);
+    arp[p] = (r < M) ? abase + (long)r * lda : nullptr;
+  }
+  const int acg = (ll & 7) ^ (nn_a_row(0, t) & 7);
+
+  // B staging: wave w covers k-rows [w*16, w*16+16); each 1-KiB glds
+  // issue is 8 rows of 128 B; lane l -> row seg*8 + (l>>3), chunk l&7,
+  // source chunk XOR'd with ((row&3)<<1).
+  const bf16* bbase = B + zb.off(z);
+  const int bcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);
+  int bcol = bcg * 8;
+  const bool bcolok = bcol + 8 <= N;
+  const bf16* brp[2];
+#pragma unroll
+  for (int p = 0; p < 2; ++p)
+    brp[p] = bbase + (long)(lw * 16 + p * 8 + (ll >> 3)) * ldb + bcol;
+
+  auto stage_glds = [&](int buf, int ks) {
+    long koff = (long)ks * NBK;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      unsigned lbase =
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 4 + p) * 1024));
+      auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemA[buf][0] + lbase);
+      auto gA = (const __attribute__((address_space(1))) unsigned int*)(
+          arp[p] ? arp[p] + koff + acg * 8 : zpage + acg * 8);
+      __builtin_amdgcn_global_load_lds(gA, ldsA, 16, 0, 0);
+    }
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      unsigned lbase =
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 2 + p) * 1024));
+      auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemB[buf][0] + lbase);
+      auto gB = (const __attribute__((address_space(1))) unsigned int*)(
+          bcolok ? brp[p] + koff * ldb : zpage + bcg * 8);
+      __builtin_amdgcn_global_load_lds(gB, ldsB, 16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int wm = (lw >> 1) * 64, wn = (lw & 1) * 32;
+  const int lrow = ll & 15, lk = ll >> 4;
+  const int ig = ll & 15, khi = ll >> 4;
+
+  auto trread = [&](const bf16* img, int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)img +
+        nn_b64_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bfr[2];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        int row = wm + mf * 16 + lrow;
+        af[mf] = *(const bf16x8*)&smemA[buf][nn_a_off(row, chunk)];
+      }
+      int rbase = kk * 32 + khi * 8 + (ig >> 2);
+      int cb = (ig & 3) * 8;
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int cbyte = (wn + nf * 16) * 2 + cb;
+        bf16x4 lo = trread(smemB[buf], rbase, cbyte);
+        bf16x4 hi = trread(smemB[buf], rbase + 4, cbyte);
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+    }
+  };
+
+  stage_glds(0, 0);
+  __syncthreads();
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    if (ks + 1 < ksteps) stage_glds(buf ^ 1, ks + 1);
+    compute(buf);
+    __syncthreads();
+  }
+
+  bf16* cz = Cout + zc.off(z);
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      int gcol = wn + nf * 16 + (ll & 15);
+      if (gcol >= N) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (ll >> 4) * 4 + i;
+        if (grow >= M) continue;
+        cz[(long)grow * ldc + gcol] = f2bf(acc[mf][nf][i]);
+      }
+    }
+  }
+}
+
 void gemm_nn_bf16(const bf16* A, ZMap za, long lda, int M,
                   const bf16* B, ZMap zb, long ldb, int N,
                   bf16* C, ZMap zc, long ldc, int K, int nz,
                   const bf16* zpage, hipStream_t stream) {
   int ksteps = K / NBK;
+  if (N <= 64) {
+    dim3 grid((unsigned)cdiv_h(M, NBM), 1, nz);
+    gemm_nn64_core<<<grid, NTHREADS, 0, stream>>>(A, za, lda, M, B, zb,
+                                                  ldb, N, C, zc, ldc,
+                                                  ksteps, zpage);
+    return;
+  }
   long bxy = (long)cdiv_h(M, NBM) * cdiv_h(N, NBN);
   dim3 grid((unsigned)bxy, 1, nz);
   gemm_nn_core<<<grid, NTHREADS, 0, stream>>>(A, za, lda, M, B, zb, ldb, N,

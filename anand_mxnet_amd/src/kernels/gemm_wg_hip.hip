@@ -362,7 +362,7 @@ struct TrIm2colRows {
   }
 };
 
-template <bool A_IM2COL>
+template <bool A_IM2COL, bool OUTBF16 = false>
 __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
     TrPlainRows aplain, TrIm2colRows aim, TrPlainRows bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
@@ -500,7 +500,9 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
         int grow = bm + wm + mf * 16 + (ll >> 4) * 4 + i;
         if (grow >= Mo) continue;
         float v = acc[mf][nf][i];
-        if (use_atomic)
+        if (OUTBF16)
+          ((bf16*)cseg)[(long)grow * ldc + gcol] = f2bf(v);
+        else if (use_atomic)
           atomicAdd(&cseg[(long)grow * ldc + gcol], v);
         else
           cseg[(long)grow * ldc + gcol] = v;
@@ -591,7 +593,7 @@ void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
 //   dV with dS := P, Q := dO).  A is the contiguous [BH, Tq, X] probs /
 //   score-grads tensor; B and the f32 output live in [B, S, H*D] layout
 //   read/written in place.
-void attn_wg_bf16(const bf16* A, const bf16* Bq, float* C, int B, int H,
+void attn_wg_bf16(const bf16* A, const bf16* Bq, bf16* C, int B, int H,
                   int Tq, int Tk, int D, const bf16* zpage,
                   hipStream_t stream) {
   long hd = (long)H * D;
@@ -604,8 +606,10 @@ void attn_wg_bf16(const bf16* A, const bf16* Bq, float* C, int B, int H,
   TrPlainRows b{Bq, zpage, Tq, hd, D};
   TrIm2colRows dummy{};
   dim3 grid((unsigned)bxy, 1, (unsigned)(B * H));
- hipLaunchKernelGGL(( gemm_wg_tr_core<false>), dim3(grid), dim3(WTHREADS), 0, stream, 
-      a, dummy, b, C, Tk, D, hd, 0, 1, 1, 1, ksteps, 1, 0,
+  // single reduce slice per z -> no atomics -> bf16 output directly
+  // (saves a full f32->bf16 cast pass over [B, S, H*D] per call)
+ hipLaunchKernelGGL(( gemm_wg_tr_core<false, true>), dim3(grid), dim3(WTHREADS), 0, stream, 
+      a, dummy, b, (float*)C, Tk, D, hd, 0, 1, 1, 1, ksteps, 1, 0,
       ZMap{1, (long)Tq * Tk, 0}, ZMap{H, Tq * hd, D}, ZMap{H, Tk * hd, D});
 }
 
