@@ -376,7 +376,8 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
   else
     aplain.base += za.off(zb_);
   bload.base += zb.off(zb_);
-  Cout += zc.off(zb_);
+  // zc.off() counts OUTPUT ELEMENTS: 2 bytes in bf16 mode, 4 in f32
+  Cout = (float*)((char*)Cout + zc.off(zb_) * (OUTBF16 ? 2 : 4));
 
   const int seg = blockIdx.y / nsplit, split = blockIdx.y % nsplit;
   const int per = (ksteps_total + nsplit - 1) / nsplit;
