@@ -524,10 +524,22 @@ static bool wg_use_zip() {
 // Host launchers
 // ---------------------------------------------------------------------------
 
+static int wg_split_target() {
+  static const int t = [] {
+    const char* e = getenv("MXA_WG_TARGET");
+    return e ? atoi(e) : 1024;
+  }();
+  return t;
+}
+
 static int pick_split(long blocks_xy, long ksteps) {
-  // enough blocks to fill 256 CUs a few times over, bounded by the reduce
+  // enough blocks to fill 256 CUs a few times over, bounded by the
+  // reduce depth.  Each extra split multiplies f32 atomic-finish traffic
+  // (64 KiB per block), so the target trades fill against atomics
+  // (MXA_WG_TARGET, default 1024).
   int target = 1;
-  while (blocks_xy * target < 1024 && target < ksteps && target < 128)
+  int goal = wg_split_target();
+  while (blocks_xy * target < goal && target < ksteps && target < 128)
     target *= 2;
   return target;
 }
