@@ -532,14 +532,27 @@ static int wg_split_target() {
   return t;
 }
 
+static int wg_min_ksteps() {
+  static const int t = [] {
+    const char* e = getenv("MXA_WG_MINK");
+    return e ? atoi(e) : 24;
+  }();
+  return t;
+}
+
 static int pick_split(long blocks_xy, long ksteps) {
-  // enough blocks to fill 256 CUs a few times over, bounded by the
-  // reduce depth.  Each extra split multiplies f32 atomic-finish traffic
-  // (64 KiB per block), so the target trades fill against atomics
-  // (MXA_WG_TARGET, default 1024).
-  int target = 1;
+  // Fill the chip (~512 blocks of 4 waves at 2 waves/SIMD), but every
+  // split adds a 64-KiB f32 atomic finish per block — so each split must
+  // also amortize over enough reduce depth (>= MXA_WG_MINK k-steps,
+  // default 24).  Measured on MI355X: dropping the old fill-only target
+  // lifted small-Mo wgrads 107->190 TF (atomic-bound), while res2-3x3
+  // (1 tile) still needs ~128 splits for occupancy.
   int goal = wg_split_target();
-  while (blocks_xy * target < goal && target < ksteps && target < 128)
+  int mink = wg_min_ksteps();
+  long depth_cap = mink > 0 ? ksteps / mink : ksteps;
+  int target = 1;
+  while (blocks_xy * target < goal && target < ksteps && target < 128 &&
+         (blocks_xy * target < 256 || target < depth_cap))
     target *= 2;
   return target;
 }
