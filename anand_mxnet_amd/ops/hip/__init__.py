@@ -164,9 +164,21 @@ class _Conv2dNHWC(torch.autograd.Function):
         K, R, S, C = w.shape
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            w_rsck = w.permute(1, 2, 3, 0).contiguous()
-            dx = _C.conv2d_dgrad(dy, w_rsck, list(x.shape), list(stride),
-                                 list(pad), list(dil))
+            if R == 1 and S == 1 and (stride[0] > 1 or stride[1] > 1) \
+                    and pad == (0, 0):
+                # strided 1x1 (ResNet downsample): only every (sh, sw)-th
+                # input pixel has gradient — dense GEMM over the P*Q
+                # subgrid + strided scatter instead of a scatter-inverse
+                # gather GEMM where 3/4 of the predicate rows are zero.
+                N, P, Q, _ = dy.shape
+                dy2 = dy.reshape(N * P * Q, K)
+                sub = _C.gemm_nn(dy2, w.reshape(K, C))  # [NPQ, C], no W^T
+                dx = torch.zeros(x.shape, dtype=dy.dtype, device=dy.device)
+                dx[:, ::stride[0], ::stride[1], :] = sub.reshape(N, P, Q, C)
+            else:
+                w_rsck = w.permute(1, 2, 3, 0).contiguous()
+                dx = _C.conv2d_dgrad(dy, w_rsck, list(x.shape),
+                                     list(stride), list(pad), list(dil))
         if ctx.needs_input_grad[1]:
             dwf = _C.conv2d_wgrad(x, dy, [R, S], list(stride), list(pad),
                                   list(dil))  # [R,S,C,K] f32
