@@ -68,7 +68,9 @@ void layernorm_bwd_bf16(const bf16*, const bf16*, bf16*, const float*,
                         int, hipStream_t);
 // misc.hip
 void dropout_fwd_bf16(const bf16*, bf16*, unsigned char*, long,
-                      unsigned long long, float, hipStream_t);
+                      const unsigned long long*, unsigned long long, float,
+                      hipStream_t);
+void rng_bump_launch(unsigned long long*, hipStream_t);
 void dropout_bwd_bf16(const bf16*, const unsigned char*, bf16*, long, float,
                       hipStream_t);
 void embedding_fwd_bf16(const long*, const bf16*, bf16*, long, int, long,
@@ -599,16 +601,40 @@ at::Tensor colsum(const at::Tensor& x) {
   return out;
 }
 
+// persistent device RNG seed word per device (graph-safe dropout)
+at::Tensor& rng_seed_for(const at::Tensor& like) {
+  static std::unordered_map<int, at::Tensor> cache;
+  int dev = like.get_device();
+  auto it = cache.find(dev);
+  if (it == cache.end()) {
+    auto z = at::full({1}, (long)0x243F6A8885A308D3LL,
+                      like.options().dtype(at::kLong));
+    it = cache.emplace(dev, z).first;
+  }
+  return it->second;
+}
+
 std::vector<at::Tensor> dropout_fwd(const at::Tensor& x, double p,
                                     long seed) {
   long n = x.numel();
   TORCH_CHECK(n % 8 == 0, "dropout: numel must be %8");
   auto y = at::empty_like(x);
   auto mask = at::empty({n}, x.options().dtype(at::kByte));
+  auto& sw = rng_seed_for(x);
   mxa::dropout_fwd_bf16(bfp(x), bfp_mut(y), mask.data_ptr<unsigned char>(),
-                        n, (unsigned long long)seed, (float)p,
-                        cur_stream());
+                        n,
+                        (const unsigned long long*)sw.data_ptr<long>(),
+                        (unsigned long long)seed, (float)p, cur_stream());
   return {y, mask};
+}
+
+void rng_step(long device) {
+  auto opts = at::TensorOptions().device(at::kCUDA, (int)device)
+                  .dtype(at::kLong);
+  auto probe = at::empty({0}, opts);
+  auto& sw = rng_seed_for(probe);
+  mxa::rng_bump_launch((unsigned long long*)sw.data_ptr<long>(),
+                       cur_stream());
 }
 
 at::Tensor dropout_bwd(const at::Tensor& dy, const at::Tensor& mask,
@@ -795,6 +821,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_ctx", &attn_ctx);
   m.def("attn_wg", &attn_wg);
   m.def("dropout_fwd", &dropout_fwd);
+  m.def("rng_step", &rng_step);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);

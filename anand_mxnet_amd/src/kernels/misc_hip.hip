@@ -20,10 +20,14 @@ DEVINL u32 hash_u32(u64 seed, u64 idx) {
   return (u32)(z >> 32);
 }
 
+// seed = *seed_ptr + salt: the device seed word evolves via rng_bump
+// (graph-safe — a captured step replays the bump, so dropout masks differ
+// per replay), the per-call salt decorrelates call sites within a step.
 __global__ __launch_bounds__(256) void dropout_fwd(
     const bf16* __restrict__ x, bf16* __restrict__ y,
-    unsigned char* __restrict__ mask, long nvec, u64 seed, float p,
-    float scale) {
+    unsigned char* __restrict__ mask, long nvec,
+    const u64* __restrict__ seed_ptr, u64 salt, float p, float scale) {
+  u64 seed = *seed_ptr + salt;
   u32 thresh = (u32)(p * 4294967296.0f);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
@@ -191,11 +195,20 @@ __global__ __launch_bounds__(256) void scale_cast_f32_bf16(
 // host launchers
 // ---------------------------------------------------------------------------
 
+__global__ void rng_bump(u64* s) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *s += 0x9e3779b97f4a7c15ULL;
+}
+
 void dropout_fwd_bf16(const bf16* x, bf16* y, unsigned char* mask, long n,
-                      u64 seed, float p, hipStream_t s) {
+                      const u64* seed_ptr, u64 salt, float p,
+                      hipStream_t s) {
   long nvec = n / 8;
- hipLaunchKernelGGL(( dropout_fwd), dim3(ew_grid(nvec)), dim3(256), 0, s, x, y, mask, nvec, seed, p,
-                                            1.f / (1.f - p));
+ hipLaunchKernelGGL(( dropout_fwd), dim3(ew_grid(nvec)), dim3(256), 0, s, x, y, mask, nvec, seed_ptr,
+                                            salt, p, 1.f / (1.f - p));
+}
+
+void rng_bump_launch(u64* seed_ptr, hipStream_t s) {
+ hipLaunchKernelGGL(( rng_bump), dim3(1), dim3(1), 0, s, seed_ptr);
 }
 
 void dropout_bwd_bf16(const bf16* dy, const unsigned char* mask, bf16* dx,

@@ -29,6 +29,8 @@ def main():
     ap.add_argument("--dtype", default="bfloat16")
     ap.add_argument("--fp8", action="store_true",
                     help="transformer_big only: fp8 e4m3 MFMA projections")
+    ap.add_argument("--capture", action="store_true",
+                    help="hipGraph whole-step capture")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -42,6 +44,7 @@ def main():
 
     import anand_mxnet_amd as mx
     from anand_mxnet_amd import gluon, autograd, nd
+    from anand_mxnet_amd import _C
 
     ctx = mx.gpu(local_rank)
     torch.cuda.set_device(local_rank)
@@ -69,6 +72,7 @@ def main():
                              dtype="int64")
 
         def step():
+            _C.rng_step(local_rank)  # graph-safe dropout seed advance
             with autograd.record():
                 mlm, nsp = net(tok, tt)
                 l = loss_fn(mlm.reshape(-1, vocab),
@@ -96,6 +100,7 @@ def main():
                        dtype="int64")
 
         def step():
+            _C.rng_step(local_rank)  # graph-safe dropout seed advance
             with autograd.record():
                 out = net(src, tgt)
                 l = loss_fn(out.reshape(-1, vocab),
@@ -109,12 +114,28 @@ def main():
     for _ in range(args.warmup):
         loss = step()
     torch.cuda.synchronize()
+    graph = None
+    if args.capture:
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                loss = step()
+            g.replay()
+            torch.cuda.synchronize()
+            graph = g
+        except Exception as e:
+            print(f"# graph capture unavailable ({e}); eager",
+                  file=sys.stderr)
+    run_step = graph.replay if graph is not None else step
+    for _ in range(2):
+        run_step()
+    torch.cuda.synchronize()
     if distributed:
         dist.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        loss = step()
+        run_step()
     torch.cuda.synchronize()
     if distributed:
         dist.barrier()
