@@ -1,2 +1,3 @@
 from . import amp
 from . import quantization
+from . import text

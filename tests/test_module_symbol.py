@@ -134,3 +134,48 @@ def test_conv_symbol_module():
     outs = ex.forward(is_train=False, data=nd.ones((2, 3, 8, 8)),
                       softmax_label=nd.zeros((2,)))
     assert outs[0].shape == (2, 10)
+
+
+def test_sequential_module_train():
+    from anand_mxnet_amd.module import SequentialModule
+    np.random.seed(3)
+    n = 400
+    x = np.random.randn(n, 20).astype(np.float32)
+    yl = (x[:, 0] > 0).astype(np.float32)
+    x[:, 1] += yl * 2
+    train = NDArrayIter(x, yl, batch_size=20, shuffle=True)
+
+    # body module (features) -> head module (classifier + loss)
+    data = sym.var("data")
+    h = sym.FullyConnected(data=data, weight=sym.var("fc1_weight"),
+                           bias=sym.var("fc1_bias"), num_hidden=32,
+                           name="fc1")
+    body_sym = sym.Activation(h, act_type="relu")
+
+    data2 = sym.var("data")
+    label = sym.var("softmax_label")
+    h2 = sym.FullyConnected(data=data2, weight=sym.var("fc2_weight"),
+                            bias=sym.var("fc2_bias"), num_hidden=2,
+                            name="fc2")
+    head_sym = sym.SoftmaxOutput(h2, label, name="softmax",
+                                 normalization="batch")
+
+    seq = SequentialModule()
+    seq.add(Module(body_sym, label_names=()))
+    seq.add(Module(head_sym), take_labels=True)
+    seq.bind(train.provide_data, train.provide_label)
+    seq.init_params()
+    seq.init_optimizer(optimizer="sgd",
+                       optimizer_params={"learning_rate": 0.1})
+
+    import anand_mxnet_amd as mx
+    metric = mx.metric.Accuracy()
+    for _ in range(4):
+        train.reset()
+        metric.reset()
+        for batch in train:
+            seq.forward(batch, is_train=True)
+            seq.backward()
+            seq.update()
+            seq.update_metric(metric, batch.label)
+    assert metric.get()[1] > 0.8, metric.get()
