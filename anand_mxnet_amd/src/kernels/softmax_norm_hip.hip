@@ -47,33 +47,38 @@ DEVINL float block_red_max(float v, float* scratch) {
 
 // vlen: optional per-row valid length (masked/causal attention); columns
 // >= vlen[row] get probability 0 (log: -inf -> stored as large negative).
+// wave-per-row (4 rows / 256-thr block): shuffle-only reductions, no
+// barriers, full lane activity at the hot widths (D=512 attention rows,
+// D<=1024 hidden rows) — a 256-thread block per row left 3/4 of the
+// lanes idle below D=2048.
 template <bool LOG>
 __global__ __launch_bounds__(256) void softmax_fwd(
     const bf16* __restrict__ x, bf16* __restrict__ y, long rows, int D,
     const int* __restrict__ vlen, float alpha) {
-  __shared__ float scratch[4];
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
+  for (long row = blockIdx.x * 4 + w; row < rows;
+       row += 4L * gridDim.x) {
     const bf16* xr = x + row * D;
     bf16* yr = y + row * D;
     const int V = vlen ? vlen[row] : D;
     float mx = -3.0e38f;
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 v = *(const bf16x8*)&xr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         if (i + j < V) mx = fmaxf(mx, bf2f(v[j]) * alpha);
     }
-    mx = block_red_max(mx, scratch);
+    mx = wave_red_max(mx);
     float sum = 0;
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 v = *(const bf16x8*)&xr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         if (i + j < V) sum += __expf(bf2f(v[j]) * alpha - mx);
     }
-    sum = block_red_sum(sum, scratch);
+    sum = wave_red_sum(sum);
     float inv = 1.f / sum, lse = __logf(sum);
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 v = *(const bf16x8*)&xr[i];
       bf16x8 o;
 #pragma unroll
@@ -95,21 +100,22 @@ template <bool LOG>
 __global__ __launch_bounds__(256) void softmax_bwd(
     const bf16* __restrict__ y, const bf16* __restrict__ dy,
     bf16* __restrict__ dx, long rows, int D, float alpha) {
-  __shared__ float scratch[4];
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
+  for (long row = blockIdx.x * 4 + w; row < rows;
+       row += 4L * gridDim.x) {
     const bf16* yr = y + row * D;
     const bf16* gr = dy + row * D;
     bf16* or_ = dx + row * D;
     float dot = 0;
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 vy = *(const bf16x8*)&yr[i];
       bf16x8 vg = *(const bf16x8*)&gr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         dot += LOG ? bf2f(vg[j]) : bf2f(vg[j]) * bf2f(vy[j]);
     }
-    dot = block_red_sum(dot, scratch);
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    dot = wave_red_sum(dot);
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 vy = *(const bf16x8*)&yr[i];
       bf16x8 vg = *(const bf16x8*)&gr[i];
       bf16x8 o;
@@ -132,12 +138,13 @@ __global__ __launch_bounds__(256) void layernorm_fwd(
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ save_mean, float* __restrict__ save_rstd, long rows,
     int D, float eps) {
-  __shared__ float scratch[4];
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
+  for (long row = blockIdx.x * 4 + w; row < rows;
+       row += 4L * gridDim.x) {
     const bf16* xr = x + row * D;
     bf16* yr = y + row * D;
     float s = 0, ss = 0;
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 v = *(const bf16x8*)&xr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -146,16 +153,16 @@ __global__ __launch_bounds__(256) void layernorm_fwd(
         ss += f * f;
       }
     }
-    s = block_red_sum(s, scratch);
-    ss = block_red_sum(ss, scratch);
+    s = wave_red_sum(s);
+    ss = wave_red_sum(ss);
     float mean = s / D;
     float var = ss / D - mean * mean;
     float rstd = rsqrtf(var > 0 ? var + eps : eps);
-    if (threadIdx.x == 0) {
+    if (l == 0) {
       save_mean[row] = mean;
       save_rstd[row] = rstd;
     }
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 v = *(const bf16x8*)&xr[i];
       bf16x8 o;
 #pragma unroll
@@ -173,14 +180,15 @@ __global__ __launch_bounds__(256) void layernorm_bwd_dx(
     bf16* __restrict__ dx, const float* __restrict__ gamma,
     const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
     long rows, int D) {
-  __shared__ float scratch[4];
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
+  for (long row = blockIdx.x * 4 + w; row < rows;
+       row += 4L * gridDim.x) {
     const bf16* xr = x + row * D;
     const bf16* gr = dy + row * D;
     bf16* or_ = dx + row * D;
     float mean = save_mean[row], rstd = save_rstd[row];
     float c1 = 0, c2 = 0;
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 vx = *(const bf16x8*)&xr[i];
       bf16x8 vg = *(const bf16x8*)&gr[i];
 #pragma unroll
@@ -191,9 +199,9 @@ __global__ __launch_bounds__(256) void layernorm_bwd_dx(
         c2 += gg * xh;
       }
     }
-    c1 = block_red_sum(c1, scratch) / D;
-    c2 = block_red_sum(c2, scratch) / D;
-    for (int i = threadIdx.x * 8; i < D; i += 256 * 8) {
+    c1 = wave_red_sum(c1) / D;
+    c2 = wave_red_sum(c2) / D;
+    for (int i = l * 8; i < D; i += 64 * 8) {
       bf16x8 vx = *(const bf16x8*)&xr[i];
       bf16x8 vg = *(const bf16x8*)&gr[i];
       bf16x8 o;
@@ -261,7 +269,8 @@ __global__ __launch_bounds__(256) void layernorm_bwd_dparams(
 // ---------------------------------------------------------------------------
 
 static int rows_grid(long rows) {
-  return (int)(rows < 2048 ? (rows < 1 ? 1 : rows) : 2048);
+  long blocks = (rows + 3) / 4;  // 4 rows (one wave each) per block
+  return (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
 }
 
 void softmax_fwd_bf16(const bf16* x, bf16* y, long rows, int D, bool log_,
