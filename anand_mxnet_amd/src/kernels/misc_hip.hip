@@ -27,7 +27,14 @@ __global__ __launch_bounds__(256) void dropout_fwd(
     const bf16* __restrict__ x, bf16* __restrict__ y,
     unsigned char* __restrict__ mask, long nvec,
     const u64* __restrict__ seed_ptr, u64 salt, float p, float scale) {
-  u64 seed = *seed_ptr + salt;
+  // finalize (seed+salt) BEFORE mixing with the element index: the raw
+  // sum is linear in the step counter, and hash(seed + c*s + c*(i+1))
+  // would make step s+1's mask a one-element SHIFT of step s's —
+  // perfectly correlated dropout across steps.
+  u64 z = *seed_ptr + salt;
+  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ULL;
+  z = (z ^ (z >> 27)) * 0x94d049bb133111ebULL;
+  u64 seed = z ^ (z >> 31);
   u32 thresh = (u32)(p * 4294967296.0f);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
