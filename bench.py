@@ -106,7 +106,10 @@ def main():
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        run_step()
+        out = run_step()
+        if out is not None:  # eager path returns the loss; graph replay
+            loss = out       # updates the captured static loss in place
+
     torch.cuda.synchronize()
     if distributed:
         dist.barrier()

@@ -135,7 +135,10 @@ def main():
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        run_step()
+        out = run_step()
+        if out is not None:  # eager path returns the loss; graph replay
+            loss = out       # updates the captured static loss in place
+
     torch.cuda.synchronize()
     if distributed:
         dist.barrier()
