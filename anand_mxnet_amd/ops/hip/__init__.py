@@ -147,13 +147,22 @@ class _Conv2dNHWC(torch.autograd.Function):
     def forward(ctx, x, w, bias, stride, pad, dil):
         x = x.contiguous()
         K, R, S, C = w.shape
-        w_rskc = w.permute(1, 2, 0, 3).contiguous()
         ctx.save_for_backward(x, w)
         ctx.conf = (stride, pad, dil)
+        ctx.has_bias = bias is not None
+        if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0):
+            # pointwise conv IS a plain GEMM: route through gemm_tn so
+            # the 256-tile 8-phase kernel picks it up (the implicit-GEMM
+            # gather loader adds address math for nothing here)
+            N, H, W_, _ = x.shape
+            y2 = _C.gemm_tn(x.reshape(N * H * W_, C), w.reshape(K, C),
+                            bias.float() if bias is not None else None,
+                            False)
+            return y2.reshape(N, H, W_, K)
+        w_rskc = w.permute(1, 2, 0, 3).contiguous()
         y = _C.conv2d_fwd(x, w_rskc,
                           bias.float() if bias is not None else None,
                           list(stride), list(pad), list(dil), False)
-        ctx.has_bias = bias is not None
         return y
 
     @staticmethod
@@ -164,7 +173,12 @@ class _Conv2dNHWC(torch.autograd.Function):
         K, R, S, C = w.shape
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            if R == 1 and S == 1 and (stride[0] > 1 or stride[1] > 1) \
+            if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0):
+                # pointwise: dx = dy @ W, the NN-form kernel (no W^T copy)
+                N, P, Q, _ = dy.shape
+                dx = _C.gemm_nn(dy.reshape(N * P * Q, K),
+                                w.reshape(K, C)).reshape(x.shape)
+            elif R == 1 and S == 1 and (stride[0] > 1 or stride[1] > 1) \
                     and pad == (0, 0):
                 # strided 1x1 (ResNet downsample): only every (sh, sw)-th
                 # input pixel has gradient — dense GEMM over the P*Q
@@ -180,9 +194,17 @@ class _Conv2dNHWC(torch.autograd.Function):
                 dx = _C.conv2d_dgrad(dy, w_rsck, list(x.shape),
                                      list(stride), list(pad), list(dil))
         if ctx.needs_input_grad[1]:
-            dwf = _C.conv2d_wgrad(x, dy, [R, S], list(stride), list(pad),
-                                  list(dil))  # [R,S,C,K] f32
-            dw = dwf.permute(3, 0, 1, 2).to(w.dtype)
+            if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0):
+                # pointwise wgrad: plain WG GEMM ([C,K] == [1,1,C,K])
+                N, P, Q, _ = dy.shape
+                dwf = _C.gemm_wg(x.reshape(N * P * Q, C),
+                                 dy.reshape(N * P * Q, K))
+                dw = dwf.reshape(1, 1, C, K).permute(3, 0, 1, 2) \
+                    .to(w.dtype)
+            else:
+                dwf = _C.conv2d_wgrad(x, dy, [R, S], list(stride),
+                                      list(pad), list(dil))  # [R,S,C,K]
+                dw = dwf.permute(3, 0, 1, 2).to(w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy)
         return dx, dw, db, None, None, None
