@@ -26,7 +26,7 @@
 
 namespace mxa {
 
-constexpr int BM = 128, BN = 128, BK = 64, THREADS = 256;
+constexpr int BM = 128, BN = 128, BK = 64, THREADS = 512;
 
 DEVINL int lds_off(int row, int chunk16) {
   // bf16 element offset of 16-byte chunk `chunk16` of row `row` in a
@@ -36,8 +36,9 @@ DEVINL int lds_off(int row, int chunk16) {
 
 DEVINL int stage_row(int p, int t) {
   // row a thread's lane-linear glds slot covers: wave w, pass p -> LDS
-  // bytes (w*4+p)*1024 + lane*16, i.e. row = (w*4+p)*8 + lane/8.
-  return (((t >> 6) * 4 + p) << 3) + ((t & 63) >> 3);
+  // bytes (w*2+p)*1024 + lane*16, i.e. row = (w*2+p)*8 + lane/8.
+  // (512-thr blocks: 8 waves x 2 passes cover the 128 tile rows.)
+  return (((t >> 6) * 2 + p) << 3) + ((t & 63) >> 3);
 }
 
 // ---------------------------------------------------------------------------
@@ -53,13 +54,13 @@ struct PlainLoader {
   ZMap zmap;      // batch offset map (ZMap{1, stride, 0} = plain batch)
   int ksteps;     // K / 64
   // device state
-  const bf16* rp[4];
+  const bf16* rp[2];
   long koff;
 
   DEVINL void init(int row0, int t, int z) {
     const bf16* b = base + zmap.off(z);
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int r = row0 + stage_row(p, t);
       rp[p] = (r < rows) ? b + (long)r * ld : nullptr;
     }
@@ -78,16 +79,16 @@ struct ConvALoader {
   int sh, sw, ph, pw, dh, dw;
   int csteps;  // C/64
   // device state
-  long nb[4];
-  int ih0[4], iw0[4];
+  long nb[2];
+  int ih0[2], iw0[2];
   int rowok;  // bitmask
-  const bf16* rp[4];
+  const bf16* rp[2];
 
   DEVINL void init(int row0, int t, int) {
     rowok = 0;
     int PQ = P * Q;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int m = row0 + stage_row(p, t);
       int ok = m < Nb * PQ;
       int mm = ok ? m : 0;
@@ -103,7 +104,7 @@ struct ConvALoader {
     int seg = ks / csteps, cb = ks % csteps;
     int r = seg / S, s = seg % S;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int ih = ih0[p] + r * dh, iw = iw0[p] + s * dw;
       bool ok = ((rowok >> p) & 1) && ih >= 0 && ih < H && iw >= 0 && iw < W;
       rp[p] = ok ? x + nb[p] + ((long)ih * W + iw) * C + cb * BK : nullptr;
@@ -120,12 +121,12 @@ struct ConvWLoader {
   const bf16* zpage;
   int Kout, C, csteps;
   // device state
-  int nrow[4];
-  const bf16* rp[4];
+  int nrow[2];
+  const bf16* rp[2];
 
   DEVINL void init(int row0, int t, int) {
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int n = row0 + stage_row(p, t);
       nrow[p] = (n < Kout) ? n : -1;
     }
@@ -134,7 +135,7 @@ struct ConvWLoader {
     int seg = ks / csteps, cb = ks % csteps;
     const bf16* base = w + (long)seg * Kout * C + cb * BK;
 #pragma unroll
-    for (int p = 0; p < 4; ++p)
+    for (int p = 0; p < 2; ++p)
       rp[p] = (nrow[p] >= 0) ? base + (long)nrow[p] * C : nullptr;
   }
   DEVINL const bf16* src(int p, int chunk16) const {
@@ -150,16 +151,16 @@ struct DgradALoader {
   int sh, sw, ph, pw, dh, dw;
   int ksteps;  // Kout/64
   // device state
-  long nb[4];
-  int h0[4], w0[4];
+  long nb[2];
+  int h0[2], w0[2];
   int rowok;
-  const bf16* rp[4];
+  const bf16* rp[2];
 
   DEVINL void init(int row0, int t, int) {
     rowok = 0;
     int HW = H * W;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int m = row0 + stage_row(p, t);
       int ok = m < Nb * HW;
       int mm = ok ? m : 0;
@@ -174,7 +175,7 @@ struct DgradALoader {
     int seg = ks / ksteps, kb = ks % ksteps;
     int r = seg / S, s = seg % S;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int pn = h0[p] - r * dh, qn = w0[p] - s * dw;
       int pi = pn / sh, qi = qn / sw;
       bool ok = ((rowok >> p) & 1) && pn >= 0 && qn >= 0 &&
@@ -194,12 +195,12 @@ struct DgradWLoader {
   const bf16* zpage;
   int C, Kout, ksteps;
   // device state
-  int nrow[4];
-  const bf16* rp[4];
+  int nrow[2];
+  const bf16* rp[2];
 
   DEVINL void init(int row0, int t, int) {
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int n = row0 + stage_row(p, t);
       nrow[p] = (n < C) ? n : -1;
     }
@@ -208,7 +209,7 @@ struct DgradWLoader {
     int seg = ks / ksteps, kb = ks % ksteps;
     const bf16* base = w + (long)seg * C * Kout + kb * BK;
 #pragma unroll
-    for (int p = 0; p < 4; ++p)
+    for (int p = 0; p < 2; ++p)
       rp[p] = (nrow[p] >= 0) ? base + (long)nrow[p] * Kout : nullptr;
   }
   DEVINL const bf16* src(int p, int chunk16) const {
@@ -254,11 +255,11 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
     aload.set_kstep(ks);
     bload.set_kstep(ks);
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       int row = stage_row(p, t);
       int cg = (ll & 7) ^ (row & 7);  // source chunk for this lane's slot
       unsigned lbase = __builtin_amdgcn_readfirstlane(
-          (unsigned)((lw * 4 + p) * 1024));
+          (unsigned)((lw * 2 + p) * 1024));
       auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&smem[buf][0][0] + lbase);
       auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
@@ -272,23 +273,26 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
     }
   };
 
-  f32x4 acc[4][4];
+  // 8 waves as 4(M) x 2(N): each wave owns a 32x64 patch (2 m-frags x
+  // 4 n-frags) — lighter acc pressure, 4 waves/SIMD to hide the gather
+  // loaders' global latency on conv shapes.
+  f32x4 acc[2][4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int w = lw, l = ll;
-  const int wm = (w >> 1) * 64, wn = (w & 1) * 64;
+  const int wm = (w & 3) * 32, wn = (w >> 2) * 64;
   const int lrow = l & 15, lk = l >> 4;  // fragment row / k-group
 
   auto compute = [&](int buf) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bfr[4];
+      bf16x8 af[2], bfr[4];
       int chunk = kk * 4 + lk;
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf) {
+      for (int mf = 0; mf < 2; ++mf) {
         int row = wm + mf * 16 + lrow;
         af[mf] = *(const bf16x8*)&smem[buf][0][lds_off(row, chunk)];
       }
@@ -298,7 +302,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
         bfr[nf] = *(const bf16x8*)&smem[buf][1][lds_off(row, chunk)];
       }
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
+      for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
         for (int nf = 0; nf < 4; ++nf)
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -321,7 +325,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
   // epilogue: C/D fragment layout (guide §3): col = l&15, row = (l>>4)*4+i
   bf16* cz = Cout + (long)z * strideC;
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
       int gcol = bn + wn + nf * 16 + (l & 15);
