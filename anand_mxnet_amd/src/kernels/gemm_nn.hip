@@ -26,14 +26,15 @@
 
 namespace mxa {
 
-constexpr int NBM = 128, NBN = 128, NBK = 64, NTHREADS = 256;
+constexpr int NBM = 128, NBN = 128, NBK = 64, NTHREADS = 512;
 
 // ---- A side (k-contiguous image, same scheme as gemm_tn) ----
 DEVINL int nn_a_off(int row, int chunk16) {
   return row * NBK + ((chunk16 ^ (row & 7)) << 3);
 }
 DEVINL int nn_a_row(int p, int t) {
-  return (((t >> 6) * 4 + p) << 3) + ((t & 63) >> 3);
+  // 512-thr blocks: 8 waves x 2 passes cover the 128 A rows
+  return (((t >> 6) * 2 + p) << 3) + ((t & 63) >> 3);
 }
 
 // ---- B side (row-major [64][128] image, tr-read fragments) ----
@@ -57,30 +58,30 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn_core(
 
   // A row pointers (TN-style: pass p covers row0 + nn_a_row(p,t))
   const bf16* abase = A + za.off(z);
-  const bf16* arp[4];
+  const bf16* arp[2];
 #pragma unroll
-  for (int p = 0; p < 4; ++p) {
+  for (int p = 0; p < 2; ++p) {
     int r = bm + nn_a_row(p, t);
     arp[p] = (r < M) ? abase + (long)r * lda : nullptr;
   }
   const int acg = (ll & 7) ^ (nn_a_row(0, t) & 7);  // row&7 == (ll>>3)&7, p-inv
 
-  // B row pointers (tr-style: pass p covers k-row lw*16 + p*4 + (ll>>4))
+  // B row pointers (tr-style: pass p covers k-row lw*8 + p*4 + (ll>>4))
   const bf16* bbase = B + zb.off(z);
   const int bcg = (ll & 15) ^ ((ll >> 4) << 1);
   int bcol = bn + bcg * 8;
   const bool bcolok = bcol + 8 <= N;
-  const bf16* brp[4];
+  const bf16* brp[2];
 #pragma unroll
-  for (int p = 0; p < 4; ++p)
-    brp[p] = bbase + (long)(lw * 16 + p * 4 + (ll >> 4)) * ldb + bcol;
+  for (int p = 0; p < 2; ++p)
+    brp[p] = bbase + (long)(lw * 8 + p * 4 + (ll >> 4)) * ldb + bcol;
 
   auto stage_glds = [&](int buf, int ks) {
     long koff = (long)ks * NBK;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       unsigned lbase =
-          __builtin_amdgcn_readfirstlane((unsigned)((lw * 4 + p) * 1024));
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 2 + p) * 1024));
       auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&smem[buf][0][0] + lbase);
       auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
@@ -94,13 +95,14 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn_core(
     }
   };
 
-  f32x4 acc[4][4];
+  f32x4 acc[2][4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int wm = (lw >> 1) * 64, wn = (lw & 1) * 64;
+  // 8 waves as 4(M) x 2(N): each wave a 32x64 patch
+  const int wm = (lw & 3) * 32, wn = (lw >> 2) * 64;
   const int lrow = ll & 15, lk = ll >> 4;
   const int ig = ll & 15, khi = ll >> 4;
 
@@ -113,10 +115,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn_core(
   auto compute = [&](int buf) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bfr[4];
+      bf16x8 af[2], bfr[4];
       int chunk = kk * 4 + lk;
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf) {
+      for (int mf = 0; mf < 2; ++mf) {
         int row = wm + mf * 16 + lrow;
         af[mf] = *(const bf16x8*)&smem[buf][0][nn_a_off(row, chunk)];
       }
@@ -130,7 +132,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn_core(
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
+      for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
         for (int nf = 0; nf < 4; ++nf)
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -149,7 +151,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn_core(
 
   bf16* cz = Cout + zc.off(z);
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
       int gcol = bn + wn + nf * 16 + (ll & 15);
@@ -186,57 +188,54 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn64_core(
   const int lw = t / WAVE, ll = t % WAVE;
 
   const bf16* abase = A + za.off(z);
-  const bf16* arp[4];
+  const bf16* arp[2];
 #pragma unroll
-  for (int p = 0; p < 4; ++p) {
+  for (int p = 0; p < 2; ++p) {
     int r = bm + nn_a_row(p, t);
     arp[p] = (r < M) ? abase + (long)r * lda : nullptr;
   }
   const int acg = (ll & 7) ^ (nn_a_row(0, t) & 7);
 
-  // B staging: wave w covers k-rows [w*16, w*16+16); each 1-KiB glds
-  // issue is 8 rows of 128 B; lane l -> row seg*8 + (l>>3), chunk l&7,
-  // source chunk XOR'd with ((row&3)<<1).
+  // B staging: 8 waves, one 1-KiB issue each (8 rows of 128 B);
+  // lane l -> row lw*8 + (l>>3), chunk l&7, source chunk XOR'd with
+  // ((row&3)<<1).
   const bf16* bbase = B + zb.off(z);
   const int bcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);
   int bcol = bcg * 8;
   const bool bcolok = bcol + 8 <= N;
-  const bf16* brp[2];
-#pragma unroll
-  for (int p = 0; p < 2; ++p)
-    brp[p] = bbase + (long)(lw * 16 + p * 8 + (ll >> 3)) * ldb + bcol;
+  const bf16* brp0 = bbase + (long)(lw * 8 + (ll >> 3)) * ldb + bcol;
 
   auto stage_glds = [&](int buf, int ks) {
     long koff = (long)ks * NBK;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       unsigned lbase =
-          __builtin_amdgcn_readfirstlane((unsigned)((lw * 4 + p) * 1024));
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 2 + p) * 1024));
       auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&smemA[buf][0] + lbase);
       auto gA = (const __attribute__((address_space(1))) unsigned int*)(
           arp[p] ? arp[p] + koff + acg * 8 : zpage + acg * 8);
       __builtin_amdgcn_global_load_lds(gA, ldsA, 16, 0, 0);
     }
-#pragma unroll
-    for (int p = 0; p < 2; ++p) {
+    {
       unsigned lbase =
-          __builtin_amdgcn_readfirstlane((unsigned)((lw * 2 + p) * 1024));
+          __builtin_amdgcn_readfirstlane((unsigned)(lw * 1024));
       auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&smemB[buf][0] + lbase);
       auto gB = (const __attribute__((address_space(1))) unsigned int*)(
-          bcolok ? brp[p] + koff * ldb : zpage + bcg * 8);
+          bcolok ? brp0 + koff * ldb : zpage + bcg * 8);
       __builtin_amdgcn_global_load_lds(gB, ldsB, 16, 0, 0);
     }
   };
 
-  f32x4 acc[4][2];
+  f32x4 acc[2][2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int wm = (lw >> 1) * 64, wn = (lw & 1) * 32;
+  // 8 waves as 4(M) x 2(N): each wave a 32x32 patch
+  const int wm = (lw & 3) * 32, wn = (lw >> 2) * 32;
   const int lrow = ll & 15, lk = ll >> 4;
   const int ig = ll & 15, khi = ll >> 4;
 
@@ -250,10 +249,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn64_core(
   auto compute = [&](int buf) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bfr[2];
+      bf16x8 af[2], bfr[2];
       int chunk = kk * 4 + lk;
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf) {
+      for (int mf = 0; mf < 2; ++mf) {
         int row = wm + mf * 16 + lrow;
         af[mf] = *(const bf16x8*)&smemA[buf][nn_a_off(row, chunk)];
       }
@@ -267,7 +266,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn64_core(
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
+      for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
         for (int nf = 0; nf < 2; ++nf)
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -286,7 +285,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn64_core(
 
   bf16* cz = Cout + zc.off(z);
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
     for (int nf = 0; nf < 2; ++nf) {
       int gcol = wn + nf * 16 + (ll & 15);

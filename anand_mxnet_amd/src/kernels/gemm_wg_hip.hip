@@ -22,6 +22,7 @@
 namespace mxa {
 
 constexpr int WBM = 128, WBN = 128, WBK = 64, WTHREADS = 256;
+constexpr int WT8 = 512;  // tr-variant thread count (8 waves per tile)
 
 DEVINL int colkey(int col) { return (col ^ (col >> 3)) & 7; }
 
@@ -285,8 +286,8 @@ struct TrPlainRows {
   long ld;
   int rowlen;
   // device state
-  const bf16* ptr[4];  // pass p -> this thread's source row pointer
-  long mrem[4];
+  const bf16* ptr[2];  // pass p -> this thread's source row pointer
+  long mrem[2];
   bool colok;
   int cgbyte;
 
@@ -295,14 +296,14 @@ struct TrPlainRows {
     int col = col0 + cg * 8;
     colok = col + 8 <= rowlen;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       ptr[p] = base + (m0 + myrow[p]) * ld + col;
       mrem[p] = mtotal - (m0 + myrow[p]);
     }
   }
   DEVINL void step() {
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       ptr[p] += (long)WBK * ld;
       mrem[p] -= WBK;
     }
@@ -320,7 +321,7 @@ struct TrIm2colRows {
   int sh, sw, ph, pw;
   // device state
   int roff, soff;
-  int n_[4], p_[4], q_[4];
+  int n_[2], p_[2], q_[2];
   bool colok;
   int col, cgbyte;
 
@@ -330,7 +331,7 @@ struct TrIm2colRows {
     colok = col + 8 <= C;
     int PQ = P * Q;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       long m = m0 + myrow[p];
       n_[p] = (int)(m / PQ);
       int pq = (int)(m % PQ);
@@ -340,7 +341,7 @@ struct TrIm2colRows {
   }
   DEVINL void step() {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < 2; ++i) {
       q_[i] += WBK;
       while (q_[i] >= Q) {
         q_[i] -= Q;
@@ -363,7 +364,7 @@ struct TrIm2colRows {
 };
 
 template <bool A_IM2COL, bool OUTBF16 = false>
-__global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
+__global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
     TrPlainRows aplain, TrIm2colRows aim, TrPlainRows bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
     int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic,
@@ -391,14 +392,14 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
   const int t = threadIdx.x;
   const int lw = t / WAVE, ll = t % WAVE;
 
-  // glds slot of pass p: LDS bytes (lw*4+p)*1024 + ll*16
-  //   -> row = lw*16 + p*4 + (ll>>4), chunk position ll&15.
+  // glds slot of pass p (8 waves x 2 passes): LDS bytes
+  // (lw*2+p)*1024 + ll*16 -> row = lw*8 + p*4 + (ll>>4), chunk ll&15.
   // Swizzle moves to the SOURCE chunk: cg = (ll&15) ^ key(row),
   // key(row) = (row&3)<<1 = (ll>>4)<<1 (uniform per lane).
   const int cg = (ll & 15) ^ ((ll >> 4) << 1);
-  int myrow[4];
+  int myrow[2];
 #pragma unroll
-  for (int p = 0; p < 4; ++p) myrow[p] = lw * 16 + p * 4 + (ll >> 4);
+  for (int p = 0; p < 2; ++p) myrow[p] = lw * 8 + p * 4 + (ll >> 4);
 
   long m0 = (long)kstep0 * WBK;
   if (A_IM2COL) {
@@ -412,9 +413,9 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
 
   auto stage_glds = [&](int buf) {
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 2; ++p) {
       unsigned lbase =
-          __builtin_amdgcn_readfirstlane((unsigned)((lw * 4 + p) * 1024));
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 2 + p) * 1024));
       auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&smem[buf][0][0] + lbase);
       auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
@@ -433,13 +434,14 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
     bload.step();
   };
 
-  f32x4 acc[4][4];
+  f32x4 acc[2][4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int wm = (lw >> 1) * 64, wn = (lw & 1) * 64;
+  // 8 waves as 4(Mo) x 2(No): each wave a 32x64 patch
+  const int wm = (lw & 3) * 32, wn = (lw >> 2) * 64;
   const int ig = ll & 15, khi = ll >> 4;
 
   auto trread = [&](const bf16* img, int row, int cbyte) {
@@ -454,9 +456,9 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
       // fragment k-rows this lane addresses: kk*32 + khi*8 + h*4 + (ig>>2)
       int rbase = kk * 32 + khi * 8 + (ig >> 2);
       int cb = (ig & 3) * 8;
-      bf16x8 af[4], bfr[4];
+      bf16x8 af[2], bfr[4];
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf) {
+      for (int mf = 0; mf < 2; ++mf) {
         int cbyte = (wm + mf * 16) * 2 + cb;
         bf16x4 lo = trread(smem[buf][0], rbase, cbyte);
         bf16x4 hi = trread(smem[buf][0], rbase + 4, cbyte);
@@ -470,7 +472,7 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
+      for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
         for (int nf = 0; nf < 4; ++nf)
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -491,7 +493,7 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_tr_core(
 
   float* cseg = Cout + (long)seg * strideSeg;
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
       int gcol = bn + wn + nf * 16 + (ll & 15);
@@ -571,7 +573,7 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, long mtotal,
     TrPlainRows a{A, zpage, mtotal, lda, Mo};
     TrPlainRows b{B, zpage, mtotal, ldb, No};
     TrIm2colRows dummy{};
-   hipLaunchKernelGGL(( gemm_wg_tr_core<false>), dim3(grid), dim3(WTHREADS), 0, stream, 
+   hipLaunchKernelGGL(( gemm_wg_tr_core<false>), dim3(grid), dim3(WT8), 0, stream, 
         a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic,
         ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0});
     return;
@@ -600,7 +602,7 @@ void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
     TrIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
     TrPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
     TrPlainRows dummy{};
-   hipLaunchKernelGGL(( gemm_wg_tr_core<true>), dim3(grid), dim3(WTHREADS), 0, stream, 
+   hipLaunchKernelGGL(( gemm_wg_tr_core<true>), dim3(grid), dim3(WT8), 0, stream, 
         dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
         ksteps, nsplit, use_atomic,
         ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0});
@@ -634,7 +636,7 @@ void attn_wg_bf16(const bf16* A, const bf16* Bq, bf16* C, int B, int H,
   dim3 grid((unsigned)bxy, 1, (unsigned)(B * H));
   // single reduce slice per z -> no atomics -> bf16 output directly
   // (saves a full f32->bf16 cast pass over [B, S, H*D] per call)
- hipLaunchKernelGGL(( gemm_wg_tr_core<false, true>), dim3(grid), dim3(WTHREADS), 0, stream, 
+ hipLaunchKernelGGL(( gemm_wg_tr_core<false, true>), dim3(grid), dim3(WT8), 0, stream, 
       a, dummy, b, (float*)C, Tk, D, hd, 0, 1, 1, 1, ksteps, 1, 0,
       ZMap{1, (long)Tq * Tk, 0}, ZMap{H, Tq * hd, D}, ZMap{H, Tk * hd, D});
 }
