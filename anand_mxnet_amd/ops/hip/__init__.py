@@ -178,6 +178,40 @@ class _Conv2dNHWC(torch.autograd.Function):
                 N, P, Q, _ = dy.shape
                 dx = _C.gemm_nn(dy.reshape(N * P * Q, K),
                                 w.reshape(K, C)).reshape(x.shape)
+            elif R == 3 and S == 3 and stride == (2, 2) and pad == (1, 1) \
+                    and dil == (1, 1) and x.shape[1] % 2 == 0 \
+                    and x.shape[2] % 2 == 0 and K % 64 == 0 and C % 8 == 0:
+                # stride-2 3x3 (ResNet transition): the scatter-inverse
+                # gather GEMM wastes 3/4 of its rows on parity predicates.
+                # Class-decompose by (h%2, w%2): each input-pixel class
+                # uses a fixed (r, s) weight subset, so the dgrad becomes
+                # 9 DENSE [NPQ,K]x[K,C] NN GEMMs (total FLOPs = fwd) plus
+                # shifted quarter-grid adds.
+                N, P, Q, _ = dy.shape
+                dy2 = dy.reshape(N * P * Q, K)
+
+                def g(r, s_):
+                    return _C.gemm_nn(dy2, w[:, r, s_, :].contiguous()) \
+                        .reshape(N, P, Q, C)
+
+                dx = torch.zeros(x.shape, dtype=dy.dtype, device=dy.device)
+                # h even, w even: r=1, s=1, (p,q)=(hh,ww)
+                dx[:, 0::2, 0::2, :] = g(1, 1)
+                # h even, w odd: s=0 -> q=ww+1 ; s=2 -> q=ww
+                c01 = dx[:, 0::2, 1::2, :]
+                c01.copy_(g(1, 2))
+                c01[:, :, :Q - 1, :] += g(1, 0)[:, :, 1:, :]
+                # h odd, w even: r=0 -> p=hh+1 ; r=2 -> p=hh
+                c10 = dx[:, 1::2, 0::2, :]
+                c10.copy_(g(2, 1))
+                c10[:, :P - 1, :, :] += g(0, 1)[:, 1:, :, :]
+                # h odd, w odd: four corner taps
+                c11 = dx[:, 1::2, 1::2, :]
+                c11.copy_(g(2, 2))
+                c11[:, :, :Q - 1, :] += g(2, 0)[:, :, 1:, :]
+                c11[:, :P - 1, :, :] += g(0, 2)[:, 1:, :, :]
+                c11[:, :P - 1, :Q - 1, :] += g(0, 0)[:, 1:, 1:, :]
+                dx = dx.contiguous()
             elif R == 1 and S == 1 and (stride[0] > 1 or stride[1] > 1) \
                     and pad == (0, 0):
                 # strided 1x1 (ResNet downsample): only every (sh, sw)-th

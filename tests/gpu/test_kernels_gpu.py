@@ -460,3 +460,34 @@ def test_bn_add_relu_fused():
     assert_close(y, ref, msg="bn_add_relu fwd")
     assert_close(x1.grad, x2.grad, rtol=5e-2, atol=5e-2, msg="bn_add dx")
     assert_close(r1.grad, r2.grad, rtol=5e-2, atol=5e-2, msg="bn_add dr")
+
+
+@pytest.mark.parametrize("cfg", [
+    ((2, 16, 16, 64), 128, 3, 2, 1),    # 3x3 s2: class-decomposed dgrad
+    ((2, 14, 14, 128), 256, 1, 1, 0),   # 1x1 s1: plain-GEMM routing
+    ((2, 14, 14, 128), 256, 1, 2, 0),   # 1x1 s2: dense-subgrid dgrad
+    ((2, 16, 16, 64), 64, 3, 1, 1),     # 3x3 s1: implicit-GEMM path
+])
+def test_conv_op_autograd_paths(cfg):
+    """The Python conv op's specialized dgrad/wgrad routings vs torch."""
+    from anand_mxnet_amd.ops.hip import conv2d_nhwc
+    xs, K, R, s, p = cfg
+    x0 = _rt(xs).to(BF)
+    w0 = _rt((K, R, R, xs[3]), 0.2).to(BF)
+
+    x1 = x0.clone().requires_grad_(True)
+    w1 = w0.clone().requires_grad_(True)
+    y1 = conv2d_nhwc(x1, w1, None, (s, s), (p, p), (1, 1), 1)
+    g = _rt(tuple(y1.shape)).to(BF)
+    y1.backward(g)
+
+    x2 = x0.clone().float().requires_grad_(True)
+    w2 = w0.clone().float().requires_grad_(True)
+    y2 = _conv_ref(x2, w2, (s, s), (p, p))
+    y2.backward(g.float())
+
+    assert_close(y1, y2, msg=f"conv op fwd {cfg}")
+    assert_close(x1.grad, x2.grad, rtol=4e-2, atol=5e-2,
+                 msg=f"conv op dx {cfg}")
+    assert_close(w1.grad, w2.grad, rtol=4e-2, atol=6e-2,
+                 msg=f"conv op dw {cfg}")
