@@ -652,34 +652,35 @@ class _MhaCore(torch.autograd.Function):
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
         B, Tq, HD = q.shape
         s = _C.attn_scores(q, k, H)                  # [BH, Tq, Tk] logits
-        if causal:
-            p = _C.softmax_masked_fwd(
-                s, _causal_vlen(B * H, Tq, s.device), scale)
+        vlen = _causal_vlen(B * H, Tq, s.device) if causal else None
+        if dropout_p > 0.0 and training:
+            # fused probs+dropout: one kernel yields p (softmax bwd),
+            # pd (ctx/dV operand) and the mask
+            p, pd, mask = _C.softmax_dropout_fwd(s, vlen, scale,
+                                                 dropout_p, _next_seed())
+        elif causal:
+            p = _C.softmax_masked_fwd(s, vlen, scale)
+            pd, mask = p, None
         else:
             p = _C.softmax_fwd(s, False, scale)
-        if dropout_p > 0.0 and training:
-            pd, mask = _C.dropout_fwd(p, dropout_p, _next_seed())
-        else:
             pd, mask = p, None
         out = _C.attn_ctx(pd, v, H)                  # [B, Tq, HD]
-        ctx.save_for_backward(q, k, v, p, mask)
+        ctx.save_for_backward(q, k, v, p, pd, mask)
         ctx.H, ctx.scale, ctx.dropout_p = H, scale, dropout_p
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        q, k, v, p, mask = ctx.saved_tensors
+        q, k, v, p, pd, mask = ctx.saved_tensors
         H, scale, dropout_p = ctx.H, ctx.scale, ctx.dropout_p
         dout = dout.contiguous()
         dpd = _C.attn_scores(dout, v, H)             # dP (post-dropout)
+        dv = _C.attn_wg(pd, dout, H)
         if mask is not None:
-            pd = _C.dropout_bwd(p, mask, dropout_p)  # recompute dropped P
-            dv = _C.attn_wg(pd, dout, H)
-            dp = _C.dropout_bwd(dpd, mask, dropout_p)
+            # fused dropout-bwd + softmax-bwd (+ scale fold)
+            ds = _C.softmax_dropout_bwd(p, dpd, mask, scale, dropout_p)
         else:
-            dv = _C.attn_wg(p, dout, H)
-            dp = dpd
-        ds = _C.softmax_bwd(p, dp, False, scale)     # [BH, Tq, Tk]
+            ds = _C.softmax_bwd(p, dpd, False, scale)
         dq = _C.attn_ctx(ds, k, H)
         dk = _C.attn_wg(ds, q, H)
         return dq, dk, dv, None, None, None, None, None

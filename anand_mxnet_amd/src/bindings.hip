@@ -61,6 +61,13 @@ void softmax_masked_fwd_bf16(const bf16*, bf16*, long, int, const int*,
                              hipStream_t);
 void softmax_bwd_bf16(const bf16*, const bf16*, bf16*, long, int, bool,
                       float, hipStream_t);
+void softmax_dropout_fwd_bf16(const bf16*, bf16*, bf16*, unsigned char*,
+                              long, int, const int*, float,
+                              const unsigned long long*,
+                              unsigned long long, float, hipStream_t);
+void softmax_dropout_bwd_bf16(const bf16*, const bf16*,
+                              const unsigned char*, bf16*, long, int,
+                              float, float, hipStream_t);
 void layernorm_fwd_bf16(const bf16*, bf16*, const float*, const float*,
                         float*, float*, long, int, float, hipStream_t);
 void layernorm_bwd_bf16(const bf16*, const bf16*, bf16*, const float*,
@@ -477,6 +484,50 @@ at::Tensor softmax_bwd(const at::Tensor& y, const at::Tensor& dy,
   return dx;
 }
 
+// persistent device RNG seed word per device (graph-safe dropout)
+at::Tensor& rng_seed_for(const at::Tensor& like) {
+  static std::unordered_map<int, at::Tensor> cache;
+  int dev = like.get_device();
+  auto it = cache.find(dev);
+  if (it == cache.end()) {
+    auto z = at::full({1}, (long)0x243F6A8885A308D3LL,
+                      like.options().dtype(at::kLong));
+    it = cache.emplace(dev, z).first;
+  }
+  return it->second;
+}
+
+std::vector<at::Tensor> softmax_dropout_fwd(
+    const at::Tensor& x, const c10::optional<at::Tensor>& vlen,
+    double alpha, double pdrop, long salt) {
+  int D = x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "softmax_dropout: last dim must be %8");
+  long rows = x.numel() / D;
+  auto p = at::empty_like(x);
+  auto pd = at::empty_like(x);
+  auto mask = at::empty({x.numel()}, x.options().dtype(at::kByte));
+  auto& sw = rng_seed_for(x);
+  mxa::softmax_dropout_fwd_bf16(
+      bfp(x), bfp_mut(p), bfp_mut(pd), mask.data_ptr<unsigned char>(),
+      rows, D, vlen.has_value() ? vlen->data_ptr<int>() : nullptr,
+      (float)alpha, (const unsigned long long*)sw.data_ptr<long>(),
+      (unsigned long long)salt, (float)pdrop, cur_stream());
+  return {p, pd, mask};
+}
+
+at::Tensor softmax_dropout_bwd(const at::Tensor& p, const at::Tensor& dpd,
+                               const at::Tensor& mask, double alpha,
+                               double pdrop) {
+  int D = p.size(-1);
+  long rows = p.numel() / D;
+  auto ds = at::empty_like(p);
+  mxa::softmax_dropout_bwd_bf16(bfp(p), bfp(dpd),
+                                mask.data_ptr<unsigned char>(),
+                                bfp_mut(ds), rows, D, (float)alpha,
+                                (float)pdrop, cur_stream());
+  return ds;
+}
+
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
                                       const at::Tensor& gamma,
                                       const at::Tensor& beta, double eps) {
@@ -599,19 +650,6 @@ at::Tensor colsum(const at::Tensor& x) {
   auto out = at::empty({N}, x.options().dtype(at::kFloat));
   mxa::colsum_bf16(bfp(x), out.data_ptr<float>(), m, N, cur_stream());
   return out;
-}
-
-// persistent device RNG seed word per device (graph-safe dropout)
-at::Tensor& rng_seed_for(const at::Tensor& like) {
-  static std::unordered_map<int, at::Tensor> cache;
-  int dev = like.get_device();
-  auto it = cache.find(dev);
-  if (it == cache.end()) {
-    auto z = at::full({1}, (long)0x243F6A8885A308D3LL,
-                      like.options().dtype(at::kLong));
-    it = cache.emplace(dev, z).first;
-  }
-  return it->second;
 }
 
 std::vector<at::Tensor> dropout_fwd(const at::Tensor& x, double p,
@@ -822,6 +860,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_wg", &attn_wg);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("rng_step", &rng_step);
+  m.def("softmax_dropout_fwd", &softmax_dropout_fwd);
+  m.def("softmax_dropout_bwd", &softmax_dropout_bwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);

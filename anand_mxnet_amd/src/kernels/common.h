@@ -37,6 +37,21 @@ struct ZMap {
   }
 };
 
+// splitmix64-finalizer counter hash (dropout RNG; statistically strong,
+// stateless — reference parity: philox counter RNG in dropout.cu)
+DEVINL u32 hash_u32(u64 seed, u64 idx) {
+  u64 z = seed + 0x9e3779b97f4a7c15ULL * (idx + 1);
+  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ULL;
+  z = (z ^ (z >> 27)) * 0x94d049bb133111ebULL;
+  return (u32)(z >> 32);
+}
+
+DEVINL u64 seed_finalize(u64 z) {
+  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ULL;
+  z = (z ^ (z >> 27)) * 0x94d049bb133111ebULL;
+  return z ^ (z >> 31);
+}
+
 // ceil-div
 DEVINL constexpr int cdiv(int a, int b) { return (a + b - 1) / b; }
 static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }

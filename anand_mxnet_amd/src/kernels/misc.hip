@@ -11,14 +11,6 @@ namespace mxa {
 // Dropout: philox-style counter hash -> mask byte + scaled output, fused.
 // ---------------------------------------------------------------------------
 
-DEVINL u32 hash_u32(u64 seed, u64 idx) {
-  // splitmix64 finalizer on (seed ^ idx): statistically strong, stateless
-  u64 z = seed + 0x9e3779b97f4a7c15ULL * (idx + 1);
-  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ULL;
-  z = (z ^ (z >> 27)) * 0x94d049bb133111ebULL;
-  return (u32)(z >> 32);
-}
-
 // seed = *seed_ptr + salt: the device seed word evolves via rng_bump
 // (graph-safe — a captured step replays the bump, so dropout masks differ
 // per replay), the per-call salt decorrelates call sites within a step.
@@ -30,10 +22,7 @@ __global__ __launch_bounds__(256) void dropout_fwd(
   // sum is linear in the step counter, and hash(seed + c*s + c*(i+1))
   // would make step s+1's mask a one-element SHIFT of step s's —
   // perfectly correlated dropout across steps.
-  u64 z = *seed_ptr + salt;
-  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ULL;
-  z = (z ^ (z >> 27)) * 0x94d049bb133111ebULL;
-  u64 seed = z ^ (z >> 31);
+  u64 seed = seed_finalize(*seed_ptr + salt);
   u32 thresh = (u32)(p * 4294967296.0f);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {

@@ -130,6 +130,106 @@ __global__ __launch_bounds__(256) void softmax_bwd(
 }
 
 // ---------------------------------------------------------------------------
+// Fused attention-probs softmax + dropout (wave-per-row).  One kernel
+// produces p (kept for softmax bwd), pd = dropout(p) (the ctx-GEMM /
+// dV operand) and the mask — replaces the softmax + dropout_fwd pair
+// and the bwd-side pd recompute (three full passes over [BH,Tq,Tk]).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void softmax_dropout_fwd(
+    const bf16* __restrict__ x, bf16* __restrict__ p,
+    bf16* __restrict__ pd, unsigned char* __restrict__ mask, long rows,
+    int D, const int* __restrict__ vlen, float alpha,
+    const u64* __restrict__ seed_ptr, u64 salt, float pdrop,
+    float dscale) {
+  const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
+  u64 seed = seed_finalize(*seed_ptr + salt);  // graph-safe (see misc.hip)
+  u32 thresh = (u32)(pdrop * 4294967296.0f);
+  for (long row = blockIdx.x * 4 + w; row < rows;
+       row += 4L * gridDim.x) {
+    const bf16* xr = x + row * D;
+    const int V = vlen ? vlen[row] : D;
+    float mx = -3.0e38f;
+    for (int i = l * 8; i < D; i += 64 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (i + j < V) mx = fmaxf(mx, bf2f(v[j]) * alpha);
+    }
+    mx = wave_red_max(mx);
+    float sum = 0;
+    for (int i = l * 8; i < D; i += 64 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (i + j < V) sum += __expf(bf2f(v[j]) * alpha - mx);
+    }
+    sum = wave_red_sum(sum);
+    float inv = 1.f / sum;
+    for (int i = l * 8; i < D; i += 64 * 8) {
+      bf16x8 v = *(const bf16x8*)&xr[i];
+      bf16x8 op, od;
+      unsigned char mb[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float pv = (i + j < V)
+                       ? __expf(bf2f(v[j]) * alpha - mx) * inv
+                       : 0.f;
+        u32 r = hash_u32(seed, row * D + i + j);
+        bool keep = r >= thresh;
+        mb[j] = keep;
+        op[j] = f2bf(pv);
+        od[j] = keep ? f2bf(pv * dscale) : (bf16)0.f;
+      }
+      *(bf16x8*)&p[row * D + i] = op;
+      *(bf16x8*)&pd[row * D + i] = od;
+      *(u64*)&mask[row * D + i] = *(const u64*)mb;
+    }
+  }
+}
+
+// ds = alpha * p * (dp - sum(dp * p)),  dp = mask * dscale * dpd
+__global__ __launch_bounds__(256) void softmax_dropout_bwd(
+    const bf16* __restrict__ p, const bf16* __restrict__ dpd,
+    const unsigned char* __restrict__ mask, bf16* __restrict__ ds,
+    long rows, int D, float alpha, float dscale) {
+  const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
+  for (long row = blockIdx.x * 4 + w; row < rows;
+       row += 4L * gridDim.x) {
+    const bf16* pr = p + row * D;
+    const bf16* gr = dpd + row * D;
+    const unsigned char* mr = mask + row * D;
+    bf16* or_ = ds + row * D;
+    float dot = 0;
+    for (int i = l * 8; i < D; i += 64 * 8) {
+      bf16x8 vp = *(const bf16x8*)&pr[i];
+      bf16x8 vg = *(const bf16x8*)&gr[i];
+      u64 m8 = *(const u64*)&mr[i];
+      const unsigned char* mb = (const unsigned char*)&m8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float dp = mb[j] ? bf2f(vg[j]) * dscale : 0.f;
+        dot += dp * bf2f(vp[j]);
+      }
+    }
+    dot = wave_red_sum(dot);
+    for (int i = l * 8; i < D; i += 64 * 8) {
+      bf16x8 vp = *(const bf16x8*)&pr[i];
+      bf16x8 vg = *(const bf16x8*)&gr[i];
+      u64 m8 = *(const u64*)&mr[i];
+      const unsigned char* mb = (const unsigned char*)&m8;
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float dp = mb[j] ? bf2f(vg[j]) * dscale : 0.f;
+        o[j] = f2bf((dp - dot) * bf2f(vp[j]) * alpha);
+      }
+      *(bf16x8*)&or_[i] = o;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // LayerNorm forward/backward: block per row, saves mean/rstd for bwd
 // ---------------------------------------------------------------------------
 
@@ -298,6 +398,24 @@ void softmax_bwd_bf16(const bf16* y, const bf16* dy, bf16* dx, long rows,
   else
    hipLaunchKernelGGL(( softmax_bwd<false>), dim3(rows_grid(rows)), dim3(256), 0, s, y, dy, dx, rows, D,
                                                        alpha);
+}
+
+void softmax_dropout_fwd_bf16(const bf16* x, bf16* p, bf16* pd,
+                              unsigned char* mask, long rows, int D,
+                              const int* vlen, float alpha,
+                              const u64* seed_ptr, u64 salt, float pdrop,
+                              hipStream_t s) {
+ hipLaunchKernelGGL(( softmax_dropout_fwd), dim3(rows_grid(rows)), dim3(256), 0, s, 
+      x, p, pd, mask, rows, D, vlen, alpha, seed_ptr, salt, pdrop,
+      1.f / (1.f - pdrop));
+}
+
+void softmax_dropout_bwd_bf16(const bf16* p, const bf16* dpd,
+                              const unsigned char* mask, bf16* ds,
+                              long rows, int D, float alpha, float pdrop,
+                              hipStream_t s) {
+ hipLaunchKernelGGL(( softmax_dropout_bwd), dim3(rows_grid(rows)), dim3(256), 0, s, 
+      p, dpd, mask, ds, rows, D, alpha, 1.f / (1.f - pdrop));
 }
 
 void layernorm_fwd_bf16(const bf16* x, bf16* y, const float* gamma,

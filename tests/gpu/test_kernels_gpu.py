@@ -491,3 +491,36 @@ def test_conv_op_autograd_paths(cfg):
                  msg=f"conv op dx {cfg}")
     assert_close(w1.grad, w2.grad, rtol=4e-2, atol=6e-2,
                  msg=f"conv op dw {cfg}")
+
+
+def test_softmax_dropout_fused():
+    rows, D = 512, 512
+    x = _rt((rows, D)).to(BF)
+    p, pd, mask = _C.softmax_dropout_fwd(x.contiguous(), None, 0.5, 0.25,
+                                         1234)
+    ref_p = torch.nn.functional.softmax(x.float() * 0.5, dim=-1)
+    assert_close(p, ref_p, msg="fused p")
+    m = mask.reshape(rows, D).bool()
+    keep = m.float().mean().item()
+    assert abs(keep - 0.75) < 0.02, keep
+    ref_pd = torch.where(m, p.float() / 0.75, torch.zeros(()).cuda())
+    assert_close(pd, ref_pd, rtol=4e-2, atol=1e-3, msg="fused pd")
+    # backward vs composed reference
+    dpd = _rt((rows, D)).to(BF)
+    ds = _C.softmax_dropout_bwd(p, dpd.contiguous(), mask, 0.5, 0.25)
+    dp = torch.where(m, dpd.float() / 0.75, torch.zeros(()).cuda())
+    dot = (dp * p.float()).sum(-1, keepdim=True)
+    ref_ds = (dp - dot) * p.float() * 0.5
+    assert_close(ds, ref_ds, rtol=4e-2, atol=2e-3, msg="fused ds")
+
+
+def test_mha_dropout_path_trains():
+    from anand_mxnet_amd.ops.hip import mha
+    B, H, T, D = 2, 4, 128, 64
+    q = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    k = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    v = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    out = mha(q, k, v, H, False, 0.125, 0.2, True)
+    out.sum().backward()
+    for t in (q, k, v):
+        assert t.grad is not None and torch.isfinite(t.grad.float()).all()
