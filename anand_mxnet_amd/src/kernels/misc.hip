@@ -229,28 +229,64 @@ void embedding_bwd_bf16(const long* idx, const bf16* dy, float* dW,
 // Column sum (bias gradient): out[c] = sum_m x[m][c], bf16 in / f32 out.
 // One thread per column, rows split over blockIdx.y, f32 atomics to finish
 // (replaces torch's materialized .float() copy + reduce_kernel pair).
-__global__ void colsum_kernel(const bf16* __restrict__ x,
-                              float* __restrict__ out, long m, int N,
-                              long rows_per) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= N) return;
-  long r0 = blockIdx.y * rows_per;
-  long r1 = min(m, r0 + rows_per);
+// 256 thr = 32 row-lanes x 8 chunks of 8 channels (same geometry as
+// bn_stats_partial): bf16x8 row loads, LDS cross-lane reduce, one atomic
+// per channel per block — the 1-thread-per-column form was latency-bound
+// on the small-N bias grads.
+__global__ __launch_bounds__(256) void colsum_kernel(
+    const bf16* __restrict__ x, float* __restrict__ out, long m, int N) {
+  const int cblk = blockIdx.x * 64;
+  const int t = threadIdx.x;
+  const int chunk = t & 7, rlane = t >> 3;
+  const int c0 = cblk + chunk * 8;
+  const bool active = c0 + 8 <= N;
+  float acc[8] = {0};
+  for (long r = blockIdx.y * 32 + rlane; active && r < m;
+       r += 32L * gridDim.y) {
+    bf16x8 v = *(const bf16x8*)&x[r * N + c0];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[i] += bf2f(v[i]);
+  }
+  __shared__ float red[32][64];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) red[rlane][chunk * 8 + i] = acc[i];
+  __syncthreads();
+  if (t < 64) {
+    float f = 0;
+#pragma unroll
+    for (int r = 0; r < 32; ++r) f += red[r][t];
+    int c = cblk + t;
+    if (c < N) {
+      if (gridDim.y > 1)
+        atomicAdd(&out[c], f);
+      else
+        out[c] = f;
+    }
+  }
+}
+
+// tail columns (N % 8 != 0 can't take bf16x8 loads): scalar finish
+__global__ void colsum_tail(const bf16* __restrict__ x,
+                            float* __restrict__ out, long m, int N,
+                            int c0) {
+  int c = c0 + blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= N) return;
   float acc = 0.f;
-  for (long r = r0; r < r1; ++r) acc += bf2f(x[r * N + col]);
-  if (gridDim.y > 1)
-    atomicAdd(&out[col], acc);
-  else
-    out[col] = acc;
+  for (long r = 0; r < m; ++r) acc += bf2f(x[r * N + c]);
+  out[c] = acc;
 }
 
 void colsum_bf16(const bf16* x, float* out, long m, int N, hipStream_t s) {
-  int bx = cdiv_h(N, 256);
-  int nsplit = 1;
-  while ((long)bx * nsplit < 1024 && nsplit < m && nsplit < 512) nsplit *= 2;
-  long rows_per = (m + nsplit - 1) / nsplit;
-  if (nsplit > 1) HIP_CHECK(hipMemsetAsync(out, 0, N * sizeof(float), s));
-  colsum_kernel<<<dim3(bx, nsplit), 256, 0, s>>>(x, out, m, N, rows_per);
+  if (N % 8 != 0) {
+    // rows not 16-B aligned -> scalar kernel over all columns
+    colsum_tail<<<cdiv_h(N, 256), min(N, 256), 0, s>>>(x, out, m, N, 0);
+    return;
+  }
+  int cb = cdiv_h(N, 64);
+  long gy = min((m + 31) / 32, (long)(1024 / cb + 1));
+  if (gy < 1) gy = 1;
+  if (gy > 1) HIP_CHECK(hipMemsetAsync(out, 0, N * sizeof(float), s));
+  colsum_kernel<<<dim3(cb, (unsigned)gy), 256, 0, s>>>(x, out, m, N);
 }
 
 void act_fwd_bf16(int op, const bf16* x, bf16* y, long n, hipStream_t s) {
