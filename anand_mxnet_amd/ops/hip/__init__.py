@@ -736,6 +736,15 @@ class _Fp8Linear(torch.autograd.Function):
     def forward(ctx, x, w, bias):
         ctx.save_for_backward(x, w)
         ctx.has_bias = bias is not None
+        # MX block-scaled fp8 (per-32-element e8m0 scales) runs on the
+        # double-rate mfma_scale path; per-tensor-scaled fp8 kept as the
+        # fallback for K % 128 != 0.
+        if x.shape[-1] % 128 == 0:
+            x8, xs = _C.mx_cast(x.contiguous())
+            w8, ws = _C.mx_cast(w.contiguous())
+            return _C.gemm_mx(x8, xs, w8, ws,
+                              bias.float() if bias is not None else None,
+                              False)
         sx = _amax_scale(x)
         sw = _amax_scale(w)
         x8 = _C.cast_to_fp8(x.contiguous(), sx)

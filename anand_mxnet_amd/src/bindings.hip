@@ -103,6 +103,12 @@ void gemm_tn_fp8(const unsigned char*, const unsigned char*, bf16*,
                  const float*, const float*, const float*, int, int, int,
                  long, long, long, long, long, long, int, bool,
                  const unsigned char*, hipStream_t);
+void gemm_mx_bf16(const unsigned char*, const unsigned char*,
+                  const unsigned char*, const unsigned char*, bf16*,
+                  const float*, int, int, int, long, int,
+                  const unsigned char*, hipStream_t);
+void mx_cast_bf16(const bf16*, unsigned char*, unsigned char*, long,
+                  hipStream_t);
 void cast_to_fp8(const bf16*, unsigned char*, long, const float*,
                  hipStream_t);
 // optim.hip
@@ -231,6 +237,40 @@ at::Tensor gemm_tn_fp8(const at::Tensor& A8, const at::Tensor& B8,
                    (batched && B8.dim() == 3) ? N * K : 0,
                    batched ? M * N : 0, batches, relu,
                    (const unsigned char*)zp(zpage_for(out)), cur_stream());
+  return out;
+}
+
+std::vector<at::Tensor> mx_cast(const at::Tensor& x) {
+  // bf16 [.., K] -> (fp8 e4m3 payload u8, e8m0 block scales u8 [.., K/32])
+  TORCH_CHECK(x.is_contiguous() && x.size(-1) % 32 == 0,
+              "mx_cast: last dim must be %32");
+  long nblk = x.numel() / 32;
+  auto q = at::empty(x.sizes(), x.options().dtype(at::kByte));
+  auto shp = x.sizes().vec();
+  shp.back() /= 32;
+  auto sc = at::empty(shp, x.options().dtype(at::kByte));
+  mxa::mx_cast_bf16(bfp(x), q.data_ptr<unsigned char>(),
+                    sc.data_ptr<unsigned char>(), nblk, cur_stream());
+  return {q, sc};
+}
+
+at::Tensor gemm_mx(const at::Tensor& A8, const at::Tensor& Asc,
+                   const at::Tensor& B8, const at::Tensor& Bsc,
+                   const c10::optional<at::Tensor>& bias, bool relu) {
+  // C[M][N] bf16 = MX-fp8 A[M,K] @ B[N,K]^T (both k-contiguous)
+  TORCH_CHECK(A8.dim() == 2 && B8.dim() == 2);
+  long M = A8.size(0), K = A8.size(1), N = B8.size(0);
+  TORCH_CHECK(B8.size(1) == K && K % 128 == 0, "gemm_mx: K % 128 != 0");
+  auto opts = at::TensorOptions().device(A8.device()).dtype(at::kBFloat16);
+  auto out = at::empty({M, N}, opts);
+  auto zprobe = at::empty({0}, opts);
+  mxa::gemm_mx_bf16(A8.data_ptr<unsigned char>(),
+                    Asc.data_ptr<unsigned char>(),
+                    B8.data_ptr<unsigned char>(),
+                    Bsc.data_ptr<unsigned char>(), bfp_mut(out),
+                    bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                    (int)M, (int)N, (int)K, N, relu,
+                    (const unsigned char*)zp(zprobe), cur_stream());
   return out;
 }
 
@@ -828,6 +868,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_wg", &gemm_wg, "C(f32) = sum_m A[m,:] outer B[m,:]");
   m.def("gemm_tn_fp8", &gemm_tn_fp8);
   m.def("cast_to_fp8", &cast_to_fp8);
+  m.def("mx_cast", &mx_cast);
+  m.def("gemm_mx", &gemm_mx, py::arg("A8"), py::arg("Asc"), py::arg("B8"),
+        py::arg("Bsc"), py::arg("bias") = py::none(),
+        py::arg("relu") = false);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);

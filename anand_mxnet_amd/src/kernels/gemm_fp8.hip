@@ -194,4 +194,217 @@ void cast_to_fp8(const bf16* x, u8* y, long n, const float* scale,
   cast_bf16_to_fp8<<<ew_grid(n / 8), 256, 0, stream>>>(x, y, n / 8, scale);
 }
 
+
+// ---------------------------------------------------------------------------
+// MX block-scaled fp8 GEMM — v_mfma_scale_f32_16x16x128_f8f6f4 (the CDNA4
+// double-rate fp8 path, ≈4.6 PF vs 2.5 PF bf16).  Operand layout probed
+// empirically (tools/probe_mx.hip / probe_mx2.hip):
+//   * lane l, byte 16h+j  ->  element (row l&15, k = h*64 + (l>>4)*16 + j)
+//   * scale operand byte0 of lane l = e8m0 scale of (row l&15, k-block l>>4)
+//   * C/D layout = the usual 16x16 map (col l&15, row (l>>4)*4 + i)
+// Tile 128x128, BK=128 (one scaled MFMA consumes the whole K-tile), 512
+// threads (8 waves as 4Mx2N), glds staging with source-side XOR swizzle.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+
+constexpr int MX_THREADS = 512;
+
+DEVINL int mx_stage_row(int p, int t) {
+  return (((t >> 6) * 2 + p) << 3) + ((t & 63) >> 3);
+}
+
+struct MxLoader {
+  const u8* base;
+  const u8* zpage;
+  int rows;
+  long ld;
+  // device state
+  const u8* rp[2];
+  long koff;
+
+  DEVINL void init(int row0, int t) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int r = row0 + mx_stage_row(p, t);
+      rp[p] = (r < rows) ? base + (long)r * ld : nullptr;
+    }
+  }
+  DEVINL void set_kstep(int ks) { koff = (long)ks * F8_BK; }
+  DEVINL const u8* src(int p, int cg) const {
+    return rp[p] ? rp[p] + koff + cg * 16 : zpage + cg * 16;
+  }
+};
+
+__global__ __launch_bounds__(MX_THREADS, 2) void gemm_mx_core(
+    MxLoader aload, MxLoader bload, const u8* __restrict__ Asc,
+    const u8* __restrict__ Bsc, long lsa, long lsb, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, int ksteps,
+    int relu) {
+  __shared__ __align__(16) u8 smem[2][2][F8_BM * F8_BK];
+
+  const int ntn = (N + F8_BN - 1) / F8_BN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * F8_BM, bn = (bid % ntn) * F8_BN;
+  const int t = threadIdx.x;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  aload.init(bm, t);
+  bload.init(bn, t);
+
+  const int cg = (ll & 7) ^ ((ll >> 3) & 7);  // source chunk (row&7 = ll>>3)
+
+  auto stage_glds = [&](int buf, int ks) {
+    aload.set_kstep(ks);
+    bload.set_kstep(ks);
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      unsigned lbase = __builtin_amdgcn_readfirstlane(
+          (unsigned)((lw * 2 + p) * 1024));
+      auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[buf][0][0] + lbase);
+      auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[buf][1][0] + lbase);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)
+              aload.src(p, cg),
+          ldsA, 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)
+              bload.src(p, cg),
+          ldsB, 16, 0, 0);
+    }
+  };
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int wm = (lw & 3) * 32, wn = (lw >> 2) * 64;
+  const int lrow = ll & 15, lk = ll >> 4;  // fragment row / k-group
+
+  auto compute = [&](int buf, int ks) {
+    int kb0 = ks * 4;
+    i32x8 af[2], bfr[4];
+    int sa[2], sb[4];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+      int row = wm + mf * 16 + lrow;
+      i32x4 lo = *(const i32x4*)&smem[buf][0][f8_lds_off(row, lk)];
+      i32x4 hi = *(const i32x4*)&smem[buf][0][f8_lds_off(row, 4 + lk)];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        af[mf][j] = lo[j];
+        af[mf][4 + j] = hi[j];
+      }
+      int rg = bm + row;
+      sa[mf] = rg < M ? Asc[(long)rg * lsa + kb0 + lk] : 127;
+    }
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int row = wn + nf * 16 + lrow;
+      i32x4 lo = *(const i32x4*)&smem[buf][1][f8_lds_off(row, lk)];
+      i32x4 hi = *(const i32x4*)&smem[buf][1][f8_lds_off(row, 4 + lk)];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        bfr[nf][j] = lo[j];
+        bfr[nf][4 + j] = hi[j];
+      }
+      int rg = bn + row;
+      sb[nf] = rg < N ? Bsc[(long)rg * lsb + kb0 + lk] : 127;
+    }
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af[mf], bfr[nf], acc[mf][nf], 0, 0, 0, sa[mf], 0, sb[nf]);
+  };
+
+  stage_glds(0, 0);
+  __syncthreads();
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    if (ks + 1 < ksteps) stage_glds(buf ^ 1, ks + 1);
+    compute(buf, ks);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wn + nf * 16 + (ll & 15);
+      if (gcol >= N) continue;
+      float bv = bias ? bias[gcol] : 0.f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (ll >> 4) * 4 + i;
+        if (grow >= M) continue;
+        float v = acc[mf][nf][i] + bv;
+        if (relu) v = v > 0.f ? v : 0.f;
+        Cout[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
+// Block-quantize bf16 -> MX fp8: per 32-element K-block, e8m0 scale such
+// that amax/scale fits e4m3 range; one thread per block.
+__global__ __launch_bounds__(256) void mx_cast_kernel(
+    const bf16* __restrict__ x, u8* __restrict__ q, u8* __restrict__ sc,
+    long nblk) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nblk;
+       i += (long)gridDim.x * blockDim.x) {
+    const bf16* xp = x + i * 32;
+    float amax = 0.f;
+    float vals[32];
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      bf16x8 b = *(const bf16x8*)&xp[v * 8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(b[j]);
+        vals[v * 8 + j] = f;
+        amax = fmaxf(amax, fabsf(f));
+      }
+    }
+    int s_exp = 0;
+    if (amax > 0.f) {
+      int e;
+      float m = frexpf(amax, &e);  // amax = m * 2^e, m in [0.5, 1)
+      s_exp = e - (m <= 0.875f ? 9 : 8);  // amax/2^s <= 448
+      if (s_exp < -127) s_exp = -127;
+      if (s_exp > 127) s_exp = 127;
+    }
+    sc[i] = (u8)(s_exp + 127);
+    float inv = ldexpf(1.f, -s_exp);
+    u8 out[32];
+#pragma unroll
+    for (int j = 0; j < 32; ++j)
+      out[j] = __hip_fp8_e4m3(vals[j] * inv).__x;
+#pragma unroll
+    for (int v = 0; v < 4; ++v)
+      *(u64*)&q[i * 32 + v * 8] = *(const u64*)&out[v * 8];
+  }
+}
+
+void gemm_mx_bf16(const u8* A, const u8* Asc, const u8* B, const u8* Bsc,
+                  bf16* C, const float* bias, int M, int N, int K, long ldc,
+                  int relu, const u8* zpage, hipStream_t stream) {
+  MxLoader a{A, zpage, M, K};
+  MxLoader b{B, zpage, N, K};
+  dim3 grid(cdiv_h(M, F8_BM) * cdiv_h(N, F8_BN));
+  gemm_mx_core<<<grid, MX_THREADS, 0, stream>>>(
+      a, b, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, K / F8_BK, relu);
+}
+
+void mx_cast_bf16(const bf16* x, u8* q, u8* sc, long nblk,
+                  hipStream_t stream) {
+  mx_cast_kernel<<<ew_grid(nblk, 256, 4096), 256, 0, stream>>>(x, q, sc,
+                                                               nblk);
+}
+
 }  // namespace mxa

@@ -524,3 +524,35 @@ def test_mha_dropout_path_trains():
     out.sum().backward()
     for t in (q, k, v):
         assert t.grad is not None and torch.isfinite(t.grad.float()).all()
+
+
+def test_mx_fp8_gemm():
+    M, N, K = 256, 512, 768
+    a = _rt((M, K)).to(BF)
+    b = _rt((N, K)).to(BF)
+    a8, asc = _C.mx_cast(a.contiguous())
+    b8, bsc = _C.mx_cast(b.contiguous())
+    out = _C.gemm_mx(a8, asc, b8, bsc)
+    ref = a.float() @ b.float().t()
+    # fp8 e4m3 block quantization: ~3 mantissa bits per element
+    err = (out.float() - ref).abs()
+    rel = (err / (ref.abs() + 1.0)).max().item()
+    assert rel < 0.15, rel
+    # correlation check (quantization noise averages out over K)
+    cos = torch.nn.functional.cosine_similarity(
+        out.float().flatten(), ref.flatten(), dim=0).item()
+    assert cos > 0.995, cos
+
+
+def test_mx_cast_roundtrip():
+    x = _rt((64, 128)).to(BF)
+    q, sc = _C.mx_cast(x.contiguous())
+    assert q.shape == (64, 128) and sc.shape == (64, 4)
+    # reconstruct and compare
+    e = sc.int().float() - 127
+    scale = torch.pow(2.0, e).repeat_interleave(32, dim=1)
+    xf8 = q.view(torch.float8_e4m3fn).float()
+    rec = xf8 * scale
+    err = (rec - x.float()).abs().max().item()
+    amax = x.float().abs().max().item()
+    assert err < 0.08 * amax, (err, amax)

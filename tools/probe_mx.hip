@@ -19,12 +19,17 @@ __global__ void k(const unsigned char* A, const unsigned char* Asc,
                   float* D) {
   int l = threadIdx.x;
   i32x8 a, b;
-  const unsigned char* ap = A + (l & 15) * 128 + (l >> 4) * 32;
-  const unsigned char* bp = B + (l & 15) * 128 + (l >> 4) * 32;
+  // CONFIRMED (probe_mx2): two K-64 halves — lane l, byte 16h+j holds
+  // element (row l&15, k = h*64 + (l>>4)*16 + j); scale byte0 of lane l
+  // is the e8m0 for (row l&15, k-block l>>4).
+  const unsigned char* ap = A + (l & 15) * 128 + (l >> 4) * 16;
+  const unsigned char* bp = B + (l & 15) * 128 + (l >> 4) * 16;
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    a[i] = *(const int*)(ap + i * 4);
-    b[i] = *(const int*)(bp + i * 4);
+  for (int j = 0; j < 4; ++j) {
+    a[j] = *(const int*)(ap + j * 4);
+    a[4 + j] = *(const int*)(ap + 64 + j * 4);
+    b[j] = *(const int*)(bp + j * 4);
+    b[4 + j] = *(const int*)(bp + 64 + j * 4);
   }
   int sa = Asc[(l & 15) * 4 + (l >> 4)];
   int sb = Bsc[(l & 15) * 4 + (l >> 4)];
