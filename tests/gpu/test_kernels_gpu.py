@@ -534,14 +534,15 @@ def test_mx_fp8_gemm():
     b8, bsc = _C.mx_cast(b.contiguous())
     out = _C.gemm_mx(a8, asc, b8, bsc)
     ref = a.float() @ b.float().t()
-    # fp8 e4m3 block quantization: ~3 mantissa bits per element
+    # fp8 e4m3 block quantization: per-product rms err ~0.7%, dot error
+    # std ~ sqrt(K)*0.007 ~ 0.19 here; bound mean/max accordingly and
+    # check correlation (the decisive signal for a layout bug)
     err = (out.float() - ref).abs()
-    rel = (err / (ref.abs() + 1.0)).max().item()
-    assert rel < 0.15, rel
-    # correlation check (quantization noise averages out over K)
+    assert err.mean().item() < 0.5, err.mean().item()
+    assert err.max().item() < 2.5, err.max().item()
     cos = torch.nn.functional.cosine_similarity(
         out.float().flatten(), ref.flatten(), dim=0).item()
-    assert cos > 0.995, cos
+    assert cos > 0.999, cos
 
 
 def test_mx_cast_roundtrip():

@@ -40,6 +40,18 @@ def bench_gemm(M, N, K):
             "rocblas_TF": round(tf_blas, 1), "ms": round(t * 1e3, 3)}
 
 
+def bench_mx(M, N, K):
+    a = (torch.rand((M, K), device=DEV) * 2 - 1).to(BF)
+    b = (torch.rand((N, K), device=DEV) * 2 - 1).to(BF)
+    a8, asc = _C.mx_cast(a.contiguous())
+    b8, bsc = _C.mx_cast(b.contiguous())
+    t = timeit(lambda: _C.gemm_mx(a8, asc, b8, bsc))
+    tf = 2.0 * M * N * K / t / 1e12
+    tc = timeit(lambda: _C.mx_cast(a.contiguous()))
+    return {"op": f"gemm_mx_fp8 {M}x{N}x{K}", "ours_TF": round(tf, 1),
+            "ms": round(t * 1e3, 3), "cast_ms": round(tc * 1e3, 3)}
+
+
 def bench_wg(m, Mo, No):
     a = (torch.rand((m, Mo), device=DEV) * 2 - 1).to(BF)
     b = (torch.rand((m, No), device=DEV) * 2 - 1).to(BF)
@@ -104,6 +116,11 @@ def bench_conv_wg(N, H, W, C, K, R, s, p, name=""):
 
 def main():
     results = []
+    if "--mx-only" in sys.argv:
+        for shape in [(4096, 4096, 4096), (8192, 8192, 8192),
+                      (8192, 4096, 1024), (8192, 1024, 4096)]:
+            print(json.dumps(bench_mx(*shape)))
+        return
     if "--wg-only" in sys.argv:
         results.append(bench_wg(100352, 128, 128))
         results.append(bench_wg(25088, 512, 256))
