@@ -255,6 +255,26 @@ __global__ __launch_bounds__(MX_THREADS, 2) void gemm_mx_core(
 
   const int cg = (ll & 7) ^ ((ll >> 3) & 7);  // source chunk (row&7 = ll>>3)
 
+  const int wm0 = (lw & 3) * 32, wn0 = (lw >> 2) * 64;
+  const int lrow0 = ll & 15, lk0 = ll >> 4;
+  int sa_buf[2][2], sb_buf[2][4];
+
+  // prefetch the step's e8m0 scale bytes along with its DMA — a scale
+  // load issued inside compute() would stall the MFMA on global latency
+  auto stage_scales = [&](int buf, int ks) {
+    int kb0 = ks * 4;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+      int rg = bm + wm0 + mf * 16 + lrow0;
+      sa_buf[buf][mf] = rg < M ? Asc[(long)rg * lsa + kb0 + lk0] : 127;
+    }
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int rg = bn + wn0 + nf * 16 + lrow0;
+      sb_buf[buf][nf] = rg < N ? Bsc[(long)rg * lsb + kb0 + lk0] : 127;
+    }
+  };
+
   auto stage_glds = [&](int buf, int ks) {
     aload.set_kstep(ks);
     bload.set_kstep(ks);
@@ -275,6 +295,7 @@ __global__ __launch_bounds__(MX_THREADS, 2) void gemm_mx_core(
               bload.src(p, cg),
           ldsB, 16, 0, 0);
     }
+    stage_scales(buf, ks);
   };
 
   f32x4 acc[2][4];
@@ -286,10 +307,8 @@ __global__ __launch_bounds__(MX_THREADS, 2) void gemm_mx_core(
   const int wm = (lw & 3) * 32, wn = (lw >> 2) * 64;
   const int lrow = ll & 15, lk = ll >> 4;  // fragment row / k-group
 
-  auto compute = [&](int buf, int ks) {
-    int kb0 = ks * 4;
+  auto compute = [&](int buf) {
     i32x8 af[2], bfr[4];
-    int sa[2], sb[4];
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf) {
       int row = wm + mf * 16 + lrow;
@@ -300,8 +319,6 @@ __global__ __launch_bounds__(MX_THREADS, 2) void gemm_mx_core(
         af[mf][j] = lo[j];
         af[mf][4 + j] = hi[j];
       }
-      int rg = bm + row;
-      sa[mf] = rg < M ? Asc[(long)rg * lsa + kb0 + lk] : 127;
     }
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
@@ -313,15 +330,16 @@ __global__ __launch_bounds__(MX_THREADS, 2) void gemm_mx_core(
         bfr[nf][j] = lo[j];
         bfr[nf][4 + j] = hi[j];
       }
-      int rg = bn + row;
-      sb[nf] = rg < N ? Bsc[(long)rg * lsb + kb0 + lk] : 127;
     }
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf)
         acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-            af[mf], bfr[nf], acc[mf][nf], 0, 0, 0, sa[mf], 0, sb[nf]);
+            af[mf], bfr[nf], acc[mf][nf], 0, 0, 0, sa_buf[buf][mf], 0,
+            sb_buf[buf][nf]);
+    __builtin_amdgcn_s_setprio(0);
   };
 
   stage_glds(0, 0);
@@ -329,7 +347,7 @@ __global__ __launch_bounds__(MX_THREADS, 2) void gemm_mx_core(
   for (int ks = 0; ks < ksteps; ++ks) {
     int buf = ks & 1;
     if (ks + 1 < ksteps) stage_glds(buf ^ 1, ks + 1);
-    compute(buf, ks);
+    compute(buf);
     __syncthreads();
   }
 
