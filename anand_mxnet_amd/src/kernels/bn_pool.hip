@@ -397,39 +397,65 @@ __global__ __launch_bounds__(256) void maxpool_bwd(
 
 // Materialized im2col for the small-C stem conv (7x7 C=3): writes the
 // padded [N*P*Q, kdim_pad] GEMM operand directly (k order (r, s, c),
-// zeros in the %64 pad tail and outside the image) — replaces a chain of
-// torch pad + advanced-indexing + pad + contiguous full-tensor copies.
+// zeros in the %64 pad tail and outside the image).  One block per
+// output row (n, p): the R input rows are staged once through LDS
+// (x read once per (n,p) instead of once per output element) and the
+// whole [Q][kpadded] slab is emitted with bf16x8 stores.
 __global__ __launch_bounds__(256) void im2col_stem(
     const bf16* __restrict__ x, bf16* __restrict__ out, int Nb, int H,
     int W, int C, int P, int Q, int R, int S, int sh, int sw, int ph,
     int pw, int dh, int dw, int kdim, int kpadded) {
-  int kv = kpadded / 8;
-  long total = (long)Nb * P * Q * kv;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
-    int j0 = (int)(i % kv) * 8;
-    long npq = i / kv;
-    int q = (int)(npq % Q);
-    int p = (int)((npq / Q) % P);
-    int n = (int)(npq / ((long)P * Q));
-    int ih0 = p * sh - ph, iw0 = q * sw - pw;
-    const bf16* xn = x + (long)n * H * W * C;
-    bf16x8 o;
-#pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      int k = j0 + e;
-      float v = 0.f;
-      if (k < kdim) {
-        int c = k % C;
-        int rs = k / C;
-        int sx = rs % S, r = rs / S;
-        int ih = ih0 + r * dh, iw = iw0 + sx * dw;
-        if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-          v = bf2f(xn[((long)ih * W + iw) * C + c]);
+  extern __shared__ bf16 xrow[];  // [R][W*C]
+  const int WC = W * C;
+  for (long np = blockIdx.x; np < (long)Nb * P; np += gridDim.x) {
+    int p = (int)(np % P);
+    int n = (int)(np / P);
+    int ih0 = p * sh - ph;
+    // stage R input rows (u32 copies, zero-fill OOB rows)
+    const bf16* xn = x + (long)n * H * WC;
+    int wcw = (WC + 1) / 2;  // u32 words per row
+    for (int idx = threadIdx.x; idx < R * wcw; idx += 256) {
+      int r = idx / wcw, wd = idx % wcw;
+      int ih = ih0 + r * dh;
+      u32 v = 0;
+      if (ih >= 0 && ih < H) {
+        const bf16* rowp = xn + (long)ih * WC;
+        if (wd * 2 + 1 < WC) {
+          v = *(const u32*)(rowp + wd * 2);
+        } else {  // odd-WC tail: avoid reading past the row
+          u32 lo = *(const unsigned short*)(rowp + wd * 2);
+          v = lo;
+        }
       }
-      o[e] = f2bf(v);
+      ((u32*)&xrow[0])[r * wcw + wd] = v;
     }
-    *(bf16x8*)&out[npq * kpadded + j0] = o;
+    __syncthreads();
+    // emit [Q][kpadded]
+    bf16* orow = out + np * (long)Q * kpadded;
+    int kv = kpadded / 8;
+    for (int i = threadIdx.x; i < Q * kv; i += 256) {
+      int q = i / kv;
+      int j0 = (i % kv) * 8;
+      int iw0 = q * sw - pw;
+      bf16x8 o;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int k = j0 + e;
+        float v = 0.f;
+        if (k < kdim) {
+          int c = k % C;
+          int rs = k / C;
+          int sx = rs % S, r = rs / S;
+          int iw = iw0 + sx * dw;
+          int ih = ih0 + r * dh;
+          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+            v = bf2f(xrow[r * WC + iw * C + c]);
+        }
+        o[e] = f2bf(v);
+      }
+      *(bf16x8*)&orow[(long)q * kpadded + j0] = o;
+    }
+    __syncthreads();
   }
 }
 
@@ -610,9 +636,12 @@ void im2col_stem_bf16(const bf16* x, bf16* out, int Nb, int H, int W,
                       int C, int P, int Q, int R, int S, int sh, int sw,
                       int ph, int pw, int dh, int dw, int kdim, int kpadded,
                       hipStream_t s) {
-  dim3 g = ew_grid((long)Nb * P * Q * (kpadded / 8), 256, 8192);
-  im2col_stem<<<g, 256, 0, s>>>(x, out, Nb, H, W, C, P, Q, R, S, sh, sw,
-                                ph, pw, dh, dw, kdim, kpadded);
+  long blocks = (long)Nb * P;
+  if (blocks > 16384) blocks = 16384;
+  int lds = R * (((W * C + 1) / 2) * 2) * sizeof(bf16) + 16;
+  im2col_stem<<<dim3((unsigned)blocks), 256, lds, s>>>(
+      x, out, Nb, H, W, C, P, Q, R, S, sh, sw, ph, pw, dh, dw, kdim,
+      kpadded);
 }
 
 void maxpool_nhwc_bwd(const bf16* dy, const unsigned char* arg, bf16* dx,
