@@ -557,3 +557,51 @@ def test_mx_cast_roundtrip():
     err = (rec - x.float()).abs().max().item()
     amax = x.float().abs().max().item()
     assert err < 0.08 * amax, (err, amax)
+
+
+def test_gemm_shape_fuzz():
+    """Randomized shape sweep over the three GEMM forms (dispatch edges:
+    tile tails, tn8 threshold, split-K, 64-wide NN)."""
+    import random
+    rng = random.Random(11)
+    for _ in range(12):
+        M = rng.choice([8, 24, 100, 128, 129, 256, 300, 512])
+        N = rng.choice([8, 16, 64, 72, 128, 200, 256, 512])
+        K = rng.choice([64, 128, 192, 512])
+        a = _rt((M, K)).to(BF).float()
+        b = _rt((N, K)).to(BF).float()
+        out = _C.gemm_tn(a.to(BF).contiguous(), b.to(BF).contiguous(),
+                         None, False)
+        assert_close(out, a @ b.t(), msg=f"tn {M}x{N}x{K}")
+        if N % 8 == 0:
+            bn = _rt((K, N)).to(BF).float()
+            out2 = _C.gemm_nn(a.to(BF).contiguous(),
+                              bn.to(BF).contiguous())
+            assert_close(out2, a @ bn, msg=f"nn {M}x{N}x{K}")
+        if M % 8 == 0 and N % 8 == 0:
+            aw = _rt((K, M)).to(BF).float()
+            bw = _rt((K, N)).to(BF).float()
+            outw = _C.gemm_wg(aw.to(BF).contiguous(),
+                              bw.to(BF).contiguous())
+            assert_close(outw, aw.t() @ bw, rtol=3e-2, atol=5e-2,
+                         msg=f"wg {M}x{N}x{K}")
+
+
+def test_conv_shape_fuzz():
+    import random
+    rng = random.Random(5)
+    for _ in range(6):
+        C = rng.choice([64, 128])
+        K = rng.choice([64, 128, 192])
+        H = rng.choice([7, 9, 14, 17])
+        R = rng.choice([1, 3])
+        s_ = rng.choice([1, 2])
+        p_ = 1 if R == 3 else 0
+        N = rng.choice([1, 2, 3])
+        x = _rt((N, H, H, C)).to(BF).float()
+        w = _rt((K, R, R, C), 0.2).to(BF).float()
+        ref = _conv_ref(x, w, (s_, s_), (p_, p_))
+        w_rskc = w.to(BF).permute(1, 2, 0, 3).contiguous()
+        y = _C.conv2d_fwd(x.to(BF).contiguous(), w_rskc, None,
+                          [s_, s_], [p_, p_], [1, 1], False)
+        assert_close(y, ref, msg=f"conv fuzz {N}x{H}x{C}->K{K} r{R}s{s_}")

@@ -190,3 +190,41 @@ def test_transformer_causal_masking_gpu():
     v = q.asnumpy().reshape(2, 16, 2, 64)
     o = out.asnumpy().reshape(2, 16, 2, 64)
     assert np.abs(o[:, 0] - v[:, 0]).max() < 0.05
+
+
+@pytest.mark.parametrize("family", ["vgg11", "alexnet", "densenet121",
+                                    "squeezenet1_0", "inception_v3",
+                                    "mobilenet_v1"])
+def test_model_zoo_families_train_step(family):
+    """Every vision model-zoo family does one fwd+bwd+step on the GPU."""
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import gluon, nd, autograd
+    from anand_mxnet_amd.gluon.model_zoo import vision as zoo
+
+    factory = {
+        "vgg11": lambda: zoo.vgg11(),
+        "alexnet": lambda: zoo.alexnet(),
+        "densenet121": lambda: zoo.densenet121(),
+        "squeezenet1_0": lambda: zoo.squeezenet1_0(),
+        "inception_v3": lambda: zoo.inception_v3(),
+        "mobilenet_v1": lambda: zoo.mobilenet1_0(),
+    }[family]
+    ctx = mx.gpu(0)
+    net = factory()
+    net.cast("bfloat16")
+    net.initialize(mx.init.Xavier(), ctx=ctx)
+    loss_fn = gluon.loss.SoftmaxCrossEntropyLoss()
+    trainer = gluon.Trainer(net.collect_params(), "sgd",
+                            {"learning_rate": 0.01,
+                             "multi_precision": True})
+    size = 299 if family == "inception_v3" else 128
+    x = nd.random.uniform(-1, 1, shape=(2, 3, size, size),
+                          dtype="bfloat16", ctx=ctx)
+    y = nd.array(np.random.randint(0, 1000, (2,)), ctx=ctx)
+    with autograd.record():
+        out = net(x)
+        loss = loss_fn(out, y)
+    loss.backward()
+    trainer.step(2)
+    v = float(loss.mean().asscalar())
+    assert np.isfinite(v), v
