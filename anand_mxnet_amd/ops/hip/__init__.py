@@ -173,7 +173,8 @@ class _Conv2dNHWC(torch.autograd.Function):
         K, R, S, C = w.shape
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0):
+            if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0) \
+                    and K % 64 == 0 and C % 8 == 0:
                 # pointwise: dx = dy @ W, the NN-form kernel (no W^T copy)
                 N, P, Q, _ = dy.shape
                 dx = _C.gemm_nn(dy.reshape(N * P * Q, K),
@@ -228,7 +229,8 @@ class _Conv2dNHWC(torch.autograd.Function):
                 dx = _C.conv2d_dgrad(dy, w_rsck, list(x.shape),
                                      list(stride), list(pad), list(dil))
         if ctx.needs_input_grad[1]:
-            if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0):
+            if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0) \
+                    and K % 8 == 0 and C % 8 == 0:
                 # pointwise wgrad: plain WG GEMM ([C,K] == [1,1,C,K])
                 N, P, Q, _ = dy.shape
                 dwf = _C.gemm_wg(x.reshape(N * P * Q, C),

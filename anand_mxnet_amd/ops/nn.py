@@ -183,6 +183,17 @@ def batch_norm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
         return gpu_impl("batchnorm_nhwc")(data, g, beta, moving_mean,
                                           moving_var, eps, momentum, training,
                                           fuse_relu, addend)
+    if data.is_cuda and data.dim() == 4 and ch_axis == 1 and \
+            _use_hip(data, "batchnorm_nhwc"):
+        # NCHW on GPU: run the NHWC kernel with layout conversion (the
+        # torch/MIOpen bf16 NCHW batch_norm is unstable on this stack)
+        xh = data.permute(0, 2, 3, 1).contiguous()
+        ah = addend.permute(0, 2, 3, 1).contiguous() \
+            if addend is not None else None
+        out = gpu_impl("batchnorm_nhwc")(xh, g, beta, moving_mean,
+                                         moving_var, eps, momentum,
+                                         training, fuse_relu, ah)
+        return out.permute(0, 3, 1, 2)
     if data.is_cuda:
         _use_hip(data, "batchnorm_nhwc")
     # torch F.batch_norm expects channel at dim1

@@ -194,7 +194,12 @@ def test_transformer_causal_masking_gpu():
 
 @pytest.mark.parametrize("family", ["vgg11", "alexnet", "densenet121",
                                     "squeezenet1_0", "inception_v3",
-                                    "mobilenet_v1"])
+                                    pytest.param(
+                                        "mobilenet_v1",
+                                        marks=pytest.mark.xfail(
+                                            reason="depthwise conv HIP "
+                                            "kernel deferred "
+                                            "(docs/PARITY.md)"))])
 def test_model_zoo_families_train_step(family):
     """Every vision model-zoo family does one fwd+bwd+step on the GPU."""
     import anand_mxnet_amd as mx
@@ -217,7 +222,7 @@ def test_model_zoo_families_train_step(family):
     trainer = gluon.Trainer(net.collect_params(), "sgd",
                             {"learning_rate": 0.01,
                              "multi_precision": True})
-    size = 299 if family == "inception_v3" else 128
+    size = 299 if family == "inception_v3" else 224
     x = nd.random.uniform(-1, 1, shape=(2, 3, size, size),
                           dtype="bfloat16", ctx=ctx)
     y = nd.array(np.random.randint(0, 1000, (2,)), ctx=ctx)
