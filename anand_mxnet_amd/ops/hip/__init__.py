@@ -226,8 +226,21 @@ class _Conv2dNHWC(torch.autograd.Function):
                 dx[:, ::stride[0], ::stride[1], :] = sub.reshape(N, P, Q, C)
             else:
                 w_rsck = w.permute(1, 2, 3, 0).contiguous()
-                dx = _C.conv2d_dgrad(dy, w_rsck, list(x.shape),
-                                     list(stride), list(pad), list(dil))
+                if K % 64 != 0:
+                    # dgrad reduces over K: zero-pad dy/w channels to the
+                    # tile width (zeros contribute nothing)
+                    Kp = (K + 63) // 64 * 64
+                    dy_p = torch.nn.functional.pad(dy, (0, Kp - K)) \
+                        .contiguous()
+                    w_p = torch.nn.functional.pad(w_rsck, (0, Kp - K)) \
+                        .contiguous()
+                    dx = _C.conv2d_dgrad(dy_p, w_p, list(x.shape),
+                                         list(stride), list(pad),
+                                         list(dil))
+                else:
+                    dx = _C.conv2d_dgrad(dy, w_rsck, list(x.shape),
+                                         list(stride), list(pad),
+                                         list(dil))
         if ctx.needs_input_grad[1]:
             if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0) \
                     and K % 8 == 0 and C % 8 == 0:
@@ -269,6 +282,9 @@ class _Conv2dIm2col(torch.autograd.Function):
             B = torch.nn.functional.pad(B, (0, kpad))
         ctx.save_for_backward(A, w)
         ctx.dims = (N, P, Q, K, kdim)
+        ctx.xshape = tuple(x.shape)
+        ctx.rs = (R, S)
+        ctx.conf = (stride, pad, dil)
         ctx.has_bias = bias is not None
         y = _C.gemm_tn(A, B.contiguous(),
                        bias.float() if bias is not None else None, False)
@@ -285,11 +301,23 @@ class _Conv2dIm2col(torch.autograd.Function):
             dw = dwf[:, :kdim].reshape(w.shape).to(w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy2)
+        dx = None
         if ctx.needs_input_grad[0]:
-            raise RuntimeError(
-                "im2col conv path has no dgrad (stem layer input needs no "
-                "gradient); use C%64 shapes for inner convs")
-        return None, dw, db, None, None, None
+            # col2im dgrad (inner small-C convs, e.g. Inception branches):
+            # dcols = dy @ W, scatter-add via F.fold.  Our k-order is
+            # (r, s, c); fold expects (c, r, s).
+            stride, pad, dil = ctx.conf
+            N2, H2, W2, C2 = ctx.xshape
+            R2, S2 = ctx.rs
+            dcols = torch.matmul(dy2.float(),
+                                 w.reshape(K, kdim).float())
+            dcols = dcols.reshape(N2, P * Q, R2, S2, C2) \
+                .permute(0, 4, 2, 3, 1).reshape(N2, C2 * R2 * S2, P * Q)
+            dxn = torch.nn.functional.fold(
+                dcols, (H2, W2), (R2, S2), dilation=tuple(dil),
+                padding=tuple(pad), stride=tuple(stride))  # [N, C, H, W]
+            dx = dxn.permute(0, 2, 3, 1).contiguous().to(dy.dtype)
+        return dx, dw, db, None, None, None
 
 
 @register_gpu("conv2d_nhwc")
