@@ -320,12 +320,48 @@ class _Conv2dIm2col(torch.autograd.Function):
         return dx, dw, db, None, None, None
 
 
+class _DwConvNHWC(torch.autograd.Function):
+    """Depthwise conv (groups == C), weight (C, R, S, 1) -> [R,S,C]."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pad, dil):
+        x = x.contiguous()
+        w_rsc = w[:, :, :, 0].permute(1, 2, 0).contiguous()  # [R,S,C]
+        ctx.save_for_backward(x, w_rsc)
+        ctx.conf = (stride, pad, dil, w.shape)
+        ctx.has_bias = bias is not None
+        return _C.dwconv_fwd(x, w_rsc,
+                             bias.float() if bias is not None else None,
+                             list(stride), list(pad), list(dil))
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w_rsc = ctx.saved_tensors
+        stride, pad, dil, wshape = ctx.conf
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = _C.dwconv_dgrad(dy, w_rsc, list(x.shape), list(stride),
+                                 list(pad))
+        if ctx.needs_input_grad[1]:
+            dwf = _C.dwconv_wgrad(x, dy, [w_rsc.shape[0], w_rsc.shape[1]],
+                                  list(stride), list(pad))  # [R,S,C] f32
+            dw = dwf.permute(2, 0, 1).unsqueeze(-1).to(x.dtype)                 .contiguous()
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = _C.colsum(dy)
+        return dx, dw, db, None, None, None
+
+
 @register_gpu("conv2d_nhwc")
 def conv2d_nhwc(x, w, bias, stride, pad, dil, groups):
     if groups != 1:
+        C_in = x.shape[3]
+        if groups == C_in and w.shape[0] == C_in and w.shape[3] == 1                 and C_in % 8 == 0 and tuple(dil) == (1, 1):
+            return _DwConvNHWC.apply(x, w, bias, tuple(stride),
+                                     tuple(pad), tuple(dil))
         raise NotImplementedError(
-            "grouped/depthwise conv HIP kernel not implemented yet "
-            "(docs/PARITY.md)")
+            "grouped conv HIP kernel: only depthwise (groups == C) "
+            "is implemented (docs/PARITY.md)")
     stride, pad, dil = tuple(stride), tuple(pad), tuple(dil)
     if w.shape[3] % 64 == 0:
         return _Conv2dNHWC.apply(x, w, bias, stride, pad, dil)

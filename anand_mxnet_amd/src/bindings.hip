@@ -88,6 +88,15 @@ void colsum_bf16(const bf16*, float*, long, int, hipStream_t);
 void im2col_stem_bf16(const bf16*, bf16*, int, int, int, int, int, int,
                       int, int, int, int, int, int, int, int, int, int,
                       hipStream_t);
+void dwconv_fwd_bf16(const bf16*, const bf16*, const float*, bf16*, int,
+                     int, int, int, int, int, int, int, int, int, int,
+                     int, int, int, hipStream_t);
+void dwconv_dgrad_bf16(const bf16*, const bf16*, bf16*, int, int, int,
+                       int, int, int, int, int, int, int, int, int,
+                       hipStream_t);
+void dwconv_wgrad_bf16(const bf16*, const bf16*, float*, int, int, int,
+                       int, int, int, int, int, int, int, int, int,
+                       hipStream_t);
 // gemm_nn.hip / attention entry points (gemm_tn.hip, gemm_wg.hip)
 void gemm_nn_bf16(const bf16*, ZMap, long, int, const bf16*, ZMap, long,
                   int, bf16*, ZMap, long, int, int, const bf16*,
@@ -662,6 +671,53 @@ at::Tensor attn_wg(const at::Tensor& p, const at::Tensor& q, long H) {
   return out;
 }
 
+at::Tensor dwconv_fwd(const at::Tensor& x, const at::Tensor& w,
+                      const c10::optional<at::Tensor>& bias,
+                      std::vector<long> stride, std::vector<long> pad,
+                      std::vector<long> dil) {
+  // x [N,H,W,C], w [R,S,C] (depthwise, groups == C)
+  TORCH_CHECK(x.dim() == 4 && w.dim() == 3 && x.size(3) == w.size(2));
+  TORCH_CHECK(x.size(3) % 8 == 0, "dwconv: C % 8 != 0");
+  int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int R = w.size(0), S = w.size(1);
+  int sh = stride[0], sw = stride[1], ph = pad[0], pw = pad[1];
+  int dh = dil[0], dw_ = dil[1];
+  int P = (H + 2 * ph - (R - 1) * dh - 1) / sh + 1;
+  int Q = (W + 2 * pw - (S - 1) * dw_ - 1) / sw + 1;
+  auto y = at::empty({Nb, P, Q, C}, x.options());
+  mxa::dwconv_fwd_bf16(bfp(x), bfp(w),
+                       bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                       bfp_mut(y), Nb, H, W, C, P, Q, R, S, sh, sw, ph, pw,
+                       dh, dw_, cur_stream());
+  return y;
+}
+
+at::Tensor dwconv_dgrad(const at::Tensor& dy, const at::Tensor& w,
+                        std::vector<long> xshape, std::vector<long> stride,
+                        std::vector<long> pad) {
+  int Nb = xshape[0], H = xshape[1], W = xshape[2], C = xshape[3];
+  int P = dy.size(1), Q = dy.size(2);
+  int R = w.size(0), S = w.size(1);
+  auto dx = at::empty({Nb, H, W, C}, dy.options());
+  mxa::dwconv_dgrad_bf16(bfp(dy), bfp(w), bfp_mut(dx), Nb, H, W, C, P, Q,
+                         R, S, stride[0], stride[1], pad[0], pad[1],
+                         cur_stream());
+  return dx;
+}
+
+at::Tensor dwconv_wgrad(const at::Tensor& x, const at::Tensor& dy,
+                        std::vector<long> rs, std::vector<long> stride,
+                        std::vector<long> pad) {
+  int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int P = dy.size(1), Q = dy.size(2);
+  int R = rs[0], S = rs[1];
+  auto dwf = at::empty({R, S, C}, x.options().dtype(at::kFloat));
+  mxa::dwconv_wgrad_bf16(bfp(x), bfp(dy), dwf.data_ptr<float>(), Nb, H, W,
+                         C, P, Q, R, S, stride[0], stride[1], pad[0],
+                         pad[1], cur_stream());
+  return dwf;
+}
+
 at::Tensor im2col_pad(const at::Tensor& x, std::vector<long> rs,
                       std::vector<long> stride, std::vector<long> pad,
                       std::vector<long> dil, long kpadded) {
@@ -898,6 +954,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("colsum", &colsum);
   m.def("im2col_pad", &im2col_pad);
+  m.def("dwconv_fwd", &dwconv_fwd);
+  m.def("dwconv_dgrad", &dwconv_dgrad);
+  m.def("dwconv_wgrad", &dwconv_wgrad);
   m.def("gemm_nn", &gemm_nn);
   m.def("attn_scores", &attn_scores);
   m.def("attn_ctx", &attn_ctx);

@@ -675,4 +675,160 @@ void gap_nhwc_bwd(const bf16* dy, bf16* dx, int Nb, long HW, int C,
   gap_bwd<<<g, 256, 0, s>>>(dy, dx, Nb, HW, C);
 }
 
+
+// ---------------------------------------------------------------------------
+// Depthwise conv (groups == C), NHWC bf16: channels are the fast dim, so
+// each thread owns 8 channels of one output pixel and walks the RxS
+// window (reference parity: depthwise_convolution.cu).  Memory-bound.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void dwconv_fwd(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,  // w [R,S,C]
+    const float* __restrict__ bias, bf16* __restrict__ y, int Nb, int H,
+    int W, int C, int P, int Q, int R, int S, int sh, int sw, int ph,
+    int pw, int dh, int dw_) {
+  int cv = C / 8;
+  long total = (long)Nb * P * Q * cv;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    long npq = i / cv;
+    int q = (int)(npq % Q);
+    int p = (int)((npq / Q) % P);
+    int n = (int)(npq / ((long)P * Q));
+    float acc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = bias ? bias[c0 + j] : 0.f;
+    for (int r = 0; r < R; ++r) {
+      int ih = p * sh - ph + r * dh;
+      if (ih < 0 || ih >= H) continue;
+      for (int s = 0; s < S; ++s) {
+        int iw = q * sw - pw + s * dw_;
+        if (iw < 0 || iw >= W) continue;
+        bf16x8 xv = *(const bf16x8*)&x[(((long)n * H + ih) * W + iw) * C
+                                       + c0];
+        bf16x8 wv = *(const bf16x8*)&w[((long)r * S + s) * C + c0];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += bf2f(xv[j]) * bf2f(wv[j]);
+      }
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[j]);
+    *(bf16x8*)&y[npq * C + c0] = o;
+  }
+}
+
+// dgrad: gather form over covering output windows (cf. maxpool_bwd)
+__global__ __launch_bounds__(256) void dwconv_dgrad(
+    const bf16* __restrict__ dy, const bf16* __restrict__ w,
+    bf16* __restrict__ dx, int Nb, int H, int W, int C, int P, int Q,
+    int R, int S, int sh, int sw, int ph, int pw) {
+  int cv = C / 8;
+  long total = (long)Nb * H * W * cv;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % cv) * 8;
+    long nhw = i / cv;
+    int w_ = (int)(nhw % W);
+    int h = (int)((nhw / W) % H);
+    int n = (int)(nhw / ((long)H * W));
+    float acc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+    int p_lo = (h + ph - R + sh) >= 0 ? (h + ph - R + sh) / sh : 0;
+    int p_hi = min((h + ph) / sh, P - 1);
+    int q_lo = (w_ + pw - S + sw) >= 0 ? (w_ + pw - S + sw) / sw : 0;
+    int q_hi = min((w_ + pw) / sw, Q - 1);
+    for (int p = p_lo; p <= p_hi; ++p) {
+      int r = h + ph - p * sh;
+      if (r < 0 || r >= R) continue;
+      for (int q = q_lo; q <= q_hi; ++q) {
+        int sx = w_ + pw - q * sw;
+        if (sx < 0 || sx >= S) continue;
+        bf16x8 g = *(const bf16x8*)&dy[(((long)n * P + p) * Q + q) * C
+                                       + c0];
+        bf16x8 wv = *(const bf16x8*)&w[((long)r * S + sx) * C + c0];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += bf2f(g[j]) * bf2f(wv[j]);
+      }
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[j]);
+    *(bf16x8*)&dx[(((long)n * H + h) * W + w_) * C + c0] = o;
+  }
+}
+
+// wgrad: per-(r,s,c) reduce over (n,p,q); stats-partial geometry + atomics
+__global__ __launch_bounds__(256) void dwconv_wgrad(
+    const bf16* __restrict__ x, const bf16* __restrict__ dy,
+    float* __restrict__ dwf, int Nb, int H, int W, int C, int P, int Q,
+    int R, int S, int sh, int sw, int ph, int pw) {
+  // block: 256 thr = 32 row-lanes x 8 chunks; grid.x = rs * C/64 blocks
+  const int nc64 = (C + 63) / 64;
+  const int rs = blockIdx.x / nc64;
+  const int cblk = (blockIdx.x % nc64) * 64;
+  const int r = rs / S, s = rs % S;
+  const int t = threadIdx.x;
+  const int chunk = t & 7, rlane = t >> 3;
+  const int c0 = cblk + chunk * 8;
+  const bool active = c0 + 8 <= C;
+  long M = (long)Nb * P * Q;
+  float acc[8] = {0};
+  for (long m = blockIdx.y * 32 + rlane; active && m < M;
+       m += 32L * gridDim.y) {
+    int q = (int)(m % Q);
+    int p = (int)((m / Q) % P);
+    int n = (int)(m / ((long)P * Q));
+    int ih = p * sh - ph + r;
+    int iw = q * sw - pw + s;
+    if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
+    bf16x8 xv = *(const bf16x8*)&x[(((long)n * H + ih) * W + iw) * C + c0];
+    bf16x8 g = *(const bf16x8*)&dy[m * C + c0];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf2f(xv[j]) * bf2f(g[j]);
+  }
+  __shared__ float red[32][64];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[rlane][chunk * 8 + j] = acc[j];
+  __syncthreads();
+  if (t < 64) {
+    float f = 0;
+#pragma unroll
+    for (int rr = 0; rr < 32; ++rr) f += red[rr][t];
+    int c = cblk + t;
+    if (c < C) atomicAdd(&dwf[(long)rs * C + c], f);
+  }
+}
+
+void dwconv_fwd_bf16(const bf16* x, const bf16* w, const float* bias,
+                     bf16* y, int Nb, int H, int W, int C, int P, int Q,
+                     int R, int S, int sh, int sw, int ph, int pw, int dh,
+                     int dw_, hipStream_t s) {
+  dim3 g = ew_grid((long)Nb * P * Q * (C / 8), 256, 8192);
+  dwconv_fwd<<<g, 256, 0, s>>>(x, w, bias, y, Nb, H, W, C, P, Q, R, S, sh,
+                               sw, ph, pw, dh, dw_);
+}
+
+void dwconv_dgrad_bf16(const bf16* dy, const bf16* w, bf16* dx, int Nb,
+                       int H, int W, int C, int P, int Q, int R, int S,
+                       int sh, int sw, int ph, int pw, hipStream_t s) {
+  dim3 g = ew_grid((long)Nb * H * W * (C / 8), 256, 8192);
+  dwconv_dgrad<<<g, 256, 0, s>>>(dy, w, dx, Nb, H, W, C, P, Q, R, S, sh,
+                                 sw, ph, pw);
+}
+
+void dwconv_wgrad_bf16(const bf16* x, const bf16* dy, float* dwf, int Nb,
+                       int H, int W, int C, int P, int Q, int R, int S,
+                       int sh, int sw, int ph, int pw, hipStream_t s) {
+  int nc64 = (C + 63) / 64;
+  long M = (long)Nb * P * Q;
+  long gy = min((M + 31) / 32, (long)(1024 / (R * S * nc64) + 1));
+  if (gy < 1) gy = 1;
+  HIP_CHECK(hipMemsetAsync(dwf, 0, sizeof(float) * R * S * C, s));
+  dwconv_wgrad<<<dim3(R * S * nc64, (unsigned)gy), 256, 0, s>>>(
+      x, dy, dwf, Nb, H, W, C, P, Q, R, S, sh, sw, ph, pw);
+}
+
 }  // namespace mxa

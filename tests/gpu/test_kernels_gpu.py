@@ -605,3 +605,29 @@ def test_conv_shape_fuzz():
         y = _C.conv2d_fwd(x.to(BF).contiguous(), w_rskc, None,
                           [s_, s_], [p_, p_], [1, 1], False)
         assert_close(y, ref, msg=f"conv fuzz {N}x{H}x{C}->K{K} r{R}s{s_}")
+
+
+def test_depthwise_conv_fwd_bwd():
+    N, H, W, C, R = 2, 14, 14, 32, 3
+    for s_ in (1, 2):
+        x0 = _rt((N, H, W, C)).to(BF)
+        w0 = _rt((C, R, R, 1), 0.3).to(BF)
+        x1 = x0.clone().requires_grad_(True)
+        w1 = w0.clone().requires_grad_(True)
+        from anand_mxnet_amd.ops.hip import conv2d_nhwc
+        y1 = conv2d_nhwc(x1, w1, None, (s_, s_), (1, 1), (1, 1), C)
+        g = _rt(tuple(y1.shape)).to(BF)
+        y1.backward(g)
+
+        x2 = x0.clone().float().requires_grad_(True)
+        w2 = w0.clone().float().requires_grad_(True)
+        xn = x2.permute(0, 3, 1, 2)
+        wn = w2.permute(0, 3, 1, 2)  # (C,1,R,R)
+        y2 = torch.nn.functional.conv2d(xn, wn, None, (s_, s_), (1, 1),
+                                        groups=C).permute(0, 2, 3, 1)
+        y2.backward(g.float())
+        assert_close(y1, y2, msg=f"dw fwd s{s_}")
+        assert_close(x1.grad, x2.grad, rtol=4e-2, atol=5e-2,
+                     msg=f"dw dx s{s_}")
+        assert_close(w1.grad, w2.grad, rtol=4e-2, atol=6e-2,
+                     msg=f"dw dw s{s_}")

@@ -194,12 +194,7 @@ def test_transformer_causal_masking_gpu():
 
 @pytest.mark.parametrize("family", ["vgg11", "alexnet", "densenet121",
                                     "squeezenet1_0", "inception_v3",
-                                    pytest.param(
-                                        "mobilenet_v1",
-                                        marks=pytest.mark.xfail(
-                                            reason="depthwise conv HIP "
-                                            "kernel deferred "
-                                            "(docs/PARITY.md)"))])
+                                    "mobilenet_v1"])
 def test_model_zoo_families_train_step(family):
     """Every vision model-zoo family does one fwd+bwd+step on the GPU."""
     import anand_mxnet_amd as mx
