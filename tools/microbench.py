@@ -134,7 +134,8 @@ def main():
             print(json.dumps(r))
         return
     for shape in [(4096, 4096, 4096), (8192, 8192, 8192),
-                  (12544, 256, 1152), (3136, 512, 2304), (25088, 128, 576)]:
+                  (12544, 256, 1152), (3136, 512, 2304), (25088, 128, 576),
+                  (8192, 768, 768), (8192, 3072, 768), (8192, 768, 3072)]:
         results.append(bench_gemm(*shape))
     results.append(bench_wg(100352, 128, 128))
     results.append(bench_wg(25088, 512, 256))
