@@ -17,6 +17,9 @@ namespace mxa {
 void gemm_tn_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
                   int, long, long, long, long, long, long, int, bool,
                   const bf16*, hipStream_t);
+void gemm_tn_splitk_bf16(const bf16*, const bf16*, bf16*, float*,
+                         const float*, int, int, int, bool, const bf16*,
+                         hipStream_t);
 void conv2d_fwd_bf16(const bf16*, const bf16*, bf16*, float*, const float*,
                      int, int, int, int, int, int, int, int, int, int, int,
                      int, int, int, int, bool, const bf16*, hipStream_t);
@@ -198,9 +201,18 @@ at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
                        (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
                        cur_stream());
   } else {
-    mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
-                      (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
-                      cur_stream());
+    long blocks = ((M + 127) / 128) * ((N + 127) / 128) * (long)batches;
+    if (!batched && blocks < 256 && K / 64 >= 16) {
+      // deep-K underfilled grid: split the reduce over extra blocks
+      auto accbuf = at::empty({M, N}, A.options().dtype(at::kFloat));
+      mxa::gemm_tn_splitk_bf16(bfp(A), bfp(B), bfp_mut(out),
+                               accbuf.data_ptr<float>(), bptr, (int)M,
+                               (int)N, (int)K, relu, zp(A), cur_stream());
+    } else {
+      mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M,
+                        (int)N, (int)K, K, K, N, sA, sB, sC, batches,
+                        relu, zp(A), cur_stream());
+    }
   }
   return out;
 }

@@ -394,6 +394,27 @@ void gemm_tn_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
   launch(a, b, C, bias, M, N, ldc, strideC, K / BK, relu, batches, stream);
 }
 
+// split-K variant for deep-K plain GEMMs that underfill the chip
+// (e.g. [3136, 512] x K=2304: 100 tiles on 256 CUs without a split)
+void gemm_tn_splitk_bf16(const bf16* A, const bf16* B, bf16* C,
+                         float* accbuf, const float* bias, int M, int N,
+                         int K, bool relu, const bf16* zpage,
+                         hipStream_t stream) {
+  PlainLoader a{A, zpage, M, (long)K, ZMap{1, 0, 0}, K / BK};
+  PlainLoader b{B, zpage, N, (long)K, ZMap{1, 0, 0}, K / BK};
+  int ksteps = K / BK;
+  long blocks = (long)cdiv_h(M, BM) * cdiv_h(N, BN);
+  int split = (int)min((long)ksteps, max(1L, 512 / (blocks < 1 ? 1 : blocks)));
+  long total = (long)M * N;
+  HIP_CHECK(hipMemsetAsync(accbuf, 0, total * sizeof(float), stream));
+  dim3 grid((unsigned)blocks, split, 1);
+  gemm_tn_core<PlainLoader, PlainLoader, false, false, true>
+      <<<grid, THREADS, 0, stream>>>(a, b, (bf16*)accbuf, nullptr, M, N, N,
+                                     0, ksteps);
+  splitk_finish<<<ew_grid(total), 256, 0, stream>>>(accbuf, C, bias, total,
+                                                    N, relu ? 1 : 0);
+}
+
 // Batched attention scores/probs GEMM over per-(batch, head) slices of
 // [B, S, H*D] activations read IN PLACE (z = b*H + h):
 //   S[z][tq][tk] = sum_d q[b, tq, h*D + d] * k[b, tk, h*D + d]
