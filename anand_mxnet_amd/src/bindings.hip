@@ -201,18 +201,13 @@ at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
                        (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
                        cur_stream());
   } else {
-    long blocks = ((M + 127) / 128) * ((N + 127) / 128) * (long)batches;
-    if (!batched && blocks < 256 && K / 64 >= 16) {
-      // deep-K underfilled grid: split the reduce over extra blocks
-      auto accbuf = at::empty({M, N}, A.options().dtype(at::kFloat));
-      mxa::gemm_tn_splitk_bf16(bfp(A), bfp(B), bfp_mut(out),
-                               accbuf.data_ptr<float>(), bptr, (int)M,
-                               (int)N, (int)K, relu, zp(A), cur_stream());
-    } else {
-      mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M,
-                        (int)N, (int)K, K, K, N, sA, sB, sC, batches,
-                        relu, zp(A), cur_stream());
-    }
+    // (measured: a split-K variant exists — gemm_tn_splitk_bf16 — but
+    // auto-splitting these mid-size grids LOWERED throughput: the
+    // atomic finish + shorter per-split pipelines cost more than the
+    // extra fill buys.  Kept callable, not auto-dispatched.)
+    mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
+                      (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
+                      cur_stream());
   }
   return out;
 }
