@@ -194,7 +194,14 @@ at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
   long sB = (batched && B.dim() == 3) ? N * K : 0;
   long sC = batched ? M * N : 0;
   long blocks8 = ((M + 255) / 256) * ((N + 255) / 256) * (long)batches;
-  if (M >= 256 && N >= 256 && blocks8 >= 240) {
+  static const long tn8_mink = [] {
+    const char* e = getenv("MXA_TN8_MINK");
+    // short-K GEMMs fill better on the 128-tile kernel: tn8 runs 1
+    // block/CU, and a K=768 FFN wgrad-shape grid of ~384 tiles loses a
+    // third of the chip to the tail (562 vs 803 TF measured)
+    return e ? atol(e) : 1536L;
+  }();
+  if (M >= 256 && N >= 256 && blocks8 >= 240 && K >= tn8_mink) {
     // deep-pipelined 256² 8-phase kernel (needs a full chip of tiles;
     // smaller grids run the 128² kernel which fills better)
     mxa::gemm_tn8_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
