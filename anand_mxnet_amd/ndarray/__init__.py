@@ -7,6 +7,7 @@ import sys as _sys
 from .ndarray import (NDArray, array, zeros, ones, empty, full, arange,
                       from_torch, waitall, concat_nd)
 from . import random  # noqa: F401
+from . import sparse  # noqa: F401 (RowSparseNDArray / row_sparse_array)
 
 from .. import ops as _ops
 

@@ -211,10 +211,10 @@ class NDArray:
                         squeeze_axis=squeeze_axis)
 
     def tostype(self, stype):
-        if stype != "default":
-            raise MXNetError("only dense ('default') storage is implemented; "
-                             "sparse falls back to dense (see docs/PARITY.md)")
-        return self
+        if stype == "default":
+            return self
+        from .sparse import _dense_tostype
+        return _dense_tostype(self, stype)
 
     # ---------------- indexing ----------------
     def _idx(self, key):

@@ -159,3 +159,38 @@ def test_waitall():
     nd.waitall()
     a = nd.ones((4,))
     a.wait_to_read()
+
+
+def test_row_sparse_roundtrip_and_retain():
+    from anand_mxnet_amd.ndarray.sparse import (RowSparseNDArray,
+                                                row_sparse_array)
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import nd
+    dense = np.zeros((6, 3), np.float32)
+    dense[1] = [1, 2, 3]
+    dense[4] = [4, 5, 6]
+    rs = nd.array(dense).tostype("row_sparse")
+    assert isinstance(rs, RowSparseNDArray)
+    assert rs.stype == "row_sparse"
+    assert sorted(rs.indices.asnumpy().tolist()) == [1, 4]
+    np.testing.assert_allclose(rs.tostype("default").asnumpy(), dense)
+    kept = rs.retain([4])
+    np.testing.assert_allclose(kept.asnumpy()[4], [4, 5, 6])
+    assert kept.asnumpy()[1].sum() == 0
+    # explicit (data, indices) constructor
+    rs2 = row_sparse_array((nd.array(np.ones((2, 3), np.float32)),
+                            [0, 5]), shape=(6, 3))
+    assert rs2.asnumpy().sum() == 6
+
+
+def test_sparse_row_update():
+    from anand_mxnet_amd.ndarray.sparse import (row_sparse_array,
+                                                sparse_row_update)
+    from anand_mxnet_amd import nd
+    w = nd.array(np.ones((5, 2), np.float32))
+    g = row_sparse_array((nd.array(np.full((2, 2), 2.0, np.float32)),
+                          [1, 3]), shape=(5, 2))
+    sparse_row_update(w, g, lr=0.5)
+    out = w.asnumpy()
+    np.testing.assert_allclose(out[1], [0, 0])   # 1 - 0.5*2
+    np.testing.assert_allclose(out[0], [1, 1])   # untouched
