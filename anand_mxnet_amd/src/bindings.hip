@@ -108,6 +108,11 @@ void attn_scores_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
                       int, const bf16*, hipStream_t);
 void attn_wg_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
                   int, const bf16*, hipStream_t);
+// flash.hip
+void flash_fwd_bf16(const bf16*, const bf16*, const bf16*, bf16*, float*,
+                    float*, int, int, int, int, float, bool,
+                    const unsigned long long*, unsigned long long, float,
+                    hipStream_t);
 void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
 void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
 // gemm_fp8.hip
@@ -670,6 +675,31 @@ at::Tensor attn_ctx(const at::Tensor& p, const at::Tensor& v, long H) {
   return out;
 }
 
+std::vector<at::Tensor> flash_fwd(const at::Tensor& q,
+                                  const at::Tensor& k,
+                                  const at::Tensor& v, long H,
+                                  double alpha, bool causal, double pdrop,
+                                  long salt) {
+  // q/k/v [B, T, H*D] bf16, D == 64, T % 128 == 0 -> (out, m, l)
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  long B = q.size(0), Tq = q.size(1), HD = q.size(2), Tk = k.size(1);
+  long D = HD / H;
+  TORCH_CHECK(D == 64 && Tq % 128 == 0 && Tk % 128 == 0,
+              "flash_fwd: D==64 and seq%128 required");
+  auto out = at::empty_like(q);
+  auto opts = q.options().dtype(at::kFloat);
+  auto m = at::empty({B * H, Tq}, opts);
+  auto l = at::empty({B * H, Tq}, opts);
+  auto& sw = rng_seed_for(q);
+  mxa::flash_fwd_bf16(bfp(q), bfp(k), bfp(v), bfp_mut(out),
+                      m.data_ptr<float>(), l.data_ptr<float>(), (int)B,
+                      (int)H, (int)Tq, (int)Tk, (float)alpha, causal,
+                      (const unsigned long long*)sw.data_ptr<long>(),
+                      (unsigned long long)salt, (float)pdrop,
+                      cur_stream());
+  return {out, m, l};
+}
+
 at::Tensor attn_wg(const at::Tensor& p, const at::Tensor& q, long H) {
   // dK[b,tk,h*D+d] = sum_tq dS[b*H+h][tq][tk] q[b,tq,h*D+d] (WG form,
   // f32 out in [B,S,H*D] layout).  Also dV with (P, dO).
@@ -975,6 +1005,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_scores", &attn_scores);
   m.def("attn_ctx", &attn_ctx);
   m.def("attn_wg", &attn_wg);
+  m.def("flash_fwd", &flash_fwd);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("rng_step", &rng_step);
   m.def("softmax_dropout_fwd", &softmax_dropout_fwd);

@@ -1,0 +1,364 @@
+#include "hip/hip_runtime.h"
+// Fused flash-attention for gfx950 (MI355X), D = 64, seq % 128 == 0.
+//
+// Forward: one workgroup per (batch*head, 128-query block); K/V stream
+// through double-buffered LDS in 128-row chunks; scores, online softmax,
+// dropout and the P@V accumulation all stay on-chip — S/P/mask are never
+// materialized in HBM (reference parity: contrib/transformer.cu composed
+// GEMMs + softmax; here the MI355X-native fused form).  Saves per-row
+// softmax stats (m, l) for the exact backward.
+//
+// Layouts (empirically validated building blocks from gemm_tn/gemm_nn):
+//   Q/K images  [128 rows][64 k]  — TN style, glds + T2 XOR swizzle,
+//                                   ds_read_b128 fragments
+//   V image     [128 tk][64 d]    — row-major + (row&3)<<1 chunk swizzle,
+//                                   ds_read_b64_tr_b16 fragments
+//   P image     [128 q][128 tk]   — TN style with 16-key swizzle
+//
+// Wave layout (512 threads): 8 waves as 4(M=q) x 2(N); each wave owns the
+// SAME 32 query rows in the S phase (N = tk half) and the O phase
+// (N = d half), so the online-softmax rescale of the O accumulator is a
+// per-lane multiply.  Cross-wave row stats go through a 2-KiB LDS
+// scratch (the two tk-half waves of each row pair).
+#include "common.h"
+
+namespace mxa {
+
+DEVINL int fa_qk_off(int row, int chunk16) {
+  // [128][64 bf16] image, 128-B rows (T2 swizzle, 8 chunks/row)
+  return row * 64 + ((chunk16 ^ (row & 7)) << 3);
+}
+
+DEVINL int fa_p_byte(int row, int chunk16) {
+  // [128][128 bf16] image, 256-B rows, 16 chunks/row, 16-key swizzle
+  return row * 256 + ((chunk16 ^ (row & 15)) << 4);
+}
+
+DEVINL int fa_v_byte(int row, int cbyte) {
+  // [128][64 bf16] row-major tr image (see gemm_nn nn_b64_byte)
+  return row * 128 + ((((cbyte >> 4) ^ ((row & 3) << 1)) << 4) | (cbyte & 15));
+}
+
+__global__ __launch_bounds__(512, 1) void flash_fwd_core(
+    const bf16* __restrict__ Q, const bf16* __restrict__ K,
+    const bf16* __restrict__ V, bf16* __restrict__ O,
+    float* __restrict__ Ms, float* __restrict__ Ls, int B, int H, int Tq,
+    int Tk, float alpha, int causal, const u64* __restrict__ seed_ptr,
+    u64 salt, float pdrop, float dscale) {
+  constexpr int D = 64;
+  __shared__ __align__(16) bf16 qimg[128 * D];
+  __shared__ __align__(16) bf16 kimg[2][128 * D];
+  __shared__ __align__(16) bf16 vimg[2][128 * D];
+  __shared__ __align__(16) bf16 pimg[128 * 128];
+  __shared__ float sred[2][128][2];  // [stat m/l][row][tk-half wave]
+
+  const int z = blockIdx.z;          // b*H + h
+  const int b = z / H, h = z % H;
+  const int qb = blockIdx.x;         // query block
+  const long hd = (long)H * D;
+  const int t = threadIdx.x;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  const u64 seed = seed_ptr ? seed_finalize(*seed_ptr + salt) : 0;
+  const u32 thresh = (u32)(pdrop * 4294967296.0f);
+
+  // ---- stage Q once: rows qb*128 + r, k = d (glds lane-linear, source
+  // chunk XOR; geometry identical to the 512-thread TN kernel) ----
+  const int srow = (((t >> 6) * 2) << 3) + ((t & 63) >> 3);  // pass-0 row
+  const int scg = (ll & 7) ^ ((ll >> 3) & 7);
+  {
+    const bf16* qbase = Q + (long)b * Tq * hd + (long)h * D;
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int row = srow + p * 8;
+      unsigned lbase = __builtin_amdgcn_readfirstlane(
+          (unsigned)((lw * 2 + p) * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&qimg[0] + lbase);
+      const bf16* gp = qbase + (long)(qb * 128 + row) * hd + scg * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds,
+          16, 0, 0);
+    }
+  }
+
+  // K / V chunk staging (kv chunk c -> rows c*128 + row)
+  const bf16* kbase = K + (long)b * Tk * hd + (long)h * D;
+  const bf16* vbase = V + (long)b * Tk * hd + (long)h * D;
+  auto stage_kv = [&](int buf, int c) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int row = srow + p * 8;
+      unsigned lbase = __builtin_amdgcn_readfirstlane(
+          (unsigned)((lw * 2 + p) * 1024));
+      auto ldsK = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&kimg[buf][0] + lbase);
+      auto ldsV = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&vimg[buf][0] + lbase);
+      const bf16* gk = kbase + (long)(c * 128 + row) * hd + scg * 8;
+      // V tr image: source chunk key is (row&3)<<1 on 8-bf16 chunks —
+      // row&3 here = (ll>>3)&3, chunk pos = ll&7
+      int vcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);
+      const bf16* gv = vbase + (long)(c * 128 + row) * hd + vcg * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gk, ldsK,
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gv, ldsV,
+          16, 0, 0);
+    }
+  };
+
+  const int wmq = (lw & 3) * 32;     // this wave's 32 query rows
+  const int wnt = (lw >> 2) * 64;    // S phase: tk half
+  const int wno = (lw >> 2) * 32;    // O phase: d half
+  const int lrow = ll & 15, lk = ll >> 4;
+
+  // persistent per-lane state: rows wmq + mf*16 + (ll>>4)*4 + i
+  float m_i[2][4], l_i[2][4];
+  f32x4 acc_o[2][2];
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      m_i[mf][i] = -3.0e38f;
+      l_i[mf][i] = 0.f;
+    }
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) acc_o[mf][nf] = f32x4{0, 0, 0, 0};
+
+  const int nchunks_all = Tk / 128;
+  const int nchunks = causal ? (qb + 1) : nchunks_all;
+
+  stage_kv(0, 0);
+  __syncthreads();
+
+  for (int c = 0; c < nchunks; ++c) {
+    int buf = c & 1;
+    if (c + 1 < nchunks) stage_kv(buf ^ 1, c + 1);
+
+    // ---- S = alpha * Q K^T for this chunk (wave: 32 q x 64 tk) ----
+    f32x4 acc_s[2][4];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) acc_s[mf][nf] = f32x4{0, 0, 0, 0};
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[2], bfr[4];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        af[mf] = *(const bf16x8*)&qimg[fa_qk_off(wmq + mf * 16 + lrow,
+                                                 chunk)];
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        bfr[nf] = *(const bf16x8*)&kimg[buf][fa_qk_off(
+            wnt + nf * 16 + lrow, chunk)];
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc_s[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc_s[mf][nf], 0, 0, 0);
+    }
+
+    // ---- scale + causal mask + chunk row-max ----
+    float cmax[2][4];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) cmax[mf][i] = -3.0e38f;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int col = c * 128 + wnt + nf * 16 + (ll & 15);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int row = qb * 128 + wmq + mf * 16 + (ll >> 4) * 4 + i;
+          float v = acc_s[mf][nf][i] * alpha;
+          if (causal && col > row) v = -3.0e38f;
+          acc_s[mf][nf][i] = v;
+          cmax[mf][i] = fmaxf(cmax[mf][i], v);
+        }
+      }
+    // reduce over the 16 column lanes (rows replicated in lane groups)
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float v = cmax[mf][i];
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          v = fmaxf(v, __shfl_xor(v, off, 64));
+        cmax[mf][i] = v;
+      }
+    // cross tk-half waves via LDS
+    if ((ll & 15) == 0) {
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          sred[0][wmq + mf * 16 + (ll >> 4) * 4 + i][lw >> 2] =
+              cmax[mf][i];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int r = wmq + mf * 16 + (ll >> 4) * 4 + i;
+        cmax[mf][i] = fmaxf(sred[0][r][0], sred[0][r][1]);
+      }
+
+    // ---- online rescale + P = exp(s - m_new) + chunk row-sum ----
+    float csum[2][4];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float m_new = fmaxf(m_i[mf][i], cmax[mf][i]);
+        float f = __expf(m_i[mf][i] - m_new);
+        m_i[mf][i] = m_new;
+        l_i[mf][i] *= f;
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) acc_o[mf][nf][i] *= f;
+        csum[mf][i] = 0.f;
+      }
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          float p = __expf(acc_s[mf][nf][i] - m_i[mf][i]);
+          acc_s[mf][nf][i] = p;
+          csum[mf][i] += p;
+        }
+      }
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float v = csum[mf][i];
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          v += __shfl_xor(v, off, 64);
+        csum[mf][i] = v;
+      }
+    if ((ll & 15) == 0) {
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          sred[1][wmq + mf * 16 + (ll >> 4) * 4 + i][lw >> 2] =
+              csum[mf][i];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int r = wmq + mf * 16 + (ll >> 4) * 4 + i;
+        l_i[mf][i] += sred[1][r][0] + sred[1][r][1];
+      }
+
+    // ---- dropout (unstored; bwd replays the hash) + write P image ----
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int col = wnt + nf * 16 + (ll & 15);
+        int gcol = c * 128 + col;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+          float p = acc_s[mf][nf][i];
+          if (thresh) {
+            long grow = (long)qb * 128 + row;
+            u32 r = hash_u32(seed, ((long)z * Tq + grow) * Tk + gcol);
+            p = r >= thresh ? p * dscale : 0.f;
+          }
+          *(bf16*)((char*)pimg + fa_p_byte(row, col >> 3) +
+                   (col & 7) * 2) = f2bf(p);
+        }
+      }
+    __syncthreads();
+
+    // ---- acc_o += P~ V (wave: 32 q x 32 d), K dim = 128 tk ----
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      bf16x8 af[2], bfr[2];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        af[mf] = *(const bf16x8*)((const char*)pimg +
+                                  fa_p_byte(wmq + mf * 16 + lrow, chunk));
+      int rbase = kk * 32 + (ll >> 4) * 8 + ((ll & 15) >> 2);
+      int cb = ((ll & 15) & 3) * 8;
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int cbyte = (wno + nf * 16) * 2 + cb;
+        auto pv = (__attribute__((address_space(3))) bf16x4*)(
+            (__attribute__((address_space(3))) char*)&vimg[buf][0] +
+            fa_v_byte(rbase, cbyte));
+        bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(pv);
+        auto pv2 = (__attribute__((address_space(3))) bf16x4*)(
+            (__attribute__((address_space(3))) char*)&vimg[buf][0] +
+            fa_v_byte(rbase + 4, cbyte));
+        bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(pv2);
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc_o[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc_o[mf][nf], 0, 0, 0);
+    }
+    __syncthreads();  // pimg reused; kv buffers swap
+  }
+
+  // ---- epilogue: O = acc_o / l, write strided [B, Tq, H*D]; stats ----
+  bf16* obase = O + (long)b * Tq * hd + (long)h * D;
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+      float inv = l_i[mf][i] > 0.f ? 1.f / l_i[mf][i] : 0.f;
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int col = wno + nf * 16 + (ll & 15);
+        obase[(long)(qb * 128 + row) * hd + col] =
+            f2bf(acc_o[mf][nf][i] * inv);
+      }
+    }
+  }
+  if ((ll & 15) == 0 && (lw >> 2) == 0) {
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+        Ms[(long)z * Tq + qb * 128 + row] = m_i[mf][i];
+        Ls[(long)z * Tq + qb * 128 + row] = l_i[mf][i];
+      }
+  }
+}
+
+void flash_fwd_bf16(const bf16* Q, const bf16* K, const bf16* V, bf16* O,
+                    float* Ms, float* Ls, int B, int H, int Tq, int Tk,
+                    float alpha, bool causal, const u64* seed_ptr, u64 salt,
+                    float pdrop, hipStream_t stream) {
+  dim3 grid(Tq / 128, 1, B * H);
+  float dscale = pdrop > 0.f ? 1.f / (1.f - pdrop) : 1.f;
+ hipLaunchKernelGGL(( flash_fwd_core), dim3(grid), dim3(512), 0, stream, 
+      Q, K, V, O, Ms, Ls, B, H, Tq, Tk, alpha, causal ? 1 : 0,
+      pdrop > 0.f ? seed_ptr : nullptr, salt, pdrop, dscale);
+}
+
+}  // namespace mxa

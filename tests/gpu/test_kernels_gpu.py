@@ -631,3 +631,51 @@ def test_depthwise_conv_fwd_bwd():
                      msg=f"dw dx s{s_}")
         assert_close(w1.grad, w2.grad, rtol=4e-2, atol=6e-2,
                      msg=f"dw dw s{s_}")
+
+
+def test_flash_fwd_vs_torch():
+    B, H, T, D = 2, 3, 256, 64
+    scale = 1.0 / (D ** 0.5)
+    for causal in (False, True):
+        q = _rt((B, T, H * D)).to(BF)
+        k = _rt((B, T, H * D)).to(BF)
+        v = _rt((B, T, H * D)).to(BF)
+        out, m, l = _C.flash_fwd(q.contiguous(), k.contiguous(),
+                                 v.contiguous(), H, scale, causal, 0.0, 0)
+        qh = _heads(q, H).float()
+        kh = _heads(k, H).float()
+        vh = _heads(v, H).float()
+        s = qh @ kh.transpose(-1, -2) * scale
+        if causal:
+            mask = torch.triu(torch.ones(T, T, device=_dev(),
+                                         dtype=torch.bool), 1)
+            s = s.masked_fill(mask, float("-inf"))
+        p = torch.nn.functional.softmax(s, dim=-1)
+        ref = (p @ vh).transpose(1, 2).reshape(B, T, H * D)
+        assert_close(out, ref, rtol=5e-2, atol=5e-2,
+                     msg=f"flash out c={causal}")
+        # stats: l = sum exp(s - m) rowwise
+        ref_m = s.max(-1).values.reshape(B * H, T)
+        ref_l = torch.exp(s - ref_m.reshape(B, H, T, 1)).sum(-1) \
+            .reshape(B * H, T)
+        assert_close(m, ref_m, rtol=1e-2, atol=1e-2, msg="flash m")
+        assert_close(l, ref_l, rtol=2e-2, atol=1e-2, msg="flash l")
+
+
+def test_flash_fwd_dropout_stats():
+    B, H, T, D = 1, 2, 128, 64
+    q = _rt((B, T, H * D)).to(BF)
+    k = _rt((B, T, H * D)).to(BF)
+    v = _rt((B, T, H * D)).to(BF)
+    o0, m0, l0 = _C.flash_fwd(q.contiguous(), k.contiguous(),
+                              v.contiguous(), H, 0.125, False, 0.0, 7)
+    o1, m1, l1 = _C.flash_fwd(q.contiguous(), k.contiguous(),
+                              v.contiguous(), H, 0.125, False, 0.3, 7)
+    # stats are dropout-independent; outputs differ
+    assert_close(m0, m1, rtol=1e-5, atol=1e-5, msg="m stable")
+    assert_close(l0, l1, rtol=1e-5, atol=1e-5, msg="l stable")
+    d = (o0.float() - o1.float()).abs().mean().item()
+    assert d > 1e-3, d
+    # E[dropout(p)] = p: means should be close
+    r = (o1.float().mean() / o0.float().mean()).item()
+    assert 0.7 < r < 1.3, r
