@@ -676,6 +676,7 @@ def test_flash_fwd_dropout_stats():
     assert_close(l0, l1, rtol=1e-5, atol=1e-5, msg="l stable")
     d = (o0.float() - o1.float()).abs().mean().item()
     assert d > 1e-3, d
-    # E[dropout(p)] = p: means should be close
-    r = (o1.float().mean() / o0.float().mean()).item()
-    assert 0.7 < r < 1.3, r
+    # E[dropout(p)] = p: magnitudes stay comparable (signed means of
+    # random attention outputs are ~0 and unusable as a ratio)
+    r = (o1.float().abs().mean() / o0.float().abs().mean()).item()
+    assert 0.6 < r < 1.8, r
