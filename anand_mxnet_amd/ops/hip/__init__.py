@@ -752,11 +752,43 @@ class _MhaCore(torch.autograd.Function):
         return dq, dk, dv, None, None, None, None, None
 
 
+class _FlashMha(torch.autograd.Function):
+    """Fully-fused flash attention (flash.hip): S/P/mask never touch HBM;
+    backward recomputes probabilities from the saved (m, l) stats and
+    replays the dropout hash."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, H, causal, scale, dropout_p, training):
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        dp = dropout_p if training else 0.0
+        salt = _next_seed()
+        out, m, l = _C.flash_fwd(q, k, v, H, scale, causal, dp, salt)
+        ctx.save_for_backward(q, k, v, out, m, l)
+        ctx.conf = (H, scale, causal, dp, salt)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, m, l = ctx.saved_tensors
+        H, scale, causal, dp, salt = ctx.conf
+        dq, dk, dv = _C.flash_bwd(q, k, v, out, dout.contiguous(), m, l,
+                                  H, scale, causal, dp, salt)
+        return dq, dk, dv, None, None, None, None, None
+
+
+import os as _os
+_USE_FLASH = _os.environ.get("MXA_FLASH", "1") != "0"
+
+
 @register_gpu("mha")
 def mha(q, k, v, num_heads, causal, scale, dropout_p=0.0, training=False):
     B, Tq, HD = q.shape
     Tk = k.shape[1]
     D = HD // num_heads
+    if (_USE_FLASH and D == 64 and Tq % 128 == 0 and Tk % 128 == 0
+            and Tq == Tk and q.dtype == torch.bfloat16):
+        return _FlashMha.apply(q, k, v, num_heads, causal, scale,
+                               dropout_p, training)
     if (D % 64 == 0 and Tk % 64 == 0 and Tq % 8 == 0
             and q.dtype == torch.bfloat16):
         return _MhaCore.apply(q, k, v, num_heads, causal, scale,

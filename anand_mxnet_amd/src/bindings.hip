@@ -113,6 +113,11 @@ void flash_fwd_bf16(const bf16*, const bf16*, const bf16*, bf16*, float*,
                     float*, int, int, int, int, float, bool,
                     const unsigned long long*, unsigned long long, float,
                     hipStream_t);
+void flash_bwd_bf16(const bf16*, const bf16*, const bf16*, const bf16*,
+                    const bf16*, const float*, const float*, float*,
+                    float*, bf16*, bf16*, int, int, int, int, float, bool,
+                    const unsigned long long*, unsigned long long, float,
+                    hipStream_t);
 void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
 void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
 // gemm_fp8.hip
@@ -700,6 +705,33 @@ std::vector<at::Tensor> flash_fwd(const at::Tensor& q,
   return {out, m, l};
 }
 
+std::vector<at::Tensor> flash_bwd(const at::Tensor& q,
+                                  const at::Tensor& k,
+                                  const at::Tensor& v,
+                                  const at::Tensor& o,
+                                  const at::Tensor& dout,
+                                  const at::Tensor& m, const at::Tensor& l,
+                                  long H, double alpha, bool causal,
+                                  double pdrop, long salt) {
+  long B = q.size(0), Tq = q.size(1), HD = q.size(2), Tk = k.size(1);
+  auto opts = q.options().dtype(at::kFloat);
+  auto drow = at::empty({B * H, Tq}, opts);
+  auto dqacc = at::zeros({B, Tq, HD}, opts);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto& sw = rng_seed_for(q);
+  mxa::flash_bwd_bf16(bfp(q), bfp(k), bfp(v), bfp(o),
+                      bfp(dout.contiguous()), m.data_ptr<float>(),
+                      l.data_ptr<float>(), drow.data_ptr<float>(),
+                      dqacc.data_ptr<float>(), bfp_mut(dk), bfp_mut(dv),
+                      (int)B, (int)H, (int)Tq, (int)Tk, (float)alpha,
+                      causal,
+                      (const unsigned long long*)sw.data_ptr<long>(),
+                      (unsigned long long)salt, (float)pdrop,
+                      cur_stream());
+  return {dqacc.to(at::kBFloat16), dk, dv};
+}
+
 at::Tensor attn_wg(const at::Tensor& p, const at::Tensor& q, long H) {
   // dK[b,tk,h*D+d] = sum_tq dS[b*H+h][tq][tk] q[b,tq,h*D+d] (WG form,
   // f32 out in [B,S,H*D] layout).  Also dV with (P, dO).
@@ -1006,6 +1038,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_ctx", &attn_ctx);
   m.def("attn_wg", &attn_wg);
   m.def("flash_fwd", &flash_fwd);
+  m.def("flash_bwd", &flash_bwd);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("rng_step", &rng_step);
   m.def("softmax_dropout_fwd", &softmax_dropout_fwd);

@@ -39,6 +39,11 @@ DEVINL int fa_v_byte(int row, int cbyte) {
   return row * 128 + ((((cbyte >> 4) ^ ((row & 3) << 1)) << 4) | (cbyte & 15));
 }
 
+DEVINL int fa_tr128_byte(int row, int cbyte) {
+  // [128][128 bf16] row-major tr image (256-B rows), same chunk key
+  return row * 256 + ((((cbyte >> 4) ^ ((row & 3) << 1)) << 4) | (cbyte & 15));
+}
+
 __global__ __launch_bounds__(512, 1) void flash_fwd_core(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O,
@@ -359,6 +364,387 @@ void flash_fwd_bf16(const bf16* Q, const bf16* K, const bf16* V, bf16* O,
  hipLaunchKernelGGL(( flash_fwd_core), dim3(grid), dim3(512), 0, stream, 
       Q, K, V, O, Ms, Ls, B, H, Tq, Tk, alpha, causal ? 1 : 0,
       pdrop > 0.f ? seed_ptr : nullptr, salt, pdrop, dscale);
+}
+
+// ---------------------------------------------------------------------------
+// Backward.  Grid (Tk/128 kv-blocks, 1, B*H): each block owns dK/dV for
+// its 128 kv rows (register accumulators across the q-chunk loop); dQ is
+// accumulated into a global f32 buffer with atomics.  Probabilities and
+// the dropout mask are RECOMPUTED from the saved (m, l) stats and the
+// RNG seed — S/P/mask never hit HBM in either direction.
+//
+// LDS is a managed arena (146 KiB peak): the Q-TN/dO-TN images are dead
+// after the two TN GEMMs and their 32 KiB is reused for the dS tr-image.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(512, 1) void flash_bwd_core(
+    const bf16* __restrict__ Q, const bf16* __restrict__ K,
+    const bf16* __restrict__ V, const bf16* __restrict__ dO,
+    const float* __restrict__ Ms, const float* __restrict__ Ls,
+    const float* __restrict__ Drow, float* __restrict__ dQacc,
+    bf16* __restrict__ dKout, bf16* __restrict__ dVout, int B, int H,
+    int Tq, int Tk, float alpha, int causal,
+    const u64* __restrict__ seed_ptr, u64 salt, float pdrop,
+    float dscale) {
+  constexpr int D = 64;
+  // arena layout (bf16 elems): ktn 8K | ktr 8K | vtn 8K | qtr 8K |
+  // dotr 8K | shared: [qtn 8K + dotn 8K] == [dstr 16K] | ptn 16K
+  __shared__ __align__(16) bf16 arena[8192 * 5 + 16384 + 16384];
+  bf16* ktn = arena;
+  bf16* ktr = arena + 8192;
+  bf16* vtn = arena + 8192 * 2;
+  bf16* qtr = arena + 8192 * 3;
+  bf16* dotr = arena + 8192 * 4;
+  bf16* qtn = arena + 8192 * 5;        // dies after S_T
+  bf16* dotn = arena + 8192 * 5 + 8192;  // dies after dP_T
+  bf16* dstr = arena + 8192 * 5;       // aliases qtn+dotn (32 KiB)
+  bf16* ptn = arena + 8192 * 7;        // pd then ds (TN image, 32 KiB)
+  __shared__ float stats[3][128];      // m, l, Drow for the q-chunk
+
+  const int z = blockIdx.z;
+  const int b = z / H, h = z % H;
+  const int kvb = blockIdx.x;
+  const long hd = (long)H * D;
+  const int t = threadIdx.x;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  const u64 seed = seed_ptr ? seed_finalize(*seed_ptr + salt) : 0;
+  const u32 thresh = (u32)(pdrop * 4294967296.0f);
+
+  const int srow = (((t >> 6) * 2) << 3) + ((t & 63) >> 3);
+  const int scg = (ll & 7) ^ ((ll >> 3) & 7);           // TN source chunk
+  const int tcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);    // tr source chunk
+
+  // one 16-KiB [128][64] image stage: 2 glds per thread
+  auto stage64 = [&](bf16* img, const bf16* base, int row0, int cg) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int row = srow + p * 8;
+      unsigned lbase = __builtin_amdgcn_readfirstlane(
+          (unsigned)((lw * 2 + p) * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)img + lbase);
+      const bf16* gp = base + (long)(row0 + row) * hd + cg * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds,
+          16, 0, 0);
+    }
+  };
+
+  const bf16* qbase = Q + (long)b * Tq * hd + (long)h * D;
+  const bf16* kbase = K + (long)b * Tk * hd + (long)h * D;
+  const bf16* vbase = V + (long)b * Tk * hd + (long)h * D;
+  const bf16* dobase = dO + (long)b * Tq * hd + (long)h * D;
+
+  // K/V resident for the whole block
+  stage64(ktn, kbase, kvb * 128, scg);
+  stage64(ktr, kbase, kvb * 128, tcg);
+  stage64(vtn, vbase, kvb * 128, scg);
+
+  const int wmq = (lw & 3) * 32;   // 32 tk rows (TN phases) / 32 q (WG)
+  const int wnt = (lw >> 2) * 64;  // 64-col half (q dim in TN phases)
+  const int wno = (lw >> 2) * 32;  // 32-col half (d dim)
+  const int lrow = ll & 15, lk = ll >> 4;
+
+  f32x4 acc_dv[2][2], acc_dk[2][2];
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      acc_dv[mf][nf] = f32x4{0, 0, 0, 0};
+      acc_dk[mf][nf] = f32x4{0, 0, 0, 0};
+    }
+
+  const int qc0 = causal ? kvb : 0;
+  const int nq = Tq / 128;
+
+  for (int qc = qc0; qc < nq; ++qc) {
+    // ---- stage q-chunk images + stats ----
+    stage64(qtn, qbase, qc * 128, scg);
+    stage64(qtr, qbase, qc * 128, tcg);
+    stage64(dotn, dobase, qc * 128, scg);
+    stage64(dotr, dobase, qc * 128, tcg);
+    for (int i = t; i < 128; i += 512) {
+      long qg = (long)z * Tq + qc * 128 + i;
+      stats[0][i] = Ms[qg];
+      stats[1][i] = Ls[qg];
+      stats[2][i] = Drow[qg];
+    }
+    __syncthreads();
+
+    // ---- S_T = K Q^T and dP_T = V dO^T, processed in two 32-q-column
+    // passes so only [2][2] accumulators are live (the full [2][4] pair
+    // spilled); pd -> ptn (TN image for dV), ds -> dstr (tr image) ----
+#pragma unroll 1
+    for (int qh = 0; qh < 2; ++qh) {
+      f32x4 acc_st[2][2], acc_dp[2][2];
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          acc_st[mf][nf] = f32x4{0, 0, 0, 0};
+          acc_dp[mf][nf] = f32x4{0, 0, 0, 0};
+        }
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bf16x8 ak[2], bq[2];
+        int chunk = kk * 4 + lk;
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+          ak[mf] = *(const bf16x8*)&ktn[fa_qk_off(wmq + mf * 16 + lrow,
+                                                  chunk)];
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          bq[nf] = *(const bf16x8*)&qtn[fa_qk_off(
+              wnt + (qh * 2 + nf) * 16 + lrow, chunk)];
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc_st[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                ak[mf], bq[nf], acc_st[mf][nf], 0, 0, 0);
+      }
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bf16x8 av[2], bdo[2];
+        int chunk = kk * 4 + lk;
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+          av[mf] = *(const bf16x8*)&vtn[fa_qk_off(wmq + mf * 16 + lrow,
+                                                  chunk)];
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          bdo[nf] = *(const bf16x8*)&dotn[fa_qk_off(
+              wnt + (qh * 2 + nf) * 16 + lrow, chunk)];
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc_dp[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                av[mf], bdo[nf], acc_dp[mf][nf], 0, 0, 0);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          int qcol = wnt + (qh * 2 + nf) * 16 + (ll & 15);
+          long qg = (long)qc * 128 + qcol;
+          float m = stats[0][qcol], linv = stats[1][qcol];
+          linv = linv > 0.f ? 1.f / linv : 0.f;
+          float dr = stats[2][qcol];
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            int tkrow = wmq + mf * 16 + (ll >> 4) * 4 + i;
+            long kg = (long)kvb * 128 + tkrow;
+            float sv = acc_st[mf][nf][i] * alpha;
+            float pv = (causal && kg > qg) ? 0.f : __expf(sv - m) * linv;
+            float keep = 1.f;
+            if (thresh) {
+              u32 r = hash_u32(seed, ((long)z * Tq + qg) * Tk + kg);
+              keep = r >= thresh ? dscale : 0.f;
+            }
+            float pd = pv * keep;
+            float dp = acc_dp[mf][nf][i] * keep;
+            float ds = pv * (dp - dr) * alpha;
+            *(bf16*)((char*)ptn + fa_p_byte(tkrow, qcol >> 3) +
+                     (qcol & 7) * 2) = f2bf(pd);
+            *(bf16*)((char*)dstr + fa_tr128_byte(tkrow, qcol * 2)) =
+                f2bf(ds);
+          }
+        }
+    }
+    __syncthreads();
+
+    // ---- dV += pd · dO  (A: ptn rows tk, k = q; B: dotr tr image) ----
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      bf16x8 af[2], bfr[2];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        af[mf] = *(const bf16x8*)((const char*)ptn +
+                                  fa_p_byte(wmq + mf * 16 + lrow, chunk));
+      int rbase = kk * 32 + (ll >> 4) * 8 + ((ll & 15) >> 2);
+      int cb = ((ll & 15) & 3) * 8;
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int cbyte = (wno + nf * 16) * 2 + cb;
+        auto p1 = (__attribute__((address_space(3))) bf16x4*)(
+            (__attribute__((address_space(3))) char*)dotr +
+            fa_v_byte(rbase, cbyte));
+        auto p2 = (__attribute__((address_space(3))) bf16x4*)(
+            (__attribute__((address_space(3))) char*)dotr +
+            fa_v_byte(rbase + 4, cbyte));
+        bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+        bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc_dv[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc_dv[mf][nf], 0, 0, 0);
+    }
+
+    // ---- dK += ds · Q  (A: dstr rows tk, k = q, b128 reads on the
+    // tr-keyed image — mild bank sharing, acceptable; B: qtr) ----
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      bf16x8 af[2], bfr[2];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        af[mf] = *(const bf16x8*)((const char*)dstr +
+                                  fa_tr128_byte(wmq + mf * 16 + lrow,
+                                                chunk * 16));
+      int rbase = kk * 32 + (ll >> 4) * 8 + ((ll & 15) >> 2);
+      int cb = ((ll & 15) & 3) * 8;
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int cbyte = (wno + nf * 16) * 2 + cb;
+        auto p1 = (__attribute__((address_space(3))) bf16x4*)(
+            (__attribute__((address_space(3))) char*)qtr +
+            fa_v_byte(rbase, cbyte));
+        auto p2 = (__attribute__((address_space(3))) bf16x4*)(
+            (__attribute__((address_space(3))) char*)qtr +
+            fa_v_byte(rbase + 4, cbyte));
+        bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+        bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc_dk[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc_dk[mf][nf], 0, 0, 0);
+    }
+
+    // ---- dQ partial: WG over tk — A = dstr [tk][128 q] tr image,
+    //      B = ktr [tk][64 d] tr image; wave = 32 q x 32 d ----
+    {
+      f32x4 acc_dq[2][2];
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) acc_dq[mf][nf] = f32x4{0, 0, 0, 0};
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 af[2], bfr[2];
+        int rbase = kk * 32 + (ll >> 4) * 8 + ((ll & 15) >> 2);
+        int cb = ((ll & 15) & 3) * 8;
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          int cbyte = (wmq + mf * 16) * 2 + cb;
+          auto p1 = (__attribute__((address_space(3))) bf16x4*)(
+              (__attribute__((address_space(3))) char*)dstr +
+              fa_tr128_byte(rbase, cbyte));
+          auto p2 = (__attribute__((address_space(3))) bf16x4*)(
+              (__attribute__((address_space(3))) char*)dstr +
+              fa_tr128_byte(rbase + 4, cbyte));
+          bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+          bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+          af[mf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6,
+                                           7);
+        }
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          int cbyte = (wno + nf * 16) * 2 + cb;
+          auto p1 = (__attribute__((address_space(3))) bf16x4*)(
+              (__attribute__((address_space(3))) char*)ktr +
+              fa_v_byte(rbase, cbyte));
+          auto p2 = (__attribute__((address_space(3))) bf16x4*)(
+              (__attribute__((address_space(3))) char*)ktr +
+              fa_v_byte(rbase + 4, cbyte));
+          bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+          bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+          bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6,
+                                            7);
+        }
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc_dq[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mf], bfr[nf], acc_dq[mf][nf], 0, 0, 0);
+      }
+      float* dqb = dQacc + ((long)b * Tq + qc * 128) * hd + (long)h * D;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          int col = wno + nf * 16 + (ll & 15);
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+            atomicAdd(&dqb[(long)row * hd + col], acc_dq[mf][nf][i]);
+          }
+        }
+    }
+    __syncthreads();  // qtr/dotr/dstr rewritten next q-chunk
+  }
+
+  // ---- epilogue: dK, dV (strided bf16) ----
+  bf16* dkb = dKout + ((long)b * Tk + kvb * 128) * hd + (long)h * D;
+  bf16* dvb = dVout + ((long)b * Tk + kvb * 128) * hd + (long)h * D;
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      int col = wno + nf * 16 + (ll & 15);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+        dkb[(long)row * hd + col] = f2bf(acc_dk[mf][nf][i]);
+        dvb[(long)row * hd + col] = f2bf(acc_dv[mf][nf][i]);
+      }
+    }
+}
+
+// rowwise dot for the backward: Drow[z][q] = sum_d dO * O  (wave-per-row)
+__global__ __launch_bounds__(256) void flash_drow(
+    const bf16* __restrict__ dO, const bf16* __restrict__ O,
+    float* __restrict__ Drow, int B, int H, int Tq) {
+  constexpr int D = 64;
+  const long hd = (long)H * D;
+  const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
+  long rows = (long)B * H * Tq;
+  for (long r = blockIdx.x * 4 + w; r < rows; r += 4L * gridDim.x) {
+    int z = (int)(r / Tq);
+    int q = (int)(r % Tq);
+    int b = z / H, h = z % H;
+    const bf16* dop = dO + ((long)b * Tq + q) * hd + (long)h * D;
+    const bf16* op = O + ((long)b * Tq + q) * hd + (long)h * D;
+    // D = 64: one bf16x8 per lane-of-8
+    float acc = 0.f;
+    if (l < 8) {
+      bf16x8 a = *(const bf16x8*)&dop[l * 8];
+      bf16x8 c = *(const bf16x8*)&op[l * 8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += bf2f(a[j]) * bf2f(c[j]);
+    }
+#pragma unroll
+    for (int off = 1; off < 8; off <<= 1) acc += __shfl_xor(acc, off, 64);
+    if (l == 0) Drow[r] = acc;
+  }
+}
+
+void flash_bwd_bf16(const bf16* Q, const bf16* K, const bf16* V,
+                    const bf16* O, const bf16* dOp, const float* Ms,
+                    const float* Ls, float* Drow, float* dQacc, bf16* dK,
+                    bf16* dV, int B, int H, int Tq, int Tk, float alpha,
+                    bool causal, const u64* seed_ptr, u64 salt,
+                    float pdrop, hipStream_t stream) {
+  long rows = (long)B * H * Tq;
+  int blocks = (int)min((rows + 3) / 4, 2048L);
+ hipLaunchKernelGGL(( flash_drow), dim3(blocks), dim3(256), 0, stream, dOp, O, Drow, B, H, Tq);
+  float dscale = pdrop > 0.f ? 1.f / (1.f - pdrop) : 1.f;
+  dim3 grid(Tk / 128, 1, B * H);
+ hipLaunchKernelGGL(( flash_bwd_core), dim3(grid), dim3(512), 0, stream, 
+      Q, K, V, dOp, Ms, Ls, Drow, dQacc, dK, dV, B, H, Tq, Tk, alpha,
+      causal ? 1 : 0, pdrop > 0.f ? seed_ptr : nullptr, salt, pdrop,
+      dscale);
 }
 
 }  // namespace mxa

@@ -680,3 +680,51 @@ def test_flash_fwd_dropout_stats():
     # random attention outputs are ~0 and unusable as a ratio)
     r = (o1.float().abs().mean() / o0.float().abs().mean()).item()
     assert 0.6 < r < 1.8, r
+
+
+def test_flash_bwd_vs_torch():
+    B, H, T, D = 2, 3, 256, 64
+    scale = 1.0 / (D ** 0.5)
+    for causal in (False, True):
+        q0 = _rt((B, T, H * D)).to(BF)
+        k0 = _rt((B, T, H * D)).to(BF)
+        v0 = _rt((B, T, H * D)).to(BF)
+        g = _rt((B, T, H * D)).to(BF)
+
+        out, m, l = _C.flash_fwd(q0.contiguous(), k0.contiguous(),
+                                 v0.contiguous(), H, scale, causal, 0.0, 0)
+        dq, dk, dv = _C.flash_bwd(q0.contiguous(), k0.contiguous(),
+                                  v0.contiguous(), out, g.contiguous(),
+                                  m, l, H, scale, causal, 0.0, 0)
+
+        q2 = q0.clone().float().requires_grad_(True)
+        k2 = k0.clone().float().requires_grad_(True)
+        v2 = v0.clone().float().requires_grad_(True)
+        qh, kh, vh = _heads(q2, H), _heads(k2, H), _heads(v2, H)
+        s = qh @ kh.transpose(-1, -2) * scale
+        if causal:
+            mask = torch.triu(torch.ones(T, T, device=_dev(),
+                                         dtype=torch.bool), 1)
+            s = s.masked_fill(mask, float("-inf"))
+        p = torch.nn.functional.softmax(s, dim=-1)
+        ref = (p @ vh).transpose(1, 2).reshape(B, T, H * D)
+        ref.backward(g.float())
+
+        assert_close(dq, q2.grad, rtol=6e-2, atol=6e-2,
+                     msg=f"flash dq c={causal}")
+        assert_close(dk, k2.grad, rtol=6e-2, atol=6e-2,
+                     msg=f"flash dk c={causal}")
+        assert_close(dv, v2.grad, rtol=6e-2, atol=6e-2,
+                     msg=f"flash dv c={causal}")
+
+
+def test_flash_mha_autograd_dropout_trains():
+    from anand_mxnet_amd.ops.hip import mha
+    B, H, T, D = 2, 4, 128, 64
+    q = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    k = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    v = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    out = mha(q, k, v, H, True, 0.125, 0.1, True)
+    out.sum().backward()
+    for t in (q, k, v):
+        assert t.grad is not None and torch.isfinite(t.grad.float()).all()
