@@ -387,19 +387,19 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
     const u64* __restrict__ seed_ptr, u64 salt, float pdrop,
     float dscale) {
   constexpr int D = 64;
-  // arena layout (bf16 elems): ktn 8K | ktr 8K | vtn 8K | qtr 8K |
-  // dotr 8K | shared: [qtn 8K + dotn 8K] == [dstr 16K] | ptn 16K
-  __shared__ __align__(16) bf16 arena[8192 * 5 + 16384 + 16384];
-  bf16* ktn = arena;
-  bf16* ktr = arena + 8192;
-  bf16* vtn = arena + 8192 * 2;
-  bf16* qtr = arena + 8192 * 3;
-  bf16* dotr = arena + 8192 * 4;
-  bf16* qtn = arena + 8192 * 5;        // dies after S_T
-  bf16* dotn = arena + 8192 * 5 + 8192;  // dies after dP_T
-  bf16* dstr = arena + 8192 * 5;       // aliases qtn+dotn (32 KiB)
-  bf16* ptn = arena + 8192 * 7;        // pd then ds (TN image, 32 KiB)
-  __shared__ float stats[3][128];      // m, l, Drow for the q-chunk
+  // q processed in 64-row chunks so no LDS region needs to alias:
+  // K/V resident (48 KiB) + per-chunk q/dO images (32 KiB) + pd/ds
+  // images (32 KiB) = 112 KiB.
+  __shared__ __align__(16) bf16 ktn[128 * D];
+  __shared__ __align__(16) bf16 ktr[128 * D];
+  __shared__ __align__(16) bf16 vtn[128 * D];
+  __shared__ __align__(16) bf16 qtn[64 * D];
+  __shared__ __align__(16) bf16 qtr[64 * D];
+  __shared__ __align__(16) bf16 dotn[64 * D];
+  __shared__ __align__(16) bf16 dotr[64 * D];
+  __shared__ __align__(16) bf16 ptn[128 * 64];   // pd, [tk][64 q]
+  __shared__ __align__(16) bf16 dstr[128 * 64];  // ds, tr image [tk][64 q]
+  __shared__ float stats[3][64];                 // m, l, Drow per q-chunk
 
   const int z = blockIdx.z;
   const int b = z / H, h = z % H;
@@ -411,12 +411,12 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
   const u64 seed = seed_ptr ? seed_finalize(*seed_ptr + salt) : 0;
   const u32 thresh = (u32)(pdrop * 4294967296.0f);
 
-  const int srow = (((t >> 6) * 2) << 3) + ((t & 63) >> 3);
-  const int scg = (ll & 7) ^ ((ll >> 3) & 7);           // TN source chunk
-  const int tcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);    // tr source chunk
+  const int srow = (((t >> 6) * 2) << 3) + ((t & 63) >> 3);  // 128-row
+  const int srow64 = ((t >> 6) << 3) + ((t & 63) >> 3);      // 64-row
+  const int scg = (ll & 7) ^ ((ll >> 3) & 7);
+  const int tcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);
 
-  // one 16-KiB [128][64] image stage: 2 glds per thread
-  auto stage64 = [&](bf16* img, const bf16* base, int row0, int cg) {
+  auto stage128 = [&](bf16* img, const bf16* base, int row0, int cg) {
 #pragma unroll
     for (int p = 0; p < 2; ++p) {
       int row = srow + p * 8;
@@ -430,20 +430,29 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
           16, 0, 0);
     }
   };
+  auto stage64r = [&](bf16* img, const bf16* base, int row0, int cg) {
+    unsigned lbase =
+        __builtin_amdgcn_readfirstlane((unsigned)(lw * 1024));
+    auto lds = (__attribute__((address_space(3))) unsigned int*)(
+        (__attribute__((address_space(3))) char*)img + lbase);
+    const bf16* gp = base + (long)(row0 + srow64) * hd + cg * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+        0, 0);
+  };
 
   const bf16* qbase = Q + (long)b * Tq * hd + (long)h * D;
   const bf16* kbase = K + (long)b * Tk * hd + (long)h * D;
   const bf16* vbase = V + (long)b * Tk * hd + (long)h * D;
   const bf16* dobase = dO + (long)b * Tq * hd + (long)h * D;
 
-  // K/V resident for the whole block
-  stage64(ktn, kbase, kvb * 128, scg);
-  stage64(ktr, kbase, kvb * 128, tcg);
-  stage64(vtn, vbase, kvb * 128, scg);
+  stage128(ktn, kbase, kvb * 128, scg);
+  stage128(ktr, kbase, kvb * 128, tcg);
+  stage128(vtn, vbase, kvb * 128, scg);
 
-  const int wmq = (lw & 3) * 32;   // 32 tk rows (TN phases) / 32 q (WG)
-  const int wnt = (lw >> 2) * 64;  // 64-col half (q dim in TN phases)
-  const int wno = (lw >> 2) * 32;  // 32-col half (d dim)
+  const int wmq = (lw & 3) * 32;   // 32 tk rows of this wave
+  const int wnt = (lw >> 2) * 32;  // 32 q cols (of the 64-chunk)
+  const int wno = (lw >> 2) * 32;  // 32 d cols
   const int lrow = ll & 15, lk = ll >> 4;
 
   f32x4 acc_dv[2][2], acc_dk[2][2];
@@ -455,115 +464,110 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
       acc_dk[mf][nf] = f32x4{0, 0, 0, 0};
     }
 
-  const int qc0 = causal ? kvb : 0;
-  const int nq = Tq / 128;
+  const int qc0 = causal ? kvb * 2 : 0;
+  const int nq = Tq / 64;
 
   for (int qc = qc0; qc < nq; ++qc) {
-    // ---- stage q-chunk images + stats ----
-    stage64(qtn, qbase, qc * 128, scg);
-    stage64(qtr, qbase, qc * 128, tcg);
-    stage64(dotn, dobase, qc * 128, scg);
-    stage64(dotr, dobase, qc * 128, tcg);
-    for (int i = t; i < 128; i += 512) {
-      long qg = (long)z * Tq + qc * 128 + i;
+    stage64r(qtn, qbase, qc * 64, scg);
+    stage64r(qtr, qbase, qc * 64, tcg);
+    stage64r(dotn, dobase, qc * 64, scg);
+    stage64r(dotr, dobase, qc * 64, tcg);
+    for (int i = t; i < 64; i += 512) {
+      long qg = (long)z * Tq + qc * 64 + i;
       stats[0][i] = Ms[qg];
       stats[1][i] = Ls[qg];
       stats[2][i] = Drow[qg];
     }
     __syncthreads();
 
-    // ---- S_T = K Q^T and dP_T = V dO^T, processed in two 32-q-column
-    // passes so only [2][2] accumulators are live (the full [2][4] pair
-    // spilled); pd -> ptn (TN image for dV), ds -> dstr (tr image) ----
-#pragma unroll 1
-    for (int qh = 0; qh < 2; ++qh) {
-      f32x4 acc_st[2][2], acc_dp[2][2];
+    // ---- S_T = K Q^T and dP_T = V dO^T ([128 tk][64 q]) ----
+    f32x4 acc_st[2][2], acc_dp[2][2];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        acc_st[mf][nf] = f32x4{0, 0, 0, 0};
+        acc_dp[mf][nf] = f32x4{0, 0, 0, 0};
+      }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 ak[2], bq[2];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        ak[mf] = *(const bf16x8*)&ktn[fa_qk_off(wmq + mf * 16 + lrow,
+                                                chunk)];
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        bq[nf] = *(const bf16x8*)&qtn[fa_qk_off(wnt + nf * 16 + lrow,
+                                                chunk)];
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
-        for (int nf = 0; nf < 2; ++nf) {
-          acc_st[mf][nf] = f32x4{0, 0, 0, 0};
-          acc_dp[mf][nf] = f32x4{0, 0, 0, 0};
-        }
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-        bf16x8 ak[2], bq[2];
-        int chunk = kk * 4 + lk;
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
-          ak[mf] = *(const bf16x8*)&ktn[fa_qk_off(wmq + mf * 16 + lrow,
-                                                  chunk)];
-#pragma unroll
         for (int nf = 0; nf < 2; ++nf)
-          bq[nf] = *(const bf16x8*)&qtn[fa_qk_off(
-              wnt + (qh * 2 + nf) * 16 + lrow, chunk)];
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
-#pragma unroll
-          for (int nf = 0; nf < 2; ++nf)
-            acc_st[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                ak[mf], bq[nf], acc_st[mf][nf], 0, 0, 0);
-      }
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-        bf16x8 av[2], bdo[2];
-        int chunk = kk * 4 + lk;
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
-          av[mf] = *(const bf16x8*)&vtn[fa_qk_off(wmq + mf * 16 + lrow,
-                                                  chunk)];
-#pragma unroll
-        for (int nf = 0; nf < 2; ++nf)
-          bdo[nf] = *(const bf16x8*)&dotn[fa_qk_off(
-              wnt + (qh * 2 + nf) * 16 + lrow, chunk)];
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
-#pragma unroll
-          for (int nf = 0; nf < 2; ++nf)
-            acc_dp[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                av[mf], bdo[nf], acc_dp[mf][nf], 0, 0, 0);
-      }
-#pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
-#pragma unroll
-        for (int nf = 0; nf < 2; ++nf) {
-          int qcol = wnt + (qh * 2 + nf) * 16 + (ll & 15);
-          long qg = (long)qc * 128 + qcol;
-          float m = stats[0][qcol], linv = stats[1][qcol];
-          linv = linv > 0.f ? 1.f / linv : 0.f;
-          float dr = stats[2][qcol];
-#pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            int tkrow = wmq + mf * 16 + (ll >> 4) * 4 + i;
-            long kg = (long)kvb * 128 + tkrow;
-            float sv = acc_st[mf][nf][i] * alpha;
-            float pv = (causal && kg > qg) ? 0.f : __expf(sv - m) * linv;
-            float keep = 1.f;
-            if (thresh) {
-              u32 r = hash_u32(seed, ((long)z * Tq + qg) * Tk + kg);
-              keep = r >= thresh ? dscale : 0.f;
-            }
-            float pd = pv * keep;
-            float dp = acc_dp[mf][nf][i] * keep;
-            float ds = pv * (dp - dr) * alpha;
-            *(bf16*)((char*)ptn + fa_p_byte(tkrow, qcol >> 3) +
-                     (qcol & 7) * 2) = f2bf(pd);
-            *(bf16*)((char*)dstr + fa_tr128_byte(tkrow, qcol * 2)) =
-                f2bf(ds);
-          }
-        }
+          acc_st[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ak[mf], bq[nf], acc_st[mf][nf], 0, 0, 0);
     }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 av[2], bdo[2];
+      int chunk = kk * 4 + lk;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        av[mf] = *(const bf16x8*)&vtn[fa_qk_off(wmq + mf * 16 + lrow,
+                                                chunk)];
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        bdo[nf] = *(const bf16x8*)&dotn[fa_qk_off(wnt + nf * 16 + lrow,
+                                                  chunk)];
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc_dp[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              av[mf], bdo[nf], acc_dp[mf][nf], 0, 0, 0);
+    }
+
+    // ---- p, pd, ds; pd -> ptn (TN image), ds -> dstr (tr image) ----
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int qcol = wnt + nf * 16 + (ll & 15);
+        long qg = (long)qc * 64 + qcol;
+        float m = stats[0][qcol], linv = stats[1][qcol];
+        linv = linv > 0.f ? 1.f / linv : 0.f;
+        float dr = stats[2][qcol];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int tkrow = wmq + mf * 16 + (ll >> 4) * 4 + i;
+          long kg = (long)kvb * 128 + tkrow;
+          float sv = acc_st[mf][nf][i] * alpha;
+          float pv = (causal && kg > qg) ? 0.f : __expf(sv - m) * linv;
+          float keep = 1.f;
+          if (thresh) {
+            u32 r = hash_u32(seed, ((long)z * Tq + qg) * Tk + kg);
+            keep = r >= thresh ? dscale : 0.f;
+          }
+          float pd = pv * keep;
+          float dp = acc_dp[mf][nf][i] * keep;
+          float ds = pv * (dp - dr) * alpha;
+          *(bf16*)((char*)ptn + fa_qk_off(tkrow, qcol >> 3) * 2 +
+                   (qcol & 7) * 2) = f2bf(pd);
+          *(bf16*)((char*)dstr + fa_v_byte(tkrow, qcol * 2)) = f2bf(ds);
+        }
+      }
     __syncthreads();
 
-    // ---- dV += pd · dO  (A: ptn rows tk, k = q; B: dotr tr image) ----
+    // ---- dV += pd · dO (K dim = 64 q) ----
 #pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {
+    for (int kk = 0; kk < 2; ++kk) {
       bf16x8 af[2], bfr[2];
       int chunk = kk * 4 + lk;
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
-        af[mf] = *(const bf16x8*)((const char*)ptn +
-                                  fa_p_byte(wmq + mf * 16 + lrow, chunk));
+        af[mf] = *(const bf16x8*)&ptn[fa_qk_off(wmq + mf * 16 + lrow,
+                                                chunk)];
       int rbase = kk * 32 + (ll >> 4) * 8 + ((ll & 15) >> 2);
       int cb = ((ll & 15) & 3) * 8;
 #pragma unroll
@@ -587,17 +591,17 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
               af[mf], bfr[nf], acc_dv[mf][nf], 0, 0, 0);
     }
 
-    // ---- dK += ds · Q  (A: dstr rows tk, k = q, b128 reads on the
-    // tr-keyed image — mild bank sharing, acceptable; B: qtr) ----
+    // ---- dK += ds · Q (ds via b128 on the tr-keyed image) ----
 #pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {
+    for (int kk = 0; kk < 2; ++kk) {
       bf16x8 af[2], bfr[2];
-      int chunk = kk * 4 + lk;
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
+      for (int mf = 0; mf < 2; ++mf) {
+        int row = wmq + mf * 16 + lrow;
+        int chunk = kk * 4 + lk;
         af[mf] = *(const bf16x8*)((const char*)dstr +
-                                  fa_tr128_byte(wmq + mf * 16 + lrow,
-                                                chunk * 16));
+                                  fa_v_byte(row, chunk * 16));
+      }
       int rbase = kk * 32 + (ll >> 4) * 8 + ((ll & 15) >> 2);
       int cb = ((ll & 15) & 3) * 8;
 #pragma unroll
@@ -621,32 +625,29 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
               af[mf], bfr[nf], acc_dk[mf][nf], 0, 0, 0);
     }
 
-    // ---- dQ partial: WG over tk — A = dstr [tk][128 q] tr image,
-    //      B = ktr [tk][64 d] tr image; wave = 32 q x 32 d ----
+    // ---- dQ partial: WG over tk (A = dstr tr image [tk][64 q],
+    //      B = ktr [tk][64 d]); wave = 16 q x 32 d ----
     {
-      f32x4 acc_dq[2][2];
+      f32x4 acc_dq[2];
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
-#pragma unroll
-        for (int nf = 0; nf < 2; ++nf) acc_dq[mf][nf] = f32x4{0, 0, 0, 0};
+      for (int nf = 0; nf < 2; ++nf) acc_dq[nf] = f32x4{0, 0, 0, 0};
+      const int wq16 = (lw & 3) * 16;  // Mo offset (64 q total)
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
-        bf16x8 af[2], bfr[2];
+        bf16x8 af, bfr[2];
         int rbase = kk * 32 + (ll >> 4) * 8 + ((ll & 15) >> 2);
         int cb = ((ll & 15) & 3) * 8;
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf) {
-          int cbyte = (wmq + mf * 16) * 2 + cb;
+        {
+          int cbyte = wq16 * 2 + cb;
           auto p1 = (__attribute__((address_space(3))) bf16x4*)(
               (__attribute__((address_space(3))) char*)dstr +
-              fa_tr128_byte(rbase, cbyte));
+              fa_v_byte(rbase, cbyte));
           auto p2 = (__attribute__((address_space(3))) bf16x4*)(
               (__attribute__((address_space(3))) char*)dstr +
-              fa_tr128_byte(rbase + 4, cbyte));
+              fa_v_byte(rbase + 4, cbyte));
           bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
           bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
-          af[mf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6,
-                                           7);
+          af = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
         }
 #pragma unroll
         for (int nf = 0; nf < 2; ++nf) {
@@ -663,26 +664,22 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
                                             7);
         }
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
-#pragma unroll
-          for (int nf = 0; nf < 2; ++nf)
-            acc_dq[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[mf], bfr[nf], acc_dq[mf][nf], 0, 0, 0);
+        for (int nf = 0; nf < 2; ++nf)
+          acc_dq[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfr[nf], acc_dq[nf], 0, 0, 0);
       }
-      float* dqb = dQacc + ((long)b * Tq + qc * 128) * hd + (long)h * D;
+      float* dqb = dQacc + ((long)b * Tq + qc * 64) * hd + (long)h * D;
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
+      for (int nf = 0; nf < 2; ++nf) {
+        int col = wno + nf * 16 + (ll & 15);
 #pragma unroll
-        for (int nf = 0; nf < 2; ++nf) {
-          int col = wno + nf * 16 + (ll & 15);
-#pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
-            atomicAdd(&dqb[(long)row * hd + col], acc_dq[mf][nf][i]);
-          }
+        for (int i = 0; i < 4; ++i) {
+          int row = wq16 + (ll >> 4) * 4 + i;
+          atomicAdd(&dqb[(long)row * hd + col], acc_dq[nf][i]);
         }
+      }
     }
-    __syncthreads();  // qtr/dotr/dstr rewritten next q-chunk
+    __syncthreads();
   }
 
   // ---- epilogue: dK, dV (strided bf16) ----
