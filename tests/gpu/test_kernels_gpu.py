@@ -408,8 +408,8 @@ def test_mha_fwd_bwd_vs_torch():
                                 ("dv", v1.grad, v2.grad)):
             cos = torch.nn.functional.cosine_similarity(
                 got.float().flatten(), want.float().flatten(), dim=0).item()
-            assert cos > 0.995, (name, causal, cos)
-            assert_close(got, want, rtol=0.15, atol=0.1,
+            assert cos > 0.9995, (name, causal, cos)
+            assert_close(got, want, rtol=6e-2, atol=6e-2,
                          msg=f"mha {name} c={causal}")
 
 
@@ -712,18 +712,12 @@ def test_flash_bwd_vs_torch():
         ref = (p @ vh).transpose(1, 2).reshape(B, T, H * D)
         ref.backward(g.float())
 
-        # flash recomputes p from stats and Drow from the bf16 O, which
-        # adds ~1 ulp-of-bf16 noise per row vs the fp32 reference; check
-        # correlation + mean error rather than elementwise max
         for name, got, want in (("dq", dq, q2.grad), ("dk", dk, k2.grad),
                                 ("dv", dv, v2.grad)):
             cos = torch.nn.functional.cosine_similarity(
                 got.float().flatten(), want.flatten(), dim=0).item()
-            assert cos > 0.995, (name, causal, cos)
-            me = (got.float() - want).abs().mean().item()
-            scale_w = want.abs().mean().item() + 1e-6
-            assert me < 0.08 * scale_w + 2e-3, (name, causal, me, scale_w)
-            assert_close(got, want, rtol=0.15, atol=0.12,
+            assert cos > 0.9995, (name, causal, cos)
+            assert_close(got, want, rtol=6e-2, atol=6e-2,
                          msg=f"flash {name} c={causal}")
 
 
