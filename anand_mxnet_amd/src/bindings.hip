@@ -31,8 +31,7 @@ void gemm_tn8_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
                    int, long, long, long, long, long, long, int, bool,
                    const bf16*, hipStream_t);
 // gemm_wg.hip
-int gemm_wg_nsplit(long, int, int);
-void gemm_wg_bf16(const bf16*, const bf16*, float*, float*, long, int, int, long,
+void gemm_wg_bf16(const bf16*, const bf16*, float*, long, int, int, long,
                   long, long, const bf16*, hipStream_t);
 void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, int, int, int, int,
                        int, int, int, int, int, int, int, int, int, int,
@@ -237,16 +236,7 @@ at::Tensor gemm_wg(const at::Tensor& A, const at::Tensor& B) {
   TORCH_CHECK(B.size(0) == m, "gemm_wg: reduce dim mismatch");
   TORCH_CHECK(Mo % 8 == 0 && No % 8 == 0, "gemm_wg: dims must be %8");
   auto out = at::empty({Mo, No}, A.options().dtype(at::kFloat));
-  int nsp = mxa::gemm_wg_nsplit(m, (int)Mo, (int)No);
-  at::Tensor scratch;
-  float* sp = nullptr;
-  if (nsp > 1) {
-    // two-stage split reduce: per-slice partials, no memset/atomics
-    scratch = at::empty({(long)nsp * Mo * No},
-                        A.options().dtype(at::kFloat));
-    sp = scratch.data_ptr<float>();
-  }
-  mxa::gemm_wg_bf16(bfp(A), bfp(B), out.data_ptr<float>(), sp, m, (int)Mo,
+  mxa::gemm_wg_bf16(bfp(A), bfp(B), out.data_ptr<float>(), m, (int)Mo,
                     (int)No, Mo, No, No, zp(A), cur_stream());
   return out;
 }

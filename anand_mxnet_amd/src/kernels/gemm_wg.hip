@@ -378,10 +378,6 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
   bload.base += zb.off(zb_);
   // zc.off() counts OUTPUT ELEMENTS: 2 bytes in bf16 mode, 4 in f32
   Cout = (float*)((char*)Cout + zc.off(zb_) * (OUTBF16 ? 2 : 4));
-  // use_atomic == 2: two-stage split reduce — each y-slice writes its own
-  // scratch image (no pre-zero, no atomic contention); a reduce kernel
-  // sums the slices afterwards
-  if (use_atomic == 2) Cout += (long)blockIdx.y * strideSeg;
 
   const int seg = blockIdx.y / nsplit, split = blockIdx.y % nsplit;
   const int per = (ksteps_total + nsplit - 1) / nsplit;
@@ -562,62 +558,30 @@ static int pick_split(long blocks_xy, long ksteps) {
   return target;
 }
 
-int gemm_wg_nsplit(long mtotal, int Mo, int No) {
+void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, long mtotal,
+                  int Mo, int No, long lda, long ldb, long ldc,
+                  const bf16* zpage, hipStream_t stream) {
   int ksteps = (int)((mtotal + WBK - 1) / WBK);
   long bxy = (long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN);
   int nsplit = pick_split(bxy, ksteps);
-  if (nsplit <= 1) return 1;
-  int per = (ksteps + nsplit - 1) / nsplit;
-  return (ksteps + per - 1) / per;  // effective non-empty slices
-}
-
-__global__ __launch_bounds__(256) void wg_split_reduce(
-    const float* __restrict__ scratch, float* __restrict__ C, long total,
-    int nsplit) {
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
-    float acc = 0.f;
-    for (int s = 0; s < nsplit; ++s) acc += scratch[(long)s * total + i];
-    C[i] = acc;
-  }
-}
-
-void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, float* scratch,
-                  long mtotal, int Mo, int No, long lda, long ldb,
-                  long ldc, const bf16* zpage, hipStream_t stream) {
-  int ksteps = (int)((mtotal + WBK - 1) / WBK);
-  long bxy = (long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN);
-  int nsplit = pick_split(bxy, ksteps);
-  int n_eff = gemm_wg_nsplit(mtotal, Mo, No);
-  bool two_stage = nsplit > 1 && scratch != nullptr;
   dim3 grid((unsigned)bxy, nsplit, 1);
-  int mode = nsplit > 1 ? (two_stage ? 2 : 1) : 0;
-  float* out0 = two_stage ? scratch : C;
-  long slice = (long)Mo * ldc;
-  if (mode == 1)
-    HIP_CHECK(hipMemsetAsync(C, 0, sizeof(float) * slice, stream));
+  int use_atomic = nsplit > 1;
+  if (use_atomic) HIP_CHECK(hipMemsetAsync(C, 0, sizeof(float) * Mo * ldc,
+                                           stream));
   if (!wg_use_zip() && lda % 8 == 0 && ldb % 8 == 0) {
     TrPlainRows a{A, zpage, mtotal, lda, Mo};
     TrPlainRows b{B, zpage, mtotal, ldb, No};
     TrIm2colRows dummy{};
     gemm_wg_tr_core<false><<<grid, WT8, 0, stream>>>(
-        a, dummy, b, out0, Mo, No, ldc, two_stage ? slice : 0, 1, 1, 1,
-        ksteps, nsplit, mode, ZMap{1, 0, 0}, ZMap{1, 0, 0},
-        ZMap{1, 0, 0});
-  } else {
-    WgPlainRows a{A, zpage, mtotal, lda, Mo};
-    WgPlainRows b{B, zpage, mtotal, ldb, No};
-    WgIm2colRows dummy{};
-    if (two_stage) mode = 1;  // zip core has no mode 2: atomics fallback
-    if (two_stage)
-      HIP_CHECK(hipMemsetAsync(C, 0, sizeof(float) * slice, stream));
-    gemm_wg_core<false><<<grid, WTHREADS, 0, stream>>>(
-        a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, mode);
+        a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic,
+        ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0});
     return;
   }
-  if (two_stage)
-    wg_split_reduce<<<ew_grid(slice, 256, 4096), 256, 0, stream>>>(
-        scratch, C, slice, n_eff);
+  WgPlainRows a{A, zpage, mtotal, lda, Mo};
+  WgPlainRows b{B, zpage, mtotal, ldb, No};
+  WgIm2colRows dummy{};
+  gemm_wg_core<false><<<grid, WTHREADS, 0, stream>>>(
+      a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic);
 }
 
 void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
