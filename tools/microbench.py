@@ -121,6 +121,16 @@ def main():
                       (8192, 4096, 1024), (8192, 1024, 4096)]:
             print(json.dumps(bench_mx(*shape)))
         return
+    if "--nn-only" in sys.argv:
+        for (M, K, N) in [(8192, 768, 768), (8192, 768, 3072),
+                          (8192, 3072, 768), (4096, 4096, 4096)]:
+            a = (torch.rand((M, K), device=DEV) * 2 - 1).to(BF)
+            b = (torch.rand((K, N), device=DEV) * 2 - 1).to(BF)
+            t = timeit(lambda: _C.gemm_nn(a, b))
+            print(json.dumps({"op": f"gemm_nn {M}x{N}x{K}",
+                              "ours_TF": round(2.0 * M * N * K / t / 1e12, 1),
+                              "ms": round(t * 1e3, 3)}))
+        return
     if "--wg-only" in sys.argv:
         results.append(bench_wg(100352, 128, 128))
         results.append(bench_wg(25088, 512, 256))
