@@ -171,6 +171,13 @@ class BERTForPretrain(HybridBlock):
     hybrid_forward = None
 
 
+def bert_tiny(vocab_size=30528, max_length=512, dropout=0.1, **kwargs):
+    """CPU-smoke-sized configuration (examples/tests)."""
+    return BERTModel(vocab_size=vocab_size, units=128, num_layers=2,
+                     hidden_size=512, num_heads=2, dropout=dropout,
+                     max_length=max_length, **kwargs)
+
+
 def bert_base(vocab_size=30528, max_length=512, dropout=0.1, **kwargs):
     return BERTModel(vocab_size=vocab_size, units=768, num_layers=12,
                      hidden_size=3072, num_heads=12, dropout=dropout,
