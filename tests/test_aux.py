@@ -173,3 +173,25 @@ def test_custom_op():
     ref = 1 / (1 + np.exp(-x.asnumpy()))
     np.testing.assert_allclose(y.asnumpy(), ref, rtol=1e-5)
     np.testing.assert_allclose(x.grad.asnumpy(), ref * (1 - ref), rtol=1e-5)
+
+
+def test_examples_smoke(tmp_path):
+    """The shipped example CLIs run end-to-end on CPU."""
+    import subprocess
+    import sys as _sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [_sys.executable, os.path.join(root, "examples", "bert",
+                                       "pretrain.py"),
+         "--steps", "2", "--log-interval", "1"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = subprocess.run(
+        [_sys.executable,
+         os.path.join(root, "examples", "image-classification",
+                      "train.py"),
+         "--network", "resnet18_v1", "--data", "synthetic",
+         "--num-epochs", "1", "--batch-size", "8", "--num-examples", "16",
+         "--image-shape", "3,64,64"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
