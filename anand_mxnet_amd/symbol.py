@@ -196,17 +196,11 @@ def _as_sym(x):
     raise TypeError(f"cannot compose Symbol with {type(x)}")
 
 
-_name_counter = {}
-
-
-def _auto_name(op):
-    c = _name_counter.get(op, 0)
-    _name_counter[op] = c + 1
-    return f"{op.lower()}{c}"
-
 
 def _make_op(op, inputs, attrs=None, name=None):
-    return Symbol(op, name or _auto_name(op), inputs, attrs)
+    from . import attribute, name as _nm
+    name = _nm.current().get(name, op.lower())
+    return Symbol(op, name, inputs, attribute.current().get(attrs))
 
 
 def var(name, shape=None, dtype=None, **kwargs):

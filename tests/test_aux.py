@@ -195,3 +195,17 @@ def test_examples_smoke(tmp_path):
          "--image-shape", "3,64,64"],
         capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_name_attribute_libinfo():
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import name as nm, attribute, libinfo
+    with nm.Prefix("stage1_"):
+        s = mx.sym.var("x", shape=(2, 4))
+        y = mx.sym.FullyConnected(s, num_hidden=8)
+    assert y.name.startswith("stage1_")
+    with attribute.AttrScope(ctx_group="dev1"):
+        z = mx.sym.var("w") * mx.sym.var("v")
+    assert z.attrs.get("ctx_group") == "dev1"
+    d = libinfo.feature_dict()
+    assert d["BF16"] and "RCCL" in d
