@@ -514,12 +514,287 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Ring-pipelined tr-variant: 32-row half-tiles on a counted-vmcnt glds ring
+// ---------------------------------------------------------------------------
+// The 2-phase schedule above keeps only ONE 64-row tile in flight while the
+// per-tile MFMA work is short — rocprofv3 PMC showed the wgrad kernels
+// 25% MFMA-busy / 41-52% wave-parked (profiles/RESULTS.md).  Same cure as
+// gemm_tn8.hip: half the staging quantum (32 k-rows = 1 glds per operand
+// per lane), put RING-1 halves in flight, and wait with a COUNTED
+// s_waitcnt vmcnt across raw s_barriers so the wait never drains the
+// younger prefetches.  RING=4 keeps LDS at 64 KiB (2 blocks/CU); RING=8
+// matches tn8's depth at 1 block/CU.
+
+struct TrPlainRows1 {
+  const bf16* base;
+  const bf16* zpage;
+  long mtotal;
+  long ld;
+  int rowlen;
+  // device state
+  const bf16* ptr;
+  long mrem;
+  bool colok;
+  int cgbyte;
+
+  DEVINL void init(int col0, long m0, int myrow, int cg) {
+    cgbyte = cg * 16;
+    int col = col0 + cg * 8;
+    colok = col + 8 <= rowlen;
+    ptr = base + (m0 + myrow) * ld + col;
+    mrem = mtotal - (m0 + myrow);
+  }
+  DEVINL void step() {
+    ptr += 32 * ld;
+    mrem -= 32;
+  }
+  DEVINL const bf16* src() const {
+    return (colok && mrem > 0) ? ptr
+                               : (const bf16*)((const char*)zpage + cgbyte);
+  }
+};
+
+struct TrIm2colRows1 {
+  const bf16* x;
+  const bf16* zpage;
+  int Nb, H, W, C, P, Q;
+  int sh, sw, ph, pw;
+  // device state
+  int roff, soff;
+  int n_, p_, q_;
+  bool colok;
+  int col, cgbyte;
+
+  DEVINL void init(int col0, long m0, int myrow, int cg) {
+    cgbyte = cg * 16;
+    col = col0 + cg * 8;
+    colok = col + 8 <= C;
+    int PQ = P * Q;
+    long m = m0 + myrow;
+    n_ = (int)(m / PQ);
+    int pq = (int)(m % PQ);
+    p_ = pq / Q;
+    q_ = pq % Q;
+  }
+  DEVINL void step() {
+    q_ += 32;
+    while (q_ >= Q) {
+      q_ -= Q;
+      p_ += 1;
+    }
+    while (p_ >= P) {
+      p_ -= P;
+      n_ += 1;
+    }
+  }
+  DEVINL const bf16* src() const {
+    const bf16* zp = (const bf16*)((const char*)zpage + cgbyte);
+    if (n_ >= Nb || !colok) return zp;
+    int ih = p_ * sh - ph + roff;
+    int iw = q_ * sw - pw + soff;
+    if (ih < 0 || ih >= H || iw < 0 || iw >= W) return zp;
+    return x + (((long)n_ * H + ih) * W + iw) * C + col;
+  }
+};
+
+// counted wait: `rem` halves still wanted in flight after this wait
+// (2 glds per half); immediates only, so a small switch.
+DEVINL void wg_ring_wait(int rem) {
+  switch (rem) {
+    case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+    case 1: asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); break;
+    case 2: asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); break;
+    case 3: asm volatile("s_waitcnt vmcnt(6)" ::: "memory"); break;
+    case 4: asm volatile("s_waitcnt vmcnt(8)" ::: "memory"); break;
+    case 5: asm volatile("s_waitcnt vmcnt(10)" ::: "memory"); break;
+    case 6: asm volatile("s_waitcnt vmcnt(12)" ::: "memory"); break;
+    default: asm volatile("s_waitcnt vmcnt(14)" ::: "memory"); break;
+  }
+}
+
+template <bool A_IM2COL, bool OUTBF16, int RING>
+__global__ __launch_bounds__(WT8, RING <= 4 ? 2 : 1) void gemm_wg_ring_core(
+    TrPlainRows1 aplain, TrIm2colRows1 aim, TrPlainRows1 bload,
+    float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
+    int S, int dh, int dw, int hsteps_total, int nsplit, int use_atomic,
+    ZMap za, ZMap zb, ZMap zc) {
+  // [ring slot][op][32 rows][128 cols]
+  __shared__ __align__(16) bf16 smem[RING][2][32 * 128];
+
+  const int zb_ = blockIdx.z;
+  if (A_IM2COL)
+    aim.x += za.off(zb_);
+  else
+    aplain.base += za.off(zb_);
+  bload.base += zb.off(zb_);
+  Cout = (float*)((char*)Cout + zc.off(zb_) * (OUTBF16 ? 2 : 4));
+
+  const int seg = blockIdx.y / nsplit, split = blockIdx.y % nsplit;
+  const int per = (hsteps_total + nsplit - 1) / nsplit;
+  const int h0 = split * per;
+  const int nh = min(per, hsteps_total - h0);
+  if (nh <= 0) return;
+
+  const int ntn = (No + WBN - 1) / WBN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * WBM, bn = (bid % ntn) * WBN;
+  const int t = threadIdx.x;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  // glds slot (8 waves x 32 rows): LDS bytes lw*1024 + ll*16 ->
+  // row = lw*4 + (ll>>4), chunk ll&15; swizzle on the SOURCE chunk.
+  const int cg = (ll & 15) ^ ((ll >> 4) << 1);
+  const int myrow = lw * 4 + (ll >> 4);
+
+  long m0 = (long)h0 * 32;
+  if (A_IM2COL) {
+    aim.roff = (seg / S) * dh;
+    aim.soff = (seg % S) * dw;
+    aim.init(bm, m0, myrow, cg);
+  } else {
+    aplain.init(bm, m0, myrow, cg);
+  }
+  bload.init(bn, m0, myrow, cg);
+
+  auto issue_half = [&](int slot) {
+    unsigned lbase = __builtin_amdgcn_readfirstlane((unsigned)(lw * 1024));
+    auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
+        (__attribute__((address_space(3))) char*)&smem[slot][0][0] + lbase);
+    auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
+        (__attribute__((address_space(3))) char*)&smem[slot][1][0] + lbase);
+    auto gA = (const __attribute__((address_space(1))) unsigned int*)(
+        A_IM2COL ? aim.src() : aplain.src());
+    auto gB =
+        (const __attribute__((address_space(1))) unsigned int*)bload.src();
+    __builtin_amdgcn_global_load_lds(gA, ldsA, 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(gB, ldsB, 16, 0, 0);
+    if (A_IM2COL)
+      aim.step();
+    else
+      aplain.step();
+    bload.step();
+  };
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int wm = (lw & 3) * 32, wn = (lw >> 2) * 64;
+  const int ig = ll & 15, khi = ll >> 4;
+
+  auto trread = [&](const bf16* img, int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)img + tr_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+
+  auto compute_half = [&](int slot) {
+    int rbase = khi * 8 + (ig >> 2);
+    int cb = (ig & 3) * 8;
+    bf16x8 af[2], bfr[4];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+      int cbyte = (wm + mf * 16) * 2 + cb;
+      bf16x4 lo = trread(smem[slot][0], rbase, cbyte);
+      bf16x4 hi = trread(smem[slot][0], rbase + 4, cbyte);
+      af[mf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+    }
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int cbyte = (wn + nf * 16) * 2 + cb;
+      bf16x4 lo = trread(smem[slot][1], rbase, cbyte);
+      bf16x4 hi = trread(smem[slot][1], rbase + 4, cbyte);
+      bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  // prologue: RING-1 halves in flight
+#pragma unroll
+  for (int i = 0; i < RING - 1; ++i)
+    if (i < nh) issue_half(i);
+
+  for (int h = 0; h < nh; ++h) {
+    int ahead = nh - 1 - h;  // younger halves wanted in flight
+    if (ahead >= RING - 1) issue_half((h + RING - 1) % RING);
+    wg_ring_wait(min(ahead, RING - 1));
+    __builtin_amdgcn_s_barrier();
+    compute_half(h % RING);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  float* cseg = Cout + (long)seg * strideSeg;
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wn + nf * 16 + (ll & 15);
+      if (gcol >= No) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (ll >> 4) * 4 + i;
+        if (grow >= Mo) continue;
+        float v = acc[mf][nf][i];
+        if (OUTBF16)
+          ((bf16*)cseg)[(long)grow * ldc + gcol] = f2bf(v);
+        else if (use_atomic)
+          atomicAdd(&cseg[(long)grow * ldc + gcol], v);
+        else
+          cseg[(long)grow * ldc + gcol] = v;
+      }
+    }
+  }
+}
+
 static bool wg_use_zip() {
   static const bool z = [] {
     const char* e = getenv("MXA_WG_ZIP");
     return e && e[0] == '1';
   }();
   return z;
+}
+
+static int wg_ring() {
+  // 0 (default) = 2-phase tr kernel; 4 = 64-KiB ring (2 blocks/CU);
+  // 8 = tn8-depth ring (128 KiB, 1 block/CU).  Measured on MI355X
+  // (profiles/ring{0,4,8}_wg.log): the ring LOSES ~10% (RING=4) to ~2x
+  // (RING=8) on every wgrad shape — the 2-phase schedule already overlaps
+  // one full tile of DMA with compute, and halving the barrier quantum to
+  // 32 rows doubles the s_barrier rate for only 8 MFMA per wave between
+  // barriers.  The small-tile wgrad wait time is split-K epilogue atomics
+  // and launch overhead, not in-loop DMA depth.  Kept as a measured
+  // negative result and for future schedule experiments.
+  static const int r = [] {
+    const char* e = getenv("MXA_WG_RING");
+    return e ? atoi(e) : 0;
+  }();
+  return (r == 4 || r == 8) ? r : 0;
+}
+
+template <bool A_IM2COL, bool OUTBF16>
+static void launch_wg_ring(const TrPlainRows1& a, const TrIm2colRows1& ai,
+                           const TrPlainRows1& b, float* C, int Mo, int No,
+                           long ldc, long strideSeg, int S, int dh, int dw,
+                           int hsteps, int nsplit, int use_atomic, ZMap za,
+                           ZMap zb, ZMap zc, dim3 grid, hipStream_t stream) {
+  if (wg_ring() == 8)
+   hipLaunchKernelGGL(( gemm_wg_ring_core<A_IM2COL, OUTBF16, 8>), dim3(grid), dim3(WT8), 0, stream, 
+        a, ai, b, C, Mo, No, ldc, strideSeg, S, dh, dw, hsteps, nsplit,
+        use_atomic, za, zb, zc);
+  else
+   hipLaunchKernelGGL(( gemm_wg_ring_core<A_IM2COL, OUTBF16, 4>), dim3(grid), dim3(WT8), 0, stream, 
+        a, ai, b, C, Mo, No, ldc, strideSeg, S, dh, dw, hsteps, nsplit,
+        use_atomic, za, zb, zc);
 }
 
 // ---------------------------------------------------------------------------
@@ -570,6 +845,17 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, long mtotal,
   if (use_atomic) HIP_CHECK(hipMemsetAsync(C, 0, sizeof(float) * Mo * ldc,
                                            stream));
   if (!wg_use_zip() && lda % 8 == 0 && ldb % 8 == 0) {
+    if (wg_ring()) {
+      TrPlainRows1 a{A, zpage, mtotal, lda, Mo};
+      TrPlainRows1 b{B, zpage, mtotal, ldb, No};
+      TrIm2colRows1 dummy{};
+      int hsteps = (int)((mtotal + 31) / 32);
+      launch_wg_ring<false, false>(a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1,
+                                   hsteps, nsplit, use_atomic, ZMap{1, 0, 0},
+                                   ZMap{1, 0, 0}, ZMap{1, 0, 0}, grid,
+                                   stream);
+      return;
+    }
     TrPlainRows a{A, zpage, mtotal, lda, Mo};
     TrPlainRows b{B, zpage, mtotal, ldb, No};
     TrIm2colRows dummy{};
@@ -599,6 +885,17 @@ void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
     HIP_CHECK(hipMemsetAsync(dw_rsck, 0,
                              sizeof(float) * (long)R * S * C * Kout, stream));
   if (!wg_use_zip() && C % 8 == 0 && Kout % 8 == 0) {
+    if (wg_ring()) {
+      TrIm2colRows1 a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
+      TrPlainRows1 b{dy, zpage, mtotal, (long)Kout, Kout};
+      TrPlainRows1 dummy{};
+      int hsteps = (int)((mtotal + 31) / 32);
+      launch_wg_ring<true, false>(dummy, a, b, dw_rsck, C, Kout, Kout,
+                                  (long)C * Kout, S, dh, dw_, hsteps, nsplit,
+                                  use_atomic, ZMap{1, 0, 0}, ZMap{1, 0, 0},
+                                  ZMap{1, 0, 0}, grid, stream);
+      return;
+    }
     TrIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
     TrPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
     TrPlainRows dummy{};
@@ -627,15 +924,25 @@ void attn_wg_bf16(const bf16* A, const bf16* Bq, bf16* C, int B, int H,
   long hd = (long)H * D;
   int ksteps = cdiv_h(Tq, WBK);
   long bxy = (long)cdiv_h(Tk, WBM) * cdiv_h(D, WBN);
+  dim3 grid((unsigned)bxy, 1, (unsigned)(B * H));
+  // single reduce slice per z -> no atomics -> bf16 output directly
+  // (saves a full f32->bf16 cast pass over [B, S, H*D] per call)
+  if (wg_ring()) {
+    TrPlainRows1 a{A, zpage, Tq, (long)Tk, Tk};
+    TrPlainRows1 b{Bq, zpage, Tq, hd, D};
+    TrIm2colRows1 dummy{};
+    launch_wg_ring<false, true>(a, dummy, b, (float*)C, Tk, D, hd, 0, 1, 1,
+                                1, cdiv_h(Tq, 32), 1, 0,
+                                ZMap{1, (long)Tq * Tk, 0}, ZMap{H, Tq * hd, D},
+                                ZMap{H, Tk * hd, D}, grid, stream);
+    return;
+  }
   TrPlainRows a{A, zpage, Tq, (long)Tk, Tk};
   // rowlen D (not H*D): the per-head slice is only D columns wide — a
   // wider guard would let edge lanes read past the allocation on the
   // final row of the final batch.
   TrPlainRows b{Bq, zpage, Tq, hd, D};
   TrIm2colRows dummy{};
-  dim3 grid((unsigned)bxy, 1, (unsigned)(B * H));
-  // single reduce slice per z -> no atomics -> bf16 output directly
-  // (saves a full f32->bf16 cast pass over [B, S, H*D] per call)
  hipLaunchKernelGGL(( gemm_wg_tr_core<false, true>), dim3(grid), dim3(WT8), 0, stream, 
       a, dummy, b, (float*)C, Tk, D, hd, 0, 1, 1, 1, ksteps, 1, 0,
       ZMap{1, (long)Tq * Tk, 0}, ZMap{H, Tq * hd, D}, ZMap{H, Tk * hd, D});
