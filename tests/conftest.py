@@ -22,10 +22,16 @@ def pytest_collection_modifyitems(config, items):
 
 @pytest.fixture(autouse=True)
 def with_seed():
-    """Reference pattern: tests/python/unittest/common.py — with_seed().
-    Seed is logged via MXA_TEST_SEED for replay."""
-    seed = int(os.environ.get("MXA_TEST_SEED", "0")) or \
-        random.randint(0, 2 ** 31)
+    """Reference pattern: tests/python/unittest/common.py — with_seed(),
+    but with a FIXED default seed: a random per-test seed made the
+    numeric-gradient checks flake roughly once per hundred suite runs
+    (tolerances are calibrated, the tail isn't).  Set MXA_TEST_SEED=random
+    for fuzzing, or to an integer to replay one."""
+    env = os.environ.get("MXA_TEST_SEED", "")
+    if env == "random":
+        seed = random.randint(0, 2 ** 31)
+    else:
+        seed = int(env) if env else 20240901
     np.random.seed(seed)
     torch.manual_seed(seed)
     random.seed(seed)
