@@ -290,10 +290,24 @@ def get_resnet(version, num_layers, pretrained=False, ctx=None,
     extra = {"stride_in_3x3": stride_in_3x3} if version == 1 else {}
     net = resnet_class(block_class, layers, channels, layout=layout,
                        **extra, **kwargs)
+    if version == 1:
+        _mark_conv_bn_stats(net)
     if pretrained:
         raise RuntimeError("no network access: pretrained weights "
                            "unavailable; load explicit .params instead")
     return net
+
+
+def _mark_conv_bn_stats(block):
+    """In v1 nets every conv output feeds a BatchNorm, so ask the conv
+    kernels to fold the BN per-channel sum/sumsq into their GEMM epilogue
+    (ops/hip conv2d_nhwc collect_stats; the BN picks the sums up from the
+    tensor and skips its stats pass)."""
+    for child in block._children.values():
+        if isinstance(child, nn.Conv2D):
+            child._collect_bn_stats = True
+        else:
+            _mark_conv_bn_stats(child)
 
 
 def resnet18_v1(**kw):

@@ -74,7 +74,9 @@ class Conv2D(HybridBlock):
                             stride=self._strides, dilate=self._dilation,
                             pad=self._padding, num_filter=self._channels,
                             num_group=self._groups, no_bias=bias is None,
-                            layout=self._layout)
+                            layout=self._layout,
+                            collect_bn_stats=getattr(
+                                self, "_collect_bn_stats", False))
         if self.act is not None:
             out = self.act(out)
         return out

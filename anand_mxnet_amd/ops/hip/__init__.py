@@ -144,7 +144,11 @@ class _Conv2dNHWC(torch.autograd.Function):
     """Implicit-GEMM conv; weight logical layout (K, R, S, C)."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, stride, pad, dil):
+    def forward(ctx, x, w, bias, stride, pad, dil, stats_out=None):
+        # stats_out: optional list; when the kernel route supports it, the
+        # per-channel BN sum/sumsq accumulated in the GEMM epilogue is
+        # appended so the caller can hand it to the following BatchNorm
+        # (skips bn_stats_partial's re-read of y; see gemm_tn.hip).
         x = x.contiguous()
         K, R, S, C = w.shape
         ctx.save_for_backward(x, w)
@@ -155,13 +159,25 @@ class _Conv2dNHWC(torch.autograd.Function):
             # the 256-tile 8-phase kernel picks it up (the implicit-GEMM
             # gather loader adds address math for nothing here)
             N, H, W_, _ = x.shape
+            b32 = bias.float() if bias is not None else None
+            if stats_out is not None:
+                res = _C.gemm_tn_stats(x.reshape(N * H * W_, C),
+                                       w.reshape(K, C), b32, False)
+                if len(res) == 2:
+                    stats_out.append(res[1])
+                return res[0].reshape(N, H, W_, K)
             y2 = _C.gemm_tn(x.reshape(N * H * W_, C), w.reshape(K, C),
-                            bias.float() if bias is not None else None,
-                            False)
+                            b32, False)
             return y2.reshape(N, H, W_, K)
         w_rskc = w.permute(1, 2, 0, 3).contiguous()
-        y = _C.conv2d_fwd(x, w_rskc,
-                          bias.float() if bias is not None else None,
+        b32 = bias.float() if bias is not None else None
+        if stats_out is not None:
+            res = _C.conv2d_fwd_stats(x, w_rskc, b32, list(stride),
+                                      list(pad), list(dil), False)
+            if len(res) == 2:
+                stats_out.append(res[1])
+            return res[0]
+        y = _C.conv2d_fwd(x, w_rskc, b32,
                           list(stride), list(pad), list(dil), False)
         return y
 
@@ -349,11 +365,16 @@ class _DwConvNHWC(torch.autograd.Function):
             dw = dwf.permute(2, 0, 1).unsqueeze(-1).to(x.dtype)                 .contiguous()
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy)
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None
+
+
+import os
+
+_CONV_BN_STATS = os.environ.get("MXA_CONV_BN_STATS", "1") != "0"
 
 
 @register_gpu("conv2d_nhwc")
-def conv2d_nhwc(x, w, bias, stride, pad, dil, groups):
+def conv2d_nhwc(x, w, bias, stride, pad, dil, groups, collect_stats=False):
     if groups != 1:
         C_in = x.shape[3]
         if groups == C_in and w.shape[0] == C_in and w.shape[3] == 1                 and C_in % 8 == 0 and tuple(dil) == (1, 1):
@@ -364,7 +385,12 @@ def conv2d_nhwc(x, w, bias, stride, pad, dil, groups):
             "is implemented (docs/PARITY.md)")
     stride, pad, dil = tuple(stride), tuple(pad), tuple(dil)
     if w.shape[3] % 64 == 0:
-        return _Conv2dNHWC.apply(x, w, bias, stride, pad, dil)
+        holder = [] if (collect_stats and _CONV_BN_STATS
+                        and torch.is_grad_enabled()) else None
+        y = _Conv2dNHWC.apply(x, w, bias, stride, pad, dil, holder)
+        if holder:
+            y._mxa_bn_stats = holder[0]
+        return y
     return _Conv2dIm2col.apply(x, w, bias, stride, pad, dil)
 
 
@@ -375,7 +401,7 @@ def conv2d_nhwc(x, w, bias, stride, pad, dil, groups):
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, resid, gamma, beta, rmean, rvar, eps, momentum,
-                training, fuse_relu):
+                training, fuse_relu, psums=None):
         x = x.contiguous()
         g32 = gamma.float().contiguous()
         b32 = beta.float().contiguous()
@@ -383,7 +409,7 @@ class _BatchNormNHWC(torch.autograd.Function):
         if training:
             y, save_mean, save_rstd = _C.bn_fwd_train(
                 x, g32, b32, rmean, rvar, eps, momentum, fuse_relu,
-                resid.contiguous() if fuse_add else None)
+                resid.contiguous() if fuse_add else None, psums)
             if fuse_add:
                 # junction form relu(bn(x)+r): mask comes from y > 0
                 ctx.save_for_backward(x, g32, b32, save_mean, save_rstd, y)
@@ -408,12 +434,12 @@ class _BatchNormNHWC(torch.autograd.Function):
             dx, dgamma, dbeta = _C.bn_bwd(x, dr, g32, b32,
                                           save_mean, save_rstd, False)
             return (dx, dr, dgamma.to(ctx.gdtype), dbeta.to(ctx.gdtype),
-                    None, None, None, None, None, None)
+                    None, None, None, None, None, None, None)
         x, g32, b32, save_mean, save_rstd = ctx.saved_tensors
         dx, dgamma, dbeta = _C.bn_bwd(x, dy.contiguous(), g32, b32,
                                       save_mean, save_rstd, ctx.relu)
         return (dx, None, dgamma.to(ctx.gdtype), dbeta.to(ctx.gdtype), None,
-                None, None, None, None, None)
+                None, None, None, None, None, None)
 
 
 @register_gpu("batchnorm_nhwc")
@@ -430,8 +456,15 @@ def batchnorm_nhwc(x, gamma, beta, rmean, rvar, eps, momentum, training,
         elif fuse_relu:
             out = torch.relu(out)
         return out.permute(0, 2, 3, 1).contiguous()
+    # fused conv->BN statistics: the producing conv left its per-channel
+    # sum/sumsq on the tensor (conv2d_nhwc collect_stats); x.contiguous()
+    # below is a no-op for the conv output, so the attribute's tensor
+    # identity is the one the stats describe.
+    psums = getattr(x, "_mxa_bn_stats", None) if training else None
+    if psums is not None and psums.numel() != 2 * x.shape[-1]:
+        psums = None
     return _BatchNormNHWC.apply(x, addend, gamma, beta, rmean, rvar, eps,
-                                momentum, training, fuse_relu)
+                                momentum, training, fuse_relu, psums)
 
 
 # ---------------------------------------------------------------------------

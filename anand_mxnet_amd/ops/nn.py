@@ -90,7 +90,7 @@ def fp8_linear_op(data, weight, bias=None, num_hidden=None, no_bias=False,
 def convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
                 dilate=(1, 1), pad=(0, 0), num_filter=None, num_group=1,
                 no_bias=False, layout="NCHW", cudnn_tune=None, cudnn_off=False,
-                workspace=None):
+                workspace=None, collect_bn_stats=False):
     """2-D convolution.  GPU path: implicit-GEMM MFMA kernel, NHWC-native.
 
     Weight conventions (reference parity): layout NCHW -> (K, C, R, S);
@@ -100,7 +100,8 @@ def convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
     if layout == "NHWC":
         if _use_hip(data, "conv2d_nhwc") and data.dtype == torch.bfloat16:
             return gpu_impl("conv2d_nhwc")(data, weight, b, stride, pad,
-                                           dilate, num_group)
+                                           dilate, num_group,
+                                           collect_bn_stats)
         # CPU oracle: permute to NCHW for torch
         x = data.permute(0, 3, 1, 2)
         w = weight.permute(0, 3, 1, 2)  # (K,R,S,C) -> (K,C,R,S)

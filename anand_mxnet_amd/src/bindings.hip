@@ -16,13 +16,14 @@ namespace mxa {
 // gemm_tn.hip
 void gemm_tn_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
                   int, long, long, long, long, long, long, int, bool,
-                  const bf16*, hipStream_t);
+                  const bf16*, hipStream_t, float* stats = nullptr);
 void gemm_tn_splitk_bf16(const bf16*, const bf16*, bf16*, float*,
                          const float*, int, int, int, bool, const bf16*,
                          hipStream_t);
-void conv2d_fwd_bf16(const bf16*, const bf16*, bf16*, float*, const float*,
+bool conv2d_fwd_bf16(const bf16*, const bf16*, bf16*, float*, const float*,
                      int, int, int, int, int, int, int, int, int, int, int,
-                     int, int, int, int, bool, const bf16*, hipStream_t);
+                     int, int, int, int, bool, const bf16*, hipStream_t,
+                     float* stats = nullptr);
 void conv2d_dgrad_bf16(const bf16*, const bf16*, bf16*, float*, int, int,
                        int, int, int, int, int, int, int, int, int, int,
                        int, int, int, const bf16*, hipStream_t);
@@ -39,7 +40,8 @@ void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, int, int, int, int,
 // bn_pool.hip
 void bn_fwd_train(const bf16*, bf16*, float*, float*, float*, float*, float*,
                   float*, const float*, const float*, long, int, float,
-                  float, bool, const bf16*, hipStream_t);
+                  float, bool, const bf16*, hipStream_t,
+                  const float* ext_sums = nullptr);
 void bn_fwd_infer(const bf16*, bf16*, const float*, const float*,
                   const float*, const float*, long, int, float, bool,
                   const bf16*, hipStream_t);
@@ -184,8 +186,9 @@ const bf16* zp(const at::Tensor& like) { return bfp(zpage_for(like)); }
 // ---------------------------------------------------------------------------
 
 // C[M,N] = A[M,K] @ B[N,K]^T (+bias) (+relu); 3-d inputs batch over dim 0.
-at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
-                   c10::optional<at::Tensor> bias, bool relu) {
+static at::Tensor gemm_tn_impl(const at::Tensor& A, const at::Tensor& B,
+                               c10::optional<at::Tensor> bias, bool relu,
+                               float* stats) {
   TORCH_CHECK(A.is_contiguous() && B.is_contiguous(), "gemm_tn: contiguous");
   bool batched = A.dim() == 3;
   long M = A.size(-2), K = A.size(-1), N = B.size(-2);
@@ -211,7 +214,7 @@ at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
     // third of the chip to the tail (562 vs 803 TF measured)
     return e ? atol(e) : 1536L;
   }();
-  if (M >= 256 && N >= 256 && blocks8 >= 240 && K >= tn8_mink) {
+  if (M >= 256 && N >= 256 && blocks8 >= 240 && K >= tn8_mink && !stats) {
     // deep-pipelined 256² 8-phase kernel (needs a full chip of tiles;
     // smaller grids run the 128² kernel which fills better)
     mxa::gemm_tn8_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
@@ -224,9 +227,38 @@ at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
     // extra fill buys.  Kept callable, not auto-dispatched.)
     mxa::gemm_tn_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
                       (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
-                      cur_stream());
+                      cur_stream(), stats);
   }
   return out;
+}
+
+at::Tensor gemm_tn(const at::Tensor& A, const at::Tensor& B,
+                   c10::optional<at::Tensor> bias, bool relu) {
+  return gemm_tn_impl(A, B, bias, relu, nullptr);
+}
+
+// gemm_tn with fused per-column BN statistics (the pointwise-conv -> BN
+// fusion): returns {out} when the dispatch route cannot fill stats (the
+// caller's BN then computes them itself), {out, stats[2,N]} when it can.
+// tn8 wins by more than the stats pass saves on the deep-K shapes it
+// owns, so stats are only produced on the 128-tile route when tn8 would
+// not have been chosen anyway.
+std::vector<at::Tensor> gemm_tn_stats(const at::Tensor& A,
+                                      const at::Tensor& B,
+                                      c10::optional<at::Tensor> bias,
+                                      bool relu) {
+  long M = A.size(-2), K = A.size(-1), N = B.size(-2);
+  long blocks8 = ((M + 255) / 256) * ((N + 255) / 256);
+  static const long tn8_mink2 = [] {
+    const char* e = getenv("MXA_TN8_MINK");
+    return e ? atol(e) : 1536L;
+  }();
+  if (A.dim() == 3 ||
+      (M >= 256 && N >= 256 && blocks8 >= 240 && K >= tn8_mink2))
+    return {gemm_tn_impl(A, B, bias, relu, nullptr)};
+  auto stats = at::zeros({2, N}, A.options().dtype(at::kFloat));
+  auto out = gemm_tn_impl(A, B, bias, relu, stats.data_ptr<float>());
+  return {out, stats};
 }
 
 // C[Mo,No] (f32) = sum_m A[m,Mo] * B[m,No]
@@ -321,10 +353,13 @@ at::Tensor cast_to_fp8(const at::Tensor& x, const at::Tensor& scale) {
 // Convolution (NHWC)
 // ---------------------------------------------------------------------------
 
-at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w_rskc,
-                      c10::optional<at::Tensor> bias,
-                      std::vector<long> stride, std::vector<long> pad,
-                      std::vector<long> dil, bool relu) {
+static at::Tensor conv2d_fwd_impl(const at::Tensor& x,
+                                  const at::Tensor& w_rskc,
+                                  c10::optional<at::Tensor> bias,
+                                  std::vector<long> stride,
+                                  std::vector<long> pad,
+                                  std::vector<long> dil, bool relu,
+                                  float* stats, bool* stats_filled) {
   TORCH_CHECK(x.is_contiguous() && w_rskc.is_contiguous());
   int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int R = w_rskc.size(0), S = w_rskc.size(1), Kout = w_rskc.size(2);
@@ -349,10 +384,38 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w_rskc,
     accbuf = at::empty({M, Kout}, x.options().dtype(at::kFloat));
     accptr = accbuf.data_ptr<float>();
   }
-  mxa::conv2d_fwd_bf16(bfp(x), bfp(w_rskc), bfp_mut(y), accptr, bptr, Nb, H,
-                       W, C, Kout, R, S, P, Q, sh, sw, ph, pw, dh, dw, relu,
-                       zp(x), cur_stream());
+  bool filled = mxa::conv2d_fwd_bf16(
+      bfp(x), bfp(w_rskc), bfp_mut(y), accptr, bptr, Nb, H, W, C, Kout, R,
+      S, P, Q, sh, sw, ph, pw, dh, dw, relu, zp(x), cur_stream(), stats);
+  if (stats_filled) *stats_filled = filled;
   return y;
+}
+
+at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w_rskc,
+                      c10::optional<at::Tensor> bias,
+                      std::vector<long> stride, std::vector<long> pad,
+                      std::vector<long> dil, bool relu) {
+  return conv2d_fwd_impl(x, w_rskc, bias, stride, pad, dil, relu, nullptr,
+                         nullptr);
+}
+
+// conv2d_fwd with fused per-channel BN statistics accumulated in the GEMM
+// epilogue: {y, stats[2,K]} when the unsplit 128-tile route ran, {y} when
+// the split-K path was taken (its finish kernel does not fold stats; the
+// caller's BN computes them itself).
+std::vector<at::Tensor> conv2d_fwd_stats(const at::Tensor& x,
+                                         const at::Tensor& w_rskc,
+                                         c10::optional<at::Tensor> bias,
+                                         std::vector<long> stride,
+                                         std::vector<long> pad,
+                                         std::vector<long> dil, bool relu) {
+  int Kout = w_rskc.size(2);
+  auto stats = at::zeros({2, Kout}, x.options().dtype(at::kFloat));
+  bool filled = false;
+  auto y = conv2d_fwd_impl(x, w_rskc, bias, stride, pad, dil, relu,
+                           stats.data_ptr<float>(), &filled);
+  if (!filled) return {y};
+  return {y, stats};
 }
 
 at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w_rsck,
@@ -406,7 +469,8 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                                      at::Tensor running_mean,
                                      at::Tensor running_var, double eps,
                                      double momentum, bool relu,
-                                     const c10::optional<at::Tensor>& resid) {
+                                     const c10::optional<at::Tensor>& resid,
+                                     const c10::optional<at::Tensor>& psums) {
   int C = x.size(-1);
   long M = x.numel() / C;
   auto y = at::empty_like(x);
@@ -414,6 +478,13 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
   auto acc = at::empty({2, C}, opts);   // [psum; psumsq], one memset
   auto psum = acc[0], psumsq = acc[1];
   auto save_mean = at::empty({C}, opts), save_rstd = at::empty({C}, opts);
+  const float* ext = nullptr;
+  if (psums.has_value() && psums->defined()) {
+    TORCH_CHECK(psums->is_contiguous() && psums->numel() == 2 * C &&
+                    psums->scalar_type() == at::kFloat,
+                "bn_fwd_train: psums must be contiguous f32 [2,C]");
+    ext = psums->data_ptr<float>();
+  }
   mxa::bn_fwd_train(bfp(x), bfp_mut(y), psum.data_ptr<float>(),
                     psumsq.data_ptr<float>(), save_mean.data_ptr<float>(),
                     save_rstd.data_ptr<float>(),
@@ -422,7 +493,7 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                     beta.data_ptr<float>(), M, C, (float)eps,
                     (float)momentum, relu,
                     resid.has_value() ? bfp(*resid) : nullptr,
-                    cur_stream());
+                    cur_stream(), ext);
   return {y, save_mean, save_rstd};
 }
 
@@ -997,6 +1068,8 @@ long engine_new_stream(long device) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn", &gemm_tn, "C = A @ B^T (bf16, MFMA)");
+  m.def("gemm_tn_stats", &gemm_tn_stats,
+        "gemm_tn + fused per-column BN sums: {out[,stats]}");
   m.def("gemm_wg", &gemm_wg, "C(f32) = sum_m A[m,:] outer B[m,:]");
   m.def("gemm_tn_fp8", &gemm_tn_fp8);
   m.def("cast_to_fp8", &cast_to_fp8);
@@ -1005,12 +1078,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("Bsc"), py::arg("bias") = py::none(),
         py::arg("relu") = false);
   m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_fwd_stats", &conv2d_fwd_stats,
+        "conv2d_fwd + fused per-channel BN sums: {y[,stats]}");
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("bn_fwd_train", &bn_fwd_train, py::arg("x"), py::arg("gamma"),
         py::arg("beta"), py::arg("running_mean"), py::arg("running_var"),
         py::arg("eps"), py::arg("momentum"), py::arg("relu"),
-        py::arg("resid") = py::none());
+        py::arg("resid") = py::none(), py::arg("psums") = py::none());
   m.def("bn_fwd_infer", &bn_fwd_infer, py::arg("x"), py::arg("gamma"),
         py::arg("beta"), py::arg("rmean"), py::arg("rvar"),
         py::arg("eps"), py::arg("relu"), py::arg("resid") = py::none());

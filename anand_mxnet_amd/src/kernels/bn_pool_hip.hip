@@ -561,16 +561,25 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
                   float* save_mean, float* save_rstd, float* running_mean,
                   float* running_var, const float* gamma, const float* beta,
                   long M, int C, float eps, float momentum, bool relu,
-                  const bf16* resid, hipStream_t s) {
-  // psum/psumsq are one contiguous [2,C] buffer: single fill
-  HIP_CHECK(hipMemsetAsync(psum, 0, 2L * C * sizeof(float), s));
-  int cb = cdiv_h(C, 64);
+                  const bf16* resid, hipStream_t s, const float* ext_sums) {
   int gy = bn_gy(M, C);
-  dim3 g1(cb, gy);
- hipLaunchKernelGGL(( bn_stats_partial), dim3(g1), dim3(256), 0, s, x, psum, psumsq, M, C);
- hipLaunchKernelGGL(( bn_finalize), dim3(cdiv_h(C, 256)), dim3(256), 0, s, 
-      psum, psumsq, save_mean, save_rstd, running_mean, running_var, M, C,
-      gy, eps, momentum, 1);
+  if (ext_sums) {
+    // per-channel sum / sum-of-squares were already accumulated by the
+    // producing conv's GEMM epilogue (gemm_tn.hip fused-stats path):
+    // skip the bn_stats_partial re-read of x entirely.
+   hipLaunchKernelGGL(( bn_finalize), dim3(cdiv_h(C, 256)), dim3(256), 0, s, 
+        ext_sums, ext_sums + C, save_mean, save_rstd, running_mean,
+        running_var, M, C, gy, eps, momentum, 1);
+  } else {
+    // psum/psumsq are one contiguous [2,C] buffer: single fill
+    HIP_CHECK(hipMemsetAsync(psum, 0, 2L * C * sizeof(float), s));
+    int cb = cdiv_h(C, 64);
+    dim3 g1(cb, gy);
+   hipLaunchKernelGGL(( bn_stats_partial), dim3(g1), dim3(256), 0, s, x, psum, psumsq, M, C);
+   hipLaunchKernelGGL(( bn_finalize), dim3(cdiv_h(C, 256)), dim3(256), 0, s, 
+        psum, psumsq, save_mean, save_rstd, running_mean, running_var, M, C,
+        gy, eps, momentum, 1);
+  }
   dim3 g2 = bn_apply_grid(M, C);
   if (resid)
    hipLaunchKernelGGL(( bn_apply_add), dim3(g2), dim3(256), 0, s, x, resid, y, save_mean, save_rstd,

@@ -731,3 +731,85 @@ def test_flash_mha_autograd_dropout_trains():
     out.sum().backward()
     for t in (q, k, v):
         assert t.grad is not None and torch.isfinite(t.grad.float()).all()
+
+
+@pytest.mark.gpu
+def test_conv_bn_fused_stats():
+    """conv2d_fwd_stats / gemm_tn_stats sums must equal the direct
+    per-channel sums of the conv output, and bn_fwd_train(psums=...)
+    must match plain bn_fwd_train."""
+    from anand_mxnet_amd import _C
+    torch.manual_seed(5)
+    dev = "cuda:0"
+    # implicit-GEMM 3x3 route
+    x = (torch.rand(8, 14, 14, 128, device=dev) * 2 - 1).to(torch.bfloat16)
+    w = (torch.rand(3, 3, 128, 128, device=dev) * 0.1).to(torch.bfloat16)
+    res = _C.conv2d_fwd_stats(x, w, None, [1, 1], [1, 1], [1, 1], False)
+    y = res[0]
+    assert len(res) == 2, "unsplit conv route should fill stats"
+    stats = res[1]
+    y2 = y.reshape(-1, 128).float()
+    ref_s = y2.sum(0)
+    ref_q = (y2 * y2).sum(0)
+    assert torch.allclose(stats[0], ref_s, rtol=1e-3, atol=2.0)
+    assert torch.allclose(stats[1], ref_q, rtol=1e-3, atol=2.0)
+    # pointwise route via gemm_tn_stats
+    a = (torch.rand(4096, 256, device=dev) * 2 - 1).to(torch.bfloat16)
+    b = (torch.rand(128, 256, device=dev) * 2 - 1).to(torch.bfloat16)
+    res = _C.gemm_tn_stats(a, b, None, False)
+    assert len(res) == 2
+    o2 = res[0].float()
+    assert torch.allclose(res[1][0], o2.sum(0), rtol=1e-3, atol=2.0)
+    assert torch.allclose(res[1][1], (o2 * o2).sum(0), rtol=1e-3, atol=4.0)
+    # bn consuming psums == bn computing its own stats
+    g = torch.rand(128, device=dev) + 0.5
+    bt = torch.rand(128, device=dev) - 0.5
+    rm1, rv1 = torch.zeros(128, device=dev), torch.ones(128, device=dev)
+    rm2, rv2 = torch.zeros(128, device=dev), torch.ones(128, device=dev)
+    stats = _C.conv2d_fwd_stats(x, w, None, [1, 1], [1, 1], [1, 1],
+                                False)[1]
+    out_a = _C.bn_fwd_train(y, g, bt, rm1, rv1, 1e-5, 0.9, True, None, None)
+    out_b = _C.bn_fwd_train(y, g, bt, rm2, rv2, 1e-5, 0.9, True, None,
+                            stats)
+    assert torch.allclose(out_a[0].float(), out_b[0].float(),
+                          rtol=1e-2, atol=1e-2)
+    assert torch.allclose(rm1, rm2, rtol=1e-4, atol=1e-4)
+    assert torch.allclose(rv1, rv2, rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.gpu
+def test_resnet_fused_stats_end_to_end():
+    """ResNet-50 fwd/bwd with conv->BN fused stats matches the same model
+    with MXA_CONV_BN_STATS disabled (same seed, same grads)."""
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import autograd, nd
+
+    def run(collect):
+        torch.manual_seed(11)
+        import numpy as np
+        np.random.seed(11)
+        net = mx.gluon.model_zoo.vision.resnet18_v1(layout="NHWC")
+        if not collect:
+            def unmark(b):
+                for c in b._children.values():
+                    if hasattr(c, "_collect_bn_stats"):
+                        c._collect_bn_stats = False
+                    unmark(c)
+            unmark(net)
+        net.initialize(ctx=mx.gpu(0))
+        net.cast("bfloat16")
+        x = nd.array(np.random.uniform(-1, 1, (8, 64, 64, 3)),
+                     ctx=mx.gpu(0), dtype="bfloat16")
+        with autograd.record():
+            out = net(x)
+            loss = (out.astype("float32") ** 2).mean()
+        loss.backward()
+        g = net.collect_params()["resnetv10_conv1_weight"].grad()
+        return loss.asnumpy(), g.asnumpy()
+
+    l1, g1 = run(True)
+    l2, g2 = run(False)
+    assert abs(float(l1) - float(l2)) < 1e-3
+    import numpy as np
+    assert np.allclose(g1.astype(np.float32), g2.astype(np.float32),
+                       rtol=1e-2, atol=1e-3)
