@@ -804,7 +804,7 @@ def test_resnet_fused_stats_end_to_end():
             out = net(x)
             loss = (out.astype("float32") ** 2).mean()
         loss.backward()
-        g = net.collect_params()["resnetv10_conv1_weight"].grad()
+        g = net.collect_params()["resnetv10_stage1_conv2d0_weight"].grad()
         return loss.asnumpy(), g.asnumpy()
 
     l1, g1 = run(True)
