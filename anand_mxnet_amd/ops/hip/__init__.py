@@ -272,7 +272,7 @@ class _Conv2dNHWC(torch.autograd.Function):
                 dw = dwf.permute(3, 0, 1, 2).to(w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy)
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None
 
 
 class _Conv2dIm2col(torch.autograd.Function):
@@ -365,7 +365,7 @@ class _DwConvNHWC(torch.autograd.Function):
             dw = dwf.permute(2, 0, 1).unsqueeze(-1).to(x.dtype)                 .contiguous()
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy)
-        return dx, dw, db, None, None, None, None
+        return dx, dw, db, None, None, None
 
 
 import os
