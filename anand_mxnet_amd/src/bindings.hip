@@ -256,7 +256,7 @@ std::vector<at::Tensor> gemm_tn_stats(const at::Tensor& A,
   if (A.dim() == 3 ||
       (M >= 256 && N >= 256 && blocks8 >= 240 && K >= tn8_mink2))
     return {gemm_tn_impl(A, B, bias, relu, nullptr)};
-  auto stats = at::zeros({2, N}, A.options().dtype(at::kFloat));
+  auto stats = at::zeros({32, 2, N}, A.options().dtype(at::kFloat));
   auto out = gemm_tn_impl(A, B, bias, relu, stats.data_ptr<float>());
   return {out, stats};
 }
@@ -410,7 +410,7 @@ std::vector<at::Tensor> conv2d_fwd_stats(const at::Tensor& x,
                                          std::vector<long> pad,
                                          std::vector<long> dil, bool relu) {
   int Kout = w_rskc.size(2);
-  auto stats = at::zeros({2, Kout}, x.options().dtype(at::kFloat));
+  auto stats = at::zeros({32, 2, Kout}, x.options().dtype(at::kFloat));
   bool filled = false;
   auto y = conv2d_fwd_impl(x, w_rskc, bias, stride, pad, dil, relu,
                            stats.data_ptr<float>(), &filled);
@@ -480,9 +480,9 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
   auto save_mean = at::empty({C}, opts), save_rstd = at::empty({C}, opts);
   const float* ext = nullptr;
   if (psums.has_value() && psums->defined()) {
-    TORCH_CHECK(psums->is_contiguous() && psums->numel() == 2 * C &&
+    TORCH_CHECK(psums->is_contiguous() && psums->numel() == 64 * C &&
                     psums->scalar_type() == at::kFloat,
-                "bn_fwd_train: psums must be contiguous f32 [2,C]");
+                "bn_fwd_train: psums must be contiguous f32 [32,2,C]");
     ext = psums->data_ptr<float>();
   }
   mxa::bn_fwd_train(bfp(x), bfp_mut(y), psum.data_ptr<float>(),

@@ -65,11 +65,17 @@ __global__ void bn_finalize(const float* __restrict__ psum,
                             float* __restrict__ running_mean,
                             float* __restrict__ running_var, long M, int C,
                             int gy, float eps, float momentum,
-                            int update_running) {
+                            int update_running, int nslots,
+                            long slot_stride) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = psum[c] / M;
-  float var = psumsq[c] / M - mean * mean;
+  float s = 0.f, q = 0.f;
+  for (int k = 0; k < nslots; ++k) {
+    s += psum[(long)k * slot_stride + c];
+    q += psumsq[(long)k * slot_stride + c];
+  }
+  float mean = s / M;
+  float var = q / M - mean * mean;
   if (var < 0) var = 0;
   save_mean[c] = mean;
   save_rstd[c] = rsqrtf(var + eps);
@@ -564,11 +570,12 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
   int gy = bn_gy(M, C);
   if (ext_sums) {
     // per-channel sum / sum-of-squares were already accumulated by the
-    // producing conv's GEMM epilogue (gemm_tn.hip fused-stats path):
-    // skip the bn_stats_partial re-read of x entirely.
+    // producing conv's GEMM epilogue (gemm_tn.hip fused-stats path) in
+    // STATS_SLOTS slot copies: fold + finalize, skipping the
+    // bn_stats_partial re-read of x entirely.
     bn_finalize<<<cdiv_h(C, 256), 256, 0, s>>>(
         ext_sums, ext_sums + C, save_mean, save_rstd, running_mean,
-        running_var, M, C, gy, eps, momentum, 1);
+        running_var, M, C, gy, eps, momentum, 1, STATS_SLOTS, 2L * C);
   } else {
     // psum/psumsq are one contiguous [2,C] buffer: single fill
     HIP_CHECK(hipMemsetAsync(psum, 0, 2L * C * sizeof(float), s));
@@ -577,7 +584,7 @@ void bn_fwd_train(const bf16* x, bf16* y, float* psum, float* psumsq,
     bn_stats_partial<<<g1, 256, 0, s>>>(x, psum, psumsq, M, C);
     bn_finalize<<<cdiv_h(C, 256), 256, 0, s>>>(
         psum, psumsq, save_mean, save_rstd, running_mean, running_var, M, C,
-        gy, eps, momentum, 1);
+        gy, eps, momentum, 1, 1, 0);
   }
   dim3 g2 = bn_apply_grid(M, C);
   if (resid)

@@ -8,6 +8,8 @@
 #include <hip/hip_bf16.h>
 
 #define WAVE 64
+// slot count for the fused BN-stats atomic accumulators ([SLOTS][2][C])
+#define STATS_SLOTS 32
 #define DEVINL __device__ __forceinline__
 
 typedef __bf16 bf16;

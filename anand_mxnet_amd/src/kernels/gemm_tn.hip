@@ -348,12 +348,15 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_tn_core(
 
   // Fused BN statistics (reference parity: the separate
   // batch_norm mean/var pass in nn/batch_norm.cu): per-column
-  // sum/sum-of-squares of the values just written, accumulated into
-  // stats[0..N) / stats[N..2N) with one pair of global atomics per
-  // column per block.  Saves the bn_stats_partial re-read of the conv
-  // output (~3.5% of a ResNet-50 step).  Host only passes `stats` for
-  // unbatched, unsplit launches.
+  // sum/sum-of-squares of the values just written, accumulated with one
+  // pair of global atomics per column per block into one of
+  // STATS_SLOTS copies of [2][N] (slot by block id — thousands of
+  // blocks hammering a single 2-KiB accumulator serialized on L2;
+  // bn_finalize folds the slots).  Saves the bn_stats_partial re-read
+  // of the conv output.  Host only passes `stats` for unbatched,
+  // unsplit launches.
   if (!ACCF32 && stats != nullptr) {
+    stats += (size_t)(bid & (STATS_SLOTS - 1)) * 2 * N;
     float* ssum = (float*)&smem[0][0][0];  // loop is done; smem is free
     float* ssq = ssum + BN;
     if (t < BN) {

@@ -742,25 +742,26 @@ def test_conv_bn_fused_stats():
     torch.manual_seed(5)
     dev = "cuda:0"
     # implicit-GEMM 3x3 route
-    x = (torch.rand(8, 14, 14, 128, device=dev) * 2 - 1).to(torch.bfloat16)
+    x = (torch.rand(32, 28, 28, 128, device=dev) * 2 - 1).to(torch.bfloat16)
     w = (torch.rand(3, 3, 128, 128, device=dev) * 0.1).to(torch.bfloat16)
     res = _C.conv2d_fwd_stats(x, w, None, [1, 1], [1, 1], [1, 1], False)
     y = res[0]
     assert len(res) == 2, "unsplit conv route should fill stats"
-    stats = res[1]
+    stats = res[1].reshape(-1, 2, 128).sum(0)  # fold atomic slots
     y2 = y.reshape(-1, 128).float()
     ref_s = y2.sum(0)
     ref_q = (y2 * y2).sum(0)
-    assert torch.allclose(stats[0], ref_s, rtol=1e-3, atol=2.0)
-    assert torch.allclose(stats[1], ref_q, rtol=1e-3, atol=2.0)
+    assert torch.allclose(stats[0], ref_s, rtol=1e-3, atol=4.0)
+    assert torch.allclose(stats[1], ref_q, rtol=1e-3, atol=8.0)
     # pointwise route via gemm_tn_stats
     a = (torch.rand(4096, 256, device=dev) * 2 - 1).to(torch.bfloat16)
     b = (torch.rand(128, 256, device=dev) * 2 - 1).to(torch.bfloat16)
     res = _C.gemm_tn_stats(a, b, None, False)
     assert len(res) == 2
     o2 = res[0].float()
-    assert torch.allclose(res[1][0], o2.sum(0), rtol=1e-3, atol=2.0)
-    assert torch.allclose(res[1][1], (o2 * o2).sum(0), rtol=1e-3, atol=4.0)
+    st = res[1].reshape(-1, 2, 128).sum(0)
+    assert torch.allclose(st[0], o2.sum(0), rtol=1e-3, atol=4.0)
+    assert torch.allclose(st[1], (o2 * o2).sum(0), rtol=1e-3, atol=8.0)
     # bn consuming psums == bn computing its own stats
     g = torch.rand(128, device=dev) + 0.5
     bt = torch.rand(128, device=dev) - 0.5
@@ -804,7 +805,8 @@ def test_resnet_fused_stats_end_to_end():
             out = net(x)
             loss = (out.astype("float32") ** 2).mean()
         loss.backward()
-        g = net.collect_params()["resnetv10_stage1_conv2d0_weight"].grad()
+        g = [v for k, v in net.collect_params().items()
+                 if k.endswith("_stage1_conv2d0_weight")][0].grad()
         return loss.asnumpy(), g.asnumpy()
 
     l1, g1 = run(True)
