@@ -370,7 +370,12 @@ class _DwConvNHWC(torch.autograd.Function):
 
 import os
 
-_CONV_BN_STATS = os.environ.get("MXA_CONV_BN_STATS", "1") != "0"
+# conv->BN fused statistics: measured NEGATIVE on MI355X (profiles/
+# RESULTS.md): the cross-block per-column atomic reduction in the GEMM
+# epilogue costs more than the near-roofline bn_stats_partial re-read it
+# saves (7142 vs 7328 samples/s on the b256 flagship even with 32-way
+# slotted accumulators).  Machinery kept correct + tested; off by default.
+_CONV_BN_STATS = os.environ.get("MXA_CONV_BN_STATS", "0") == "1"
 
 
 @register_gpu("conv2d_nhwc")
