@@ -407,19 +407,38 @@ __global__ __launch_bounds__(256) void maxpool_bwd(
 // output row (n, p): the R input rows are staged once through LDS
 // (x read once per (n,p) instead of once per output element) and the
 // whole [Q][kpadded] slab is emitted with bf16x8 stores.
+// Table-driven gather: the k -> (r, s, c) unpack is FIXED across every
+// output row, so it is computed once per block into an LDS offset table
+// (the per-element %C / /C / %S divisions made the first version
+// VALU-bound at ~19x the HBM roofline).  The staged rows carry zeroed
+// left/right column pads sized for the filter overhang, and OOB input
+// rows stage as zeros, so the inner loop is branch-free: one LDS read
+// per output element at a precomputed offset.
 __global__ __launch_bounds__(256) void im2col_stem(
     const bf16* __restrict__ x, bf16* __restrict__ out, int Nb, int H,
     int W, int C, int P, int Q, int R, int S, int sh, int sw, int ph,
-    int pw, int dh, int dw, int kdim, int kpadded) {
-  extern __shared__ bf16 xrow[];  // [R][W*C]
+    int pw, int dh, int dw, int kdim, int kpadded, int lpad, int wpc) {
+  extern __shared__ bf16 xrow[];  // [R][wpc] rows, then int koff[kpadded]
+  int* koff = (int*)&xrow[(long)R * wpc];
   const int WC = W * C;
+  for (int k = threadIdx.x; k < kpadded; k += 256) {
+    int c = k % C, rs = k / C;
+    int sx = rs % S, r = rs / S;
+    koff[k] = r * wpc + sx * dw * C + c + lpad;
+  }
+  // zero the column pads once; staging only ever writes the interior
+  for (int i = threadIdx.x; i < R * wpc; i += 256) {
+    int col = i % wpc;
+    if (col < lpad || col >= lpad + WC) xrow[i] = (bf16)0;
+  }
+  const int wcw = (WC + 1) / 2;  // u32 words per interior row
+  const int kv = kpadded / 8;
+  __syncthreads();
   for (long np = blockIdx.x; np < (long)Nb * P; np += gridDim.x) {
     int p = (int)(np % P);
     int n = (int)(np / P);
     int ih0 = p * sh - ph;
-    // stage R input rows (u32 copies, zero-fill OOB rows)
     const bf16* xn = x + (long)n * H * WC;
-    int wcw = (WC + 1) / 2;  // u32 words per row
     for (int idx = threadIdx.x; idx < R * wcw; idx += 256) {
       int r = idx / wcw, wd = idx % wcw;
       int ih = ih0 + r * dh;
@@ -433,31 +452,26 @@ __global__ __launch_bounds__(256) void im2col_stem(
           v = lo;
         }
       }
-      ((u32*)&xrow[0])[r * wcw + wd] = v;
+      // lpad is even -> 4-byte aligned u32 store
+      *(u32*)&xrow[r * wpc + lpad + wd * 2] = v;
     }
     __syncthreads();
     // emit [Q][kpadded]
     bf16* orow = out + np * (long)Q * kpadded;
-    int kv = kpadded / 8;
     for (int i = threadIdx.x; i < Q * kv; i += 256) {
       int q = i / kv;
       int j0 = (i % kv) * 8;
-      int iw0 = q * sw - pw;
+      int base = (q * sw - pw) * C;
       bf16x8 o;
+      if (j0 + 8 <= kdim) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        int k = j0 + e;
-        float v = 0.f;
-        if (k < kdim) {
-          int c = k % C;
-          int rs = k / C;
-          int sx = rs % S, r = rs / S;
-          int iw = iw0 + sx * dw;
-          int ih = ih0 + r * dh;
-          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-            v = bf2f(xrow[r * WC + iw * C + c]);
+        for (int e = 0; e < 8; ++e) o[e] = xrow[koff[j0 + e] + base];
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          int k = j0 + e;
+          o[e] = k < kdim ? xrow[koff[k] + base] : (bf16)0;
         }
-        o[e] = f2bf(v);
       }
       *(bf16x8*)&orow[(long)q * kpadded + j0] = o;
     }
@@ -654,10 +668,18 @@ void im2col_stem_bf16(const bf16* x, bf16* out, int Nb, int H, int W,
                       hipStream_t s) {
   long blocks = (long)Nb * P;
   if (blocks > 16384) blocks = 16384;
-  int lds = R * (((W * C + 1) / 2) * 2) * sizeof(bf16) + 16;
+  // padded row image: left pad covers the filter's left overhang (pw),
+  // right pad its right overhang; lpad even keeps u32 staging aligned
+  int lpad = pw * C;
+  lpad += lpad & 1;
+  int iw_max = (Q - 1) * sw - pw + (S - 1) * dw;
+  int rext = iw_max >= W ? (iw_max - W + 1) * C : 0;
+  int wpc = lpad + W * C + rext;
+  wpc += wpc & 1;
+  int lds = R * wpc * sizeof(bf16) + kpadded * sizeof(int);
   im2col_stem<<<dim3((unsigned)blocks), 256, lds, s>>>(
       x, out, Nb, H, W, C, P, Q, R, S, sh, sw, ph, pw, dh, dw, kdim,
-      kpadded);
+      kpadded, lpad, wpc);
 }
 
 void maxpool_nhwc_bwd(const bf16* dy, const unsigned char* arg, bf16* dx,
