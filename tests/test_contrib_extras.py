@@ -142,3 +142,42 @@ def test_sync_batchnorm_two_procs():
         p.join(110)
     for p in procs:
         assert p.exitcode == 0, f"syncbn worker failed: {p.exitcode}"
+
+
+def test_svrg_module():
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.contrib.svrg_optimization import SVRGModule
+
+    np.random.seed(0)
+    data = mx.sym.var("data")
+    out = mx.sym.FullyConnected(
+        data=data, weight=mx.sym.var("fc_weight"),
+        bias=mx.sym.var("fc_bias"), num_hidden=4, name="fc")
+    out = mx.sym.SoftmaxOutput(out, mx.sym.var("softmax_label"),
+                               name="softmax")
+    mod = SVRGModule(out, data_names=("data",),
+                     label_names=("softmax_label",), context=mx.cpu(),
+                     update_freq=1)
+    x = np.random.normal(size=(32, 8)).astype("float32")
+    y = np.random.randint(0, 4, size=(32,)).astype("float32")
+    it = mx.io.NDArrayIter(x, y, batch_size=8, label_name="softmax_label")
+    mod.fit(it, num_epoch=2, optimizer="sgd",
+            optimizer_params={"learning_rate": 0.1})
+    assert mod._full_grads is not None and len(mod._full_grads) > 0
+
+
+def test_runtime_and_tensorboard(tmp_path):
+    from anand_mxnet_amd import runtime
+    feats = runtime.Features()
+    assert feats.is_enabled("BF16")
+    from anand_mxnet_amd.contrib.tensorboard import LogMetricsCallback
+    import anand_mxnet_amd as mx
+
+    class P:
+        eval_metric = mx.metric.Accuracy()
+    cb = LogMetricsCallback(str(tmp_path))
+    import numpy as np
+    P.eval_metric.update(mx.nd.array([0, 1]), mx.nd.array(
+        np.array([[0.9, 0.1], [0.2, 0.8]], dtype="float32")))
+    cb(P)
