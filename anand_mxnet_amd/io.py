@@ -12,6 +12,7 @@ from .base import MXNetError
 from .ndarray.ndarray import NDArray, array
 
 __all__ = ["DataDesc", "DataBatch", "DataIter", "NDArrayIter", "ResizeIter",
+           "LibSVMIter", "ImageRecordIter",
            "PrefetchingIter", "MNISTIter", "CSVIter"]
 
 
@@ -243,3 +244,47 @@ class CSVIter(NDArrayIter):
             label = np.loadtxt(label_csv, delimiter=",", dtype=np.float32)
             label = label.reshape((-1,) + tuple(label_shape))
         super().__init__(data, label, batch_size, **kwargs)
+
+
+class LibSVMIter(NDArrayIter):
+    """LibSVM-format reader (reference: src/io/iter_libsvm.cc).
+
+    The reference yields CSRNDArray batches; CSR math is out of scope
+    here (docs/PARITY.md), so rows are densified — same iterator
+    contract, dense storage."""
+
+    def __init__(self, data_libsvm, data_shape, label_shape=(1,),
+                 batch_size=1, **kwargs):
+        dim = int(np.prod(data_shape))
+        rows, labels = [], []
+        with open(data_libsvm) as f:
+            for line in f:
+                parts = line.split()
+                if not parts:
+                    continue
+                labels.append(float(parts[0]))
+                row = np.zeros(dim, dtype=np.float32)
+                for tok in parts[1:]:
+                    k, v = tok.split(":")
+                    row[int(k)] = float(v)
+                rows.append(row)
+        data = np.stack(rows).reshape((-1,) + tuple(data_shape))
+        label = np.asarray(labels, dtype=np.float32) \
+            .reshape((-1,) + tuple(label_shape))
+        super().__init__(data, label, batch_size, **kwargs)
+
+
+def ImageRecordIter(path_imgrec, data_shape, batch_size, shuffle=False,
+                    preprocess_threads=None, rand_crop=False,
+                    rand_mirror=False, mean_r=0.0, mean_g=0.0, mean_b=0.0,
+                    label_width=1, **kwargs):
+    """RecordIO image iterator (reference API: mx.io.ImageRecordIter from
+    src/io/iter_image_recordio_2.cc).  Maps the reference's flat kwargs
+    onto the Python ImageIter pipeline over the same .rec format."""
+    from .image import ImageIter, CreateAugmenter
+    aug = CreateAugmenter(data_shape, rand_crop=rand_crop,
+                          rand_mirror=rand_mirror,
+                          mean=np.array([mean_r, mean_g, mean_b])
+                          if (mean_r or mean_g or mean_b) else None)
+    return ImageIter(batch_size, data_shape, path_imgrec=path_imgrec,
+                     shuffle=shuffle, aug_list=aug)

@@ -153,3 +153,30 @@ def test_mnist_iter_from_raw(tmp_path):
     b = next(it)
     assert b.data[0].shape == (2, 1, 28, 28)
     assert b.label[0].asnumpy().tolist() == [0, 1]
+
+
+def test_libsvm_and_imagerecord_iters(tmp_path):
+    import numpy as np
+    import anand_mxnet_amd as mx
+    p = tmp_path / "d.libsvm"
+    p.write_text("1 0:0.5 3:1.5\n0 1:2.0\n1 2:0.25 3:0.75\n")
+    it = mx.io.LibSVMIter(str(p), data_shape=(4,), batch_size=3)
+    b = next(iter(it))
+    assert b.data[0].shape == (3, 4)
+    assert float(b.data[0].asnumpy()[1, 1]) == 2.0
+    # ImageRecordIter over an im2rec-style .rec
+    from anand_mxnet_amd import recordio
+    rec = str(tmp_path / "data.rec")
+    idx = str(tmp_path / "data.idx")
+    w = recordio.MXIndexedRecordIO(idx, rec, "w")
+    import io as _io
+    for i in range(4):
+        buf = _io.BytesIO()
+        np.save(buf, (np.random.rand(8, 8, 3) * 255).astype("uint8"))
+        hdr = recordio.IRHeader(0, float(i % 2), i, 0)
+        w.write_idx(i, recordio.pack(hdr, buf.getvalue()))
+    w.close()
+    rit = mx.io.ImageRecordIter(path_imgrec=rec, data_shape=(3, 8, 8),
+                                batch_size=2)
+    batch = next(iter(rit))
+    assert batch.data[0].shape == (2, 3, 8, 8)
