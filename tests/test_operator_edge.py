@@ -1,0 +1,153 @@
+"""Operator edge coverage (reference parity:
+tests/python/unittest/test_operator.py — the long tail beyond the
+hot-path ops covered in test_ndarray/test_gluon)."""
+import numpy as np
+import pytest
+
+import anand_mxnet_amd as mx
+from anand_mxnet_amd import nd
+
+
+def _np(x):
+    return x.asnumpy()
+
+
+def test_pad_tile_repeat_flip():
+    a = nd.array(np.arange(12, dtype="float32").reshape(1, 1, 3, 4))
+    p = nd.pad(a, mode="constant", pad_width=(0, 0, 0, 0, 1, 1, 2, 2),
+               constant_value=7.0)
+    assert p.shape == (1, 1, 5, 8)
+    assert float(_np(p)[0, 0, 0, 0]) == 7.0
+    assert float(_np(p)[0, 0, 1, 2]) == 0.0
+    t = nd.tile(nd.array([[1, 2]]), reps=(2, 3))
+    assert t.shape == (2, 6)
+    r = nd.repeat(nd.array([1, 2, 3]), repeats=2)
+    np.testing.assert_allclose(_np(r), [1, 1, 2, 2, 3, 3])
+    f = nd.flip(nd.array([[1, 2], [3, 4]]), axis=1)
+    np.testing.assert_allclose(_np(f), [[2, 1], [4, 3]])
+
+
+def test_topk_sort_argsort():
+    x = nd.array([[3.0, 1.0, 2.0], [0.0, 5.0, 4.0]])
+    t = nd.topk(x, k=2, axis=1, ret_typ="value")
+    np.testing.assert_allclose(_np(t), [[3, 2], [5, 4]])
+    s = nd.sort(x, axis=1)
+    np.testing.assert_allclose(_np(s), [[1, 2, 3], [0, 4, 5]])
+    a = nd.argsort(x, axis=1)
+    np.testing.assert_allclose(_np(a), [[1, 2, 0], [0, 2, 1]])
+
+
+def test_one_hot_pick_diag():
+    oh = nd.one_hot(nd.array([0, 2]), depth=3)
+    np.testing.assert_allclose(_np(oh), [[1, 0, 0], [0, 0, 1]])
+    x = nd.array([[1.0, 2.0], [3.0, 4.0]])
+    p = nd.pick(x, nd.array([1, 0]), axis=1)
+    np.testing.assert_allclose(_np(p), [2, 3])
+    d = nd.diag(x)
+    np.testing.assert_allclose(_np(d), [1, 4])
+
+
+def test_sequence_ops():
+    # [seq, batch, feat] layout (reference SequenceMask semantics)
+    x = nd.array(np.arange(2 * 3 * 1, dtype="float32").reshape(3, 2, 1)
+                 + 1.0)
+    m = nd.SequenceMask(x, sequence_length=nd.array([2, 1]),
+                        use_sequence_length=True, value=-1.0)
+    out = _np(m)
+    assert out[2, 0, 0] == -1.0 and out[1, 1, 0] == -1.0
+    assert out[1, 0, 0] != -1.0
+    last = nd.SequenceLast(x, sequence_length=nd.array([2, 1]),
+                           use_sequence_length=True)
+    np.testing.assert_allclose(_np(last).ravel(),
+                               [_np(x)[1, 0, 0], _np(x)[0, 1, 0]])
+    rev = nd.SequenceReverse(x, sequence_length=nd.array([2, 3]),
+                             use_sequence_length=True)
+    assert _np(rev)[0, 0, 0] == _np(x)[1, 0, 0]
+
+
+def test_gather_scatter_boolean():
+    x = nd.array(np.arange(6, dtype="float32").reshape(2, 3))
+    idx = nd.array(np.array([[0, 1], [2, 0]]), dtype="int64")
+    g = nd.gather_nd(x, idx)
+    np.testing.assert_allclose(_np(g), [2.0, 3.0])
+    s = nd.scatter_nd(nd.array([9.0, 8.0]), idx, shape=(2, 3))
+    assert _np(s)[0, 2] == 9.0 and _np(s)[1, 0] == 8.0
+    bm = nd.boolean_mask(x, nd.array([0, 1]))
+    np.testing.assert_allclose(_np(bm), [[3, 4, 5]])
+
+
+def test_norm_l2_normalize_smoothl1():
+    x = nd.array([[3.0, 4.0], [6.0, 8.0]])
+    n = nd.norm(x, ord=2, axis=1)
+    np.testing.assert_allclose(_np(n), [5, 10])
+    l2 = nd.L2Normalization(x, mode="instance")
+    np.testing.assert_allclose(_np(l2)[0], [0.6, 0.8], rtol=1e-5)
+    s = nd.smooth_l1(nd.array([-2.0, 0.5]), scalar=1.0)
+    np.testing.assert_allclose(_np(s), [1.5, 0.125], rtol=1e-5)
+
+
+def test_norm_layers():
+    x = nd.array(np.random.RandomState(0)
+                 .normal(size=(2, 4, 3)).astype("float32"))
+    ga = nd.ones((4,))
+    be = nd.zeros((4,))
+    out = nd.InstanceNorm(x, ga, be, eps=1e-5)
+    m = _np(out).mean(axis=2)
+    np.testing.assert_allclose(m, np.zeros_like(m), atol=1e-5)
+    g = nd.GroupNorm(x.reshape(2, 4, 3, 1), nd.ones((4,)), nd.zeros((4,)),
+                     num_groups=2, eps=1e-5)
+    assert g.shape == (2, 4, 3, 1)
+    r = nd.RMSNorm(x, nd.ones((3,)), eps=1e-6)
+    ref = _np(x) / np.sqrt((_np(x) ** 2).mean(-1, keepdims=True) + 1e-6)
+    np.testing.assert_allclose(_np(r), ref, rtol=1e-4)
+
+
+def test_leakyrelu_modes():
+    x = nd.array([-1.0, 2.0])
+    lk = nd.LeakyReLU(x, act_type="leaky", slope=0.1)
+    np.testing.assert_allclose(_np(lk), [-0.1, 2.0], rtol=1e-6)
+    el = nd.LeakyReLU(x, act_type="elu", slope=1.0)
+    np.testing.assert_allclose(_np(el), [np.expm1(-1.0), 2.0], rtol=1e-5)
+    pr = nd.LeakyReLU(x, nd.array([0.25]), act_type="prelu")
+    np.testing.assert_allclose(_np(pr), [-0.25, 2.0], rtol=1e-6)
+
+
+def test_broadcast_axis_slice_like_batch_take():
+    x = nd.array(np.arange(4, dtype="float32").reshape(1, 4))
+    b = nd.broadcast_axis(x, axis=0, size=3)
+    assert b.shape == (3, 4)
+    y = nd.zeros((2, 3))
+    sl = nd.slice_like(nd.array(np.arange(12).reshape(3, 4)), y)
+    assert sl.shape == (2, 3)
+    bt = nd.batch_take(nd.array([[1.0, 2.0], [3.0, 4.0]]),
+                       nd.array([1, 0], dtype="int64"))
+    np.testing.assert_allclose(_np(bt), [2, 3])
+
+
+def test_arctan2_hypot_erf():
+    a = nd.array([1.0, -1.0])
+    b = nd.array([1.0, 1.0])
+    np.testing.assert_allclose(_np(nd.arctan2(a, b)),
+                               np.arctan2([1, -1], [1, 1]), rtol=1e-6)
+    np.testing.assert_allclose(_np(nd.broadcast_hypot(a, b)),
+                               np.hypot([1, -1], [1, 1]), rtol=1e-6)
+    np.testing.assert_allclose(_np(nd.erf(a)),
+                               [0.8427008, -0.8427008], rtol=1e-5)
+    np.testing.assert_allclose(_np(nd.erfinv(nd.array([0.5]))),
+                               [0.4769363], rtol=1e-4)
+
+
+def test_linalg_ops():
+    rs = np.random.RandomState(1)
+    a = rs.normal(size=(3, 3)).astype("float32")
+    spd = a @ a.T + 3 * np.eye(3, dtype="float32")
+    l = nd.linalg_potrf(nd.array(spd))
+    np.testing.assert_allclose(_np(l) @ _np(l).T, spd, rtol=1e-4,
+                               atol=1e-4)
+    g = nd.linalg_gemm2(nd.array(a), nd.array(a), transpose_b=True)
+    np.testing.assert_allclose(_np(g), a @ a.T, rtol=1e-5, atol=1e-5)
+    # solve L x = b via trsm
+    bvec = rs.normal(size=(3, 1)).astype("float32")
+    x = nd.linalg_trsm(l, nd.array(bvec), rightside=False, lower=True)
+    np.testing.assert_allclose(_np(l) @ _np(x), bvec, rtol=1e-4,
+                               atol=1e-4)
