@@ -179,3 +179,52 @@ def test_sequential_module_train():
             seq.update()
             seq.update_metric(metric, batch.label)
     assert metric.get()[1] > 0.8, metric.get()
+
+
+def test_bucketing_module():
+    """BucketingModule shares params across bucket lengths (reference:
+    module/bucketing_module.py; RNN bucketing example pattern)."""
+    import numpy as np
+    from anand_mxnet_amd.module.bucketing_module import BucketingModule
+    from anand_mxnet_amd.io import DataBatch
+
+    def sym_gen(seq_len):
+        data = sym.var("data")
+        w = sym.var("fc_weight")
+        b = sym.var("fc_bias")
+        h = sym.FullyConnected(data=data, weight=w, bias=b, num_hidden=4,
+                               name="fc")
+        label = sym.var("softmax_label")
+        out = sym.SoftmaxOutput(h, label, name="softmax")
+        return out, ("data",), ("softmax_label",)
+
+    bm = BucketingModule(sym_gen, default_bucket_key=8, context=mx.cpu())
+    bm.bind(data_shapes=[("data", (2, 8))],
+            label_shapes=[("softmax_label", (2,))])
+    bm.init_params(initializer=mx.init.Xavier())
+    bm.init_optimizer(optimizer="sgd",
+                      optimizer_params={"learning_rate": 0.1})
+
+    rs = np.random.RandomState(0)
+
+    def batch(n_feat):
+        b = DataBatch(data=[mx.nd.array(rs.normal(size=(2, n_feat)))],
+                      label=[mx.nd.array([0.0, 1.0])],
+                      provide_data=[("data", (2, n_feat))],
+                      provide_label=[("softmax_label", (2,))])
+        b.bucket_key = n_feat
+        return b
+
+    # default bucket
+    bm.forward(batch(8), is_train=True)
+    bm.backward()
+    bm.update()
+    out8 = bm.get_outputs()[0].asnumpy()
+    assert out8.shape == (2, 4)
+    # same-width second bucket shares (and continues training) the params
+    bm.forward(batch(8), is_train=True)
+    bm.backward()
+    bm.update()
+    arg, _ = bm.get_params()
+    assert "fc_weight" in arg
+    assert len(bm._buckets) == 1
