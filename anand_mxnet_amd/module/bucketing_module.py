@@ -68,6 +68,12 @@ class BucketingModule(BaseModule):
                             allow_missing=False, force_init=True)
             if self.optimizer_initialized and not mod.optimizer_initialized:
                 mod.init_optimizer(**self._opt_kwargs)
+                # all buckets must share ONE optimizer + updater state
+                # (reference: a single updater over the shared params —
+                # per-bucket momentum would diverge otherwise).  Param
+                # order matches across buckets (same symbol structure).
+                mod._optimizer = self._curr_module._optimizer
+                mod._updater = self._curr_module._updater
         self._curr_module = mod
 
     def forward(self, data_batch, is_train=None):

@@ -228,3 +228,51 @@ def test_bucketing_module():
     arg, _ = bm.get_params()
     assert "fc_weight" in arg
     assert len(bm._buckets) == 1
+
+
+def test_bucketing_module_switch_sync():
+    """Params flow across bucket switches and the optimizer state is
+    SHARED (reference: one updater over shared params)."""
+    import numpy as np
+    from anand_mxnet_amd.module.bucketing_module import BucketingModule
+    from anand_mxnet_amd.io import DataBatch
+
+    def sym_gen(key):
+        data = sym.var("data")
+        h = sym.FullyConnected(data=data, weight=sym.var("fc_weight"),
+                               bias=sym.var("fc_bias"), num_hidden=4,
+                               name="fc")
+        return (sym.SoftmaxOutput(h, sym.var("softmax_label"),
+                                  name="softmax"),
+                ("data",), ("softmax_label",))
+
+    bm = BucketingModule(sym_gen, default_bucket_key=8, context=mx.cpu())
+    bm.bind(data_shapes=[("data", (2, 8))],
+            label_shapes=[("softmax_label", (2,))])
+    bm.init_params(initializer=mx.init.Xavier())
+    bm.init_optimizer(optimizer="sgd",
+                      optimizer_params={"learning_rate": 0.1,
+                                        "momentum": 0.9})
+
+    def batch(key):
+        b = DataBatch(data=[mx.nd.array(np.ones((2, 8), dtype="float32"))],
+                      label=[mx.nd.array([0.0, 1.0])],
+                      provide_data=[("data", (2, 8))],
+                      provide_label=[("softmax_label", (2,))])
+        b.bucket_key = key
+        return b
+
+    bm.forward(batch(8), is_train=True)
+    bm.backward()
+    bm.update()
+    bm.forward(batch(99), is_train=True)
+    bm.backward()
+    bm.update()
+    assert sorted(bm._buckets) == [8, 99]
+    assert bm._buckets[8]._updater is bm._buckets[99]._updater
+    # switching back syncs bucket 99's training into bucket 8
+    bm.forward(batch(8), is_train=True)
+    a8, _ = bm._buckets[8].get_params()
+    a99, _ = bm._buckets[99].get_params()
+    np.testing.assert_allclose(a8["fc_weight"].asnumpy(),
+                               a99["fc_weight"].asnumpy())
