@@ -443,3 +443,108 @@ def rnn(data, parameters, state, state_cell=None, state_size=None,
     raise NotImplementedError(
         "fused RNN op not implemented; use gluon.rnn cells "
         "(reference rnn.cu is cuDNN-bound; see docs/PARITY.md)")
+
+
+# ---------------- misc nn long tail (reference src/operator/) ----------------
+
+@register("UpSampling")
+def upsampling(data, scale=2, sample_type="nearest", num_filter=0,
+               num_args=1):
+    """UpSampling (reference: src/operator/upsampling.cc).  NCHW."""
+    mode = "nearest" if sample_type == "nearest" else "bilinear"
+    return F.interpolate(data, scale_factor=float(scale), mode=mode,
+                         **({} if mode == "nearest"
+                            else {"align_corners": False}))
+
+
+@register("LRN")
+def lrn(data, alpha=1e-4, beta=0.75, knorm=2.0, nsize=5):
+    """Local response normalization (reference: src/operator/lrn.cc).
+    MXNet alpha is TOTAL (torch divides by size internally)."""
+    return F.local_response_norm(data, size=int(nsize),
+                                 alpha=float(alpha), beta=float(beta),
+                                 k=float(knorm))
+
+
+@register("GridGenerator")
+def grid_generator(data, transform_type="affine", target_shape=(0, 0)):
+    """Affine grid (reference: src/operator/grid_generator.cc).  Takes
+    [N, 6] affine params, emits [N, 2, H, W] sampling grid in MXNet's
+    channel-first convention."""
+    if transform_type != "affine":
+        raise NotImplementedError("GridGenerator: warp type not supported")
+    n = data.shape[0]
+    theta = data.reshape(n, 2, 3)
+    h, w = int(target_shape[0]), int(target_shape[1])
+    grid = F.affine_grid(theta, [n, 1, h, w], align_corners=False)
+    return grid.permute(0, 3, 1, 2)  # [N, 2, H, W], (x, y) channels
+
+
+@register("BilinearSampler")
+def bilinear_sampler(data, grid, cudnn_off=False):
+    """Bilinear sampling (reference: src/operator/bilinear_sampler.cu).
+    data NCHW, grid [N, 2, H, W] -> NCHW output."""
+    g = grid.permute(0, 2, 3, 1)  # to torch's [N, H, W, 2]
+    return F.grid_sample(data, g, mode="bilinear", padding_mode="zeros",
+                         align_corners=False)
+
+
+@register("SpatialTransformer")
+def spatial_transformer(data, loc, target_shape=(0, 0),
+                        transform_type="affine", sampler_type="bilinear",
+                        cudnn_off=False):
+    """STN = GridGenerator + BilinearSampler (reference:
+    src/operator/spatial_transformer.cu)."""
+    grid = grid_generator(loc, transform_type, target_shape)
+    return bilinear_sampler(data, grid)
+
+
+@register("ROIPooling")
+def roi_pooling(data, rois, pooled_size=(1, 1), spatial_scale=1.0):
+    """ROI max pooling (reference: src/operator/roi_pooling.cu).
+    rois [R, 5] = (batch_idx, x1, y1, x2, y2) in image coordinates."""
+    ph, pw = int(pooled_size[0]), int(pooled_size[1])
+    n_roi = rois.shape[0]
+    out = data.new_zeros((n_roi, data.shape[1], ph, pw))
+    H, W = data.shape[2], data.shape[3]
+    r = rois.detach().cpu()
+    for i in range(n_roi):
+        b = int(r[i, 0])
+        x1 = int(torch.round(r[i, 1] * spatial_scale))
+        y1 = int(torch.round(r[i, 2] * spatial_scale))
+        x2 = int(torch.round(r[i, 3] * spatial_scale))
+        y2 = int(torch.round(r[i, 4] * spatial_scale))
+        x1, y1 = max(x1, 0), max(y1, 0)
+        x2, y2 = min(max(x2, x1), W - 1), min(max(y2, y1), H - 1)
+        patch = data[b:b + 1, :, y1:y2 + 1, x1:x2 + 1]
+        out[i] = F.adaptive_max_pool2d(patch, (ph, pw))[0]
+    return out
+
+
+@register("ctc_loss")
+@register("CTCLoss")
+def ctc_loss(data, label, data_lengths=None, label_lengths=None,
+             use_data_lengths=False, use_label_lengths=False,
+             blank_label="first"):
+    """CTC loss (reference: src/operator/contrib/ctc_loss.cc — warp-ctc).
+    data [seq, batch, alphabet] unnormalized; label [batch, L] padded
+    with 0 (blank='first' convention) or -1."""
+    T, N, A = data.shape
+    log_probs = F.log_softmax(data.float(), dim=-1)
+    if blank_label == "first":
+        blank = 0
+        lab = label.long()
+    else:
+        blank = A - 1
+        lab = label.long()
+    if use_data_lengths and data_lengths is not None:
+        in_lens = data_lengths.long()
+    else:
+        in_lens = torch.full((N,), T, dtype=torch.long)
+    if use_label_lengths and label_lengths is not None:
+        tgt_lens = label_lengths.long()
+    else:
+        tgt_lens = (lab > 0).sum(dim=1).long() if blank == 0 else \
+            (lab >= 0).sum(dim=1).long()
+    return F.ctc_loss(log_probs, lab, in_lens, tgt_lens, blank=blank,
+                      reduction="none", zero_infinity=True)

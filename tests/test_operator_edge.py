@@ -151,3 +151,45 @@ def test_linalg_ops():
     x = nd.linalg_trsm(l, nd.array(bvec), rightside=False, lower=True)
     np.testing.assert_allclose(_np(l) @ _np(x), bvec, rtol=1e-4,
                                atol=1e-4)
+
+
+def test_upsampling_lrn():
+    x = nd.array(np.arange(16, dtype="float32").reshape(1, 1, 4, 4))
+    u = nd.UpSampling(x, scale=2, sample_type="nearest")
+    assert u.shape == (1, 1, 8, 8)
+    assert float(_np(u)[0, 0, 0, 1]) == 0.0 and \
+        float(_np(u)[0, 0, 0, 2]) == 1.0
+    l = nd.LRN(x, nsize=3)
+    assert l.shape == x.shape
+
+
+def test_spatial_transformer_identity():
+    x = nd.array(np.random.RandomState(0)
+                 .normal(size=(2, 3, 5, 5)).astype("float32"))
+    loc = nd.array(np.tile(np.array([1, 0, 0, 0, 1, 0], dtype="float32"),
+                           (2, 1)))
+    out = nd.SpatialTransformer(x, loc, target_shape=(5, 5))
+    np.testing.assert_allclose(_np(out), _np(x), rtol=1e-4, atol=1e-4)
+    grid = nd.GridGenerator(loc, transform_type="affine",
+                            target_shape=(5, 5))
+    assert grid.shape == (2, 2, 5, 5)
+    samp = nd.BilinearSampler(x, grid)
+    np.testing.assert_allclose(_np(samp), _np(x), rtol=1e-4, atol=1e-4)
+
+
+def test_roi_pooling():
+    x = nd.array(np.arange(16, dtype="float32").reshape(1, 1, 4, 4))
+    rois = nd.array(np.array([[0, 0, 0, 3, 3], [0, 2, 2, 3, 3]],
+                             dtype="float32"))
+    out = nd.ROIPooling(x, rois, pooled_size=(2, 2), spatial_scale=1.0)
+    np.testing.assert_allclose(_np(out)[0].ravel(), [5, 7, 13, 15])
+    np.testing.assert_allclose(_np(out)[1].ravel(), [10, 11, 14, 15])
+
+
+def test_ctc_loss():
+    rs = np.random.RandomState(3)
+    data = nd.array(rs.normal(size=(10, 2, 5)).astype("float32"))
+    lab = nd.array(np.array([[1, 2, 0], [3, 0, 0]], dtype="float32"))
+    c = nd.ctc_loss(data, lab)
+    assert c.shape == (2,)
+    assert np.all(_np(c) > 0)
