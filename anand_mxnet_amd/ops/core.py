@@ -662,3 +662,71 @@ def linalg_trsm(a, b, transpose=False, rightside=False, lower=True, alpha=1.0):
 def diag(x, k=0):
     return torch.diagonal(x, offset=k, dim1=-2, dim2=-1) if x.dim() >= 2 \
         else torch.diag(x, k)
+
+
+@register("index_copy")
+def index_copy(old, index, new_tensor, out=None):
+    """Copy rows of new_tensor into old at index (reference:
+    src/operator/contrib/index_copy.cc)."""
+    res = old.clone()
+    res.index_copy_(0, index.long(), new_tensor.to(old.dtype))
+    return res
+
+
+@register("box_nms")
+def box_nms(data, overlap_thresh=0.5, valid_thresh=0.0, topk=-1,
+            coord_start=2, score_index=1, id_index=-1, force_suppress=False,
+            in_format="corner", out_format="corner", background_id=-1):
+    """Non-maximum suppression over [..., N, K] boxes (reference:
+    src/operator/contrib/bounding_box.cc).  Suppressed entries get
+    score -1 (reference convention); boxes kept in score order."""
+    shape = data.shape
+    flat = data.reshape(-1, shape[-2], shape[-1]).clone()
+    for b in range(flat.shape[0]):
+        page = flat[b]
+        scores = page[:, score_index]
+        order = torch.argsort(scores, descending=True)
+        if topk > 0:
+            order = order[:topk]
+        keep = []
+        for i in order.tolist():
+            if scores[i] <= valid_thresh:
+                continue
+            box_i = page[i, coord_start:coord_start + 4]
+            if in_format == "center":
+                box_i = torch.stack([box_i[0] - box_i[2] / 2,
+                                     box_i[1] - box_i[3] / 2,
+                                     box_i[0] + box_i[2] / 2,
+                                     box_i[1] + box_i[3] / 2])
+            ok = True
+            for j in keep:
+                if not force_suppress and id_index >= 0 and \
+                        page[i, id_index] != page[j, id_index]:
+                    continue
+                box_j = page[j, coord_start:coord_start + 4]
+                if in_format == "center":
+                    box_j = torch.stack([box_j[0] - box_j[2] / 2,
+                                         box_j[1] - box_j[3] / 2,
+                                         box_j[0] + box_j[2] / 2,
+                                         box_j[1] + box_j[3] / 2])
+                ix1 = torch.maximum(box_i[0], box_j[0])
+                iy1 = torch.maximum(box_i[1], box_j[1])
+                ix2 = torch.minimum(box_i[2], box_j[2])
+                iy2 = torch.minimum(box_i[3], box_j[3])
+                iw = (ix2 - ix1).clamp(min=0)
+                ih = (iy2 - iy1).clamp(min=0)
+                inter = iw * ih
+                a_i = (box_i[2] - box_i[0]) * (box_i[3] - box_i[1])
+                a_j = (box_j[2] - box_j[0]) * (box_j[3] - box_j[1])
+                iou = inter / (a_i + a_j - inter + 1e-12)
+                if float(iou) > overlap_thresh:
+                    ok = False
+                    break
+            if ok:
+                keep.append(i)
+        mask = torch.ones(page.shape[0], dtype=torch.bool,
+                          device=page.device)
+        mask[torch.tensor(keep, dtype=torch.long)
+             if keep else torch.zeros(0, dtype=torch.long)] = False
+        page[mask, score_index] = -1.0
+    return flat.reshape(shape)

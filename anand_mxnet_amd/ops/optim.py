@@ -196,3 +196,44 @@ def multi_sum_sq(arrays):
     if arrays and _maybe_hip("multi_sum_sq", arrays[0]):
         return gpu_impl("multi_sum_sq")(arrays)
     return [a.float().pow(2).sum() for a in arrays]
+
+
+@register("adamw_update")
+def adamw_update(weight, grad, mean, var, rescale_grad, lr, beta1=0.9,
+                 beta2=0.999, epsilon=1e-8, wd=0.0, eta=1.0,
+                 clip_gradient=-1.0, out=None):
+    """AdamW with decoupled weight decay (reference:
+    src/operator/contrib/adamw.cc — note rescale_grad is a TENSOR there
+    so a dynamic loss scale can divide on device; scalar also accepted).
+    update = eta * (lr * m/(sqrt(v)+eps) + wd * w)."""
+    with torch.no_grad():
+        rs = rescale_grad if not torch.is_tensor(rescale_grad) \
+            else rescale_grad.item()
+        g = grad.to(mean.dtype) * rs
+        if clip_gradient > 0:
+            g = g.clamp(-clip_gradient, clip_gradient)
+        mean.mul_(beta1).add_(g, alpha=1 - beta1)
+        var.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        w32 = weight.to(mean.dtype)
+        upd = lr * mean / (var.sqrt() + epsilon) + wd * w32
+        weight.copy_((w32 - eta * upd).to(weight.dtype))
+    return weight
+
+
+@register("mp_adamw_update")
+def mp_adamw_update(weight, grad, mean, var, weight32, rescale_grad, lr,
+                    beta1=0.9, beta2=0.999, epsilon=1e-8, wd=0.0, eta=1.0,
+                    clip_gradient=-1.0, out=None):
+    """Multi-precision AdamW: fp32 master weights (reference contrib)."""
+    with torch.no_grad():
+        rs = rescale_grad if not torch.is_tensor(rescale_grad) \
+            else rescale_grad.item()
+        g = grad.float() * rs
+        if clip_gradient > 0:
+            g = g.clamp(-clip_gradient, clip_gradient)
+        mean.mul_(beta1).add_(g, alpha=1 - beta1)
+        var.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        upd = lr * mean / (var.sqrt() + epsilon) + wd * weight32
+        weight32.sub_(eta * upd)
+        weight.copy_(weight32.to(weight.dtype))
+    return weight

@@ -193,3 +193,27 @@ def test_ctc_loss():
     c = nd.ctc_loss(data, lab)
     assert c.shape == (2,)
     assert np.all(_np(c) > 0)
+
+
+def test_box_nms_index_copy_adamw():
+    d = nd.array(np.array([[[0, 0.9, 0, 0, 10, 10],
+                            [0, 0.8, 1, 1, 11, 11],
+                            [0, 0.7, 50, 50, 60, 60]]], dtype="float32"))
+    o = nd.box_nms(d, overlap_thresh=0.5, coord_start=2, score_index=1,
+                   id_index=0)
+    np.testing.assert_allclose(_np(o)[0, :, 1], [0.9, -1.0, 0.7])
+    x = nd.zeros((4, 2))
+    ic = nd.index_copy(x, nd.array([1, 3], dtype="int64"),
+                       nd.array(np.ones((2, 2), dtype="float32")))
+    np.testing.assert_allclose(_np(ic)[[1, 3]], np.ones((2, 2)))
+    np.testing.assert_allclose(_np(ic)[[0, 2]], np.zeros((2, 2)))
+    w = nd.array(np.ones(4, dtype="float32"))
+    g = nd.array(np.full(4, 0.1, dtype="float32"))
+    m, v = nd.zeros((4,)), nd.zeros((4,))
+    nd.adamw_update(w, g, m, v, 1.0, lr=0.1, wd=0.01)
+    assert np.all(_np(w) < 1.0)
+    w32 = nd.array(np.ones(4, dtype="float32"))
+    wb = w32.astype("float16")
+    m2, v2 = nd.zeros((4,)), nd.zeros((4,))
+    nd.mp_adamw_update(wb, g.astype("float16"), m2, v2, w32, 1.0, lr=0.1)
+    assert np.all(_np(w32) < 1.0)
