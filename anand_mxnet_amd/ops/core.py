@@ -730,3 +730,55 @@ def box_nms(data, overlap_thresh=0.5, valid_thresh=0.0, topk=-1,
              if keep else torch.zeros(0, dtype=torch.long)] = False
         page[mask, score_index] = -1.0
     return flat.reshape(shape)
+
+
+@register("_contrib_fft")
+def contrib_fft(data, compute_size=128):
+    """FFT on the last axis (reference: src/operator/contrib/fft.cc —
+    cuFFT).  Real input [..., d] -> interleaved re/im [..., 2d]."""
+    spec = torch.fft.fft(data.float(), dim=-1)
+    return torch.stack([spec.real, spec.imag], dim=-1) \
+        .reshape(*data.shape[:-1], 2 * data.shape[-1]).to(data.dtype)
+
+
+@register("_contrib_ifft")
+def contrib_ifft(data, compute_size=128):
+    """Inverse FFT (reference contrib/ifft.cc): interleaved re/im
+    [..., 2d] -> real [..., d].  Reference returns the UNSCALED inverse
+    (caller divides by d)."""
+    d = data.shape[-1] // 2
+    z = data.float().reshape(*data.shape[:-1], d, 2)
+    spec = torch.complex(z[..., 0], z[..., 1])
+    return (torch.fft.ifft(spec, dim=-1).real * d).to(data.dtype)
+
+
+@register("_contrib_ROIAlign")
+def roi_align(data, rois, pooled_size=(1, 1), spatial_scale=1.0,
+              sample_ratio=-1, position_sensitive=False, aligned=False):
+    """ROI align (reference: src/operator/contrib/roi_align.cu) via
+    grid_sample bilinear interpolation.  rois [R, 5]."""
+    ph, pw = int(pooled_size[0]), int(pooled_size[1])
+    n_roi = rois.shape[0]
+    C, H, W = data.shape[1], data.shape[2], data.shape[3]
+    out = data.new_zeros((n_roi, C, ph, pw))
+    off = 0.5 if aligned else 0.0
+    r = rois.detach().float()
+    for i in range(n_roi):
+        b = int(r[i, 0])
+        x1 = float(r[i, 1]) * spatial_scale - off
+        y1 = float(r[i, 2]) * spatial_scale - off
+        x2 = float(r[i, 3]) * spatial_scale - off
+        y2 = float(r[i, 4]) * spatial_scale - off
+        sr = sample_ratio if sample_ratio > 0 else 2
+        gh, gw = ph * sr, pw * sr
+        ys = torch.linspace(y1, y2, 2 * gh + 1, device=data.device)[1::2]
+        xs = torch.linspace(x1, x2, 2 * gw + 1, device=data.device)[1::2]
+        gy = (ys / max(H - 1, 1)) * 2 - 1
+        gx = (xs / max(W - 1, 1)) * 2 - 1
+        grid = torch.stack(torch.meshgrid(gy, gx, indexing="ij"),
+                           dim=-1)[..., [1, 0]].unsqueeze(0)
+        samp = torch.nn.functional.grid_sample(
+            data[b:b + 1].float(), grid, mode="bilinear",
+            padding_mode="zeros", align_corners=True)
+        out[i] = torch.nn.functional.avg_pool2d(samp, sr)[0].to(data.dtype)
+    return out

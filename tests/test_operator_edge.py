@@ -217,3 +217,18 @@ def test_box_nms_index_copy_adamw():
     m2, v2 = nd.zeros((4,)), nd.zeros((4,))
     nd.mp_adamw_update(wb, g.astype("float16"), m2, v2, w32, 1.0, lr=0.1)
     assert np.all(_np(w32) < 1.0)
+
+
+def test_fft_ifft_roialign():
+    x = nd.array(np.array([[1.0, 2.0, 3.0, 4.0]], dtype="float32"))
+    f = nd._contrib_fft(x)
+    np.testing.assert_allclose(_np(f)[0],
+                               [10, 0, -2, 2, -2, 0, -2, -2], atol=1e-5)
+    inv = nd._contrib_ifft(f)
+    np.testing.assert_allclose(_np(inv)[0] / 4, [1, 2, 3, 4], atol=1e-5)
+    xx = nd.array(np.arange(16, dtype="float32").reshape(1, 1, 4, 4))
+    rois = nd.array(np.array([[0, 0, 0, 3, 3]], dtype="float32"))
+    ra = nd._contrib_ROIAlign(xx, rois, pooled_size=(2, 2),
+                              spatial_scale=1.0)
+    np.testing.assert_allclose(_np(ra).ravel(),
+                               [3.75, 5.25, 9.75, 11.25], atol=1e-4)
