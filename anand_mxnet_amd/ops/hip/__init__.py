@@ -385,9 +385,23 @@ def conv2d_nhwc(x, w, bias, stride, pad, dil, groups, collect_stats=False):
         if groups == C_in and w.shape[0] == C_in and w.shape[3] == 1                 and C_in % 8 == 0 and tuple(dil) == (1, 1):
             return _DwConvNHWC.apply(x, w, bias, tuple(stride),
                                      tuple(pad), tuple(dil))
-        raise NotImplementedError(
-            "grouped conv HIP kernel: only depthwise (groups == C) "
-            "is implemented (docs/PARITY.md)")
+        # general grouped conv (ResNeXt-style): per-group channel slices
+        # through the dense implicit-GEMM kernels.  Each group is a full
+        # conv of (K/G, R, S, C/G); channel-contiguous NHWC slicing makes
+        # the per-group inputs cheap views + one concat at the end.
+        K = w.shape[0]
+        if K % groups or C_in % groups:
+            raise NotImplementedError(
+                "grouped conv: channels must divide groups")
+        kg = K // groups
+        outs = []
+        for g in range(groups):
+            xg = x[..., g * (C_in // groups):(g + 1) * (C_in // groups)] \
+                .contiguous()
+            wg = w[g * kg:(g + 1) * kg]
+            bg = bias[g * kg:(g + 1) * kg] if bias is not None else None
+            outs.append(conv2d_nhwc(xg, wg, bg, stride, pad, dil, 1))
+        return torch.cat(outs, dim=3)
     stride, pad, dil = tuple(stride), tuple(pad), tuple(dil)
     if w.shape[3] % 64 == 0:
         holder = [] if (collect_stats and _CONV_BN_STATS
