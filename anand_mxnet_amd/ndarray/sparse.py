@@ -12,8 +12,14 @@ is needed (CSRNDArray remains out of scope, documented in docs/PARITY.md).
 """
 from __future__ import annotations
 
+import warnings
+
 import numpy as np
 import torch
+
+# torch's CSR support warns 'beta' on every construction
+warnings.filterwarnings("ignore",
+                        message="Sparse CSR tensor support is in beta")
 
 from ..base import MXNetError
 from .ndarray import NDArray
@@ -113,13 +119,14 @@ def row_sparse_array(arg1, shape=None, ctx=None, dtype=None):
 
 
 def _dense_tostype(nd_arr: NDArray, stype):
-    """tostype('row_sparse') on a dense NDArray."""
+    """tostype('row_sparse'/'csr') on a dense NDArray."""
     if stype == "default":
         return nd_arr
     if stype == "row_sparse":
         return row_sparse_array(nd_arr)
-    raise MXNetError(f"unsupported stype {stype} (CSR is out of scope, "
-                     "see docs/PARITY.md)")
+    if stype == "csr":
+        return csr_matrix(nd_arr)
+    raise MXNetError(f"unsupported stype {stype}")
 
 
 def sparse_row_update(weight: NDArray, grad: RowSparseNDArray, lr: float,
@@ -133,3 +140,94 @@ def sparse_row_update(weight: NDArray, grad: RowSparseNDArray, lr: float,
     rows = w[idx]
     rows.add_(vals + wd * rows, alpha=-lr)
     w[idx] = rows
+
+
+class CSRNDArray:
+    """Compressed sparse row matrix (reference parity:
+    python/mxnet/ndarray/sparse.py — CSRNDArray; src/operator/tensor/
+    cast_storage / dot CSR kernels).  Backed by torch.sparse_csr."""
+
+    stype = "csr"
+
+    def __init__(self, data: torch.Tensor, indices: torch.Tensor,
+                 indptr: torch.Tensor, shape):
+        self._shape = tuple(shape)
+        self._csr = torch.sparse_csr_tensor(
+            indptr.long(), indices.long(), data, size=self._shape)
+
+    # ---------------- reference API surface ----------------
+    @property
+    def shape(self):
+        return self._shape
+
+    @property
+    def dtype(self):
+        from ..base import torch_dtype_to_np
+        return torch_dtype_to_np(self._csr.values().dtype)
+
+    @property
+    def context(self):
+        from ..context import Context
+        return Context.from_torch(self._csr.device)
+
+    @property
+    def data(self) -> NDArray:
+        return NDArray(self._csr.values())
+
+    @property
+    def indices(self) -> NDArray:
+        return NDArray(self._csr.col_indices())
+
+    @property
+    def indptr(self) -> NDArray:
+        return NDArray(self._csr.crow_indices())
+
+    def tostype(self, stype):
+        if stype == "csr":
+            return self
+        if stype == "default":
+            return NDArray(self._csr.to_dense())
+        if stype == "row_sparse":
+            return row_sparse_array(NDArray(self._csr.to_dense()))
+        raise MXNetError(f"unsupported stype {stype}")
+
+    def asnumpy(self):
+        return self._csr.to_dense().numpy()
+
+    def dot(self, rhs) -> NDArray:
+        """csr @ dense (reference: dot(csr, default) forward)."""
+        r = rhs._t if isinstance(rhs, NDArray) else torch.as_tensor(rhs)
+        return NDArray(self._csr.matmul(r.to(self._csr.values().dtype)))
+
+    def __getitem__(self, key):
+        if isinstance(key, slice):
+            dense = self._csr.to_dense()[key]
+            return csr_matrix(NDArray(dense))
+        raise MXNetError("CSRNDArray: only row-slice indexing")
+
+    def wait_to_read(self):
+        pass
+
+    def __repr__(self):
+        return (f"<CSRNDArray {self._shape} "
+                f"nnz={self._csr.values().numel()}>")
+
+
+def csr_matrix(arg1, shape=None, ctx=None, dtype=None):
+    """Create a CSRNDArray from (data, indices, indptr) or dense input
+    (reference API mx.nd.sparse.csr_matrix)."""
+    from .. import nd as _nd
+    if isinstance(arg1, tuple) and len(arg1) == 3:
+        data, indices, indptr = arg1
+        to_t = lambda v, dt=None: (
+            v._t if isinstance(v, NDArray)
+            else torch.as_tensor(np.asarray(v), dtype=dt))
+        assert shape is not None, "csr_matrix needs an explicit shape"
+        return CSRNDArray(to_t(data, torch.float32),
+                          to_t(indices, torch.int64),
+                          to_t(indptr, torch.int64), shape)
+    dense = arg1 if isinstance(arg1, NDArray) else _nd.array(
+        arg1, ctx=ctx, dtype=dtype)
+    sp = dense._t.to_sparse_csr()
+    return CSRNDArray(sp.values(), sp.col_indices(), sp.crow_indices(),
+                      dense.shape)

@@ -194,3 +194,25 @@ def test_sparse_row_update():
     out = w.asnumpy()
     np.testing.assert_allclose(out[1], [0, 0])   # 1 - 0.5*2
     np.testing.assert_allclose(out[0], [1, 1])   # untouched
+
+
+def test_csr_matrix():
+    from anand_mxnet_amd.ndarray.sparse import csr_matrix
+    m = csr_matrix(([1.0, 2.0, 3.0], [0, 2, 1], [0, 2, 3]), shape=(2, 3))
+    np.testing.assert_allclose(m.asnumpy(), [[1, 0, 2], [0, 3, 0]])
+    assert m.stype == "csr"
+    d = nd.array(np.arange(12, dtype="float32").reshape(3, 4))
+    out = m.dot(d)
+    np.testing.assert_allclose(out.asnumpy(),
+                               m.asnumpy() @ d.asnumpy())
+    dense = nd.array(np.array([[0, 1.0], [2.0, 0]], dtype="float32"))
+    c = dense.tostype("csr")
+    np.testing.assert_allclose(c.tostype("default").asnumpy(),
+                               dense.asnumpy())
+    np.testing.assert_allclose(c[0:1].asnumpy(), [[0, 1.0]])
+    rs = c.tostype("row_sparse")
+    assert rs.stype == "row_sparse"
+    np.testing.assert_allclose(rs.tostype("default").asnumpy(),
+                               dense.asnumpy())
+    assert list(m.indptr.asnumpy()) == [0, 2, 3]
+    assert list(m.indices.asnumpy()) == [0, 2, 1]
