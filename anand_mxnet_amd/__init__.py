@@ -43,6 +43,11 @@ from . import monitor
 from . import operator
 from . import image
 from . import visualization
+from . import libinfo
+from . import runtime
+from . import attribute
+from . import name
+from . import kvstore_server
 from . import test_utils
 from . import util
 
