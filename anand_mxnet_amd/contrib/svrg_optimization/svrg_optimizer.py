@@ -5,7 +5,7 @@ grad_corrected = grad - special_grad + full_batch_grad_snapshot; the
 corrected gradient feeds a base optimizer (default SGD)."""
 from __future__ import annotations
 
-from ...optimizer import Optimizer, SGD, register
+from ...optimizer import Optimizer, register
 
 
 @register

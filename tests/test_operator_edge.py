@@ -2,7 +2,6 @@
 tests/python/unittest/test_operator.py — the long tail beyond the
 hot-path ops covered in test_ndarray/test_gluon)."""
 import numpy as np
-import pytest
 
 import anand_mxnet_amd as mx
 from anand_mxnet_amd import nd
