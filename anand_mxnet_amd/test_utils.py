@@ -132,3 +132,22 @@ def check_consistency(fn, inputs_np, ctx_list=None, rtol=None, atol=None,
                                         names=(f"grad {ctx}/{dt}",
                                                "baseline"))
     return base_out
+
+
+def retry(n):
+    """Retry decorator for stochastic tests (reference:
+    tests/python/unittest/common.py — retry)."""
+    def deco(fn):
+        import functools
+
+        @functools.wraps(fn)
+        def wrapped(*args, **kwargs):
+            last = None
+            for _ in range(n):
+                try:
+                    return fn(*args, **kwargs)
+                except AssertionError as e:
+                    last = e
+            raise last
+        return wrapped
+    return deco
