@@ -6,7 +6,8 @@ from .block import HybridBlock
 __all__ = ["Loss", "L2Loss", "L1Loss", "SigmoidBinaryCrossEntropyLoss",
            "SigmoidBCELoss", "SoftmaxCrossEntropyLoss", "SoftmaxCELoss",
            "KLDivLoss", "HuberLoss", "HingeLoss", "SquaredHingeLoss",
-           "LogisticLoss", "TripletLoss"]
+           "LogisticLoss", "TripletLoss", "CTCLoss", "CosineEmbeddingLoss",
+           "PoissonNLLLoss"]
 
 
 def _apply_weighting(F, loss, weight=None, sample_weight=None):
@@ -181,3 +182,75 @@ class TripletLoss(Loss):
                      axis=self._batch_axis, exclude=True)
         loss = F.relu(loss + self._margin)
         return _apply_weighting(F, loss, self._weight, sample_weight)
+
+
+class CTCLoss(Loss):
+    """Connectionist temporal classification loss (reference
+    gluon/loss.py — CTCLoss over the ctc_loss op).  layout 'NTC' or
+    'TNC'; label [batch, L] padded."""
+
+    def __init__(self, layout="NTC", label_layout="NT", weight=None,
+                 **kwargs):
+        assert layout in ("NTC", "TNC")
+        assert label_layout in ("NT", "TN")
+        self._layout = layout
+        self._label_layout = label_layout
+        super().__init__(weight, 0, **kwargs)
+
+    def hybrid_forward(self, F, pred, label, pred_lengths=None,
+                       label_lengths=None, sample_weight=None):
+        if self._layout == "NTC":
+            pred = F.transpose(pred, axes=(1, 0, 2))
+        if self._label_layout == "TN":
+            label = F.transpose(label, axes=(1, 0))
+        loss = F.ctc_loss(pred, label, pred_lengths, label_lengths,
+                          use_data_lengths=pred_lengths is not None,
+                          use_label_lengths=label_lengths is not None,
+                          blank_label="first")
+        return _apply_weighting(F, loss, self._weight, sample_weight)
+
+
+class CosineEmbeddingLoss(Loss):
+    """1 - cos(x1, x2) when label=1; max(0, cos - margin) when label=-1
+    (reference gluon/loss.py — CosineEmbeddingLoss)."""
+
+    def __init__(self, weight=None, batch_axis=0, margin=0, **kwargs):
+        super().__init__(weight, batch_axis, **kwargs)
+        self._margin = margin
+
+    def hybrid_forward(self, F, input1, input2, label,
+                       sample_weight=None):
+        num = F.sum(input1 * input2, axis=-1)
+        den = F.sqrt(F.sum(input1 ** 2, axis=-1)) * \
+            F.sqrt(F.sum(input2 ** 2, axis=-1)) + 1e-12
+        cos = num / den
+        label = label.reshape(cos.shape)
+        pos = 1 - cos
+        neg = F.relu(cos - self._margin)
+        loss = F.where(label == 1, pos, neg)
+        return _apply_weighting(F, loss, self._weight, sample_weight)
+
+
+class PoissonNLLLoss(Loss):
+    """Poisson negative log likelihood (reference gluon/loss.py):
+    L = pred - target*log(pred + eps) [+ Stirling]."""
+
+    def __init__(self, weight=None, from_logits=True, batch_axis=0,
+                 compute_full=False, **kwargs):
+        super().__init__(weight, batch_axis, **kwargs)
+        self._from_logits = from_logits
+        self._compute_full = compute_full
+
+    def hybrid_forward(self, F, pred, target, sample_weight=None,
+                       epsilon=1e-08):
+        if self._from_logits:
+            loss = F.exp(pred) - target * pred
+        else:
+            loss = pred - target * F.log(pred + epsilon)
+        if self._compute_full:
+            stirling = target * F.log(target + epsilon) - target + \
+                0.5 * F.log(2 * 3.1415926535 * (target + epsilon))
+            stirling = F.where(target <= 1, F.zeros_like(target), stirling)
+            loss = loss + stirling
+        loss = _apply_weighting(F, loss, self._weight, sample_weight)
+        return F.mean(loss)

@@ -143,3 +143,22 @@ def test_softmax_output_op():
     oh = np.eye(10)[label.asnumpy().astype(int)]
     assert_almost_equal(data.grad.asnumpy(), (prob - oh) / 4,
                         rtol=1e-4, atol=1e-5)
+
+
+def test_ctc_cosine_poisson_losses():
+    from anand_mxnet_amd import gluon
+    rs = np.random.RandomState(7)
+    ce = gluon.loss.CosineEmbeddingLoss(margin=0.5)
+    a = nd.array(np.array([[1.0, 0.0], [0.0, 1.0]], dtype="float32"))
+    b = nd.array(np.array([[1.0, 0.0], [1.0, 0.0]], dtype="float32"))
+    lab = nd.array(np.array([1.0, -1.0], dtype="float32"))
+    out = ce(a, b, lab).asnumpy()
+    np.testing.assert_allclose(out, [0.0, 0.0], atol=1e-5)  # cos=1 / cos=0
+    pl = gluon.loss.PoissonNLLLoss(from_logits=True)
+    v = float(pl(nd.array([[0.0]]), nd.array([[2.0]])).asnumpy())
+    np.testing.assert_allclose(v, 1.0, rtol=1e-5)  # e^0 - 2*0
+    ctc = gluon.loss.CTCLoss()
+    pred = nd.array(rs.randn(2, 10, 5).astype("float32"))
+    labels = nd.array(np.array([[1, 2, 0], [3, 0, 0]], dtype="float32"))
+    out = ctc(pred, labels).asnumpy()
+    assert out.shape == (2,) and np.all(out > 0)
