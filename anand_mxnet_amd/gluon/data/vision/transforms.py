@@ -11,7 +11,9 @@ from ....ndarray.ndarray import NDArray
 
 __all__ = ["Compose", "Cast", "ToTensor", "Normalize", "Resize",
            "CenterCrop", "RandomResizedCrop", "RandomFlipLeftRight",
-           "RandomFlipTopBottom"]
+           "RandomFlipTopBottom",
+           "RandomBrightness", "RandomContrast", "RandomSaturation",
+           "RandomHue", "RandomColorJitter", "RandomLighting"]
 
 
 class Compose(Compose_):
@@ -131,3 +133,108 @@ class RandomFlipTopBottom(Block):
             axis = 0 if x._t.dim() == 3 and x._t.shape[0] > 4 else -2
             return NDArray(torch.flip(x._t, (axis,)))
         return x
+
+
+def _jitter(img_t, factor):
+    """Uniform multiplicative jitter in [1-f, 1+f] (reference
+    image.py brightness/saturation/contrast aug convention)."""
+    alpha = 1.0 + float(np.random.uniform(-factor, factor))
+    return img_t.float() * alpha
+
+
+class RandomBrightness(Block):
+    """(reference: gluon/data/vision/transforms.py — RandomBrightness)"""
+
+    def __init__(self, brightness):
+        super().__init__()
+        self._b = brightness
+
+    def forward(self, x):
+        return NDArray(_jitter(x._t, self._b).to(x._t.dtype))
+
+
+class RandomContrast(Block):
+    def __init__(self, contrast):
+        super().__init__()
+        self._c = contrast
+
+    def forward(self, x):
+        t = x._t.float()
+        alpha = 1.0 + float(np.random.uniform(-self._c, self._c))
+        mean = t.mean()
+        return NDArray((t * alpha + mean * (1 - alpha)).to(x._t.dtype))
+
+
+class RandomSaturation(Block):
+    """HWC images; grayscale mix (ITU-R BT.601 weights, reference
+    image.py SaturationJitterAug)."""
+
+    def __init__(self, saturation):
+        super().__init__()
+        self._s = saturation
+
+    def forward(self, x):
+        t = x._t.float()
+        alpha = 1.0 + float(np.random.uniform(-self._s, self._s))
+        coef = torch.tensor([0.299, 0.587, 0.114], dtype=t.dtype)
+        gray = (t * coef).sum(-1, keepdim=True)
+        return NDArray((t * alpha + gray * (1 - alpha)).to(x._t.dtype))
+
+
+class RandomHue(Block):
+    """Approximate hue rotation in YIQ space (reference image.py
+    HueJitterAug linear approximation)."""
+
+    def __init__(self, hue):
+        super().__init__()
+        self._h = hue
+
+    def forward(self, x):
+        t = x._t.float()
+        alpha = float(np.random.uniform(-self._h, self._h))
+        u, w = np.cos(alpha * np.pi), np.sin(alpha * np.pi)
+        bt = np.array([[1.0, 0.0, 0.0], [0.0, u, -w], [0.0, w, u]])
+        tyiq = np.array([[0.299, 0.587, 0.114], [0.596, -0.274, -0.321],
+                         [0.211, -0.523, 0.311]])
+        m = np.linalg.inv(tyiq) @ bt @ tyiq
+        mt = torch.tensor(m.T, dtype=t.dtype)
+        return NDArray((t @ mt).to(x._t.dtype))
+
+
+class RandomColorJitter(Block):
+    def __init__(self, brightness=0, contrast=0, saturation=0, hue=0):
+        super().__init__()
+        self._ts = []
+        if brightness:
+            self._ts.append(RandomBrightness(brightness))
+        if contrast:
+            self._ts.append(RandomContrast(contrast))
+        if saturation:
+            self._ts.append(RandomSaturation(saturation))
+        if hue:
+            self._ts.append(RandomHue(hue))
+
+    def forward(self, x):
+        for t in np.random.permutation(self._ts):
+            x = t.forward(x)
+        return x
+
+
+class RandomLighting(Block):
+    """AlexNet-style PCA lighting noise (reference transforms)."""
+
+    _eigval = np.array([55.46, 4.794, 1.148])
+    _eigvec = np.array([[-0.5675, 0.7192, 0.4009],
+                        [-0.5808, -0.0045, -0.8140],
+                        [-0.5836, -0.6948, 0.4203]])
+
+    def __init__(self, alpha):
+        super().__init__()
+        self._a = alpha
+
+    def forward(self, x):
+        a = np.random.normal(0, self._a, size=(3,))
+        rgb = (self._eigvec * a * self._eigval).sum(axis=1)
+        return NDArray((x._t.float() +
+                        torch.tensor(rgb, dtype=torch.float32))
+                       .to(x._t.dtype))

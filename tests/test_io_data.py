@@ -180,3 +180,25 @@ def test_libsvm_and_imagerecord_iters(tmp_path):
                                 batch_size=2)
     batch = next(iter(rit))
     assert batch.data[0].shape == (2, 3, 8, 8)
+
+
+def test_color_transforms():
+    import numpy as np
+    from anand_mxnet_amd import nd
+    from anand_mxnet_amd.gluon.data.vision import transforms as T
+    np.random.seed(0)
+    img = nd.array(np.random.rand(8, 8, 3).astype("float32"))
+    for t in [T.RandomBrightness(0.5), T.RandomContrast(0.5),
+              T.RandomSaturation(0.5), T.RandomHue(0.2),
+              T.RandomColorJitter(0.3, 0.3, 0.3, 0.1),
+              T.RandomLighting(0.1)]:
+        out = t(img)
+        assert out.shape == img.shape
+    # hue with alpha=0 is identity
+    h = T.RandomHue(0.0)
+    np.testing.assert_allclose(h(img).asnumpy(), img.asnumpy(), atol=1e-5)
+    # saturation of a gray image is identity
+    gray = nd.array(np.full((4, 4, 3), 0.5, dtype="float32"))
+    s = T.RandomSaturation(0.9)
+    np.testing.assert_allclose(s(gray).asnumpy(), gray.asnumpy(),
+                               atol=1e-6)
