@@ -13,7 +13,10 @@ from ..block import HybridBlock
 from .basic_layers import Activation
 
 __all__ = ["Conv2D", "Conv1D", "Conv2DTranspose", "MaxPool2D", "AvgPool2D",
-           "GlobalAvgPool2D", "GlobalMaxPool2D", "MaxPool1D", "AvgPool1D"]
+           "GlobalAvgPool2D", "GlobalMaxPool2D", "MaxPool1D", "AvgPool1D",
+           "Conv3D", "Conv1DTranspose", "MaxPool3D", "AvgPool3D",
+           "GlobalMaxPool1D", "GlobalAvgPool1D", "GlobalMaxPool3D",
+           "GlobalAvgPool3D", "ReflectionPad2D"]
 
 
 def _pair(x):
@@ -247,3 +250,155 @@ class GlobalMaxPool2D(HybridBlock):
     def hybrid_forward(self, F, x):
         return F.Pooling(x, global_pool=True, pool_type="max",
                          layout=self._layout, kernel=(1, 1))
+
+
+class Conv3D(HybridBlock):
+    """3-D convolution, NCDHW (reference conv_layers.py — Conv3D)."""
+
+    def __init__(self, channels, kernel_size, strides=(1, 1, 1),
+                 padding=(0, 0, 0), dilation=(1, 1, 1), groups=1,
+                 layout="NCDHW", activation=None, use_bias=True,
+                 weight_initializer=None, bias_initializer="zeros",
+                 in_channels=0, **kwargs):
+        super().__init__(**kwargs)
+        assert layout == "NCDHW", "Conv3D: NCDHW only"
+        with self.name_scope():
+            self._channels = channels
+            self._kernel = _triple(kernel_size)
+            self._strides = _triple(strides)
+            self._padding = _triple(padding)
+            self._dilation = _triple(dilation)
+            self._groups = groups
+            self.weight = self.params.get(
+                "weight", shape=(channels,
+                                 in_channels // groups if in_channels else 0)
+                + self._kernel,
+                init=weight_initializer, allow_deferred_init=True)
+            self.bias = self.params.get("bias", shape=(channels,),
+                                        init=bias_initializer,
+                                        allow_deferred_init=True) \
+                if use_bias else None
+            self.act = Activation(activation) if activation else None
+
+    def shape_infer(self, x, *args):
+        self.weight.shape = (self._channels,
+                             x.shape[1] // self._groups) + self._kernel
+        if self.bias is not None:
+            self.bias.shape = (self._channels,)
+
+    def hybrid_forward(self, F, x, weight, bias=None):
+        out = F.Convolution(x, weight, bias, kernel=self._kernel,
+                            stride=self._strides, dilate=self._dilation,
+                            pad=self._padding, num_filter=self._channels,
+                            num_group=self._groups, no_bias=bias is None,
+                            layout="NCDHW")
+        if self.act is not None:
+            out = self.act(out)
+        return out
+
+
+class Conv1DTranspose(HybridBlock):
+    """1-D transposed conv (reference conv_layers.py)."""
+
+    def __init__(self, channels, kernel_size, strides=1, padding=0,
+                 output_padding=0, dilation=1, groups=1, use_bias=True,
+                 weight_initializer=None, bias_initializer="zeros",
+                 in_channels=0, **kwargs):
+        super().__init__(**kwargs)
+        with self.name_scope():
+            self._channels = channels
+            k = kernel_size if isinstance(kernel_size, int) \
+                else kernel_size[0]
+            self._kernel = (k,)
+            self._strides = (strides if isinstance(strides, int)
+                             else strides[0],)
+            self._padding = (padding if isinstance(padding, int)
+                             else padding[0],)
+            self._adj = (output_padding if isinstance(output_padding, int)
+                         else output_padding[0],)
+            self._dilation = (dilation if isinstance(dilation, int)
+                              else dilation[0],)
+            self._groups = groups
+            self.weight = self.params.get(
+                "weight",
+                shape=(in_channels, channels // groups) + self._kernel
+                if in_channels else (0, channels // groups) + self._kernel,
+                init=weight_initializer, allow_deferred_init=True)
+            self.bias = self.params.get("bias", shape=(channels,),
+                                        init=bias_initializer,
+                                        allow_deferred_init=True) \
+                if use_bias else None
+
+    def shape_infer(self, x, *args):
+        self.weight.shape = (x.shape[1],
+                             self._channels // self._groups) + self._kernel
+        if self.bias is not None:
+            self.bias.shape = (self._channels,)
+
+    def hybrid_forward(self, F, x, weight, bias=None):
+        return F.Deconvolution(x, weight, bias, kernel=self._kernel,
+                               stride=self._strides, dilate=self._dilation,
+                               pad=self._padding, adj=self._adj,
+                               num_filter=self._channels,
+                               num_group=self._groups,
+                               no_bias=bias is None)
+
+
+class MaxPool3D(HybridBlock):
+    def __init__(self, pool_size=(2, 2, 2), strides=None, padding=0,
+                 ceil_mode=False, **kwargs):
+        super().__init__(**kwargs)
+        self._k = _triple(pool_size)
+        self._s = _triple(strides) if strides is not None else self._k
+        self._p = _triple(padding)
+        self._conv = "full" if ceil_mode else "valid"
+
+    def hybrid_forward(self, F, x):
+        return F.Pooling(x, kernel=self._k, stride=self._s, pad=self._p,
+                         pool_type="max", pooling_convention=self._conv)
+
+
+class AvgPool3D(MaxPool3D):
+    def hybrid_forward(self, F, x):
+        return F.Pooling(x, kernel=self._k, stride=self._s, pad=self._p,
+                         pool_type="avg", pooling_convention=self._conv)
+
+
+class GlobalMaxPool1D(HybridBlock):
+    def hybrid_forward(self, F, x):
+        return F.Pooling(x, kernel=(1,), global_pool=True, pool_type="max")
+
+
+class GlobalAvgPool1D(HybridBlock):
+    def hybrid_forward(self, F, x):
+        return F.Pooling(x, kernel=(1,), global_pool=True, pool_type="avg")
+
+
+class GlobalMaxPool3D(HybridBlock):
+    def hybrid_forward(self, F, x):
+        return F.Pooling(x, kernel=(1, 1, 1), global_pool=True,
+                         pool_type="max")
+
+
+class GlobalAvgPool3D(HybridBlock):
+    def hybrid_forward(self, F, x):
+        return F.Pooling(x, kernel=(1, 1, 1), global_pool=True,
+                         pool_type="avg")
+
+
+class ReflectionPad2D(HybridBlock):
+    """Reflection padding on H/W of NCHW (reference conv_layers.py)."""
+
+    def __init__(self, padding=0, **kwargs):
+        super().__init__(**kwargs)
+        self._pad = padding if isinstance(padding, (list, tuple)) \
+            else (padding,) * 4
+
+    def hybrid_forward(self, F, x):
+        l, r, t, b = (int(p) for p in self._pad)
+        return F.pad(x, mode="reflect",
+                     pad_width=(0, 0, 0, 0, t, b, l, r))
+
+
+def _triple(v):
+    return (v, v, v) if isinstance(v, int) else tuple(v)

@@ -383,8 +383,15 @@ def pad(x, mode="constant", pad_width=None, constant_value=0):
     pw = []
     for i in range(len(pad_width) // 2 - 1, -1, -1):
         pw += [pad_width[2 * i], pad_width[2 * i + 1]]
-    return torch.nn.functional.pad(x, pw, mode=mode if mode != "edge" else "replicate",
-                                   value=constant_value)
+    tmode = mode if mode != "edge" else "replicate"
+    if tmode == "constant":
+        return torch.nn.functional.pad(x, pw, mode=tmode,
+                                       value=constant_value)
+    # non-constant torch pad only accepts trailing-dim pads: strip
+    # leading zero pairs (MXNet requires zeros on N/C anyway)
+    while len(pw) > 2 and pw[-1] == 0 and pw[-2] == 0:
+        pw = pw[:-2]
+    return torch.nn.functional.pad(x, pw, mode=tmode)
 
 
 @register("Cast")

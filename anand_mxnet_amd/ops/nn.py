@@ -115,8 +115,11 @@ def convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
         out = gpu_impl("conv2d_nhwc")(x, w, b, stride, pad, dilate, num_group)
         return out.permute(0, 3, 1, 2)
     _use_hip(data, "conv2d_nhwc")  # raises on GPU without native ext
-    return F.conv2d(data, weight, b, stride=tuple(stride), padding=tuple(pad),
-                    dilation=tuple(dilate), groups=num_group)
+    nd_ = data.dim() - 2
+    fn = {1: F.conv1d, 2: F.conv2d, 3: F.conv3d}[nd_]
+    return fn(data, weight, b, stride=tuple(stride)[:nd_],
+              padding=tuple(pad)[:nd_], dilation=tuple(dilate)[:nd_],
+              groups=num_group)
 
 
 @register("Deconvolution")
@@ -124,9 +127,12 @@ def deconvolution(data, weight, bias=None, kernel=None, stride=(1, 1),
                   dilate=(1, 1), pad=(0, 0), adj=(0, 0), num_filter=None,
                   num_group=1, no_bias=True, layout="NCHW"):
     b = None if no_bias else bias
-    return F.conv_transpose2d(data, weight, b, stride=tuple(stride),
-                              padding=tuple(pad), output_padding=tuple(adj),
-                              dilation=tuple(dilate), groups=num_group)
+    nd_ = data.dim() - 2
+    fn = {1: F.conv_transpose1d, 2: F.conv_transpose2d,
+          3: F.conv_transpose3d}[nd_]
+    return fn(data, weight, b, stride=tuple(stride)[:nd_],
+              padding=tuple(pad)[:nd_], output_padding=tuple(adj)[:nd_],
+              dilation=tuple(dilate)[:nd_], groups=num_group)
 
 
 # ---------------- Pooling ----------------
@@ -140,25 +146,35 @@ def pooling(data, kernel=(2, 2), pool_type="max", global_pool=False,
         return gpu_impl("pool_nhwc")(data, kernel, pool_type, global_pool,
                                      stride, pad, count_include_pad)
     x = data.permute(0, 3, 1, 2) if nhwc else data
-    if not nhwc and x.is_cuda:
+    if not nhwc and x.dim() == 4 and x.is_cuda:
         _use_hip(x.permute(0, 2, 3, 1), "pool_nhwc")
+    nd_ = x.dim() - 2
+    mp = {1: F.max_pool1d, 2: F.max_pool2d, 3: F.max_pool3d}[nd_]
+    ap = {1: F.avg_pool1d, 2: F.avg_pool2d, 3: F.avg_pool3d}[nd_]
+    amp = {1: F.adaptive_max_pool1d, 2: F.adaptive_max_pool2d,
+           3: F.adaptive_max_pool3d}[nd_]
+    aap = {1: F.adaptive_avg_pool1d, 2: F.adaptive_avg_pool2d,
+           3: F.adaptive_avg_pool3d}[nd_]
     if global_pool:
-        if pool_type == "max":
-            out = F.adaptive_max_pool2d(x, 1)
-        else:
-            out = F.adaptive_avg_pool2d(x, 1)
+        out = amp(x, 1) if pool_type == "max" else aap(x, 1)
     else:
         stride = stride or kernel
         ceil = pooling_convention == "full"
+        k = tuple(kernel)[:nd_]
+        st = tuple(stride)[:nd_]
+        pd = tuple(pad)[:nd_]
+        if nd_ == 1:
+            k, st, pd = k[0], st[0], pd[0]
         if pool_type == "max":
-            out = F.max_pool2d(x, tuple(kernel), tuple(stride), tuple(pad),
-                               ceil_mode=ceil)
+            out = mp(x, k, st, pd, ceil_mode=ceil)
         elif pool_type == "avg":
-            out = F.avg_pool2d(x, tuple(kernel), tuple(stride), tuple(pad),
-                               ceil_mode=ceil, count_include_pad=count_include_pad)
+            out = ap(x, k, st, pd, ceil_mode=ceil,
+                     count_include_pad=count_include_pad)
         elif pool_type == "sum":
-            out = F.avg_pool2d(x, tuple(kernel), tuple(stride), tuple(pad),
-                               ceil_mode=ceil, divisor_override=1)
+            if nd_ == 1:  # avg_pool1d has no divisor_override
+                out = ap(x, k, st, pd, ceil_mode=ceil) * k
+            else:
+                out = ap(x, k, st, pd, ceil_mode=ceil, divisor_override=1)
         else:
             raise ValueError(pool_type)
     return out.permute(0, 2, 3, 1).contiguous() if nhwc else out

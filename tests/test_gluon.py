@@ -212,3 +212,32 @@ def test_clip_global_norm():
     assert abs(norm - np.sqrt(9 * 2 + 16 * 2)) < 1e-4
     total = sum((a.asnumpy() ** 2).sum() for a in arrays)
     assert abs(np.sqrt(total) - 1.0) < 1e-3
+
+
+def test_conv_pool_3d_1d_family():
+    from anand_mxnet_amd.gluon import nn as gnn
+    rs = np.random.RandomState(2)
+    x3 = nd.array(rs.randn(2, 3, 4, 5, 6).astype("float32"))
+    c3 = gnn.Conv3D(4, 3, padding=1)
+    c3.initialize()
+    assert c3(x3).shape == (2, 4, 4, 5, 6)
+    assert gnn.MaxPool3D()(x3).shape == (2, 3, 2, 2, 3)
+    assert gnn.AvgPool3D()(x3).shape == (2, 3, 2, 2, 3)
+    assert gnn.GlobalAvgPool3D()(x3).shape == (2, 3, 1, 1, 1)
+    assert gnn.GlobalMaxPool3D()(x3).shape == (2, 3, 1, 1, 1)
+    x1 = nd.array(rs.randn(2, 3, 10).astype("float32"))
+    ct = gnn.Conv1DTranspose(5, 3, strides=2)
+    ct.initialize()
+    assert ct(x1).shape == (2, 5, 21)
+    assert gnn.GlobalMaxPool1D()(x1).shape == (2, 3, 1)
+    assert gnn.GlobalAvgPool1D()(x1).shape == (2, 3, 1)
+    x2 = nd.array(np.arange(16, dtype="float32").reshape(1, 1, 4, 4))
+    out = gnn.ReflectionPad2D(1)(x2)
+    assert out.shape == (1, 1, 6, 6)
+    assert float(out.asnumpy()[0, 0, 0, 0]) == 5.0  # reflected corner
+    # 3D autograd flows
+    from anand_mxnet_amd import autograd
+    with autograd.record():
+        loss = (c3(x3) ** 2).sum()
+    loss.backward()
+    assert c3.weight.grad() is not None
