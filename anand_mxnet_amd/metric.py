@@ -292,3 +292,35 @@ def np_fn(numpy_feval, name=None, allow_extra_outputs=False):
 
 
 np_metric = np_fn
+
+
+@register
+class MCC(EvalMetric):
+    """Matthews correlation coefficient for binary classification
+    (reference: python/mxnet/metric.py — MCC)."""
+
+    def __init__(self, name="mcc", output_names=None, label_names=None):
+        super().__init__(name, output_names=output_names,
+                         label_names=label_names)
+        self._tp = self._tn = self._fp = self._fn = 0
+
+    def reset(self):
+        super().reset()
+        self._tp = self._tn = self._fp = self._fn = 0
+
+    def update(self, labels, preds):
+        labels, preds = _as_list(labels), _as_list(preds)
+        for label, pred in zip(labels, preds):
+            p = _np(pred)
+            pred_cls = p.argmax(axis=-1) if p.ndim > 1 else (p > 0.5)
+            lab = _np(label).astype("int64").ravel()
+            pred_cls = pred_cls.astype("int64").ravel()
+            self._tp += int(((pred_cls == 1) & (lab == 1)).sum())
+            self._tn += int(((pred_cls == 0) & (lab == 0)).sum())
+            self._fp += int(((pred_cls == 1) & (lab == 0)).sum())
+            self._fn += int(((pred_cls == 0) & (lab == 1)).sum())
+            self.num_inst += lab.size
+        tp, tn, fp, fn = self._tp, self._tn, self._fp, self._fn
+        denom = ((tp + fp) * (tp + fn) * (tn + fp) * (tn + fn)) ** 0.5
+        self.sum_metric = self.num_inst * (
+            (tp * tn - fp * fn) / denom if denom else 0.0)

@@ -162,3 +162,21 @@ def test_ctc_cosine_poisson_losses():
     labels = nd.array(np.array([[1, 2, 0], [3, 0, 0]], dtype="float32"))
     out = ctc(pred, labels).asnumpy()
     assert out.shape == (2,) and np.all(out > 0)
+
+
+def test_mcc_metric():
+    import anand_mxnet_amd as mx
+    m = mx.metric.MCC()
+    # perfect binary predictions
+    lab = nd.array(np.array([1, 0, 1, 0], dtype="float32"))
+    pred = nd.array(np.array([[0.1, 0.9], [0.8, 0.2], [0.2, 0.8],
+                              [0.7, 0.3]], dtype="float32"))
+    m.update(lab, pred)
+    assert abs(m.get()[1] - 1.0) < 1e-6
+    m.reset()
+    # anti-correlated -> -1
+    m.update(nd.array(np.array([0, 1], dtype="float32")),
+             nd.array(np.array([[0.1, 0.9], [0.9, 0.1]],
+                               dtype="float32")))
+    assert abs(m.get()[1] + 1.0) < 1e-6
+    assert mx.metric.create("mcc").name == "mcc"
