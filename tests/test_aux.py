@@ -188,6 +188,12 @@ def test_examples_smoke(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     r = subprocess.run(
         [_sys.executable,
+         os.path.join(root, "examples", "machine-translation", "train.py"),
+         "--steps", "2", "--log-interval", "1"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = subprocess.run(
+        [_sys.executable,
          os.path.join(root, "examples", "image-classification",
                       "train.py"),
          "--network", "resnet18_v1", "--data", "synthetic",
