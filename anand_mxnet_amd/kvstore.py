@@ -106,6 +106,13 @@ class KVStore:
     def push(self, key, value, priority=0):
         keys, values = self._key_value(key, value)
         for k, v in zip(keys, values):
+            gc = getattr(self, "_gc", None)
+            if gc is not None:
+                vs = _as_list(v)
+                v = [NDArray(gc.compress((k, i), x._t.detach()))
+                     for i, x in enumerate(vs)]
+                if len(v) == 1:
+                    v = v[0]
             agg = self._reduce(v)
             stored = self._store.get(k)
             if stored is None:
@@ -153,9 +160,16 @@ class KVStore:
         self.set_updater(_opt.get_updater(optimizer))
 
     def set_gradient_compression(self, compression_params):
-        raise MXNetError(
-            "2-bit gradient compression is not implemented on the RCCL path "
-            "(reference gradient_compression.cu; see docs/PARITY.md)")
+        """2-bit threshold compression with error feedback (reference:
+        src/kvstore/gradient_compression.cc).  Applied to each device's
+        contribution before the reduce, exactly like each worker's push
+        is quantized on the reference's wire."""
+        ctype = compression_params.get("type", "2bit")
+        if ctype != "2bit":
+            raise MXNetError(f"unsupported compression type {ctype!r}")
+        from .gradient_compression import GradientCompression
+        self._gc = GradientCompression(
+            float(compression_params.get("threshold", 0.5)))
 
     def save_optimizer_states(self, fname, dump_optimizer=False):
         if self._updater is None:

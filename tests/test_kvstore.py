@@ -122,3 +122,35 @@ def test_dist_kvstore_two_procs():
         p.join(110)
     for p in procs:
         assert p.exitcode == 0, f"dist worker failed: {p.exitcode}"
+
+
+def test_gradient_compression_2bit():
+    """Reference semantics: values clip to {-t, 0, +t}; residual error
+    feedback recovers the truncation over repeated pushes."""
+    import torch
+    from anand_mxnet_amd.gradient_compression import GradientCompression
+    gc = GradientCompression(threshold=0.5)
+    g = torch.tensor([0.7, -0.9, 0.2, 0.0])
+    q1 = gc.compress("k", g)
+    assert q1.tolist() == [0.5, -0.5, 0.0, 0.0]
+    # residual carries: second identical push pushes the remainder over t
+    q2 = gc.compress("k", g)
+    assert q2.tolist() == [0.5, -0.5, 0.0, 0.0]  # 0.2+0.2 < 0.5 still 0
+    q3 = gc.compress("k", g)
+    assert q3[2].item() == 0.5  # 0.6 accumulated > t
+    # error-feedback invariant: truth - sent == residual, exactly
+    total_q = q1 + q2 + q3
+    assert torch.allclose(3 * g - total_q, gc._residuals["k"])
+    # wire format round trip
+    words = GradientCompression.pack(q1, 0.5)
+    back = GradientCompression.unpack(words, q1.numel(), 0.5)
+    assert torch.equal(back, q1)
+    # kvstore integration
+    kv = kvs.create("local")
+    kv.set_gradient_compression({"type": "2bit", "threshold": 0.5})
+    kv.init("w", nd.zeros((4,)))
+    kv.push("w", nd.array(np.array([1.0, -1.0, 0.1, 0.0],
+                                   dtype="float32")))
+    out = nd.zeros((4,))
+    kv.pull("w", out=out)
+    np.testing.assert_allclose(out.asnumpy(), [0.5, -0.5, 0.0, 0.0])
