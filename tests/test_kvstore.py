@@ -146,11 +146,11 @@ def test_gradient_compression_2bit():
     back = GradientCompression.unpack(words, q1.numel(), 0.5)
     assert torch.equal(back, q1)
     # kvstore integration
-    kv = kvs.create("local")
-    kv.set_gradient_compression({"type": "2bit", "threshold": 0.5})
-    kv.init("w", nd.zeros((4,)))
-    kv.push("w", nd.array(np.array([1.0, -1.0, 0.1, 0.0],
+    kv_store = mx.kv.create("local")
+    kv_store.set_gradient_compression({"type": "2bit", "threshold": 0.5})
+    kv_store.init("w", nd.zeros((4,)))
+    kv_store.push("w", nd.array(np.array([1.0, -1.0, 0.1, 0.0],
                                    dtype="float32")))
     out = nd.zeros((4,))
-    kv.pull("w", out=out)
+    kv_store.pull("w", out=out)
     np.testing.assert_allclose(out.asnumpy(), [0.5, -0.5, 0.0, 0.0])
