@@ -789,3 +789,55 @@ def roi_align(data, rois, pooled_size=(1, 1), spatial_scale=1.0,
             padding_mode="zeros", align_corners=True)
         out[i] = torch.nn.functional.avg_pool2d(samp, sr)[0].to(data.dtype)
     return out
+
+
+@register("swapaxes")
+@register("SwapAxis")
+def swapaxes(data, dim1=0, dim2=0):
+    return data.transpose(int(dim1), int(dim2))
+
+
+@register("reverse")
+def reverse(data, axis=0):
+    axes = [axis] if isinstance(axis, int) else list(axis)
+    return torch.flip(data, axes)
+
+
+@register("space_to_depth")
+def space_to_depth(data, block_size):
+    b = int(block_size)
+    n, c, h, w = data.shape
+    x = data.reshape(n, c, h // b, b, w // b, b)
+    return x.permute(0, 3, 5, 1, 2, 4).reshape(n, c * b * b, h // b, w // b)
+
+
+@register("depth_to_space")
+def depth_to_space(data, block_size):
+    b = int(block_size)
+    n, c, h, w = data.shape
+    x = data.reshape(n, b, b, c // (b * b), h, w)
+    return x.permute(0, 3, 4, 1, 5, 2).reshape(n, c // (b * b), h * b, w * b)
+
+
+@register("SliceChannel")
+def slice_channel(data, num_outputs, axis=1, squeeze_axis=False):
+    """Legacy split (reference: src/operator/slice_channel.cc)."""
+    outs = torch.chunk(data, int(num_outputs), dim=int(axis))
+    if squeeze_axis:
+        outs = [o.squeeze(int(axis)) for o in outs]
+    return list(outs)
+
+
+@register("broadcast_like")
+def broadcast_like(lhs, rhs):
+    return lhs.expand(rhs.shape)
+
+
+@register("shape_array")
+def shape_array(data):
+    return torch.tensor(list(data.shape), dtype=torch.int64)
+
+
+@register("size_array")
+def size_array(data):
+    return torch.tensor([data.numel()], dtype=torch.int64)

@@ -564,3 +564,45 @@ def ctc_loss(data, label, data_lengths=None, label_lengths=None,
             (lab >= 0).sum(dim=1).long()
     return F.ctc_loss(log_probs, lab, in_lens, tgt_lens, blank=blank,
                       reduction="none", zero_infinity=True)
+
+
+# ---------------- Module-era regression heads ----------------
+
+class _RegressionOutputFn(torch.autograd.Function):
+    """Identity/sigmoid/mae heads whose backward injects (pred - label)
+    (reference: src/operator/regression_output{.cc,-inl.h})."""
+
+    @staticmethod
+    def forward(ctx, data, label, grad_scale, kind):
+        pred = torch.sigmoid(data) if kind == "logistic" else data
+        ctx.save_for_backward(pred, label)
+        ctx.grad_scale = grad_scale
+        ctx.kind = kind
+        return pred
+
+    @staticmethod
+    def backward(ctx, gout):
+        pred, label = ctx.saved_tensors
+        lab = label.reshape(pred.shape).to(pred.dtype)
+        n = pred.shape[0] if pred.dim() else 1
+        if ctx.kind == "mae":
+            g = torch.sign(pred - lab)
+        else:  # linear and logistic share (pred - label)
+            g = pred - lab
+        g = g * (ctx.grad_scale / max(n, 1))
+        return g.to(pred.dtype), None, None, None
+
+
+@register("LinearRegressionOutput")
+def linear_regression_output(data, label, grad_scale=1.0):
+    return _RegressionOutputFn.apply(data, label, grad_scale, "linear")
+
+
+@register("LogisticRegressionOutput")
+def logistic_regression_output(data, label, grad_scale=1.0):
+    return _RegressionOutputFn.apply(data, label, grad_scale, "logistic")
+
+
+@register("MAERegressionOutput")
+def mae_regression_output(data, label, grad_scale=1.0):
+    return _RegressionOutputFn.apply(data, label, grad_scale, "mae")

@@ -231,3 +231,49 @@ def test_fft_ifft_roialign():
                               spatial_scale=1.0)
     np.testing.assert_allclose(_np(ra).ravel(),
                                [3.75, 5.25, 9.75, 11.25], atol=1e-4)
+
+
+def test_space_depth_swap_reverse():
+    x = nd.array(np.random.RandomState(4)
+                 .randn(2, 3, 4, 4).astype("float32"))
+    s = nd.space_to_depth(x, 2)
+    assert s.shape == (2, 12, 2, 2)
+    d = nd.depth_to_space(s, 2)
+    np.testing.assert_allclose(_np(d), _np(x))
+    assert nd.swapaxes(x, 1, 3).shape == (2, 4, 4, 3)
+    np.testing.assert_allclose(_np(nd.reverse(x, axis=2)),
+                               _np(x)[:, :, ::-1])
+    np.testing.assert_allclose(_np(nd.broadcast_like(
+        nd.ones((1, 4)), nd.zeros((3, 4)))), np.ones((3, 4)))
+    np.testing.assert_allclose(_np(nd.shape_array(x)), [2, 3, 4, 4])
+    np.testing.assert_allclose(_np(nd.size_array(x)), [96])
+    outs = nd.SliceChannel(x, 3, axis=1, squeeze_axis=True)
+    assert len(outs) == 3 and outs[0].shape == (2, 4, 4)
+
+
+def test_regression_outputs():
+    from anand_mxnet_amd import autograd
+    pred = nd.array(np.array([[1.0], [2.0]], dtype="float32"))
+    pred.attach_grad()
+    lab = nd.array(np.array([0.5, 0.5], dtype="float32"))
+    with autograd.record():
+        out = nd.LinearRegressionOutput(pred, lab)
+    out.backward()
+    np.testing.assert_allclose(_np(pred.grad).ravel(), [0.25, 0.75])
+    pred2 = nd.array(np.array([[0.0]], dtype="float32"))
+    pred2.attach_grad()
+    with autograd.record():
+        out = nd.LogisticRegressionOutput(pred2,
+                                          nd.array(np.array([1.0],
+                                                   dtype="float32")))
+    np.testing.assert_allclose(_np(out), [[0.5]])
+    out.backward()
+    np.testing.assert_allclose(_np(pred2.grad), [[-0.5]])
+    pred3 = nd.array(np.array([[2.0]], dtype="float32"))
+    pred3.attach_grad()
+    with autograd.record():
+        out = nd.MAERegressionOutput(pred3,
+                                     nd.array(np.array([0.0],
+                                              dtype="float32")))
+    out.backward()
+    np.testing.assert_allclose(_np(pred3.grad), [[1.0]])
