@@ -46,3 +46,22 @@ def concat(*args, dim=1, axis=None):
     if axis is not None:
         dim = axis
     return _ops.OPS["concat"](*args, dim=dim)
+
+
+class _ContribNS:
+    """mx.nd.contrib.* namespace (reference: ndarray/contrib.py —
+    contrib ops are registered with a `_contrib_` prefix and re-exposed
+    bare under nd.contrib)."""
+
+    def __getattr__(self, item):
+        fn = globals().get(f"_contrib_{item}")
+        if fn is None:
+            import sys
+            mod = sys.modules[__name__]
+            fn = getattr(mod, f"_contrib_{item}", None)
+        if fn is None:
+            raise AttributeError(f"nd.contrib has no op {item!r}")
+        return fn
+
+
+contrib = _ContribNS()

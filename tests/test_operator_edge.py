@@ -277,3 +277,12 @@ def test_regression_outputs():
                                               dtype="float32")))
     out.backward()
     np.testing.assert_allclose(_np(pred3.grad), [[1.0]])
+
+
+def test_nd_contrib_namespace():
+    x = nd.array(np.array([[1.0, 2.0, 3.0, 4.0]], dtype="float32"))
+    assert nd.contrib.fft(x).shape == (1, 8)
+    assert abs(float(_np(nd.contrib.div_sqrt_dim(x))[0, 0]) - 0.5) < 1e-6
+    import pytest as _pytest
+    with _pytest.raises(AttributeError):
+        nd.contrib.nonexistent_op
