@@ -202,3 +202,30 @@ def test_color_transforms():
     s = T.RandomSaturation(0.9)
     np.testing.assert_allclose(s(gray).asnumpy(), gray.asnumpy(),
                                atol=1e-6)
+
+
+def test_image_det_iter():
+    import numpy as np
+    from anand_mxnet_amd.image_detection import (ImageDetIter,
+                                                 DetHorizontalFlipAug)
+    from anand_mxnet_amd.ndarray.ndarray import NDArray
+    import torch
+    rs = np.random.RandomState(0)
+    items = []
+    for i in range(5):
+        img = rs.rand(8, 8, 3).astype("float32")
+        boxes = np.array([[1, 0.1, 0.2, 0.4, 0.6],
+                          [2, 0.5, 0.5, 0.9, 0.8]][:1 + i % 2],
+                         dtype="float32")
+        items.append((img, boxes))
+    it = ImageDetIter(batch_size=2, data_shape=(3, 8, 8), imglist=items)
+    b = next(iter(it))
+    assert b.data[0].shape == (2, 3, 8, 8)
+    assert b.label[0].shape[0] == 2 and b.label[0].shape[2] == 5
+    # flip aug maps x coords correctly and deterministically at p=1
+    aug = DetHorizontalFlipAug(p=1.1)
+    img = NDArray(torch.rand(4, 4, 3))
+    lab = np.array([[0, 0.1, 0.2, 0.4, 0.6]], dtype="float32")
+    img2, lab2 = aug(img, lab)
+    np.testing.assert_allclose(lab2[0], [0, 0.6, 0.2, 0.9, 0.6],
+                               rtol=1e-6)
