@@ -42,7 +42,9 @@ def _is_dist_env() -> bool:
 def _ensure_pg():
     if dist.is_initialized():
         return
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    backend = os.environ.get(
+        "MXA_DIST_BACKEND",
+        "nccl" if torch.cuda.is_available() else "gloo")
     dist.init_process_group(backend=backend, init_method="env://")
 
 
