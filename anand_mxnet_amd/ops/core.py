@@ -841,3 +841,66 @@ def shape_array(data):
 @register("size_array")
 def size_array(data):
     return torch.tensor([data.numel()], dtype=torch.int64)
+
+
+# ---------------- int8 quantization ops (reference src/operator/quantization)
+
+@register("_contrib_quantize")
+@register("quantize")
+def quantize(data, min_range, max_range, out_type="uint8"):
+    """Affine quantization (reference: quantization/quantize.cc):
+    returns (q, min, max)."""
+    lo = float(min_range.reshape(-1)[0]) \
+        if torch.is_tensor(min_range) else float(min_range)
+    hi = float(max_range.reshape(-1)[0]) \
+        if torch.is_tensor(max_range) else float(max_range)
+    if out_type == "uint8":
+        scale = 255.0 / max(hi - lo, 1e-12)
+        q = ((data.float() - lo) * scale).round().clamp(0, 255) \
+            .to(torch.uint8)
+    else:  # int8: symmetric around 0 (reference quantize_v2 int8 mode)
+        amax = max(abs(lo), abs(hi), 1e-12)
+        scale = 127.0 / amax
+        q = (data.float() * scale).round().clamp(-127, 127).to(torch.int8)
+        lo, hi = -amax, amax
+    return [q, torch.tensor([lo]), torch.tensor([hi])]
+
+
+@register("_contrib_quantize_v2")
+@register("quantize_v2")
+def quantize_v2(data, out_type="int8", min_calib_range=None,
+                max_calib_range=None):
+    lo = min_calib_range if min_calib_range is not None \
+        else float(data.min())
+    hi = max_calib_range if max_calib_range is not None \
+        else float(data.max())
+    return quantize(data, torch.tensor([lo]), torch.tensor([hi]),
+                    out_type=out_type)
+
+
+@register("_contrib_dequantize")
+@register("dequantize")
+def dequantize(data, min_range, max_range, out_type="float32"):
+    lo = float(min_range.reshape(-1)[0])
+    hi = float(max_range.reshape(-1)[0])
+    if data.dtype == torch.uint8:
+        scale = (hi - lo) / 255.0
+        return data.float() * scale + lo
+    amax = max(abs(lo), abs(hi))
+    return data.float() * (amax / 127.0)
+
+
+@register("_contrib_requantize")
+@register("requantize")
+def requantize(data, min_range, max_range, min_calib_range=None,
+               max_calib_range=None):
+    """int32 accum -> int8 with new range (reference requantize.cc)."""
+    f = dequantize(data.to(torch.float32),
+                   min_range, max_range) if data.dtype != torch.int32 \
+        else data.float() * (max(abs(float(min_range.reshape(-1)[0])),
+                                 abs(float(max_range.reshape(-1)[0])))
+                             / float(2 ** 31 - 1))
+    lo = min_calib_range if min_calib_range is not None else float(f.min())
+    hi = max_calib_range if max_calib_range is not None else float(f.max())
+    return quantize(f, torch.tensor([lo]), torch.tensor([hi]),
+                    out_type="int8")

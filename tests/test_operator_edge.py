@@ -286,3 +286,16 @@ def test_nd_contrib_namespace():
     import pytest as _pytest
     with _pytest.raises(AttributeError):
         nd.contrib.nonexistent_op
+
+
+def test_int8_quantize_roundtrip():
+    x = nd.array(np.array([0.0, 0.25, 0.5, 1.0], dtype="float32"))
+    q, lo, hi = nd.quantize(x, nd.array([0.0]), nd.array([1.0]),
+                            out_type="uint8")
+    d = nd.dequantize(q, lo, hi)
+    np.testing.assert_allclose(_np(d), _np(x), atol=1.0 / 255)
+    x2 = nd.array(np.array([-2.0, 0.5, 2.0], dtype="float32"))
+    q2, l2, h2 = nd.quantize_v2(x2, out_type="int8")
+    d2 = nd.dequantize(q2, l2, h2)
+    np.testing.assert_allclose(_np(d2), _np(x2), atol=2.0 / 127)
+    assert float(_np(q2)[0]) == -127 and float(_np(q2)[2]) == 127
