@@ -241,3 +241,21 @@ def test_conv_pool_3d_1d_family():
         loss = (c3(x3) ** 2).sum()
     loss.backward()
     assert c3.weight.grad() is not None
+
+
+def test_new_layers_save_load_roundtrip(tmp_path):
+    from anand_mxnet_amd.gluon import nn as gnn
+    net = gnn.HybridSequential()
+    net.add(gnn.Conv3D(4, 3, padding=1))
+    net.add(gnn.Conv1DTranspose(2, 3, in_channels=0))
+    rs = np.random.RandomState(1)
+    x = nd.array(rs.randn(1, 2, 4, 4, 4).astype("float32"))
+    net.initialize()
+    # build conv3d only (shapes differ per child); save/load params
+    y = net._children["0"](x)
+    f = str(tmp_path / "c3.params")
+    net._children["0"].save_parameters(f)
+    net2 = gnn.Conv3D(4, 3, padding=1)
+    net2.load_parameters(f)
+    y2 = net2(x)
+    np.testing.assert_allclose(y.asnumpy(), y2.asnumpy(), rtol=1e-6)
