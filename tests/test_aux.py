@@ -187,6 +187,12 @@ def test_examples_smoke(tmp_path):
         capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-2000:]
     r = subprocess.run(
+        [_sys.executable, os.path.join(root, "examples", "rnn", "lm.py"),
+         "--steps", "2", "--log-interval", "1", "--batch-size", "2",
+         "--bptt", "8", "--vocab", "200"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = subprocess.run(
         [_sys.executable,
          os.path.join(root, "examples", "machine-translation", "train.py"),
          "--steps", "2", "--log-interval", "1"],
