@@ -133,3 +133,57 @@ def test_wd_mult_lr_mult():
     o.set_wd_mult({0: 0.0})
     assert o._get_lr(0) == 0.5
     assert o._get_wd(0) == 0.0
+
+
+def test_lamb_vs_numpy():
+    """LAMB update vs a slow numpy reference (reference: optimizer.py
+    LAMB / lamb_update_phase1+2)."""
+    from anand_mxnet_amd import optimizer as opt
+    rs = np.random.RandomState(0)
+    w0 = rs.normal(size=(6,)).astype("float32")
+    g0 = rs.normal(size=(6,)).astype("float32")
+    lamb = opt.create("lamb", learning_rate=0.01, beta1=0.9, beta2=0.999,
+                      epsilon=1e-6, wd=0.01)
+    w = nd.array(w0.copy())
+    grad = nd.array(g0.copy())
+    state = lamb.create_state(0, w)
+    lamb.update(0, w, grad, state)
+    # numpy reference
+    m = (1 - 0.9) * g0
+    v = (1 - 0.999) * g0 * g0
+    mh = m / (1 - 0.9)
+    vh = v / (1 - 0.999)
+    upd = mh / (np.sqrt(vh) + 1e-6) + 0.01 * w0
+    r1 = np.linalg.norm(w0)
+    r2 = np.linalg.norm(upd)
+    trust = r1 / r2 if (r1 > 0 and r2 > 0) else 1.0
+    ref = w0 - 0.01 * trust * upd
+    np.testing.assert_allclose(w.asnumpy(), ref, rtol=1e-4, atol=1e-5)
+
+
+def test_ftrl_signum_nag_run():
+    from anand_mxnet_amd import optimizer as opt
+    for name, kw in [("ftrl", {"learning_rate": 0.1}),
+                     ("signum", {"learning_rate": 0.01}),
+                     ("nag", {"learning_rate": 0.1, "momentum": 0.9})]:
+        o = opt.create(name, **kw)
+        w = nd.array(np.ones(4, dtype="float32"))
+        g = nd.array(np.full(4, 0.5, dtype="float32"))
+        st = o.create_state(0, w)
+        before = w.asnumpy().copy()
+        o.update(0, w, g, st)
+        o.update(0, w, g, st)
+        assert not np.allclose(w.asnumpy(), before), name
+
+
+def test_adamw_decoupled_wd():
+    """AdamW: wd acts on the weight directly, not through the adaptive
+    scaling (reference contrib adamw)."""
+    from anand_mxnet_amd import optimizer as opt
+    o = opt.create("adamw", learning_rate=0.1, wd=0.5)
+    w = nd.array(np.array([1.0], dtype="float32"))
+    g = nd.array(np.array([0.0], dtype="float32"))
+    st = o.create_state(0, w)
+    o.update(0, w, g, st)
+    # zero grad: only decay moves the weight: w -= lr * wd * w
+    np.testing.assert_allclose(w.asnumpy(), [1.0 - 0.1 * 0.5], rtol=1e-5)
