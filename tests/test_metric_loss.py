@@ -180,3 +180,45 @@ def test_mcc_metric():
                                dtype="float32")))
     assert abs(m.get()[1] + 1.0) < 1e-6
     assert mx.metric.create("mcc").name == "mcc"
+
+
+def test_f1_mcc_vs_sklearn():
+    from sklearn.metrics import f1_score, matthews_corrcoef
+    import anand_mxnet_amd as mx
+    rs = np.random.RandomState(11)
+    lab = rs.randint(0, 2, 64).astype("float32")
+    prob = rs.rand(64, 2).astype("float32")
+    pred_cls = prob.argmax(axis=1)
+    f1 = mx.metric.F1()
+    f1.update(nd.array(lab), nd.array(prob))
+    np.testing.assert_allclose(f1.get()[1], f1_score(lab, pred_cls),
+                               rtol=1e-6)
+    mcc = mx.metric.MCC()
+    mcc.update(nd.array(lab), nd.array(prob))
+    np.testing.assert_allclose(mcc.get()[1],
+                               matthews_corrcoef(lab, pred_cls),
+                               rtol=1e-6, atol=1e-12)
+
+
+def test_topk_perplexity_pearson():
+    import anand_mxnet_amd as mx
+    m = mx.metric.TopKAccuracy(top_k=2)
+    lab = nd.array(np.array([0, 1, 2], dtype="float32"))
+    prob = nd.array(np.array([[0.5, 0.3, 0.2],    # top2 {0,1} hit
+                              [0.1, 0.2, 0.7],    # top2 {2,1} hit
+                              [0.6, 0.3, 0.1]],   # top2 {0,1} miss
+                             dtype="float32"))
+    m.update(lab, prob)
+    np.testing.assert_allclose(m.get()[1], 2.0 / 3, rtol=1e-6)
+    p = mx.metric.Perplexity(ignore_label=None)
+    probs = nd.array(np.array([[0.5, 0.5], [0.25, 0.75]],
+                              dtype="float32"))
+    p.update(nd.array(np.array([0, 1], dtype="float32")), probs)
+    expect = np.exp(-(np.log(0.5) + np.log(0.75)) / 2)
+    np.testing.assert_allclose(p.get()[1], expect, rtol=1e-5)
+    pc = mx.metric.PearsonCorrelation()
+    a = np.array([1.0, 2, 3, 4], dtype="float32")
+    b = np.array([2.0, 4, 6.2, 7.9], dtype="float32")
+    pc.update(nd.array(a), nd.array(b))
+    np.testing.assert_allclose(pc.get()[1], np.corrcoef(a, b)[0, 1],
+                               rtol=1e-4)
