@@ -126,3 +126,15 @@ def test_grad_add_req():
             y = (x * 2).sum()
         y.backward()
     assert_almost_equal(x.grad.asnumpy(), [6.0])
+
+
+def test_numeric_gradient_layernorm_gelu():
+    x = mx.test_utils.rand_ndarray((4, 8))
+    g = mx.test_utils.rand_ndarray((8,), scale=0.5)
+    b = mx.test_utils.rand_ndarray((8,), scale=0.5)
+    check_numeric_gradient(
+        lambda a, gg, bb: nd.LayerNorm(a, gg, bb, eps=1e-5), [x, g, b],
+        rtol=2e-2, atol=1e-3)
+    y = mx.test_utils.rand_ndarray((3, 5))
+    check_numeric_gradient(lambda a: nd.gelu(a), [y], rtol=2e-2,
+                           atol=1e-3)
