@@ -73,3 +73,37 @@ class BatchSampler(Sampler):
         if self._last_batch == "rollover":
             return (len(self._prev) + len(self._sampler)) // self._batch_size
         raise ValueError(self._last_batch)
+
+
+class IntervalSampler(Sampler):
+    """Sample i, i+interval, i+2*interval, ... for each start i
+    (reference: gluon/data/sampler.py — IntervalSampler)."""
+
+    def __init__(self, length, interval, rollover=True):
+        assert 0 < interval <= length
+        self._length = length
+        self._interval = interval
+        self._rollover = rollover
+
+    def __iter__(self):
+        starts = range(self._interval) if self._rollover else [0]
+        for i in starts:
+            yield from range(i, self._length, self._interval)
+
+    def __len__(self):
+        return self._length if self._rollover \
+            else (self._length + self._interval - 1) // self._interval
+
+
+class FilterSampler(Sampler):
+    """Indices of dataset elements passing a predicate (reference:
+    gluon/data/sampler.py — FilterSampler)."""
+
+    def __init__(self, fn, dataset):
+        self._indices = [i for i in range(len(dataset)) if fn(dataset[i])]
+
+    def __iter__(self):
+        return iter(self._indices)
+
+    def __len__(self):
+        return len(self._indices)

@@ -229,3 +229,16 @@ def test_image_det_iter():
     img2, lab2 = aug(img, lab)
     np.testing.assert_allclose(lab2[0], [0, 0.6, 0.2, 0.9, 0.6],
                                rtol=1e-6)
+
+
+def test_interval_filter_samplers():
+    from anand_mxnet_amd.gluon.data.sampler import (IntervalSampler,
+                                                    FilterSampler)
+    s = IntervalSampler(6, 2)
+    assert list(s) == [0, 2, 4, 1, 3, 5] and len(s) == 6
+    s2 = IntervalSampler(6, 2, rollover=False)
+    assert list(s2) == [0, 2, 4] and len(s2) == 3
+    from anand_mxnet_amd.gluon.data.dataset import SimpleDataset
+    ds = SimpleDataset([1, 2, 3, 4, 5])
+    f = FilterSampler(lambda v: v % 2 == 1, ds)
+    assert list(f) == [0, 2, 4]
