@@ -91,3 +91,33 @@ def test_bad_magic(tmp_path):
     open(f, "wb").write(b"\x00" * 32)
     with pytest.raises(mx.MXNetError):
         load_ndarrays(f)
+
+
+def test_roundtrip_dtype_matrix(tmp_path):
+    """Every dtype the reference serializer carries (TShape + type flag +
+    raw blob) must round-trip byte-faithfully."""
+    from anand_mxnet_amd.serializer import save_ndarrays, load_ndarrays
+    path = str(tmp_path / "dt.params")
+    arrays = {}
+    for i, dt in enumerate(["float32", "float64", "float16", "bfloat16",
+                            "uint8", "int32", "int64", "int8"]):
+        arr = nd.array(np.arange(6).reshape(2, 3), dtype=dt)
+        arrays[f"arg:w{i}_{dt}"] = arr
+    save_ndarrays(path, arrays)
+    back = load_ndarrays(path)
+    assert set(back) == set(arrays)
+    for k, v in arrays.items():
+        got = back[k]
+        assert got._t.dtype == v._t.dtype, k
+        np.testing.assert_array_equal(
+            got.astype("float64").asnumpy(),
+            v.astype("float64").asnumpy())
+
+
+def test_large_name_and_many_arrays(tmp_path):
+    from anand_mxnet_amd.serializer import save_ndarrays, load_ndarrays
+    path = str(tmp_path / "many.params")
+    arrays = {f"arg:{'x' * 200}_{i}": nd.ones((1,)) for i in range(64)}
+    save_ndarrays(path, arrays)
+    back = load_ndarrays(path)
+    assert len(back) == 64
