@@ -125,6 +125,18 @@ class Trainer:
     def set_learning_rate(self, lr):
         self._optimizer.set_learning_rate(lr)
 
+    def no_sync(self):
+        """Context manager for gradient accumulation: backwards inside it
+        skip the hook-launched bucket collectives (DDP no_sync parity).
+        The final (unwrapped) backward + step() reduces the accumulated
+        grads once."""
+        if not self._kv_initialized:
+            self._init_kvstore()
+        if self._reducer is not None and hasattr(self._reducer, "no_sync"):
+            return self._reducer.no_sync()
+        import contextlib
+        return contextlib.nullcontext()
+
     def step(self, batch_size, ignore_stale_grad=False):
         """rescale grads by 1/batch_size, all-reduce, apply optimizer."""
         if not self._kv_initialized:
@@ -164,6 +176,14 @@ class Trainer:
             if param.grad_req == "null":
                 continue
             for arr, grad in zip(param.list_data(), param.list_grad()):
+                if arr._t.grad is None:
+                    # no backward reached this param since the last step
+                    if ignore_stale_grad:
+                        continue
+                    raise MXNetError(
+                        f"Gradient of Parameter '{param.name}' has not "
+                        "been updated by backward since the last step; "
+                        "pass ignore_stale_grad=True to skip it")
                 updater(i, grad, arr)
 
     # ---------------- fused multi-tensor optimizer path ----------------
@@ -196,6 +216,13 @@ class Trainer:
             self._fused = None  # allocator moved a grad: rebuild
             return self._try_fused_update()
         opt = self._optimizer
+        # Adam bias correction below uses ONE step count for the whole
+        # group — valid only while every param's count agrees (a step that
+        # fell back to the per-param path for a subset would desync them)
+        counts = {opt._index_update_count.get(i, opt.begin_num_update)
+                  for i in fused["indices"]}
+        if len(counts) > 1:
+            return False  # desynced: per-param path applies exact factors
         opt._update_count(fused["indices"])
         tcount = opt._index_update_count[fused["indices"][0]]
         from ..ops import hip as hipops
@@ -224,6 +251,12 @@ class Trainer:
             return False
         if opt.lr_mult or opt.wd_mult or opt.lr_scheduler is not None:
             return False  # per-param lr/wd or schedule: per-param path
+        # per-Parameter multipliers (wd_mult=0 on bias/BN is common) are
+        # honored by _get_lr/_get_wd, which the fused kernel bypasses
+        for p in self._params:
+            if getattr(p, "lr_mult", 1.0) != 1.0 or \
+                    getattr(p, "wd_mult", 1.0) != 1.0:
+                return False
         is_adam = type(opt).__name__ == "Adam"
         updater = self._updaters[0]
         groups = {}

@@ -39,6 +39,7 @@ class GradBucket:
         self.pending = 0
         self.expected = len(tensors)
         self.work = None
+        self.ready_flag = False
 
     def attach(self):
         for t, v in zip(self.tensors, self.views):
@@ -55,6 +56,7 @@ class GradBucket:
     def reset(self):
         self.pending = 0
         self.work = None
+        self.ready_flag = False
 
 
 class GradFlattener:
@@ -113,11 +115,38 @@ class BucketedAllReducer(GradFlattener):
         assert dist.is_initialized()
         self.world_size = dist.get_world_size()
         self.average = average
-        self.overlap = overlap and all(t.is_cuda for t in param_tensors)
+        # hooks fire on GPU tensors (the RCCL path); MXA_FORCE_OVERLAP=1
+        # exercises the same hook machinery over gloo on CPU in tests
+        force = os.environ.get("MXA_FORCE_OVERLAP", "0") == "1"
+        self.overlap = overlap and (force or
+                                    all(t.is_cuda for t in param_tensors))
+        self._next_launch = 0
+        self._sync = True
+        self._overrun = False
         super().__init__(param_tensors, bucket_bytes)
         if self.overlap:
             for t in param_tensors:
                 t.register_post_accumulate_grad_hook(self._hook)
+
+    def no_sync(self):
+        """Context manager disabling hook-time collectives (DDP parity).
+
+        Gradient accumulation (several backwards per step) MUST wrap the
+        non-final backwards in ``no_sync()``: otherwise the first backward
+        would all-reduce partial grads in place and later local
+        contributions could never be folded into the already-mixed sums.
+        An un-synced extra backward is detected and raises in finish()."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def _cm():
+            prev = self._sync
+            self._sync = False
+            try:
+                yield
+            finally:
+                self._sync = prev
+        return _cm()
 
     def _launch(self, bucket: GradBucket):
         if self.average:
@@ -127,26 +156,43 @@ class BucketedAllReducer(GradFlattener):
 
     def _hook(self, param: torch.Tensor):
         b = self._t2bucket[param]
+        if b.work is not None:
+            # grads accumulated AFTER this bucket's collective launched:
+            # the in-place sum is corrupted beyond repair — flag and fail
+            # loudly in finish() instead of silently training on bad grads.
+            self._overrun = True
+            return
         if b.ready():
+            # a full backward pass landed on this bucket: rearm the
+            # counter so a later (post-no_sync) pass is counted afresh
+            b.pending = 0
+            if not self._sync:
+                return
             b.ready_flag = True
             # launch in fixed bucket order: every rank must issue the same
             # collective sequence on the communicator (DDP-style guard
             # against autograd scheduling divergence -> NCCL hang)
             while self._next_launch < len(self.buckets) and \
-                    getattr(self.buckets[self._next_launch], "ready_flag",
-                            False):
+                    self.buckets[self._next_launch].ready_flag:
                 self._launch(self.buckets[self._next_launch])
                 self._next_launch += 1
 
-    _next_launch = 0
+    def zero_grads(self):
+        super().zero_grads()
+        self._next_launch = 0
 
     def finish(self):
         """Complete all reductions (called from Trainer before update)."""
+        if self._overrun:
+            self._overrun = False
+            raise RuntimeError(
+                "backward ran after a gradient bucket was already "
+                "all-reduced this step; wrap non-final backwards in "
+                "trainer/reducer no_sync() for gradient accumulation")
         for b in self.buckets:
-            if b.work is None:  # non-overlap mode / hook missed / stragglers
+            if b.work is None:  # no_sync mode / hook missed / stragglers
                 self._launch(b)
             b.work.wait()
         for b in self.buckets:
             b.reset()
-            b.ready_flag = False
         self._next_launch = 0

@@ -7,8 +7,13 @@ file := uint64 kMXAPINDArrayListMagic(0x112) | uint64 reserved(0)
         | uint64 n_arrays | n_arrays * ndarray
         | uint64 n_names  | n_names  * (uint64 len | bytes)
 ndarray (V2) := uint32 0xF993FAC9 | int32 stype(-1 dense... see note)
-        | shape(uint32 ndim | uint32*ndim) | int32 dev_type | int32 dev_id
+        | shape(uint32 ndim | int64*ndim) | int32 dev_type | int32 dev_id
         | int32 dtype | raw data bytes (C-order, prod(shape)*sizeof(dtype))
+
+TShape dims are int64 for V1/V2 (V1 magic 0xF993FAC8 is "version 1, with
+int64_t TShape"); only the legacy V0 layout (magic==ndim) used uint32 dims.
+ndim==0 on the wire means an EMPTY/none array (upstream semantics), so
+saving a genuine 0-dim scalar raises instead of silently dropping its value.
 
 stype: NDArrayStorageType (include/mxnet/ndarray.h): kUndefinedStorage=-1,
 kDefaultStorage=0 (dense), kRowSparseStorage=1, kCSRStorage=2.  Dense writes 0.
@@ -51,13 +56,15 @@ def _tensor_of(x):
 
 def _write_ndarray(out: bytearray, t: torch.Tensor):
     t = t.detach().contiguous().cpu()
+    if t.dim() == 0:
+        raise MXNetError(
+            "cannot save a 0-dim array: on the wire ndim==0 means an "
+            "empty/none array (upstream semantics) — reshape to (1,) first")
     out += struct.pack("<I", NDARRAY_V2_MAGIC)
     out += struct.pack("<i", 0)  # stype: dense (kDefaultStorage)
     out += struct.pack("<I", t.dim())
     for s in t.shape:
-        out += struct.pack("<I", s)
-    if t.dim() == 0:
-        return
+        out += struct.pack("<q", s)  # int64 TShape dims (V1+ layout)
     out += struct.pack("<ii", 1, 0)  # ctx: cpu(0) (reference saves via cpu copy)
     out += struct.pack("<i", _DTYPE_CODE[t.dtype])
     if t.dtype == torch.bfloat16:
@@ -92,6 +99,7 @@ class _Reader:
 
 def _read_ndarray(r: _Reader) -> torch.Tensor:
     magic = r.u32()
+    legacy_v0 = False
     if magic == NDARRAY_V2_MAGIC or magic == NDARRAY_V1_MAGIC + 2:  # V2/V3
         stype = r.i32()
         if stype not in (0, -1):
@@ -101,11 +109,15 @@ def _read_ndarray(r: _Reader) -> torch.Tensor:
     elif magic == NDARRAY_V1_MAGIC:
         ndim = r.u32()
     else:
-        # legacy V0: magic itself was ndim (TShape saved directly)
+        # legacy V0: magic itself was ndim (TShape saved directly, uint32)
+        legacy_v0 = True
         ndim = magic
         if ndim > 32:
             raise MXNetError(f"bad .params magic 0x{magic:x}")
-    shape = [r.u32() for _ in range(ndim)]
+    if legacy_v0:
+        shape = [r.u32() for _ in range(ndim)]
+    else:
+        shape = [struct.unpack("<q", r.take(8))[0] for _ in range(ndim)]
     if ndim == 0:
         return torch.zeros(())
     r.i32()  # dev_type (ignored: arrays load to cpu, reference parity)

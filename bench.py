@@ -34,11 +34,15 @@ def main():
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    # more ranks than GPUs (e.g. 2-rank overlap validation on a 1-GPU box):
+    # fold ranks onto the available devices
+    local_rank = local_rank % max(torch.cuda.device_count(), 1)
     distributed = world_size > 1
     if distributed:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl")
+        backend = os.environ.get("MXA_DIST_BACKEND", "nccl")
+        dist.init_process_group(backend=backend)
 
     import anand_mxnet_amd as mx
     from anand_mxnet_amd import gluon, autograd, nd
@@ -126,6 +130,26 @@ def main():
     samples_per_sec = global_batch * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    # Self-evidencing GPU-busy proof (outside the timed region): profile two
+    # extra steps and report the summed device kernel time per step.  A
+    # GPU-bound step has kernel_ms_per_step ≈ ms_per_step, independent of
+    # any driver-side SMI sampling.
+    kernel_ms = None
+    n_kernels = None
+    try:
+        from torch.profiler import profile, ProfilerActivity
+        with profile(activities=[ProfilerActivity.CUDA]) as prof:
+            for _ in range(2):
+                run_step()
+            torch.cuda.synchronize()
+        evs = prof.key_averages()
+        total_us = sum(e.device_time_total for e in evs
+                       if e.device_time_total > 0)
+        n_kernels = sum(e.count for e in evs if e.device_time_total > 0)
+        kernel_ms = total_us / 1000.0 / 2.0
+    except Exception:
+        pass
+
     if rank == 0:
         print(json.dumps({
             "metric": "samples/sec (whole node) ResNet-50 ImageNet-shape "
@@ -145,6 +169,13 @@ def main():
                        "image": S, "layout": "NHWC",
                        "parallelism": f"dp{world_size}",
                        "loss": round(lval, 4)},
+            "gpu_evidence": {
+                "kernel_ms_per_step": (round(kernel_ms, 3)
+                                       if kernel_ms is not None else None),
+                "kernel_launches_per_step": (n_kernels // 2
+                                             if n_kernels else None),
+                "mem_gb": round(
+                    torch.cuda.max_memory_allocated() / 2**30, 2)},
         }))
     if distributed:
         dist.destroy_process_group()

@@ -24,12 +24,23 @@ def test_file_header_bytes(tmp_path):
     assert struct.unpack("<I", raw[24:28])[0] == 0xF993FAC9
     # stype dense = 0
     assert struct.unpack("<i", raw[28:32])[0] == 0
-    # ndim=2, dims 2,3
-    assert struct.unpack("<III", raw[32:44]) == (2, 2, 3)
+    # ndim=2 (uint32), then dims 2,3 as int64 (V1+ TShape is int64_t)
+    assert struct.unpack("<I", raw[32:36])[0] == 2
+    assert struct.unpack("<qq", raw[36:52]) == (2, 3)
     # ctx cpu(1,0), dtype f32 (0)
-    assert struct.unpack("<iii", raw[44:56]) == (1, 0, 0)
+    assert struct.unpack("<iii", raw[52:64]) == (1, 0, 0)
     # 24 bytes of zeros data
-    assert raw[56:80] == b"\x00" * 24
+    assert raw[64:88] == b"\x00" * 24
+
+
+def test_zero_dim_save_raises(tmp_path):
+    """ndim==0 on the wire means an empty array upstream; saving a genuine
+    scalar must raise rather than silently drop its value."""
+    import torch
+    from anand_mxnet_amd.serializer import save_ndarrays
+    f = str(tmp_path / "z.params")
+    with pytest.raises(mx.MXNetError):
+        save_ndarrays(f, {"s": nd.NDArray(torch.tensor(3.5))})
 
 
 def test_roundtrip_dict(tmp_path):
