@@ -117,7 +117,16 @@ class _Linear(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             if w.shape[0] % 64 == 0 and w.shape[1] % 8 == 0:
                 dx = _C.gemm_nn(dy, w)  # no W^T copy (NN-form kernel)
-            else:  # e.g. the 1000-class head: library GEMM for dgrad
+            elif w.shape[1] % 8 == 0:
+                # non-%64 class dim (e.g. the 1000-way head): pad the
+                # reduce dim to 64 — zero rows of W contribute nothing —
+                # and stay on the NN MFMA kernel (no rocBLAS in the step)
+                N = w.shape[0]
+                npad = (-N) % 64
+                dyp = torch.nn.functional.pad(dy, (0, npad))
+                wp = torch.nn.functional.pad(w, (0, 0, 0, npad))
+                dx = _C.gemm_nn(dyp, wp)
+            else:
                 dx = torch.matmul(dy, w)
         if ctx.needs_input_grad[1]:
             dw = _wg_or_matmul(dy, x, w.dtype)
