@@ -74,10 +74,20 @@ def _tn_or_matmul(a, b):
 
 
 def _wg_or_matmul(dy, x, out_dtype):
-    """WG MFMA kernel when output dims are %8, else library GEMM."""
-    if dy.size(-1) % 8 == 0 and x.size(-1) % 8 == 0:
-        return _C.gemm_wg(dy, x).to(out_dtype)
-    return torch.matmul(dy.t(), x).to(out_dtype)
+    """WG MFMA kernel for dW = dy^T x.
+
+    The WG kernel emits bf16 directly (split slots folded + cast in its
+    epilogue), so .to() is a no-op for the bf16 weights of every config.
+    Narrow heads (e.g. the BERT NSP 2-class dy) are zero-padded to the
+    %8 output quantum and sliced back — a row prefix stays contiguous —
+    instead of falling back to a library GEMM."""
+    if x.size(-1) % 8 != 0:
+        return torch.matmul(dy.t(), x).to(out_dtype)
+    n = dy.size(-1)
+    if n % 8 != 0:
+        dyp = torch.nn.functional.pad(dy, (0, (-n) % 8))
+        return _C.gemm_wg(dyp.contiguous(), x)[:n].to(out_dtype)
+    return _C.gemm_wg(dy, x).to(out_dtype)
 
 
 @register_gpu("gemm")
@@ -269,16 +279,16 @@ class _Conv2dNHWC(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0) \
                     and K % 8 == 0 and C % 8 == 0:
-                # pointwise wgrad: plain WG GEMM ([C,K] == [1,1,C,K])
+                # pointwise wgrad: plain WG GEMM; the transposed fold
+                # emits [K,C] bf16 == (K,1,1,C) contiguous directly
                 N, P, Q, _ = dy.shape
-                dwf = _C.gemm_wg(x.reshape(N * P * Q, C),
-                                 dy.reshape(N * P * Q, K))
-                dw = dwf.reshape(1, 1, C, K).permute(3, 0, 1, 2) \
-                    .to(w.dtype)
+                dw = _C.gemm_wg(x.reshape(N * P * Q, C),
+                                dy.reshape(N * P * Q, K),
+                                True).reshape(K, 1, 1, C)
             else:
-                dwf = _C.conv2d_wgrad(x, dy, [R, S], list(stride),
-                                      list(pad), list(dil))  # [R,S,C,K]
-                dw = dwf.permute(3, 0, 1, 2).to(w.dtype)
+                # fold epilogue returns (K,R,S,C) bf16 — no permute/cast
+                dw = _C.conv2d_wgrad(x, dy, [R, S], list(stride),
+                                     list(pad), list(dil))
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy)
         return dx, dw, db, None, None, None, None
@@ -322,7 +332,7 @@ class _Conv2dIm2col(torch.autograd.Function):
         dy2 = dy.reshape(N * P * Q, K).contiguous()
         dw = db = None
         if ctx.needs_input_grad[1]:
-            dwf = _C.gemm_wg(dy2, A)       # [K, kdim_padded] f32
+            dwf = _C.gemm_wg(dy2, A)       # [K, kdim_padded] bf16
             dw = dwf[:, :kdim].reshape(w.shape).to(w.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy2)

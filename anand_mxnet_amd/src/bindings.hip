@@ -31,12 +31,14 @@ void conv2d_dgrad_bf16(const bf16*, const bf16*, bf16*, float*, int, int,
 void gemm_tn8_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
                    int, long, long, long, long, long, long, int, bool,
                    const bf16*, hipStream_t);
-// gemm_wg.hip
-void gemm_wg_bf16(const bf16*, const bf16*, float*, long, int, int, long,
-                  long, long, const bf16*, hipStream_t);
-void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, int, int, int, int,
+// gemm_wg.hip (slot-buffer split-K + fold epilogue)
+int gemm_wg_nslots(long, int, int, bool);
+void gemm_wg_bf16(const bf16*, const bf16*, float*, bf16*, bool, long, int,
+                  int, long, long, const bf16*, hipStream_t);
+int conv2d_wgrad_nslots(long, int, int, int, int);
+void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, bf16*, int, int,
                        int, int, int, int, int, int, int, int, int, int,
-                       int, const bf16*, hipStream_t);
+                       int, int, int, const bf16*, hipStream_t);
 // bn_pool.hip
 void bn_fwd_train(const bf16*, bf16*, float*, float*, float*, float*, float*,
                   float*, const float*, const float*, long, int, float,
@@ -261,15 +263,27 @@ std::vector<at::Tensor> gemm_tn_stats(const at::Tensor& A,
   return {out, stats};
 }
 
-// C[Mo,No] (f32) = sum_m A[m,Mo] * B[m,No]
-at::Tensor gemm_wg(const at::Tensor& A, const at::Tensor& B) {
+// C[Mo,No] (bf16) = sum_m A[m,Mo] * B[m,No]; transpose -> C[No,Mo].
+// Split-K goes through per-split f32 slot buffers folded (and cast, and
+// optionally transposed) in one pass — no HBM atomics, no memset, no
+// separate .to(bf16) launch.
+at::Tensor gemm_wg(const at::Tensor& A, const at::Tensor& B,
+                   bool transpose) {
   TORCH_CHECK(A.is_contiguous() && B.is_contiguous(), "gemm_wg: contiguous");
   long m = A.size(0), Mo = A.size(1), No = B.size(1);
   TORCH_CHECK(B.size(0) == m, "gemm_wg: reduce dim mismatch");
   TORCH_CHECK(Mo % 8 == 0 && No % 8 == 0, "gemm_wg: dims must be %8");
-  auto out = at::empty({Mo, No}, A.options().dtype(at::kFloat));
-  mxa::gemm_wg_bf16(bfp(A), bfp(B), out.data_ptr<float>(), m, (int)Mo,
-                    (int)No, Mo, No, No, zp(A), cur_stream());
+  int ns = mxa::gemm_wg_nslots(m, (int)Mo, (int)No, transpose);
+  auto out = transpose ? at::empty({No, Mo}, A.options())
+                       : at::empty({Mo, No}, A.options());
+  float* workp = nullptr;
+  at::Tensor work;
+  if (ns > 0) {
+    work = at::empty({ns, Mo, No}, A.options().dtype(at::kFloat));
+    workp = work.data_ptr<float>();
+  }
+  mxa::gemm_wg_bf16(bfp(A), bfp(B), workp, bfp_mut(out), transpose, m,
+                    (int)Mo, (int)No, Mo, No, zp(A), cur_stream());
   return out;
 }
 
@@ -444,6 +458,9 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w_rsck,
   return dx;
 }
 
+// Returns dW in (K, R, S, C) bf16 directly: the split slots are folded,
+// cast and transposed in one epilogue pass (the old path emitted f32
+// [R,S,C,K] + a torch permute + a torch cast per conv per step).
 at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy,
                         std::vector<long> kernel, std::vector<long> stride,
                         std::vector<long> pad, std::vector<long> dil) {
@@ -451,11 +468,16 @@ at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy,
   int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
   int R = kernel[0], S = kernel[1];
-  auto dw = at::empty({R, S, C, Kout}, x.options().dtype(at::kFloat));
-  mxa::conv2d_wgrad_bf16(bfp(x), bfp(dy), dw.data_ptr<float>(), Nb, H, W, C,
-                         Kout, R, S, P, Q, (int)stride[0], (int)stride[1],
-                         (int)pad[0], (int)pad[1], (int)dil[0], (int)dil[1],
-                         zp(x), cur_stream());
+  long mtotal = (long)Nb * P * Q;
+  int ns = mxa::conv2d_wgrad_nslots(mtotal, C, Kout, R, S);
+  auto work = at::empty({ns, (long)R * S * C, (long)Kout},
+                        x.options().dtype(at::kFloat));
+  auto dw = at::empty({Kout, R, S, C}, x.options());
+  mxa::conv2d_wgrad_bf16(bfp(x), bfp(dy), work.data_ptr<float>(),
+                         bfp_mut(dw), Nb, H, W, C, Kout, R, S, P, Q,
+                         (int)stride[0], (int)stride[1], (int)pad[0],
+                         (int)pad[1], (int)dil[0], (int)dil[1], zp(x),
+                         cur_stream());
   return dw;
 }
 
@@ -1070,7 +1092,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn", &gemm_tn, "C = A @ B^T (bf16, MFMA)");
   m.def("gemm_tn_stats", &gemm_tn_stats,
         "gemm_tn + fused per-column BN sums: {out[,stats]}");
-  m.def("gemm_wg", &gemm_wg, "C(f32) = sum_m A[m,:] outer B[m,:]");
+  m.def("gemm_wg", &gemm_wg,
+        "C(bf16) = sum_m A[m,:] outer B[m,:] (transpose: C[No,Mo])",
+        py::arg("A"), py::arg("B"), py::arg("transpose") = false);
   m.def("gemm_tn_fp8", &gemm_tn_fp8);
   m.def("cast_to_fp8", &cast_to_fp8);
   m.def("mx_cast", &mx_cast);

@@ -108,7 +108,8 @@ template <bool A_IM2COL>
 __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_core(
     WgPlainRows aplain, WgIm2colRows aim, WgPlainRows bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
-    int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic) {
+    int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic,
+    long strideSlot) {
   __shared__ __align__(16) bf16 smem[2][2][WBM * WBK];
 
   const int seg = blockIdx.y / nsplit, split = blockIdx.y % nsplit;
@@ -236,7 +237,9 @@ __global__ __launch_bounds__(WTHREADS, 2) void gemm_wg_core(
     __syncthreads();
   }
 
-  float* cseg = Cout + (long)seg * strideSeg;
+  // slot-buffer epilogue: each split owns a private f32 slot (plain
+  // stores, no HBM atomics, no pre-zero memset); wg_fold sums the slots
+  float* cseg = Cout + (long)seg * strideSeg + (long)split * strideSlot;
 #pragma unroll
   for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
@@ -367,7 +370,7 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
     TrPlainRows aplain, TrIm2colRows aim, TrPlainRows bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
     int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic,
-    ZMap za, ZMap zb, ZMap zc) {
+    ZMap za, ZMap zb, ZMap zc, long strideSlot = 0) {
   __shared__ __align__(16) bf16 smem[2][2][WBK * WBM];  // [buf][op][64][128]
 
   const int zb_ = blockIdx.z;
@@ -490,7 +493,7 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
     __syncthreads();
   }
 
-  float* cseg = Cout + (long)seg * strideSeg;
+  float* cseg = Cout + (long)seg * strideSeg + (long)split * strideSlot;
 #pragma unroll
   for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
@@ -617,7 +620,7 @@ __global__ __launch_bounds__(WT8, RING <= 4 ? 2 : 1) void gemm_wg_ring_core(
     TrPlainRows1 aplain, TrIm2colRows1 aim, TrPlainRows1 bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
     int S, int dh, int dw, int hsteps_total, int nsplit, int use_atomic,
-    ZMap za, ZMap zb, ZMap zc) {
+    ZMap za, ZMap zb, ZMap zc, long strideSlot = 0) {
   // [ring slot][op][32 rows][128 cols]
   __shared__ __align__(16) bf16 smem[RING][2][32 * 128];
 
@@ -732,7 +735,7 @@ __global__ __launch_bounds__(WT8, RING <= 4 ? 2 : 1) void gemm_wg_ring_core(
     __builtin_amdgcn_s_barrier();
   }
 
-  float* cseg = Cout + (long)seg * strideSeg;
+  float* cseg = Cout + (long)seg * strideSeg + (long)split * strideSlot;
 #pragma unroll
   for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
@@ -752,6 +755,129 @@ __global__ __launch_bounds__(WT8, RING <= 4 ? 2 : 1) void gemm_wg_ring_core(
           cseg[(long)grow * ldc + gcol] = v;
       }
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Split-slot fold epilogue
+// ---------------------------------------------------------------------------
+// Split-K blocks write per-split f32 slot buffers with plain stores (no
+// HBM atomics, no pre-zero memset — the round-1 PMC evidence showed the
+// small-tile wgrads wait-bound on exactly those).  This single pass then
+// sums the slots in f32 (deterministic, fixed order) and emits the final
+// bf16 tensor, fusing the f32->bf16 cast and, in the TRANSPOSE variant,
+// the [RSC,K] -> [K,RSC] weight-grad permute that used to be separate
+// torch launches after every wgrad.
+
+// At high split counts the final tensor is small (one or a few tiles), so
+// a one-stage fold would be a tiny latency-bound grid doing up to 128
+// serial slot loads per element.  Two levels instead: a WIDE partial stage
+// (grid.y = slot groups of 16) halves into f32 partial slots appended
+// after the input slots, then the final fold reads <= 8 of those.
+constexpr int WG_FOLD_GROUP = 16;
+
+__global__ __launch_bounds__(256) void wg_fold_partial(
+    const float* __restrict__ work, long slotStride, int nslots, long n,
+    float* __restrict__ outslots) {
+  long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (base >= n) return;
+  int cnt = min(WG_FOLD_GROUP, nslots - (int)blockIdx.y * WG_FOLD_GROUP);
+  const float* p =
+      work + (long)blockIdx.y * WG_FOLD_GROUP * slotStride + base;
+  f32x4 s0{0.f, 0.f, 0.f, 0.f}, s1{0.f, 0.f, 0.f, 0.f};
+  int sl = 0;
+  for (; sl + 2 <= cnt; sl += 2) {
+    s0 += *(const f32x4*)p;
+    s1 += *(const f32x4*)(p + slotStride);
+    p += 2 * slotStride;
+  }
+  if (sl < cnt) s0 += *(const f32x4*)p;
+  s0 += s1;
+  *(f32x4*)(outslots + (long)blockIdx.y * slotStride + base) = s0;
+}
+
+__global__ __launch_bounds__(256) void wg_fold_flat(
+    const float* __restrict__ work, long slotStride, int nslots, long n,
+    bf16* __restrict__ out) {
+  long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (base >= n) return;
+  f32x4 s0{0.f, 0.f, 0.f, 0.f}, s1{0.f, 0.f, 0.f, 0.f};
+  const float* p = work + base;
+  int sl = 0;
+  for (; sl + 2 <= nslots; sl += 2) {
+    s0 += *(const f32x4*)p;
+    s1 += *(const f32x4*)(p + slotStride);
+    p += 2 * slotStride;
+  }
+  if (sl < nslots) s0 += *(const f32x4*)p;
+  s0 += s1;
+  bf16x4 o{f2bf(s0.x), f2bf(s0.y), f2bf(s0.z), f2bf(s0.w)};
+  *(bf16x4*)(out + base) = o;
+}
+
+__global__ __launch_bounds__(256) void wg_fold_tr(
+    const float* __restrict__ work, long slotStride, int nslots, long rows,
+    long cols, bf16* __restrict__ out) {
+  __shared__ float tile[32][33];  // +1: no bank conflicts on the transpose
+  long ntr = (rows + 31) / 32;
+  long tr = (long)(blockIdx.x % ntr) * 32;
+  long tc = (long)(blockIdx.x / ntr) * 32;
+  int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;  // 32 x 8
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int r = ty + p * 8;
+    long gr = tr + r, gc = tc + tx;
+    float s0 = 0.f, s1 = 0.f;
+    if (gr < rows && gc < cols) {
+      const float* q = work + gr * cols + gc;
+      int sl = 0;
+      for (; sl + 2 <= nslots; sl += 2) {
+        s0 += q[0];
+        s1 += q[slotStride];
+        q += 2 * slotStride;
+      }
+      if (sl < nslots) s0 += q[0];
+    }
+    tile[r][tx] = s0 + s1;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int c = ty + p * 8;
+    long gc = tc + c, gr = tr + tx;
+    if (gc < cols && gr < rows) out[gc * rows + gr] = f2bf(tile[tx][c]);
+  }
+}
+
+// slots the CALLER must allocate for `used` split slots (adds headroom
+// for the partial stage)
+static int wg_alloc_slots(int used) {
+  return used + (used > WG_FOLD_GROUP
+                     ? (used + WG_FOLD_GROUP - 1) / WG_FOLD_GROUP
+                     : 0);
+}
+
+void wg_fold(const float* work, long slotStride, int nslots, long rows,
+             long cols, bf16* out, bool transpose, hipStream_t stream) {
+  long n = rows * cols;  // cols % 8 == 0 -> n % 4 == 0
+  const float* src = work;
+  if (nslots > WG_FOLD_GROUP) {
+    int nchunk = (nslots + WG_FOLD_GROUP - 1) / WG_FOLD_GROUP;
+    float* partial = const_cast<float*>(work) + (long)nslots * slotStride;
+    long blocks = (n / 4 + 255) / 256;
+    wg_fold_partial<<<dim3((unsigned)blocks, nchunk), 256, 0, stream>>>(
+        work, slotStride, nslots, n, partial);
+    src = partial;
+    nslots = nchunk;
+  }
+  if (transpose) {
+    long ntr = (rows + 31) / 32, ntc = (cols + 31) / 32;
+    wg_fold_tr<<<dim3((unsigned)(ntr * ntc)), 256, 0, stream>>>(
+        src, slotStride, nslots, rows, cols, out);
+  } else {
+    long blocks = (n / 4 + 255) / 256;
+    wg_fold_flat<<<dim3((unsigned)blocks), 256, 0, stream>>>(
+        src, slotStride, nslots, n, out);
   }
 }
 
@@ -833,83 +959,127 @@ static int pick_split(long blocks_xy, long ksteps) {
   return target;
 }
 
-void gemm_wg_bf16(const bf16* A, const bf16* B, float* C, long mtotal,
-                  int Mo, int No, long lda, long ldb, long ldc,
-                  const bf16* zpage, hipStream_t stream) {
-  int ksteps = (int)((mtotal + WBK - 1) / WBK);
+struct WgPlanT {
+  int nsplit;  // grid.y factor
+  int used;    // splits that actually produce output (slots to fold)
+  int ksteps;  // 64-row reduce steps
+  int hsteps;  // 32-row reduce steps (ring kernel)
+};
+
+static WgPlanT wg_plan_(long bxy, long mtotal) {
+  WgPlanT p;
+  p.ksteps = (int)((mtotal + WBK - 1) / WBK);
+  p.hsteps = (int)((mtotal + 31) / 32);
+  p.nsplit = pick_split(bxy, p.ksteps);
+  int steps = wg_ring() ? p.hsteps : p.ksteps;
+  int per = (steps + p.nsplit - 1) / p.nsplit;
+  p.used = (steps + per - 1) / per;
+  return p;
+}
+
+int gemm_wg_nslots(long mtotal, int Mo, int No, bool transpose) {
+  // 0 = direct bf16 output, no workspace (single split, plain layout,
+  // kernels that have the OUTBF16 epilogue)
+  WgPlanT p = wg_plan_((long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN), mtotal);
+  bool tr_path = !wg_use_zip() && Mo % 8 == 0 && No % 8 == 0;
+  if (p.used == 1 && !transpose && tr_path) return 0;
+  return wg_alloc_slots(p.used);
+}
+
+void gemm_wg_bf16(const bf16* A, const bf16* B, float* work, bf16* out,
+                  bool transpose, long mtotal, int Mo, int No, long lda,
+                  long ldb, const bf16* zpage, hipStream_t stream) {
   long bxy = (long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN);
-  int nsplit = pick_split(bxy, ksteps);
-  dim3 grid((unsigned)bxy, nsplit, 1);
-  int use_atomic = nsplit > 1;
-  if (use_atomic) HIP_CHECK(hipMemsetAsync(C, 0, sizeof(float) * Mo * ldc,
-                                           stream));
+  WgPlanT pl = wg_plan_(bxy, mtotal);
+  dim3 grid((unsigned)bxy, pl.nsplit, 1);
+  long slotStride = (long)Mo * No;
+  const bool direct = work == nullptr;  // OUTBF16 epilogue, no fold pass
+  float* Cdst = direct ? (float*)out : work;
   if (!wg_use_zip() && lda % 8 == 0 && ldb % 8 == 0) {
     if (wg_ring()) {
       TrPlainRows1 a{A, zpage, mtotal, lda, Mo};
       TrPlainRows1 b{B, zpage, mtotal, ldb, No};
       TrIm2colRows1 dummy{};
-      int hsteps = (int)((mtotal + 31) / 32);
-      launch_wg_ring<false, false>(a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1,
-                                   hsteps, nsplit, use_atomic, ZMap{1, 0, 0},
-                                   ZMap{1, 0, 0}, ZMap{1, 0, 0}, grid,
-                                   stream);
-      return;
+      if (direct)
+        launch_wg_ring<false, true>(a, dummy, b, Cdst, Mo, No, No, 0, 1, 1,
+                                    1, pl.hsteps, pl.nsplit, 0, ZMap{1, 0, 0},
+                                    ZMap{1, 0, 0}, ZMap{1, 0, 0}, grid,
+                                    stream);
+      else
+        gemm_wg_ring_core<false, false, 4><<<grid, WT8, 0, stream>>>(
+            a, dummy, b, Cdst, Mo, No, No, 0, 1, 1, 1, pl.hsteps, pl.nsplit,
+            0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, slotStride);
+    } else {
+      TrPlainRows a{A, zpage, mtotal, lda, Mo};
+      TrPlainRows b{B, zpage, mtotal, ldb, No};
+      TrIm2colRows dummy{};
+      if (direct)
+        gemm_wg_tr_core<false, true><<<grid, WT8, 0, stream>>>(
+            a, dummy, b, Cdst, Mo, No, No, 0, 1, 1, 1, pl.ksteps, pl.nsplit,
+            0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, 0);
+      else
+        gemm_wg_tr_core<false, false><<<grid, WT8, 0, stream>>>(
+            a, dummy, b, Cdst, Mo, No, No, 0, 1, 1, 1, pl.ksteps, pl.nsplit,
+            0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, slotStride);
     }
-    TrPlainRows a{A, zpage, mtotal, lda, Mo};
-    TrPlainRows b{B, zpage, mtotal, ldb, No};
-    TrIm2colRows dummy{};
-    gemm_wg_tr_core<false><<<grid, WT8, 0, stream>>>(
-        a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic,
-        ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0});
-    return;
+  } else {
+    WgPlainRows a{A, zpage, mtotal, lda, Mo};
+    WgPlainRows b{B, zpage, mtotal, ldb, No};
+    WgIm2colRows dummy{};
+    gemm_wg_core<false><<<grid, WTHREADS, 0, stream>>>(
+        a, dummy, b, Cdst, Mo, No, No, 0, 1, 1, 1, pl.ksteps, pl.nsplit, 0,
+        slotStride);
   }
-  WgPlainRows a{A, zpage, mtotal, lda, Mo};
-  WgPlainRows b{B, zpage, mtotal, ldb, No};
-  WgIm2colRows dummy{};
-  gemm_wg_core<false><<<grid, WTHREADS, 0, stream>>>(
-      a, dummy, b, C, Mo, No, ldc, 0, 1, 1, 1, ksteps, nsplit, use_atomic);
+  if (!direct)
+    wg_fold(work, slotStride, pl.used, Mo, No, out, transpose, stream);
 }
 
-void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* dw_rsck,
-                       int Nb, int H, int W, int C, int Kout, int R, int S,
-                       int P, int Q, int sh, int sw, int ph, int pw, int dh,
-                       int dw_, const bf16* zpage, hipStream_t stream) {
+int conv2d_wgrad_nslots(long mtotal, int C, int Kout, int R, int S) {
+  WgPlanT p =
+      wg_plan_((long)cdiv_h(C, WBM) * cdiv_h(Kout, WBN) * R * S, mtotal);
+  return wg_alloc_slots(p.used);
+}
+
+void conv2d_wgrad_bf16(const bf16* x, const bf16* dy, float* work,
+                       bf16* out_krsc, int Nb, int H, int W, int C, int Kout,
+                       int R, int S, int P, int Q, int sh, int sw, int ph,
+                       int pw, int dh, int dw_, const bf16* zpage,
+                       hipStream_t stream) {
   long mtotal = (long)Nb * P * Q;
-  int ksteps = (int)((mtotal + WBK - 1) / WBK);
   long bxy = (long)cdiv_h(C, WBM) * cdiv_h(Kout, WBN);
-  int nsplit = pick_split(bxy * R * S, ksteps);
-  dim3 grid((unsigned)bxy, R * S * nsplit, 1);
-  int use_atomic = nsplit > 1;
-  if (use_atomic)
-    HIP_CHECK(hipMemsetAsync(dw_rsck, 0,
-                             sizeof(float) * (long)R * S * C * Kout, stream));
+  WgPlanT pl = wg_plan_(bxy * R * S, mtotal);
+  dim3 grid((unsigned)bxy, R * S * pl.nsplit, 1);
+  long strideSeg = (long)C * Kout;          // within a slot: [R*S][C][K]
+  long slotStride = strideSeg * R * S;      // between slots
   if (!wg_use_zip() && C % 8 == 0 && Kout % 8 == 0) {
     if (wg_ring()) {
       TrIm2colRows1 a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
       TrPlainRows1 b{dy, zpage, mtotal, (long)Kout, Kout};
       TrPlainRows1 dummy{};
-      int hsteps = (int)((mtotal + 31) / 32);
-      launch_wg_ring<true, false>(dummy, a, b, dw_rsck, C, Kout, Kout,
-                                  (long)C * Kout, S, dh, dw_, hsteps, nsplit,
-                                  use_atomic, ZMap{1, 0, 0}, ZMap{1, 0, 0},
-                                  ZMap{1, 0, 0}, grid, stream);
-      return;
+      gemm_wg_ring_core<true, false, 4><<<grid, WT8, 0, stream>>>(
+          dummy, a, b, work, C, Kout, Kout, strideSeg, S, dh, dw_, pl.hsteps,
+          pl.nsplit, 0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0},
+          slotStride);
+    } else {
+      TrIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
+      TrPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
+      TrPlainRows dummy{};
+      gemm_wg_tr_core<true><<<grid, WT8, 0, stream>>>(
+          dummy, a, b, work, C, Kout, Kout, strideSeg, S, dh, dw_, pl.ksteps,
+          pl.nsplit, 0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0},
+          slotStride);
     }
-    TrIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
-    TrPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
-    TrPlainRows dummy{};
-    gemm_wg_tr_core<true><<<grid, WT8, 0, stream>>>(
-        dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
-        ksteps, nsplit, use_atomic,
-        ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0});
-    return;
+  } else {
+    WgIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
+    WgPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
+    WgPlainRows dummy{};
+    gemm_wg_core<true><<<grid, WTHREADS, 0, stream>>>(
+        dummy, a, b, work, C, Kout, Kout, strideSeg, S, dh, dw_, pl.ksteps,
+        pl.nsplit, 0, slotStride);
   }
-  WgIm2colRows a{x, zpage, Nb, H, W, C, P, Q, sh, sw, ph, pw};
-  WgPlainRows b{dy, zpage, mtotal, (long)Kout, Kout};
-  WgPlainRows dummy{};
-  gemm_wg_core<true><<<grid, WTHREADS, 0, stream>>>(
-      dummy, a, b, dw_rsck, C, Kout, Kout, (long)C * Kout, S, dh, dw_,
-      ksteps, nsplit, use_atomic);
+  // fold all slots, transposing [R*S*C][K] -> [K][R*S*C] == (K,R,S,C)
+  wg_fold(work, slotStride, pl.used, (long)R * S * C, Kout, out_krsc, true,
+          stream);
 }
 
 // Batched attention dK/dV over per-(batch, head) slices (z = b*H + h):

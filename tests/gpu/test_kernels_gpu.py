@@ -67,6 +67,11 @@ def test_gemm_wg_vs_fp32():
         ref = a.t() @ b
         out = _C.gemm_wg(a.to(BF).contiguous(), b.to(BF).contiguous())
         assert_close(out, ref, rtol=3e-2, atol=5e-2, msg=f"wg {m}x{Mo}x{No}")
+        # transposed fold epilogue (the conv weight-grad permute path)
+        outT = _C.gemm_wg(a.to(BF).contiguous(), b.to(BF).contiguous(),
+                          True)
+        assert_close(outT, ref.t(), rtol=3e-2, atol=5e-2,
+                     msg=f"wgT {m}x{Mo}x{No}")
 
 
 def _conv_ref(x, w, stride, pad):
@@ -114,9 +119,8 @@ def test_conv2d_dgrad_wgrad(cfg):
 
     dwf = _C.conv2d_wgrad(x.detach().to(BF).contiguous(),
                           dy.to(BF).contiguous(), [R, R], [s, s], [p, p],
-                          [1, 1])
-    dw_ref = w.grad.permute(1, 2, 3, 0)  # (R,S,C,K)
-    assert_close(dwf, dw_ref, rtol=3e-2, atol=5e-2, msg=f"conv wgrad {cfg}")
+                          [1, 1])  # (K,R,S,C) bf16 (fold epilogue)
+    assert_close(dwf, w.grad, rtol=3e-2, atol=5e-2, msg=f"conv wgrad {cfg}")
 
 
 def test_bn_train_fwd_bwd():
