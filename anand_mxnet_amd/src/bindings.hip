@@ -39,6 +39,13 @@ int conv2d_wgrad_nslots(long, int, int, int, int);
 void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, bf16*, int, int,
                        int, int, int, int, int, int, int, int, int, int,
                        int, int, int, const bf16*, hipStream_t);
+bool conv_wg_c9_eligible(int, int, int, int, int, int, int, int, int, int,
+                         int, int, int, int);
+long conv_wg_c9_xp_elems(int, int, int, int);
+void conv_wg_pad_image(const bf16*, bf16*, int, int, int, int, hipStream_t);
+int conv2d_wgrad_c9_nslots(int, int, int, int, int);
+void conv2d_wgrad_c9_bf16(const bf16*, const bf16*, float*, bf16*, int, int,
+                          int, int, int, const bf16*, hipStream_t);
 // bn_pool.hip
 void bn_fwd_train(const bf16*, bf16*, float*, float*, float*, float*, float*,
                   float*, const float*, const float*, long, int, float,
@@ -468,6 +475,28 @@ at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy,
   int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
   int R = kernel[0], S = kernel[1];
+  static const bool c9_on = [] {
+    const char* e = getenv("MXA_WG_C9");
+    return !e || e[0] != '0';  // default on; MXA_WG_C9=0 -> per-seg path
+  }();
+  if (c9_on &&
+      mxa::conv_wg_c9_eligible(R, S, (int)stride[0], (int)stride[1],
+                               (int)pad[0], (int)pad[1], (int)dil[0],
+                               (int)dil[1], H, W, P, Q, C, Kout)) {
+    // seg-merged path: all 9 (r,s) tiles from one staged stream of a
+    // padded copy (see gemm_wg.hip "c9"); x+dy read once instead of 9x
+    auto xp = at::empty({mxa::conv_wg_c9_xp_elems(Nb, H, W, C)},
+                        x.options());
+    mxa::conv_wg_pad_image(bfp(x), bfp_mut(xp), Nb, H, W, C, cur_stream());
+    int ns = mxa::conv2d_wgrad_c9_nslots(Nb, H, W, C, Kout);
+    auto work = at::empty({ns, 9L * C, (long)Kout},
+                          x.options().dtype(at::kFloat));
+    auto dw = at::empty({Kout, 3, 3, C}, x.options());
+    mxa::conv2d_wgrad_c9_bf16(bfp(xp), bfp(dy), work.data_ptr<float>(),
+                              bfp_mut(dw), Nb, H, W, C, Kout, zp(x),
+                              cur_stream());
+    return dw;
+  }
   long mtotal = (long)Nb * P * Q;
   int ns = mxa::conv2d_wgrad_nslots(mtotal, C, Kout, R, S);
   auto work = at::empty({ns, (long)R * S * C, (long)Kout},

@@ -759,6 +759,300 @@ __global__ __launch_bounds__(WT8, RING <= 4 ? 2 : 1) void gemm_wg_ring_core(
 }
 
 // ---------------------------------------------------------------------------
+// Seg-merged 3x3 stride-1 conv wgrad ("c9"): all 9 (r,s) tiles per block
+// ---------------------------------------------------------------------------
+// The per-seg wgrad above re-reads x AND dy once per (r,s) — 9x the HBM
+// traffic — and the round-1 PMC evidence showed the 3x3 wgrads
+// staging-bound (25% MFMA busy).  For the same-size stride-1 case
+// (P==H, Q==W, pad 1, dilation 1 — every ResNet 3x3) there is a clean
+// restructuring: walk the reduce dimension m' over the PADDED output grid
+// (P+2)x(Q+2).  Then
+//   * dy rows are a gather with zeros at the padding positions (read ONCE),
+//   * the im2col of a 1-padded COPY of x becomes a pure LINEAR stream:
+//     xp_linear(m', r, s) = m' + (r-1)*(W+2) + (s-1)  — every (r,s) is a
+//     row-SHIFT of one staged stream, so x is staged ONCE per k-step into
+//     a 4-slot circular LDS window and all 9 accumulators are fed from it.
+//   * every x row that falls outside a seg's valid window pairs with a
+//     zero dy row by construction (margins of xp are zeroed so no NaNs).
+// Traffic drops ~3-6x and MFMA work per staged byte grows 9x; the cost is
+// (P+2)(Q+2)/(P*Q) extra (zero-row) reduce steps and the padded copy.
+//
+// Tile: 64(C) x 128(K); 8 waves as 2(C) x 4(K), wave patch 32x32;
+// acc[9][2][2] f32x4 = 144 acc VGPRs -> 1 block/CU (2 waves/SIMD), LDS
+// 4x8 KiB circular x + 2x16 KiB dy = 64 KiB.  Requires W <= 61 so the
+// 9-seg read window [m'-(W+3), m'+63+(W+3)] spans <= 3 of the 4 slots.
+
+constexpr int C9M = 64, C9N = 128;
+// xp margin (bf16 elements worth of rows*C): staged blocks run one 64-row
+// block past each end and fragment reads reach W+3 rows past that.
+constexpr int C9_MARGIN_ROWS = 256;
+
+DEVINL int c9x_byte(int row, int cbyte) {
+  // [256 rows][64 ch] bf16 image, 128-B rows, chunk16 XOR swizzle
+  return (row & 255) * 128 +
+         ((((cbyte >> 4) ^ ((row & 3) << 1)) & 7) << 4 | (cbyte & 15));
+}
+
+// dy rows over the padded (P+2)x(Q+2) walk: zero at padding positions
+struct DyPadRows {
+  const bf16* dy;
+  const bf16* zpage;
+  int N, P, Q, K;
+  // device state
+  int n_[2], p_[2], q_[2];
+  bool colok;
+  int col, cgbyte;
+
+  DEVINL void init(int col0, long m0, const int* myrow, int cg) {
+    cgbyte = cg * 16;
+    col = col0 + cg * 8;
+    colok = col + 8 <= K;
+    int PQ = (P + 2) * (Q + 2);
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      long m = m0 + myrow[p];
+      n_[p] = (int)(m / PQ);
+      int pq = (int)(m % PQ);
+      p_[p] = pq / (Q + 2);
+      q_[p] = pq % (Q + 2);
+    }
+  }
+  DEVINL void step() {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      q_[i] += WBK;
+      while (q_[i] >= Q + 2) {
+        q_[i] -= Q + 2;
+        p_[i] += 1;
+      }
+      while (p_[i] >= P + 2) {
+        p_[i] -= P + 2;
+        n_[i] += 1;
+      }
+    }
+  }
+  DEVINL const bf16* src(int p) const {
+    const bf16* zp = (const bf16*)((const char*)zpage + cgbyte);
+    if (n_[p] >= N || !colok) return zp;
+    int pp = p_[p] - 1, qq = q_[p] - 1;
+    if (pp < 0 || pp >= P || qq < 0 || qq >= Q) return zp;
+    return dy + (((long)n_[p] * P + pp) * Q + qq) * K + col;
+  }
+};
+
+__global__ __launch_bounds__(WT8, 1) void gemm_wg_c9_core(
+    const bf16* __restrict__ xp, DyPadRows dyld, float* __restrict__ Cout,
+    int C, int Ctot, int Kout, int W, long msteps_total, int nsplit,
+    long strideSlot) {
+  // xp: padded-copy base (margins zeroed); this block's channel tile c0
+  __shared__ __align__(16) bf16 xbuf[4][64 * 64];   // circular, 4x8 KiB
+  __shared__ __align__(16) bf16 dybuf[3][WBK * C9N];  // 3x16 KiB
+
+  const int split = blockIdx.y;
+  const long per = (msteps_total + nsplit - 1) / nsplit;
+  const long k0 = split * per;
+  const long ksteps = min(per, msteps_total - k0);
+  if (ksteps <= 0) return;
+
+  const int ntn = (Kout + C9N - 1) / C9N;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int c0 = (bid / ntn) * C9M, kn0 = (bid % ntn) * C9N;
+  const int t = threadIdx.x;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  // ---- x staging: one contiguous 64-row (x 64-ch) block per step ----
+  // 512 lanes x 16 B = 8 KiB: row = t/8, chunk = t%8, XOR-swizzled source
+  const int xrow = t >> 3;
+  const int xcg = (t & 7) ^ ((xrow & 3) << 1);
+  // global source: xp + (m_abs + xrow)*Ctot + c0 + xcg*8
+  const bf16* xsrc0 = xp + (long)(k0 * WBK + xrow) * Ctot + c0 + xcg * 8;
+  const long xstep = (long)WBK * Ctot;
+
+  auto stage_x = [&](long blk_rel) {  // block index relative to k0 (can be -1)
+    unsigned slot = (unsigned)((k0 + blk_rel) & 3);
+    unsigned lbase = __builtin_amdgcn_readfirstlane(
+        (unsigned)(slot * 64 * 128 + lw * 1024));
+    auto lds = (__attribute__((address_space(3))) unsigned int*)(
+        (__attribute__((address_space(3))) char*)&xbuf[0][0] + lbase);
+    auto g = (const __attribute__((address_space(1))) unsigned int*)(
+        xsrc0 + blk_rel * xstep);
+    __builtin_amdgcn_global_load_lds(g, lds, 16, 0, 0);
+  };
+
+  // ---- dy staging: gather rows, 2 glds passes (16 KiB) ----
+  const int cg = (ll & 15) ^ ((ll >> 4) << 1);
+  int myrow[2];
+#pragma unroll
+  for (int p = 0; p < 2; ++p) myrow[p] = lw * 8 + p * 4 + (ll >> 4);
+  dyld.init(kn0, k0 * WBK, myrow, cg);
+
+  auto stage_dy = [&](int buf) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      unsigned lbase =
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 2 + p) * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&dybuf[buf][0] + lbase);
+      auto g =
+          (const __attribute__((address_space(1))) unsigned int*)dyld.src(p);
+      __builtin_amdgcn_global_load_lds(g, lds, 16, 0, 0);
+    }
+    dyld.step();
+  };
+
+  f32x4 acc[9][2][2];
+#pragma unroll
+  for (int s = 0; s < 9; ++s)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) acc[s][i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // wave patch: 2(C) x 4(K) waves, each 32x32
+  const int wm = (lw & 1) * 32, wn = (lw >> 1) * 32;
+  const int ig = ll & 15, khi = ll >> 4;
+
+  // seg row shifts in m': (r-1)*(W+2) + (s-1)
+  int dseg[9];
+#pragma unroll
+  for (int r = 0; r < 3; ++r)
+#pragma unroll
+    for (int s = 0; s < 3; ++s) dseg[r * 3 + s] = (r - 1) * (W + 2) + (s - 1);
+
+  auto xtrread = [&](int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)&xbuf[0][0] +
+        c9x_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+  auto dytrread = [&](const bf16* img, int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)img + tr_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+
+  auto compute = [&](long ks) {
+    int dbuf = (int)(ks % 3);
+    int mbase = (int)((k0 + ks) & 3) * 64;  // circular base row of this step
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int rloc = kk * 32 + khi * 8 + (ig >> 2);
+      int cb = (ig & 3) * 8;
+      bf16x8 bfr[2];
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int cbyte = (wn + nf * 16) * 2 + cb;
+        bf16x4 lo = dytrread(dybuf[dbuf], rloc, cbyte);
+        bf16x4 hi = dytrread(dybuf[dbuf], rloc + 4, cbyte);
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      // double-buffer the x fragments over segs: seg s+1's tr reads issue
+      // before seg s's MFMAs so the ~50-cycle LDS latency hides under the
+      // 4 MFMAs (the naive order measured 14% MFMA-busy / 23% issue-stall)
+      bf16x8 af[2][2];
+      auto lda = [&](int s, int buf) {
+        int xr = mbase + rloc + dseg[s];
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          int cbyte = (wm + mf * 16) * 2 + cb;
+          bf16x4 lo = xtrread(xr, cbyte);
+          bf16x4 hi = xtrread(xr + 4, cbyte);
+          af[buf][mf] =
+              __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+        }
+      };
+      lda(0, 0);
+#pragma unroll
+      for (int s = 0; s < 9; ++s) {
+        if (s + 1 < 9) lda(s + 1, (s + 1) & 1);
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc[s][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[s & 1][mf], bfr[nf], acc[s][mf][nf], 0, 0, 0);
+      }
+    }
+  };
+
+  // prologue: x blocks -1, 0, 1 (margins make -1 and past-end valid), dy 0
+  stage_x(-1);
+  stage_x(0);
+  stage_x(1);
+  stage_dy(0);
+  if (ksteps > 1) stage_dy(1);
+  __syncthreads();
+  for (long ks = 0; ks < ksteps; ++ks) {
+    stage_x(ks + 2);
+    if (ks + 2 < ksteps) stage_dy((int)((ks + 2) % 3));
+    compute(ks);
+    __syncthreads();
+  }
+
+  // epilogue: 9 seg tiles into this split's slot [seg][C][K]
+  float* cslot = Cout + (long)split * strideSlot;
+#pragma unroll
+  for (int s = 0; s < 9; ++s) {
+    float* cseg = cslot + (long)s * C * Kout;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int gcol = kn0 + wn + nf * 16 + (ll & 15);
+        if (gcol >= Kout) continue;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int grow = c0 + wm + mf * 16 + (ll >> 4) * 4 + i;
+          if (grow >= C) continue;
+          cseg[(long)grow * Kout + gcol] = acc[s][mf][nf][i];
+        }
+      }
+    }
+  }
+}
+
+// padded copy: xp[(n, hp, wp, c)] over (H+2)x(W+2) with 1-px zero ring;
+// front/back margins of C9_MARGIN_ROWS*C elements are zeroed separately.
+__global__ __launch_bounds__(256) void pad_image_kernel(
+    const bf16* __restrict__ x, bf16* __restrict__ xp, int N, int H, int W,
+    int C, long total8) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total8) return;
+  long e = i * 8;  // 8 channels per thread; C % 8 == 0
+  int c = (int)(e % C);
+  long pix = e / C;
+  int wp = (int)(pix % (W + 2));
+  long t2 = pix / (W + 2);
+  int hp = (int)(t2 % (H + 2));
+  int n = (int)(t2 / (H + 2));
+  bf16x8 v = {};
+  if (hp >= 1 && hp <= H && wp >= 1 && wp <= W)
+    v = *(const bf16x8*)(x + (((long)n * H + hp - 1) * W + wp - 1) * C + c);
+  *(bf16x8*)(xp + e) = v;
+}
+
+__global__ __launch_bounds__(256) void zero_bf16_kernel(bf16* p, long n8) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n8) ((bf16x8*)p)[i] = bf16x8{};
+}
+
+bool conv_wg_c9_eligible(int R, int S, int sh, int sw, int ph, int pw,
+                         int dh, int dw, int H, int W, int P, int Q, int C,
+                         int K) {
+  // W >= 24: the padded walk costs (H+2)(W+2)/(HW) extra reduce steps —
+  // measured (profiles, r2 A/B at b256): +14% at 56x56, tie at 28x28,
+  // -15% at 14x14 (1.31x pad overhead) — so small images keep the
+  // per-seg path.
+  return R == 3 && S == 3 && sh == 1 && sw == 1 && ph == 1 && pw == 1 &&
+         dh == 1 && dw == 1 && P == H && Q == W && W <= 61 && W >= 24 &&
+         C % 8 == 0 && K % 8 == 0;
+}
+
+long conv_wg_c9_xp_elems(int N, int H, int W, int C) {
+  return ((long)N * (H + 2) * (W + 2) + 2L * C9_MARGIN_ROWS) * C;
+}
+
+// ---------------------------------------------------------------------------
 // Split-slot fold epilogue
 // ---------------------------------------------------------------------------
 // Split-K blocks write per-split f32 slot buffers with plain stores (no
@@ -784,15 +1078,23 @@ __global__ __launch_bounds__(256) void wg_fold_partial(
   int cnt = min(WG_FOLD_GROUP, nslots - (int)blockIdx.y * WG_FOLD_GROUP);
   const float* p =
       work + (long)blockIdx.y * WG_FOLD_GROUP * slotStride + base;
+  // 4 independent accumulators keep >=4 loads in flight per lane — the
+  // 2-deep version measured latency-bound (95% wave-parked)
   f32x4 s0{0.f, 0.f, 0.f, 0.f}, s1{0.f, 0.f, 0.f, 0.f};
+  f32x4 s2{0.f, 0.f, 0.f, 0.f}, s3{0.f, 0.f, 0.f, 0.f};
   int sl = 0;
-  for (; sl + 2 <= cnt; sl += 2) {
+  for (; sl + 4 <= cnt; sl += 4) {
     s0 += *(const f32x4*)p;
     s1 += *(const f32x4*)(p + slotStride);
-    p += 2 * slotStride;
+    s2 += *(const f32x4*)(p + 2 * slotStride);
+    s3 += *(const f32x4*)(p + 3 * slotStride);
+    p += 4 * slotStride;
   }
-  if (sl < cnt) s0 += *(const f32x4*)p;
-  s0 += s1;
+  for (; sl < cnt; ++sl) {
+    s0 += *(const f32x4*)p;
+    p += slotStride;
+  }
+  s0 += s1 + s2 + s3;
   *(f32x4*)(outslots + (long)blockIdx.y * slotStride + base) = s0;
 }
 
@@ -802,15 +1104,21 @@ __global__ __launch_bounds__(256) void wg_fold_flat(
   long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (base >= n) return;
   f32x4 s0{0.f, 0.f, 0.f, 0.f}, s1{0.f, 0.f, 0.f, 0.f};
+  f32x4 s2{0.f, 0.f, 0.f, 0.f}, s3{0.f, 0.f, 0.f, 0.f};
   const float* p = work + base;
   int sl = 0;
-  for (; sl + 2 <= nslots; sl += 2) {
+  for (; sl + 4 <= nslots; sl += 4) {
     s0 += *(const f32x4*)p;
     s1 += *(const f32x4*)(p + slotStride);
-    p += 2 * slotStride;
+    s2 += *(const f32x4*)(p + 2 * slotStride);
+    s3 += *(const f32x4*)(p + 3 * slotStride);
+    p += 4 * slotStride;
   }
-  if (sl < nslots) s0 += *(const f32x4*)p;
-  s0 += s1;
+  for (; sl < nslots; ++sl) {
+    s0 += *(const f32x4*)p;
+    p += slotStride;
+  }
+  s0 += s1 + s2 + s3;
   bf16x4 o{f2bf(s0.x), f2bf(s0.y), f2bf(s0.z), f2bf(s0.w)};
   *(bf16x4*)(out + base) = o;
 }
@@ -1032,6 +1340,51 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* work, bf16* out,
   }
   if (!direct)
     wg_fold(work, slotStride, pl.used, Mo, No, out, transpose, stream);
+}
+
+void conv_wg_pad_image(const bf16* x, bf16* xp_alloc, int N, int H, int W,
+                       int C, hipStream_t stream) {
+  long margin8 = (long)C9_MARGIN_ROWS * C / 8;
+  long body = (long)N * (H + 2) * (W + 2) * C;
+  zero_bf16_kernel<<<dim3((unsigned)((margin8 + 255) / 256)), 256, 0,
+                     stream>>>(xp_alloc, margin8);
+  zero_bf16_kernel<<<dim3((unsigned)((margin8 + 255) / 256)), 256, 0,
+                     stream>>>(xp_alloc + (long)C9_MARGIN_ROWS * C + body,
+                               margin8);
+  long total8 = body / 8;
+  pad_image_kernel<<<dim3((unsigned)((total8 + 255) / 256)), 256, 0,
+                     stream>>>(x, xp_alloc + (long)C9_MARGIN_ROWS * C, N, H,
+                               W, C, total8);
+}
+
+static WgPlanT c9_plan(int N, int H, int W, int C, int Kout) {
+  long m2 = (long)N * (H + 2) * (W + 2);
+  long bxy = (long)cdiv_h(C, C9M) * cdiv_h(Kout, C9N);
+  WgPlanT p;
+  p.ksteps = (int)((m2 + WBK - 1) / WBK);
+  p.hsteps = p.ksteps;
+  p.nsplit = pick_split(bxy, p.ksteps);
+  int per = (p.ksteps + p.nsplit - 1) / p.nsplit;
+  p.used = (p.ksteps + per - 1) / per;
+  return p;
+}
+
+int conv2d_wgrad_c9_nslots(int N, int H, int W, int C, int Kout) {
+  return wg_alloc_slots(c9_plan(N, H, W, C, Kout).used);
+}
+
+void conv2d_wgrad_c9_bf16(const bf16* xp_alloc, const bf16* dy, float* work,
+                          bf16* out_krsc, int N, int H, int W, int C,
+                          int Kout, const bf16* zpage, hipStream_t stream) {
+  WgPlanT pl = c9_plan(N, H, W, C, Kout);
+  long bxy = (long)cdiv_h(C, C9M) * cdiv_h(Kout, C9N);
+  dim3 grid((unsigned)bxy, pl.nsplit, 1);
+  DyPadRows dyld{dy, zpage, N, H, W, Kout};
+  long strideSlot = 9L * C * Kout;
+  gemm_wg_c9_core<<<grid, WT8, 0, stream>>>(
+      xp_alloc + (long)C9_MARGIN_ROWS * C, dyld, work, C, C, Kout, W,
+      pl.ksteps, pl.nsplit, strideSlot);
+  wg_fold(work, strideSlot, pl.used, 9L * C, Kout, out_krsc, true, stream);
 }
 
 int conv2d_wgrad_nslots(long mtotal, int C, int Kout, int R, int S) {
