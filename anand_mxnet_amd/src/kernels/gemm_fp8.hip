@@ -568,14 +568,229 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx256_core(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 256² 8-phase MX kernel: the tn8 half-tile glds ring carried over to the
+// double-rate scaled-fp8 path.  The 2-phase gemm_mx256_core above drains
+// vmcnt(0) + __syncthreads every K-step (measured 762 TF = 16% of the
+// 4661 TF MX µbench ceiling); here three half-tiles stay in flight across
+// counted s_waitcnt vmcnt + raw barriers.  e8m0 scales for the whole
+// K-tile (256 A rows + 256 B rows x 4 k-blocks = 2 KiB) are staged to LDS
+// by ONE dword-per-lane glds issued with half A0 — so each tile is 9 VMEM
+// ops [A0+sc(3), B0(2), A1(2), B1(2)] and the steady-state waits become
+// vmcnt(7) at phases 0/1/2 (tail drains 4/2/0), the same invariant
+// accounting that yields tn8's vmcnt(6) with 2-op halves.
+// ---------------------------------------------------------------------------
+
+struct Mx8Loader {
+  const u8* base;
+  const u8* zpage;
+  int rows;
+  long ld;
+  // device state: 4 row-slots (half = p>>1, glds pass j = p&1)
+  const u8* rp[4];
+  int cg[4];
+
+  DEVINL void init(int row0, int t) {
+    int w = t / WAVE, l = t % WAVE;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      int local = (p >> 1) * 128 + w * 16 + (p & 1) * 8 + (l >> 3);
+      int r = row0 + local;
+      rp[p] = (r < rows) ? base + (long)r * ld : nullptr;
+      cg[p] = (l & 7) ^ (local & 7);  // source chunk for f8_lds_off
+    }
+  }
+  DEVINL const u8* src(int p, long koff) const {
+    return rp[p] ? rp[p] + koff + cg[p] * 16 : zpage + cg[p] * 16;
+  }
+};
+
+__global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
+    Mx8Loader aload, Mx8Loader bload, const u8* __restrict__ Asc,
+    const u8* __restrict__ Bsc, long lsa, long lsb, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, int ksteps,
+    int relu) {
+  // [tilebuf][op][half][128 rows x 128 B]  (128 KiB) + staged scales
+  __shared__ __align__(16) u8 smem[2][2][2][128 * 128];
+  __shared__ __align__(16) u8 scbuf[2][2048];  // [buf][A 1K | B 1K]
+
+  const int ntn = (N + 255) / 256;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * 256, bn = (bid % ntn) * 256;
+  const int t = threadIdx.x;
+  const int w = t / WAVE, l = t % WAVE;
+
+  aload.init(bm, t);
+  bload.init(bn, t);
+
+  // per-lane scale source: lanes 0-255 stage A rows, 256-511 B rows
+  const bool sc_is_a = t < 256;
+  const int sc_row = sc_is_a ? t : t - 256;
+  const int sc_g = sc_is_a ? bm + sc_row : bn + sc_row;
+  const u8* sc_base = sc_is_a
+                          ? (sc_g < M ? Asc + (long)sc_g * lsa : aload.zpage)
+                          : (sc_g < N ? Bsc + (long)sc_g * lsb : aload.zpage);
+  const long sc_ld = (sc_is_a ? (sc_g < M) : (sc_g < N)) ? 4 : 0;
+
+  auto issue_group = [&](int issue_idx, int tile) {
+    int op = issue_idx & 1;  // A0, B0, A1, B1
+    int half = issue_idx >> 1;
+    long koff = (long)tile * F8_BK;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[tile & 1][op][half]
+              [0] + seg);
+      const u8* gp = op ? bload.src(p, koff) : aload.src(p, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+    if (issue_idx == 0) {  // the tile's scale stage rides with half A0
+      unsigned seg = __builtin_amdgcn_readfirstlane((unsigned)(w * 256));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&scbuf[tile & 1][0] +
+          seg);
+      const u8* gp = sc_base + (sc_ld ? (long)tile * 4 : 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 4,
+          0, 0);
+    }
+  };
+
+  f32x4 acc[4][8];  // [quadrant][mf*2+nf]
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[q][i] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int lrow = l & 15, lk = l >> 4;
+  const int wmq = (w >> 2) * 64;  // within-quadrant M offset (2 M-waves)
+  const int wnq = (w & 3) * 32;   // within-quadrant N offset (4 N-waves)
+
+  auto vm_wait = [&](int n) {
+    if (n == 7)
+      asm volatile("s_waitcnt vmcnt(7)" ::: "memory");
+    else if (n == 4)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (n == 2)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else if (n == 0)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  };
+
+  auto phase = [&](int tile, int qa, int qb, int q, int waitn) {
+    vm_wait(waitn);
+    __builtin_amdgcn_s_barrier();
+    const u8* As = smem[tile & 1][0][qa];
+    const u8* Bs = smem[tile & 1][1][qb];
+    const u8* sc = scbuf[tile & 1];
+    i32x8 af[4];
+    int sa[4];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      int row = wmq + mf * 16 + lrow;
+      i32x4 lo = *(const i32x4*)&As[f8_lds_off(row, lk)];
+      i32x4 hi = *(const i32x4*)&As[f8_lds_off(row, 4 + lk)];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        af[mf][j] = lo[j];
+        af[mf][4 + j] = hi[j];
+      }
+      sa[mf] = sc[(qa * 128 + row) * 4 + lk];
+    }
+    i32x8 bfr[2];
+    int sb[2];
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      int row = wnq + nf * 16 + lrow;
+      i32x4 lo = *(const i32x4*)&Bs[f8_lds_off(row, lk)];
+      i32x4 hi = *(const i32x4*)&Bs[f8_lds_off(row, 4 + lk)];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        bfr[nf][j] = lo[j];
+        bfr[nf][4 + j] = hi[j];
+      }
+      sb[nf] = sc[1024 + (qb * 128 + row) * 4 + lk];
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        acc[q][mf * 2 + nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af[mf], bfr[nf], acc[q][mf * 2 + nf], 0, 0, 0, sa[mf], 0,
+            sb[nf]);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  };
+
+  // prologue: tile 0's 4 groups (9 VMEM ops)
+#pragma unroll
+  for (int h = 0; h < 4; ++h) issue_group(h, 0);
+
+  for (int tile = 0; tile < ksteps; ++tile) {
+    int nxt = tile + 1;
+    bool pre = nxt < ksteps;
+    if (pre) issue_group(0, nxt);
+    phase(tile, 0, 0, 0, pre ? 7 : 4);
+    if (pre) issue_group(1, nxt);
+    phase(tile, 1, 0, 1, pre ? 7 : 2);
+    if (pre) issue_group(2, nxt);
+    phase(tile, 1, 1, 2, pre ? 7 : 0);
+    if (pre) issue_group(3, nxt);
+    phase(tile, 0, 1, 3, -1);
+  }
+
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    int qa = (q == 1 || q == 2) ? 1 : 0;
+    int qb = (q >= 2) ? 1 : 0;
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int gcol = bn + qb * 128 + wnq + nf * 16 + (l & 15);
+        if (gcol >= N) continue;
+        float bv = bias ? bias[gcol] : 0.f;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int grow = bm + qa * 128 + wmq + mf * 16 + (l >> 4) * 4 + i;
+          if (grow >= M) continue;
+          float v = acc[q][mf * 2 + nf][i] + bv;
+          if (relu) v = v > 0.f ? v : 0.f;
+          Cout[(long)grow * ldc + gcol] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
 void gemm_mx_bf16(const u8* A, const u8* Asc, const u8* B, const u8* Bsc,
                   bf16* C, const float* bias, int M, int N, int K, long ldc,
                   int relu, const u8* zpage, hipStream_t stream) {
   MxLoader a{A, zpage, M, K};
   MxLoader b{B, zpage, N, K};
   long blocks256 = (long)cdiv_h(M, 256) * cdiv_h(N, 256);
-  if (M >= 256 && N >= 256 && blocks256 >= 200) {
+  static const bool mx8 = [] {
+    const char* e = getenv("MXA_MX8");
+    return !e || e[0] != '0';  // default on; =0 -> 2-phase 256 kernel
+  }();
+  // mx8 threshold 120: at 1 block/CU even a half-filled chip with deep K
+  // beats the 128-tile 2-phase kernel (8192x1024x4096 measured 550 there)
+  if (M >= 256 && N >= 256 && blocks256 >= (mx8 ? 120 : 200)) {
     dim3 grid((unsigned)blocks256);
+    if (mx8) {
+      Mx8Loader a8{A, zpage, M, K};
+      Mx8Loader b8{B, zpage, N, K};
+      gemm_mx8_core<<<grid, MX_THREADS, 0, stream>>>(
+          a8, b8, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, K / F8_BK,
+          relu);
+      return;
+    }
     gemm_mx256_core<<<grid, MX_THREADS, 0, stream>>>(
         a, b, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, K / F8_BK,
         relu);

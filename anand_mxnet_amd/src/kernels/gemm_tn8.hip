@@ -54,7 +54,7 @@ struct Plain8Loader {
 };
 
 template <bool RELU, bool BIAS>
-__global__ __launch_bounds__(T8_THREADS, 2) void gemm_tn8_core(
+__global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8_core(
     Plain8Loader aload, Plain8Loader bload, bf16* __restrict__ Cout,
     const float* __restrict__ bias, int M, int N, long ldc, long strideC,
     int ksteps, long kstride) {
