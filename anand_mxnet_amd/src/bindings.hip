@@ -138,10 +138,15 @@ void gemm_tn_fp8(const unsigned char*, const unsigned char*, bf16*,
                  const unsigned char*, hipStream_t);
 void gemm_mx_bf16(const unsigned char*, const unsigned char*,
                   const unsigned char*, const unsigned char*, bf16*,
-                  const float*, int, int, int, long, int,
+                  const float*, int, int, int, long, int, int, int, float*,
                   const unsigned char*, hipStream_t);
+int gemm_mx_nslots(int, int, int, int, int);
 void mx_cast_bf16(const bf16*, unsigned char*, unsigned char*, long,
                   hipStream_t);
+void mx_cast_bf8(const bf16*, unsigned char*, unsigned char*, long,
+                 hipStream_t);
+void mx_cast_t(const bf16*, unsigned char*, unsigned char*, int, int, bool,
+               hipStream_t);
 void cast_to_fp8(const bf16*, unsigned char*, long, const float*,
                  hipStream_t);
 // optim.hip
@@ -326,8 +331,9 @@ at::Tensor gemm_tn_fp8(const at::Tensor& A8, const at::Tensor& B8,
   return out;
 }
 
-std::vector<at::Tensor> mx_cast(const at::Tensor& x) {
-  // bf16 [.., K] -> (fp8 e4m3 payload u8, e8m0 block scales u8 [.., K/32])
+std::vector<at::Tensor> mx_cast(const at::Tensor& x, bool e5) {
+  // bf16 [.., K] -> (fp8 payload u8, e8m0 block scales u8 [.., K/32]);
+  // e5 -> e5m2 (gradients), else e4m3
   TORCH_CHECK(x.is_contiguous() && x.size(-1) % 32 == 0,
               "mx_cast: last dim must be %32");
   long nblk = x.numel() / 32;
@@ -335,14 +341,34 @@ std::vector<at::Tensor> mx_cast(const at::Tensor& x) {
   auto shp = x.sizes().vec();
   shp.back() /= 32;
   auto sc = at::empty(shp, x.options().dtype(at::kByte));
-  mxa::mx_cast_bf16(bfp(x), q.data_ptr<unsigned char>(),
-                    sc.data_ptr<unsigned char>(), nblk, cur_stream());
+  if (e5)
+    mxa::mx_cast_bf8(bfp(x), q.data_ptr<unsigned char>(),
+                     sc.data_ptr<unsigned char>(), nblk, cur_stream());
+  else
+    mxa::mx_cast_bf16(bfp(x), q.data_ptr<unsigned char>(),
+                      sc.data_ptr<unsigned char>(), nblk, cur_stream());
+  return {q, sc};
+}
+
+std::vector<at::Tensor> mx_cast_t(const at::Tensor& x, bool e5) {
+  // bf16 [R, C] -> transposed MX: (Q u8 [C, R], SC u8 [C, R/32]),
+  // quantization blocks of 32 along R — grads/operands for the backward
+  // GEMMs expressed as plain TN MX GEMMs over transposed operands
+  TORCH_CHECK(x.is_contiguous() && x.dim() == 2 && x.size(0) % 32 == 0,
+              "mx_cast_t: [R, C] with R % 32 == 0");
+  int R = (int)x.size(0), C = (int)x.size(1);
+  auto q = at::empty({(long)C, (long)R}, x.options().dtype(at::kByte));
+  auto sc = at::empty({(long)C, (long)R / 32},
+                      x.options().dtype(at::kByte));
+  mxa::mx_cast_t(bfp(x), q.data_ptr<unsigned char>(),
+                 sc.data_ptr<unsigned char>(), R, C, e5, cur_stream());
   return {q, sc};
 }
 
 at::Tensor gemm_mx(const at::Tensor& A8, const at::Tensor& Asc,
                    const at::Tensor& B8, const at::Tensor& Bsc,
-                   const c10::optional<at::Tensor>& bias, bool relu) {
+                   const c10::optional<at::Tensor>& bias, bool relu,
+                   long afmt, long bfmt) {
   // C[M][N] bf16 = MX-fp8 A[M,K] @ B[N,K]^T (both k-contiguous)
   TORCH_CHECK(A8.dim() == 2 && B8.dim() == 2);
   long M = A8.size(0), K = A8.size(1), N = B8.size(0);
@@ -350,13 +376,22 @@ at::Tensor gemm_mx(const at::Tensor& A8, const at::Tensor& Asc,
   auto opts = at::TensorOptions().device(A8.device()).dtype(at::kBFloat16);
   auto out = at::empty({M, N}, opts);
   auto zprobe = at::empty({0}, opts);
+  int ns = mxa::gemm_mx_nslots((int)M, (int)N, (int)K,
+                               bias.has_value() ? 1 : 0, relu ? 1 : 0);
+  at::Tensor work;
+  float* wp = nullptr;
+  if (ns > 0) {
+    work = at::empty({(long)ns, M, N}, opts.dtype(at::kFloat));
+    wp = work.data_ptr<float>();
+  }
   mxa::gemm_mx_bf16(A8.data_ptr<unsigned char>(),
                     Asc.data_ptr<unsigned char>(),
                     B8.data_ptr<unsigned char>(),
                     Bsc.data_ptr<unsigned char>(), bfp_mut(out),
                     bias.has_value() ? bias->data_ptr<float>() : nullptr,
-                    (int)M, (int)N, (int)K, N, relu,
-                    (const unsigned char*)zp(zprobe), cur_stream());
+                    (int)M, (int)N, (int)K, N, relu, (int)afmt,
+                    (int)bfmt, wp, (const unsigned char*)zp(zprobe),
+                    cur_stream());
   return out;
 }
 
@@ -1126,10 +1161,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("transpose") = false);
   m.def("gemm_tn_fp8", &gemm_tn_fp8);
   m.def("cast_to_fp8", &cast_to_fp8);
-  m.def("mx_cast", &mx_cast);
+  m.def("mx_cast", &mx_cast, py::arg("x"), py::arg("e5") = false);
+  m.def("mx_cast_t", &mx_cast_t, py::arg("x"), py::arg("e5") = false);
   m.def("gemm_mx", &gemm_mx, py::arg("A8"), py::arg("Asc"), py::arg("B8"),
         py::arg("Bsc"), py::arg("bias") = py::none(),
-        py::arg("relu") = false);
+        py::arg("relu") = false, py::arg("afmt") = 0, py::arg("bfmt") = 0);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_fwd_stats", &conv2d_fwd_stats,
         "conv2d_fwd + fused per-channel BN sums: {y[,stats]}");

@@ -208,6 +208,9 @@ void cast_to_fp8(const bf16* x, u8* y, long n, const float* scale,
 
 typedef __attribute__((ext_vector_type(8))) int i32x8;
 
+// split-K slot fold (gemm_wg.hip)
+void wg_fold(const float*, long, int, long, long, bf16*, bool, hipStream_t);
+
 constexpr int MX_THREADS = 512;
 
 DEVINL int mx_stage_row(int p, int t) {
@@ -605,11 +608,13 @@ struct Mx8Loader {
   }
 };
 
+template <int AFMT = 0, int BFMT = 0>  // 0 = e4m3 (fp8), 1 = e5m2 (bf8)
 __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
     Mx8Loader aload, Mx8Loader bload, const u8* __restrict__ Asc,
     const u8* __restrict__ Bsc, long lsa, long lsb, bf16* __restrict__ Cout,
-    const float* __restrict__ bias, int M, int N, long ldc, int ksteps,
-    int relu) {
+    const float* __restrict__ bias, int M, int N, long ldc, int ksteps_total,
+    int relu, int nsplit = 1, long strideSlot = 0,
+    float* __restrict__ work = nullptr) {
   // [tilebuf][op][half][128 rows x 128 B]  (128 KiB) + staged scales
   __shared__ __align__(16) u8 smem[2][2][2][128 * 128];
   __shared__ __align__(16) u8 scbuf[2][2048];  // [buf][A 1K | B 1K]
@@ -619,6 +624,14 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
   const int bm = (bid / ntn) * 256, bn = (bid % ntn) * 256;
   const int t = threadIdx.x;
   const int w = t / WAVE, l = t % WAVE;
+
+  // split-K (deep-reduce small-output grad GEMMs): each split owns a
+  // private f32 slot; wg_fold sums them afterwards
+  const int split = blockIdx.y;
+  const int per = (ksteps_total + nsplit - 1) / nsplit;
+  const int kstep0 = split * per;
+  const int ksteps = min(per, ksteps_total - kstep0);
+  if (ksteps <= 0) return;
 
   aload.init(bm, t);
   bload.init(bn, t);
@@ -635,7 +648,7 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
   auto issue_group = [&](int issue_idx, int tile) {
     int op = issue_idx & 1;  // A0, B0, A1, B1
     int half = issue_idx >> 1;
-    long koff = (long)tile * F8_BK;
+    long koff = (long)(kstep0 + tile) * F8_BK;
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       int p = half * 2 + j;
@@ -654,7 +667,7 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
       auto lds = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&scbuf[tile & 1][0] +
           seg);
-      const u8* gp = sc_base + (sc_ld ? (long)tile * 4 : 0);
+      const u8* gp = sc_base + (sc_ld ? (long)(kstep0 + tile) * 4 : 0);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gp, lds, 4,
           0, 0);
@@ -722,7 +735,7 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
 #pragma unroll
       for (int nf = 0; nf < 2; ++nf)
         acc[q][mf * 2 + nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-            af[mf], bfr[nf], acc[q][mf * 2 + nf], 0, 0, 0, sa[mf], 0,
+            af[mf], bfr[nf], acc[q][mf * 2 + nf], AFMT, BFMT, 0, sa[mf], 0,
             sb[nf]);
     __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
@@ -745,6 +758,7 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
     phase(tile, 0, 1, 3, -1);
   }
 
+  float* wslot = work ? work + (long)split * strideSlot : nullptr;
 #pragma unroll
   for (int q = 0; q < 4; ++q) {
     int qa = (q == 1 || q == 2) ? 1 : 0;
@@ -762,16 +776,37 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
           if (grow >= M) continue;
           float v = acc[q][mf * 2 + nf][i] + bv;
           if (relu) v = v > 0.f ? v : 0.f;
-          Cout[(long)grow * ldc + gcol] = f2bf(v);
+          if (wslot)
+            wslot[(long)grow * ldc + gcol] = v;
+          else
+            Cout[(long)grow * ldc + gcol] = f2bf(v);
         }
       }
     }
   }
 }
 
+static int mx8_pick_split(long blocks, int ksteps, bool can_split) {
+  if (!can_split) return 1;
+  int ns = 1;
+  // fill ~256 CUs (1 block/CU kernel), keep >= 8 K-tiles per split
+  while (blocks * ns < 240 && ns * 2 <= ksteps / 8 && ns < 64) ns *= 2;
+  return ns;
+}
+
+int gemm_mx_nslots(int M, int N, int K, int has_bias, int relu) {
+  long blocks = (long)cdiv_h(M, 256) * cdiv_h(N, 256);
+  int ns = mx8_pick_split(blocks, K / F8_BK, !has_bias && !relu);
+  if (ns == 1) return 0;
+  int per = (K / F8_BK + ns - 1) / ns;
+  int used = (K / F8_BK + per - 1) / per;
+  return used + (used > 16 ? (used + 15) / 16 : 0);  // wg_fold headroom
+}
+
 void gemm_mx_bf16(const u8* A, const u8* Asc, const u8* B, const u8* Bsc,
                   bf16* C, const float* bias, int M, int N, int K, long ldc,
-                  int relu, const u8* zpage, hipStream_t stream) {
+                  int relu, int afmt, int bfmt, float* work,
+                  const u8* zpage, hipStream_t stream) {
   MxLoader a{A, zpage, M, K};
   MxLoader b{B, zpage, N, K};
   long blocks256 = (long)cdiv_h(M, 256) * cdiv_h(N, 256);
@@ -780,17 +815,38 @@ void gemm_mx_bf16(const u8* A, const u8* Asc, const u8* B, const u8* Bsc,
     return !e || e[0] != '0';  // default on; =0 -> 2-phase 256 kernel
   }();
   // mx8 threshold 120: at 1 block/CU even a half-filled chip with deep K
-  // beats the 128-tile 2-phase kernel (8192x1024x4096 measured 550 there)
-  if (M >= 256 && N >= 256 && blocks256 >= (mx8 ? 120 : 200)) {
-    dim3 grid((unsigned)blocks256);
-    if (mx8) {
-      Mx8Loader a8{A, zpage, M, K};
-      Mx8Loader b8{B, zpage, N, K};
-      gemm_mx8_core<<<grid, MX_THREADS, 0, stream>>>(
-          a8, b8, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, K / F8_BK,
-          relu);
-      return;
+  // beats the 128-tile 2-phase kernel (8192x1024x4096 measured 550 there).
+  // Non-default formats (e5m2 grads) ALWAYS go through mx8 — the other
+  // kernels decode e4m3 only.
+  bool use8 = afmt || bfmt ||
+              (mx8 && M >= 256 && N >= 256 && blocks256 >= 120);
+  if (use8) {
+    int ksteps = K / F8_BK;
+    int ns = work ? mx8_pick_split(blocks256, ksteps,
+                                   bias == nullptr && !relu)
+                  : 1;
+    dim3 grid((unsigned)blocks256, ns, 1);
+    Mx8Loader a8{A, zpage, M, K};
+    Mx8Loader b8{B, zpage, N, K};
+    long slotStride = (long)M * ldc;
+    float* wk = ns > 1 ? work : nullptr;
+    if (afmt == 1)
+      gemm_mx8_core<1, 0><<<grid, MX_THREADS, 0, stream>>>(
+          a8, b8, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, ksteps,
+          relu, ns, slotStride, wk);
+    else
+      gemm_mx8_core<0, 0><<<grid, MX_THREADS, 0, stream>>>(
+          a8, b8, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, ksteps,
+          relu, ns, slotStride, wk);
+    if (ns > 1) {
+      int per = (ksteps + ns - 1) / ns;
+      int used = (ksteps + per - 1) / per;
+      wg_fold(work, slotStride, used, M, ldc, C, false, stream);
     }
+    return;
+  }
+  if (M >= 256 && N >= 256 && blocks256 >= 200) {
+    dim3 grid((unsigned)blocks256);
     gemm_mx256_core<<<grid, MX_THREADS, 0, stream>>>(
         a, b, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, K / F8_BK,
         relu);
@@ -799,6 +855,108 @@ void gemm_mx_bf16(const u8* A, const u8* Asc, const u8* B, const u8* Bsc,
   dim3 grid(cdiv_h(M, F8_BM) * cdiv_h(N, F8_BN));
   gemm_mx_core<<<grid, MX_THREADS, 0, stream>>>(
       a, b, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, K / F8_BK, relu);
+}
+
+// Row-major MX cast, e5m2 variant (gradients): amax/2^s <= 57344
+__global__ __launch_bounds__(256) void mx_cast_bf8_kernel(
+    const bf16* __restrict__ x, u8* __restrict__ q, u8* __restrict__ sc,
+    long nblk) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nblk;
+       i += (long)gridDim.x * blockDim.x) {
+    const bf16* xp = x + i * 32;
+    float amax = 0.f;
+    float vals[32];
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      bf16x8 b = *(const bf16x8*)&xp[v * 8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(b[j]);
+        vals[v * 8 + j] = f;
+        amax = fmaxf(amax, fabsf(f));
+      }
+    }
+    int s_exp = 0;
+    if (amax > 0.f) {
+      int e;
+      float m = frexpf(amax, &e);
+      s_exp = e - (m <= 0.875f ? 16 : 15);  // amax/2^s <= 57344 (e5m2 max)
+      if (s_exp < -127) s_exp = -127;
+      if (s_exp > 127) s_exp = 127;
+    }
+    sc[i] = (u8)(s_exp + 127);
+    float inv = ldexpf(1.f, -s_exp);
+    u8 out[32];
+#pragma unroll
+    for (int j = 0; j < 32; ++j)
+      out[j] = __hip_fp8_e5m2(vals[j] * inv).__x;
+#pragma unroll
+    for (int v = 0; v < 4; ++v)
+      *(u64*)&q[i * 32 + v * 8] = *(const u64*)&out[v * 8];
+  }
+}
+
+// Transposing MX cast: X [R][C] bf16 -> Q [C][R] u8 + SC [C][R/32],
+// quantization blocks of 32 along R (the reduce dim of the grad GEMMs:
+// dW = mx8(dyT, xT) and dx = mx8(dy, WT) then reuse the plain TN kernel
+// instead of needing WG/NN-form MX kernels).  R % 32 == 0.
+template <bool E5>
+__global__ __launch_bounds__(256) void mx_cast_t_kernel(
+    const bf16* __restrict__ x, u8* __restrict__ q, u8* __restrict__ sc,
+    int R, int C) {
+  __shared__ bf16 tile[32][257];
+  int r0 = blockIdx.y * 32;
+  int c0 = blockIdx.x * 256;
+  int t = threadIdx.x;
+#pragma unroll 4
+  for (int j = 0; j < 32; ++j) {
+    int c = c0 + t;
+    tile[j][t] = (c < C) ? x[(long)(r0 + j) * C + c] : (bf16)0.f;
+  }
+  __syncthreads();
+  int c = c0 + t;
+  if (c >= C) return;
+  float vals[32];
+  float amax = 0.f;
+#pragma unroll
+  for (int j = 0; j < 32; ++j) {
+    float f = bf2f(tile[j][t]);
+    vals[j] = f;
+    amax = fmaxf(amax, fabsf(f));
+  }
+  int s_exp = 0;
+  if (amax > 0.f) {
+    int e;
+    float m = frexpf(amax, &e);
+    s_exp = e - (E5 ? (m <= 0.875f ? 16 : 15) : (m <= 0.875f ? 9 : 8));
+    if (s_exp < -127) s_exp = -127;
+    if (s_exp > 127) s_exp = 127;
+  }
+  sc[(long)c * (R / 32) + blockIdx.y] = (u8)(s_exp + 127);
+  float inv = ldexpf(1.f, -s_exp);
+  u8 out[32];
+#pragma unroll
+  for (int j = 0; j < 32; ++j)
+    out[j] = E5 ? __hip_fp8_e5m2(vals[j] * inv).__x
+                : __hip_fp8_e4m3(vals[j] * inv).__x;
+#pragma unroll
+  for (int v = 0; v < 4; ++v)
+    *(u64*)&q[(long)c * R + r0 + v * 8] = *(const u64*)&out[v * 8];
+}
+
+void mx_cast_bf8(const bf16* x, u8* q, u8* sc, long nblk,
+                 hipStream_t stream) {
+  mx_cast_bf8_kernel<<<ew_grid(nblk, 256, 4096), 256, 0, stream>>>(x, q, sc,
+                                                                   nblk);
+}
+
+void mx_cast_t(const bf16* x, u8* q, u8* sc, int R, int C, bool e5,
+               hipStream_t stream) {
+  dim3 grid((unsigned)((C + 255) / 256), (unsigned)(R / 32));
+  if (e5)
+    mx_cast_t_kernel<true><<<grid, 256, 0, stream>>>(x, q, sc, R, C);
+  else
+    mx_cast_t_kernel<false><<<grid, 256, 0, stream>>>(x, q, sc, R, C);
 }
 
 void mx_cast_bf16(const bf16* x, u8* q, u8* sc, long nblk,
