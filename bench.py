@@ -43,6 +43,19 @@ def main():
         torch.cuda.set_device(local_rank)
         backend = os.environ.get("MXA_DIST_BACKEND", "nccl")
         dist.init_process_group(backend=backend)
+        try:
+            # probe: RCCL refuses two ranks on one device ("Duplicate GPU
+            # detected") — overlap-path validation on a 1-GPU lease falls
+            # back to gloo over CUDA tensors (same hook/bucket code)
+            t = torch.ones(8, device=f"cuda:{local_rank}")
+            dist.all_reduce(t)
+            torch.cuda.synchronize()
+        except Exception:
+            dist.destroy_process_group()
+            dist.init_process_group(backend="gloo")
+            if rank == 0:
+                print("# nccl refused (shared device?); gloo fallback",
+                      file=sys.stderr)
 
     import anand_mxnet_amd as mx
     from anand_mxnet_amd import gluon, autograd, nd

@@ -36,11 +36,20 @@ def main():
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    local_rank = local_rank % max(torch.cuda.device_count(), 1)
     distributed = world_size > 1
     if distributed:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl")
+        dist.init_process_group(
+            backend=os.environ.get("MXA_DIST_BACKEND", "nccl"))
+        try:
+            t = torch.tensor([1.0], device=f"cuda:{local_rank}")
+            dist.all_reduce(t)
+            torch.cuda.synchronize()
+        except Exception:
+            dist.destroy_process_group()
+            dist.init_process_group(backend="gloo")
 
     import anand_mxnet_amd as mx
     from anand_mxnet_amd import gluon, autograd, nd
