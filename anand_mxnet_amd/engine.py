@@ -91,6 +91,74 @@ class Engine:
             torch.cuda.synchronize()
 
 
+class DepEngine:
+    """The native variable-dependency scheduler (src/kernels/engine.hip —
+    reference parity: ThreadedEngine var queues, ThreadedEnginePerDevice
+    workers, NaiveEngine oracle via MXA_ENGINE_TYPE=NaiveEngine).
+
+    Operations are pushed with explicit read/write variable lists and run
+    on worker threads (per-device HIP streams for GPU payloads); the
+    public sync points are wait_for_var / wait_all, and exceptions from
+    payloads resurface there (reference exception contract)."""
+
+    _inst = None
+
+    def __init__(self):
+        from .base import lib
+        self._c = lib()
+        if self._c is None or not hasattr(self._c, "dep_new_var"):
+            raise RuntimeError("native engine not built")
+
+    @classmethod
+    def get(cls) -> "DepEngine":
+        if cls._inst is None:
+            cls._inst = DepEngine()
+        return cls._inst
+
+    @classmethod
+    def available(cls) -> bool:
+        try:
+            cls.get()
+            return True
+        except Exception:
+            return False
+
+    @property
+    def engine_type(self) -> str:
+        return self._c.dep_engine_type()
+
+    def new_var(self) -> int:
+        return self._c.dep_new_var()
+
+    def delete_var(self, var: int):
+        self._c.dep_delete_var(var)
+
+    def var_version(self, var: int) -> int:
+        return self._c.dep_var_version(var)
+
+    def push(self, fn, reads=(), writes=(), device=-1):
+        self._c.dep_push(fn, list(reads), list(writes), device)
+
+    def push_allreduce(self, tensors, average, reads, writes, device=-1):
+        """Native no-GIL payload: in-place sum/average + broadcast."""
+        self._c.dep_push_allreduce(tensors, average, list(reads),
+                                   list(writes), device)
+
+    def wait_for_var(self, var: int):
+        self._c.dep_wait_for_var(var)
+
+    def wait_all(self):
+        self._c.dep_wait_all()
+
+    def var_of(self, tensor) -> int:
+        """Lazy engine variable attached to a tensor's storage."""
+        v = getattr(tensor, "_mxa_dep_var", None)
+        if v is None:
+            v = self.new_var()
+            tensor._mxa_dep_var = v
+        return v
+
+
 def get() -> Engine:
     if Engine._inst is None:
         Engine._inst = Engine()

@@ -48,6 +48,10 @@ def _ensure_pg():
     dist.init_process_group(backend=backend, init_method="env://")
 
 
+def _engine_kvstore_enabled() -> bool:
+    return os.environ.get("MXA_ENGINE_KVSTORE", "1") != "0"
+
+
 def _as_list(x):
     return x if isinstance(x, (list, tuple)) else [x]
 
@@ -142,7 +146,30 @@ class KVStore:
 
     # ---------------- allreduce-style helper ----------------
     def allreduce(self, tensors, average=False):
-        """Sum (or average) per-device tensor lists in place."""
+        """Sum (or average) per-device tensor lists in place.
+
+        Multi-key calls are scheduled on the native dependency engine
+        (engine.DepEngine): one reduce op per key with read/write vars per
+        tensor, executing concurrently on the worker pool (the reference's
+        CommDevice reduce was engine-ordered the same way); the call
+        returns after a wait so the public in-place semantics are
+        unchanged.  MXA_ENGINE_KVSTORE=0 keeps the inline loop."""
+        tensors = list(tensors)
+        if (len(tensors) > 1 and _engine_kvstore_enabled()):
+            from . import engine as _eng
+            if _eng.DepEngine.available():
+                eng = _eng.DepEngine.get()
+                for lst in tensors:
+                    lst = _as_list(lst)
+                    if len(lst) < 2:
+                        continue
+                    ts = [v._t.detach() for v in lst]
+                    wvars = [eng.var_of(t) for t in ts]
+                    dev = ts[0].device.index if ts[0].is_cuda else -1
+                    eng.push_allreduce(ts, average, [], wvars,
+                                       -1 if dev is None else dev)
+                eng.wait_all()
+                return
         for lst in tensors:
             lst = _as_list(lst)
             acc = self._reduce(lst)

@@ -1115,6 +1115,67 @@ at::Tensor multi_sumsq_run(const at::Tensor& blob, long ntens, long nchunks,
 
 // hipEvent pool: WaitForVar/cross-stream ordering primitives (reference
 // engine.h VarHandle semantics collapse to event waits under stream order)
+// ---- dependency engine (engine.hip: var-queue scheduler) ----
+namespace mxa_engine {
+uint64_t engine_new_var();
+void engine_delete_var(uint64_t);
+uint64_t engine_var_version(uint64_t);
+void engine_push(std::function<void()>, const std::vector<uint64_t>&,
+                 const std::vector<uint64_t>&, int);
+void engine_wait_for_var(uint64_t);
+void engine_wait_all();
+const char* engine_type();
+hipStream_t engine_current_stream();
+}  // namespace mxa_engine
+
+static void dep_push_py(py::function fn, std::vector<uint64_t> reads,
+                        std::vector<uint64_t> writes, int device) {
+  // py::function must be destroyed under the GIL; the payload runs on an
+  // engine worker thread (GIL acquired around the call)
+  std::shared_ptr<py::function> fp(
+      new py::function(std::move(fn)), [](py::function* p) {
+        py::gil_scoped_acquire g;
+        delete p;
+      });
+  mxa_engine::engine_push(
+      [fp]() {
+        py::gil_scoped_acquire g;
+        (*fp)();
+      },
+      reads, writes, device);
+}
+
+// native payload: y += sum(xs) elementwise (KVStore 'device' reduce) —
+// no GIL on the hot path; runs on the engine worker's stream
+static void dep_push_sum_py(at::Tensor dst, std::vector<at::Tensor> srcs,
+                            std::vector<uint64_t> reads,
+                            std::vector<uint64_t> writes, int device) {
+  auto payload = [dst, srcs]() {
+    c10::InferenceMode guard;
+    for (auto& s : srcs) dst.add_(s);
+  };
+  mxa_engine::engine_push(payload, reads, writes, device);
+}
+
+// native payload: in-place allreduce of a tensor list (KVStore local /
+// device reduce): acc into t0, optional average, broadcast back.  Runs on
+// an engine worker thread with no GIL; cross-device copies go through
+// torch's peer copy (xGMI P2P).
+static void dep_push_allreduce_py(std::vector<at::Tensor> ts, bool average,
+                                  std::vector<uint64_t> reads,
+                                  std::vector<uint64_t> writes, int device) {
+  auto payload = [ts, average]() {
+    c10::InferenceMode guard;
+    auto& t0 = const_cast<at::Tensor&>(ts[0]);
+    for (size_t i = 1; i < ts.size(); ++i)
+      t0.add_(ts[i].to(t0.device(), /*non_blocking=*/true));
+    if (average) t0.div_((double)ts.size());
+    for (size_t i = 1; i < ts.size(); ++i)
+      const_cast<at::Tensor&>(ts[i]).copy_(t0, /*non_blocking=*/true);
+  };
+  mxa_engine::engine_push(payload, reads, writes, device);
+}
+
 long engine_new_event() {
   hipEvent_t e;
   TORCH_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming) ==
@@ -1216,6 +1277,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_opt_run", &multi_opt_run);
   m.def("multi_sumsq_run", &multi_sumsq_run);
   m.def("engine_new_stream", &engine_new_stream);
+  m.def("dep_new_var", &mxa_engine::engine_new_var);
+  m.def("dep_delete_var", &mxa_engine::engine_delete_var);
+  m.def("dep_var_version", &mxa_engine::engine_var_version);
+  m.def("dep_push", &dep_push_py, py::arg("fn"), py::arg("reads"),
+        py::arg("writes"), py::arg("device") = -1);
+  m.def("dep_push_sum", &dep_push_sum_py);
+  m.def("dep_push_allreduce", &dep_push_allreduce_py, py::arg("tensors"),
+        py::arg("average"), py::arg("reads"), py::arg("writes"),
+        py::arg("device") = -1);
+  m.def("dep_wait_for_var", &mxa_engine::engine_wait_for_var,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("dep_wait_all", &mxa_engine::engine_wait_all,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("dep_engine_type",
+        []() { return std::string(mxa_engine::engine_type()); });
   m.def("engine_new_event", &engine_new_event);
   m.def("engine_record_event", &engine_record_event);
   m.def("engine_stream_wait_event", &engine_stream_wait_event);
