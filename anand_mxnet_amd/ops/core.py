@@ -890,6 +890,73 @@ def dequantize(data, min_range, max_range, out_type="float32"):
     return data.float() * (amax / 127.0)
 
 
+def _i8_gemm(x8, w8):
+    """Exact int32 x8 @ w8^T; CDNA4 i8 MFMA kernel on GPU (K % 128), torch
+    integer matmul as the exact CPU oracle."""
+    if x8.is_cuda and x8.shape[-1] % 128 == 0:
+        from . import hip as _h  # noqa: F401 (registers/loads _C)
+        from ..base import lib
+        return lib().gemm_i8(x8.contiguous(), w8.contiguous())
+    return torch.matmul(x8.to(torch.int32), w8.to(torch.int32).t())
+
+
+@register("_contrib_quantized_fully_connected")
+@register("quantized_fully_connected")
+def quantized_fully_connected(data, weight, bias, min_data, max_data,
+                              min_weight, max_weight, min_bias=None,
+                              max_bias=None, num_hidden=None, no_bias=True,
+                              flatten=True):
+    """int8 x int8 -> int32 FC (reference
+    quantization/quantized_fully_connected.cc).  Inputs are symmetric
+    int8 with float ranges; output int32 with the combined range so a
+    following requantize/dequantize recovers floats."""
+    x = data.reshape(data.shape[0], -1) if flatten and data.dim() > 2 \
+        else data
+    out32 = _i8_gemm(x, weight)
+    sd = max(abs(float(min_data.reshape(-1)[0])),
+             abs(float(max_data.reshape(-1)[0]))) / 127.0
+    sw = max(abs(float(min_weight.reshape(-1)[0])),
+             abs(float(max_weight.reshape(-1)[0]))) / 127.0
+    so = sd * sw
+    if not no_bias and bias is not None:
+        sb = max(abs(float(min_bias.reshape(-1)[0])),
+                 abs(float(max_bias.reshape(-1)[0]))) / 127.0
+        b32 = (bias.to(torch.int32).float() * (sb / so)).round() \
+            .to(torch.int32)
+        out32 = out32 + b32
+    lim = float(2 ** 31 - 1) * so
+    return [out32, torch.tensor([-lim]), torch.tensor([lim])]
+
+
+@register("_contrib_quantized_conv")
+@register("quantized_conv")
+def quantized_conv(data, weight, bias, min_data, max_data, min_weight,
+                   max_weight, kernel=None, stride=(1, 1), pad=(0, 0),
+                   num_filter=None, no_bias=True, layout="NHWC"):
+    """int8 convolution (reference quantized_conv.cc), NHWC.  Pointwise
+    (1x1 s1) runs as the plain int8 GEMM; general windows go through an
+    int32 unfold composition (exact)."""
+    assert layout == "NHWC", "quantized_conv: NHWC only"
+    N, H, W, C = data.shape
+    K = weight.shape[0]
+    R, S = (kernel if kernel else (weight.shape[1], weight.shape[2]))
+    if R == 1 and S == 1 and tuple(stride) == (1, 1) and tuple(pad) == (0, 0):
+        out32 = _i8_gemm(data.reshape(-1, C), weight.reshape(K, C)) \
+            .reshape(N, H, W, K)
+    else:
+        xi = data.permute(0, 3, 1, 2).to(torch.int32).float()
+        wi = weight.permute(0, 3, 1, 2).to(torch.int32).float()
+        out32 = torch.nn.functional.conv2d(
+            xi, wi, None, tuple(stride), tuple(pad)) \
+            .round().to(torch.int32).permute(0, 2, 3, 1)
+    sd = max(abs(float(min_data.reshape(-1)[0])),
+             abs(float(max_data.reshape(-1)[0]))) / 127.0
+    sw = max(abs(float(min_weight.reshape(-1)[0])),
+             abs(float(max_weight.reshape(-1)[0]))) / 127.0
+    lim = float(2 ** 31 - 1) * sd * sw
+    return [out32, torch.tensor([-lim]), torch.tensor([lim])]
+
+
 @register("_contrib_requantize")
 @register("requantize")
 def requantize(data, min_range, max_range, min_calib_range=None,

@@ -357,10 +357,14 @@ class _SoftmaxOutputFn(torch.autograd.Function):
 
 
 @register("SoftmaxOutput")
-def softmax_output(data, label, grad_scale=1.0, ignore_label=-1,
+def softmax_output(data, label=None, grad_scale=1.0, ignore_label=-1,
                    use_ignore=False, normalization="null",
                    multi_output=False, preserve_shape=False,
                    out_grad=False, smooth_alpha=0.0):
+    if label is None:
+        # inference binding without a label (reference semantics: the op
+        # is a plain softmax at predict time — c_predict_api path)
+        return F.softmax(data.float(), dim=-1).to(data.dtype)
     return _SoftmaxOutputFn.apply(data, label, grad_scale, normalization,
                                   ignore_label, use_ignore)
 

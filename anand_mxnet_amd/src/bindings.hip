@@ -3,6 +3,7 @@
 // native boundary is a torch extension: tensors in, stream-ordered HIP
 // kernel launches out, zero Python in the hot loop.
 #include <torch/extension.h>
+#include <pybind11/numpy.h>
 
 #include <hip/hip_runtime.h>
 #include <ATen/hip/HIPContext.h>
@@ -141,6 +142,9 @@ void gemm_mx_bf16(const unsigned char*, const unsigned char*,
                   const float*, int, int, int, long, int, int, int, float*,
                   const unsigned char*, hipStream_t);
 int gemm_mx_nslots(int, int, int, int, int);
+void gemm_tn_i8(const unsigned char*, const unsigned char*, int*, bf16*,
+                const float*, const float*, int, int, int, long, long, long,
+                const unsigned char*, hipStream_t);
 void mx_cast_bf16(const bf16*, unsigned char*, unsigned char*, long,
                   hipStream_t);
 void mx_cast_bf8(const bf16*, unsigned char*, unsigned char*, long,
@@ -392,6 +396,38 @@ at::Tensor gemm_mx(const at::Tensor& A8, const at::Tensor& Asc,
                     (int)M, (int)N, (int)K, N, relu, (int)afmt,
                     (int)bfmt, wp, (const unsigned char*)zp(zprobe),
                     cur_stream());
+  return out;
+}
+
+// int8 TN GEMM (quantized FC / pointwise conv compute).  inv_scale none
+// -> exact int32 accumulator out; else fused dequant to bf16 (+bias).
+at::Tensor gemm_i8(const at::Tensor& A8, const at::Tensor& B8,
+                   const c10::optional<at::Tensor>& inv_scale,
+                   const c10::optional<at::Tensor>& bias) {
+  TORCH_CHECK(A8.scalar_type() == at::kChar && B8.scalar_type() == at::kChar,
+              "gemm_i8: int8 operands");
+  TORCH_CHECK(A8.is_contiguous() && B8.is_contiguous());
+  long M = A8.size(0), K = A8.size(1), N = B8.size(0);
+  TORCH_CHECK(B8.size(1) == K && K % 128 == 0, "gemm_i8: K % 128 != 0");
+  auto zprobe = at::empty(
+      {0}, at::TensorOptions().device(A8.device()).dtype(at::kBFloat16));
+  if (!inv_scale.has_value()) {
+    auto out = at::empty({M, N},
+                         A8.options().dtype(at::kInt));
+    mxa::gemm_tn_i8((const unsigned char*)A8.data_ptr<signed char>(),
+                    (const unsigned char*)B8.data_ptr<signed char>(),
+                    out.data_ptr<int>(), nullptr, nullptr, nullptr, (int)M,
+                    (int)N, (int)K, K, K, N,
+                    (const unsigned char*)zp(zprobe), cur_stream());
+    return out;
+  }
+  auto out = at::empty({M, N}, A8.options().dtype(at::kBFloat16));
+  mxa::gemm_tn_i8((const unsigned char*)A8.data_ptr<signed char>(),
+                  (const unsigned char*)B8.data_ptr<signed char>(), nullptr,
+                  bfp_mut(out), inv_scale->data_ptr<float>(),
+                  bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                  (int)M, (int)N, (int)K, K, K, N,
+                  (const unsigned char*)zp(zprobe), cur_stream());
   return out;
 }
 
@@ -1213,6 +1249,106 @@ long engine_new_stream(long device) {
   return (long)(uintptr_t)s;
 }
 
+// ---------------------------------------------------------------------------
+// C predict API (reference src/c_api/c_predict_api.cc).  extern "C"
+// symbols exported from this .so; they drive anand_mxnet_amd.predict
+// through the embedded interpreter.  Thread-local error string via
+// MXAGetLastError (reference MXGetLastError contract).
+// ---------------------------------------------------------------------------
+static thread_local std::string mxa_last_error;
+
+extern "C" {
+
+const char* MXAGetLastError() { return mxa_last_error.c_str(); }
+
+int MXAPredCreate(const char* symbol_json, const char* params_file,
+                  int dev_type, int dev_id, unsigned num_inputs,
+                  const char** input_names, void** out) {
+  py::gil_scoped_acquire g;
+  try {
+    py::module m = py::module::import("anand_mxnet_amd.predict");
+    py::list names;
+    for (unsigned i = 0; i < num_inputs; ++i)
+      names.append(py::str(input_names[i]));
+    py::object p = m.attr("CPredictor")(py::str(symbol_json),
+                                        py::str(params_file), dev_type,
+                                        dev_id, names);
+    *out = new py::object(std::move(p));
+    return 0;
+  } catch (const std::exception& e) {
+    mxa_last_error = e.what();
+    return -1;
+  }
+}
+
+int MXAPredSetInput(void* h, const char* name, const float* data,
+                    const long* shape, unsigned ndim) {
+  py::gil_scoped_acquire g;
+  try {
+    auto& p = *(py::object*)h;
+    std::vector<long> dims(shape, shape + ndim);
+    long n = 1;
+    for (long d : dims) n *= d;
+    py::array_t<float> arr(dims);
+    std::memcpy(arr.mutable_data(), data, sizeof(float) * n);
+    p.attr("set_input")(py::str(name), arr);
+    return 0;
+  } catch (const std::exception& e) {
+    mxa_last_error = e.what();
+    return -1;
+  }
+}
+
+int MXAPredForward(void* h) {
+  py::gil_scoped_acquire g;
+  try {
+    ((py::object*)h)->attr("forward")();
+    return 0;
+  } catch (const std::exception& e) {
+    mxa_last_error = e.what();
+    return -1;
+  }
+}
+
+int MXAPredGetOutputShape(void* h, unsigned idx, long* shape,
+                          unsigned* ndim) {
+  py::gil_scoped_acquire g;
+  try {
+    py::list sh = ((py::object*)h)->attr("output_shape")(idx);
+    *ndim = (unsigned)py::len(sh);
+    for (unsigned i = 0; i < *ndim; ++i) shape[i] = sh[i].cast<long>();
+    return 0;
+  } catch (const std::exception& e) {
+    mxa_last_error = e.what();
+    return -1;
+  }
+}
+
+int MXAPredGetOutput(void* h, unsigned idx, float* out, long size) {
+  py::gil_scoped_acquire g;
+  try {
+    py::array_t<float> a =
+        ((py::object*)h)->attr("output_data")(idx).cast<py::array_t<float>>();
+    if ((long)a.size() > size) {
+      mxa_last_error = "output buffer too small";
+      return -1;
+    }
+    std::memcpy(out, a.data(), sizeof(float) * a.size());
+    return 0;
+  } catch (const std::exception& e) {
+    mxa_last_error = e.what();
+    return -1;
+  }
+}
+
+int MXAPredFree(void* h) {
+  py::gil_scoped_acquire g;
+  delete (py::object*)h;
+  return 0;
+}
+
+}  // extern "C"
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn", &gemm_tn, "C = A @ B^T (bf16, MFMA)");
   m.def("gemm_tn_stats", &gemm_tn_stats,
@@ -1223,6 +1359,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn_fp8", &gemm_tn_fp8);
   m.def("cast_to_fp8", &cast_to_fp8);
   m.def("mx_cast", &mx_cast, py::arg("x"), py::arg("e5") = false);
+  m.def("gemm_i8", &gemm_i8, py::arg("A8"), py::arg("B8"),
+        py::arg("inv_scale") = py::none(), py::arg("bias") = py::none());
   m.def("mx_cast_t", &mx_cast_t, py::arg("x"), py::arg("e5") = false);
   m.def("gemm_mx", &gemm_mx, py::arg("A8"), py::arg("Asc"), py::arg("B8"),
         py::arg("Bsc"), py::arg("bias") = py::none(),

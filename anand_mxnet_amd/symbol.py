@@ -240,14 +240,46 @@ def loads(json_str):
 
 # ---------------- op functions on symbols ----------------
 
+# parameter slots auto-created as variables when not passed explicitly
+# (reference UX: sym.FullyConnected(x, num_hidden=8, name='fc1') creates
+# fc1_weight / fc1_bias automatically)
+_AUTO_PARAMS = {
+    "FullyConnected": ("weight", "bias"),
+    "Convolution": ("weight", "bias"),
+    "Deconvolution": ("weight", "bias"),
+    "Embedding": ("weight",),
+    "BatchNorm": ("gamma", "beta", "moving_mean", "moving_var"),
+    "LayerNorm": ("gamma", "beta"),
+}
+
+
 def _sym_op_factory(op_name):
     def fn(*args, name=None, **kwargs):
         inputs = [a for a in args if isinstance(a, Symbol)]
         # keyword-NDArray-style: data=..., weight=...
+        explicit = {}
         for key in ("data", "weight", "bias", "gamma", "beta", "moving_mean",
                     "moving_var", "label"):
             if key in kwargs and isinstance(kwargs[key], Symbol):
-                inputs.append(kwargs.pop(key))
+                explicit[key] = kwargs.pop(key)
+        auto = _AUTO_PARAMS.get(op_name, ())
+        for key in ("data", "label"):  # non-parameter inputs keep their slot
+            if key in explicit:
+                inputs.append(explicit.pop(key))
+        if auto:
+            # resolve the node name now so auto vars get '{name}_{slot}'
+            from . import name as _nm
+            name = _nm.current().get(name, op_name.lower())
+            no_bias = str(kwargs.get("no_bias", False)).lower() in \
+                ("true", "1")
+            for slot in auto:
+                if slot == "bias" and no_bias:
+                    continue
+                s = explicit.pop(slot, None)
+                if s is None:
+                    s = var(f"{name}_{slot}")
+                inputs.append(s)
+        inputs.extend(explicit.values())
         attrs = {k: _attr_str(v) for k, v in kwargs.items()}
         return _make_op(op_name, inputs, attrs, name)
     fn.__name__ = op_name

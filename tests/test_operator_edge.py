@@ -299,3 +299,48 @@ def test_int8_quantize_roundtrip():
     d2 = nd.dequantize(q2, l2, h2)
     np.testing.assert_allclose(_np(d2), _np(x2), atol=2.0 / 127)
     assert float(_np(q2)[0]) == -127 and float(_np(q2)[2]) == 127
+
+
+def test_quantized_fully_connected_pipeline():
+    """quantize -> int8 FC -> dequantize recovers the float FC within
+    int8 quantization error (reference quantized_fully_connected)."""
+    import torch
+    from anand_mxnet_amd.ops.registry import get as get_op
+    rng = np.random.RandomState(0)
+    x = torch.tensor(rng.uniform(-1, 1, (8, 128)).astype(np.float32))
+    w = torch.tensor(rng.uniform(-1, 1, (16, 128)).astype(np.float32))
+    q = get_op("quantize_v2")
+    qfc = get_op("quantized_fully_connected")
+    deq = get_op("dequantize")
+    xq, xmin, xmax = q(x, out_type="int8")
+    wq, wmin, wmax = q(w, out_type="int8")
+    out32, omin, omax = qfc(xq, wq, None, xmin, xmax, wmin, wmax)
+    o = out32.asnumpy()
+    assert o.dtype == np.int32
+    # combined-scale dequant
+    sd = float(xmax.asscalar()) / 127.0
+    sw = float(wmax.asscalar()) / 127.0
+    f = o.astype(np.float64) * (sd * sw)
+    ref = (x @ w.t()).numpy()
+    err = np.abs(f - ref).max()
+    assert err < 0.15, err
+
+
+def test_quantized_conv_pointwise():
+    import torch
+    from anand_mxnet_amd.ops.registry import get as get_op
+    rng = np.random.RandomState(1)
+    x = torch.tensor(rng.uniform(-1, 1, (2, 4, 4, 128)).astype(np.float32))
+    w = torch.tensor(rng.uniform(-1, 1, (32, 1, 1, 128)).astype(np.float32))
+    q = get_op("quantize_v2")
+    qc = get_op("quantized_conv")
+    xq, xmin, xmax = q(x, out_type="int8")
+    wq, wmin, wmax = q(w, out_type="int8")
+    out32, omin, omax = qc(xq, wq, None, xmin, xmax, wmin, wmax,
+                           kernel=(1, 1))
+    sd = float(xmax.asscalar()) / 127.0
+    sw = float(wmax.asscalar()) / 127.0
+    f = out32.asnumpy().astype(np.float64) * (sd * sw)
+    ref = torch.nn.functional.conv2d(
+        x.permute(0, 3, 1, 2), w.permute(0, 3, 1, 2)).permute(0, 2, 3, 1)
+    assert np.abs(f - ref.numpy()).max() < 0.15
