@@ -68,19 +68,32 @@ def softrelu(x):
     return torch.nn.functional.softplus(x)
 
 
+def _act_or_torch(x, name, fallback):
+    # GPU: the fused CDNA4 activation kernels (misc.hip) — nd.gelu in the
+    # BERT MLM head was silently running torch's gelu (substrate audit)
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.numel() % 8 == 0:
+        from .registry import GPU_IMPL
+        impl = GPU_IMPL.get("act_" + name)
+        if impl is not None:
+            return impl(x)
+    return fallback(x)
+
+
 @register("gelu")
 def gelu(x):
-    return torch.nn.functional.gelu(x)
+    return _act_or_torch(x, "gelu", torch.nn.functional.gelu)
 
 
 @register("gelu_tanh")
 def gelu_tanh(x):
-    return torch.nn.functional.gelu(x, approximate="tanh")
+    return _act_or_torch(
+        x, "gelu_tanh",
+        lambda t: torch.nn.functional.gelu(t, approximate="tanh"))
 
 
 @register("silu")
 def silu(x):
-    return torch.nn.functional.silu(x)
+    return _act_or_torch(x, "silu", torch.nn.functional.silu)
 
 
 @register("LeakyReLU")

@@ -913,13 +913,13 @@ def test_gemm_i8_exact_vs_integer_oracle():
         b = torch.randint(-127, 128, (N, K), dtype=torch.int8,
                           device=_dev())
         out = _C.gemm_i8(a, b)
-        ref = (a.to(torch.int32) @ b.to(torch.int32).t())
-        assert (out == ref).all(), f"i8 gemm mismatch {M}x{N}x{K}"
+        ref = (a.cpu().to(torch.int32) @ b.cpu().to(torch.int32).t())
+        assert (out.cpu() == ref).all(), f"i8 gemm mismatch {M}x{N}x{K}"
     # fused dequant epilogue
     a = torch.randint(-127, 128, (64, 128), dtype=torch.int8, device=_dev())
     b = torch.randint(-127, 128, (32, 128), dtype=torch.int8, device=_dev())
     isc = torch.tensor([1.0 / (127.0 * 127.0)], device=_dev())
     outf = _C.gemm_i8(a, b, isc)
-    reff = (a.to(torch.int32) @ b.to(torch.int32).t()).float() \
-        / (127.0 * 127.0)
+    reff = (a.cpu().to(torch.int32) @ b.cpu().to(torch.int32).t()).float() \
+        .to(_dev()) / (127.0 * 127.0)
     assert_close(outf, reff, rtol=1e-2, atol=1e-2, msg="i8 dequant")
