@@ -856,8 +856,11 @@ def mha(q, k, v, num_heads, causal, scale, dropout_p=0.0, training=False):
     B, Tq, HD = q.shape
     Tk = k.shape[1]
     D = HD // num_heads
-    if (_USE_FLASH and D == 64 and Tq % 128 == 0 and Tk % 128 == 0
-            and Tq == Tk and q.dtype == torch.bfloat16):
+    if (_USE_FLASH and D == 64 and (Tq == Tk or not causal)
+            and q.dtype == torch.bfloat16):
+        # any sequence length (tail blocks are masked in-kernel) and
+        # cross-attention Tq != Tk (decoder enc-dec attention) run fused;
+        # only causal masking assumes a square score matrix
         return _FlashMha.apply(q, k, v, num_heads, causal, scale,
                                dropout_p, training)
     if (D % 64 == 0 and Tk % 64 == 0 and Tq % 8 == 0

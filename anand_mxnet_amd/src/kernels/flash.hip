@@ -1,4 +1,6 @@
-// Fused flash-attention for gfx950 (MI355X), D = 64, seq % 128 == 0.
+// Fused flash-attention for gfx950 (MI355X), D = 64, any Tq/Tk
+// (tail blocks clamp their staging rows and mask scores at col >= Tk;
+// cross-attention Tq != Tk supported for the non-causal case).
 //
 // Forward: one workgroup per (batch*head, 128-query block); K/V stream
 // through double-buffered LDS in 128-row chunks; scores, online softmax,
@@ -79,7 +81,10 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
           (unsigned)((lw * 2 + p) * 1024));
       auto lds = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&qimg[0] + lbase);
-      const bf16* gp = qbase + (long)(qb * 128 + row) * hd + scg * 8;
+      // tail blocks: clamp the source row (finite garbage; the scores,
+      // stats and O writes of rows >= Tq are guarded/discarded)
+      int qrow = min(qb * 128 + row, Tq - 1);
+      const bf16* gp = qbase + (long)qrow * hd + scg * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gp, lds,
           16, 0, 0);
@@ -99,11 +104,12 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
           (__attribute__((address_space(3))) char*)&kimg[buf][0] + lbase);
       auto ldsV = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&vimg[buf][0] + lbase);
-      const bf16* gk = kbase + (long)(c * 128 + row) * hd + scg * 8;
+      int krow = min(c * 128 + row, Tk - 1);  // tail clamp (masked below)
+      const bf16* gk = kbase + (long)krow * hd + scg * 8;
       // V tr image: source chunk key is (row&3)<<1 on 8-bf16 chunks —
       // row&3 here = (ll>>3)&3, chunk pos = ll&7
       int vcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);
-      const bf16* gv = vbase + (long)(c * 128 + row) * hd + vcg * 8;
+      const bf16* gv = vbase + (long)krow * hd + vcg * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gk, ldsK,
           16, 0, 0);
@@ -133,8 +139,8 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
 #pragma unroll
     for (int nf = 0; nf < 2; ++nf) acc_o[mf][nf] = f32x4{0, 0, 0, 0};
 
-  const int nchunks_all = Tk / 128;
-  const int nchunks = causal ? (qb + 1) : nchunks_all;
+  const int nchunks_all = (Tk + 127) / 128;
+  const int nchunks = causal ? min(qb + 1, nchunks_all) : nchunks_all;
 
   stage_kv(0, 0);
   __syncthreads();
@@ -184,7 +190,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
         for (int i = 0; i < 4; ++i) {
           int row = qb * 128 + wmq + mf * 16 + (ll >> 4) * 4 + i;
           float v = acc_s[mf][nf][i] * alpha;
-          if (causal && col > row) v = -3.0e38f;
+          if ((causal && col > row) || col >= Tk) v = -3.0e38f;
           acc_s[mf][nf][i] = v;
           cmax[mf][i] = fmaxf(cmax[mf][i], v);
         }
@@ -333,6 +339,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+      if (qb * 128 + row >= Tq) continue;  // tail guard
       float inv = l_i[mf][i] > 0.f ? 1.f / l_i[mf][i] : 0.f;
 #pragma unroll
       for (int nf = 0; nf < 2; ++nf) {
@@ -348,6 +355,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+        if (qb * 128 + row >= Tq) continue;  // tail guard
         Ms[(long)z * Tq + qb * 128 + row] = m_i[mf][i];
         Ls[(long)z * Tq + qb * 128 + row] = l_i[mf][i];
       }
@@ -358,7 +366,7 @@ void flash_fwd_bf16(const bf16* Q, const bf16* K, const bf16* V, bf16* O,
                     float* Ms, float* Ls, int B, int H, int Tq, int Tk,
                     float alpha, bool causal, const u64* seed_ptr, u64 salt,
                     float pdrop, hipStream_t stream) {
-  dim3 grid(Tq / 128, 1, B * H);
+  dim3 grid((Tq + 127) / 128, 1, B * H);
   float dscale = pdrop > 0.f ? 1.f / (1.f - pdrop) : 1.f;
   flash_fwd_core<<<grid, 512, 0, stream>>>(
       Q, K, V, O, Ms, Ls, B, H, Tq, Tk, alpha, causal ? 1 : 0,
@@ -415,7 +423,8 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
   const int scg = (ll & 7) ^ ((ll >> 3) & 7);
   const int tcg = (ll & 7) ^ (((ll >> 3) & 3) << 1);
 
-  auto stage128 = [&](bf16* img, const bf16* base, int row0, int cg) {
+  auto stage128 = [&](bf16* img, const bf16* base, int row0, int cg,
+                      int rmax) {
 #pragma unroll
     for (int p = 0; p < 2; ++p) {
       int row = srow + p * 8;
@@ -423,18 +432,20 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
           (unsigned)((lw * 2 + p) * 1024));
       auto lds = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)img + lbase);
-      const bf16* gp = base + (long)(row0 + row) * hd + cg * 8;
+      const bf16* gp = base + (long)min(row0 + row, rmax - 1) * hd + cg * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gp, lds,
           16, 0, 0);
     }
   };
-  auto stage64r = [&](bf16* img, const bf16* base, int row0, int cg) {
+  auto stage64r = [&](bf16* img, const bf16* base, int row0, int cg,
+                      int rmax) {
     unsigned lbase =
         __builtin_amdgcn_readfirstlane((unsigned)(lw * 1024));
     auto lds = (__attribute__((address_space(3))) unsigned int*)(
         (__attribute__((address_space(3))) char*)img + lbase);
-    const bf16* gp = base + (long)(row0 + srow64) * hd + cg * 8;
+    const bf16* gp = base + (long)min(row0 + srow64, rmax - 1) * hd +
+                     cg * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
         0, 0);
@@ -445,9 +456,9 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
   const bf16* vbase = V + (long)b * Tk * hd + (long)h * D;
   const bf16* dobase = dO + (long)b * Tq * hd + (long)h * D;
 
-  stage128(ktn, kbase, kvb * 128, scg);
-  stage128(ktr, kbase, kvb * 128, tcg);
-  stage128(vtn, vbase, kvb * 128, scg);
+  stage128(ktn, kbase, kvb * 128, scg, Tk);
+  stage128(ktr, kbase, kvb * 128, tcg, Tk);
+  stage128(vtn, vbase, kvb * 128, scg, Tk);
 
   const int wmq = (lw & 3) * 32;   // 32 tk rows of this wave
   const int wnt = (lw >> 2) * 32;  // 32 q cols (of the 64-chunk)
@@ -464,15 +475,15 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
     }
 
   const int qc0 = causal ? kvb * 2 : 0;
-  const int nq = Tq / 64;
+  const int nq = (Tq + 63) / 64;
 
   for (int qc = qc0; qc < nq; ++qc) {
-    stage64r(qtn, qbase, qc * 64, scg);
-    stage64r(qtr, qbase, qc * 64, tcg);
-    stage64r(dotn, dobase, qc * 64, scg);
-    stage64r(dotr, dobase, qc * 64, tcg);
+    stage64r(qtn, qbase, qc * 64, scg, Tq);
+    stage64r(qtr, qbase, qc * 64, tcg, Tq);
+    stage64r(dotn, dobase, qc * 64, scg, Tq);
+    stage64r(dotr, dobase, qc * 64, tcg, Tq);
     for (int i = t; i < 64; i += 512) {
-      long qg = (long)z * Tq + qc * 64 + i;
+      long qg = (long)z * Tq + min(qc * 64 + i, Tq - 1);
       stats[0][i] = Ms[qg];
       stats[1][i] = Ls[qg];
       stats[2][i] = Drow[qg];
@@ -542,7 +553,9 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
           int tkrow = wmq + mf * 16 + (ll >> 4) * 4 + i;
           long kg = (long)kvb * 128 + tkrow;
           float sv = acc_st[mf][nf][i] * alpha;
-          float pv = (causal && kg > qg) ? 0.f : __expf(sv - m) * linv;
+          float pv = ((causal && kg > qg) || qg >= Tq || kg >= Tk)
+                         ? 0.f
+                         : __expf(sv - m) * linv;
           float keep = 1.f;
           if (thresh) {
             u32 r = hash_u32(seed, ((long)z * Tq + qg) * Tk + kg);
@@ -674,6 +687,7 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
           int row = wq16 + (ll >> 4) * 4 + i;
+          if (qc * 64 + row >= Tq) continue;  // tail guard
           atomicAdd(&dqb[(long)row * hd + col], acc_dq[nf][i]);
         }
       }
@@ -692,6 +706,7 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         int row = wmq + mf * 16 + (ll >> 4) * 4 + i;
+        if (kvb * 128 + row >= Tk) continue;  // tail guard
         dkb[(long)row * hd + col] = f2bf(acc_dk[mf][nf][i]);
         dvb[(long)row * hd + col] = f2bf(acc_dv[mf][nf][i]);
       }
@@ -736,7 +751,7 @@ void flash_bwd_bf16(const bf16* Q, const bf16* K, const bf16* V,
   int blocks = (int)min((rows + 3) / 4, 2048L);
   flash_drow<<<blocks, 256, 0, stream>>>(dOp, O, Drow, B, H, Tq);
   float dscale = pdrop > 0.f ? 1.f / (1.f - pdrop) : 1.f;
-  dim3 grid(Tk / 128, 1, B * H);
+  dim3 grid((Tk + 127) / 128, 1, B * H);
   flash_bwd_core<<<grid, 512, 0, stream>>>(
       Q, K, V, dOp, Ms, Ls, Drow, dQacc, dK, dV, B, H, Tq, Tk, alpha,
       causal ? 1 : 0, pdrop > 0.f ? seed_ptr : nullptr, salt, pdrop,

@@ -923,3 +923,94 @@ def test_gemm_i8_exact_vs_integer_oracle():
     reff = (a.cpu().to(torch.int32) @ b.cpu().to(torch.int32).t()).float() \
         .to(_dev()) / (127.0 * 127.0)
     assert_close(outf, reff, rtol=1e-2, atol=1e-2, msg="i8 dequant")
+
+
+def test_flash_cross_attention_fwd_bwd():
+    """Tq != Tk (decoder enc-dec attention) through the fused flash path,
+    forward AND backward vs the unfused fp32 oracle."""
+    B, H, Tq, Tk, D = 2, 4, 128, 384, 64
+    scale = 1.0 / (D ** 0.5)
+    q = _rt((B, Tq, H * D)).to(BF).requires_grad_(True)
+    k = _rt((B, Tk, H * D)).to(BF).requires_grad_(True)
+    v = _rt((B, Tk, H * D)).to(BF).requires_grad_(True)
+
+    out, m, l = _C.flash_fwd(q.detach().contiguous(),
+                             k.detach().contiguous(),
+                             v.detach().contiguous(), H, scale, False,
+                             0.0, 0)
+    qh = _heads(q.detach(), H).float()
+    kh = _heads(k.detach(), H).float()
+    vh = _heads(v.detach(), H).float()
+    s = qh @ kh.transpose(-1, -2) * scale
+    p = torch.nn.functional.softmax(s, dim=-1)
+    ref = (p @ vh).transpose(1, 2).reshape(B, Tq, H * D)
+    assert_close(out, ref, rtol=5e-2, atol=5e-2, msg="cross-attn out")
+
+    dout = _rt((B, Tq, H * D)).to(BF)
+    dq, dk, dv = _C.flash_bwd(q.detach().contiguous(),
+                              k.detach().contiguous(),
+                              v.detach().contiguous(), out.contiguous(),
+                              dout.contiguous(), m, l, H, scale, False,
+                              0.0, 0)
+    # fp32 oracle grads
+    qf = _heads(q.detach(), H).float().requires_grad_(True)
+    kf = _heads(k.detach(), H).float().requires_grad_(True)
+    vf = _heads(v.detach(), H).float().requires_grad_(True)
+    sf = qf @ kf.transpose(-1, -2) * scale
+    pf = torch.nn.functional.softmax(sf, dim=-1)
+    of = (pf @ vf).transpose(1, 2).reshape(B, Tq, H * D)
+    of.backward(dout.float())
+    dq_ref = qf.grad.transpose(1, 2).reshape(B, Tq, H * D)
+    dk_ref = kf.grad.transpose(1, 2).reshape(B, Tk, H * D)
+    dv_ref = vf.grad.transpose(1, 2).reshape(B, Tk, H * D)
+    assert_close(dq, dq_ref, rtol=6e-2, atol=6e-2, msg="cross dq")
+    assert_close(dk, dk_ref, rtol=6e-2, atol=6e-2, msg="cross dk")
+    assert_close(dv, dv_ref, rtol=6e-2, atol=6e-2, msg="cross dv")
+
+
+@pytest.mark.parametrize("cfg", [
+    (1, 2, 200, 200, True),    # causal, non-%128
+    (2, 2, 100, 100, False),
+    (1, 3, 160, 272, False),   # cross-attention with tails
+    (2, 1, 64, 512, False),
+])
+def test_flash_tail_lengths(cfg):
+    """Sequence lengths not % 128: tail blocks clamp staging and mask
+    scores; fwd + bwd vs the fp32 oracle."""
+    B, H, Tq, Tk, causal = cfg
+    D = 64
+    scale = 1.0 / (D ** 0.5)
+    q = _rt((B, Tq, H * D)).to(BF)
+    k = _rt((B, Tk, H * D)).to(BF)
+    v = _rt((B, Tk, H * D)).to(BF)
+    out, m, l = _C.flash_fwd(q.contiguous(), k.contiguous(),
+                             v.contiguous(), H, scale, causal, 0.0, 0)
+    qh, kh, vh = (_heads(x, H).float() for x in (q, k, v))
+    s = qh @ kh.transpose(-1, -2) * scale
+    if causal:
+        cm = torch.triu(torch.ones(Tq, Tk, device=_dev(),
+                                   dtype=torch.bool), 1)
+        s = s.masked_fill(cm, float("-inf"))
+    p = torch.nn.functional.softmax(s, dim=-1)
+    ref = (p @ vh).transpose(1, 2).reshape(B, Tq, H * D)
+    assert_close(out, ref, rtol=5e-2, atol=5e-2, msg=f"tail fwd {cfg}")
+
+    dout = _rt((B, Tq, H * D)).to(BF)
+    dq, dk, dv = _C.flash_bwd(q.contiguous(), k.contiguous(),
+                              v.contiguous(), out.contiguous(),
+                              dout.contiguous(), m, l, H, scale, causal,
+                              0.0, 0)
+    qf = qh.clone().requires_grad_(True)
+    kf = kh.clone().requires_grad_(True)
+    vf = vh.clone().requires_grad_(True)
+    sf = qf @ kf.transpose(-1, -2) * scale
+    if causal:
+        sf = sf.masked_fill(cm, float("-inf"))
+    pf = torch.nn.functional.softmax(sf, dim=-1)
+    (pf @ vf).transpose(1, 2).reshape(B, Tq, H * D).backward(dout.float())
+    assert_close(dq, qf.grad.transpose(1, 2).reshape(B, Tq, H * D),
+                 rtol=7e-2, atol=7e-2, msg=f"tail dq {cfg}")
+    assert_close(dk, kf.grad.transpose(1, 2).reshape(B, Tk, H * D),
+                 rtol=7e-2, atol=7e-2, msg=f"tail dk {cfg}")
+    assert_close(dv, vf.grad.transpose(1, 2).reshape(B, Tk, H * D),
+                 rtol=7e-2, atol=7e-2, msg=f"tail dv {cfg}")
