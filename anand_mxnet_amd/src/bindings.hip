@@ -1254,6 +1254,68 @@ long engine_new_stream(long device) {
 // through the embedded interpreter.  Thread-local error string via
 // MXAGetLastError (reference MXGetLastError contract).
 // ---------------------------------------------------------------------------
+// ---- raw hipGraph step capture (ROADMAP #7: torch's CUDAGraph wrapper
+// segfaults on this torch/ROCm combo; this is the direct HIP path).
+// Capture must run on a NON-default stream (python wraps the step in
+// torch.cuda.stream(side_stream)); replay launches the instantiated exec
+// on the current stream.  Allocator stability contract: capture in steady
+// state so every tensor the captured kernels touch stays alive/stable.
+static std::unordered_map<long, hipGraphExec_t> g_graph_execs;
+static long g_graph_next = 1;
+
+long graph_begin() {
+  hipStream_t s = cur_stream();
+  // relaxed: tolerate event queries from the allocator/autograd engine
+  hipError_t e = hipStreamBeginCapture(s, hipStreamCaptureModeRelaxed);
+  TORCH_CHECK(e == hipSuccess, "hipStreamBeginCapture: ",
+              hipGetErrorString(e));
+  return 0;
+}
+
+void graph_abort() {
+  // recover a stream stuck in capture after a failed step
+  hipStream_t s = cur_stream();
+  hipStreamCaptureStatus st;
+  if (hipStreamIsCapturing(s, &st) == hipSuccess &&
+      st != hipStreamCaptureStatusNone) {
+    hipGraph_t g = nullptr;
+    hipStreamEndCapture(s, &g);
+    if (g) hipGraphDestroy(g);
+  }
+  (void)hipGetLastError();  // clear sticky error
+}
+
+long graph_end() {
+  hipStream_t s = cur_stream();
+  hipGraph_t graph = nullptr;
+  hipError_t e = hipStreamEndCapture(s, &graph);
+  TORCH_CHECK(e == hipSuccess, "hipStreamEndCapture: ",
+              hipGetErrorString(e));
+  hipGraphExec_t exec = nullptr;
+  e = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+  hipGraphDestroy(graph);
+  TORCH_CHECK(e == hipSuccess, "hipGraphInstantiate: ",
+              hipGetErrorString(e));
+  long h = g_graph_next++;
+  g_graph_execs[h] = exec;
+  return h;
+}
+
+void graph_launch(long h) {
+  auto it = g_graph_execs.find(h);
+  TORCH_CHECK(it != g_graph_execs.end(), "unknown graph handle");
+  hipError_t e = hipGraphLaunch(it->second, cur_stream());
+  TORCH_CHECK(e == hipSuccess, "hipGraphLaunch: ", hipGetErrorString(e));
+}
+
+void graph_free(long h) {
+  auto it = g_graph_execs.find(h);
+  if (it != g_graph_execs.end()) {
+    hipGraphExecDestroy(it->second);
+    g_graph_execs.erase(it);
+  }
+}
+
 static thread_local std::string mxa_last_error;
 
 extern "C" {
@@ -1427,6 +1489,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("dep_wait_all", &mxa_engine::engine_wait_all,
         py::call_guard<py::gil_scoped_release>());
+  m.def("graph_begin", &graph_begin);
+  m.def("graph_abort", &graph_abort);
+  m.def("graph_end", &graph_end);
+  m.def("graph_launch", &graph_launch);
+  m.def("graph_free", &graph_free);
   m.def("dep_engine_type",
         []() { return std::string(mxa_engine::engine_type()); });
   m.def("engine_new_event", &engine_new_event);

@@ -30,9 +30,16 @@ def main():
     ap.add_argument("--model", default="resnet50_v1_5")
     ap.add_argument("--image-size", type=int, default=224)
     ap.add_argument("--capture", action="store_true",
-                    help="hipGraph whole-step capture (experimental: "
-                         "crashes under rocprofv3 on this stack)")
+                    help="raw hipGraph whole-step capture (implies "
+                         "MXA_FLAT_GRADS=1; measured ~parity with eager "
+                         "— the step is GPU-bound)")
     args = ap.parse_args()
+    if args.capture:
+        # persistent flat grad buckets: stable pointers and no per-step
+        # allocator frees — frees during capture emit event ops that HIP
+        # rejects (hipErrorStreamCaptureUnsupported); with the write-
+        # semantics default the capture fails and falls back to eager
+        os.environ.setdefault("MXA_FLAT_GRADS", "1")
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -94,29 +101,49 @@ def main():
         trainer.step(global_batch)
         return loss
 
-    # warmup
-    for _ in range(args.warmup):
-        loss = step()
+    # warmup — under --capture run EVERYTHING on one side stream so the
+    # autograd AccumulateGrad nodes live on the captured stream
+    cap_stream = torch.cuda.Stream() if args.capture else None
+    import contextlib
+    stream_ctx = (lambda: torch.cuda.stream(cap_stream)) if cap_stream \
+        else contextlib.nullcontext
+    with stream_ctx():
+        for _ in range(args.warmup):
+            loss = step()
     torch.cuda.synchronize()
 
     # hipGraph step capture (the CachedOp/static_alloc fast path: one graph
-    # replay per step instead of per-op launches).  Graph capture composes
-    # with RCCL collectives; fall back to eager if capture fails.
+    # replay per step instead of per-op launches).  Raw hipStream capture
+    # (ROADMAP #7: torch's CUDAGraph wrapper segfaults on this stack);
+    # requires a steady-state step so every captured pointer stays stable.
     graph = None
     if args.capture:
+        from anand_mxnet_amd import _C
         try:
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                loss = step()
-            g.replay()
+            loss = None  # drop the old autograd graph before capturing
             torch.cuda.synchronize()
-            graph = g
+            with torch.cuda.stream(cap_stream):
+                _C.graph_begin()
+                step()
+                handle = _C.graph_end()
+                _C.graph_launch(handle)
+            torch.cuda.synchronize()
+
+            def replay():
+                with torch.cuda.stream(cap_stream):
+                    _C.graph_launch(handle)
+            graph = replay
         except Exception as e:
+            with torch.cuda.stream(cap_stream):
+                _C.graph_abort()
+            torch.cuda.synchronize()
             if rank == 0:
                 print(f"# graph capture unavailable ({e}); eager",
                       file=sys.stderr)
             graph = None
-    run_step = graph.replay if graph is not None else step
+    run_step = graph if graph is not None else step
+    if graph is not None:
+        step_for_loss = None  # loss tensor is stale under replay
     for _ in range(2):
         run_step()
     torch.cuda.synchronize()
@@ -142,7 +169,7 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    lval = float(loss.mean().asscalar())
+    lval = float(loss.mean().asscalar()) if loss is not None else -1.0
     samples_per_sec = global_batch * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
