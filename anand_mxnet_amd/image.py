@@ -34,9 +34,21 @@ def imdecode(buf, to_rgb=True, flag=1):
         if b[:6] == b"\x93NUMPY":
             import io as _io
             return array(np.load(_io.BytesIO(b)).astype(np.float32))
-        raise MXNetError("imdecode: JPEG/PNG need an image decoder; this "
-                         "offline build reads raw/npy records "
-                         "(docs/PARITY.md)")
+        if b[:2] == b"\xff\xd8":
+            # native baseline JPEG decoder (src/kernels/jpeg.hip)
+            from .base import lib
+            C = lib()
+            if C is not None and hasattr(C, "imdecode_jpeg"):
+                t = C.imdecode_jpeg(b)  # HWC uint8 RGB
+                a = t.numpy().astype(np.float32)
+                if not to_rgb:
+                    a = a[..., ::-1].copy()  # BGR (reference cv2 default)
+                if flag == 0:
+                    a = a.mean(axis=2, keepdims=True)
+                return array(a)
+        raise MXNetError("imdecode: unsupported image format; this build "
+                         "decodes baseline JPEG natively plus raw/npy "
+                         "records (docs/PARITY.md)")
     return array(np.asarray(buf, dtype=np.float32))
 
 

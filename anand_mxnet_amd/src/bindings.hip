@@ -1150,6 +1150,24 @@ at::Tensor multi_sumsq_run(const at::Tensor& blob, long ntens, long nchunks,
 
 // hipEvent pool: WaitForVar/cross-stream ordering primitives (reference
 // engine.h VarHandle semantics collapse to event waits under stream order)
+// ---- baseline JPEG decoder (jpeg.hip) ----
+namespace mxa_jpeg {
+void decode(const unsigned char*, size_t, std::vector<unsigned char>&,
+            int*, int*);
+}
+
+at::Tensor imdecode_jpeg(py::bytes data) {
+  std::string buf = data;  // copy; decoding reads it on the CPU
+  std::vector<unsigned char> rgb;
+  int w = 0, h = 0;
+  mxa_jpeg::decode((const unsigned char*)buf.data(), buf.size(), rgb, &w,
+                   &h);
+  auto out = at::empty({(long)h, (long)w, 3},
+                       at::TensorOptions().dtype(at::kByte));
+  std::memcpy(out.data_ptr<unsigned char>(), rgb.data(), rgb.size());
+  return out;
+}
+
 // ---- dependency engine (engine.hip: var-queue scheduler) ----
 namespace mxa_engine {
 uint64_t engine_new_var();
@@ -1489,6 +1507,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("dep_wait_all", &mxa_engine::engine_wait_all,
         py::call_guard<py::gil_scoped_release>());
+  m.def("imdecode_jpeg", &imdecode_jpeg, "baseline JPEG -> HWC uint8");
   m.def("graph_begin", &graph_begin);
   m.def("graph_abort", &graph_abort);
   m.def("graph_end", &graph_end);
