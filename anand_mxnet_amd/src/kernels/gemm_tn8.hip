@@ -187,6 +187,138 @@ __global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8_core(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 32x32x16 variant: same tile/ring/phase structure, but each phase is
+// 8 x v_mfma_f32_32x32x16_bf16 instead of 16 x 16x16x32 — half the MFMA
+// issue slots (PMC showed 48% SQ_WAIT_INST_ANY on the 16x16 kernel) and
+// the 32x32 shape's µbench ceiling is ~15% higher (2382 vs 2075 TF).
+// C/D layout: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
+// ---------------------------------------------------------------------------
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8x32_core(
+    Plain8Loader aload, Plain8Loader bload, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, long strideC,
+    int ksteps, long kstride) {
+  __shared__ __align__(16) bf16 smem[2][2][2][128 * 64];
+
+  const int ntn = (N + T8_BN - 1) / T8_BN;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * T8_BM, bn = (bid % ntn) * T8_BN;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+  const int w = t / WAVE, l = t % WAVE;
+
+  aload.init(bm, t, z);
+  bload.init(bn, t, z);
+
+  auto issue_half = [&](int issue_idx, int tile) {
+    int op = issue_idx & 1;
+    int half = issue_idx >> 1;
+    long koff = (long)tile * kstride;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[tile & 1][op][half]
+              [0] + seg);
+      const bf16* gp = op ? bload.src(p, koff) : aload.src(p, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+
+  f32x16 acc[4][2];  // [quadrant][m-tile]
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+#pragma unroll
+    for (int i = 0; i < 2; ++i) acc[q][i] = f32x16{};
+
+  const int l31 = l & 31, lhi = l >> 5;
+  const int wmq = (w >> 2) * 64;
+  const int wnq = (w & 3) * 32;
+
+  auto phase = [&](int tile, int qa, int qb, int q, int wait6) {
+    if (wait6 == 1)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if (wait6 == 0)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const bf16* As = smem[tile & 1][0][qa];
+    const bf16* Bs = smem[tile & 1][1][qb];
+    bf16x8 af[2][4], bfr[4];
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      int chunk = kk * 2 + lhi;  // k = kk*16 + (l>>5)*8 .. +8
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt)
+        af[mt][kk] = *(const bf16x8*)((const char*)As +
+                                      t8_lds_byte(wmq + mt * 32 + l31,
+                                                  chunk));
+      bfr[kk] = *(const bf16x8*)((const char*)Bs +
+                                 t8_lds_byte(wnq + l31, chunk));
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt)
+        acc[q][mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            af[mt][kk], bfr[kk], acc[q][mt], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  };
+
+#pragma unroll
+  for (int h = 0; h < 4; ++h) issue_half(h, 0);
+
+  for (int tile = 0; tile < ksteps; ++tile) {
+    int nxt = tile + 1;
+    bool pre = nxt < ksteps;
+    if (pre) issue_half(0, nxt);
+    phase(tile, 0, 0, 0, pre ? 1 : 0);
+    if (pre) issue_half(1, nxt);
+    phase(tile, 1, 0, 1, pre ? 1 : -1);
+    if (pre) issue_half(2, nxt);
+    phase(tile, 1, 1, 2, pre ? 1 : -1);
+    if (pre) issue_half(3, nxt);
+    phase(tile, 0, 1, 3, -1);
+  }
+
+  bf16* cz = Cout + (long)z * strideC;
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    int qa = (q == 1 || q == 2) ? 1 : 0;
+    int qb = (q >= 2) ? 1 : 0;
+    int gcol = bn + qb * 128 + wnq + l31;
+    if (gcol >= N) continue;
+    float bv = BIAS ? bias[gcol] : 0.f;
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int grow = bm + qa * 128 + wmq + mt * 32 + (reg & 3) +
+                   8 * (reg >> 2) + 4 * lhi;
+        if (grow >= M) continue;
+        float v = acc[q][mt][reg] + bv;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        cz[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
+static bool tn8_use32() {
+  static const bool v = [] {
+    const char* e = getenv("MXA_TN8_32");
+    return e && e[0] == '1';
+  }();
+  return v;
+}
+
 void gemm_tn8_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
                    int M, int N, int K, long lda, long ldb, long ldc,
                    long strideA, long strideB, long strideC, int batches,
@@ -195,6 +327,23 @@ void gemm_tn8_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
   Plain8Loader b{B, zpage, N, ldb, strideB};
   dim3 grid(cdiv_h(M, T8_BM) * cdiv_h(N, T8_BN), 1, batches);
   int ksteps = K / T8_BK;
+  if (tn8_use32()) {
+    if (bias) {
+      if (relu)
+        gemm_tn8x32_core<true, true><<<grid, T8_THREADS, 0, stream>>>(
+            a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+      else
+        gemm_tn8x32_core<false, true><<<grid, T8_THREADS, 0, stream>>>(
+            a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    } else if (relu) {
+      gemm_tn8x32_core<true, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    } else {
+      gemm_tn8x32_core<false, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    }
+    return;
+  }
   if (bias) {
     if (relu)
       gemm_tn8_core<true, true><<<grid, T8_THREADS, 0, stream>>>(
