@@ -54,8 +54,32 @@ _UNARY = {
     "gammaln": torch.lgamma,
     "logical_not": lambda x: (x == 0).to(x.dtype),
 }
+def _ew(name, *ts):
+    """Generic elementwise engine dispatch (ops/hip ew.hip) for GPU bf16
+    tensors — the framework-owned long-tail surface; returns None when the
+    native path does not apply (CPU oracle / substrate runs instead)."""
+    t0 = ts[0]
+    if not (isinstance(t0, torch.Tensor) and t0.is_cuda):
+        return None
+    try:
+        from . import hip as _h
+    except Exception:
+        return None
+    if len(ts) == 1:
+        return _h.ew_unary_dispatch(name, t0)
+    return _h.ew_binary_dispatch(name, ts[0], ts[1])
+
+
+def _make_unary_op(name, fn):
+    def op(x):
+        r = _ew(name, x)
+        return r if r is not None else fn(x)
+    op.__name__ = name
+    return op
+
+
 for _name, _fn in _UNARY.items():
-    register(_name)(_fn)
+    register(_name)(_make_unary_op(_name, _fn))
 
 
 @register("softsign")
@@ -125,8 +149,19 @@ _BINARY = {
     "broadcast_hypot": torch.hypot,
     "arctan2": torch.atan2,
 }
+def _make_binary_op(name, fn):
+    def op(a, b):
+        if isinstance(a, torch.Tensor) and isinstance(b, torch.Tensor):
+            r = _ew(name, a, b)
+            if r is not None:
+                return r
+        return fn(a, b)
+    op.__name__ = name
+    return op
+
+
 for _name, _fn in _BINARY.items():
-    register(_name)(_fn)
+    register(_name)(_make_binary_op(_name, _fn))
 
 for _name, _cmp in [("broadcast_equal", torch.eq), ("broadcast_not_equal", torch.ne),
                     ("broadcast_greater", torch.gt), ("broadcast_greater_equal", torch.ge),
@@ -174,6 +209,17 @@ def sum_(x, axis=None, keepdims=False, exclude=False):
     if exclude and ax is not None:
         ax = tuple(i for i in range(x.dim()) if i not in
                    tuple(a % x.dim() for a in ax))
+    if (ax is not None and len(ax) == 1 and isinstance(x, torch.Tensor)
+            and x.dim() >= 1 and ax[0] % x.dim() == x.dim() - 1
+            and not keepdims and x.is_cuda
+            and x.dtype == torch.bfloat16):
+        try:
+            from . import hip as _h
+            r = _h.ew_sum_last(x.contiguous())
+            if r is not None:
+                return r
+        except Exception:
+            pass
     if ax is None:
         out = x.sum()
         return out.reshape((1,) * x.dim()) if keepdims else out

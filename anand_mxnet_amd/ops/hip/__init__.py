@@ -976,3 +976,157 @@ def fp8_linear(x, w, bias):
     out = _Fp8Linear.apply(x2.contiguous(), w,
                            bias if bias is None else bias.to(x.dtype))
     return out.reshape(x.shape[:-1] + (w.shape[0],))
+
+
+# ---------------------------------------------------------------------------
+# Generic elementwise engine (ew.hip) — the framework-owned long-tail op
+# surface (reference mxnet_op.h Kernel<OP,gpu>::Launch).  core.py consults
+# these tables for GPU bf16 tensors; everything else stays on the CPU
+# oracle / torch substrate.
+# ---------------------------------------------------------------------------
+
+EW_UNARY = {
+    "relu": 0, "sigmoid": 1, "tanh": 2, "exp": 3, "log": 4, "sqrt": 5,
+    "rsqrt": 6, "square": 7, "abs": 8, "negative": 9, "reciprocal": 10,
+    "sin": 11, "cos": 12, "erf": 13, "log1p": 14, "expm1": 15,
+    "floor": 16, "ceil": 17, "round": 18, "sign": 19, "softsign": 20,
+}
+
+EW_BINARY = {"elemwise_add": 0, "broadcast_add": 0, "elemwise_sub": 1,
+             "broadcast_sub": 1, "elemwise_mul": 2, "broadcast_mul": 2,
+             "elemwise_div": 3, "broadcast_div": 3, "broadcast_power": 4,
+             "power": 4, "broadcast_maximum": 5, "maximum": 5,
+             "broadcast_minimum": 6, "minimum": 6}
+
+
+class _EwUnary(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, opid):
+        y = _C.ew_unary(opid, x.contiguous())
+        ctx.save_for_backward(x, y)
+        ctx.opid = opid
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y = ctx.saved_tensors
+        return _C.ew_unary_grad(ctx.opid, x.contiguous(), y,
+                                dy.contiguous()), None
+
+
+class _EwBinary(torch.autograd.Function):
+    """Same-shape binary with analytic backwards on the same engine.
+    Saves only the per-op backward operands (add/sub save NOTHING — the
+    residual adds of every model go through here)."""
+
+    @staticmethod
+    def forward(ctx, a, b, opid):
+        y = _C.ew_binary(opid, a.contiguous(), b.contiguous())
+        ctx.opid = opid
+        if opid in (0, 1):
+            ctx.save_for_backward()
+        elif opid == 2:
+            ctx.save_for_backward(a, b)
+        elif opid == 3:
+            ctx.save_for_backward(b, y)
+        else:
+            ctx.save_for_backward(a, b, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        saved = ctx.saved_tensors
+        a = b = y = None
+        if ctx.opid == 2:
+            a, b = saved
+        elif ctx.opid == 3:
+            b, y = saved
+        elif ctx.opid in (4, 5, 6):
+            a, b, y = saved
+        dy = dy.contiguous()
+        op = ctx.opid
+        da = db = None
+        need_a, need_b = ctx.needs_input_grad[0], ctx.needs_input_grad[1]
+        if op == 0:    # add
+            da = dy if need_a else None
+            db = dy if need_b else None
+        elif op == 1:  # sub
+            da = dy if need_a else None
+            db = _C.ew_unary(EW_UNARY["negative"], dy) if need_b else None
+        elif op == 2:  # mul
+            da = _C.ew_binary(2, dy, b.contiguous()) if need_a else None
+            db = _C.ew_binary(2, dy, a.contiguous()) if need_b else None
+        elif op == 3:  # div
+            if need_a:
+                da = _C.ew_binary(3, dy, b.contiguous())
+            if need_b:
+                db = _C.ew_binary(
+                    2, _C.ew_unary(EW_UNARY["negative"], dy),
+                    _C.ew_binary(3, y, b.contiguous()))
+        elif op == 4:  # pow: da = dy * b * a^(b-1); db = dy * y * ln a
+            if need_a:
+                da = _C.ew_binary(
+                    2, dy, _C.ew_binary(
+                        2, b.contiguous(),
+                        _C.ew_binary(3, y, a.contiguous())))
+            if need_b:
+                db = _C.ew_binary(
+                    2, dy, _C.ew_binary(
+                        2, y, _C.ew_unary(EW_UNARY["log"],
+                                          a.contiguous())))
+        elif op in (5, 6):  # max/min: grad to the winning side
+            amask = (a >= b) if op == 5 else (a <= b)
+            if need_a:
+                da = dy * amask.to(dy.dtype)
+            if need_b:
+                db = dy * (~amask).to(dy.dtype)
+        return da, db, None
+
+
+class _EwReduceLast(torch.autograd.Function):
+    """sum over the last axis (fp32 accumulation), bf16 out."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ctx.D = x.shape[-1]
+        y = _C.ew_reduce_last(0, x.contiguous())
+        return y.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return dy.unsqueeze(-1).expand(dy.shape + (ctx.D,)).contiguous()
+
+
+def ew_unary_dispatch(name, x):
+    """Native launch when eligible, else None (caller falls back)."""
+    opid = EW_UNARY.get(name)
+    if (opid is None or not isinstance(x, torch.Tensor) or not x.is_cuda
+            or x.dtype != torch.bfloat16):
+        return None
+    return _EwUnary.apply(x, opid)
+
+
+def ew_binary_dispatch(name, a, b):
+    opid = EW_BINARY.get(name)
+    if (opid is None or not isinstance(a, torch.Tensor)
+            or not isinstance(b, torch.Tensor) or not a.is_cuda
+            or not b.is_cuda or a.dtype != torch.bfloat16
+            or b.dtype != torch.bfloat16):
+        return None
+    if a.shape == b.shape:
+        return _EwBinary.apply(a, b, opid)
+    # broadcast fwd runs native when no grad is needed; the backward of a
+    # broadcast needs reduction over the broadcast dims (substrate)
+    if not torch.is_grad_enabled() or (not a.requires_grad
+                                       and not b.requires_grad):
+        if a.dim() <= 6 and b.dim() <= 6:
+            return _C.ew_binary(opid, a.contiguous(), b.contiguous())
+    return None
+
+
+def ew_sum_last(x):
+    if (not isinstance(x, torch.Tensor) or not x.is_cuda
+            or x.dtype != torch.bfloat16 or x.dim() < 1
+            or not x.is_contiguous()):
+        return None
+    return _EwReduceLast.apply(x)

@@ -142,6 +142,14 @@ void gemm_mx_bf16(const unsigned char*, const unsigned char*,
                   const float*, int, int, int, long, int, int, int, float*,
                   const unsigned char*, hipStream_t);
 int gemm_mx_nslots(int, int, int, int, int);
+// ew.hip (generic elementwise/broadcast/reduce launcher)
+void ew_unary(int, const bf16*, bf16*, long, hipStream_t);
+void ew_unary_grad(int, const bf16*, const bf16*, const bf16*, bf16*, long,
+                   hipStream_t);
+void ew_binary(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
+void ew_binary_bcast(int, const bf16*, const bf16*, bf16*, int, const long*,
+                     const long*, const long*, long, hipStream_t);
+void ew_reduce_last(int, const bf16*, float*, long, long, hipStream_t);
 void gemm_tn_i8(const unsigned char*, const unsigned char*, int*, bf16*,
                 const float*, const float*, int, int, int, long, long, long,
                 const unsigned char*, hipStream_t);
@@ -429,6 +437,69 @@ at::Tensor gemm_i8(const at::Tensor& A8, const at::Tensor& B8,
                   (int)M, (int)N, (int)K, K, K, N,
                   (const unsigned char*)zp(zprobe), cur_stream());
   return out;
+}
+
+// ---- generic elementwise launcher (ew.hip) ----
+at::Tensor ew_unary_rt(long op, const at::Tensor& x) {
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_contiguous());
+  auto y = at::empty_like(x);
+  mxa::ew_unary((int)op, bfp(x), bfp_mut(y), x.numel(), cur_stream());
+  return y;
+}
+
+at::Tensor ew_unary_grad_rt(long op, const at::Tensor& x,
+                            const at::Tensor& y, const at::Tensor& dy) {
+  auto dx = at::empty_like(x);
+  mxa::ew_unary_grad((int)op, bfp(x), bfp(y), bfp(dy), bfp_mut(dx),
+                     x.numel(), cur_stream());
+  return dx;
+}
+
+at::Tensor ew_binary_rt(long op, const at::Tensor& a, const at::Tensor& b) {
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 &&
+              b.scalar_type() == at::kBFloat16);
+  if (a.sizes() == b.sizes() && a.is_contiguous() && b.is_contiguous()) {
+    auto y = at::empty_like(a);
+    mxa::ew_binary((int)op, bfp(a), bfp(b), bfp_mut(y), a.numel(),
+                   cur_stream());
+    return y;
+  }
+  // broadcast: align shapes right, size-1 dims get stride 0
+  auto out_sizes = at::infer_size(a.sizes(), b.sizes());
+  int nd = (int)out_sizes.size();
+  TORCH_CHECK(nd <= 6, "ew_binary: ndim > 6");
+  long shape[6], sa[6], sb[6];
+  auto aa = a, bb = b;
+  while (aa.dim() < nd) aa = aa.unsqueeze(0);
+  while (bb.dim() < nd) bb = bb.unsqueeze(0);
+  aa = aa.contiguous();
+  bb = bb.contiguous();
+  long n = 1;
+  for (int d = 0; d < nd; ++d) {
+    shape[d] = out_sizes[d];
+    n *= out_sizes[d];
+    sa[d] = aa.size(d) == 1 ? 0 : aa.stride(d);
+    sb[d] = bb.size(d) == 1 ? 0 : bb.stride(d);
+    TORCH_CHECK(aa.size(d) == 1 || aa.size(d) == out_sizes[d]);
+    TORCH_CHECK(bb.size(d) == 1 || bb.size(d) == out_sizes[d]);
+  }
+  auto y = at::empty(out_sizes, a.options());
+  mxa::ew_binary_bcast((int)op, bfp(aa), bfp(bb), bfp_mut(y), nd, shape,
+                       sa, sb, n, cur_stream());
+  return y;
+}
+
+at::Tensor ew_reduce_last_rt(long op, const at::Tensor& x) {
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_contiguous() &&
+              x.dim() >= 1);
+  long D = x.size(-1);
+  long rows = x.numel() / D;
+  auto out = at::empty({rows}, x.options().dtype(at::kFloat));
+  mxa::ew_reduce_last((int)op, bfp(x), out.data_ptr<float>(), rows, D,
+                      cur_stream());
+  auto shp = x.sizes().vec();
+  shp.pop_back();
+  return shp.empty() ? out.reshape({}) : out.reshape(shp);
 }
 
 at::Tensor cast_to_fp8(const at::Tensor& x, const at::Tensor& scale) {
@@ -1437,6 +1508,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("transpose") = false);
   m.def("gemm_tn_fp8", &gemm_tn_fp8);
   m.def("cast_to_fp8", &cast_to_fp8);
+  m.def("ew_unary", &ew_unary_rt);
+  m.def("ew_unary_grad", &ew_unary_grad_rt);
+  m.def("ew_binary", &ew_binary_rt);
+  m.def("ew_reduce_last", &ew_reduce_last_rt);
   m.def("mx_cast", &mx_cast, py::arg("x"), py::arg("e5") = false);
   m.def("gemm_i8", &gemm_i8, py::arg("A8"), py::arg("B8"),
         py::arg("inv_scale") = py::none(), py::arg("bias") = py::none());

@@ -1014,3 +1014,88 @@ def test_flash_tail_lengths(cfg):
                  rtol=7e-2, atol=7e-2, msg=f"tail dk {cfg}")
     assert_close(dv, vf.grad.transpose(1, 2).reshape(B, Tk, H * D),
                  rtol=7e-2, atol=7e-2, msg=f"tail dv {cfg}")
+
+
+def test_ew_unary_family_vs_torch():
+    """Generic elementwise engine (ew.hip): every unary functor fwd+bwd
+    vs torch fp32 (reference mxnet_op.h Kernel::Launch surface)."""
+    from anand_mxnet_amd.ops.hip import EW_UNARY, _EwUnary
+    torch.manual_seed(1)
+    refs = {
+        "relu": torch.relu, "sigmoid": torch.sigmoid, "tanh": torch.tanh,
+        "exp": torch.exp, "log": torch.log, "sqrt": torch.sqrt,
+        "rsqrt": torch.rsqrt, "square": torch.square, "abs": torch.abs,
+        "negative": torch.neg, "reciprocal": torch.reciprocal,
+        "sin": torch.sin, "cos": torch.cos, "erf": torch.erf,
+        "log1p": torch.log1p, "expm1": torch.expm1, "floor": torch.floor,
+        "ceil": torch.ceil, "round": torch.round, "sign": torch.sign,
+        "softsign": lambda t: t / (1 + t.abs()),
+    }
+    for name, opid in EW_UNARY.items():
+        pos_only = name in ("log", "sqrt", "rsqrt", "log1p", "reciprocal")
+        x = torch.rand(1000, device=_dev()) * 2 + 0.1 if pos_only else \
+            (torch.rand(1000, device=_dev()) * 4 - 2)
+        xb = x.to(BF)
+        y = _C.ew_unary(opid, xb.contiguous())
+        ref = refs[name](xb.float())
+        assert_close(y, ref, rtol=2e-2, atol=2e-2, msg=f"ew {name}")
+        # gradient (skip zero-grad step functions)
+        if name in ("floor", "ceil", "round", "sign"):
+            continue
+        xg = xb.clone().requires_grad_(True)
+        yy = _EwUnary.apply(xg, opid)
+        dy = torch.rand_like(yy)
+        yy.backward(dy)
+        xr = xb.float().clone().requires_grad_(True)
+        refs[name](xr).backward(dy.float())
+        assert_close(xg.grad, xr.grad, rtol=4e-2, atol=4e-2,
+                     msg=f"ew grad {name}")
+
+
+def test_ew_binary_and_broadcast():
+    from anand_mxnet_amd.ops.hip import _EwBinary
+    torch.manual_seed(2)
+    a = (torch.rand(64, 33, device=_dev()) * 2 - 1).to(BF)
+    b = (torch.rand(64, 33, device=_dev()) * 2 + 0.5).to(BF)
+    for opid, fn in [(0, torch.add), (1, torch.sub), (2, torch.mul),
+                     (3, torch.div), (5, torch.maximum),
+                     (6, torch.minimum)]:
+        y = _C.ew_binary(opid, a.contiguous(), b.contiguous())
+        assert_close(y, fn(a.float(), b.float()), rtol=2e-2, atol=2e-2,
+                     msg=f"ew bin {opid}")
+        # autograd path
+        ag = a.clone().requires_grad_(True)
+        bg = b.clone().requires_grad_(True)
+        yy = _EwBinary.apply(ag, bg, opid)
+        dy = torch.rand_like(yy)
+        yy.backward(dy)
+        ar = a.float().clone().requires_grad_(True)
+        br = b.float().clone().requires_grad_(True)
+        fn(ar, br).backward(dy.float())
+        assert_close(ag.grad, ar.grad, rtol=4e-2, atol=4e-2,
+                     msg=f"ew bin da {opid}")
+        assert_close(bg.grad, br.grad, rtol=4e-2, atol=4e-2,
+                     msg=f"ew bin db {opid}")
+    # broadcast forward (inference)
+    with torch.no_grad():
+        c = (torch.rand(1, 33, device=_dev()) * 2 - 1).to(BF)
+        y = _C.ew_binary(2, a.contiguous(), c.contiguous())
+        assert_close(y, a.float() * c.float(), rtol=2e-2, atol=2e-2,
+                     msg="ew bcast mul")
+        d = (torch.rand(64, 1, device=_dev())).to(BF)
+        y = _C.ew_binary(0, a.contiguous(), d.contiguous())
+        assert_close(y, a.float() + d.float(), rtol=2e-2, atol=2e-2,
+                     msg="ew bcast add")
+
+
+def test_ew_reduce_last():
+    torch.manual_seed(3)
+    x = (torch.rand(37, 515, device=_dev()) * 2 - 1).to(BF)
+    s = _C.ew_reduce_last(0, x.contiguous())
+    assert_close(s, x.float().sum(-1), rtol=2e-2, atol=5e-2, msg="ew sum")
+    m = _C.ew_reduce_last(1, x.contiguous())
+    assert_close(m, x.float().max(-1).values, rtol=1e-4, atol=1e-4,
+                 msg="ew max")
+    mn = _C.ew_reduce_last(2, x.contiguous())
+    assert_close(mn, x.float().min(-1).values, rtol=1e-4, atol=1e-4,
+                 msg="ew min")
