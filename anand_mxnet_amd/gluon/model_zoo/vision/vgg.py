@@ -8,9 +8,10 @@ __all__ = ["VGG", "vgg11", "vgg13", "vgg16", "vgg19", "vgg11_bn", "vgg13_bn",
 
 class VGG(HybridBlock):
     def __init__(self, layers, filters, classes=1000, batch_norm=False,
-                 **kwargs):
+                 layout="NCHW", **kwargs):
         super().__init__(**kwargs)
         assert len(layers) == len(filters)
+        self._layout = layout
         with self.name_scope():
             self.features = self._make_features(layers, filters, batch_norm)
             self.features.add(nn.Dense(4096, activation="relu",
@@ -27,11 +28,13 @@ class VGG(HybridBlock):
             for _ in range(num):
                 featurizer.add(nn.Conv2D(filters[i], kernel_size=3,
                                          padding=1,
-                                         weight_initializer="xavier"))
+                                         weight_initializer="xavier",
+                                         layout=self._layout))
                 if batch_norm:
-                    featurizer.add(nn.BatchNorm())
+                    featurizer.add(nn.BatchNorm(
+                        axis=3 if self._layout == "NHWC" else 1))
                 featurizer.add(nn.Activation("relu"))
-            featurizer.add(nn.MaxPool2D(strides=2))
+            featurizer.add(nn.MaxPool2D(strides=2, layout=self._layout))
         return featurizer
 
     def forward(self, x):

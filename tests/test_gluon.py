@@ -259,3 +259,35 @@ def test_new_layers_save_load_roundtrip(tmp_path):
     net2.load_parameters(f)
     y2 = net2(x)
     np.testing.assert_allclose(y.asnumpy(), y2.asnumpy(), rtol=1e-6)
+
+
+def test_model_zoo_nhwc_layouts():
+    """Every CNN zoo family except inception builds and runs in NHWC (the
+    MI355X performance layout) as well as NCHW."""
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import nd
+    from anand_mxnet_amd.gluon.model_zoo import vision as zoo
+    factories = [
+        ("resnet18_v1", 64), ("vgg11", 64), ("alexnet", 224),
+        ("mobilenet0_25", 64), ("squeezenet1_1", 64),
+    ]
+    for name, size in factories:
+        for layout in ("NCHW", "NHWC"):
+            net = getattr(zoo, name)(layout=layout)
+            net.initialize(mx.init.Xavier())
+            shape = (1, 3, size, size) if layout == "NCHW" \
+                else (1, size, size, 3)
+            y = net(nd.array(np.random.rand(*shape).astype("float32")))
+            assert y.shape == (1, 1000), (name, layout)
+
+
+def test_densenet_nhwc():
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import nd
+    from anand_mxnet_amd.gluon.model_zoo.vision.densenet import DenseNet
+    net = DenseNet(16, 8, [2, 2], classes=10, layout="NHWC")
+    net.initialize(mx.init.Xavier())
+    y = net(nd.array(np.random.rand(1, 56, 56, 3).astype("float32")))
+    assert y.shape == (1, 10)
