@@ -45,9 +45,16 @@ def score(name, factory, batch_sizes, size=224, layout="NHWC"):
 def main():
     score("resnet50_v1.5", lambda: zoo.resnet50_v1_5(layout="NHWC"),
           [1, 32, 128, 256])
+    score("resnet18_v1", lambda: zoo.resnet18_v1(layout="NHWC"), [1, 64])
     score("resnet18_v1 (NCHW)", lambda: zoo.resnet18_v1(), [1, 64],
           layout="NCHW")
+    score("vgg11", lambda: zoo.vgg11(layout="NHWC"), [1, 64])
     score("vgg11 (NCHW)", lambda: zoo.vgg11(), [1, 64], layout="NCHW")
+    score("mobilenet1_0", lambda: zoo.mobilenet1_0(layout="NHWC"),
+          [1, 64])
+    score("squeezenet1_1", lambda: zoo.squeezenet1_1(layout="NHWC"),
+          [1, 64])
+    score("densenet121", lambda: zoo.densenet121(layout="NHWC"), [64])
 
 
 if __name__ == "__main__":
