@@ -1243,9 +1243,12 @@ static int wg_split_target() {
 }
 
 static int wg_min_ksteps() {
+  // 48 since the slot-fold epilogue: fold traffic scales with the split
+  // count, so deeper splits-per-slot win (b512 sweep: mk=48 8169-8199 vs
+  // mk=24 8075-8170 samples/s across fill targets)
   static const int t = [] {
     const char* e = getenv("MXA_WG_MINK");
-    return e ? atoi(e) : 24;
+    return e ? atoi(e) : 48;
   }();
   return t;
 }
