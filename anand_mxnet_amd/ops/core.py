@@ -561,6 +561,13 @@ def take(a, indices, axis=0, mode="clip"):
         idx = idx.clamp(0, a.shape[axis] - 1)
     elif mode == "wrap":
         idx = idx % a.shape[axis]
+    if (axis == 0 and a.dim() == 2 and isinstance(a, torch.Tensor)
+            and a.is_cuda and a.dtype == torch.bfloat16):
+        # row gather == the embedding kernel (fwd gather + scatter-add
+        # grad); reference indexing_op.h Take/AddTakeGrad
+        from . import nn as _nn
+        out = _nn.embedding_impl(idx.reshape(-1), a)
+        return out.reshape(tuple(indices.shape) + (a.shape[1],))
     return torch.index_select(a, axis, idx.flatten()).reshape(
         a.shape[:axis] + tuple(indices.shape) + a.shape[axis + 1:])
 
@@ -653,14 +660,43 @@ def boolean_mask(data, index, axis=0):
 
 
 # ---------------- random ----------------
+_rng_salt = [7919]
+
+
+def _native_sample(shape, a, b, normal, dtype, device):
+    """Counter-hash sampling kernel (reference random/sample_op.* family)
+    for GPU bf16 requests; None -> torch substrate."""
+    if device is None or dtype_from_any(dtype) != torch.bfloat16:
+        return None
+    dev = torch.device(device) if not isinstance(device, torch.device)         else device
+    if dev.type != "cuda":
+        return None
+    try:
+        from ..base import lib
+        C = lib()
+        if C is None or not hasattr(C, "rng_sample"):
+            return None
+        _rng_salt[0] = (_rng_salt[0] * 2654435761 + 97) % (1 << 62)
+        return C.rng_sample(list(shape), float(a), float(b),
+                            _rng_salt[0], normal, dev.index or 0)
+    except Exception:
+        return None
+
+
 @register("random_uniform")
 def random_uniform(low=0.0, high=1.0, shape=(1,), dtype="float32", device=None):
+    r = _native_sample(shape, low, high, False, dtype, device)
+    if r is not None:
+        return r
     td = dtype_from_any(dtype)
     return torch.empty(tuple(shape), dtype=td, device=device).uniform_(low, high)
 
 
 @register("random_normal")
 def random_normal(loc=0.0, scale=1.0, shape=(1,), dtype="float32", device=None):
+    r = _native_sample(shape, loc, scale, True, dtype, device)
+    if r is not None:
+        return r
     td = dtype_from_any(dtype)
     return torch.empty(tuple(shape), dtype=td, device=device).normal_(loc, scale)
 

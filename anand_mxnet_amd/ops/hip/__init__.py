@@ -856,7 +856,7 @@ def mha(q, k, v, num_heads, causal, scale, dropout_p=0.0, training=False):
     B, Tq, HD = q.shape
     Tk = k.shape[1]
     D = HD // num_heads
-    if (_USE_FLASH and D == 64 and (Tq == Tk or not causal)
+    if (_USE_FLASH and D in (64, 128) and (Tq == Tk or not causal)
             and q.dtype == torch.bfloat16):
         # any sequence length (tail blocks are masked in-kernel) and
         # cross-attention Tq != Tk (decoder enc-dec attention) run fused;

@@ -122,14 +122,14 @@ void attn_wg_bf16(const bf16*, const bf16*, bf16*, int, int, int, int,
                   int, const bf16*, hipStream_t);
 // flash.hip
 void flash_fwd_bf16(const bf16*, const bf16*, const bf16*, bf16*, float*,
-                    float*, int, int, int, int, float, bool,
+                    float*, int, int, int, int, int, float, bool,
                     const unsigned long long*, unsigned long long, float,
                     hipStream_t);
 void flash_bwd_bf16(const bf16*, const bf16*, const bf16*, const bf16*,
                     const bf16*, const float*, const float*, float*,
-                    float*, bf16*, bf16*, int, int, int, int, float, bool,
-                    const unsigned long long*, unsigned long long, float,
-                    hipStream_t);
+                    float*, bf16*, bf16*, int, int, int, int, int, float,
+                    bool, const unsigned long long*, unsigned long long,
+                    float, hipStream_t);
 void act_fwd_bf16(int, const bf16*, bf16*, long, hipStream_t);
 void act_bwd_bf16(int, const bf16*, const bf16*, bf16*, long, hipStream_t);
 // gemm_fp8.hip
@@ -142,6 +142,10 @@ void gemm_mx_bf16(const unsigned char*, const unsigned char*,
                   const float*, int, int, int, long, int, int, int, float*,
                   const unsigned char*, hipStream_t);
 int gemm_mx_nslots(int, int, int, int, int);
+void rng_uniform_bf16(bf16*, long, const unsigned long long*,
+                      unsigned long long, float, float, hipStream_t);
+void rng_normal_bf16(bf16*, long, const unsigned long long*,
+                     unsigned long long, float, float, hipStream_t);
 // ew.hip (generic elementwise/broadcast/reduce launcher)
 void ew_unary(int, const bf16*, bf16*, long, hipStream_t);
 void ew_unary_grad(int, const bf16*, const bf16*, const bf16*, bf16*, long,
@@ -440,6 +444,31 @@ at::Tensor gemm_i8(const at::Tensor& A8, const at::Tensor& B8,
 }
 
 // ---- generic elementwise launcher (ew.hip) ----
+at::Tensor& rng_seed_for(const at::Tensor& like);  // defined below
+
+at::Tensor rng_sample(std::vector<long> shape, double a, double b,
+                      long salt, bool normal, long device) {
+  auto opts = at::TensorOptions()
+                  .device(at::Device(at::kCUDA, (int)device))
+                  .dtype(at::kBFloat16);
+  long n = 1;
+  for (long d : shape) n *= d;
+  long npad = (n + 7) / 8 * 8;
+  auto out = at::empty({npad}, opts);
+  auto& sw = rng_seed_for(out);
+  if (normal)
+    mxa::rng_normal_bf16(bfp_mut(out), npad,
+                         (const unsigned long long*)sw.data_ptr<long>(),
+                         (unsigned long long)salt, (float)a, (float)b,
+                         cur_stream());
+  else
+    mxa::rng_uniform_bf16(bfp_mut(out), npad,
+                          (const unsigned long long*)sw.data_ptr<long>(),
+                          (unsigned long long)salt, (float)a, (float)b,
+                          cur_stream());
+  return out.narrow(0, 0, n).reshape(shape);
+}
+
 at::Tensor ew_unary_rt(long op, const at::Tensor& x) {
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_contiguous());
   auto y = at::empty_like(x);
@@ -953,7 +982,8 @@ std::vector<at::Tensor> flash_fwd(const at::Tensor& q,
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   long B = q.size(0), Tq = q.size(1), HD = q.size(2), Tk = k.size(1);
   long D = HD / H;
-  TORCH_CHECK(D == 64 && Tq >= 1 && Tk >= 1, "flash_fwd: D==64 required");
+  TORCH_CHECK((D == 64 || D == 128) && Tq >= 1 && Tk >= 1,
+              "flash_fwd: D in {64, 128} required");
   auto out = at::empty_like(q);
   auto opts = q.options().dtype(at::kFloat);
   auto m = at::empty({B * H, Tq}, opts);
@@ -961,7 +991,8 @@ std::vector<at::Tensor> flash_fwd(const at::Tensor& q,
   auto& sw = rng_seed_for(q);
   mxa::flash_fwd_bf16(bfp(q), bfp(k), bfp(v), bfp_mut(out),
                       m.data_ptr<float>(), l.data_ptr<float>(), (int)B,
-                      (int)H, (int)Tq, (int)Tk, (float)alpha, causal,
+                      (int)H, (int)Tq, (int)Tk, (int)D, (float)alpha,
+                      causal,
                       (const unsigned long long*)sw.data_ptr<long>(),
                       (unsigned long long)salt, (float)pdrop,
                       cur_stream());
@@ -987,8 +1018,8 @@ std::vector<at::Tensor> flash_bwd(const at::Tensor& q,
                       bfp(dout.contiguous()), m.data_ptr<float>(),
                       l.data_ptr<float>(), drow.data_ptr<float>(),
                       dqacc.data_ptr<float>(), bfp_mut(dk), bfp_mut(dv),
-                      (int)B, (int)H, (int)Tq, (int)Tk, (float)alpha,
-                      causal,
+                      (int)B, (int)H, (int)Tq, (int)Tk, (int)(HD / H),
+                      (float)alpha, causal,
                       (const unsigned long long*)sw.data_ptr<long>(),
                       (unsigned long long)salt, (float)pdrop,
                       cur_stream());
@@ -1508,6 +1539,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("transpose") = false);
   m.def("gemm_tn_fp8", &gemm_tn_fp8);
   m.def("cast_to_fp8", &cast_to_fp8);
+  m.def("rng_sample", &rng_sample);
   m.def("ew_unary", &ew_unary_rt);
   m.def("ew_unary_grad", &ew_unary_grad_rt);
   m.def("ew_binary", &ew_binary_rt);

@@ -190,6 +190,58 @@ __global__ __launch_bounds__(256) void scale_cast_f32_bf16(
 // host launchers
 // ---------------------------------------------------------------------------
 
+// native sampling (reference src/operator/random/sample_op.* — philox-
+// style counter hash, same generator family as dropout)
+__global__ __launch_bounds__(256) void rng_uniform_kernel(
+    bf16* __restrict__ y, long nvec, const u64* __restrict__ seed_ptr,
+    u64 salt, float lo, float span) {
+  u64 seed = seed_finalize(*seed_ptr + salt);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      u32 r = hash_u32(seed, i * 8 + j);
+      o[j] = f2bf(lo + span * (float)r * 2.3283064365386963e-10f);
+    }
+    *(bf16x8*)&y[i * 8] = o;
+  }
+}
+
+__global__ __launch_bounds__(256) void rng_normal_kernel(
+    bf16* __restrict__ y, long nvec, const u64* __restrict__ seed_ptr,
+    u64 salt, float mu, float sigma) {
+  u64 seed = seed_finalize(*seed_ptr + salt);
+  const float TWO_PI = 6.28318530718f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; j += 2) {
+      u32 r1 = hash_u32(seed, (i * 8 + j) * 2);
+      u32 r2 = hash_u32(seed, (i * 8 + j) * 2 + 1);
+      float u1 = ((float)r1 + 1.0f) * 2.3283064365386963e-10f;
+      float u2 = (float)r2 * 2.3283064365386963e-10f;
+      float rad = sqrtf(-2.0f * __logf(u1)) * sigma;
+      o[j] = f2bf(mu + rad * __cosf(TWO_PI * u2));
+      o[j + 1] = f2bf(mu + rad * __sinf(TWO_PI * u2));
+    }
+    *(bf16x8*)&y[i * 8] = o;
+  }
+}
+
+void rng_uniform_bf16(bf16* y, long n, const u64* seed, u64 salt, float lo,
+                      float hi, hipStream_t s) {
+  rng_uniform_kernel<<<ew_grid(n / 8), 256, 0, s>>>(y, n / 8, seed, salt,
+                                                    lo, hi - lo);
+}
+
+void rng_normal_bf16(bf16* y, long n, const u64* seed, u64 salt, float mu,
+                     float sigma, hipStream_t s) {
+  rng_normal_kernel<<<ew_grid(n / 8), 256, 0, s>>>(y, n / 8, seed, salt,
+                                                   mu, sigma);
+}
+
 __global__ void rng_bump(u64* s) {
   if (threadIdx.x == 0 && blockIdx.x == 0) *s += 0x9e3779b97f4a7c15ULL;
 }

@@ -1099,3 +1099,89 @@ def test_ew_reduce_last():
     mn = _C.ew_reduce_last(2, x.contiguous())
     assert_close(mn, x.float().min(-1).values, rtol=1e-4, atol=1e-4,
                  msg="ew min")
+
+
+def test_native_sampling_moments():
+    """Counter-hash sampling kernels: statistical moments (reference
+    test_random.py pattern)."""
+    from anand_mxnet_amd.base import lib
+    C = lib()
+    n = 1 << 20
+    u = C.rng_sample([n], -1.0, 3.0, 12345, False, 0).float()
+    assert abs(u.mean().item() - 1.0) < 0.02
+    assert abs(u.var().item() - 16.0 / 12.0) < 0.03
+    assert u.min().item() >= -1.0 and u.max().item() <= 3.0
+    g = C.rng_sample([n], 0.5, 2.0, 999, True, 0).float()
+    assert abs(g.mean().item() - 0.5) < 0.03
+    assert abs(g.std().item() - 2.0) < 0.05
+    # skew ~ 0, kurtosis ~ 3 (loose bf16 bounds)
+    z = (g - g.mean()) / g.std()
+    assert abs((z ** 3).mean().item()) < 0.05
+    assert abs((z ** 4).mean().item() - 3.0) < 0.2
+    # different salts decorrelate
+    h = C.rng_sample([n], 0.5, 2.0, 1000, True, 0).float()
+    corr = torch.corrcoef(torch.stack([g, h]))[0, 1].item()
+    assert abs(corr) < 0.01
+
+
+def test_take_via_embedding_gather():
+    w = _rt((128, 64)).to(BF).requires_grad_(True)
+    idx = torch.randint(0, 128, (10, 7), device=_dev())
+    from anand_mxnet_amd.ops.registry import get as get_op
+    import anand_mxnet_amd  # noqa
+    from anand_mxnet_amd import nd
+    out = get_op("take")(nd.NDArray(w), nd.NDArray(idx), axis=0)
+    o = out.asnumpy()
+    ref = w.detach().float()[idx.reshape(-1)].reshape(10, 7, 64)
+    np.testing.assert_allclose(o.astype(np.float32),
+                               ref.cpu().numpy(), rtol=1e-2, atol=1e-2)
+
+
+@pytest.mark.parametrize("cfg", [
+    (2, 2, 256, 256, False),
+    (1, 4, 128, 384, False),   # cross-attention
+    (1, 2, 200, 200, True),    # causal, tails
+])
+def test_flash_fwd_d128(cfg):
+    """D = 128 heads: 64-chunk flash forward + 32-q-chunk backward."""
+    B, H, Tq, Tk, causal = cfg
+    D = 128
+    scale = 1.0 / (D ** 0.5)
+    q = _rt((B, Tq, H * D)).to(BF)
+    k = _rt((B, Tk, H * D)).to(BF)
+    v = _rt((B, Tk, H * D)).to(BF)
+    out, m, l = _C.flash_fwd(q.contiguous(), k.contiguous(),
+                             v.contiguous(), H, scale, causal, 0.0, 0)
+
+    def heads(x, T):
+        return x.reshape(B, T, H, D).transpose(1, 2).float()
+    s = heads(q, Tq) @ heads(k, Tk).transpose(-1, -2) * scale
+    if causal:
+        cm = torch.triu(torch.ones(Tq, Tk, device=_dev(),
+                                   dtype=torch.bool), 1)
+        s = s.masked_fill(cm, float("-inf"))
+    p = torch.nn.functional.softmax(s, dim=-1)
+    ref = (p @ heads(v, Tk)).transpose(1, 2).reshape(B, Tq, H * D)
+    assert_close(out, ref, rtol=5e-2, atol=5e-2, msg=f"d128 fwd {cfg}")
+    ref_m = s.max(-1).values.reshape(B * H, Tq)
+    assert_close(m, ref_m, rtol=1e-2, atol=1e-2, msg="d128 m")
+
+    dout = _rt((B, Tq, H * D)).to(BF)
+    dq, dk, dv = _C.flash_bwd(q.contiguous(), k.contiguous(),
+                              v.contiguous(), out.contiguous(),
+                              dout.contiguous(), m, l, H, scale, causal,
+                              0.0, 0)
+    qf = heads(q, Tq).clone().requires_grad_(True)
+    kf = heads(k, Tk).clone().requires_grad_(True)
+    vf = heads(v, Tk).clone().requires_grad_(True)
+    sf = qf @ kf.transpose(-1, -2) * scale
+    if causal:
+        sf = sf.masked_fill(cm, float("-inf"))
+    pf = torch.nn.functional.softmax(sf, dim=-1)
+    (pf @ vf).transpose(1, 2).reshape(B, Tq, H * D).backward(dout.float())
+    assert_close(dq, qf.grad.transpose(1, 2).reshape(B, Tq, H * D),
+                 rtol=7e-2, atol=7e-2, msg=f"d128 dq {cfg}")
+    assert_close(dk, kf.grad.transpose(1, 2).reshape(B, Tk, H * D),
+                 rtol=7e-2, atol=7e-2, msg=f"d128 dk {cfg}")
+    assert_close(dv, vf.grad.transpose(1, 2).reshape(B, Tk, H * D),
+                 rtol=7e-2, atol=7e-2, msg=f"d128 dv {cfg}")
