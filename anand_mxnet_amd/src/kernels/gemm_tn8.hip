@@ -311,6 +311,205 @@ __global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8x32_core(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Guide-faithful phase variant (§5 template discipline, MXA_TN8G):
+// fixed 128x64 wave tiles (wave reads ONE A-half + ONE B-half only —
+// halves the LDS read traffic of the quadrant-shared layout above),
+// double-barrier phases ([ds_read; glds; s_barrier; lgkmcnt(0); setprio
+// MFMA cluster; s_barrier]), ONE counted s_waitcnt vmcnt(4) per K-tile
+// window, and a race-free issue schedule proved by read-confinement:
+//   A frags fully read by p1 (A0/A1 regions rewritten by glds at p2/p3),
+//   B frags read at p0/p2 (B regions rewritten at the NEXT window's p0,
+//   which targets the other tile buffer).
+// Window T (buffer T&1):  [vmcnt(4)]
+//   p0: read a[0..3],b[0..1]; glds B0,B1 of T+1   ; MFMA mf0-3 x nf0-1
+//   p1: read a[4..7]        ; glds  -             ; MFMA mf4-7 x nf0-1
+//   p2: read b[2..3]        ; glds A0 of T+2      ; MFMA mf0-3 x nf2-3
+//   p3:  -                  ; glds A1 of T+2      ; MFMA mf4-7 x nf2-3
+// Tile T is fully resident at its window start: A halves issued at
+// window T-2 (p2,p3), B halves at window T-1 (p0); vmcnt(4) leaves only
+// tile T+1's A halves in flight.
+// ---------------------------------------------------------------------------
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8g_core(
+    Plain8Loader aload, Plain8Loader bload, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, long strideC,
+    int ksteps, long kstride) {
+  __shared__ __align__(16) bf16 smem[2][2][2][128 * 64];
+
+  const int ntn = (N + T8_BN - 1) / T8_BN;
+  const int ntm = (M + T8_BM - 1) / T8_BM;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  // 4-row tile patches: an XCD's ~32 resident tiles form a 4(M) x 8(N)
+  // patch (12 distinct A/B panels through its L2) instead of a row strip
+  // (~18 panels) — the 256^2-tile GEMM is HBM-bound without this reuse.
+  int tm, tnn;
+  // patches only pay when blocks oversubscribe the CUs (multi-wave
+  // grids); at <=1 block/CU the row-strip map measured faster (LLC)
+  if ((ntm & 3) == 0 && ntm * ntn > 256) {
+    tm = (bid / (4 * ntn)) * 4 + (bid & 3);
+    tnn = (bid >> 2) % ntn;
+  } else {
+    tm = bid / ntn;
+    tnn = bid % ntn;
+  }
+  const int bm = tm * T8_BM, bn = tnn * T8_BN;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+  const int w = t / WAVE, l = t % WAVE;
+
+  aload.init(bm, t, z);
+  bload.init(bn, t, z);
+
+  auto issue_half = [&](int op, int half, int tile) {
+    long koff = (long)tile * kstride;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[tile & 1][op][half]
+              [0] + seg);
+      const bf16* gp = op ? bload.src(p, koff) : aload.src(p, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+
+  const int lrow = l & 15, lk = l >> 4;
+  const int wr = w >> 2;        // A half (M)
+  const int wc = w & 3;         // N quarter
+  const int bh = wc >> 1;       // B half
+  const int bno = (wc & 1) * 64;  // N offset within the B half
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) acc[mf][nf] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  bf16x8 a[8][2], b[4][2];
+
+  auto read_a = [&](const bf16* As, int mf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+      a[mf][kk] = *(const bf16x8*)((const char*)As +
+                                   t8_lds_byte(mf * 16 + lrow, kk * 4 + lk));
+  };
+  auto read_b = [&](const bf16* Bs, int nf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+      b[nf][kk] = *(const bf16x8*)((const char*)Bs +
+                                   t8_lds_byte(bno + nf * 16 + lrow,
+                                               kk * 4 + lk));
+  };
+  // one 16-MFMA cluster, no sync (the window body handles all sync)
+  auto mma16 = [&](int mf0, int nf0) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[mf0 + mf][nf0 + nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf0 + mf][kk], b[nf0 + nf][kk], acc[mf0 + mf][nf0 + nf], 0,
+              0, 0);
+  };
+
+  // prologue: tile0 A0,A1,B0,B1 + tile1 A0,A1
+  issue_half(0, 0, 0);
+  issue_half(0, 1, 0);
+  issue_half(1, 0, 0);
+  issue_half(1, 1, 0);
+  if (1 < ksteps) {
+    issue_half(0, 0, 1);
+    issue_half(0, 1, 1);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();  // publish prologue landings to all waves
+
+  // Window body (2 barriers per K-tile): every fragment of the wave's
+  // A-half and B-half is register-resident before any MFMA, so the 64
+  // MFMAs run as one barrier-free setprio section.  Safety: the opening
+  // lgkmcnt(0)+barrier publishes that all waves' LDS reads are DONE, so
+  // the A-half glds of tile+2 (same buffer) issued inside the MFMA
+  // section cannot overwrite data anyone still needs; the closing
+  // vmcnt+barrier publishes every wave's glds landings for the next
+  // window's reads.
+  // Window body (2 barriers per K-tile): every fragment of the wave's
+  // A-half and B-half is register-resident before any MFMA, so the 64
+  // MFMAs run as one barrier-free setprio section.  Safety: the opening
+  // lgkmcnt(0)+barrier publishes that all waves' LDS reads are DONE, so
+  // the A-half glds of tile+2 (same buffer) issued inside the MFMA
+  // section cannot overwrite data anyone still needs; the closing
+  // vmcnt+barrier publishes every wave's glds landings for the next
+  // window's reads.  (A deeper variant issuing all 4 halves of tile+2
+  // here with vmcnt(8) measured 6-9% SLOWER — 16 loads in flight fight
+  // the read zone for L2; B halves stay on the one-window-ahead path.)
+  for (int tile = 0; tile < ksteps; ++tile) {
+    const bf16* As = smem[tile & 1][0][wr];
+    const bf16* Bs = smem[tile & 1][1][bh];
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) read_a(As, mf);
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) read_b(Bs, nf);
+    if (tile + 1 < ksteps) {
+      issue_half(1, 0, tile + 1);
+      issue_half(1, 1, tile + 1);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_s_setprio(1);
+    mma16(0, 0);
+    if (tile + 2 < ksteps) issue_half(0, 0, tile + 2);
+    mma16(4, 0);
+    if (tile + 2 < ksteps) issue_half(0, 1, tile + 2);
+    mma16(0, 2);
+    mma16(4, 2);
+    __builtin_amdgcn_s_setprio(0);
+    if (tile + 2 < ksteps)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (tile + 1 < ksteps)  // penultimate: drain tile+1 fully
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue
+  bf16* cz = Cout + (long)z * strideC;
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wc * 64 + nf * 16 + (l & 15);
+      if (gcol >= N) continue;
+      float bv = BIAS ? bias[gcol] : 0.f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wr * 128 + mf * 16 + (l >> 4) * 4 + i;
+        if (grow >= M) continue;
+        float v = acc[mf][nf][i] + bv;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        cz[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
+static bool tn8_useg() {
+  // default ON (measured +12-24% vs gemm_tn8_core on every dense shape);
+  // MXA_TN8G=0 falls back to the aligned-phase kernel
+  static const bool v = [] {
+    const char* e = getenv("MXA_TN8G");
+    return !e || e[0] != '0';
+  }();
+  return v;
+}
+
 static bool tn8_use32() {
   static const bool v = [] {
     const char* e = getenv("MXA_TN8_32");
@@ -327,6 +526,23 @@ void gemm_tn8_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
   Plain8Loader b{B, zpage, N, ldb, strideB};
   dim3 grid(cdiv_h(M, T8_BM) * cdiv_h(N, T8_BN), 1, batches);
   int ksteps = K / T8_BK;
+  if (tn8_useg()) {
+    if (bias) {
+      if (relu)
+        gemm_tn8g_core<true, true><<<grid, T8_THREADS, 0, stream>>>(
+            a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+      else
+        gemm_tn8g_core<false, true><<<grid, T8_THREADS, 0, stream>>>(
+            a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    } else if (relu) {
+      gemm_tn8g_core<true, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    } else {
+      gemm_tn8g_core<false, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    }
+    return;
+  }
   if (tn8_use32()) {
     if (bias) {
       if (relu)
