@@ -195,6 +195,15 @@ bf16* bfp_mut(at::Tensor& t) {
   return reinterpret_cast<bf16*>(t.data_ptr());
 }
 
+// minimum 256^2-tile count for the deep-pipelined tn8 path (tunable)
+static long tn8_minblocks() {
+  static const long v = [] {
+    const char* e = getenv("MXA_TN8_MINBLOCKS");
+    return e ? atol(e) : 240L;
+  }();
+  return v;
+}
+
 // 128-byte zero page per device (OOB rows of tile staging read this)
 at::Tensor& zpage_for(const at::Tensor& like) {
   static std::unordered_map<int, at::Tensor> cache;
@@ -241,10 +250,12 @@ static at::Tensor gemm_tn_impl(const at::Tensor& A, const at::Tensor& B,
     const char* e = getenv("MXA_TN8_MINK");
     // short-K GEMMs fill better on the 128-tile kernel: tn8 runs 1
     // block/CU, and a K=768 FFN wgrad-shape grid of ~384 tiles loses a
-    // third of the chip to the tail (562 vs 803 TF measured)
-    return e ? atol(e) : 1536L;
+    // third of the chip to the tail (562 vs 803 TF measured).  With the
+    // 2-barrier-window kernel the crossover moved: K=1024 grids of
+    // >=240 tiles now win on tn8 (835 vs 659 TF at 8192x4096x1024)
+    return e ? atol(e) : 1024L;
   }();
-  if (M >= 256 && N >= 256 && blocks8 >= 240 && K >= tn8_mink && !stats) {
+  if (M >= 256 && N >= 256 && blocks8 >= tn8_minblocks() && K >= tn8_mink && !stats) {
     // deep-pipelined 256² 8-phase kernel (needs a full chip of tiles;
     // smaller grids run the 128² kernel which fills better)
     mxa::gemm_tn8_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
@@ -281,10 +292,10 @@ std::vector<at::Tensor> gemm_tn_stats(const at::Tensor& A,
   long blocks8 = ((M + 255) / 256) * ((N + 255) / 256);
   static const long tn8_mink2 = [] {
     const char* e = getenv("MXA_TN8_MINK");
-    return e ? atol(e) : 1536L;
+    return e ? atol(e) : 1024L;
   }();
   if (A.dim() == 3 ||
-      (M >= 256 && N >= 256 && blocks8 >= 240 && K >= tn8_mink2))
+      (M >= 256 && N >= 256 && blocks8 >= tn8_minblocks() && K >= tn8_mink2))
     return {gemm_tn_impl(A, B, bias, relu, nullptr)};
   auto stats = at::zeros({32, 2, N}, A.options().dtype(at::kFloat));
   auto out = gemm_tn_impl(A, B, bias, relu, stats.data_ptr<float>());
