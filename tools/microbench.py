@@ -131,6 +131,18 @@ def main():
                               "ours_TF": round(2.0 * M * N * K / t / 1e12, 1),
                               "ms": round(t * 1e3, 3)}))
         return
+    if "--conv-only" in sys.argv:
+        for cfg in [(64, 56, 56, 64, 64, 3, 1, 1, "res2-3x3"),
+                    (64, 56, 56, 64, 256, 1, 1, 0, "res2-1x1"),
+                    (64, 56, 56, 256, 64, 1, 1, 0, "res2-red"),
+                    (64, 28, 28, 128, 128, 3, 1, 1, "res3-3x3"),
+                    (64, 28, 28, 512, 128, 1, 1, 0, "res3-red"),
+                    (64, 14, 14, 256, 256, 3, 1, 1, "res4-3x3"),
+                    (64, 14, 14, 1024, 256, 1, 1, 0, "res4-red"),
+                    (64, 7, 7, 512, 512, 3, 1, 1, "res5-3x3"),
+                    (64, 7, 7, 512, 2048, 1, 1, 0, "res5-exp")]:
+            print(json.dumps(bench_conv(*cfg)))
+        return
     if "--wg-only" in sys.argv:
         results.append(bench_wg(100352, 128, 128))
         results.append(bench_wg(25088, 512, 256))
