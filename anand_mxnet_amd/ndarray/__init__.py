@@ -8,6 +8,8 @@ from .ndarray import (NDArray, array, zeros, ones, empty, full, arange,
                       from_torch, waitall, concat_nd)
 from . import random  # noqa: F401
 from . import sparse  # noqa: F401 (RowSparseNDArray / row_sparse_array)
+from . import linalg  # noqa: F401 (la_op.cc surface)
+from . import utils  # noqa: F401 (save/load helpers)
 
 from .. import ops as _ops
 

@@ -216,3 +216,59 @@ def test_csr_matrix():
                                dense.asnumpy())
     assert list(m.indptr.asnumpy()) == [0, 2, 3]
     assert list(m.indices.asnumpy()) == [0, 2, 1]
+
+
+# ---------------------------------------------------------------------------
+# mx.nd.linalg surface (reference: src/operator/tensor/la_op.cc)
+# ---------------------------------------------------------------------------
+
+def test_linalg_gemm_family():
+    import numpy as np
+    import anand_mxnet_amd as mx
+    rs = np.random.RandomState(3)
+    A = mx.nd.array(rs.rand(16, 8).astype("float32"))
+    B = mx.nd.array(rs.rand(8, 12).astype("float32"))
+    C = mx.nd.array(rs.rand(16, 12).astype("float32"))
+    out = mx.nd.linalg.gemm2(A, B, alpha=2.0)
+    assert np.allclose(out.asnumpy(), 2.0 * A.asnumpy() @ B.asnumpy(),
+                       atol=1e-4)
+    out = mx.nd.linalg.gemm(A, B, C, alpha=0.5, beta=2.0)
+    assert np.allclose(out.asnumpy(),
+                       0.5 * A.asnumpy() @ B.asnumpy() + 2.0 * C.asnumpy(),
+                       atol=1e-4)
+    # transposes
+    out = mx.nd.linalg.gemm2(A, C, transpose_a=True)
+    assert np.allclose(out.asnumpy(), A.asnumpy().T @ C.asnumpy(),
+                       atol=1e-4)
+    # batched
+    Ab = mx.nd.array(rs.rand(4, 8, 8).astype("float32"))
+    Bb = mx.nd.array(rs.rand(4, 8, 8).astype("float32"))
+    out = mx.nd.linalg.gemm2(Ab, Bb)
+    assert np.allclose(out.asnumpy(), Ab.asnumpy() @ Bb.asnumpy(),
+                       atol=1e-4)
+    s = mx.nd.linalg.syrk(A)
+    assert np.allclose(s.asnumpy(), A.asnumpy() @ A.asnumpy().T, atol=1e-4)
+
+
+def test_linalg_factorizations():
+    import numpy as np
+    import anand_mxnet_amd as mx
+    rs = np.random.RandomState(5)
+    M = rs.rand(8, 8).astype("float32")
+    P = M @ M.T + 8 * np.eye(8, dtype="float32")
+    L = mx.nd.linalg.potrf(mx.nd.array(P))
+    assert np.allclose(L.asnumpy() @ L.asnumpy().T, P, atol=1e-3)
+    inv = mx.nd.linalg.potri(L)
+    assert np.allclose(inv.asnumpy(), np.linalg.inv(P), atol=1e-3)
+    X = mx.nd.linalg.trsm(L, mx.nd.array(np.eye(8, dtype="float32")))
+    assert np.allclose(L.asnumpy() @ X.asnumpy(), np.eye(8), atol=1e-3)
+    sld = mx.nd.linalg.sumlogdiag(mx.nd.array(P))
+    assert np.allclose(sld.asnumpy(), np.log(np.diag(P)).sum(), atol=1e-3)
+    d = mx.nd.linalg.det(mx.nd.array(P))
+    assert np.allclose(d.asnumpy(), np.linalg.det(P), rtol=1e-3)
+    U, lam = mx.nd.linalg.syevd(mx.nd.array(P))
+    rec = U.asnumpy().T @ np.diag(lam.asnumpy()) @ U.asnumpy()
+    assert np.allclose(rec, P, atol=1e-2)
+    Lq, Q = mx.nd.linalg.gelqf(mx.nd.array(M[:4]))
+    assert np.allclose(Lq.asnumpy() @ Q.asnumpy(), M[:4], atol=1e-3)
+    assert np.allclose(Q.asnumpy() @ Q.asnumpy().T, np.eye(4), atol=1e-3)
