@@ -175,6 +175,11 @@ class BatchNorm(HybridBlock):
 
     def hybrid_forward(self, F, x, addend=None, gamma=None, beta=None,
                        running_mean=None, running_var=None):
+        if getattr(self, "_folded", False):
+            # utils.fold_batchnorm absorbed this BN (and its fused relu)
+            # into the preceding Conv2D's weights/epilogue
+            assert addend is None, "folded BN cannot take a junction addend"
+            return x
         # optional second data arg: residual-junction form
         # relu(bn(x) + addend) fused in the kernel epilogue
         return F.BatchNorm(x, gamma, beta, running_mean, running_var,

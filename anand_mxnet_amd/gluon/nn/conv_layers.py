@@ -73,13 +73,16 @@ class Conv2D(HybridBlock):
             self.bias.shape = (self._channels,)
 
     def hybrid_forward(self, F, x, weight, bias=None):
+        if bias is None and getattr(self, "_folded_bias", None) is not None:
+            bias = self._folded_bias  # from utils.fold_batchnorm
         out = F.Convolution(x, weight, bias, kernel=self._kernel,
                             stride=self._strides, dilate=self._dilation,
                             pad=self._padding, num_filter=self._channels,
                             num_group=self._groups, no_bias=bias is None,
                             layout=self._layout,
                             collect_bn_stats=getattr(
-                                self, "_collect_bn_stats", False))
+                                self, "_collect_bn_stats", False),
+                            fold_relu=getattr(self, "_fold_relu", False))
         if self.act is not None:
             out = self.act(out)
         return out

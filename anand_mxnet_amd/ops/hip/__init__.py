@@ -398,7 +398,30 @@ _CONV_BN_STATS = os.environ.get("MXA_CONV_BN_STATS", "0") == "1"
 
 
 @register_gpu("conv2d_nhwc")
-def conv2d_nhwc(x, w, bias, stride, pad, dil, groups, collect_stats=False):
+def conv2d_nhwc(x, w, bias, stride, pad, dil, groups, collect_stats=False,
+                relu=False):
+    if relu:
+        # fused conv+relu epilogue: inference-only (conv2d_fwd's RELU
+        # template; backward has no mask).  Under autograd fall back to
+        # a separate relu so grads stay correct.
+        if torch.is_grad_enabled() and (x.requires_grad or w.requires_grad):
+            return torch.relu(conv2d_nhwc(x, w, bias, stride, pad, dil,
+                                          groups, collect_stats))
+        if groups == 1 and w.shape[3] % 64 == 0:
+            stride, pad, dil = tuple(stride), tuple(pad), tuple(dil)
+            x = x.contiguous()
+            K, R, S, C = w.shape
+            b32 = bias.float() if bias is not None else None
+            if R == 1 and S == 1 and stride == (1, 1) and pad == (0, 0):
+                N, H, W_, _ = x.shape
+                y2 = _C.gemm_tn(x.reshape(N * H * W_, C),
+                                w.reshape(K, C), b32, True)
+                return y2.reshape(N, H, W_, K)
+            w_rskc = w.permute(1, 2, 0, 3).contiguous()
+            return _C.conv2d_fwd(x, w_rskc, b32, list(stride), list(pad),
+                                 list(dil), True)
+        return torch.relu(conv2d_nhwc(x, w, bias, stride, pad, dil,
+                                      groups, collect_stats))
     if groups != 1:
         C_in = x.shape[3]
         if groups == C_in and w.shape[0] == C_in and w.shape[3] == 1                 and C_in % 8 == 0 and tuple(dil) == (1, 1):

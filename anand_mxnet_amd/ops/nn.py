@@ -107,7 +107,7 @@ def fp8_linear_op(data, weight, bias=None, num_hidden=None, no_bias=False,
 def convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
                 dilate=(1, 1), pad=(0, 0), num_filter=None, num_group=1,
                 no_bias=False, layout="NCHW", cudnn_tune=None, cudnn_off=False,
-                workspace=None, collect_bn_stats=False):
+                workspace=None, collect_bn_stats=False, fold_relu=False):
     """2-D convolution.  GPU path: implicit-GEMM MFMA kernel, NHWC-native.
 
     Weight conventions (reference parity): layout NCHW -> (K, C, R, S);
@@ -116,20 +116,25 @@ def convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
     b = None if no_bias else bias
     if layout == "NHWC":
         if _use_hip(data, "conv2d_nhwc", _BF16):
-            return gpu_impl("conv2d_nhwc")(data, weight, b, stride, pad,
-                                           dilate, num_group,
-                                           collect_bn_stats)
+            out = gpu_impl("conv2d_nhwc")(data, weight, b, stride, pad,
+                                          dilate, num_group,
+                                          collect_bn_stats,
+                                          relu=fold_relu)
+            return out
         # CPU oracle: permute to NCHW for torch
         x = data.permute(0, 3, 1, 2)
         w = weight.permute(0, 3, 1, 2)  # (K,R,S,C) -> (K,C,R,S)
         out = F.conv2d(x, w, b, stride=tuple(stride), padding=tuple(pad),
                        dilation=tuple(dilate), groups=num_group)
+        if fold_relu:
+            out = torch.relu(out)
         return out.permute(0, 2, 3, 1).contiguous()
     if data.dim() == 4 and _use_hip(data, "conv2d_nhwc", _BF16):
         # NCHW request on GPU: run the NHWC kernel with layout conversion
         x = data.permute(0, 2, 3, 1).contiguous()
         w = weight.permute(0, 2, 3, 1).contiguous()
-        out = gpu_impl("conv2d_nhwc")(x, w, b, stride, pad, dilate, num_group)
+        out = gpu_impl("conv2d_nhwc")(x, w, b, stride, pad, dilate,
+                                      num_group, relu=fold_relu)
         return out.permute(0, 3, 1, 2)
     if data.dim() != 4:
         # 1-D/3-D conv: no native kernel family (reference GPU path was
@@ -139,9 +144,10 @@ def convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
         _use_hip(data, "conv2d_nhwc")  # raises on GPU without native ext
     nd_ = data.dim() - 2
     fn = {1: F.conv1d, 2: F.conv2d, 3: F.conv3d}[nd_]
-    return fn(data, weight, b, stride=tuple(stride)[:nd_],
-              padding=tuple(pad)[:nd_], dilation=tuple(dilate)[:nd_],
-              groups=num_group)
+    out = fn(data, weight, b, stride=tuple(stride)[:nd_],
+             padding=tuple(pad)[:nd_], dilation=tuple(dilate)[:nd_],
+             groups=num_group)
+    return torch.relu(out) if fold_relu else out
 
 
 @register("Deconvolution")

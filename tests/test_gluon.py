@@ -291,3 +291,37 @@ def test_densenet_nhwc():
     net.initialize(mx.init.Xavier())
     y = net(nd.array(np.random.rand(1, 56, 56, 3).astype("float32")))
     assert y.shape == (1, 10)
+
+
+def test_fold_batchnorm_resnet_inference():
+    """utils.fold_batchnorm: conv+BN(+relu) fold matches the unfolded
+    net in inference mode (reference parity: conv+bn+relu subgraph
+    fusion, SURVEY 3.2 subgraph framework)."""
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.gluon.model_zoo import vision as zoo
+    from anand_mxnet_amd.gluon.utils import fold_batchnorm
+
+    net = zoo.resnet18_v1(layout="NHWC")
+    net.initialize(mx.init.Xavier())
+    x = mx.nd.array(np.random.RandomState(2)
+                    .rand(2, 32, 32, 3).astype("float32"))
+    net(x)  # materialize deferred shapes
+    # non-trivial running stats
+    for _, p in net.collect_params().items():
+        if p.name.endswith("running_mean"):
+            p.set_data(mx.nd.array(
+                np.random.RandomState(0).rand(p.shape[0]).astype("float32")
+                - 0.5))
+        if p.name.endswith("running_var"):
+            p.set_data(mx.nd.array(
+                0.5 + np.random.RandomState(1).rand(p.shape[0])
+                .astype("float32")))
+    ref = net(x).asnumpy()
+    n = fold_batchnorm(net)
+    # resnet18: stem BN + 2 body BNs per basic block fold; junction BNs
+    # (bn_join, addend form) must remain
+    assert n >= 9, n
+    out = net(x).asnumpy()
+    assert np.allclose(out, ref, rtol=1e-3, atol=1e-3), \
+        np.abs(out - ref).max()

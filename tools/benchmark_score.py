@@ -19,11 +19,18 @@ from anand_mxnet_amd.gluon.model_zoo import vision as zoo
 import torch
 
 
-def score(name, factory, batch_sizes, size=224, layout="NHWC"):
+def score(name, factory, batch_sizes, size=224, layout="NHWC",
+          fold=None):
+    if fold is None:
+        fold = "--fold-bn" in sys.argv
     ctx = mx.gpu(0)
     net = factory()
     net.cast("bfloat16")
     net.initialize(mx.init.Xavier(), ctx=ctx)
+    if fold:
+        from anand_mxnet_amd.gluon.utils import fold_batchnorm
+        n = fold_batchnorm(net)
+        name = f"{name} +fold{n}"
     for b in batch_sizes:
         shape = (b, size, size, 3) if layout == "NHWC" else (b, 3, size,
                                                              size)
