@@ -29,6 +29,9 @@ def score(name, factory, batch_sizes, size=224, layout="NHWC",
     net.initialize(mx.init.Xavier(), ctx=ctx)
     if fold:
         from anand_mxnet_amd.gluon.utils import fold_batchnorm
+        shape0 = (1, size, size, 3) if layout == "NHWC" else (1, 3, size,
+                                                              size)
+        net(nd.zeros(shape0, dtype="bfloat16", ctx=ctx))  # materialize
         n = fold_batchnorm(net)
         name = f"{name} +fold{n}"
     for b in batch_sizes:
