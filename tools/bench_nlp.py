@@ -23,7 +23,7 @@ def main():
                     choices=["bert_base", "transformer_big"])
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch-size", type=int, default=16,
+    ap.add_argument("--batch-size", type=int, default=32,
                     help="per-GPU sequences")
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--dtype", default="bfloat16")
