@@ -786,6 +786,215 @@ __global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8_core(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Window variant of gemm_mx8_core (the 2-barrier-per-K-tile discipline
+// measured +23% on the bf16 twin, gemm_tn8g_core): fixed 128x64 wave
+// tiles (one A-half + one B-half per wave), every fragment and scale
+// register-resident before a barrier-free 32-MFMA setprio section, ONE
+// counted s_waitcnt vmcnt(5) per K-tile.  Issue schedule per window T
+// (9 VMEM ops/tile: A0+scales 3, A1 2, B0 2, B1 2):
+//   pre-barrier:  reads; glds B0,B1 of T+1 (other buffer)
+//   post-barrier: MFMAs with glds A0+sc, A1 of T+2 (this buffer, whose
+//                 reads the opening lgkmcnt(0)+barrier just published)
+//   close:        vmcnt(5) leaves only tile T+2's A group in flight
+// ---------------------------------------------------------------------------
+
+template <int AFMT = 0, int BFMT = 0>
+__global__ __launch_bounds__(MX_THREADS, 1) void gemm_mx8g_core(
+    Mx8Loader aload, Mx8Loader bload, const u8* __restrict__ Asc,
+    const u8* __restrict__ Bsc, long lsa, long lsb, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, int ksteps_total,
+    int relu, int nsplit = 1, long strideSlot = 0,
+    float* __restrict__ work = nullptr) {
+  __shared__ __align__(16) u8 smem[2][2][2][128 * 128];
+  __shared__ __align__(16) u8 scbuf[2][2048];  // [buf][A 1K | B 1K]
+
+  const int ntn = (N + 255) / 256;
+  const int ntm = (M + 255) / 256;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm, tnn;
+  if ((ntm & 3) == 0 && ntm * ntn > 256) {  // 2D XCD patches (tn8g)
+    tm = (bid / (4 * ntn)) * 4 + (bid & 3);
+    tnn = (bid >> 2) % ntn;
+  } else {
+    tm = bid / ntn;
+    tnn = bid % ntn;
+  }
+  const int bm = tm * 256, bn = tnn * 256;
+  const int t = threadIdx.x;
+  const int w = t / WAVE, l = t % WAVE;
+
+  const int split = blockIdx.y;
+  const int per = (ksteps_total + nsplit - 1) / nsplit;
+  const int kstep0 = split * per;
+  const int ksteps = min(per, ksteps_total - kstep0);
+  if (ksteps <= 0) return;
+
+  aload.init(bm, t);
+  bload.init(bn, t);
+
+  const bool sc_is_a = t < 256;
+  const int sc_row = sc_is_a ? t : t - 256;
+  const int sc_g = sc_is_a ? bm + sc_row : bn + sc_row;
+  const u8* sc_base = sc_is_a
+                          ? (sc_g < M ? Asc + (long)sc_g * lsa : aload.zpage)
+                          : (sc_g < N ? Bsc + (long)sc_g * lsb : aload.zpage);
+  const long sc_ld = (sc_is_a ? (sc_g < M) : (sc_g < N)) ? 4 : 0;
+
+  auto issue_group = [&](int issue_idx, int tile) {
+    int op = issue_idx & 1;  // A0, B0, A1, B1
+    int half = issue_idx >> 1;
+    long koff = (long)(kstep0 + tile) * F8_BK;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[tile & 1][op][half]
+              [0] + seg);
+      const u8* gp = op ? bload.src(p, koff) : aload.src(p, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+    if (issue_idx == 0) {  // the tile's scale stage rides with half A0
+      unsigned seg = __builtin_amdgcn_readfirstlane((unsigned)(w * 256));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&scbuf[tile & 1][0] +
+          seg);
+      const u8* gp = sc_base + (sc_ld ? (long)(kstep0 + tile) * 4 : 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 4,
+          0, 0);
+    }
+  };
+
+  const int lrow = l & 15, lk = l >> 4;
+  const int wr = w >> 2;          // A half
+  const int wc = w & 3;           // N quarter
+  const int bh = wc >> 1;         // B half
+  const int bno = (wc & 1) * 64;  // N offset within the half
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) acc[mf][nf] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  i32x8 af[4], bfr[4];
+  int sa[8], sb[4];
+
+  auto read_a4 = [&](const u8* As, const u8* sc, int mf0) {
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      int row = (mf0 + mf) * 16 + lrow;
+      i32x4 lo = *(const i32x4*)&As[f8_lds_off(row, lk)];
+      i32x4 hi = *(const i32x4*)&As[f8_lds_off(row, 4 + lk)];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        af[mf][j] = lo[j];
+        af[mf][4 + j] = hi[j];
+      }
+      sa[mf0 + mf] = sc[(wr * 128 + row) * 4 + lk];
+    }
+  };
+
+  // prologue: all 9 ops of tile0 (1-window prefetch depth: the long
+  // scaled-MFMA section of each window covers the next tile's latency,
+  // so the close drains vmcnt(0) cheaply)
+#pragma unroll
+  for (int h = 0; h < 4; ++h) issue_group(h, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int tile = 0; tile < ksteps; ++tile) {
+    const u8* As = smem[tile & 1][0][wr];
+    const u8* Bs = smem[tile & 1][1][bh];
+    const u8* sc = scbuf[tile & 1];
+    read_a4(As, sc, 0);
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int row = bno + nf * 16 + lrow;
+      i32x4 lo = *(const i32x4*)&Bs[f8_lds_off(row, lk)];
+      i32x4 hi = *(const i32x4*)&Bs[f8_lds_off(row, 4 + lk)];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        bfr[nf][j] = lo[j];
+        bfr[nf][4 + j] = hi[j];
+      }
+      sb[nf] = sc[1024 + (bh * 128 + row) * 4 + lk];
+    }
+    // the whole next tile (tile+1, other buffer: its last readers
+    // finished at window tile-1's close) rides in the pre-zone
+    if (tile + 1 < ksteps) {
+#pragma unroll
+      for (int h = 0; h < 4; ++h) issue_group(h, tile + 1);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af[mf], bfr[nf], acc[mf][nf], AFMT, BFMT, 0, sa[mf], 0,
+            sb[nf]);
+    // re-use the af registers for rows 64..127 (A glds this window all
+    // target the OTHER buffer, so this buffer's A image is stable)
+    read_a4(As, sc, 4);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[4 + mf][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af[mf], bfr[nf], acc[4 + mf][nf], AFMT, BFMT, 0, sa[4 + mf], 0,
+            sb[nf]);
+    __builtin_amdgcn_s_setprio(0);
+    if (tile + 1 < ksteps)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  float* wslot = work ? work + (long)split * strideSlot : nullptr;
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wc * 64 + nf * 16 + (l & 15);
+      if (gcol >= N) continue;
+      float bv = bias ? bias[gcol] : 0.f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wr * 128 + mf * 16 + (l >> 4) * 4 + i;
+        if (grow >= M) continue;
+        float v = acc[mf][nf][i] + bv;
+        if (relu) v = v > 0.f ? v : 0.f;
+        if (wslot)
+          wslot[(long)grow * ldc + gcol] = v;
+        else
+          Cout[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
+static bool mx8_useg() {
+  // default OFF: the window port measured 672-701 TF vs the aligned
+  // 8-phase kernel's 1034-1111 on the same shapes — fp8's i32x8
+  // fragments (2x the bf16 width) push full-window residency to ~320
+  // VGPRs on the 512-reg file: 66 hot-loop spills (132 B scratch/lane).
+  // The quadrant-phase structure keeps only 48 frag regs live per phase,
+  // which is why it fits; kept for the record (profiles/RESULTS.md).
+  static const bool v = [] {
+    const char* e = getenv("MXA_MX8G");
+    return e && e[0] == '1';
+  }();
+  return v;
+}
+
 static int mx8_pick_split(long blocks, int ksteps, bool can_split) {
   if (!can_split) return 1;
   int ns = 1;
@@ -830,7 +1039,16 @@ void gemm_mx_bf16(const u8* A, const u8* Asc, const u8* B, const u8* Bsc,
     Mx8Loader b8{B, zpage, N, K};
     long slotStride = (long)M * ldc;
     float* wk = ns > 1 ? work : nullptr;
-    if (afmt == 1)
+    if (mx8_useg()) {
+      if (afmt == 1)
+        gemm_mx8g_core<1, 0><<<grid, MX_THREADS, 0, stream>>>(
+            a8, b8, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, ksteps,
+            relu, ns, slotStride, wk);
+      else
+        gemm_mx8g_core<0, 0><<<grid, MX_THREADS, 0, stream>>>(
+            a8, b8, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, ksteps,
+            relu, ns, slotStride, wk);
+    } else if (afmt == 1)
       gemm_mx8_core<1, 0><<<grid, MX_THREADS, 0, stream>>>(
           a8, b8, Asc, Bsc, K / 32, K / 32, C, bias, M, N, ldc, ksteps,
           relu, ns, slotStride, wk);
