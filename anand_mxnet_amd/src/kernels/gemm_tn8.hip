@@ -500,6 +500,158 @@ __global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8g_core(
   }
 }
 
+// 32x32x16 window variant: identical sync/issue structure to
+// gemm_tn8g_core but 32 x v_mfma_f32_32x32x16_bf16 per K-tile instead
+// of 64 x 16x16x32 — half the MFMA issue slots for the same FLOPs (the
+// aligned-structure PMC showed 46-48% SQ_WAIT_INST_ANY).  C/D layout:
+// col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8g32_core(
+    Plain8Loader aload, Plain8Loader bload, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, long strideC,
+    int ksteps, long kstride) {
+  __shared__ __align__(16) bf16 smem[2][2][2][128 * 64];
+
+  const int ntn = (N + T8_BN - 1) / T8_BN;
+  const int ntm = (M + T8_BM - 1) / T8_BM;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm, tnn;
+  if ((ntm & 3) == 0 && ntm * ntn > 256) {
+    tm = (bid / (4 * ntn)) * 4 + (bid & 3);
+    tnn = (bid >> 2) % ntn;
+  } else {
+    tm = bid / ntn;
+    tnn = bid % ntn;
+  }
+  const int bm = tm * T8_BM, bn = tnn * T8_BN;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+  const int w = t / WAVE, l = t % WAVE;
+
+  aload.init(bm, t, z);
+  bload.init(bn, t, z);
+
+  auto issue_half = [&](int op, int half, int tile) {
+    long koff = (long)tile * kstride;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smem[tile & 1][op][half]
+              [0] + seg);
+      const bf16* gp = op ? bload.src(p, koff) : aload.src(p, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+
+  const int l31 = l & 31, lhi = l >> 5;
+  const int wr = w >> 2;          // A half (M)
+  const int wc = w & 3;           // N quarter
+  const int bh = wc >> 1;         // B half
+  const int bno = (wc & 1) * 64;  // N offset within the B half
+
+  f32x16 acc[4][2];  // [m-frag 32][n-frag 32]
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) acc[mf][nf] = f32x16{};
+
+  bf16x8 a[4][4], b[2][4];  // [frag][kstep of 16]
+
+  auto read_a = [&](const bf16* As, int mf) {
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+      a[mf][kk] = *(const bf16x8*)((const char*)As +
+                                   t8_lds_byte(mf * 32 + l31,
+                                               kk * 2 + lhi));
+  };
+  auto read_b = [&](const bf16* Bs, int nf) {
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+      b[nf][kk] = *(const bf16x8*)((const char*)Bs +
+                                   t8_lds_byte(bno + nf * 32 + l31,
+                                               kk * 2 + lhi));
+  };
+
+  issue_half(0, 0, 0);
+  issue_half(0, 1, 0);
+  issue_half(1, 0, 0);
+  issue_half(1, 1, 0);
+  if (1 < ksteps) {
+    issue_half(0, 0, 1);
+    issue_half(0, 1, 1);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int tile = 0; tile < ksteps; ++tile) {
+    const bf16* As = smem[tile & 1][0][wr];
+    const bf16* Bs = smem[tile & 1][1][bh];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) read_a(As, mf);
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) read_b(Bs, nf);
+    if (tile + 1 < ksteps) {
+      issue_half(1, 0, tile + 1);
+      issue_half(1, 1, tile + 1);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      if (kk == 1 && tile + 2 < ksteps) issue_half(0, 0, tile + 2);
+      if (kk == 2 && tile + 2 < ksteps) issue_half(0, 1, tile + 2);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a[mf][kk], b[nf][kk], acc[mf][nf], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (tile + 2 < ksteps)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (tile + 1 < ksteps)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  bf16* cz = Cout + (long)z * strideC;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      int gcol = bn + wc * 64 + nf * 32 + l31;
+      if (gcol >= N) continue;
+      float bv = BIAS ? bias[gcol] : 0.f;
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int grow = bm + wr * 128 + mf * 32 + (reg & 3) + 8 * (reg >> 2) +
+                   4 * lhi;
+        if (grow >= M) continue;
+        float v = acc[mf][nf][reg] + bv;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        cz[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
+static bool tn8_useg32() {
+  static const bool v = [] {
+    const char* e = getenv("MXA_TN8G32");
+    return e && e[0] == '1';
+  }();
+  return v;
+}
+
 static bool tn8_useg() {
   // default ON (measured +12-24% vs gemm_tn8_core on every dense shape);
   // MXA_TN8G=0 falls back to the aligned-phase kernel
@@ -526,6 +678,23 @@ void gemm_tn8_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
   Plain8Loader b{B, zpage, N, ldb, strideB};
   dim3 grid(cdiv_h(M, T8_BM) * cdiv_h(N, T8_BN), 1, batches);
   int ksteps = K / T8_BK;
+  if (tn8_useg32()) {
+    if (bias) {
+      if (relu)
+        gemm_tn8g32_core<true, true><<<grid, T8_THREADS, 0, stream>>>(
+            a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+      else
+        gemm_tn8g32_core<false, true><<<grid, T8_THREADS, 0, stream>>>(
+            a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    } else if (relu) {
+      gemm_tn8g32_core<true, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    } else {
+      gemm_tn8g32_core<false, false><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    }
+    return;
+  }
   if (tn8_useg()) {
     if (bias) {
       if (relu)

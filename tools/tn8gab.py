@@ -19,7 +19,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(
 from anand_mxnet_amd import _C
 BF = torch.bfloat16
 DEV = torch.device("cuda", 0)
-out = {"arm": os.environ.get("MXA_TN8G", "0")}
+out = {"arm": os.environ.get("ARM32", os.environ.get("MXA_TN8G", "0"))}
 # correctness at a shape the tn8 router actually takes (>=240 tiles of
 # 256^2 and K>=1536): 4096^3 vs fp32 matmul of the same bf16 inputs
 torch.manual_seed(0)
@@ -52,8 +52,13 @@ print(json.dumps(out))
 def main():
     env = dict(os.environ)
     env["AB_SHAPES"] = json.dumps(SHAPES)
-    for arm in ("0", "1"):
-        env["MXA_TN8G"] = arm
+    arms = ["0", "1"]
+    if os.environ.get("AB_G32"):
+        arms.append("32")
+    for arm in arms:
+        env["MXA_TN8G"] = "1" if arm == "32" else arm
+        env["MXA_TN8G32"] = "1" if arm == "32" else "0"
+        env["ARM32"] = arm
         r = subprocess.run([sys.executable, "-c", CHILD], env=env,
                            capture_output=True, text=True,
                            cwd=os.path.dirname(os.path.dirname(
