@@ -255,7 +255,10 @@ static at::Tensor gemm_tn_impl(const at::Tensor& A, const at::Tensor& B,
     // >=240 tiles now win on tn8 (835 vs 659 TF at 8192x4096x1024)
     return e ? atol(e) : 1024L;
   }();
-  if (M >= 256 && N >= 256 && blocks8 >= tn8_minblocks() && K >= tn8_mink && !stats) {
+  bool tn8_take = (blocks8 >= tn8_minblocks() && K >= tn8_mink) ||
+                  (blocks8 >= 640 && K >= 768);  // huge grids: window
+                  // kernel wins even at K=768 (744 vs 664 TF measured)
+  if (M >= 256 && N >= 256 && tn8_take && !stats) {
     // deep-pipelined 256² 8-phase kernel (needs a full chip of tiles;
     // smaller grids run the 128² kernel which fills better)
     mxa::gemm_tn8_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
@@ -295,7 +298,9 @@ std::vector<at::Tensor> gemm_tn_stats(const at::Tensor& A,
     return e ? atol(e) : 1024L;
   }();
   if (A.dim() == 3 ||
-      (M >= 256 && N >= 256 && blocks8 >= tn8_minblocks() && K >= tn8_mink2))
+      (M >= 256 && N >= 256 &&
+       ((blocks8 >= tn8_minblocks() && K >= tn8_mink2) ||
+        (blocks8 >= 640 && K >= 768))))
     return {gemm_tn_impl(A, B, bias, relu, nullptr)};
   auto stats = at::zeros({32, 2, N}, A.options().dtype(at::kFloat));
   auto out = gemm_tn_impl(A, B, bias, relu, stats.data_ptr<float>());

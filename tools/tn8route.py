@@ -8,9 +8,9 @@ import subprocess
 import sys
 
 # (M, N, K): bert_base / transformer_big fwd+wgrad-shaped dense GEMMs
-SHAPES = [(8192, 1024, 1024), (8192, 4096, 1024), (8192, 1024, 4096),
-          (8192, 3072, 768), (8192, 768, 3072), (8192, 768, 768),
-          (16384, 1024, 1024), (16384, 4096, 1024)]
+SHAPES = [(16384, 3072, 768), (16384, 768, 3072), (16384, 768, 768),
+          (16384, 4096, 1024), (16384, 1024, 4096), (32768, 3072, 768),
+          (32768, 768, 768)]
 
 CHILD = r"""
 import json, os, sys, time
@@ -41,7 +41,7 @@ def main():
     env = dict(os.environ)
     env["AB_SHAPES"] = json.dumps(SHAPES)
     arms = [("128sq", {"MXA_TN8_MINBLOCKS": "99999"}),
-            ("tn8g-lowgate", {"MXA_TN8_MINK": "512",
+            ("tn8g-lowgate", {"MXA_TN8_MINK": "256",
                               "MXA_TN8_MINBLOCKS": "64"})]
     for name, overrides in arms:
         e = dict(env)
