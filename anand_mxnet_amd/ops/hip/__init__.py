@@ -139,8 +139,17 @@ class _Linear(torch.autograd.Function):
             else:
                 dx = torch.matmul(dy, w)
         if ctx.needs_input_grad[1]:
-            dw = _wg_or_matmul(dy, x, w.dtype)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+            want_db = ctx.has_bias and ctx.needs_input_grad[2]
+            if (want_db and x.size(-1) % 8 == 0 and dy.size(-1) % 8 == 0):
+                # fused dW + db: the wg kernel folds colsum(dy) from the
+                # A fragments it already holds (no extra dy read)
+                dw, db = _C.gemm_wg_bias(dy, x)
+                dw = dw.to(w.dtype)
+            else:
+                dw = _wg_or_matmul(dy, x, w.dtype)
+                if want_db:
+                    db = _C.colsum(dy)
+        elif ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy)
         return dx, dw, db
 

@@ -35,7 +35,8 @@ void gemm_tn8_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
 // gemm_wg.hip (slot-buffer split-K + fold epilogue)
 int gemm_wg_nslots(long, int, int, bool);
 void gemm_wg_bf16(const bf16*, const bf16*, float*, bf16*, bool, long, int,
-                  int, long, long, const bf16*, hipStream_t);
+                  int, long, long, const bf16*, hipStream_t,
+                  float* biasg = nullptr);
 int conv2d_wgrad_nslots(long, int, int, int, int);
 void conv2d_wgrad_bf16(const bf16*, const bf16*, float*, bf16*, int, int,
                        int, int, int, int, int, int, int, int, int, int,
@@ -329,6 +330,32 @@ at::Tensor gemm_wg(const at::Tensor& A, const at::Tensor& B,
   mxa::gemm_wg_bf16(bfp(A), bfp(B), workp, bfp_mut(out), transpose, m,
                     (int)Mo, (int)No, Mo, No, zp(A), cur_stream());
   return out;
+}
+
+// dW = A^T B with the bias gradient (colsum of A) folded into the same
+// pass: returns {dW bf16, db f32[Mo]}.  Used by Linear/pointwise-conv
+// backward where A is dy (reference: FullyConnectedGradCompute's
+// separate bias-reduce in nn/fully_connected-inl.h).
+std::vector<at::Tensor> gemm_wg_bias(const at::Tensor& A,
+                                     const at::Tensor& B) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous(),
+              "gemm_wg_bias: contiguous");
+  long m = A.size(0), Mo = A.size(1), No = B.size(1);
+  TORCH_CHECK(B.size(0) == m, "gemm_wg_bias: reduce dim mismatch");
+  TORCH_CHECK(Mo % 8 == 0 && No % 8 == 0, "gemm_wg_bias: dims must be %8");
+  int ns = mxa::gemm_wg_nslots(m, (int)Mo, (int)No, false);
+  auto out = at::empty({Mo, No}, A.options());
+  auto biasg = at::zeros({Mo}, A.options().dtype(at::kFloat));
+  float* workp = nullptr;
+  at::Tensor work;
+  if (ns > 0) {
+    work = at::empty({ns, Mo, No}, A.options().dtype(at::kFloat));
+    workp = work.data_ptr<float>();
+  }
+  mxa::gemm_wg_bf16(bfp(A), bfp(B), workp, bfp_mut(out), false, m, (int)Mo,
+                    (int)No, Mo, No, zp(A), cur_stream(),
+                    biasg.data_ptr<float>());
+  return {out, biasg};
 }
 
 // fp8 (e4m3) GEMM: A8 [.., M, K] u8, B8 [.., N, K] u8 -> bf16 C
@@ -1550,6 +1577,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn", &gemm_tn, "C = A @ B^T (bf16, MFMA)");
   m.def("gemm_tn_stats", &gemm_tn_stats,
         "gemm_tn + fused per-column BN sums: {out[,stats]}");
+  m.def("gemm_wg_bias", &gemm_wg_bias);
   m.def("gemm_wg", &gemm_wg,
         "C(bf16) = sum_m A[m,:] outer B[m,:] (transpose: C[No,Mo])",
         py::arg("A"), py::arg("B"), py::arg("transpose") = false);

@@ -370,7 +370,8 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
     TrPlainRows aplain, TrIm2colRows aim, TrPlainRows bload,
     float* __restrict__ Cout, int Mo, int No, long ldc, long strideSeg,
     int S, int dh, int dw, int ksteps_total, int nsplit, int use_atomic,
-    ZMap za, ZMap zb, ZMap zc, long strideSlot = 0) {
+    ZMap za, ZMap zb, ZMap zc, long strideSlot = 0,
+    float* __restrict__ biasg = nullptr) {
   __shared__ __align__(16) bf16 smem[2][2][WBK * WBM];  // [buf][op][64][128]
 
   const int zb_ = blockIdx.z;
@@ -446,6 +447,15 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
   const int wm = (lw & 3) * 32, wn = (lw >> 2) * 64;
   const int ig = ll & 15, khi = ll >> 4;
 
+  // fused bias gradient (A = dy): blocks in the first No-tile column of
+  // seg 0 also fold their A fragments over the reduce dim — kills the
+  // standalone colsum pass over dy (the A tile is already in registers)
+  // only the wn==0 wave group folds (each mo column is read by BOTH
+  // N-halves' waves; counting one avoids the 2x)
+  const bool do_bias = biasg != nullptr && bn == 0 && seg == 0 &&
+                       (lw >> 2) == 0;
+  float bsum[2] = {0.f, 0.f};
+
   auto trread = [&](const bf16* img, int row, int cbyte) {
     auto p = (__attribute__((address_space(3))) bf16x4*)(
         (__attribute__((address_space(3))) char*)img + tr_byte(row, cbyte));
@@ -479,6 +489,12 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
         for (int nf = 0; nf < 4; ++nf)
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+      if (do_bias) {
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) bsum[mf] += bf2f(af[mf][j]);
+      }
     }
   };
 
@@ -512,6 +528,20 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
         else
           cseg[(long)grow * ldc + gcol] = v;
       }
+    }
+  }
+
+  if (do_bias) {
+    // A-fragment lane layout: column = wm + mf*16 + (ll&15); the four
+    // ll>>4 groups hold disjoint reduce slices — fold then one atomic
+    // per column per block (splits sum via the atomic)
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+      float v = bsum[mf];
+      v += __shfl_xor(v, 16, 64);
+      v += __shfl_xor(v, 32, 64);
+      int col = bm + wm + mf * 16 + (ll & 15);
+      if ((ll >> 4) == 0 && col < Mo) atomicAdd(&biasg[col], v);
     }
   }
 }
@@ -756,6 +786,7 @@ __global__ __launch_bounds__(WT8, RING <= 4 ? 2 : 1) void gemm_wg_ring_core(
       }
     }
   }
+
 }
 
 // ---------------------------------------------------------------------------
@@ -1297,15 +1328,21 @@ int gemm_wg_nslots(long mtotal, int Mo, int No, bool transpose) {
   return wg_alloc_slots(p.used);
 }
 
+void colsum_bf16(const bf16* x, float* out, long m, int N, hipStream_t s);
+
 void gemm_wg_bf16(const bf16* A, const bf16* B, float* work, bf16* out,
                   bool transpose, long mtotal, int Mo, int No, long lda,
-                  long ldb, const bf16* zpage, hipStream_t stream) {
+                  long ldb, const bf16* zpage, hipStream_t stream,
+                  float* biasg) {
   long bxy = (long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN);
   WgPlanT pl = wg_plan_(bxy, mtotal);
   dim3 grid((unsigned)bxy, pl.nsplit, 1);
   long slotStride = (long)Mo * No;
   const bool direct = work == nullptr;  // OUTBF16 epilogue, no fold pass
   float* Cdst = direct ? (float*)out : work;
+  // bias-grad fusion lives in the tr core; other paths fall back to the
+  // standalone colsum over A (same result, one extra read of A)
+  bool bias_fused = false;
   if (!wg_use_zip() && lda % 8 == 0 && ldb % 8 == 0) {
     if (wg_ring()) {
       TrPlainRows1 a{A, zpage, mtotal, lda, Mo};
@@ -1327,11 +1364,13 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* work, bf16* out,
       if (direct)
         gemm_wg_tr_core<false, true><<<grid, WT8, 0, stream>>>(
             a, dummy, b, Cdst, Mo, No, No, 0, 1, 1, 1, pl.ksteps, pl.nsplit,
-            0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, 0);
+            0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, 0, biasg);
       else
         gemm_wg_tr_core<false, false><<<grid, WT8, 0, stream>>>(
             a, dummy, b, Cdst, Mo, No, No, 0, 1, 1, 1, pl.ksteps, pl.nsplit,
-            0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, slotStride);
+            0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, slotStride,
+            biasg);
+      bias_fused = true;
     }
   } else {
     WgPlainRows a{A, zpage, mtotal, lda, Mo};
@@ -1343,6 +1382,7 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* work, bf16* out,
   }
   if (!direct)
     wg_fold(work, slotStride, pl.used, Mo, No, out, transpose, stream);
+  if (biasg && !bias_fused) colsum_bf16(A, biasg, mtotal, Mo, stream);
 }
 
 void conv_wg_pad_image(const bf16* x, bf16* xp_alloc, int N, int H, int W,

@@ -1185,3 +1185,20 @@ def test_flash_fwd_d128(cfg):
                  rtol=7e-2, atol=7e-2, msg=f"d128 dk {cfg}")
     assert_close(dv, vf.grad.transpose(1, 2).reshape(B, Tk, H * D),
                  rtol=7e-2, atol=7e-2, msg=f"d128 dv {cfg}")
+
+
+def test_gemm_wg_bias_fused():
+    """gemm_wg_bias: dW and the folded bias gradient (colsum of A) in one
+    pass vs fp32 oracle, with and without split-K slots."""
+    torch.manual_seed(7)
+    for (m, Mo, No) in [(4096, 768, 768), (100352, 128, 128),
+                        (8192, 3072, 768), (777, 64, 64)]:
+        dy = _rt((m, Mo)).to(BF)
+        x = _rt((m, No)).to(BF)
+        dw, db = _C.gemm_wg_bias(dy.contiguous(), x.contiguous())
+        ref_dw = dy.float().t() @ x.float()
+        ref_db = dy.float().sum(0)
+        assert_close(dw, ref_dw, rtol=5e-2, atol=5e-2,
+                     msg=f"wg_bias dw {m}x{Mo}x{No}")
+        assert_close(db, ref_db, rtol=2e-2, atol=2e-2,
+                     msg=f"wg_bias db {m}x{Mo}x{No}")
