@@ -341,9 +341,12 @@ class _Conv2dIm2col(torch.autograd.Function):
         dy2 = dy.reshape(N * P * Q, K).contiguous()
         dw = db = None
         if ctx.needs_input_grad[1]:
-            dwf = _C.gemm_wg(dy2, A)       # [K, kdim_padded] bf16
+            if ctx.has_bias and ctx.needs_input_grad[2]:
+                dwf, db = _C.gemm_wg_bias(dy2, A)  # fused colsum(dy)
+            else:
+                dwf = _C.gemm_wg(dy2, A)   # [K, kdim_padded] bf16
             dw = dwf[:, :kdim].reshape(w.shape).to(w.dtype)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+        elif ctx.has_bias and ctx.needs_input_grad[2]:
             db = _C.colsum(dy2)
         dx = None
         if ctx.needs_input_grad[0]:

@@ -1202,3 +1202,67 @@ def test_gemm_wg_bias_fused():
                      msg=f"wg_bias dw {m}x{Mo}x{No}")
         assert_close(db, ref_db, rtol=2e-2, atol=2e-2,
                      msg=f"wg_bias db {m}x{Mo}x{No}")
+
+
+def test_fold_batchnorm_gpu_bf16():
+    """fold_batchnorm on a bf16 NHWC conv-bn-relu stack on the GPU:
+    folded (fused conv relu epilogue) matches unfolded inference."""
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.gluon import nn
+    from anand_mxnet_amd.gluon.utils import fold_batchnorm
+    net = nn.HybridSequential()
+    with net.name_scope():
+        net.add(nn.Conv2D(64, 3, padding=1, layout="NHWC", in_channels=64,
+                          use_bias=False))
+        net.add(nn.BatchNorm(axis=3, fuse_relu=True, in_channels=64))
+        net.add(nn.Conv2D(128, 1, layout="NHWC", in_channels=64))
+        net.add(nn.BatchNorm(axis=3, in_channels=128))
+    net.cast("bfloat16")
+    net.initialize(mx.init.Xavier(), ctx=mx.gpu(0))
+    x = mx.nd.random.uniform(-1, 1, shape=(2, 16, 16, 64),
+                             dtype="bfloat16", ctx=mx.gpu(0))
+    net(x)  # materialize
+    rs = np.random.RandomState(0)
+    for blk in net._children.values():
+        if isinstance(blk, nn.BatchNorm):
+            n = blk.running_mean.shape[0]
+            blk.running_mean.set_data(mx.nd.array(
+                (rs.rand(n) - 0.5).astype("float32"), ctx=mx.gpu(0)))
+            blk.running_var.set_data(mx.nd.array(
+                (0.5 + rs.rand(n)).astype("float32"), ctx=mx.gpu(0)))
+    ref = net(x).asnumpy().astype("float32")
+    assert fold_batchnorm(net) == 2
+    out = net(x).asnumpy().astype("float32")
+    assert np.allclose(out, ref, rtol=5e-2, atol=5e-2), \
+        np.abs(out - ref).max()
+
+
+def test_mha_d128_op():
+    """nd.mha with 128-dim heads dispatches the fused flash path
+    (forward+backward) and matches the unfused fp32 oracle."""
+    B, T, H, D = 2, 200, 2, 128
+    q = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    k = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    v = _rt((B, T, H * D)).to(BF).requires_grad_(True)
+    from anand_mxnet_amd.ops.hip import mha
+    scale = 1.0 / (D ** 0.5)
+    out = mha(q, k, v, H, True, scale)
+    dout = _rt((B, T, H * D)).to(BF)
+    out.backward(dout)
+    qf = _heads(q.detach(), H).float().requires_grad_(True)
+    kf = _heads(k.detach(), H).float().requires_grad_(True)
+    vf = _heads(v.detach(), H).float().requires_grad_(True)
+    s = qf @ kf.transpose(-1, -2) * scale
+    cm = torch.triu(torch.ones(T, T, device=_dev(), dtype=torch.bool), 1)
+    s = s.masked_fill(cm, float("-inf"))
+    p = torch.nn.functional.softmax(s, dim=-1)
+    (p @ vf).transpose(1, 2).reshape(B, T, H * D).backward(dout.float())
+    assert_close(out, (p @ vf).transpose(1, 2).reshape(B, T, H * D),
+                 rtol=5e-2, atol=5e-2, msg="mha d128 out")
+    assert_close(q.grad, qf.grad.transpose(1, 2).reshape(B, T, H * D),
+                 rtol=7e-2, atol=7e-2, msg="mha d128 dq")
+    assert_close(k.grad, kf.grad.transpose(1, 2).reshape(B, T, H * D),
+                 rtol=7e-2, atol=7e-2, msg="mha d128 dk")
+    assert_close(v.grad, vf.grad.transpose(1, 2).reshape(B, T, H * D),
+                 rtol=7e-2, atol=7e-2, msg="mha d128 dv")
