@@ -300,6 +300,214 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn64_core(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 256^2 window variant (the 2-barrier-per-K-tile discipline measured
+// +23% on gemm_tn8g_core): fixed 128x64 wave tiles, A fragments via
+// plain b128 reads of the row-XOR image, B fragments via tr reads of
+// row-major [64 k][128 n] images, all register-resident before a
+// barrier-free 64-MFMA setprio section.  Same issue/vmcnt accounting as
+// tn8g: B halves of tile+1 in the pre-zone (other buffer), A halves of
+// tile+2 inside the MFMA section (this buffer, reads published by the
+// opening lgkmcnt(0)+barrier), one vmcnt(4) per window.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(NTHREADS, 1) void gemm_nn8g_core(
+    const bf16* __restrict__ A, ZMap za, long lda, int M,
+    const bf16* __restrict__ B, ZMap zb, long ldb, int N,
+    bf16* __restrict__ Cout, ZMap zc, long ldc,
+    int ksteps, const bf16* __restrict__ zpage) {
+  __shared__ __align__(16) bf16 smemA[2][2][128 * 64];
+  __shared__ __align__(16) bf16 smemB[2][2][64 * 128];
+
+  const int ntn = (N + 255) / 256;
+  const int ntm = (M + 255) / 256;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm, tnn;
+  if ((ntm & 3) == 0 && ntm * ntn > 256) {  // 2D XCD patches (tn8g)
+    tm = (bid / (4 * ntn)) * 4 + (bid & 3);
+    tnn = (bid >> 2) % ntn;
+  } else {
+    tm = bid / ntn;
+    tnn = bid % ntn;
+  }
+  const int bm = tm * 256, bn = tnn * 256;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  // A loader: 4 row-slots (half = p>>1, glds pass j = p&1), tn8 scheme
+  const bf16* abase = A + za.off(z);
+  const bf16* arp[4];
+  int acg[4];
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int local = (p >> 1) * 128 + lw * 16 + (p & 1) * 8 + (ll >> 3);
+    int r = bm + local;
+    arp[p] = (r < M) ? abase + (long)r * lda : nullptr;
+    acg[p] = (ll & 7) ^ (local & 7);
+  }
+  auto issue_a = [&](int half, int tile) {
+    long koff = (long)tile * NBK;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(lw * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemA[tile & 1][half][0] +
+          seg);
+      const bf16* gp = arp[p] ? arp[p] + koff + acg[p] * 8
+                              : zpage + acg[p] * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+
+  // B loader: [64 k][128 n] halves, 256-B rows, tr key (row&3)<<1.
+  // Stage pass p covers rows p*32 + (lw*4 + ll>>4); per-lane source
+  // column clamped against N (zpage holds 128 zero bf16).
+  const int bsrow = lw * 4 + (ll >> 4);
+  const int bcg = (ll & 15) ^ ((bsrow & 3) << 1);
+  const bf16* brp0 = B + zb.off(z) + (long)bsrow * ldb + bn + bcg * 8;
+  bool bok[2];
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
+    bok[h] = bn + h * 128 + bcg * 8 + 8 <= N;
+  auto issue_b = [&](int half, int tile) {
+    long koff = (long)tile * NBK * ldb + (long)half * 128;
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)((p * 32 + lw * 4) * 256));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemB[tile & 1][half][0] +
+          seg);
+      const bf16* gp = bok[half] ? brp0 + koff + (long)(p * 32) * ldb
+                                 : zpage + bcg * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+
+  const int lrow = ll & 15, lk = ll >> 4;
+  const int ig = ll & 15, khi = ll >> 4;
+  const int wr = lw >> 2;          // A half (M)
+  const int wc = lw & 3;           // N quarter
+  const int bh = wc >> 1;          // B half
+  const int bno = (wc & 1) * 64;   // N offset within the half
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) acc[mf][nf] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  bf16x8 a[8][2], b[4][2];
+
+  auto trread = [&](const bf16* img, int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)img + nn_b_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+
+  // prologue: tile0 A+B + tile1 A (vmcnt(4) leaves tile1's A in flight)
+  issue_a(0, 0);
+  issue_a(1, 0);
+  issue_b(0, 0);
+  issue_b(1, 0);
+  if (1 < ksteps) {
+    issue_a(0, 1);
+    issue_a(1, 1);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int tile = 0; tile < ksteps; ++tile) {
+    const bf16* As = smemA[tile & 1][wr];
+    const bf16* Bs = smemB[tile & 1][bh];
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        a[mf][kk] = *(const bf16x8*)&As[nn_a_off(mf * 16 + lrow,
+                                                 kk * 4 + lk)];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int rbase = kk * 32 + khi * 8 + (ig >> 2);
+      int cb = (ig & 3) * 8;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int cbyte = (bno + nf * 16) * 2 + cb;
+        bf16x4 lo = trread(Bs, rbase, cbyte);
+        bf16x4 hi = trread(Bs, rbase + 4, cbyte);
+        b[nf][kk] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+    }
+    if (tile + 1 < ksteps) {
+      issue_b(0, tile + 1);
+      issue_b(1, tile + 1);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      if (kk == 0 && tile + 2 < ksteps) issue_a(0, tile + 2);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf][kk], b[nf][kk], acc[mf][nf], 0, 0, 0);
+      if (kk == 0 && tile + 2 < ksteps) issue_a(1, tile + 2);
+#pragma unroll
+      for (int mf = 4; mf < 8; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf][kk], b[nf][kk], acc[mf][nf], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (tile + 2 < ksteps)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (tile + 1 < ksteps)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  bf16* cz = Cout + zc.off(z);
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wc * 64 + nf * 16 + (ll & 15);
+      if (gcol >= N) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wr * 128 + mf * 16 + (ll >> 4) * 4 + i;
+        if (grow >= M) continue;
+        cz[(long)grow * ldc + gcol] = f2bf(acc[mf][nf][i]);
+      }
+    }
+  }
+}
+
+static bool nn8_useg() {
+  // default ON for K >= 1024 grids of >= 240 tiles (A/B: +7-30% — 1041
+  // vs 799 TF at 8192^3, +12% at 16384x4096x1024, +7.5% at the ResNet
+  // dgrad shape 200704x512x2048; K=768 shapes LOSE ~15% and stay on
+  // gemm_nn_core).  MXA_NN8G=0 reverts.
+  static const bool v = [] {
+    const char* e = getenv("MXA_NN8G");
+    return !e || e[0] != '0';
+  }();
+  return v;
+}
+
 void gemm_nn_bf16(const bf16* A, ZMap za, long lda, int M,
                   const bf16* B, ZMap zb, long ldb, int N,
                   bf16* C, ZMap zc, long ldc, int K, int nz,
@@ -308,6 +516,14 @@ void gemm_nn_bf16(const bf16* A, ZMap za, long lda, int M,
   if (N <= 64) {
     dim3 grid((unsigned)cdiv_h(M, NBM), 1, nz);
     gemm_nn64_core<<<grid, NTHREADS, 0, stream>>>(A, za, lda, M, B, zb,
+                                                  ldb, N, C, zc, ldc,
+                                                  ksteps, zpage);
+    return;
+  }
+  long bxy256 = (long)cdiv_h(M, 256) * cdiv_h(N, 256) * (long)nz;
+  if (nn8_useg() && M >= 256 && N >= 256 && bxy256 >= 240 && K >= 1024) {
+    dim3 grid((unsigned)(cdiv_h(M, 256) * cdiv_h(N, 256)), 1, nz);
+    gemm_nn8g_core<<<grid, NTHREADS, 0, stream>>>(A, za, lda, M, B, zb,
                                                   ldb, N, C, zc, ldc,
                                                   ksteps, zpage);
     return;

@@ -123,7 +123,9 @@ def main():
         return
     if "--nn-only" in sys.argv:
         for (M, K, N) in [(8192, 768, 768), (8192, 768, 3072),
-                          (8192, 3072, 768), (4096, 4096, 4096)]:
+                          (8192, 3072, 768), (4096, 4096, 4096),
+                          (16384, 1024, 4096), (16384, 4096, 1024),
+                          (8192, 8192, 8192), (200704, 2048, 512)]:
             a = (torch.rand((M, K), device=DEV) * 2 - 1).to(BF)
             b = (torch.rand((K, N), device=DEV) * 2 - 1).to(BF)
             t = timeit(lambda: _C.gemm_nn(a, b))
