@@ -311,6 +311,15 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nn64_core(
 // opening lgkmcnt(0)+barrier), one vmcnt(4) per window.
 // ---------------------------------------------------------------------------
 
+// tr-image swizzle with a WIDER key than nn_b_byte: the fragment read
+// set spans rows {q + 4h + 8*khi} — row&3 alone leaves the four khi
+// groups on identical chunk columns (PMC: 100M 4-way conflicts at
+// 8192^3); folding (row>>3)&3 into the key separates them.
+DEVINL int nn_b2_byte(int row, int cbyte) {
+  int key = ((row & 3) << 1) ^ ((row >> 3) & 3);
+  return row * 256 + ((((cbyte >> 4) ^ key) << 4) | (cbyte & 15));
+}
+
 __global__ __launch_bounds__(NTHREADS, 1) void gemm_nn8g_core(
     const bf16* __restrict__ A, ZMap za, long lda, int M,
     const bf16* __restrict__ B, ZMap zb, long ldb, int N,
@@ -368,12 +377,23 @@ __global__ __launch_bounds__(NTHREADS, 1) void gemm_nn8g_core(
   // Stage pass p covers rows p*32 + (lw*4 + ll>>4); per-lane source
   // column clamped against N (zpage holds 128 zero bf16).
   const int bsrow = lw * 4 + (ll >> 4);
-  const int bcg = (ll & 15) ^ ((bsrow & 3) << 1);
-  const bf16* brp0 = B + zb.off(z) + (long)bsrow * ldb + bn + bcg * 8;
-  bool bok[2];
+  // per-pass source chunk for nn_b2_byte's wider key (row bits 0-1 and
+  // 3-4; pass p only sets row bit 5, so the key is in fact p-invariant
+  // -- kept per-pass for clarity, the unroller folds the duplicates)
+  const bf16* bbase = B + zb.off(z);
+  int bcg[2];
+  const bf16* brp0[2];
+  bool bok[2][2];
 #pragma unroll
-  for (int h = 0; h < 2; ++h)
-    bok[h] = bn + h * 128 + bcg * 8 + 8 <= N;
+  for (int p = 0; p < 2; ++p) {
+    int row = p * 32 + bsrow;
+    int key = ((row & 3) << 1) ^ ((row >> 3) & 3);
+    bcg[p] = (ll & 15) ^ key;
+    brp0[p] = bbase + (long)row * ldb + bn + bcg[p] * 8;
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+      bok[p][h] = bn + h * 128 + bcg[p] * 8 + 8 <= N;
+  }
   auto issue_b = [&](int half, int tile) {
     long koff = (long)tile * NBK * ldb + (long)half * 128;
 #pragma unroll
@@ -383,8 +403,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void gemm_nn8g_core(
       auto lds = (__attribute__((address_space(3))) unsigned int*)(
           (__attribute__((address_space(3))) char*)&smemB[tile & 1][half][0] +
           seg);
-      const bf16* gp = bok[half] ? brp0 + koff + (long)(p * 32) * ldb
-                                 : zpage + bcg * 8;
+      const bf16* gp = bok[p][half] ? brp0[p] + koff : zpage + bcg[p] * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
           0, 0);
@@ -408,7 +427,8 @@ __global__ __launch_bounds__(NTHREADS, 1) void gemm_nn8g_core(
 
   auto trread = [&](const bf16* img, int row, int cbyte) {
     auto p = (__attribute__((address_space(3))) bf16x4*)(
-        (__attribute__((address_space(3))) char*)img + nn_b_byte(row, cbyte));
+        (__attribute__((address_space(3))) char*)img +
+        nn_b2_byte(row, cbyte));
     return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
   };
 

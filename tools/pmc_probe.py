@@ -29,5 +29,10 @@ elif which == "tn":
     b = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
     for _ in range(iters):
         _C.gemm_tn(a, b, None, False)
+elif which == "nn":
+    a = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
+    b = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
+    for _ in range(iters):
+        _C.gemm_nn(a, b)
 torch.cuda.synchronize()
 print("done", which)
