@@ -516,6 +516,16 @@ __global__ __launch_bounds__(NTHREADS, 1) void gemm_nn8g_core(
   }
 }
 
+static long nn8_minblocks() {
+  static const long v = [] {
+    const char* e = getenv("MXA_NN8_MINBLOCKS");
+    // 160: BERT's 192-tile K=3072 dx shapes win on the window kernel
+    // (506 -> 517k tok/s e2e); tbig unchanged
+    return e ? atol(e) : 160L;
+  }();
+  return v;
+}
+
 static bool nn8_useg() {
   // default ON for K >= 1024 grids of >= 240 tiles (A/B: +7-30% — 1041
   // vs 799 TF at 8192^3, +12% at 16384x4096x1024, +7.5% at the ResNet
@@ -541,7 +551,8 @@ void gemm_nn_bf16(const bf16* A, ZMap za, long lda, int M,
     return;
   }
   long bxy256 = (long)cdiv_h(M, 256) * cdiv_h(N, 256) * (long)nz;
-  if (nn8_useg() && M >= 256 && N >= 256 && bxy256 >= 240 && K >= 1024) {
+  if (nn8_useg() && M >= 256 && N >= 256 && bxy256 >= nn8_minblocks() &&
+      K >= 1024) {
     dim3 grid((unsigned)(cdiv_h(M, 256) * cdiv_h(N, 256)), 1, nz);
     gemm_nn8g_core<<<grid, NTHREADS, 0, stream>>>(A, za, lda, M, B, zb,
                                                   ldb, N, C, zc, ldc,
