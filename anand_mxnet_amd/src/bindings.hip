@@ -32,6 +32,9 @@ void conv2d_dgrad_bf16(const bf16*, const bf16*, bf16*, float*, int, int,
 void gemm_tn8_bf16(const bf16*, const bf16*, bf16*, const float*, int, int,
                    int, long, long, long, long, long, long, int, bool,
                    const bf16*, hipStream_t);
+void gemm_tn8h_bf16(const bf16*, const bf16*, bf16*, const float*, int,
+                    int, int, long, long, long, long, long, long, int,
+                    bool, const bf16*, hipStream_t);
 // gemm_wg.hip (slot-buffer split-K + fold epilogue)
 int gemm_wg_nslots(long, int, int, bool);
 void gemm_wg_bf16(const bf16*, const bf16*, float*, bf16*, bool, long, int,
@@ -259,12 +262,23 @@ static at::Tensor gemm_tn_impl(const at::Tensor& A, const at::Tensor& B,
   bool tn8_take = (blocks8 >= tn8_minblocks() && K >= tn8_mink) ||
                   (blocks8 >= 640 && K >= 768);  // huge grids: window
                   // kernel wins even at K=768 (744 vs 664 TF measured)
+  static const bool tn8h_on = [] {
+    const char* e = getenv("MXA_TN8H");
+    return e && e[0] == '1';  // off until A/B'd
+  }();
+  long blocksh = ((M + 255) / 256) * ((N + 127) / 128) * (long)batches;
   if (M >= 256 && N >= 256 && tn8_take && !stats) {
     // deep-pipelined 256² 8-phase kernel (needs a full chip of tiles;
     // smaller grids run the 128² kernel which fills better)
     mxa::gemm_tn8_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M, (int)N,
                        (int)K, K, K, N, sA, sB, sC, batches, relu, zp(A),
                        cur_stream());
+  } else if (tn8h_on && M >= 256 && N >= 128 && K >= 768 &&
+             blocksh >= 240 && !stats) {
+    // 256x128 window variant: short-K grids the 256² tile can't fill
+    mxa::gemm_tn8h_bf16(bfp(A), bfp(B), bfp_mut(out), bptr, (int)M,
+                        (int)N, (int)K, K, K, N, sA, sB, sC, batches,
+                        relu, zp(A), cur_stream());
   } else {
     // (measured: a split-K variant exists — gemm_tn_splitk_bf16 — but
     // auto-splitting these mid-size grids LOWERED throughput: the

@@ -644,6 +644,156 @@ __global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8g32_core(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Half-width window variant (256 x 128 tile): for short-K (K=768-class)
+// grids the 256^2 kernel can't fill — halving BN doubles the tile count
+// (8192x3072x768: 384 -> 768 blocks) while keeping the 2-barrier window
+// discipline.  Waves are 4(M) x 2(N), 64x64 patches, acc[4][4]; B is a
+// single [128][64] row-XOR image per buffer.
+// ---------------------------------------------------------------------------
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(T8_THREADS, 1) void gemm_tn8h_core(
+    Plain8Loader aload, Plain8Loader bload, bf16* __restrict__ Cout,
+    const float* __restrict__ bias, int M, int N, long ldc, long strideC,
+    int ksteps, long kstride) {
+  __shared__ __align__(16) bf16 smemA[2][2][128 * 64];
+  __shared__ __align__(16) bf16 smemB[2][128 * 64];
+
+  const int ntn = (N + 127) / 128;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * T8_BM, bn = (bid % ntn) * 128;
+  const int t = threadIdx.x;
+  const int z = blockIdx.z;
+  const int w = t / WAVE, l = t % WAVE;
+
+  aload.init(bm, t, z);
+  bload.init(bn, t, z);
+
+  auto issue_a = [&](int half, int tile) {
+    long koff = (long)tile * kstride;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int p = half * 2 + j;
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemA[tile & 1][half][0] +
+          seg);
+      const bf16* gp = aload.src(p, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+  auto issue_b = [&](int tile) {  // B: single 128-row half (slots 0,1)
+    long koff = (long)tile * kstride;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      unsigned seg = __builtin_amdgcn_readfirstlane(
+          (unsigned)(w * 2048 + j * 1024));
+      auto lds = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemB[tile & 1][0] + seg);
+      const bf16* gp = bload.src(j, koff);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp, lds, 16,
+          0, 0);
+    }
+  };
+
+  const int lrow = l & 15, lk = l >> 4;
+  const int wr2 = w >> 1;         // M quarter (64 rows)
+  const int wc2 = w & 1;          // N half (64 cols)
+  const int ah = wr2 >> 1;        // A half image
+  const int amo = (wr2 & 1) * 64; // row offset within the half
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) acc[mf][nf] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  bf16x8 a[4][2], b[4][2];
+
+  // prologue: tile0 A+B, tile1 A (vmcnt(4) leaves tile1's A in flight)
+  issue_a(0, 0);
+  issue_a(1, 0);
+  issue_b(0);
+  if (1 < ksteps) {
+    issue_a(0, 1);
+    issue_a(1, 1);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int tile = 0; tile < ksteps; ++tile) {
+    const bf16* As = smemA[tile & 1][ah];
+    const bf16* Bs = smemB[tile & 1];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        a[mf][kk] = *(const bf16x8*)((const char*)As +
+                                     t8_lds_byte(amo + mf * 16 + lrow,
+                                                 kk * 4 + lk));
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        b[nf][kk] = *(const bf16x8*)((const char*)Bs +
+                                     t8_lds_byte(wc2 * 64 + nf * 16 + lrow,
+                                                 kk * 4 + lk));
+    if (tile + 1 < ksteps) issue_b(tile + 1);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      if (kk == 0 && tile + 2 < ksteps) issue_a(0, tile + 2);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf][kk], b[nf][kk], acc[mf][nf], 0, 0, 0);
+      if (kk == 0 && tile + 2 < ksteps) issue_a(1, tile + 2);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 2; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf][kk], b[nf][kk], acc[mf][nf], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (tile + 2 < ksteps)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (tile + 1 < ksteps)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  bf16* cz = Cout + (long)z * strideC;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wc2 * 64 + nf * 16 + (l & 15);
+      if (gcol >= N) continue;
+      float bv = BIAS ? bias[gcol] : 0.f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + ah * 128 + amo + mf * 16 + (l >> 4) * 4 + i;
+        if (grow >= M) continue;
+        float v = acc[mf][nf][i] + bv;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        cz[(long)grow * ldc + gcol] = f2bf(v);
+      }
+    }
+  }
+}
+
 static bool tn8_useg32() {
   static const bool v = [] {
     const char* e = getenv("MXA_TN8G32");
@@ -743,6 +893,31 @@ void gemm_tn8_bf16(const bf16* A, const bf16* B, bf16* C, const float* bias,
     else
       gemm_tn8_core<false, false><<<grid, T8_THREADS, 0, stream>>>(
           a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+  }
+}
+
+void gemm_tn8h_bf16(const bf16* A, const bf16* B, bf16* C,
+                    const float* bias, int M, int N, int K, long lda,
+                    long ldb, long ldc, long strideA, long strideB,
+                    long strideC, int batches, bool relu,
+                    const bf16* zpage, hipStream_t stream) {
+  Plain8Loader a{A, zpage, M, lda, strideA};
+  Plain8Loader b{B, zpage, N, ldb, strideB};
+  dim3 grid(cdiv_h(M, T8_BM) * cdiv_h(N, 128), 1, batches);
+  int ksteps = K / T8_BK;
+  if (bias) {
+    if (relu)
+      gemm_tn8h_core<true, true><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+    else
+      gemm_tn8h_core<false, true><<<grid, T8_THREADS, 0, stream>>>(
+          a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+  } else if (relu) {
+    gemm_tn8h_core<true, false><<<grid, T8_THREADS, 0, stream>>>(
+        a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
+  } else {
+    gemm_tn8h_core<false, false><<<grid, T8_THREADS, 0, stream>>>(
+        a, b, C, bias, M, N, ldc, strideC, ksteps, T8_BK);
   }
 }
 
