@@ -1067,7 +1067,7 @@ std::vector<at::Tensor> flash_bwd(const at::Tensor& q,
   long B = q.size(0), Tq = q.size(1), HD = q.size(2), Tk = k.size(1);
   auto opts = q.options().dtype(at::kFloat);
   auto drow = at::empty({B * H, Tq}, opts);
-  auto dqacc = at::zeros({B, Tq, HD}, opts);
+  auto dqacc = at::empty({B, Tq, HD}, opts);  // zeroed inside flash_drow
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
   auto& sw = rng_seed_for(q);

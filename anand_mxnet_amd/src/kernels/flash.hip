@@ -1042,7 +1042,16 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_core(
 // rowwise dot for the backward: Drow[z][q] = sum_d dO * O  (wave-per-row)
 __global__ __launch_bounds__(256) void flash_drow(
     const bf16* __restrict__ dO, const bf16* __restrict__ O,
-    float* __restrict__ Drow, int B, int H, int Tq, int D) {
+    float* __restrict__ Drow, int B, int H, int Tq, int D,
+    float* __restrict__ dqzero, long nzero4) {
+  // fused dQacc zeroing (float4 grid-stride) — replaces the separate
+  // at::zeros fill the backward otherwise pays before its atomics
+  if (dqzero) {
+    auto z4 = (f32x4*)dqzero;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nzero4;
+         i += (long)gridDim.x * blockDim.x)
+      z4[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
   const long hd = (long)H * D;
   const int w = threadIdx.x >> 6, l = threadIdx.x & 63;
   long rows = (long)B * H * Tq;
@@ -1354,7 +1363,8 @@ void flash_bwd_bf16(const bf16* Q, const bf16* K, const bf16* V,
                     u64 salt, float pdrop, hipStream_t stream) {
   long rows = (long)B * H * Tq;
   int blocks = (int)min((rows + 3) / 4, 2048L);
-  flash_drow<<<blocks, 256, 0, stream>>>(dOp, O, Drow, B, H, Tq, D);
+  flash_drow<<<blocks, 256, 0, stream>>>(dOp, O, Drow, B, H, Tq, D, dQacc,
+                                         (long)B * Tq * H * D / 4);
   float dscale = pdrop > 0.f ? 1.f / (1.f - pdrop) : 1.f;
   dim3 grid((Tk + 127) / 128, 1, B * H);
   if (D == 128) {
