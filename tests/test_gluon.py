@@ -325,3 +325,43 @@ def test_fold_batchnorm_resnet_inference():
     out = net(x).asnumpy()
     assert np.allclose(out, ref, rtol=1e-3, atol=1e-3), \
         np.abs(out - ref).max()
+
+
+def test_fold_batchnorm_edge_cases():
+    """fold with a biased conv folds into the existing bias; a BN whose
+    axis does not match the conv layout's channel axis is skipped."""
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.gluon import nn
+    from anand_mxnet_amd.gluon.utils import fold_batchnorm
+
+    # biased conv: fold lands in conv.bias, outputs must match
+    net = nn.HybridSequential()
+    with net.name_scope():
+        net.add(nn.Conv2D(8, 3, padding=1, layout="NHWC", in_channels=4,
+                          use_bias=True))
+        net.add(nn.BatchNorm(axis=3, in_channels=8))
+    net.initialize(mx.init.Xavier())
+    x = mx.nd.array(np.random.RandomState(1)
+                    .rand(2, 6, 6, 4).astype("float32"))
+    net(x)
+    bn = list(net._children.values())[1]
+    bn.running_mean.set_data(mx.nd.array(
+        np.random.RandomState(2).rand(8).astype("float32") - 0.5))
+    bn.running_var.set_data(mx.nd.array(
+        0.5 + np.random.RandomState(3).rand(8).astype("float32")))
+    ref = net(x).asnumpy()
+    assert fold_batchnorm(net) == 1
+    conv = list(net._children.values())[0]
+    assert conv.bias is not None  # folded INTO the parameter
+    assert getattr(conv, "_folded_bias", None) is None
+    out = net(x).asnumpy()
+    assert np.allclose(out, ref, atol=2e-3), np.abs(out - ref).max()
+
+    # axis mismatch: nothing folds
+    net2 = nn.HybridSequential()
+    with net2.name_scope():
+        net2.add(nn.Conv2D(8, 1, layout="NHWC", in_channels=4))
+        net2.add(nn.BatchNorm(axis=1, in_channels=6))
+    net2.initialize(mx.init.Xavier())
+    assert fold_batchnorm(net2) == 0

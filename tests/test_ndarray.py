@@ -272,3 +272,21 @@ def test_linalg_factorizations():
     Lq, Q = mx.nd.linalg.gelqf(mx.nd.array(M[:4]))
     assert np.allclose(Lq.asnumpy() @ Q.asnumpy(), M[:4], atol=1e-3)
     assert np.allclose(Q.asnumpy() @ Q.asnumpy().T, np.eye(4), atol=1e-3)
+
+
+def test_linalg_triangular_and_diag():
+    import numpy as np
+    import anand_mxnet_amd as mx
+    rs = np.random.RandomState(9)
+    A = rs.rand(6, 6).astype("float32")
+    B = rs.rand(6, 4).astype("float32")
+    # trmm: lower-triangular multiply, left and right, with transpose
+    out = mx.nd.linalg.trmm(mx.nd.array(A), mx.nd.array(B), alpha=2.0)
+    assert np.allclose(out.asnumpy(), 2.0 * np.tril(A) @ B, atol=1e-4)
+    out = mx.nd.linalg.trmm(mx.nd.array(A), mx.nd.array(B.T),
+                            rightside=True, transpose=True)
+    assert np.allclose(out.asnumpy(), B.T @ np.tril(A).T, atol=1e-4)
+    d = mx.nd.linalg.extractdiag(mx.nd.array(A))
+    assert np.allclose(d.asnumpy(), np.diag(A))
+    m = mx.nd.linalg.makediag(d)
+    assert np.allclose(m.asnumpy(), np.diag(np.diag(A)))
