@@ -52,10 +52,10 @@ class Executor:
 
         self._input_names = set(shapes.keys()) if shapes else set()
         if args_grad is not None:
-            self.grad_dict = dict(zip(arg_names, args_grad)) \
+            self._grad_dict = dict(zip(arg_names, args_grad)) \
                 if isinstance(args_grad, (list, tuple)) else dict(args_grad)
         else:
-            self.grad_dict = {}
+            self._grad_dict = None  # lazily exposes the attached .grad
         if grad_req != "null":
             for name, arr in self.arg_dict.items():
                 arr.attach_grad(grad_req if isinstance(grad_req, str)
@@ -65,6 +65,19 @@ class Executor:
     @property
     def arg_arrays(self):
         return [self.arg_dict[n] for n in self._symbol.list_arguments()]
+
+    @property
+    def grad_dict(self):
+        """name -> gradient array (reference executor.py grad_dict).
+        When no explicit args_grad was bound, the autograd-attached
+        .grad arrays are exposed (populated after backward())."""
+        if self._grad_dict is not None:
+            return self._grad_dict
+        if self._grad_req == "null":
+            return {}
+        return {n: self.arg_dict[n].grad
+                for n in self._symbol.list_arguments()
+                if getattr(self.arg_dict[n], "grad", None) is not None}
 
     @property
     def grad_arrays(self):

@@ -221,3 +221,57 @@ def test_name_attribute_libinfo():
     assert z.attrs.get("ctx_group") == "dev1"
     d = libinfo.feature_dict()
     assert d["BF16"] and "RCCL" in d
+
+
+# ---------------------------------------------------------------------------
+# callback.Speedometer / executor / util (previously untested modules)
+# ---------------------------------------------------------------------------
+
+def test_speedometer_and_callbacks(caplog):
+    import logging
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.callback import Speedometer, BatchEndParam, \
+        log_train_metric
+    from anand_mxnet_amd import metric as met
+    m = met.Accuracy()
+    m.update(mx.nd.array([0, 1]), mx.nd.array([[0.9, 0.1], [0.2, 0.8]]))
+    sp = Speedometer(batch_size=4, frequent=1, auto_reset=False)
+    with caplog.at_level(logging.INFO):
+        for nb in range(3):
+            sp(BatchEndParam(epoch=0, nbatch=nb, eval_metric=m,
+                             locals=None))
+        log_train_metric(1)(BatchEndParam(epoch=0, nbatch=1,
+                                          eval_metric=m, locals=None))
+    msgs = [r.message for r in caplog.records]
+    assert any("Speed" in s and "samples/sec" in s for s in msgs)
+    assert any("accuracy" in s for s in msgs)
+
+
+def test_executor_simple_bind_forward_backward():
+    import numpy as np
+    import anand_mxnet_amd as mx
+    x = mx.sym.Variable("data")
+    w = mx.sym.Variable("w")
+    y = mx.sym.FullyConnected(x, weight=w, no_bias=True, num_hidden=3,
+                              name="fc")
+    ex = y.simple_bind(mx.cpu(), data=(2, 4), grad_req="write")
+    ex.arg_dict["data"][:] = mx.nd.array(
+        np.arange(8).reshape(2, 4).astype("float32"))
+    ex.arg_dict["w"][:] = mx.nd.ones((3, 4))
+    out = ex.forward(is_train=True)[0]
+    ref = np.arange(8).reshape(2, 4).astype("float32") @ np.ones((4, 3))
+    assert np.allclose(out.asnumpy(), ref)
+    ex.backward(mx.nd.ones((2, 3)))
+    gw = ex.grad_dict["w"].asnumpy()
+    assert np.allclose(gw, np.tile(np.arange(8).reshape(2, 4)
+                                   .astype("float32").sum(0), (3, 1)))
+
+
+def test_util_helpers(tmp_path):
+    from anand_mxnet_amd import util
+    d = str(tmp_path / "a" / "b")
+    util.makedirs(d)
+    util.makedirs(d)  # idempotent
+    import os
+    assert os.path.isdir(d)
+    assert util.get_gpu_count() >= 0
