@@ -154,3 +154,18 @@ def test_gradient_compression_2bit():
     out = nd.zeros((4,))
     kv_store.pull("w", out=out)
     np.testing.assert_allclose(out.asnumpy(), [0.5, -0.5, 0.0, 0.0])
+
+
+def test_kvstore_server_role_contract(monkeypatch):
+    """kvstore_server: the dist server/scheduler role is a compatibility
+    shell in the all-reduce architecture — it identifies the role from
+    DMLC_ROLE and parks (reference python/mxnet/kvstore_server.py)."""
+    from anand_mxnet_amd import kvstore_server as ks
+    monkeypatch.setenv("DMLC_ROLE", "worker")
+    assert ks._init_server_module() is False
+    monkeypatch.setenv("DMLC_ROLE", "server")
+    assert ks._init_server_module() is True
+    monkeypatch.setenv("DMLC_ROLE", "scheduler")
+    assert ks._init_server_module() is True
+    srv = ks.KVStoreServer()
+    assert hasattr(srv, "run")
