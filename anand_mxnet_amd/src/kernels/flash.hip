@@ -40,6 +40,16 @@ DEVINL int fa_v_byte(int row, int cbyte) {
   return row * 128 + ((((cbyte >> 4) ^ ((row & 3) << 1)) << 4) | (cbyte & 15));
 }
 
+// raw barrier preceded by lgkmcnt(0): publishes every wave's LDS writes
+// without the vmcnt(0) drain __syncthreads folds in while a glds is in
+// flight — that drain serialized the kv prefetch against each softmax
+// reduction barrier (PMC: 45% wave-park on flash_fwd_core)
+#define FA_LDS_BARRIER()                                   \
+  do {                                                     \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");     \
+    __builtin_amdgcn_s_barrier();                          \
+  } while (0)
+
 DEVINL int fa_tr128_byte(int row, int cbyte) {
   // [128][128 bf16] row-major tr image (256-B rows), same chunk key
   return row * 256 + ((((cbyte >> 4) ^ ((row & 3) << 1)) << 4) | (cbyte & 15));
@@ -218,7 +228,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core_d128(
           sred[0][wmq + mf * 16 + (ll >> 4) * 4 + i][lw >> 2] =
               cmax[mf][i];
     }
-    __syncthreads();
+    FA_LDS_BARRIER();
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
@@ -270,7 +280,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core_d128(
           sred[1][wmq + mf * 16 + (ll >> 4) * 4 + i][lw >> 2] =
               csum[mf][i];
     }
-    __syncthreads();
+    FA_LDS_BARRIER();
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
@@ -487,12 +497,14 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
       for (int nf = 0; nf < 4; ++nf)
         bfr[nf] = *(const bf16x8*)&kimg[buf][fa_qk_off(
             wnt + nf * 16 + lrow, chunk)];
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
         for (int nf = 0; nf < 4; ++nf)
           acc_s[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mf], bfr[nf], acc_s[mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
 
     // ---- scale + causal mask + chunk row-max ----
@@ -535,7 +547,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
           sred[0][wmq + mf * 16 + (ll >> 4) * 4 + i][lw >> 2] =
               cmax[mf][i];
     }
-    __syncthreads();
+    FA_LDS_BARRIER();
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
@@ -587,7 +599,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
           sred[1][wmq + mf * 16 + (ll >> 4) * 4 + i][lw >> 2] =
               csum[mf][i];
     }
-    __syncthreads();
+    FA_LDS_BARRIER();
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
@@ -616,7 +628,7 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
                    (col & 7) * 2) = f2bf(p);
         }
       }
-    __syncthreads();
+    FA_LDS_BARRIER();
 
     // ---- acc_o += P~ V (wave: 32 q x 32 d), K dim = 128 tk ----
 #pragma unroll
@@ -642,14 +654,17 @@ __global__ __launch_bounds__(512, 1) void flash_fwd_core(
         bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(pv2);
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
         for (int nf = 0; nf < 2; ++nf)
           acc_o[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mf], bfr[nf], acc_o[mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
-    __syncthreads();  // pimg reused; kv buffers swap
+    __syncthreads();  // pimg reused; kv buffers swap — and the chunk's
+                      // prefetch (issued a full iteration ago) drains
   }
 
   // ---- epilogue: O = acc_o / l, write strided [B, Tq, H*D]; stats ----

@@ -29,6 +29,16 @@ elif which == "tn":
     b = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
     for _ in range(iters):
         _C.gemm_tn(a, b, None, False)
+elif which == "flash":
+    B, H, T, D = 32, 12, 512, 64
+    q = (torch.rand(B, T, H*D, device=DEV)*2-1).to(BF)
+    k = (torch.rand(B, T, H*D, device=DEV)*2-1).to(BF)
+    v = (torch.rand(B, T, H*D, device=DEV)*2-1).to(BF)
+    sc = 1.0/(D**0.5)
+    for _ in range(iters):
+        out, m, l = _C.flash_fwd(q, k, v, H, sc, False, 0.0, 7)
+        dq, dk, dv = _C.flash_bwd(q, k, v, out, q, m, l, H, sc, False,
+                                  0.0, 7)
 elif which == "nn":
     a = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
     b = (torch.rand((8192, 8192), device=DEV) * 2 - 1).to(BF)
