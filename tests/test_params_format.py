@@ -132,3 +132,32 @@ def test_large_name_and_many_arrays(tmp_path):
     save_ndarrays(path, arrays)
     back = load_ndarrays(path)
     assert len(back) == 64
+
+
+def test_legacy_v0_read(tmp_path):
+    """Legacy V0 records (magic == ndim, uint32 TShape dims) still load
+    (reference ndarray.cc NDArray::Load's oldest branch)."""
+    import struct
+    import numpy as np
+    from anand_mxnet_amd.serializer import load_ndarrays
+
+    data = np.arange(6, dtype=np.float32).reshape(2, 3)
+    buf = b""
+    buf += struct.pack("<Q", 0x112)      # list magic
+    buf += struct.pack("<Q", 0)          # reserved
+    buf += struct.pack("<Q", 1)          # n arrays
+    # V0 ndarray: ndim(=magic) | uint32 dims | dev_type | dev_id | dtype
+    buf += struct.pack("<I", 2)          # ndim as magic
+    buf += struct.pack("<II", 2, 3)      # uint32 dims
+    buf += struct.pack("<ii", 1, 0)      # dev_type, dev_id
+    buf += struct.pack("<i", 0)          # float32
+    buf += data.tobytes()
+    buf += struct.pack("<Q", 1)          # n names
+    name = b"arg:w"
+    buf += struct.pack("<Q", len(name)) + name
+    p = tmp_path / "v0.params"
+    p.write_bytes(buf)
+    out = load_ndarrays(str(p))
+    assert "arg:w" in out or "w" in out
+    arr = out.get("arg:w", out.get("w"))
+    assert np.allclose(arr.asnumpy(), data)
