@@ -1266,3 +1266,28 @@ def test_mha_d128_op():
                  rtol=7e-2, atol=7e-2, msg="mha d128 dk")
     assert_close(v.grad, vf.grad.transpose(1, 2).reshape(B, T, H * D),
                  rtol=7e-2, atol=7e-2, msg="mha d128 dv")
+
+
+def test_window_gemm_routes_vs_oracle():
+    """Shapes large enough to take the 2-barrier-window kernels
+    (gemm_tn8g_core: >=240 256^2-tiles & K>=1024; gemm_nn8g_core:
+    >=160 tiles & K>=1024) match the fp32 oracle."""
+    torch.manual_seed(11)
+    # TN route: 4096x4096, K=1024 -> 256 tiles
+    a = _rt((4096, 1024)).to(BF)
+    b = _rt((4096, 1024)).to(BF)
+    y = _C.gemm_tn(a.contiguous(), b.contiguous(), None, False)
+    ref = a.float() @ b.float().t()
+    assert_close(y, ref, rtol=3e-2, atol=3e-2, msg="tn8g route")
+    # NN route: 4096x2560, K=1024 -> 160 tiles
+    a = _rt((4096, 1024)).to(BF)
+    b = _rt((1024, 2560)).to(BF)
+    y = _C.gemm_nn(a.contiguous(), b.contiguous())
+    ref = a.float() @ b.float()
+    assert_close(y, ref, rtol=3e-2, atol=3e-2, msg="nn8g route")
+    # tail rows on both routes (M not % 256)
+    a = _rt((4000, 1024)).to(BF)
+    b = _rt((4096, 1024)).to(BF)
+    y = _C.gemm_tn(a.contiguous(), b.contiguous(), None, False)
+    assert_close(y, a.float() @ b.float().t(), rtol=3e-2, atol=3e-2,
+                 msg="tn8g tail")
