@@ -24,6 +24,14 @@ def _next_seed():
     return _seed_counter[0]
 
 
+def reset_seed(n):
+    """Reseed the native RNG salt stream (dropout masks, flash dropout
+    replay, sampling) — called by mx.random.seed so runs are
+    reproducible under seeding (reference: mx.random.seed seeds the
+    engine's kRandom/kParallelRandom resources)."""
+    _seed_counter[0] = int(n) % (1 << 63)
+
+
 # ---------------------------------------------------------------------------
 # GEMM / linear
 # ---------------------------------------------------------------------------

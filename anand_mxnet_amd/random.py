@@ -11,3 +11,8 @@ def seed(seed_state, ctx="all"):
     np.random.seed(seed_state % (2 ** 31))
     if torch.cuda.is_available():
         torch.cuda.manual_seed_all(seed_state)
+    try:  # native salt stream (dropout / flash replay / sampling)
+        from .ops import hip as _hip
+        _hip.reset_seed(seed_state)
+    except ImportError:  # pragma: no cover
+        pass

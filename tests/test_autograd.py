@@ -138,3 +138,18 @@ def test_numeric_gradient_layernorm_gelu():
     y = mx.test_utils.rand_ndarray((3, 5))
     check_numeric_gradient(lambda a: nd.gelu(a), [y], rtol=2e-2,
                            atol=1e-3)
+
+
+def test_random_seed_controls_native_salt():
+    """mx.random.seed resets the native dropout/sampling salt stream so
+    masks reproduce (reference: seed reaches the kRandom resources)."""
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.ops import hip as hipops
+    mx.random.seed(42)
+    a = [hipops._next_seed() for _ in range(3)]
+    mx.random.seed(42)
+    b = [hipops._next_seed() for _ in range(3)]
+    assert a == b
+    mx.random.seed(7)
+    c = [hipops._next_seed() for _ in range(3)]
+    assert c != a
