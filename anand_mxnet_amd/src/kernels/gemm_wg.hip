@@ -314,6 +314,35 @@ struct TrPlainRows {
     return (colok && mrem[p] > 0) ? ptr[p]
                                   : (const bf16*)((const char*)zpage + cgbyte);
   }
+
+  // 4-slot variant for the 256-wide w64 A image: per-slot column chunk
+  // (the tr512 swizzle key varies by row) and per-slot colok
+  const bf16* ptr4[4];
+  long mrem4[4];
+  bool colok4[4];
+  int cgbyte4[4];
+  DEVINL void init4(int col0, long m0, const int* rows, const int* cg) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      cgbyte4[p] = (cg[p] & 15) * 16;
+      int col = col0 + cg[p] * 8;
+      colok4[p] = col + 8 <= rowlen;
+      ptr4[p] = base + (m0 + rows[p]) * ld + col;
+      mrem4[p] = mtotal - (m0 + rows[p]);
+    }
+  }
+  DEVINL const bf16* src4(int p) const {
+    return (colok4[p] && mrem4[p] > 0)
+               ? ptr4[p]
+               : (const bf16*)((const char*)zpage + cgbyte4[p]);
+  }
+  DEVINL void step4() {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      ptr4[p] += (long)WBK * ld;
+      mrem4[p] -= WBK;
+    }
+  }
 };
 
 struct TrIm2colRows {
@@ -537,6 +566,191 @@ __global__ __launch_bounds__(WT8, 2) void gemm_wg_tr_core(
     // per column per block (splits sum via the atomic)
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf) {
+      float v = bsum[mf];
+      v += __shfl_xor(v, 16, 64);
+      v += __shfl_xor(v, 32, 64);
+      int col = bm + wm + mf * 16 + (ll & 15);
+      if ((ll >> 4) == 0 && col < Mo) atomicAdd(&biasg[col], v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 64x64-wave-patch tr-variant (linear wgrad only, MXA_WG64): fragment
+// reads scale with mf+nf while MFMAs scale with mf*nf, so widening the
+// wave patch from 32x64 (16 MFMA : 12 tr-pairs per K-step) to 64x64
+// (32 MFMA : 16 tr-pairs) raises MFMA per LDS instruction by 50% — the
+// tr16 ISSUE rate is what bounds the 2-operand out-contiguous wgrad
+// (profiles/RESULTS.md r3 lead #2).  Tile 256(Mo) x 128(No), 8 waves as
+// 4x2; A image [64 m][256 Mo] with a 4-bit tr-swizzle key so the
+// 16-row read sets spread across banks; B keeps the [64][128] format.
+// ---------------------------------------------------------------------------
+
+DEVINL int tr512_byte(int row, int cbyte) {
+  int key = ((row & 3) << 2) ^ ((row >> 3) & 3);
+  return row * 512 + ((((cbyte >> 4) ^ key) << 4) | (cbyte & 15));
+}
+
+template <bool OUTBF16>
+__global__ __launch_bounds__(WT8, 1) void gemm_wg_w64_core(
+    TrPlainRows aload, TrPlainRows bload, float* __restrict__ Cout,
+    int Mo, int No, long ldc, int ksteps_total, int nsplit,
+    long strideSlot, float* __restrict__ biasg) {
+  __shared__ __align__(16) bf16 smemA[2][64 * 256];
+  __shared__ __align__(16) bf16 smemB[2][64 * 128];
+
+  const int ntn = (No + 127) / 128;
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm = (bid / ntn) * 256, bn = (bid % ntn) * 128;
+  const int t = threadIdx.x;
+  const int lw = t / WAVE, ll = t % WAVE;
+
+  const int split = blockIdx.y;
+  const int per = (ksteps_total + nsplit - 1) / nsplit;
+  const int kstep0 = split * per;
+  const int ksteps = min(per, ksteps_total - kstep0);
+  if (ksteps <= 0) return;
+
+  // A staging: 4 glds passes of 16 rows x 512 B; pass p covers rows
+  // p*16 + (t>>5), 16-B chunk (t&31), source chunk XOR'd with the
+  // 4-bit key so the LDS image matches tr512_byte
+  int arow[4], acg[4];
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    arow[p] = p * 16 + (t >> 5);
+    int key = ((arow[p] & 3) << 2) ^ ((arow[p] >> 3) & 3);
+    acg[p] = (t & 31) ^ key;
+  }
+  // B staging: 2 passes of the existing [64][128] geometry
+  int brow[2];
+#pragma unroll
+  for (int p = 0; p < 2; ++p) brow[p] = lw * 8 + p * 4 + (ll >> 4);
+  const int bcg = (ll & 15) ^ ((ll >> 4) << 1);
+
+  long m0 = (long)kstep0 * WBK;
+  aload.init4(bm, m0, arow, acg);
+  bload.init(bn, m0, brow, bcg);
+
+  auto stage_glds = [&](int buf) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      unsigned lbase = __builtin_amdgcn_readfirstlane(
+          (unsigned)((p * 16 + (lw * 2)) * 512));
+      auto ldsA = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemA[buf][0] + lbase);
+      auto gA =
+          (const __attribute__((address_space(1))) unsigned int*)aload.src4(
+              p);
+      __builtin_amdgcn_global_load_lds(gA, ldsA, 16, 0, 0);
+    }
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      unsigned lbase =
+          __builtin_amdgcn_readfirstlane((unsigned)((lw * 2 + p) * 1024));
+      auto ldsB = (__attribute__((address_space(3))) unsigned int*)(
+          (__attribute__((address_space(3))) char*)&smemB[buf][0] + lbase);
+      auto gB =
+          (const __attribute__((address_space(1))) unsigned int*)bload.src(p);
+      __builtin_amdgcn_global_load_lds(gB, ldsB, 16, 0, 0);
+    }
+    aload.step4();
+    bload.step();
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int wm = (lw & 3) * 64, wn = (lw >> 2) * 64;
+  const int ig = ll & 15, khi = ll >> 4;
+
+  // wn==0 wave group only: both N-halves' waves read every mo column
+  const bool do_bias = biasg != nullptr && bn == 0 && (lw >> 2) == 0;
+  float bsum[4] = {0.f, 0.f, 0.f, 0.f};
+
+  auto trreadA = [&](const bf16* img, int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)img +
+        tr512_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+  auto trreadB = [&](const bf16* img, int row, int cbyte) {
+    auto p = (__attribute__((address_space(3))) bf16x4*)(
+        (__attribute__((address_space(3))) char*)img + tr_byte(row, cbyte));
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  };
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int rbase = kk * 32 + khi * 8 + (ig >> 2);
+      int cb = (ig & 3) * 8;
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        int cbyte = (wm + mf * 16) * 2 + cb;
+        bf16x4 lo = trreadA(smemA[buf], rbase, cbyte);
+        bf16x4 hi = trreadA(smemA[buf], rbase + 4, cbyte);
+        af[mf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int cbyte = (wn + nf * 16) * 2 + cb;
+        bf16x4 lo = trreadB(smemB[buf], rbase, cbyte);
+        bf16x4 hi = trreadB(smemB[buf], rbase + 4, cbyte);
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      if (do_bias) {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) bsum[mf] += bf2f(af[mf][j]);
+      }
+    }
+  };
+
+  stage_glds(0);
+  __syncthreads();
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    if (ks + 1 < ksteps) stage_glds(buf ^ 1);
+    compute(buf);
+    __syncthreads();
+  }
+
+  float* cseg = Cout + (long)split * strideSlot;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int gcol = bn + wn + nf * 16 + (ll & 15);
+      if (gcol >= No) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int grow = bm + wm + mf * 16 + (ll >> 4) * 4 + i;
+        if (grow >= Mo) continue;
+        float v = acc[mf][nf][i];
+        if (OUTBF16)
+          ((bf16*)cseg)[(long)grow * ldc + gcol] = f2bf(v);
+        else
+          cseg[(long)grow * ldc + gcol] = v;
+      }
+    }
+  }
+
+  if (do_bias) {
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
       float v = bsum[mf];
       v += __shfl_xor(v, 16, 64);
       v += __shfl_xor(v, 32, 64);
@@ -1319,11 +1533,27 @@ static WgPlanT wg_plan_(long bxy, long mtotal) {
   return p;
 }
 
+static bool wg_w64() {
+  static const bool v = [] {
+    const char* e = getenv("MXA_WG64");
+    return e && e[0] == '1';  // off until A/B'd
+  }();
+  return v;
+}
+
 int gemm_wg_nslots(long mtotal, int Mo, int No, bool transpose) {
   // 0 = direct bf16 output, no workspace (single split, plain layout,
   // kernels that have the OUTBF16 epilogue)
   WgPlanT p = wg_plan_((long)cdiv_h(Mo, WBM) * cdiv_h(No, WBN), mtotal);
   bool tr_path = !wg_use_zip() && Mo % 8 == 0 && No % 8 == 0;
+  if (wg_w64() && Mo >= 256 && tr_path) {
+    // the 256x128 w64 plan may pick a different split count; size the
+    // workspace for whichever path will actually run
+    WgPlanT p64 = wg_plan_((long)cdiv_h(Mo, 256) * cdiv_h(No, 128),
+                           mtotal);
+    if (p64.used == 1 && !transpose) return 0;
+    return wg_alloc_slots(p64.used);
+  }
   if (p.used == 1 && !transpose && tr_path) return 0;
   return wg_alloc_slots(p.used);
 }
@@ -1358,6 +1588,25 @@ void gemm_wg_bf16(const bf16* A, const bf16* B, float* work, bf16* out,
             a, dummy, b, Cdst, Mo, No, No, 0, 1, 1, 1, pl.hsteps, pl.nsplit,
             0, ZMap{1, 0, 0}, ZMap{1, 0, 0}, ZMap{1, 0, 0}, slotStride);
     } else {
+      if (wg_w64() && Mo >= 256) {
+        long bxy64 = (long)cdiv_h(Mo, 256) * cdiv_h(No, 128);
+        WgPlanT p64 = wg_plan_(bxy64, mtotal);
+        dim3 g64((unsigned)bxy64, p64.nsplit, 1);
+        TrPlainRows a{A, zpage, mtotal, lda, Mo};
+        TrPlainRows b{B, zpage, mtotal, ldb, No};
+        if (direct)
+          gemm_wg_w64_core<true><<<g64, WT8, 0, stream>>>(
+              a, b, Cdst, Mo, No, No, p64.ksteps, p64.nsplit, 0, biasg);
+        else
+          gemm_wg_w64_core<false><<<g64, WT8, 0, stream>>>(
+              a, b, Cdst, Mo, No, No, p64.ksteps, p64.nsplit, slotStride,
+              biasg);
+        if (!direct)
+          wg_fold(work, slotStride, p64.used, Mo, No, out, transpose,
+                  stream);
+        if (biasg) { /* fused above */ }
+        return;
+      }
       TrPlainRows a{A, zpage, mtotal, lda, Mo};
       TrPlainRows b{B, zpage, mtotal, ldb, No};
       TrIm2colRows dummy{};
