@@ -311,8 +311,11 @@ class HybridBlock(Block):
         s.save(f"{path}-symbol.json")
         params = self._collect_params_with_prefix()
         from ..serializer import save_ndarrays
-        arg_dict = {"arg:" + k.replace(".", "_"): v._reduce()
-                    for k, v in params.items()}
+        # key by Parameter.name — the traced symbol's variable nodes use
+        # parameter names, and Module/SymbolBlock loaders bind by them
+        # (the dotted child-path keys of save_parameters are a different
+        # contract); reference export writes arg:/aux: the same way
+        arg_dict = {"arg:" + v.name: v._reduce() for v in params.values()}
         save_ndarrays(f"{path}-{epoch:04d}.params", arg_dict)
         return f"{path}-symbol.json", f"{path}-{epoch:04d}.params"
 

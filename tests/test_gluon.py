@@ -365,3 +365,62 @@ def test_fold_batchnorm_edge_cases():
         net2.add(nn.BatchNorm(axis=1, in_channels=6))
     net2.initialize(mx.init.Xavier())
     assert fold_batchnorm(net2) == 0
+
+
+def test_hybridize_cached_execution():
+    """net.hybridize(): outputs identical before/after, cache reused
+    across calls, re-specializes on a new input shape, and backward
+    still flows (reference HybridBlock._build_cache -> CachedOp)."""
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import autograd as ag
+    from anand_mxnet_amd.gluon import nn
+
+    net = nn.HybridSequential()
+    with net.name_scope():
+        net.add(nn.Dense(8, in_units=6, activation="relu"))
+        net.add(nn.Dense(3, in_units=8))
+    net.initialize(mx.init.Xavier())
+    x = mx.nd.array(np.random.RandomState(0).rand(4, 6).astype("float32"))
+    ref = net(x).asnumpy()
+    net.hybridize(static_alloc=True)
+    out1 = net(x).asnumpy()
+    out2 = net(x).asnumpy()  # cached call
+    assert np.allclose(out1, ref, atol=1e-5)
+    assert np.allclose(out2, ref, atol=1e-5)
+    # new batch size: re-specialization
+    x2 = mx.nd.array(np.random.RandomState(1).rand(7, 6).astype("float32"))
+    assert net(x2).shape == (7, 3)
+    # backward through the hybridized net
+    with ag.record():
+        loss = (net(x) ** 2).mean()
+    loss.backward()
+    g = net[0].weight.data().grad
+    assert g is not None
+
+
+def test_export_and_symbolblock_imports(tmp_path):
+    """HybridBlock.export() -> -symbol.json + -0000.params ->
+    SymbolBlock.imports reproduces the outputs (reference block.py
+    export / SymbolBlock.imports deployment path)."""
+    import os
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.gluon import nn, SymbolBlock
+
+    net = nn.HybridSequential()
+    with net.name_scope():
+        net.add(nn.Dense(8, in_units=5, activation="relu"))
+        net.add(nn.Dense(2, in_units=8))
+    net.initialize(mx.init.Xavier())
+    x = mx.nd.array(np.random.RandomState(3).rand(3, 5).astype("float32"))
+    net.hybridize()
+    ref = net(x).asnumpy()
+    prefix = str(tmp_path / "m")
+    net.export(prefix)
+    assert os.path.exists(prefix + "-symbol.json")
+    assert os.path.exists(prefix + "-0000.params")
+    sb = SymbolBlock.imports(prefix + "-symbol.json", ["data"],
+                             prefix + "-0000.params")
+    out = sb(x).asnumpy()
+    assert np.allclose(out, ref, atol=1e-5), np.abs(out - ref).max()
