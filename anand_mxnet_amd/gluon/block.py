@@ -314,8 +314,13 @@ class HybridBlock(Block):
         # key by Parameter.name — the traced symbol's variable nodes use
         # parameter names, and Module/SymbolBlock loaders bind by them
         # (the dotted child-path keys of save_parameters are a different
-        # contract); reference export writes arg:/aux: the same way
-        arg_dict = {"arg:" + v.name: v._reduce() for v in params.values()}
+        # contract).  Auxiliary states (grad_req null: BN running stats)
+        # get the reference's aux: prefix so Module.load_checkpoint
+        # classifies them correctly.
+        arg_dict = {}
+        for v in params.values():
+            pfx = "aux:" if v.grad_req == "null" else "arg:"
+            arg_dict[pfx + v.name] = v._reduce()
         save_ndarrays(f"{path}-{epoch:04d}.params", arg_dict)
         return f"{path}-symbol.json", f"{path}-{epoch:04d}.params"
 

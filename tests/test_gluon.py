@@ -424,3 +424,32 @@ def test_export_and_symbolblock_imports(tmp_path):
                              prefix + "-0000.params")
     out = sb(x).asnumpy()
     assert np.allclose(out, ref, atol=1e-5), np.abs(out - ref).max()
+
+
+def test_export_imports_conv_bn_net(tmp_path):
+    """export/imports roundtrip for a conv+BN+pool net (the trace_block
+    layer set beyond Dense)."""
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.gluon import nn, SymbolBlock
+
+    net = nn.HybridSequential()
+    with net.name_scope():
+        net.add(nn.Conv2D(8, 3, padding=1, in_channels=3))
+        net.add(nn.BatchNorm(axis=1, in_channels=8))
+        net.add(nn.Activation("relu"))
+        net.add(nn.MaxPool2D(pool_size=2, strides=2))
+        net.add(nn.Flatten())
+        net.add(nn.Dense(4, in_units=8 * 8 * 8))
+    net.initialize(mx.init.Xavier())
+    x = mx.nd.array(np.random.RandomState(5)
+                    .rand(2, 3, 16, 16).astype("float32"))
+    net(x)  # materialize + set BN running stats shape
+    ref = net(x).asnumpy()
+    prefix = str(tmp_path / "cnn")
+    net.export(prefix)
+    sb = SymbolBlock.imports(prefix + "-symbol.json", ["data"],
+                             prefix + "-0000.params")
+    out = sb(x).asnumpy()
+    assert np.allclose(out, ref, rtol=1e-4, atol=1e-4), \
+        np.abs(out - ref).max()
