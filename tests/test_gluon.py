@@ -453,3 +453,44 @@ def test_export_imports_conv_bn_net(tmp_path):
     out = sb(x).asnumpy()
     assert np.allclose(out, ref, rtol=1e-4, atol=1e-4), \
         np.abs(out - ref).max()
+
+
+def test_load_parameters_flags(tmp_path):
+    """allow_missing / ignore_extra / cast_dtype contracts of
+    load_parameters (reference block.py)."""
+    import numpy as np
+    import pytest
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd.base import MXNetError
+    from anand_mxnet_amd.gluon import nn
+
+    net = nn.HybridSequential()
+    with net.name_scope():
+        net.add(nn.Dense(4, in_units=3))
+        net.add(nn.Dense(2, in_units=4))
+    net.initialize(mx.init.Constant(0.5))
+    f = str(tmp_path / "p.params")
+    net.save_parameters(f)
+
+    # ignore_extra: a smaller net loads from the full file
+    small = nn.HybridSequential()
+    with small.name_scope():
+        small.add(nn.Dense(4, in_units=3))
+    small.initialize(mx.init.Zero())
+    with pytest.raises(MXNetError):
+        small.load_parameters(f)  # extra keys present
+    small.load_parameters(f, ignore_extra=True)
+    assert float(small[0].weight.data().asnumpy().mean()) == 0.5
+
+    # allow_missing: a bigger net loads what exists
+    big = nn.HybridSequential()
+    with big.name_scope():
+        big.add(nn.Dense(4, in_units=3))
+        big.add(nn.Dense(2, in_units=4))
+        big.add(nn.Dense(1, in_units=2))
+    big.initialize(mx.init.Zero())
+    with pytest.raises(MXNetError):
+        big.load_parameters(f, ignore_extra=True)  # third layer missing
+    big.load_parameters(f, ignore_extra=True, allow_missing=True)
+    assert float(big[0].weight.data().asnumpy().mean()) == 0.5
+    assert float(big[2].weight.data().asnumpy().mean()) == 0.0
