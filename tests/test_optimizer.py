@@ -236,3 +236,30 @@ def test_trainer_save_load_states(tmp_path):
     wA = netA.weight.data().asnumpy()
     wC = netC.weight.data().asnumpy()
     assert np.allclose(wA, wC, atol=1e-6), np.abs(wA - wC).max()
+
+
+def test_trainer_lr_scheduler_progression():
+    """Trainer honors lr_scheduler across steps (reference
+    lr_scheduler.py FactorScheduler) and learning_rate reflects it."""
+    import numpy as np
+    import anand_mxnet_amd as mx
+    from anand_mxnet_amd import gluon, lr_scheduler
+    from anand_mxnet_amd.gluon import nn
+    from anand_mxnet_amd import autograd as ag
+
+    sched = lr_scheduler.FactorScheduler(step=2, factor=0.5,
+                                         base_lr=0.4)
+    net = nn.Dense(2, in_units=3)
+    net.initialize(mx.init.Constant(0.1))
+    tr = gluon.Trainer(net.collect_params(), "sgd",
+                       {"learning_rate": 0.4, "lr_scheduler": sched})
+    x = mx.nd.ones((2, 3))
+    lrs = []
+    for _ in range(5):
+        with ag.record():
+            loss = (net(x) ** 2).mean()
+        loss.backward()
+        tr.step(1)
+        lrs.append(tr.learning_rate)
+    # factor 0.5 every 2 updates: 0.4, 0.4, 0.2, 0.2, 0.1
+    assert np.allclose(lrs, [0.4, 0.4, 0.2, 0.2, 0.1]), lrs
