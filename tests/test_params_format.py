@@ -161,3 +161,32 @@ def test_legacy_v0_read(tmp_path):
     assert "arg:w" in out or "w" in out
     arr = out.get("arg:w", out.get("w"))
     assert np.allclose(arr.asnumpy(), data)
+
+
+def test_serializer_hypothesis_fuzz(tmp_path):
+    """Property fuzz: save/load round-trips arbitrary shapes/dtypes
+    bit-exactly (hypothesis, bounded examples for suite speed)."""
+    from hypothesis import given, settings, strategies as st
+    import numpy as np
+    import torch
+    from anand_mxnet_amd.serializer import save_ndarrays, load_ndarrays
+    from anand_mxnet_amd.ndarray.ndarray import NDArray
+
+    dtypes = [np.float32, np.float64, np.float16, np.uint8, np.int32,
+              np.int8, np.int64]
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.lists(st.integers(1, 7), min_size=1, max_size=4),
+           st.integers(0, len(dtypes) - 1),
+           st.integers(0, 2 ** 31 - 1))
+    def roundtrip(shape, di, seed):
+        rs = np.random.RandomState(seed)
+        a = (rs.rand(*shape) * 100 - 50).astype(dtypes[di])
+        f = str(tmp_path / "h.params")
+        save_ndarrays(f, {"x": NDArray(torch.from_numpy(a.copy()))})
+        out = load_ndarrays(f)["x"].asnumpy()
+        assert out.dtype == a.dtype
+        assert out.shape == a.shape
+        assert np.array_equal(out, a)
+
+    roundtrip()
