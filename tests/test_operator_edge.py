@@ -344,3 +344,36 @@ def test_quantized_conv_pointwise():
     ref = torch.nn.functional.conv2d(
         x.permute(0, 3, 1, 2), w.permute(0, 3, 1, 2)).permute(0, 2, 3, 1)
     assert np.abs(f - ref.numpy()).max() < 0.15
+
+
+def test_broadcast_ops_hypothesis_fuzz():
+    """Property fuzz: nd broadcast add/mul/max against numpy across
+    random broadcastable shape pairs (reference
+    elemwise_binary_broadcast_op semantics)."""
+    from hypothesis import given, settings, strategies as st
+    import numpy as np
+    import anand_mxnet_amd as mx
+
+    @st.composite
+    def bcast_pair(draw):
+        ndim = draw(st.integers(1, 4))
+        base = [draw(st.integers(1, 5)) for _ in range(ndim)]
+        a = [d if draw(st.booleans()) else 1 for d in base]
+        b = [d if (a[i] == 1 or draw(st.booleans())) else 1
+             for i, d in enumerate(base)]
+        return tuple(a), tuple(b)
+
+    @settings(max_examples=30, deadline=None)
+    @given(bcast_pair(), st.integers(0, 2 ** 31 - 1))
+    def check(shapes, seed):
+        sa, sb = shapes
+        rs = np.random.RandomState(seed)
+        a = (rs.rand(*sa) * 4 - 2).astype("float32")
+        b = (rs.rand(*sb) * 4 - 2).astype("float32")
+        na, nb = mx.nd.array(a), mx.nd.array(b)
+        assert np.allclose((na + nb).asnumpy(), a + b, atol=1e-6)
+        assert np.allclose((na * nb).asnumpy(), a * b, atol=1e-6)
+        assert np.allclose(mx.nd.broadcast_maximum(na, nb).asnumpy(),
+                           np.maximum(a, b), atol=1e-6)
+
+    check()
