@@ -377,3 +377,39 @@ def test_broadcast_ops_hypothesis_fuzz():
                            np.maximum(a, b), atol=1e-6)
 
     check()
+
+
+def test_reduce_ops_hypothesis_fuzz():
+    """Property fuzz: sum/mean/max over random axis subsets (keepdims
+    both ways) vs numpy (reference broadcast_reduce_op semantics)."""
+    from hypothesis import given, settings, strategies as st
+    import numpy as np
+    import anand_mxnet_amd as mx
+
+    @st.composite
+    def case(draw):
+        ndim = draw(st.integers(1, 4))
+        shape = tuple(draw(st.integers(1, 5)) for _ in range(ndim))
+        axes = tuple(sorted(draw(st.sets(st.integers(0, ndim - 1),
+                                         max_size=ndim))))
+        keep = draw(st.booleans())
+        return shape, (axes if axes else None), keep
+
+    @settings(max_examples=30, deadline=None)
+    @given(case(), st.integers(0, 2 ** 31 - 1))
+    def check(c, seed):
+        shape, axes, keep = c
+        rs = np.random.RandomState(seed)
+        a = (rs.rand(*shape) * 4 - 2).astype("float32")
+        na = mx.nd.array(a)
+        assert np.allclose(
+            mx.nd.sum(na, axis=axes, keepdims=keep).asnumpy(),
+            a.sum(axis=axes, keepdims=keep), atol=1e-5)
+        assert np.allclose(
+            mx.nd.mean(na, axis=axes, keepdims=keep).asnumpy(),
+            a.mean(axis=axes, keepdims=keep), atol=1e-5)
+        assert np.allclose(
+            mx.nd.max(na, axis=axes, keepdims=keep).asnumpy(),
+            a.max(axis=axes, keepdims=keep), atol=1e-6)
+
+    check()
