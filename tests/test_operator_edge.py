@@ -413,3 +413,32 @@ def test_reduce_ops_hypothesis_fuzz():
             a.max(axis=axes, keepdims=keep), atol=1e-6)
 
     check()
+
+
+def test_matrix_op_hypothesis_fuzz():
+    """Property fuzz: transpose/reshape/slice compositions vs numpy
+    (reference matrix_op semantics — views must not alias-corrupt)."""
+    from hypothesis import given, settings, strategies as st
+    import numpy as np
+    import anand_mxnet_amd as mx
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.tuples(st.integers(1, 6), st.integers(1, 6),
+                     st.integers(1, 6)),
+           st.permutations([0, 1, 2]), st.integers(0, 2 ** 31 - 1))
+    def check(shape, perm, seed):
+        rs = np.random.RandomState(seed)
+        a = (rs.rand(*shape) * 4 - 2).astype("float32")
+        na = mx.nd.array(a)
+        t = mx.nd.transpose(na, axes=tuple(perm))
+        assert np.allclose(t.asnumpy(), a.transpose(perm))
+        r = mx.nd.reshape(na, shape=(-1,))
+        assert np.allclose(r.asnumpy(), a.reshape(-1))
+        s0 = min(1, shape[0] - 1)
+        sl = na[s0:]
+        assert np.allclose(sl.asnumpy(), a[s0:])
+        # slicing then mutating the slice must write through (view)
+        sl[:] = 0
+        assert float(na.asnumpy()[s0:].sum()) == 0.0
+
+    check()
